@@ -1,0 +1,142 @@
+"""Sparse-core tests (mirrors reference test strategy, SURVEY.md §4:
+torchrec/sparse/tests/test_keyed_jagged_tensor.py)."""
+
+import torch
+
+from torchrec_amd.sparse.jagged_tensor import JaggedTensor, KeyedJaggedTensor, KeyedTensor
+
+
+def make_kjt():
+    # f1: [ [1,2], [] , [3] ], f2: [ [4], [5,6], [] ]  B=3
+    return KeyedJaggedTensor(
+        keys=["f1", "f2"],
+        values=torch.tensor([1, 2, 3, 4, 5, 6]),
+        lengths=torch.tensor([2, 0, 1, 1, 2, 0]),
+        stride=3,
+    )
+
+
+class TestJaggedTensor:
+    def test_lengths_offsets(self):
+        jt = JaggedTensor(values=torch.arange(6), lengths=torch.tensor([2, 0, 4]))
+        assert jt.offsets().tolist() == [0, 2, 2, 6]
+        jt2 = JaggedTensor(values=torch.arange(6), offsets=torch.tensor([0, 2, 2, 6]))
+        assert jt2.lengths().tolist() == [2, 0, 4]
+
+    def test_to_padded_dense(self):
+        jt = JaggedTensor(
+            values=torch.arange(6, dtype=torch.float32), lengths=torch.tensor([2, 0, 4])
+        )
+        d = jt.to_padded_dense(desired_length=3, padding_value=-1.0)
+        assert d.tolist() == [[0, 1, -1], [-1, -1, -1], [2, 3, 4]]
+
+    def test_to_padded_dense_2d(self):
+        vals = torch.arange(12, dtype=torch.float32).reshape(6, 2)
+        jt = JaggedTensor(values=vals, lengths=torch.tensor([1, 2, 3]))
+        d = jt.to_padded_dense(desired_length=3)
+        assert d.shape == (3, 3, 2)
+        assert d[0, 0].tolist() == [0.0, 1.0]
+        assert d[0, 1].tolist() == [0.0, 0.0]
+        assert d[2, 2].tolist() == [10.0, 11.0]
+
+    def test_from_dense(self):
+        jt = JaggedTensor.from_dense([torch.tensor([1.0, 2.0]), torch.tensor([3.0])])
+        assert jt.lengths().tolist() == [2, 1]
+        assert jt.values().tolist() == [1.0, 2.0, 3.0]
+
+    def test_from_dense_lengths(self):
+        vals = torch.arange(6, dtype=torch.float32).reshape(2, 3)
+        jt = JaggedTensor.from_dense_lengths(vals, torch.tensor([2, 1]))
+        assert jt.values().tolist() == [0.0, 1.0, 3.0]
+
+
+class TestKeyedJaggedTensor:
+    def test_to_dict(self):
+        kjt = make_kjt()
+        d = kjt.to_dict()
+        assert d["f1"].values().tolist() == [1, 2, 3]
+        assert d["f1"].lengths().tolist() == [2, 0, 1]
+        assert d["f2"].values().tolist() == [4, 5, 6]
+        assert d["f2"].lengths().tolist() == [1, 2, 0]
+
+    def test_split(self):
+        kjt = make_kjt()
+        a, b = kjt.split([1, 1])
+        assert a.keys() == ["f1"] and b.keys() == ["f2"]
+        assert a.values().tolist() == [1, 2, 3]
+        assert b.values().tolist() == [4, 5, 6]
+        assert b.lengths().tolist() == [1, 2, 0]
+        (whole,) = kjt.split([2])
+        assert whole.values().tolist() == [1, 2, 3, 4, 5, 6]
+
+    def test_permute(self):
+        kjt = make_kjt()
+        p = kjt.permute([1, 0])
+        assert p.keys() == ["f2", "f1"]
+        assert p.values().tolist() == [4, 5, 6, 1, 2, 3]
+        assert p.lengths().tolist() == [1, 2, 0, 2, 0, 1]
+
+    def test_permute_duplicate(self):
+        kjt = make_kjt()
+        p = kjt.permute([0, 1, 0])
+        assert p.keys() == ["f1", "f2", "f1"]
+        assert p.values().tolist() == [1, 2, 3, 4, 5, 6, 1, 2, 3]
+
+    def test_concat_roundtrip(self):
+        kjt = make_kjt()
+        a, b = kjt.split([1, 1])
+        back = KeyedJaggedTensor.concat([a, b])
+        assert back.keys() == kjt.keys()
+        assert back.values().tolist() == kjt.values().tolist()
+
+    def test_from_lengths_sync(self):
+        kjt = KeyedJaggedTensor.from_lengths_sync(
+            keys=["a"], values=torch.tensor([9, 8]), lengths=torch.tensor([1, 1]), stride=2
+        )
+        assert kjt.length_per_key() == [2]
+        assert kjt.offset_per_key() == [0, 2]
+
+    def test_weights(self):
+        kjt = KeyedJaggedTensor(
+            keys=["f1"],
+            values=torch.tensor([1, 2, 3]),
+            weights=torch.tensor([0.1, 0.2, 0.3]),
+            lengths=torch.tensor([1, 1, 1]),
+        )
+        p = kjt.permute([0])
+        assert torch.allclose(p.weights(), torch.tensor([0.1, 0.2, 0.3]))
+        jt = kjt["f1"]
+        assert torch.allclose(jt.weights(), torch.tensor([0.1, 0.2, 0.3]))
+
+    def test_empty(self):
+        kjt = KeyedJaggedTensor.empty()
+        assert kjt.keys() == []
+        assert kjt.stride() == 0
+
+
+class TestKeyedTensor:
+    def test_getitem_and_dict(self):
+        kt = KeyedTensor(
+            keys=["a", "b"],
+            length_per_key=[2, 3],
+            values=torch.arange(10, dtype=torch.float32).reshape(2, 5),
+        )
+        assert kt["a"].tolist() == [[0, 1], [5, 6]]
+        assert kt["b"].shape == (2, 3)
+        d = kt.to_dict()
+        assert set(d.keys()) == {"a", "b"}
+
+    def test_regroup(self):
+        kt1 = KeyedTensor(
+            keys=["a", "b"], length_per_key=[1, 2], values=torch.arange(6.0).reshape(2, 3)
+        )
+        kt2 = KeyedTensor(keys=["c"], length_per_key=[2], values=torch.arange(4.0).reshape(2, 2))
+        out = KeyedTensor.regroup([kt1, kt2], [["a", "c"], ["b"]])
+        assert out[0].shape == (2, 3)
+        assert out[0][0].tolist() == [0.0, 0.0, 1.0]
+        assert out[1][0].tolist() == [1.0, 2.0]
+
+    def test_from_tensor_list(self):
+        kt = KeyedTensor.from_tensor_list(["x", "y"], [torch.ones(2, 2), torch.zeros(2, 1)])
+        assert kt.length_per_key() == [2, 1]
+        assert kt.values().shape == (2, 3)

@@ -1,0 +1,167 @@
+"""EBC / EC / DLRM module tests (mirrors reference modules tests)."""
+
+import torch
+
+from torchrec_amd.datasets.random import generate_batch
+from torchrec_amd.models.dlrm import DLRM, DLRM_DCN, DLRMTrain, InteractionArch
+from torchrec_amd.modules.embedding_configs import (
+    EmbeddingBagConfig,
+    EmbeddingConfig,
+    PoolingType,
+)
+from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection, EmbeddingCollection
+from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+
+def make_kjt():
+    return KeyedJaggedTensor(
+        keys=["f1", "f2"],
+        values=torch.tensor([1, 2, 3, 4, 5, 6]),
+        lengths=torch.tensor([2, 0, 1, 1, 2, 0]),
+        stride=3,
+    )
+
+
+class TestEmbeddingBagCollection:
+    def test_forward_sum(self):
+        tables = [
+            EmbeddingBagConfig(num_embeddings=10, embedding_dim=4, name="t1", feature_names=["f1"]),
+            EmbeddingBagConfig(num_embeddings=10, embedding_dim=8, name="t2", feature_names=["f2"]),
+        ]
+        ebc = EmbeddingBagCollection(tables=tables)
+        kt = ebc(make_kjt())
+        assert kt.keys() == ["f1", "f2"]
+        assert kt.values().shape == (3, 12)
+        # numerics vs manual lookup
+        w1 = ebc.embedding_bags["t1"].weight
+        expected_row0 = w1[1] + w1[2]
+        assert torch.allclose(kt["f1"][0], expected_row0)
+        assert torch.allclose(kt["f1"][1], torch.zeros(4))
+
+    def test_forward_mean(self):
+        tables = [
+            EmbeddingBagConfig(
+                num_embeddings=10,
+                embedding_dim=4,
+                name="t1",
+                feature_names=["f1"],
+                pooling=PoolingType.MEAN,
+            ),
+        ]
+        ebc = EmbeddingBagCollection(tables=tables)
+        kjt = KeyedJaggedTensor(
+            keys=["f1"], values=torch.tensor([1, 2]), lengths=torch.tensor([2]), stride=1
+        )
+        kt = ebc(kjt)
+        w = ebc.embedding_bags["t1"].weight
+        assert torch.allclose(kt["f1"][0], (w[1] + w[2]) / 2)
+
+    def test_weighted(self):
+        tables = [
+            EmbeddingBagConfig(num_embeddings=10, embedding_dim=4, name="t1", feature_names=["f1"]),
+        ]
+        ebc = EmbeddingBagCollection(tables=tables, is_weighted=True)
+        kjt = KeyedJaggedTensor(
+            keys=["f1"],
+            values=torch.tensor([1, 2]),
+            weights=torch.tensor([2.0, 3.0]),
+            lengths=torch.tensor([2]),
+            stride=1,
+        )
+        kt = ebc(kjt)
+        w = ebc.embedding_bags["t1"].weight
+        assert torch.allclose(kt["f1"][0], 2.0 * w[1] + 3.0 * w[2])
+
+    def test_shared_table_multiple_features(self):
+        tables = [
+            EmbeddingBagConfig(
+                num_embeddings=10, embedding_dim=4, name="t1", feature_names=["f1", "f2"]
+            ),
+        ]
+        ebc = EmbeddingBagCollection(tables=tables)
+        kt = ebc(make_kjt())
+        assert kt.keys() == ["f1", "f2"]
+        assert kt.values().shape == (3, 8)
+
+
+class TestEmbeddingCollection:
+    def test_forward(self):
+        tables = [
+            EmbeddingConfig(num_embeddings=10, embedding_dim=4, name="t1", feature_names=["f1"]),
+            EmbeddingConfig(num_embeddings=10, embedding_dim=4, name="t2", feature_names=["f2"]),
+        ]
+        ec = EmbeddingCollection(tables=tables)
+        out = ec(make_kjt())
+        assert set(out.keys()) == {"f1", "f2"}
+        assert out["f1"].values().shape == (3, 4)
+        assert out["f1"].lengths().tolist() == [2, 0, 1]
+        w = ec.embeddings["t1"].weight
+        assert torch.allclose(out["f1"].values()[0], w[1])
+
+
+class TestDLRM:
+    def _tables(self, dim=8):
+        return [
+            EmbeddingBagConfig(
+                num_embeddings=100, embedding_dim=dim, name=f"t{i}", feature_names=[f"f{i}"]
+            )
+            for i in range(3)
+        ]
+
+    def test_forward_shape(self):
+        ebc = EmbeddingBagCollection(tables=self._tables())
+        model = DLRM(
+            embedding_bag_collection=ebc,
+            dense_in_features=13,
+            dense_arch_layer_sizes=[16, 8],
+            over_arch_layer_sizes=[16, 1],
+        )
+        batch = generate_batch(
+            keys=["f0", "f1", "f2"], batch_size=4, hash_sizes=[100] * 3, ids_per_feature=5
+        )
+        logits = model(batch.dense_features, batch.sparse_features)
+        assert logits.shape == (4, 1)
+
+    def test_interaction_math(self):
+        ia = InteractionArch(num_sparse_features=2)
+        dense = torch.tensor([[1.0, 0.0]])
+        sparse = torch.tensor([[[0.0, 1.0], [1.0, 1.0]]])
+        out = ia(dense, sparse)
+        # pairs: (dense,s0)=0, (dense,s1)=1, (s0,s1)=1
+        assert out.shape == (1, 2 + 3)
+        assert out[0, 2:].tolist() == [0.0, 1.0, 1.0]
+
+    def test_train_step(self):
+        ebc = EmbeddingBagCollection(tables=self._tables())
+        model = DLRMTrain(
+            DLRM(
+                embedding_bag_collection=ebc,
+                dense_in_features=13,
+                dense_arch_layer_sizes=[16, 8],
+                over_arch_layer_sizes=[16, 1],
+            )
+        )
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        batch = generate_batch(
+            keys=["f0", "f1", "f2"], batch_size=4, hash_sizes=[100] * 3, ids_per_feature=5
+        )
+        loss, _ = model(batch)
+        loss.backward()
+        opt.step()
+        assert torch.isfinite(loss)
+
+    def test_dlrm_dcn(self):
+        ebc = EmbeddingBagCollection(tables=self._tables())
+        model = DLRM_DCN(
+            embedding_bag_collection=ebc,
+            dense_in_features=13,
+            dense_arch_layer_sizes=[16, 8],
+            over_arch_layer_sizes=[16, 1],
+            dcn_num_layers=2,
+            dcn_low_rank_dim=4,
+        )
+        batch = generate_batch(
+            keys=["f0", "f1", "f2"], batch_size=4, hash_sizes=[100] * 3, ids_per_feature=5
+        )
+        logits = model(batch.dense_features, batch.sparse_features)
+        assert logits.shape == (4, 1)
