@@ -1,0 +1,36 @@
+"""Build the in-tree gfx950 HIP extension.
+
+Usage: PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+(The built .so lands in torchrec_amd/ops/ and ships to the GPU box with the
+repo snapshot; it is git-ignored so history stays source-only.)
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)), "torchrec_amd", "ops", "csrc")
+
+ext = CUDAExtension(
+    name="torchrec_amd.ops._hip_ops",
+    sources=[
+        os.path.join(CSRC, "jagged_ops.hip"),
+        os.path.join(CSRC, "tbe.hip"),
+        os.path.join(CSRC, "bindings.hip"),
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+    },
+)
+
+setup(
+    name="torchrec_amd_ext",
+    version="0.1.0",
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=True)},
+)

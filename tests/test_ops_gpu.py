@@ -1,0 +1,111 @@
+"""HIP jagged/KJT kernels vs the CPU reference implementations (@gpu)."""
+
+import pytest
+import torch
+
+from torchrec_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+
+def test_extension_loaded():
+    assert ops.extension_available(), "HIP extension must be present on a GPU box"
+    ops.hip_ops()
+
+
+def test_complete_cumsum():
+    x = torch.randint(0, 10, (1000,), dtype=torch.int64)
+    ref = ops.complete_cumsum(x)
+    got = ops.complete_cumsum(x.cuda()).cpu()
+    assert torch.equal(ref, got)
+    assert torch.equal(ops.complete_cumsum(torch.empty(0, dtype=torch.int64).cuda()).cpu(),
+                       torch.zeros(1, dtype=torch.int64))
+
+
+def test_lengths_range():
+    lengths = torch.randint(0, 6, (100,), dtype=torch.int64)
+    offsets = ops.complete_cumsum(lengths)
+    ref = ops.lengths_range(offsets)
+    got = ops.lengths_range(offsets.cuda()).cpu()
+    assert torch.equal(ref, got)
+
+
+def test_permute_2d_sparse_data():
+    torch.manual_seed(0)
+    K, B = 6, 9
+    lengths = torch.randint(0, 5, (K, B), dtype=torch.int64)
+    values = torch.randint(0, 1000, (int(lengths.sum()),), dtype=torch.int64)
+    weights = torch.rand(values.numel())
+    perm = torch.tensor([3, 0, 5, 5, 1, 2, 4])
+    rl, rv, rw = ops.permute_2d_sparse_data(perm, lengths, values, weights)
+    gl, gv, gw = ops.permute_2d_sparse_data(
+        perm.cuda(), lengths.cuda(), values.cuda(), weights.cuda()
+    )
+    assert torch.equal(rl, gl.cpu())
+    assert torch.equal(rv, gv.cpu())
+    assert torch.allclose(rw, gw.cpu())
+
+
+def test_jagged_to_padded_dense_roundtrip():
+    torch.manual_seed(0)
+    lengths = torch.randint(0, 7, (50,), dtype=torch.int64)
+    offsets = ops.complete_cumsum(lengths)
+    values = torch.randn(int(lengths.sum()), 16)
+    ref = ops.jagged_to_padded_dense(values, offsets, 7, -1.0)
+    got = ops.jagged_to_padded_dense(values.cuda(), offsets.cuda(), 7, -1.0)
+    assert torch.allclose(ref, got.cpu())
+    back = ops.dense_to_jagged(got, offsets.cuda())
+    assert torch.allclose(back.cpu(), values)
+
+
+def test_segment_sum_csr():
+    torch.manual_seed(0)
+    csr = torch.tensor([0, 3, 3, 10, 12], dtype=torch.int64)
+    values = torch.randn(12)
+    ref = ops.segment_sum_csr(1, csr, values)
+    got = ops.segment_sum_csr(1, csr.cuda(), values.cuda())
+    assert torch.allclose(ref, got.cpu(), atol=1e-6)
+
+
+def test_block_bucketize():
+    torch.manual_seed(0)
+    F, B, W = 3, 4, 4
+    lengths = torch.randint(0, 5, (F * B,), dtype=torch.int64)
+    N = int(lengths.sum())
+    rows = [100, 80, 60]
+    block_sizes = torch.tensor([(r + W - 1) // W for r in rows], dtype=torch.int64)
+    indices = torch.cat(
+        [
+            torch.randint(0, rows[f], (int(lengths[f * B : (f + 1) * B].sum()),))
+            for f in range(F)
+        ]
+    ) if N else torch.empty(0, dtype=torch.int64)
+    weights = torch.rand(N)
+    ref = ops.block_bucketize_sparse_features(
+        lengths, indices, True, True, block_sizes, W, weights=weights
+    )
+    got = ops.block_bucketize_sparse_features(
+        lengths.cuda(), indices.cuda(), True, True, block_sizes.cuda(), W,
+        weights=weights.cuda(),
+    )
+    for r, g in zip(ref, got):
+        if r is None:
+            assert g is None
+            continue
+        assert torch.equal(r.cpu(), g.cpu().to(r.dtype)), (r, g)
+
+
+def test_permute_pooled_embs():
+    torch.manual_seed(0)
+    B = 5
+    dims = [4, 8, 4]
+    vals = torch.randn(B, sum(dims), requires_grad=True)
+    vals_g = vals.detach().clone().cuda().requires_grad_(True)
+    order = torch.tensor([2, 0, 1])
+    ref = ops.permute_pooled_embs(vals, dims, order)
+    got = ops.permute_pooled_embs(vals_g, dims, order.cuda())
+    assert torch.allclose(ref, got.cpu())
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.cuda())
+    assert torch.allclose(vals.grad, vals_g.grad.cpu())

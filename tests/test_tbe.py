@@ -1,0 +1,247 @@
+"""TBE numerics tests.
+
+CPU tests validate the fused-optimizer oracle; @gpu tests validate the HIP
+kernels against the same plain-PyTorch fp32 reference (SURVEY.md §4
+golden-model pattern).
+"""
+
+import pytest
+import torch
+
+from torchrec_amd.ops.tbe import (
+    PoolingMode,
+    TableBatchedEmbeddingBags,
+    TableBatchedEmbeddings,
+)
+
+SPECS = [("t0", 100, 8), ("t1", 50, 16), ("t2", 1000, 8)]
+
+
+def make_inputs(specs, B=4, L=5, seed=0, device="cpu", feature_table_map=None):
+    g = torch.Generator().manual_seed(seed)
+    ftm = feature_table_map or list(range(len(specs)))
+    lengths = torch.randint(0, L + 1, (len(ftm) * B,), generator=g)
+    indices = torch.cat(
+        [
+            torch.randint(0, specs[t][1], (int(l),), generator=g)
+            for t, l in zip([ftm[i // B] for i in range(len(ftm) * B)], lengths)
+        ]
+    ) if int(lengths.sum()) else torch.empty(0, dtype=torch.int64)
+    offsets = torch.zeros(len(ftm) * B + 1, dtype=torch.int64)
+    torch.cumsum(lengths, 0, out=offsets[1:])
+    return indices.to(device), offsets.to(device)
+
+
+def reference_forward(tbe, indices, offsets, B, psw=None):
+    """Plain fp32 PyTorch reference over the split weights."""
+    outs = []
+    ws = tbe.split_embedding_weights()
+    for f, t in enumerate(tbe._feature_table_map):
+        w = ws[t].float().cpu()
+        out_f = []
+        for b in range(B):
+            lo, hi = int(offsets[f * B + b]), int(offsets[f * B + b + 1])
+            idx = indices[lo:hi].cpu()
+            rows = w[idx]
+            if psw is not None:
+                rows = rows * psw[lo:hi].cpu().unsqueeze(1)
+            if tbe.pooling_mode == PoolingMode.MEAN and len(idx) > 0:
+                out_f.append(rows.mean(0))
+            else:
+                out_f.append(rows.sum(0) if len(idx) else torch.zeros(w.shape[1]))
+        outs.append(torch.stack(out_f))
+    return torch.cat(outs, dim=1)
+
+
+class TestTBECpu:
+    def test_forward_sum(self):
+        tbe = TableBatchedEmbeddingBags(SPECS, optimizer="rowwise_adagrad")
+        indices, offsets = make_inputs(SPECS)
+        out = tbe(indices, offsets)
+        ref = reference_forward(tbe, indices, offsets, 4)
+        assert torch.allclose(out, ref, atol=1e-6)
+
+    def test_forward_mean(self):
+        tbe = TableBatchedEmbeddingBags(SPECS, pooling_mode=PoolingMode.MEAN)
+        indices, offsets = make_inputs(SPECS)
+        out = tbe(indices, offsets)
+        ref = reference_forward(tbe, indices, offsets, 4)
+        assert torch.allclose(out, ref, atol=1e-6)
+
+    def test_fused_rowwise_adagrad_step(self):
+        torch.manual_seed(7)
+        tbe = TableBatchedEmbeddingBags(
+            [("t0", 10, 4)], optimizer="rowwise_adagrad", learning_rate=0.1, eps=1e-8
+        )
+        w0 = tbe.split_embedding_weights()[0].clone()
+        indices = torch.tensor([1, 2, 1])
+        offsets = torch.tensor([0, 2, 3])  # B=2: bag0={1,2}, bag1={1}
+        out = tbe(indices, offsets)
+        grad = torch.ones_like(out)
+        out.backward(grad)
+        # row1 grad = 2 (both bags), row2 grad = 1
+        g = torch.zeros(10, 4)
+        g[1] = 2.0
+        g[2] = 1.0
+        m = g.pow(2).mean(1)
+        expected = w0 - 0.1 * g / (m.sqrt() + 1e-8).unsqueeze(1)
+        assert torch.allclose(tbe.split_embedding_weights()[0], expected, atol=1e-6)
+        mom = tbe.split_optimizer_states()[0][0]
+        assert torch.allclose(mom, m, atol=1e-6)
+
+    def test_sgd_step(self):
+        tbe = TableBatchedEmbeddingBags([("t0", 10, 4)], optimizer="sgd", learning_rate=0.5)
+        w0 = tbe.split_embedding_weights()[0].clone()
+        indices = torch.tensor([3])
+        offsets = torch.tensor([0, 1])
+        out = tbe(indices, offsets)
+        out.backward(torch.ones_like(out))
+        expected = w0.clone()
+        expected[3] -= 0.5 * torch.ones(4)
+        assert torch.allclose(tbe.split_embedding_weights()[0], expected, atol=1e-6)
+
+    def test_shared_table(self):
+        specs = [("t0", 20, 8)]
+        tbe = TableBatchedEmbeddingBags(specs, feature_table_map=[0, 0])
+        indices, offsets = make_inputs(specs, feature_table_map=[0, 0])
+        out = tbe(indices, offsets)
+        assert out.shape == (4, 16)
+        ref = reference_forward(tbe, indices, offsets, 4)
+        assert torch.allclose(out, ref, atol=1e-6)
+
+    def test_sequence(self):
+        specs = [("t0", 30, 8), ("t1", 40, 8)]
+        tbe = TableBatchedEmbeddings(specs, optimizer="sgd", learning_rate=1.0)
+        indices = torch.tensor([1, 2, 3, 4, 4])
+        offsets = torch.tensor([0, 2, 3, 4, 5])  # B=2, F=2
+        out = tbe(indices, offsets)
+        assert out.shape == (5, 8)
+        ws = tbe.split_embedding_weights()
+        assert torch.allclose(out[0], ws[0][1])
+        assert torch.allclose(out[3], ws[1][4])
+        w1_4 = ws[1][4].clone()
+        out.backward(torch.ones_like(out))
+        # row 4 of t1 hit twice -> grad 2
+        assert torch.allclose(tbe.split_embedding_weights()[1][4], w1_4 - 2.0)
+
+    def test_dense_optimizer_grad(self):
+        tbe = TableBatchedEmbeddingBags([("t0", 10, 4)], optimizer="dense")
+        indices = torch.tensor([1, 1, 2])
+        offsets = torch.tensor([0, 3])
+        out = tbe(indices, offsets)
+        out.sum().backward()
+        g = tbe.weights.grad.view(10, 4)
+        assert torch.allclose(g[1], torch.full((4,), 2.0))
+        assert torch.allclose(g[2], torch.full((4,), 1.0))
+
+
+@pytest.mark.gpu
+class TestTBEGpu:
+    def _pair(self, specs, optimizer="rowwise_adagrad", pooling=PoolingMode.SUM, ftm=None, lr=0.05):
+        torch.manual_seed(0)
+        cpu = TableBatchedEmbeddingBags(
+            specs, feature_table_map=ftm, pooling_mode=pooling, optimizer=optimizer,
+            learning_rate=lr,
+        )
+        gpu = TableBatchedEmbeddingBags(
+            specs, feature_table_map=ftm, pooling_mode=pooling, optimizer=optimizer,
+            learning_rate=lr, device=torch.device("cuda"),
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        return cpu, gpu
+
+    @pytest.mark.parametrize("pooling", [PoolingMode.SUM, PoolingMode.MEAN])
+    def test_forward_matches_cpu(self, pooling):
+        specs = [("t0", 100, 8), ("t1", 50, 128), ("t2", 1000, 64)]
+        cpu, gpu = self._pair(specs, pooling=pooling)
+        indices, offsets = make_inputs(specs, B=16, L=7)
+        out_c = cpu(indices, offsets)
+        out_g = gpu(indices.cuda(), offsets.cuda())
+        torch.cuda.synchronize()
+        assert torch.allclose(out_g.cpu(), out_c, atol=1e-5, rtol=1e-5)
+
+    def test_backward_fused_matches_cpu(self):
+        specs = [("t0", 100, 8), ("t1", 50, 128), ("t2", 1000, 64)]
+        cpu, gpu = self._pair(specs)
+        for step in range(3):
+            indices, offsets = make_inputs(specs, B=16, L=7, seed=step)
+            out_c = cpu(indices, offsets)
+            out_g = gpu(indices.cuda(), offsets.cuda())
+            grad = torch.randn_like(out_c)
+            out_c.backward(grad)
+            out_g.backward(grad.cuda())
+        torch.cuda.synchronize()
+        for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+            assert torch.allclose(wg.cpu(), wc, atol=1e-5, rtol=1e-4)
+        assert torch.allclose(gpu.momentum.cpu(), cpu.momentum, atol=1e-5, rtol=1e-4)
+
+    def test_backward_mean_matches_cpu(self):
+        specs = [("t0", 64, 16)]
+        cpu, gpu = self._pair(specs, pooling=PoolingMode.MEAN)
+        indices, offsets = make_inputs(specs, B=8, L=5)
+        out_c = cpu(indices, offsets)
+        out_g = gpu(indices.cuda(), offsets.cuda())
+        grad = torch.randn_like(out_c)
+        out_c.backward(grad)
+        out_g.backward(grad.cuda())
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            gpu.split_embedding_weights()[0].cpu(),
+            cpu.split_embedding_weights()[0],
+            atol=1e-5,
+            rtol=1e-4,
+        )
+
+    def test_weighted_forward(self):
+        specs = [("t0", 64, 16)]
+        cpu, gpu = self._pair(specs, optimizer="sgd")
+        indices, offsets = make_inputs(specs, B=8, L=5)
+        psw = torch.rand(indices.numel())
+        out_c = cpu(indices, offsets, psw)
+        out_g = gpu(indices.cuda(), offsets.cuda(), psw.cuda())
+        torch.cuda.synchronize()
+        assert torch.allclose(out_g.cpu(), out_c, atol=1e-5, rtol=1e-5)
+
+    def test_sequence_matches_cpu(self):
+        specs = [("t0", 30, 32), ("t1", 40, 32)]
+        torch.manual_seed(0)
+        cpu = TableBatchedEmbeddings(specs, optimizer="sgd", learning_rate=0.1)
+        gpu = TableBatchedEmbeddings(
+            specs, optimizer="sgd", learning_rate=0.1, device=torch.device("cuda")
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        indices = torch.tensor([1, 2, 3, 4, 4, 7])
+        offsets = torch.tensor([0, 2, 3, 4, 5, 5, 6])  # F=2, B=3
+        out_c = cpu(indices, offsets)
+        out_g = gpu(indices.cuda(), offsets.cuda())
+        grad = torch.randn_like(out_c)
+        out_c.backward(grad)
+        out_g.backward(grad.cuda())
+        torch.cuda.synchronize()
+        assert torch.allclose(out_g.cpu(), out_c, atol=1e-6)
+        for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+            assert torch.allclose(wg.cpu(), wc, atol=1e-5, rtol=1e-4)
+
+    def test_hot_row_long_run(self):
+        """Many duplicates of one id — exercises long segment runs."""
+        specs = [("t0", 16, 128)]
+        cpu, gpu = self._pair(specs)
+        B = 64
+        indices = torch.cat([torch.zeros(B * 10, dtype=torch.int64), torch.arange(16).repeat(4)])
+        lengths = torch.full((B,), 10, dtype=torch.int64)
+        lengths[-1] += 64
+        offsets = torch.zeros(B + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        out_c = cpu(indices, offsets)
+        out_g = gpu(indices.cuda(), offsets.cuda())
+        grad = torch.randn_like(out_c)
+        out_c.backward(grad)
+        out_g.backward(grad.cuda())
+        torch.cuda.synchronize()
+        assert torch.allclose(out_g.cpu(), out_c, atol=1e-4, rtol=1e-4)
+        assert torch.allclose(
+            gpu.split_embedding_weights()[0].cpu(),
+            cpu.split_embedding_weights()[0],
+            atol=1e-4,
+            rtol=1e-3,
+        )

@@ -1,0 +1,121 @@
+// torch.ops.trec_amd.* registration + Python module init.
+
+#include <torch/extension.h>
+
+#include <tuple>
+
+namespace trec_amd {
+
+// jagged_ops.hip
+at::Tensor complete_cumsum(const at::Tensor& lengths);
+at::Tensor lengths_range(const at::Tensor& offsets);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> permute_2d_sparse_data(
+    const at::Tensor& permute, const at::Tensor& lengths, const at::Tensor& values,
+    const at::Tensor& weights);
+at::Tensor jagged_to_padded_dense(const at::Tensor& values, const at::Tensor& offsets,
+                                  int64_t max_length, double padding_value);
+at::Tensor dense_to_jagged(const at::Tensor& dense, const at::Tensor& offsets);
+at::Tensor segment_sum_csr(const at::Tensor& csr, const at::Tensor& values);
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+block_bucketize_sparse_features(const at::Tensor& lengths, const at::Tensor& indices,
+                                const at::Tensor& block_sizes, int64_t num_buckets,
+                                bool bucketize_pos, bool sequence, const at::Tensor& weights);
+at::Tensor permute_pooled_embs(const at::Tensor& values, const at::Tensor& in_offsets,
+                               const at::Tensor& out_offsets, const at::Tensor& order);
+
+// tbe.hip
+at::Tensor tbe_forward_pooled(const at::Tensor& weights, const at::Tensor& table_elem_offsets,
+                              const at::Tensor& dims, const at::Tensor& feat_table,
+                              const at::Tensor& d_out_offsets, const at::Tensor& indices,
+                              const at::Tensor& offsets, const at::Tensor& per_sample_weights,
+                              int64_t B, int64_t total_D, int64_t max_D, bool mean_pool);
+at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_elem_offsets,
+                           const at::Tensor& dims, const at::Tensor& feat_table,
+                           const at::Tensor& feat_val_offsets, const at::Tensor& indices,
+                           int64_t D_out, int64_t max_D);
+std::tuple<at::Tensor, at::Tensor> sort_pairs(const at::Tensor& keys, int64_t end_bit);
+std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_linear);
+void tbe_backward_fused(at::Tensor weights, at::Tensor momentum, const at::Tensor& grad,
+                        const at::Tensor& sorted_linear, const at::Tensor& sort_perm,
+                        const at::Tensor& seg_offsets, const at::Tensor& num_runs,
+                        const at::Tensor& pos_row, const at::Tensor& pos_col,
+                        const at::Tensor& pos_scale, const at::Tensor& table_row_offsets,
+                        const at::Tensor& table_elem_offsets, const at::Tensor& dims,
+                        int64_t max_D, double lr, double eps, int64_t mode,
+                        at::Tensor grad_weights);
+at::Tensor tbe_grad_per_sample_weights(const at::Tensor& weights,
+                                       const at::Tensor& table_elem_offsets,
+                                       const at::Tensor& dims, const at::Tensor& grad,
+                                       const at::Tensor& indices, const at::Tensor& pos_row,
+                                       const at::Tensor& pos_col, const at::Tensor& pos_table,
+                                       int64_t max_D);
+at::Tensor bounds_check_indices(at::Tensor indices, const at::Tensor& feat_val_offsets,
+                                const at::Tensor& rows, const at::Tensor& feat_table);
+
+}  // namespace trec_amd
+
+TORCH_LIBRARY(trec_amd, m) {
+  m.def("complete_cumsum(Tensor lengths) -> Tensor");
+  m.def("lengths_range(Tensor offsets) -> Tensor");
+  m.def(
+      "permute_2d_sparse_data(Tensor permute, Tensor lengths, Tensor values, Tensor weights)"
+      " -> (Tensor, Tensor, Tensor)");
+  m.def(
+      "jagged_to_padded_dense(Tensor values, Tensor offsets, int max_length,"
+      " float padding_value) -> Tensor");
+  m.def("dense_to_jagged(Tensor dense, Tensor offsets) -> Tensor");
+  m.def("segment_sum_csr(Tensor csr, Tensor values) -> Tensor");
+  m.def(
+      "block_bucketize_sparse_features(Tensor lengths, Tensor indices, Tensor block_sizes,"
+      " int num_buckets, bool bucketize_pos, bool sequence, Tensor weights)"
+      " -> (Tensor, Tensor, Tensor, Tensor, Tensor)");
+  m.def(
+      "permute_pooled_embs(Tensor values, Tensor in_offsets, Tensor out_offsets,"
+      " Tensor order) -> Tensor");
+  m.def(
+      "tbe_forward_pooled(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
+      " Tensor feat_table, Tensor d_out_offsets, Tensor indices, Tensor offsets,"
+      " Tensor per_sample_weights, int B, int total_D, int max_D, bool mean_pool) -> Tensor");
+  m.def(
+      "tbe_forward_seq(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
+      " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"
+      " -> Tensor");
+  m.def("sort_pairs(Tensor keys, int end_bit) -> (Tensor, Tensor)");
+  m.def("tbe_backward_prep(Tensor sorted_linear) -> (Tensor, Tensor)");
+  m.def(
+      "tbe_backward_fused(Tensor(a!) weights, Tensor(b!) momentum, Tensor grad,"
+      " Tensor sorted_linear, Tensor sort_perm, Tensor seg_offsets, Tensor num_runs,"
+      " Tensor pos_row, Tensor pos_col, Tensor pos_scale, Tensor table_row_offsets,"
+      " Tensor table_elem_offsets, Tensor dims, int max_D, float lr, float eps, int mode,"
+      " Tensor(c!) grad_weights) -> ()");
+  m.def(
+      "tbe_grad_per_sample_weights(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
+      " Tensor grad, Tensor indices, Tensor pos_row, Tensor pos_col, Tensor pos_table,"
+      " int max_D) -> Tensor");
+  m.def(
+      "bounds_check_indices(Tensor(a!) indices, Tensor feat_val_offsets, Tensor rows,"
+      " Tensor feat_table) -> Tensor");
+}
+
+TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
+  m.impl("complete_cumsum", trec_amd::complete_cumsum);
+  m.impl("lengths_range", trec_amd::lengths_range);
+  m.impl("permute_2d_sparse_data", trec_amd::permute_2d_sparse_data);
+  m.impl("jagged_to_padded_dense", trec_amd::jagged_to_padded_dense);
+  m.impl("dense_to_jagged", trec_amd::dense_to_jagged);
+  m.impl("segment_sum_csr", trec_amd::segment_sum_csr);
+  m.impl("block_bucketize_sparse_features", trec_amd::block_bucketize_sparse_features);
+  m.impl("permute_pooled_embs", trec_amd::permute_pooled_embs);
+  m.impl("tbe_forward_pooled", trec_amd::tbe_forward_pooled);
+  m.impl("tbe_forward_seq", trec_amd::tbe_forward_seq);
+  m.impl("sort_pairs", trec_amd::sort_pairs);
+  m.impl("tbe_backward_prep", trec_amd::tbe_backward_prep);
+  m.impl("tbe_backward_fused", trec_amd::tbe_backward_fused);
+  m.impl("tbe_grad_per_sample_weights", trec_amd::tbe_grad_per_sample_weights);
+  m.impl("bounds_check_indices", trec_amd::bounds_check_indices);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("loaded", []() { return true; });
+  m.attr("__hip_arch__") = "gfx950";
+}
