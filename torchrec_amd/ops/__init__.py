@@ -69,6 +69,7 @@ def complete_cumsum(lengths: torch.Tensor) -> torch.Tensor:
     (reference: torchrec/sparse/jagged_tensor.py:157).
     """
     if lengths.is_cuda:
+        hip_ops()
         return torch.ops.trec_amd.complete_cumsum(lengths)
     out = torch.zeros(lengths.numel() + 1, dtype=lengths.dtype, device=lengths.device)
     torch.cumsum(lengths, dim=0, out=out[1:])
@@ -87,6 +88,7 @@ def lengths_range(offsets: torch.Tensor) -> torch.Tensor:
     """
     lengths = offsets_to_lengths(offsets)
     if offsets.is_cuda:
+        hip_ops()
         return torch.ops.trec_amd.lengths_range(offsets)
     seq = torch.arange(int(offsets[-1]), device=offsets.device, dtype=offsets.dtype)
     starts = torch.repeat_interleave(offsets[:-1], lengths)
@@ -121,6 +123,7 @@ def permute_2d_sparse_data(
     Returns (permuted_lengths [K',B], permuted_values, permuted_weights).
     """
     if values.is_cuda:
+        hip_ops()
         pl, pv, pw = torch.ops.trec_amd.permute_2d_sparse_data(
             permute, lengths, values, weights if weights is not None else torch.empty(0)
         )
@@ -184,6 +187,7 @@ def jagged_to_padded_dense(
     two_d = values.dim() == 2
     vals2 = values if two_d else values.unsqueeze(1)
     if values.is_cuda:
+        hip_ops()
         out = torch.ops.trec_amd.jagged_to_padded_dense(vals2, offsets, max_length, padding_value)
         return out if two_d else out.squeeze(-1)
     B = offsets.numel() - 1
@@ -199,6 +203,7 @@ def jagged_to_padded_dense(
 def dense_to_jagged(dense: torch.Tensor, offsets: torch.Tensor) -> torch.Tensor:
     """[B, max_length, D] dense -> [sum_L, D] jagged (inverse of padding)."""
     if dense.is_cuda:
+        hip_ops()
         return torch.ops.trec_amd.dense_to_jagged(dense, offsets)
     B = offsets.numel() - 1
     chunks = []
@@ -221,6 +226,7 @@ def segment_sum_csr(batch_size: int, csr_seg: torch.Tensor, values: torch.Tensor
     = sum(values[csr_seg[s]:csr_seg[s+1]]).
     """
     if values.is_cuda:
+        hip_ops()
         return torch.ops.trec_amd.segment_sum_csr(csr_seg, values)
     out = values.new_empty(csr_seg.numel() - 1)
     for s in range(csr_seg.numel() - 1):
@@ -260,6 +266,7 @@ def block_bucketize_sparse_features(
     torchrec/distributed/embedding_sharding.py:315).
     """
     if indices.is_cuda:
+        hip_ops()
         w = weights if weights is not None else torch.empty(0, device=indices.device)
         bl, bi, bw, bp, up = torch.ops.trec_amd.block_bucketize_sparse_features(
             lengths, indices, block_sizes, num_buckets, bucketize_pos, sequence, w
@@ -331,6 +338,7 @@ class _PermutePooledEmbs(torch.autograd.Function):
     def forward(ctx, values, in_offsets, out_offsets, order):  # type: ignore[override]
         ctx.save_for_backward(in_offsets, out_offsets, order)
         if values.is_cuda:
+            hip_ops()
             return torch.ops.trec_amd.permute_pooled_embs(values, in_offsets, out_offsets, order)
         cols = []
         for i in range(order.numel()):
