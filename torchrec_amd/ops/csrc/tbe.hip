@@ -269,7 +269,129 @@ std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_li
 }
 
 // ---------------------------------------------------------------------------
-// backward + fused optimizer. One wave-slot per duplicate-index run.
+// chunking of long duplicate-index runs.
+//
+// Hot rows (tiny tables: 3..128 rows under batch 8192 -> runs of thousands of
+// occurrences) would serialize a single wave-slot. Runs longer than
+// `chunk_size` are split into chunks processed by independent slots writing
+// fp32 partials to scratch (phase 1), then one slot per run sums its chunks
+// in order and applies the optimizer update (phase 2) — deterministic, no
+// atomics. Mirrors the reference's short/long-run split (triton TBE
+// :787/:1292) redesigned for wave-slot scheduling.
+// ---------------------------------------------------------------------------
+
+__global__ void chunk_prep_kernel(const int32_t* __restrict__ seg_offsets,
+                                  const int32_t* __restrict__ num_runs_ptr, int64_t n,
+                                  int chunk_size, int32_t* __restrict__ nchunks) {
+  int32_t num_runs = *num_runs_ptr;
+  for (int64_t r = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; r < n;
+       r += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    int32_t nc = 0;
+    if (r < num_runs) {
+      int32_t len = seg_offsets[r + 1] - seg_offsets[r];
+      if (len > chunk_size) nc = (len + chunk_size - 1) / chunk_size;
+    }
+    nchunks[r] = nc;
+  }
+}
+
+__global__ void total_chunks_kernel(const int32_t* __restrict__ chunk_offsets,
+                                    const int32_t* __restrict__ num_runs_ptr,
+                                    int32_t* __restrict__ total) {
+  *total = chunk_offsets[*num_runs_ptr];
+}
+
+std::tuple<at::Tensor, at::Tensor> tbe_backward_chunk_prep(const at::Tensor& seg_offsets,
+                                                           const at::Tensor& num_runs,
+                                                           int64_t chunk_size) {
+  int64_t n = seg_offsets.numel() - 1;  // max possible runs
+  auto opts = seg_offsets.options();
+  auto chunk_offsets = at::zeros({n + 1}, opts);
+  auto total = at::zeros({1}, opts);
+  if (n == 0) return {chunk_offsets, total};
+  auto nchunks = at::empty({n}, opts);
+  auto stream = tbe_stream();
+  int grid = grid_for(n, kBlockThreads);
+  hipLaunchKernelGGL(chunk_prep_kernel, dim3(grid), dim3(kBlockThreads), 0, stream,
+                     seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(), n,
+                     (int)chunk_size, nchunks.data_ptr<int32_t>());
+  size_t tmp_bytes = 0;
+  hipcub::DeviceScan::InclusiveSum(nullptr, tmp_bytes, nchunks.data_ptr<int32_t>(),
+                                   chunk_offsets.data_ptr<int32_t>() + 1, n, stream);
+  auto tmp = at::empty({static_cast<int64_t>(tmp_bytes)}, opts.dtype(at::kByte));
+  hipcub::DeviceScan::InclusiveSum(tmp.data_ptr(), tmp_bytes, nchunks.data_ptr<int32_t>(),
+                                   chunk_offsets.data_ptr<int32_t>() + 1, n, stream);
+  hipLaunchKernelGGL(total_chunks_kernel, dim3(1), dim3(1), 0, stream,
+                     chunk_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),
+                     total.data_ptr<int32_t>());
+  return {chunk_offsets, total};
+}
+
+__device__ __forceinline__ int upper_bound_segment_i32(const int32_t* offs, int n, int32_t x) {
+  int lo = 0, hi = n;
+  while (hi - lo > 1) {
+    int mid = (lo + hi) >> 1;
+    if (offs[mid] <= x) lo = mid; else hi = mid;
+  }
+  return lo;
+}
+
+template <int LPS, int CHUNKS>
+__global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
+    const float* __restrict__ grad, int64_t grad_stride,
+    const int64_t* __restrict__ sorted_linear, const int32_t* __restrict__ sort_perm,
+    const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
+    const int32_t* __restrict__ chunk_offsets, const int32_t* __restrict__ total_chunks_ptr,
+    const int32_t* __restrict__ pos_row, const int64_t* __restrict__ pos_col,
+    const float* __restrict__ pos_scale, const int64_t* __restrict__ table_row_offsets,
+    const int32_t* __restrict__ dims, int T, int chunk_size, int64_t max_D,
+    float* __restrict__ scratch) {
+  int sl = threadIdx.x % LPS;
+  int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
+  int64_t n_slots = (static_cast<int64_t>(gridDim.x) * blockDim.x) / LPS;
+  int32_t total_chunks = *total_chunks_ptr;
+  int32_t num_runs = *num_runs_ptr;
+  for (int64_t c = slot; c < total_chunks; c += n_slots) {
+    int r = upper_bound_segment_i32(chunk_offsets, num_runs, (int32_t)c);
+    int32_t rk0 = seg_offsets[r], rk1 = seg_offsets[r + 1];
+    int32_t k0 = rk0 + (c - chunk_offsets[r]) * chunk_size;
+    int32_t k1 = min(rk1, k0 + chunk_size);
+    int64_t lin = sorted_linear[rk0];
+    int t = upper_bound_segment(table_row_offsets, T, lin);
+    int D = dims[t];
+    float4 acc[CHUNKS];
+#pragma unroll
+    for (int cc = 0; cc < CHUNKS; ++cc) acc[cc] = make_float4(0.f, 0.f, 0.f, 0.f);
+    for (int32_t k = k0; k < k1; ++k) {
+      int32_t p = sort_perm[k];
+      const float4* grow = reinterpret_cast<const float4*>(
+          grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p]);
+      float s = pos_scale ? pos_scale[p] : 1.f;
+#pragma unroll
+      for (int cc = 0; cc < CHUNKS; ++cc) {
+        int col4 = cc * LPS + sl;
+        if (col4 * 4 < D) {
+          float4 g = grow[col4];
+          acc[cc].x += s * g.x;
+          acc[cc].y += s * g.y;
+          acc[cc].z += s * g.z;
+          acc[cc].w += s * g.w;
+        }
+      }
+    }
+    float4* srow = reinterpret_cast<float4*>(scratch + c * max_D);
+#pragma unroll
+    for (int cc = 0; cc < CHUNKS; ++cc) {
+      int col4 = cc * LPS + sl;
+      if (col4 * 4 < D) srow[col4] = acc[cc];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward + fused optimizer. One wave-slot per duplicate-index run; long
+// runs (chunk_offsets[r+1] > chunk_offsets[r]) sum pre-computed chunk
+// partials instead of walking occurrences.
 // mode: 0 = SGD, 1 = rowwise Adagrad, 2 = dense grad (write grad_weights).
 // ---------------------------------------------------------------------------
 
@@ -279,6 +401,8 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     const float* __restrict__ grad, int64_t grad_stride,
     const int64_t* __restrict__ sorted_linear, const int32_t* __restrict__ sort_perm,
     const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
+    const int32_t* __restrict__ chunk_offsets, const float* __restrict__ scratch,
+    int64_t max_D,
     const int32_t* __restrict__ pos_row, const int64_t* __restrict__ pos_col,
     const float* __restrict__ pos_scale,
     const int64_t* __restrict__ table_row_offsets,   // [T+1] rows
@@ -298,20 +422,39 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     float4 acc[CHUNKS];
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) acc[c] = make_float4(0.f, 0.f, 0.f, 0.f);
-    for (int32_t k = k0; k < k1; ++k) {
-      int32_t p = sort_perm[k];
-      const float4* grow = reinterpret_cast<const float4*>(
-          grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p]);
-      float s = pos_scale ? pos_scale[p] : 1.f;
+    int32_t c0 = chunk_offsets[r], c1 = chunk_offsets[r + 1];
+    if (c1 > c0) {
+      // long run: sum phase-1 chunk partials in order (deterministic)
+      for (int32_t c = c0; c < c1; ++c) {
+        const float4* srow = reinterpret_cast<const float4*>(scratch + c * max_D);
 #pragma unroll
-      for (int c = 0; c < CHUNKS; ++c) {
-        int col4 = c * LPS + sl;
-        if (col4 * 4 < D) {
-          float4 g = grow[col4];
-          acc[c].x += s * g.x;
-          acc[c].y += s * g.y;
-          acc[c].z += s * g.z;
-          acc[c].w += s * g.w;
+        for (int cc = 0; cc < CHUNKS; ++cc) {
+          int col4 = cc * LPS + sl;
+          if (col4 * 4 < D) {
+            float4 g = srow[col4];
+            acc[cc].x += g.x;
+            acc[cc].y += g.y;
+            acc[cc].z += g.z;
+            acc[cc].w += g.w;
+          }
+        }
+      }
+    } else {
+      for (int32_t k = k0; k < k1; ++k) {
+        int32_t p = sort_perm[k];
+        const float4* grow = reinterpret_cast<const float4*>(
+            grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p]);
+        float s = pos_scale ? pos_scale[p] : 1.f;
+#pragma unroll
+        for (int c = 0; c < CHUNKS; ++c) {
+          int col4 = c * LPS + sl;
+          if (col4 * 4 < D) {
+            float4 g = grow[col4];
+            acc[c].x += s * g.x;
+            acc[c].y += s * g.y;
+            acc[c].z += s * g.z;
+            acc[c].w += s * g.w;
+          }
         }
       }
     }
@@ -386,17 +529,36 @@ void tbe_backward_fused(
   int grid = grid_for(n * lps, kBlockThreads);  // upper bound: runs <= n
   const float* scale_ptr = pos_scale.numel() > 0 ? pos_scale.data_ptr<float>() : nullptr;
   float* gw_ptr = grad_weights.numel() > 0 ? grad_weights.data_ptr<float>() : nullptr;
+
+  // chunk long duplicate-runs: partials scratch sized by the host upper bound
+  constexpr int kChunkSize = 32;
+  auto [chunk_offsets, total_chunks] = tbe_backward_chunk_prep(seg_offsets, num_runs, kChunkSize);
+  int64_t max_chunks = 2 * ((n + kChunkSize - 1) / kChunkSize) + 2;
+  auto scratch = at::empty({max_chunks * max_D}, weights.options());
+  int grid_long = grid_for(max_chunks * lps, kBlockThreads);
+
 #define TBE_BWD_LAUNCH(LPS, CHUNKS)                                                          \
-  hipLaunchKernelGGL((tbe_bwd_fused_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads),   \
-                     0, stream, weights.data_ptr<float>(),                                   \
-                     momentum.numel() ? momentum.data_ptr<float>() : nullptr,                \
-                     grad.data_ptr<float>(), grad.size(1), sorted_linear.data_ptr<int64_t>(),\
-                     sort_perm.data_ptr<int32_t>(), seg_offsets.data_ptr<int32_t>(),         \
-                     num_runs.data_ptr<int32_t>(), pos_row.data_ptr<int32_t>(),              \
-                     pos_col.data_ptr<int64_t>(), scale_ptr,                                 \
-                     table_row_offsets.data_ptr<int64_t>(),                                  \
-                     table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,    \
-                     (float)lr, (float)eps, (int)mode, gw_ptr)
+  do {                                                                                       \
+    hipLaunchKernelGGL((tbe_bwd_long_partial_kernel<LPS, CHUNKS>), dim3(grid_long),          \
+                       dim3(kBlockThreads), 0, stream, grad.data_ptr<float>(), grad.size(1), \
+                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
+                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
+                       chunk_offsets.data_ptr<int32_t>(), total_chunks.data_ptr<int32_t>(),  \
+                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
+                       table_row_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,   \
+                       kChunkSize, max_D, scratch.data_ptr<float>());                        \
+    hipLaunchKernelGGL((tbe_bwd_fused_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads), \
+                       0, stream, weights.data_ptr<float>(),                                 \
+                       momentum.numel() ? momentum.data_ptr<float>() : nullptr,              \
+                       grad.data_ptr<float>(), grad.size(1),                                 \
+                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
+                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
+                       chunk_offsets.data_ptr<int32_t>(), scratch.data_ptr<float>(), max_D,  \
+                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
+                       table_row_offsets.data_ptr<int64_t>(),                                \
+                       table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,  \
+                       (float)lr, (float)eps, (int)mode, gw_ptr);                            \
+  } while (0)
   if (lps == 16) TBE_BWD_LAUNCH(16, 1);
   else if (lps == 32) TBE_BWD_LAUNCH(32, 1);
   else switch (chunks) {
