@@ -1,0 +1,304 @@
+"""Collective modules over KJTs and pooled/sequence embeddings.
+
+Reference parity: torchrec/distributed/dist_data.py — KJTAllToAll (:1139,
+splits-then-tensors protocol, SplitsAllToAllAwaitable :421,
+KJTAllToAllTensorsAwaitable :670), PooledEmbeddingsAllToAll (:1341),
+PooledEmbeddingsReduceScatter (:1733), SequenceEmbeddingsAllToAll (:1976).
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from torchrec_amd import ops
+from torchrec_amd.distributed.comm_ops import (
+    alltoall_pooled,
+    alltoall_sequence,
+    all_gather_base_pooled,
+    reduce_scatter_base_pooled,
+)
+from torchrec_amd.distributed.types import Awaitable, LazyAwaitable, NoWait
+from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+
+class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
+    """Phase 2: async lengths/values/weights a2a -> recat -> KJT.
+
+    Received layout is source-rank-major ((r, f, b)); `wait` permutes to the
+    feature-major global-batch layout ((f, r, b)) with the HIP permute kernel.
+    """
+
+    def __init__(
+        self,
+        pg: dist.ProcessGroup,
+        keys: List[str],
+        works: List[dist.Work],
+        out_lengths: torch.Tensor,
+        out_values: torch.Tensor,
+        out_weights: Optional[torch.Tensor],
+        B_local: int,
+        in_value_splits: Optional[List[int]] = None,
+        out_value_splits: Optional[List[int]] = None,
+    ) -> None:
+        super().__init__()
+        self._pg = pg
+        self._keys = keys
+        self._works = works
+        self._out_lengths = out_lengths
+        self._out_values = out_values
+        self._out_weights = out_weights
+        self._B = B_local
+        self._W = dist.get_world_size(pg)
+        # host-side a2a splits; sequence output_dist mirrors them
+        self.in_value_splits = in_value_splits or []
+        self.out_value_splits = out_value_splits or []
+
+    def _wait_impl(self) -> KeyedJaggedTensor:
+        for w in self._works:
+            if w is not None:
+                w.wait()
+        F = len(self._keys)
+        W = self._W
+        if F == 0:
+            return KeyedJaggedTensor.empty(device=self._out_values.device)
+        # recat (r, f) -> (f, r)
+        perm = torch.tensor(
+            [r * F + f for f in range(F) for r in range(W)],
+            dtype=torch.int64,
+            device=self._out_values.device,
+        )
+        lengths2d = self._out_lengths.view(W * F, self._B)
+        pl, pv, pw = ops.permute_2d_sparse_data(
+            perm, lengths2d, self._out_values, self._out_weights
+        )
+        kjt = KeyedJaggedTensor(
+            keys=self._keys,
+            values=pv,
+            weights=pw,
+            lengths=pl.reshape(-1),
+            stride=W * self._B,
+        )
+        kjt._dist_value_splits = (self.in_value_splits, self.out_value_splits)
+        return kjt
+
+
+class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
+    """Phase 1: async exchange of value counts, then issue the tensor a2a.
+
+    ``wait`` syncs the small splits tensor to host (unavoidable: RCCL a2a
+    needs host split sizes), then launches the big async transfers.
+    """
+
+    def __init__(
+        self,
+        pg: dist.ProcessGroup,
+        input: KeyedJaggedTensor,
+        splits: List[int],  # features per destination rank (input already ordered)
+        keys: List[str],  # my features post-exchange
+        stagger: int = 1,
+    ) -> None:
+        super().__init__()
+        self._pg = pg
+        self._input = input
+        self._splits = splits
+        self._keys = keys
+        self._W = dist.get_world_size(pg)
+        B = input.stride()
+        self._B = B
+        device = input.device()
+        # per-dest-rank value counts (device, async)
+        lengths = input.lengths()
+        feat_sums = lengths.view(len(input.keys()), B).sum(dim=1)
+        boundaries = torch.tensor(
+            [sum(splits[:i]) for i in range(len(splits) + 1)], device=device
+        )
+        send_counts = torch.stack(
+            [feat_sums[boundaries[i] : boundaries[i + 1]].sum() for i in range(self._W)]
+        )
+        self._in_splits_t = send_counts  # device
+        self._out_splits_t = torch.empty_like(send_counts)
+        self._splits_work = dist.all_to_all_single(
+            self._out_splits_t, send_counts, group=pg, async_op=True
+        )
+
+    def _wait_impl(self) -> KJTAllToAllTensorsAwaitable:
+        self._splits_work.wait()
+        in_value_splits = self._in_splits_t.cpu().tolist()  # sync (small)
+        out_value_splits = self._out_splits_t.cpu().tolist()
+        kjt = self._input
+        B = self._B
+        W = self._W
+        device = kjt.device()
+        F_mine = len(self._keys)
+        works = []
+        # lengths a2a: splits known statically
+        len_in_splits = [s * B for s in self._splits]
+        len_out_splits = [F_mine * B] * W
+        out_lengths = kjt.lengths().new_empty(sum(len_out_splits))
+        works.append(
+            dist.all_to_all_single(
+                out_lengths,
+                kjt.lengths().contiguous(),
+                len_out_splits,
+                len_in_splits,
+                group=self._pg,
+                async_op=True,
+            )
+        )
+        out_values = kjt.values().new_empty(sum(out_value_splits))
+        works.append(
+            dist.all_to_all_single(
+                out_values,
+                kjt.values().contiguous(),
+                out_value_splits,
+                in_value_splits,
+                group=self._pg,
+                async_op=True,
+            )
+        )
+        out_weights = None
+        if kjt.weights_or_none() is not None:
+            out_weights = kjt.weights().new_empty(sum(out_value_splits))
+            works.append(
+                dist.all_to_all_single(
+                    out_weights,
+                    kjt.weights().contiguous(),
+                    out_value_splits,
+                    in_value_splits,
+                    group=self._pg,
+                    async_op=True,
+                )
+            )
+        return KJTAllToAllTensorsAwaitable(
+            self._pg,
+            self._keys,
+            works,
+            out_lengths,
+            out_values,
+            out_weights,
+            B,
+            in_value_splits=in_value_splits,
+            out_value_splits=out_value_splits,
+        )
+
+
+class KJTAllToAll(nn.Module):
+    """Redistributes KJT features to their owning ranks (reference :1139).
+
+    ``splits[i]`` = number of (already-ordered) features destined to rank i.
+    forward(kjt) -> Awaitable[Awaitable[KJT]] (splits phase, tensors phase).
+    """
+
+    def __init__(self, pg: dist.ProcessGroup, splits: List[int], stagger: int = 1) -> None:
+        super().__init__()
+        self._pg = pg
+        self._splits = splits
+        self._stagger = stagger
+        self._splits_cumsum = [0]
+        for s in splits:
+            self._splits_cumsum.append(self._splits_cumsum[-1] + s)
+
+    def forward(self, input: KeyedJaggedTensor) -> Awaitable[KJTAllToAllTensorsAwaitable]:
+        rank = dist.get_rank(self._pg)
+        local_keys = input.keys()[
+            self._splits_cumsum[rank] : self._splits_cumsum[rank + 1]
+        ]
+        return KJTAllToAllSplitsAwaitable(
+            self._pg, input, self._splits, local_keys, self._stagger
+        )
+
+
+class PooledEmbeddingsAwaitable(LazyAwaitable[torch.Tensor]):
+    def __init__(self, tensor_awaitable: Awaitable[torch.Tensor]) -> None:
+        super().__init__()
+        self._tensor_awaitable = tensor_awaitable
+
+    def _wait_impl(self) -> torch.Tensor:
+        return self._tensor_awaitable.wait()
+
+
+class PooledEmbeddingsAllToAll(nn.Module):
+    """[W*B, D_local] -> awaitable [B, sum_D] (reference dist_data.py:1341)."""
+
+    def __init__(
+        self,
+        pg: dist.ProcessGroup,
+        dim_sum_per_rank: List[int],
+        device: Optional[torch.device] = None,
+        callbacks: Optional[List[Callable[[torch.Tensor], torch.Tensor]]] = None,
+        codec=None,
+    ) -> None:
+        super().__init__()
+        self._pg = pg
+        self._dim_sum_per_rank = dim_sum_per_rank
+        self._callbacks = callbacks or []
+        self._codec = codec
+
+    def forward(self, local_embs: torch.Tensor) -> PooledEmbeddingsAwaitable:
+        aw = alltoall_pooled(local_embs, self._dim_sum_per_rank, self._pg, codec=self._codec)
+        out = PooledEmbeddingsAwaitable(aw)
+        for cb in self._callbacks:
+            out.callbacks.append(cb)
+        return out
+
+    @property
+    def callbacks(self) -> List[Callable[[torch.Tensor], torch.Tensor]]:
+        return self._callbacks
+
+
+class PooledEmbeddingsReduceScatter(nn.Module):
+    """[W*B, D] partial sums -> awaitable [B, D] (reference :1733)."""
+
+    def __init__(self, pg: dist.ProcessGroup, codec=None) -> None:
+        super().__init__()
+        self._pg = pg
+        self._codec = codec
+
+    def forward(self, local_embs: torch.Tensor) -> PooledEmbeddingsAwaitable:
+        return PooledEmbeddingsAwaitable(reduce_scatter_base_pooled(local_embs, self._pg))
+
+
+class PooledEmbeddingsAllGather(nn.Module):
+    """[B, D] -> awaitable [W*B, D] (reference :1881)."""
+
+    def __init__(self, pg: dist.ProcessGroup) -> None:
+        super().__init__()
+        self._pg = pg
+
+    def forward(self, local_embs: torch.Tensor) -> PooledEmbeddingsAwaitable:
+        return PooledEmbeddingsAwaitable(all_gather_base_pooled(local_embs, self._pg))
+
+
+class SequenceEmbeddingsAwaitable(LazyAwaitable[torch.Tensor]):
+    def __init__(self, tensor_awaitable: Awaitable[torch.Tensor]) -> None:
+        super().__init__()
+        self._tensor_awaitable = tensor_awaitable
+
+    def _wait_impl(self) -> torch.Tensor:
+        return self._tensor_awaitable.wait()
+
+
+class SequenceEmbeddingsAllToAll(nn.Module):
+    """Per-row embedding exchange for sequence shardings (reference :1976).
+
+    forward splits = rows this rank sends to each peer (the ids it received
+    in input_dist, per source rank); output splits = rows it gets back.
+    """
+
+    def __init__(self, pg: dist.ProcessGroup) -> None:
+        super().__init__()
+        self._pg = pg
+
+    def forward(
+        self,
+        local_embs: torch.Tensor,
+        fwd_in_splits: List[int],
+        fwd_out_splits: List[int],
+    ) -> SequenceEmbeddingsAwaitable:
+        return SequenceEmbeddingsAwaitable(
+            alltoall_sequence(local_embs, fwd_in_splits, fwd_out_splits, self._pg)
+        )

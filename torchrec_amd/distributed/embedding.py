@@ -1,0 +1,372 @@
+"""Sharded EmbeddingCollection (sequence / non-pooled embeddings).
+
+Reference parity: torchrec/distributed/embedding.py
+(ShardedEmbeddingCollection :441 — input_dist :1560, compute :1645 returns
+per-row embeddings, output_dist :1662 = SequenceEmbeddingsAllToAll,
+EmbeddingCollectionAwaitable :349 building Dict[str, JaggedTensor]) and the
+sequence shardings (torchrec/distributed/sharding/tw_sequence_sharding.py,
+rw_sequence_sharding.py).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Type
+
+import torch
+import torch.nn as nn
+
+from torchrec_amd import ops
+from torchrec_amd.distributed.dist_data import (
+    KJTAllToAll,
+    SequenceEmbeddingsAllToAll,
+)
+from torchrec_amd.distributed.embedding_sharding import (
+    ShardedTableLocal,
+    bucketize_kjt_before_all2all,
+    group_tables_by_kernel,
+)
+from torchrec_amd.distributed.sharding.rw_sharding import rw_shard_rows
+from torchrec_amd.distributed.types import (
+    Awaitable,
+    EmbeddingModuleShardingPlan,
+    LazyAwaitable,
+    ModuleSharder,
+    NoWait,
+    ShardingEnv,
+    ShardingType,
+)
+from torchrec_amd.modules.embedding_configs import EmbeddingConfig, PoolingType
+from torchrec_amd.modules.embedding_modules import (
+    EmbeddingCollection,
+    get_embedding_names_by_table,
+)
+from torchrec_amd.optim.keyed import FusedOptimizer
+from torchrec_amd.ops.tbe import TableBatchedEmbeddings
+from torchrec_amd.sparse.jagged_tensor import JaggedTensor, KeyedJaggedTensor
+
+
+@dataclass
+class EmbeddingCollectionContext:
+    """SequenceShardingContext analogue (reference sharding/sequence_sharding.py)."""
+
+    input_kjts: List[Optional[KeyedJaggedTensor]] = field(default_factory=list)
+    unbucketize_permutes: List[Optional[torch.Tensor]] = field(default_factory=list)
+    local_features: List[Optional[KeyedJaggedTensor]] = field(default_factory=list)
+
+
+class EmbeddingCollectionAwaitable(LazyAwaitable[Dict[str, JaggedTensor]]):
+    """Waits per-sharding row tensors, reassembles Dict[str, JaggedTensor]."""
+
+    def __init__(
+        self,
+        awaitables: List[Awaitable[torch.Tensor]],
+        ctx: EmbeddingCollectionContext,
+        features_per_sharding: List[List[str]],
+        need_indices: bool = False,
+    ) -> None:
+        super().__init__()
+        self._awaitables = awaitables
+        self._ctx = ctx
+        self._features_per_sharding = features_per_sharding
+        self._need_indices = need_indices
+
+    def _wait_impl(self) -> Dict[str, JaggedTensor]:
+        out: Dict[str, JaggedTensor] = {}
+        for aw, feats, local_kjt, unbucketize in zip(
+            self._awaitables,
+            self._features_per_sharding,
+            self._ctx.local_features,
+            self._ctx.unbucketize_permutes,
+        ):
+            rows = aw.wait()
+            if unbucketize is not None:
+                # RW: rows arrived in bucketized order; map back
+                rows = rows.index_select(0, unbucketize)
+            lengths = local_kjt.lengths()
+            opk = local_kjt.offset_per_key()
+            B = local_kjt.stride()
+            for i, f in enumerate(feats):
+                out[f] = JaggedTensor(
+                    values=rows[opk[i] : opk[i + 1]],
+                    lengths=lengths[i * B : (i + 1) * B],
+                    weights=local_kjt.values()[opk[i] : opk[i + 1]]
+                    if self._need_indices
+                    else None,
+                )
+        return out
+
+
+class ShardedEmbeddingCollection(nn.Module):
+    """TW / RW / DP sequence sharding of an EmbeddingCollection."""
+
+    def __init__(
+        self,
+        module: EmbeddingCollection,
+        table_name_to_parameter_sharding: EmbeddingModuleShardingPlan,
+        env: ShardingEnv,
+        fused_params: Optional[Dict[str, Any]] = None,
+        device: Optional[torch.device] = None,
+    ) -> None:
+        super().__init__()
+        self._env = env
+        self._device = device or torch.device("cpu")
+        self._fused_params = dict(fused_params or {})
+        self._need_indices = module.need_indices()
+        W = env.world_size
+        rank = env.rank
+
+        emb_names_by_table = get_embedding_names_by_table(module.embedding_configs())
+        self._emb_name_per_feature: Dict[str, str] = {}
+        for cfg, names in zip(module.embedding_configs(), emb_names_by_table):
+            for f, n in zip(cfg.feature_names, names):
+                self._emb_name_per_feature[f] = n
+
+        by_type: Dict[str, List[EmbeddingConfig]] = {}
+        self._ps = table_name_to_parameter_sharding
+        for cfg in module.embedding_configs():
+            ps = table_name_to_parameter_sharding[cfg.name]
+            by_type.setdefault(ps.sharding_type, []).append(cfg)
+        self._sharding_types: List[str] = list(by_type.keys())
+
+        self._features_per_sharding: List[List[str]] = []
+        self._emb_names_per_sharding: List[List[str]] = []
+        self._lookups = nn.ModuleList()
+        self._input_splits_per_sharding: List[Optional[List[int]]] = []
+        self._block_sizes: List[Optional[torch.Tensor]] = []
+        self._a2a_modules = nn.ModuleList()
+        self._seq_a2a = nn.ModuleList()
+        self._dims_per_sharding: List[int] = []
+
+        for st, cfgs in by_type.items():
+            dims = {c.embedding_dim for c in cfgs}
+            assert len(dims) == 1, "sequence sharding requires uniform dim per group"
+            D = next(iter(dims))
+            self._dims_per_sharding.append(D)
+            feats = [f for c in cfgs for f in c.feature_names]
+            self._features_per_sharding.append(feats)
+            self._emb_names_per_sharding.append(
+                [self._emb_name_per_feature[f] for f in feats]
+            )
+            if st == ShardingType.TABLE_WISE.value:
+                tables_per_rank: List[List[ShardedTableLocal]] = [[] for _ in range(W)]
+                for c in cfgs:
+                    r = (self._ps[c.name].ranks or [0])[0]
+                    tables_per_rank[r].append(self._seq_table(c, c.num_embeddings))
+                # feature order: rank-major
+                feats_tw = [
+                    f for r in range(W) for t in tables_per_rank[r] for f in t.feature_names
+                ]
+                self._features_per_sharding[-1] = feats_tw
+                self._emb_names_per_sharding[-1] = [
+                    self._emb_name_per_feature[f] for f in feats_tw
+                ]
+                local = tables_per_rank[rank]
+                self._lookups.append(self._make_lookup(local, D))
+                self._input_splits_per_sharding.append(
+                    [len([f for t in tables_per_rank[r] for f in t.feature_names]) for r in range(W)]
+                )
+                self._block_sizes.append(None)
+            elif st == ShardingType.ROW_WISE.value:
+                local = [
+                    self._seq_table(c, rw_shard_rows(c.num_embeddings, W, rank))
+                    for c in cfgs
+                ]
+                self._lookups.append(self._make_lookup(local, D))
+                self._input_splits_per_sharding.append([len(feats)] * W)
+                self._block_sizes.append(
+                    torch.tensor(
+                        [(c.num_embeddings + W - 1) // W for c in cfgs for _ in c.feature_names],
+                        dtype=torch.int64,
+                    )
+                )
+            elif st == ShardingType.DATA_PARALLEL.value:
+                local = [self._seq_table(c, c.num_embeddings) for c in cfgs]
+                self._lookups.append(self._make_lookup(local, D, dense=True))
+                self._input_splits_per_sharding.append(None)
+                self._block_sizes.append(None)
+            else:
+                raise ValueError(f"sequence sharding {st} unsupported")
+            if st != ShardingType.DATA_PARALLEL.value and W > 1:
+                self._a2a_modules.append(
+                    KJTAllToAll(env.process_group, self._input_splits_per_sharding[-1])
+                )
+                self._seq_a2a.append(SequenceEmbeddingsAllToAll(env.process_group))
+            else:
+                self._a2a_modules.append(nn.Identity())
+                self._seq_a2a.append(nn.Identity())
+
+        self._fused_optimizer = _ECFusedOptimizer(self)
+
+    @staticmethod
+    def _seq_table(cfg: EmbeddingConfig, rows: int) -> ShardedTableLocal:
+        return ShardedTableLocal(
+            name=cfg.name,
+            local_rows=rows,
+            local_dim=cfg.embedding_dim,
+            pooling=PoolingType.NONE,
+            kernel="fused",
+            feature_names=list(cfg.feature_names),
+            full_dim=cfg.embedding_dim,
+        )
+
+    def _make_lookup(
+        self, tables: List[ShardedTableLocal], D: int, dense: bool = False
+    ) -> nn.Module:
+        specs = [(t.name, max(t.local_rows, 1), t.local_dim) for t in tables]
+        ftm = [i for i, t in enumerate(tables) for _ in t.feature_names]
+        return TableBatchedEmbeddings(
+            specs,
+            feature_table_map=ftm,
+            optimizer="dense" if dense else self._fused_params.get("optimizer", "rowwise_adagrad"),
+            learning_rate=self._fused_params.get("learning_rate", 0.01),
+            eps=self._fused_params.get("eps", 1.0e-8),
+            device=self._device,
+        )
+
+    # -- forward ------------------------------------------------------------
+
+    def create_context(self) -> EmbeddingCollectionContext:
+        n = len(self._sharding_types)
+        return EmbeddingCollectionContext(
+            input_kjts=[None] * n,
+            unbucketize_permutes=[None] * n,
+            local_features=[None] * n,
+        )
+
+    def input_dist(
+        self, ctx: EmbeddingCollectionContext, features: KeyedJaggedTensor
+    ) -> Awaitable[Awaitable[List[KeyedJaggedTensor]]]:
+        from torchrec_amd.distributed.embeddingbag import (
+            KJTListSplitsAwaitable,
+            KJTListTensorsAwaitable,
+        )
+
+        wanted = [f for feats in self._features_per_sharding for f in feats]
+        kjt_keys = features.keys()
+        if kjt_keys != wanted:
+            order = [kjt_keys.index(f) for f in wanted]
+            features = features.permute(order)
+        splits = [len(f) for f in self._features_per_sharding]
+        kjts = features.split(splits)
+        awaitables = []
+        for si, (st, kjt) in enumerate(zip(self._sharding_types, kjts)):
+            ctx.local_features[si] = kjt
+            if st == ShardingType.DATA_PARALLEL.value or self._env.world_size == 1:
+                awaitables.append(NoWait(NoWait(kjt)))
+                continue
+            if st == ShardingType.ROW_WISE.value:
+                bucketized, unbucketize = bucketize_kjt_before_all2all(
+                    kjt,
+                    num_buckets=self._env.world_size,
+                    block_sizes=self._block_sizes[si].to(kjt.device()),
+                    output_permute=True,
+                )
+                ctx.unbucketize_permutes[si] = unbucketize
+                awaitables.append(self._a2a_modules[si](bucketized))
+            else:
+                awaitables.append(self._a2a_modules[si](kjt))
+        return KJTListSplitsAwaitable(awaitables)
+
+    def compute_and_output_dist(
+        self, ctx: EmbeddingCollectionContext, dist_input: List[KeyedJaggedTensor]
+    ) -> EmbeddingCollectionAwaitable:
+        awaitables: List[Awaitable[torch.Tensor]] = []
+        for si, (st, kjt, lookup) in enumerate(
+            zip(self._sharding_types, dist_input, self._lookups)
+        ):
+            rows = lookup(kjt.values(), kjt.offsets())
+            if st == ShardingType.DATA_PARALLEL.value or self._env.world_size == 1:
+                awaitables.append(NoWait(rows))
+                continue
+            W = self._env.world_size
+            F = len(kjt.keys())
+            B_local = kjt.stride() // W
+            # permute rows (f, r) -> (r, f) segments, then a2a back to sources
+            lengths = kjt.lengths().view(F, W, B_local)
+            seg_counts = lengths.sum(dim=2)  # [F, W]
+            perm = torch.tensor(
+                [f * W + r for r in range(W) for f in range(F)],
+                dtype=torch.int64,
+                device=rows.device,
+            )
+            positions = torch.arange(rows.shape[0], device=rows.device)
+            _, perm_positions, _ = ops.permute_2d_sparse_data(
+                perm, seg_counts.reshape(-1, 1), positions
+            )
+            rows_rank_major = rows.index_select(0, perm_positions)
+            in_splits, out_splits = kjt._dist_value_splits
+            # output_dist mirrors input KJT a2a: send back what was received
+            awaitables.append(
+                self._seq_a2a[si](rows_rank_major, out_splits, in_splits)
+            )
+        return EmbeddingCollectionAwaitable(
+            awaitables, ctx, self._emb_names_per_sharding, self._need_indices
+        )
+
+    def compute(self, ctx, dist_input):
+        raise NotImplementedError("use compute_and_output_dist")
+
+    def output_dist(self, ctx, output):
+        raise NotImplementedError("use compute_and_output_dist")
+
+    def forward(self, features: KeyedJaggedTensor) -> LazyAwaitable[Dict[str, JaggedTensor]]:
+        ctx = self.create_context()
+        dist_input = self.input_dist(ctx, features).wait().wait()
+        return self.compute_and_output_dist(ctx, dist_input)
+
+    @property
+    def fused_optimizer(self) -> FusedOptimizer:
+        return self._fused_optimizer
+
+    def tbes(self) -> List[TableBatchedEmbeddings]:
+        return list(self._lookups)
+
+
+class _ECFusedOptimizer(FusedOptimizer):
+    def __init__(self, sharded_ec: ShardedEmbeddingCollection) -> None:
+        params: Dict[str, torch.Tensor] = {}
+        state: Dict[torch.Tensor, Any] = {}
+        param_groups: List[Dict[str, Any]] = []
+        for tbe in sharded_ec.tbes():
+            inner = tbe._bags
+            if inner.optimizer == 2:
+                continue
+            for spec, w, st in zip(
+                inner.embedding_specs,
+                inner.split_embedding_weights(),
+                inner.split_optimizer_states(),
+            ):
+                key = f"embeddings.{spec.name}.weight"
+                params[key] = w
+                if st:
+                    state[w] = {f"{spec.name}.momentum1": st[0]}
+                param_groups.append({"params": [w], "lr": inner.learning_rate})
+        super().__init__(params, state, param_groups)
+
+
+class EmbeddingCollectionSharder(ModuleSharder[EmbeddingCollection]):
+    def __init__(self, fused_params: Optional[Dict[str, Any]] = None) -> None:
+        self._fused_params = fused_params or {}
+
+    def shard(
+        self,
+        module: EmbeddingCollection,
+        params: EmbeddingModuleShardingPlan,
+        env: ShardingEnv,
+        device: Optional[torch.device] = None,
+    ) -> ShardedEmbeddingCollection:
+        return ShardedEmbeddingCollection(
+            module, params, env, fused_params=self._fused_params, device=device
+        )
+
+    @property
+    def module_type(self) -> Type[EmbeddingCollection]:
+        return EmbeddingCollection
+
+    def sharding_types(self, compute_device_type: str) -> List[str]:
+        return [
+            ShardingType.DATA_PARALLEL.value,
+            ShardingType.TABLE_WISE.value,
+            ShardingType.ROW_WISE.value,
+        ]

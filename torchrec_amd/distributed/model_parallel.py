@@ -1,0 +1,197 @@
+"""DistributedModelParallel — the training entry point.
+
+Reference parity: torchrec/distributed/model_parallel.py
+(DistributedModelParallel :255 — default planner :343-356, recursive module
+swap via sharder map, DDP wrap of the dense remainder :142, CombinedOptimizer
+of fused optims, state_dict passthrough).
+"""
+
+from __future__ import annotations
+
+import copy
+from typing import Any, Dict, List, Optional, Type
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+from torchrec_amd.distributed.types import (
+    ModuleSharder,
+    ShardingEnv,
+    ShardingPlan,
+)
+from torchrec_amd.optim.keyed import CombinedOptimizer, KeyedOptimizer
+
+
+def get_default_sharders() -> List[ModuleSharder[nn.Module]]:
+    """Reference parity: sharding_plan.py:49."""
+    from torchrec_amd.distributed.embedding import EmbeddingCollectionSharder
+
+    return [EmbeddingBagCollectionSharder(), EmbeddingCollectionSharder()]
+
+
+class DistributedModelParallel(nn.Module):
+    """Shards embedding modules per plan, DDP-wraps the dense remainder."""
+
+    def __init__(
+        self,
+        module: nn.Module,
+        env: Optional[ShardingEnv] = None,
+        device: Optional[torch.device] = None,
+        plan: Optional[ShardingPlan] = None,
+        sharders: Optional[List[ModuleSharder[nn.Module]]] = None,
+        init_data_parallel: bool = True,
+        init_parameters: bool = True,
+    ) -> None:
+        super().__init__()
+        torch._C._log_api_usage_once("torchrec_amd.DistributedModelParallel")
+        if env is None:
+            assert dist.is_initialized(), "need torch.distributed or explicit env"
+            env = ShardingEnv.from_process_group(dist.group.WORLD)
+        self._env = env
+        self.device = device or torch.device("cpu")
+        if sharders is None:
+            sharders = get_default_sharders()
+        self._sharder_map: Dict[Type[nn.Module], ModuleSharder[nn.Module]] = {
+            s.module_type: s for s in sharders
+        }
+        if plan is None:
+            planner = EmbeddingShardingPlanner(
+                topology=None if self.device.type == "cuda" else _cpu_topology(env)
+            )
+            plan = planner.collective_plan(module, sharders, env.process_group)
+        self._plan = plan
+
+        self._dmp_wrapped_module = module
+        self._sharded_modules: Dict[str, nn.Module] = {}
+        self._shard_modules_impl(module, "")
+
+        # move the dense remainder to device
+        self._move_dense(module)
+
+        self._optim = self._init_optim()
+
+        self._ddp_wrapped = False
+        if init_data_parallel and env.world_size > 1:
+            self.init_data_parallel()
+
+    # -- sharding ----------------------------------------------------------
+
+    def _shard_modules_impl(self, module: nn.Module, path: str) -> None:
+        for name, child in list(module.named_children()):
+            fqn = f"{path}.{name}" if path else name
+            sharder = self._sharder_map.get(type(child))
+            mplan = self._plan.get_plan_for_module(fqn) if sharder else None
+            if sharder is not None and mplan is not None:
+                sharded = sharder.shard(child, mplan, self._env, self.device)
+                setattr(module, name, sharded)
+                self._sharded_modules[fqn] = sharded
+            else:
+                self._shard_modules_impl(child, fqn)
+
+    def _move_dense(self, module: nn.Module) -> None:
+        sharded = set(self._sharded_modules.values())
+
+        def move(m: nn.Module) -> None:
+            for child in m.children():
+                if child in sharded:
+                    continue
+                move(child)
+            for p in m.parameters(recurse=False):
+                if p.device != self.device:
+                    p.data = p.data.to(self.device)
+            for b in m.buffers(recurse=False):
+                if b.device != self.device:
+                    b.data = b.data.to(self.device)
+
+        move(module)
+
+    def init_data_parallel(self) -> None:
+        """DDP over dense params; sharded-module params/buffers ignored
+        (reference model_parallel.py:142-254 DefaultDataParallelWrapper)."""
+        if self._ddp_wrapped or self._env.process_group is None:
+            return
+        ignore: List[str] = []
+        for fqn, sharded in self._sharded_modules.items():
+            for n, p in sharded.named_parameters():
+                # DP-sharded dense-kernel tables DO participate in DDP allreduce
+                if getattr(p, "_ddp_include", False):
+                    continue
+                ignore.append(f"{fqn}.{n}")
+            for n, _ in sharded.named_buffers():
+                ignore.append(f"{fqn}.{n}")
+        ignore_set = set(ignore)
+        dense_params = [
+            p
+            for name, p in self._dmp_wrapped_module.named_parameters()
+            if name not in ignore_set
+        ]
+        if not any(p.requires_grad for p in dense_params):
+            self._ddp_wrapped = True
+            return
+        nn.parallel.DistributedDataParallel._set_params_and_buffers_to_ignore_for_model(
+            self._dmp_wrapped_module, ignore
+        )
+        self._dmp_wrapped_module = nn.parallel.DistributedDataParallel(
+            self._dmp_wrapped_module,
+            device_ids=[self.device] if self.device.type == "cuda" else None,
+            process_group=self._env.process_group,
+            gradient_as_bucket_view=True,
+            static_graph=False,
+        )
+        self._ddp_wrapped = True
+
+    # -- optimizer ---------------------------------------------------------
+
+    def _init_optim(self) -> CombinedOptimizer:
+        optims: List = []
+        for fqn, sharded in self._sharded_modules.items():
+            if hasattr(sharded, "fused_optimizer"):
+                optims.append((fqn, sharded.fused_optimizer))
+        return CombinedOptimizer(optims)
+
+    @property
+    def fused_optimizer(self) -> KeyedOptimizer:
+        return self._optim
+
+    @property
+    def plan(self) -> ShardingPlan:
+        return self._plan
+
+    @property
+    def module(self) -> nn.Module:
+        if isinstance(self._dmp_wrapped_module, nn.parallel.DistributedDataParallel):
+            return self._dmp_wrapped_module.module
+        return self._dmp_wrapped_module
+
+    def sharded_modules(self) -> Dict[str, nn.Module]:
+        return self._sharded_modules
+
+    # -- nn.Module ---------------------------------------------------------
+
+    def forward(self, *args, **kwargs) -> Any:
+        return self._dmp_wrapped_module(*args, **kwargs)
+
+    def state_dict(self, *args, **kwargs):
+        return self.module.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, state_dict, strict: bool = True):
+        return self.module.load_state_dict(state_dict, strict=strict)
+
+    def named_parameters(self, prefix: str = "", recurse: bool = True, remove_duplicate: bool = True):
+        return self.module.named_parameters(prefix, recurse, remove_duplicate=remove_duplicate)
+
+    def bare_named_parameters(self):
+        return super().named_parameters()
+
+
+def _cpu_topology(env: ShardingEnv):
+    from torchrec_amd.distributed.planner import constants
+    from torchrec_amd.distributed.planner.types import Topology
+
+    # gloo/CPU runs: treat host RAM as the device memory pool
+    return Topology(
+        world_size=env.world_size, compute_device="cpu", hbm_cap=constants.DDR_CAP
+    )

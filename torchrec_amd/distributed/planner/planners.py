@@ -1,0 +1,372 @@
+"""Sharding planner: enumerate -> propose -> partition -> score -> best.
+
+Reference parity: torchrec/distributed/planner/ — EmbeddingShardingPlanner
+(planners.py:668), EmbeddingEnumerator (enumerators.py:81), proposers
+(proposers.py:34,137), GreedyPerfPartitioner (partitioners.py:176), perf /
+storage estimation (shard_estimators.py:71,126). Condensed into one module;
+the cost model uses the MI355X constants (constants.py).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Tuple, cast
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from torchrec_amd.distributed.planner import constants
+from torchrec_amd.distributed.planner.types import (
+    DeviceHardware,
+    ParameterConstraints,
+    Perf,
+    PlannerError,
+    Shard,
+    ShardingOption,
+    Storage,
+    Topology,
+)
+from torchrec_amd.distributed.types import (
+    EmbeddingComputeKernel,
+    EmbeddingModuleShardingPlan,
+    ModuleSharder,
+    ParameterSharding,
+    ShardMetadata,
+    ShardingPlan,
+    ShardingPlanner,
+    ShardingType,
+)
+
+logger = logging.getLogger(__name__)
+
+OPTIMIZER_MULTIPLIER = {"rowwise_adagrad": 1.0 / 128, "sgd": 0.0, "adagrad": 1.0}
+
+
+def _shard_bytes(rows: int, cols: int, elem: int = 4, opt_mult: float = 1.0 / 128) -> int:
+    return int(rows * cols * elem * (1 + opt_mult))
+
+
+def calculate_shards(
+    rows: int, cols: int, sharding_type: str, world_size: int, min_partition: int = 64
+) -> List[Tuple[List[int], List[int]]]:
+    """(sizes, offsets) per shard (reference sharding_plan.py:94)."""
+    if sharding_type in (ShardingType.TABLE_WISE.value, ShardingType.DATA_PARALLEL.value):
+        n = 1 if sharding_type == ShardingType.TABLE_WISE.value else world_size
+        return [([rows, cols], [0, 0]) for _ in range(n)]
+    if sharding_type == ShardingType.ROW_WISE.value:
+        block = (rows + world_size - 1) // world_size
+        out = []
+        for r in range(world_size):
+            lo = min(r * block, rows)
+            hi = min((r + 1) * block, rows)
+            out.append(([hi - lo, cols], [lo, 0]))
+        return out
+    if sharding_type == ShardingType.COLUMN_WISE.value:
+        # split into shards of >= min_partition cols, at most world_size shards
+        n_shards = min(world_size, max(1, cols // max(4, min_partition)))
+        base = cols // n_shards
+        base -= base % 4
+        out = []
+        off = 0
+        for i in range(n_shards):
+            w = cols - off if i == n_shards - 1 else base
+            out.append(([rows, w], [0, off]))
+            off += w
+        return out
+    raise PlannerError(f"unsupported sharding type {sharding_type}")
+
+
+class EmbeddingEnumerator:
+    """Cross product table x sharding_type x kernel (reference enumerators.py:81)."""
+
+    def __init__(
+        self,
+        topology: Topology,
+        constraints: Optional[Dict[str, ParameterConstraints]] = None,
+    ) -> None:
+        self._topology = topology
+        self._constraints = constraints or {}
+
+    def enumerate(
+        self, module: nn.Module, sharders: List[ModuleSharder[nn.Module]]
+    ) -> List[ShardingOption]:
+        options: List[ShardingOption] = []
+        W = self._topology.world_size
+        sharder_by_type = {s.module_type: s for s in sharders}
+        for fqn, child in module.named_modules():
+            sharder = sharder_by_type.get(type(child))
+            if sharder is None:
+                continue
+            configs = child.embedding_bag_configs() if hasattr(child, "embedding_bag_configs") else child.embedding_configs()
+            is_weighted = child.is_weighted() if hasattr(child, "is_weighted") else False
+            for cfg in configs:
+                cons = self._constraints.get(cfg.name)
+                allowed_types = (
+                    cons.sharding_types
+                    if cons and cons.sharding_types
+                    else sharder.sharding_types(self._topology.compute_device)
+                )
+                for st in allowed_types:
+                    if W == 1 and st not in (
+                        ShardingType.TABLE_WISE.value,
+                        ShardingType.DATA_PARALLEL.value,
+                    ):
+                        continue
+                    kernels = (
+                        cons.compute_kernels
+                        if cons and cons.compute_kernels
+                        else sharder.compute_kernels(st, self._topology.compute_device)
+                    )
+                    for kernel in kernels:
+                        shards = [
+                            Shard(size=list(sz), offset=list(off))
+                            for sz, off in calculate_shards(
+                                cfg.num_embeddings,
+                                cfg.embedding_dim,
+                                st,
+                                W,
+                                min_partition=(cons.min_partition if cons and cons.min_partition else 64),
+                            )
+                        ]
+                        opt = ShardingOption(
+                            name=cfg.name,
+                            module_fqn=fqn,
+                            config=cfg,
+                            sharding_type=st,
+                            compute_kernel=kernel,
+                            shards=shards,
+                            is_weighted=is_weighted,
+                        )
+                        self._estimate(opt, cons)
+                        options.append(opt)
+        return options
+
+    def _estimate(self, opt: ShardingOption, cons: Optional[ParameterConstraints]) -> None:
+        """Perf + storage per shard (reference shard_estimators.py:71,126)."""
+        topo = self._topology
+        B = topo.batch_size
+        W = topo.world_size
+        pooling = cons.pooling_factors[0] if cons else constants.POOLING_FACTOR
+        D = opt.config.embedding_dim
+        elem = 4  # fp32
+        opt_mult = (
+            0.0
+            if opt.compute_kernel == EmbeddingComputeKernel.DENSE.value
+            else 1.0 / D  # rowwise adagrad: one momentum scalar per row
+        )
+        for shard in opt.shards:
+            rows, cols = shard.size
+            # storage: weights + optimizer + a slice of activation/grad buffers
+            weight_bytes = int(rows * cols * elem * (1 + opt_mult))
+            act_bytes = int(B * W * cols * elem * 4)
+            shard.storage = Storage(hbm=weight_bytes + act_bytes, ddr=0)
+            # perf: bytes moved / bandwidth
+            if opt.sharding_type == ShardingType.DATA_PARALLEL.value:
+                local_B = B
+                fwd_bytes = local_B * pooling * cols * elem
+                comms = 2 * rows * cols * elem / topo.intra_host_bw  # allreduce-ish
+            elif opt.sharding_type == ShardingType.ROW_WISE.value:
+                local_B = B * W
+                fwd_bytes = local_B * pooling / W * cols * elem
+                comms = local_B * cols * elem / topo.intra_host_bw
+            else:  # TW / CW: global batch through this shard
+                local_B = B * W
+                fwd_bytes = local_B * pooling * cols * elem
+                comms = local_B * cols * elem / topo.intra_host_bw
+            fwd_compute = fwd_bytes / topo.hbm_mem_bw
+            shard.perf = Perf(
+                fwd_compute=fwd_compute,
+                fwd_comms=comms,
+                bwd_compute=fwd_compute * constants.BWD_COMPUTE_MULTIPLIER,
+                bwd_comms=comms,
+            )
+
+
+class GreedyPerfPartitioner:
+    """Bin-pack shards onto devices by perf (reference partitioners.py:176)."""
+
+    def partition(
+        self, proposal: List[ShardingOption], topology: Topology
+    ) -> List[ShardingOption]:
+        devices = [
+            DeviceHardware(d.rank, Storage(d.storage.hbm, d.storage.ddr), Perf())
+            for d in topology.devices
+        ]
+        # fixed-rank types first (RW/DP span all devices)
+        for opt in proposal:
+            if opt.sharding_type in (
+                ShardingType.ROW_WISE.value,
+                ShardingType.DATA_PARALLEL.value,
+            ):
+                for r, shard in enumerate(opt.shards):
+                    shard.rank = r
+                    dev = devices[r]
+                    dev.storage = dev.storage - shard.storage
+                    dev.perf = dev.perf + shard.perf
+                    if dev.storage.hbm < 0:
+                        raise PlannerError(f"OOM on rank {r} for {opt.name}")
+        # greedy for TW/CW: biggest perf first onto least-loaded feasible device
+        movable = [
+            (shard, opt)
+            for opt in proposal
+            if opt.sharding_type
+            in (ShardingType.TABLE_WISE.value, ShardingType.COLUMN_WISE.value)
+            for shard in opt.shards
+        ]
+        movable.sort(key=lambda x: x[0].perf.total if x[0].perf else 0, reverse=True)
+        for shard, opt in movable:
+            feasible = [d for d in devices if shard.storage.fits_in(d.storage)]
+            if not feasible:
+                raise PlannerError(f"no device fits shard of {opt.name}")
+            dev = min(feasible, key=lambda d: d.perf.total)
+            shard.rank = dev.rank
+            dev.storage = dev.storage - shard.storage
+            dev.perf = dev.perf + shard.perf
+        return proposal
+
+
+class GreedyProposer:
+    """Per-table best-perf choice, tables in size-desc order
+    (reference proposers.py:34)."""
+
+    def propose(self, options: List[ShardingOption]) -> List[List[ShardingOption]]:
+        by_table: Dict[Tuple[str, str], List[ShardingOption]] = {}
+        for o in options:
+            by_table.setdefault((o.module_fqn, o.name), []).append(o)
+        proposals: List[List[ShardingOption]] = []
+        # proposal k: per table, k-th best option by estimated perf
+        max_k = max(len(v) for v in by_table.values()) if by_table else 0
+        for k in range(min(max_k, 4)):
+            prop = []
+            for opts in by_table.values():
+                ranked = sorted(opts, key=lambda o: o.total_perf)
+                prop.append(ranked[min(k, len(ranked) - 1)])
+            proposals.append([self._clone(o) for o in prop])
+        return proposals
+
+    @staticmethod
+    def _clone(o: ShardingOption) -> ShardingOption:
+        return ShardingOption(
+            name=o.name,
+            module_fqn=o.module_fqn,
+            config=o.config,
+            sharding_type=o.sharding_type,
+            compute_kernel=o.compute_kernel,
+            shards=[
+                Shard(list(s.size), list(s.offset), None, s.storage, s.perf)
+                for s in o.shards
+            ],
+            is_weighted=o.is_weighted,
+        )
+
+
+class UniformProposer(GreedyProposer):
+    """All tables use the same sharding type (reference proposers.py:137)."""
+
+    def propose(self, options: List[ShardingOption]) -> List[List[ShardingOption]]:
+        proposals = []
+        for st in [
+            ShardingType.TABLE_WISE.value,
+            ShardingType.ROW_WISE.value,
+            ShardingType.COLUMN_WISE.value,
+            ShardingType.DATA_PARALLEL.value,
+        ]:
+            by_table: Dict[Tuple[str, str], ShardingOption] = {}
+            ok = True
+            for o in options:
+                if o.sharding_type != st:
+                    continue
+                key = (o.module_fqn, o.name)
+                if key not in by_table or o.total_perf < by_table[key].total_perf:
+                    by_table[key] = o
+            n_tables = len({(o.module_fqn, o.name) for o in options})
+            if by_table and len(by_table) == n_tables:
+                proposals.append([self._clone(o) for o in by_table.values()])
+        return proposals
+
+
+class EmbeddingShardingPlanner(ShardingPlanner):
+    """Reference parity: planners.py:668."""
+
+    def __init__(
+        self,
+        topology: Optional[Topology] = None,
+        constraints: Optional[Dict[str, ParameterConstraints]] = None,
+        batch_size: Optional[int] = None,
+    ) -> None:
+        if topology is None:
+            topology = Topology(
+                world_size=dist.get_world_size() if dist.is_initialized() else 1,
+                compute_device="cuda" if torch.cuda.is_available() else "cpu",
+            )
+        if batch_size is not None:
+            topology.batch_size = batch_size
+        self._topology = topology
+        self._enumerator = EmbeddingEnumerator(topology, constraints)
+        self._partitioner = GreedyPerfPartitioner()
+        self._proposers = [GreedyProposer(), UniformProposer()]
+
+    def plan(
+        self, module: nn.Module, sharders: List[ModuleSharder[nn.Module]]
+    ) -> ShardingPlan:
+        options = self._enumerator.enumerate(module, sharders)
+        if not options:
+            return ShardingPlan({})
+        best: Optional[List[ShardingOption]] = None
+        best_score = float("inf")
+        errors = []
+        for proposer in self._proposers:
+            for proposal in proposer.propose(options):
+                try:
+                    placed = self._partitioner.partition(proposal, self._topology)
+                except PlannerError as e:
+                    errors.append(str(e))
+                    continue
+                # score = max per-device perf (critical path)
+                per_dev = [0.0] * self._topology.world_size
+                for opt in placed:
+                    for shard in opt.shards:
+                        per_dev[shard.rank] += shard.perf.total if shard.perf else 0
+                score = max(per_dev)
+                if score < best_score:
+                    best_score = score
+                    best = placed
+        if best is None:
+            raise PlannerError(
+                f"no feasible sharding plan found; partition errors: {errors[:3]}"
+            )
+        return self._to_sharding_plan(best)
+
+    def collective_plan(
+        self, module: nn.Module, sharders: List[ModuleSharder[nn.Module]], pg
+    ) -> ShardingPlan:
+        """Plan on rank 0, broadcast (reference planners.py collective_plan)."""
+        if pg is None or dist.get_world_size(pg) == 1:
+            return self.plan(module, sharders)
+        if dist.get_rank(pg) == 0:
+            plan = self.plan(module, sharders)
+            obj = [plan]
+        else:
+            obj = [None]
+        dist.broadcast_object_list(obj, src=0, group=pg)
+        return obj[0]
+
+    def _to_sharding_plan(self, options: List[ShardingOption]) -> ShardingPlan:
+        plan: Dict[str, EmbeddingModuleShardingPlan] = {}
+        for opt in options:
+            mplan = plan.setdefault(opt.module_fqn, EmbeddingModuleShardingPlan())
+            mplan.plan[opt.name] = ParameterSharding(
+                sharding_type=opt.sharding_type,
+                compute_kernel=opt.compute_kernel,
+                ranks=[s.rank for s in opt.shards],
+                sharding_spec=[
+                    ShardMetadata(
+                        shard_offsets=list(s.offset),
+                        shard_sizes=list(s.size),
+                        placement_rank=s.rank,
+                    )
+                    for s in opt.shards
+                ],
+            )
+        return ShardingPlan(plan)

@@ -1,0 +1,157 @@
+"""Row-wise sharding (reference: torchrec/distributed/sharding/rw_sharding.py
+RwPooledEmbeddingSharding :661, RwSparseFeaturesDist :361 (bucketize + a2a),
+RwPooledEmbeddingDist :534 (reduce-scatter))."""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+
+from torchrec_amd.distributed.dist_data import (
+    KJTAllToAll,
+    PooledEmbeddingsReduceScatter,
+)
+from torchrec_amd.distributed.embedding_sharding import (
+    BaseEmbeddingDist,
+    BaseSparseFeaturesDist,
+    EmbeddingSharding,
+    EmbeddingShardingInfo,
+    GroupedPooledEmbeddingsLookup,
+    OutputColumnGroup,
+    ShardedTableLocal,
+    bucketize_kjt_before_all2all,
+    group_tables_by_kernel,
+)
+from torchrec_amd.distributed.types import Awaitable, NoWait, ShardingEnv
+from torchrec_amd.modules.embedding_configs import PoolingType
+from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+
+def rw_shard_rows(rows: int, world_size: int, rank: int) -> int:
+    """Even row split: block size ceil(rows/W); last rank takes the remainder."""
+    block = (rows + world_size - 1) // world_size
+    lo = min(rank * block, rows)
+    hi = min((rank + 1) * block, rows)
+    return hi - lo
+
+
+class RwSparseFeaturesDist(BaseSparseFeaturesDist):
+    """Bucketize ids by row block, then a2a (reference rw_sharding.py:361)."""
+
+    def __init__(self, pg, num_features: int, block_sizes: torch.Tensor) -> None:
+        super().__init__()
+        self._pg = pg
+        import torch.distributed as dist
+
+        self._W = dist.get_world_size(pg)
+        self._block_sizes = block_sizes
+        self._a2a = KJTAllToAll(pg, splits=[num_features] * self._W)
+
+    def forward(self, sparse_features: KeyedJaggedTensor):
+        bucketized, _ = bucketize_kjt_before_all2all(
+            sparse_features,
+            num_buckets=self._W,
+            block_sizes=self._block_sizes.to(sparse_features.device()),
+        )
+        return self._a2a(bucketized)
+
+
+class RwPooledEmbeddingDist(BaseEmbeddingDist):
+    def __init__(self, pg) -> None:
+        super().__init__()
+        self._rs = PooledEmbeddingsReduceScatter(pg)
+
+    def forward(self, local_embs: torch.Tensor):
+        return self._rs(local_embs)
+
+
+class RwPooledEmbeddingSharding(EmbeddingSharding):
+    """Rows of every table split across all ranks; partial pools reduce-scatter.
+
+    MEAN-pooled tables run the SUM kernel here; the mean divisor is applied
+    by the sharded EBC from pre-dist lengths (reference
+    embeddingbag.py _create_mean_pooling_divisor pattern).
+    """
+
+    def __init__(
+        self,
+        infos: List[EmbeddingShardingInfo],
+        env: ShardingEnv,
+        device: Optional[torch.device] = None,
+    ) -> None:
+        self._env = env
+        self._pg = env.process_group
+        W = env.world_size
+        rank = env.rank
+        self._fused_params: Dict = infos[0].fused_params if infos else {}
+        tables: List[ShardedTableLocal] = []
+        for info in infos:
+            cfg = info.embedding_config
+            block = (cfg.num_embeddings + W - 1) // W
+            tables.append(
+                ShardedTableLocal(
+                    name=cfg.name,
+                    local_rows=rw_shard_rows(cfg.num_embeddings, W, rank),
+                    local_dim=cfg.embedding_dim,
+                    pooling=cfg.pooling,
+                    kernel=info.param_sharding.compute_kernel,
+                    feature_names=list(cfg.feature_names),
+                    row_offset=min(rank * block, cfg.num_embeddings),
+                    full_dim=cfg.embedding_dim,
+                    full_rows=cfg.num_embeddings,
+                    use_sum_kernel=(cfg.pooling == PoolingType.MEAN and W > 1),
+                )
+            )
+        self._grouped = group_tables_by_kernel(tables)
+        self._features: List[str] = [
+            f for g in self._grouped for t in g for f in t.feature_names
+        ]
+        # block size per feature in grouped feature order
+        rows_by_name = {i.embedding_config.name: i.embedding_config.num_embeddings for i in infos}
+        self._block_sizes = torch.tensor(
+            [
+                (rows_by_name[t.name] + W - 1) // W
+                for g in self._grouped
+                for t in g
+                for _ in t.feature_names
+            ],
+            dtype=torch.int64,
+        )
+
+    def features_to_send(self) -> List[str]:
+        return self._features
+
+    def output_column_groups(self) -> List[OutputColumnGroup]:
+        return [
+            OutputColumnGroup(f, 0, t.local_dim)
+            for g in self._grouped
+            for t in g
+            for f in t.feature_names
+        ]
+
+    def mean_feature_names(self) -> List[str]:
+        return [
+            f
+            for g in self._grouped
+            for t in g
+            if t.use_sum_kernel
+            for f in t.feature_names
+        ]
+
+    def create_input_dist(self, device: torch.device) -> BaseSparseFeaturesDist:
+        if self._env.world_size == 1:
+            from torchrec_amd.distributed.sharding.tw_sharding import _NoOpFeaturesDist
+
+            return _NoOpFeaturesDist()
+        return RwSparseFeaturesDist(self._pg, len(self._features), self._block_sizes)
+
+    def create_lookup(self, device: torch.device) -> GroupedPooledEmbeddingsLookup:
+        return GroupedPooledEmbeddingsLookup(self._grouped, self._fused_params, device)
+
+    def create_output_dist(self, device: torch.device) -> BaseEmbeddingDist:
+        if self._env.world_size == 1:
+            from torchrec_amd.distributed.sharding.tw_sharding import _NoOpEmbeddingDist
+
+            return _NoOpEmbeddingDist()
+        return RwPooledEmbeddingDist(self._pg)

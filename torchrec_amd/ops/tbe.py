@@ -204,7 +204,12 @@ class TableBatchedEmbeddingBags(nn.Module):
             torch.tensor(d_out[:-1], dtype=torch.int64),
         )
         reg("_empty_f", torch.empty(0, dtype=torch.float32))
-        self._dummy = nn.Parameter(torch.zeros(1, device=device))
+        # autograd anchor for the fused path: requires-grad, non-persistent
+        # (not an nn.Parameter so it stays out of checkpoints / optimizers)
+        dummy = torch.zeros(1, device=device)
+        if device.type != "meta":
+            dummy.requires_grad_(True)
+        self.register_buffer("_dummy", dummy, persistent=False)
 
     # -- public API --------------------------------------------------------
 
