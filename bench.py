@@ -1,14 +1,14 @@
 """Flagship benchmark: DLRM training step on synthetic Criteo-TB data.
 
 Metric (BASELINE.json): "samples/sec (whole node) DLRM Criteo-TB synthetic at
-1/2/4/8 MI355X". Weak scaling: per-GPU batch is fixed, global batch = B * N.
+1/2/4/8 MI355X". Weak scaling: per-GPU batch fixed, global batch = B * N.
 Table shapes are the reference's DLRM-EMB config (reference
-benchmarks/README.md:14 — 26 tables, dim 128, full sizes; ~91 GB fp32, fully
-HBM-resident on one MI355X). Data is synthetic (random ids / dense / labels),
-weights random-init.
+benchmarks/README.md:14 — 26 tables, dim 128, full sizes; ~91 GB fp32 —
+fully HBM-resident on one MI355X, sharded by the planner at N>1).
 
-Single GPU runs the fused-TBE DLRM directly; N>1 runs DistributedModelParallel
-(table-wise sharding) over RCCL once the distributed stack is wired in.
+Path under test: DistributedModelParallel (planner-sharded EmbeddingBagCollection
+-> HIP TBE with fused rowwise-Adagrad) + TrainPipelineSparseDist (H2D ‖ KJT
+a2a ‖ compute on HIP streams) + DDP dense over RCCL.
 """
 
 from __future__ import annotations
@@ -17,7 +17,7 @@ import argparse
 import json
 import os
 import time
-from typing import List, Optional
+from typing import List
 
 import torch
 
@@ -32,106 +32,140 @@ NUM_DENSE = 13
 DENSE_ARCH = [512, 256, 128]
 OVER_ARCH = [1024, 1024, 512, 256, 1]
 IDS_PER_FEATURE = 1  # Criteo is one-hot per categorical feature
+LR = 0.05
 
 
-def build_tables(scale: float):
+def scaled_rows(scale: float) -> List[int]:
+    return [max(10, int(r * scale)) for r in DLRM_EMB_ROWS]
+
+
+def build_model(scale: float):
+    """DLRM over a meta-device EBC: only local shards materialize on GPU."""
+    from torchrec_amd.models.dlrm import DLRM, DLRMTrain
     from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
 
-    rows = [max(10, int(r * scale)) for r in DLRM_EMB_ROWS]
-    return [
+    tables = [
         EmbeddingBagConfig(
             num_embeddings=r,
             embedding_dim=EMB_DIM,
             name=f"t_cat_{i}",
             feature_names=[f"cat_{i}"],
         )
-        for i, r in enumerate(rows)
+        for i, r in enumerate(scaled_rows(scale))
     ]
-
-
-def build_model(device: torch.device, scale: float, learning_rate: float = 0.05):
-    """Single-process flagship: DLRM with fused-TBE sparse arch."""
-    from torchrec_amd.models.dlrm import DLRM, DLRMTrain
-    from torchrec_amd.modules.fused_embedding_modules import FusedEmbeddingBagCollection
-
-    tables = build_tables(scale)
-    ebc = FusedEmbeddingBagCollection(
-        tables,
-        optimizer="rowwise_adagrad",
-        learning_rate=learning_rate,
-        device=device,
+    ebc = EmbeddingBagCollection(tables=tables, device=torch.device("meta"))
+    model = DLRMTrain(
+        DLRM(
+            embedding_bag_collection=ebc,
+            dense_in_features=NUM_DENSE,
+            dense_arch_layer_sizes=DENSE_ARCH,
+            over_arch_layer_sizes=OVER_ARCH,
+        )
     )
-    model = DLRM(
-        embedding_bag_collection=ebc,
-        dense_in_features=NUM_DENSE,
-        dense_arch_layer_sizes=DENSE_ARCH,
-        over_arch_layer_sizes=OVER_ARCH,
-        dense_device=device,
-    )
-    return DLRMTrain(model)
+    return model
 
 
-def make_batches(n_batches: int, batch_size: int, scale: float, device: torch.device, seed: int):
+def make_host_batches(n_batches: int, batch_size: int, scale: float, seed: int, pin: bool):
     from torchrec_amd.datasets.random import generate_batch
 
-    rows = [max(10, int(r * scale)) for r in DLRM_EMB_ROWS]
+    rows = scaled_rows(scale)
     keys = [f"cat_{i}" for i in range(len(rows))]
-    gen = torch.Generator(device=device).manual_seed(seed)
-    return [
-        generate_batch(
+    gen = torch.Generator().manual_seed(seed)
+    out = []
+    for _ in range(n_batches):
+        b = generate_batch(
             keys,
             batch_size,
             rows,
             ids_per_feature=IDS_PER_FEATURE,
             num_dense=NUM_DENSE,
-            device=device,
             generator=gen,
+            pinned=pin,
         )
-        for _ in range(n_batches)
-    ]
+        out.append(b)
+    return out
 
 
-def _dist_ctx():
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    local = int(os.environ.get("LOCAL_RANK", rank))
-    return rank, world, local
+class _CyclingIterator:
+    def __init__(self, batches):
+        self._batches = batches
+        self._i = 0
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        b = self._batches[self._i % len(self._batches)]
+        self._i += 1
+        return b
 
 
 def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float) -> None:
     import torch.distributed as dist
 
-    rank, world, local_rank = _dist_ctx()
-    if world > 1:
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.train_pipeline import TrainPipelineSparseDist
+    from torchrec_amd.distributed.types import ShardingEnv
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)
-    torch.cuda.set_device(device)
-
     if world > 1:
-        loss_hist = _run_distributed(device, rank, world, steps, warmup, batch_size, scale)
-        return loss_hist
+        dist.init_process_group("nccl")
+        env = ShardingEnv.from_process_group(dist.group.WORLD)
+        pg = dist.group.WORLD
+    else:
+        env = ShardingEnv.from_local(1, 0)
+        pg = None
 
-    model = build_model(device, scale)
-    dense_params = [p for n, p in model.named_parameters() if "_tbe" not in n]
-    opt = torch.optim.SGD(dense_params, lr=0.05)
-    batches = make_batches(8, batch_size, scale, device, seed=1234 + rank)
+    model = build_model(scale)
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world, compute_device="cuda", batch_size=batch_size)
+    )
+    plan = planner.collective_plan(model, [sharder], pg)
+    dmp = DistributedModelParallel(
+        model, env=env, plan=plan, sharders=[sharder], device=device
+    )
+    if rank == 0:
+        counts = {}
+        for mplan in plan.plan.values():
+            for ps in mplan.values():
+                counts[ps.sharding_type] = counts.get(ps.sharding_type, 0) + 1
+        print(f"# plan sharding mix: {counts}", flush=True)
+    dense_opt = torch.optim.SGD(
+        [p for p in dmp.parameters() if p.requires_grad], lr=LR
+    )
+    pipeline = TrainPipelineSparseDist(dmp, dense_opt, device)
 
-    def step(i: int) -> None:
-        batch = batches[i % len(batches)]
-        loss, _ = model(batch)
-        opt.zero_grad(set_to_none=True)
-        loss.backward()
-        opt.step()
+    batches = make_host_batches(8, batch_size, scale, seed=1234 + rank, pin=True)
+    it = _CyclingIterator(batches)
 
-    for i in range(warmup):
-        step(i)
+    for _ in range(warmup):
+        pipeline.progress(it)
+    if world > 1:
+        dist.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for i in range(steps):
-        step(warmup + i)
+    for _ in range(steps):
+        pipeline.progress(it)
     torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
     dt = time.perf_counter() - t0
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([dt], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dt = float(t.item())
 
     ms_per_step = dt / steps * 1e3
     samples_per_sec = batch_size * world * steps / dt
@@ -149,31 +183,42 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float)
         "dtype": "fp32",
         "data": "synthetic (random ids/dense/labels, Criteo-TB shapes)",
         "config": {
-            "model": "DLRM (dot interaction, fused rowwise-Adagrad TBE)",
+            "model": "DLRM (dot interaction, fused rowwise-Adagrad HIP TBE)",
             "global_batch": batch_size * world,
             "local_batch": batch_size,
             "tables": len(DLRM_EMB_ROWS),
             "embedding_dim": EMB_DIM,
             "row_scale": scale,
-            "parallelism": "single" if world == 1 else f"tw{world}",
+            "parallelism": f"planner/dmp x{world} + pipeline",
         },
     }
     if rank == 0:
         print(json.dumps(result))
 
 
-def _run_distributed(device, rank, world, steps, warmup, batch_size, scale):
-    raise NotImplementedError(
-        "multi-GPU path lands with DistributedModelParallel (next milestone)"
-    )
-
-
 def run_smoke() -> None:
     """One tiny forward+backward of the flagship on cuda:0 (driver contract)."""
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.types import ShardingEnv
+
     device = torch.device("cuda", 0)
-    model = build_model(device, scale=1e-4)
-    batches = make_batches(1, 32, 1e-4, device, seed=0)
-    loss, _ = model(batches[0])
+    torch.cuda.set_device(device)
+    model = build_model(1e-4)
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=1, compute_device="cuda", batch_size=32)
+    )
+    plan = planner.plan(model, [sharder])
+    dmp = DistributedModelParallel(
+        model, env=ShardingEnv.from_local(1, 0), plan=plan, sharders=[sharder], device=device
+    )
+    batch = make_host_batches(1, 32, 1e-4, seed=0, pin=False)[0].to(device)
+    loss, _ = dmp(batch)
     loss.backward()
     torch.cuda.synchronize()
     assert torch.isfinite(loss), "smoke loss is not finite"

@@ -1,0 +1,104 @@
+"""TrainPipelineSparseDist semantics: pipelined losses == non-pipelined losses
+(reference pattern: train_pipeline/tests/test_train_pipelines.py)."""
+
+import torch
+import torch.distributed as dist
+
+from tests.dist_utils import run_multi_process
+from torchrec_amd.datasets.random import generate_batch
+from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+from torchrec_amd.distributed.planner.types import Topology
+from torchrec_amd.distributed.train_pipeline import TrainPipelineBase, TrainPipelineSparseDist
+from torchrec_amd.models.dlrm import DLRM, DLRMTrain
+from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+
+
+def _build_dmp(world_size, seed=11):
+    torch.manual_seed(seed)
+    tables = [
+        EmbeddingBagConfig(
+            num_embeddings=50 + 20 * i, embedding_dim=8, name=f"t{i}", feature_names=[f"f{i}"]
+        )
+        for i in range(3)
+    ]
+    model = DLRMTrain(
+        DLRM(
+            embedding_bag_collection=EmbeddingBagCollection(tables=tables),
+            dense_in_features=4,
+            dense_arch_layer_sizes=[8, 8],
+            over_arch_layer_sizes=[8, 1],
+        )
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40)
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.05}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(model, plan=plan, sharders=[sharder])
+    opt = torch.optim.SGD([p for p in dmp.parameters() if p.requires_grad], lr=0.05)
+    return dmp, opt, tables
+
+
+def _batches(rank, tables, n=6):
+    return [
+        generate_batch(
+            keys=[f"f{i}" for i in range(3)],
+            batch_size=4,
+            hash_sizes=[t.num_embeddings for t in tables],
+            ids_per_feature=3,
+            num_dense=4,
+            generator=torch.Generator().manual_seed(500 + 10 * s + rank),
+        )
+        for s in range(n)
+    ]
+
+
+def _run_pipeline_test(rank, world_size):
+    # reference run (no pipeline)
+    dmp_a, opt_a, tables = _build_dmp(world_size)
+    ref_losses = []
+    for batch in _batches(rank, tables):
+        loss, _ = dmp_a(batch)
+        opt_a.zero_grad()
+        loss.backward()
+        opt_a.step()
+        ref_losses.append(float(loss.detach()))
+
+    # pipelined run on an identical model
+    dmp_b, opt_b, _ = _build_dmp(world_size)
+    pipe = TrainPipelineSparseDist(dmp_b, opt_b, torch.device("cpu"))
+    it = iter(_batches(rank, tables))
+    pipe_losses = []
+    for _ in range(6):
+        out = pipe.progress(it)
+        pipe_losses.append(float(out[0]))
+    torch.testing.assert_close(
+        torch.tensor(pipe_losses), torch.tensor(ref_losses), atol=1e-5, rtol=1e-5
+    )
+
+
+def test_pipeline_sparse_dist_gloo():
+    run_multi_process(_run_pipeline_test, 2, "gloo")
+
+
+def test_pipeline_base_cpu_single():
+    import os
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29541")
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        dmp, opt, tables = _build_dmp(1)
+        pipe = TrainPipelineBase(dmp, opt, torch.device("cpu"))
+        it = iter(_batches(0, tables, n=3))
+        for _ in range(3):
+            out = pipe.progress(it)
+            assert torch.isfinite(out[0])
+    finally:
+        dist.destroy_process_group()

@@ -1,0 +1,220 @@
+"""Stream-overlapped train pipelines.
+
+Reference parity: torchrec/distributed/train_pipeline/train_pipelines.py
+(TrainPipelineBase :260 — 2-stage H2D overlap; TrainPipelineSparseDist :530 —
+3-stage, 2 batches in flight: H2D on a memcpy stream ‖ sparse input_dist on a
+data-dist stream ‖ fwd/bwd on the default stream, with sharded-module
+forwards swapped for PipelinedForward consuming the pre-started input_dist).
+
+MI355X notes: streams are HIP streams; the KJT a2a's splits phase syncs a
+small tensor to host — it runs one batch ahead so the sync hides under the
+current batch's compute. RCCL collectives ordered via work.wait() on the
+consuming stream.
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Any, Callable, Dict, Iterator, List, Optional, Tuple
+
+import torch
+
+from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+from torchrec_amd.streamable import Pipelineable
+
+logger = logging.getLogger(__name__)
+
+
+def _wait_for_batch(batch: Pipelineable, stream: Optional[torch.cuda.Stream]) -> None:
+    """Reference parity: train_pipeline/utils.py:151."""
+    if stream is None:
+        return
+    torch.cuda.current_stream().wait_stream(stream)
+    cur = torch.cuda.current_stream()
+    batch.record_stream(cur)
+
+
+class TrainPipelineBase:
+    """2-stage: H2D copy of batch i+1 overlaps fwd/bwd of batch i
+    (reference train_pipelines.py:260)."""
+
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        optimizer: torch.optim.Optimizer,
+        device: torch.device,
+    ) -> None:
+        self._model = model
+        self._optimizer = optimizer
+        self._device = device
+        self._memcpy_stream: Optional[torch.cuda.Stream] = (
+            torch.cuda.Stream(priority=-1) if device.type == "cuda" else None
+        )
+        self._cur_batch: Optional[Pipelineable] = None
+        self._connected = False
+
+    def _connect(self, dataloader_iter: Iterator[Pipelineable]) -> None:
+        cur_batch = next(dataloader_iter)
+        with torch.cuda.stream(self._memcpy_stream):
+            self._cur_batch = cur_batch.to(self._device, non_blocking=True)
+        self._connected = True
+
+    def progress(self, dataloader_iter: Iterator[Pipelineable]) -> Any:
+        if not self._connected:
+            self._connect(dataloader_iter)
+        cur_batch = self._cur_batch
+        assert cur_batch is not None
+        _wait_for_batch(cur_batch, self._memcpy_stream)
+
+        with torch.autograd.profiler.record_function("## next-batch H2D ##"):
+            try:
+                next_batch = next(dataloader_iter)
+            except StopIteration:
+                next_batch = None
+            if next_batch is not None:
+                with torch.cuda.stream(self._memcpy_stream):
+                    self._cur_batch = next_batch.to(self._device, non_blocking=True)
+            else:
+                self._cur_batch = None
+
+        self._optimizer.zero_grad(set_to_none=True)
+        with torch.autograd.profiler.record_function("## forward ##"):
+            losses, output = self._model(cur_batch)
+        with torch.autograd.profiler.record_function("## backward ##"):
+            torch.sum(losses).backward()
+        with torch.autograd.profiler.record_function("## optimizer ##"):
+            self._optimizer.step()
+        return output
+
+
+class PipelinedForward:
+    """Replaces a sharded module's forward: consumes the input_dist that the
+    pipeline started one step ahead (reference runtime_forwards.py:106)."""
+
+    def __init__(self, fqn: str, module: torch.nn.Module, pipeline: "TrainPipelineSparseDist") -> None:
+        self._fqn = fqn
+        self._module = module
+        self._pipeline = pipeline
+
+    def __call__(self, *args, **kwargs):
+        ctx, tensors_awaitable = self._pipeline._fetch_dist(self._fqn)
+        with torch.autograd.profiler.record_function(f"## wait_sparse_data_dist {self._fqn} ##"):
+            dist_input = tensors_awaitable.wait()
+        if self._pipeline._data_dist_stream is not None:
+            torch.cuda.current_stream().wait_stream(self._pipeline._data_dist_stream)
+            for kjt in dist_input if isinstance(dist_input, list) else [dist_input]:
+                if hasattr(kjt, "record_stream"):
+                    kjt.record_stream(torch.cuda.current_stream())
+        return self._module.compute_and_output_dist(ctx, dist_input)
+
+
+class TrainPipelineSparseDist:
+    """3-stage pipeline, 2 batches in flight (reference train_pipelines.py:530).
+
+    batch i:   fwd/bwd/opt on the default stream
+    batch i+1: sparse input_dist (KJT a2a) on the data-dist stream
+    batch i+2: H2D copy on the memcpy stream
+    """
+
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        optimizer: torch.optim.Optimizer,
+        device: torch.device,
+        execute_all_batches: bool = True,
+        extract_kjt: Optional[Callable[[Any], KeyedJaggedTensor]] = None,
+    ) -> None:
+        self._model = model
+        self._optimizer = optimizer
+        self._device = device
+        self._execute_all_batches = execute_all_batches
+        self._extract_kjt = extract_kjt or (lambda batch: batch.sparse_features)
+        is_cuda = device.type == "cuda"
+        self._memcpy_stream = torch.cuda.Stream(priority=-1) if is_cuda else None
+        self._data_dist_stream = torch.cuda.Stream(priority=-1) if is_cuda else None
+        self._batches: List[Optional[Pipelineable]] = []
+        self._dist_contexts: List[Dict[str, Tuple[Any, Any]]] = []
+        self._pipelined: Dict[str, torch.nn.Module] = {}
+        self._rewritten = False
+
+    # -- model rewrite -----------------------------------------------------
+
+    def _rewrite_model(self) -> None:
+        """Swap sharded-module forwards for PipelinedForward
+        (reference utils.py:508 _rewrite_model)."""
+        for fqn, module in self._model.named_modules():
+            if hasattr(module, "compute_and_output_dist") and hasattr(module, "input_dist"):
+                self._pipelined[fqn] = module
+                module.forward = PipelinedForward(fqn, module, self)
+        if not self._pipelined:
+            logger.warning("TrainPipelineSparseDist: no sharded modules found")
+        self._rewritten = True
+
+    def _fetch_dist(self, fqn: str) -> Tuple[Any, Any]:
+        return self._dist_contexts[0][fqn]
+
+    # -- stages ------------------------------------------------------------
+
+    def _copy_batch_to_gpu(self, dataloader_iter) -> Optional[Pipelineable]:
+        with torch.autograd.profiler.record_function("## copy_batch_to_gpu ##"):
+            try:
+                batch = next(dataloader_iter)
+            except StopIteration:
+                return None
+            with torch.cuda.stream(self._memcpy_stream):
+                return batch.to(self._device, non_blocking=True)
+
+    def _start_sparse_data_dist(self, batch: Optional[Pipelineable]) -> Dict[str, Tuple[Any, Any]]:
+        """Reference parity: utils.py:248 _start_data_dist."""
+        if batch is None:
+            return {}
+        out: Dict[str, Tuple[Any, Any]] = {}
+        with torch.autograd.profiler.record_function("## start_sparse_data_dist ##"):
+            with torch.cuda.stream(self._data_dist_stream):
+                _wait_for_batch(batch, self._memcpy_stream)
+                kjt = self._extract_kjt(batch)
+                for fqn, module in self._pipelined.items():
+                    ctx = module.create_context()
+                    splits_aw = module.input_dist(ctx, kjt)
+                    tensors_aw = splits_aw.wait()  # syncs small splits; one batch ahead
+                    out[fqn] = (ctx, tensors_aw)
+        return out
+
+    def _fill_pipeline(self, dataloader_iter) -> None:
+        b0 = self._copy_batch_to_gpu(dataloader_iter)
+        if b0 is None:
+            raise StopIteration
+        if not self._rewritten:
+            self._rewrite_model()
+        d0 = self._start_sparse_data_dist(b0)
+        b1 = self._copy_batch_to_gpu(dataloader_iter)
+        self._batches = [b0, b1]
+        self._dist_contexts = [d0]
+
+    def progress(self, dataloader_iter: Iterator[Pipelineable]) -> Any:
+        if not self._batches:
+            self._fill_pipeline(dataloader_iter)
+        cur_batch = self._batches[0]
+        next_batch = self._batches[1]
+        if cur_batch is None:
+            raise StopIteration
+
+        # stage 3 fill: batch i+2 H2D
+        batch_ip2 = self._copy_batch_to_gpu(dataloader_iter)
+
+        _wait_for_batch(cur_batch, self._memcpy_stream)
+        self._optimizer.zero_grad(set_to_none=True)
+
+        # batch i+1 sparse dist overlaps batch i compute
+        d_next = self._start_sparse_data_dist(next_batch)
+
+        with torch.autograd.profiler.record_function("## forward ##"):
+            losses, output = self._model(cur_batch)
+        with torch.autograd.profiler.record_function("## backward ##"):
+            torch.sum(losses).backward()
+        with torch.autograd.profiler.record_function("## optimizer ##"):
+            self._optimizer.step()
+
+        self._batches = [next_batch, batch_ip2]
+        self._dist_contexts = [d_next]
+        return output
