@@ -1,0 +1,91 @@
+"""Single-GPU end-to-end: DMP + HIP TBE + pipeline on cuda:0 (@gpu)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_dmp_pipeline_cuda():
+    import bench
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.train_pipeline import TrainPipelineSparseDist
+    from torchrec_amd.distributed.types import ShardingEnv
+
+    device = torch.device("cuda", 0)
+    model = bench.build_model(1e-4)
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.05}
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=1, compute_device="cuda", batch_size=64)
+    )
+    plan = planner.plan(model, [sharder])
+    dmp = DistributedModelParallel(
+        model, env=ShardingEnv.from_local(1, 0), plan=plan, sharders=[sharder], device=device
+    )
+    opt = torch.optim.SGD([p for p in dmp.parameters() if p.requires_grad], lr=0.05)
+    pipe = TrainPipelineSparseDist(dmp, opt, device)
+    it = bench._CyclingIterator(bench.make_host_batches(4, 64, 1e-4, seed=3, pin=True))
+    losses = []
+    for _ in range(6):
+        out = pipe.progress(it)
+        losses.append(float(out[0]))
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    # training should reduce loss on repeated identical data eventually; at
+    # least assert weights actually changed (native update ran)
+    sd = dmp.state_dict()
+    w = sd["model.sparse_arch.embedding_bag_collection.embedding_bags.t_cat_0.weight"]
+    assert w.abs().sum() > 0
+
+
+def test_cuda_vs_cpu_dmp_losses():
+    """Golden-model: cuda DMP losses match cpu DMP losses on the same data."""
+    import bench
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.types import ShardingEnv
+
+    def build(device_str):
+        torch.manual_seed(7)
+        model = bench.build_model(1e-4)
+        sharder = EmbeddingBagCollectionSharder(
+            fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.05}
+        )
+        planner = EmbeddingShardingPlanner(
+            topology=Topology(world_size=1, compute_device=device_str, batch_size=32, hbm_cap=1 << 40)
+        )
+        plan = planner.plan(model, [sharder])
+        return DistributedModelParallel(
+            model,
+            env=ShardingEnv.from_local(1, 0),
+            plan=plan,
+            sharders=[sharder],
+            device=torch.device(device_str),
+        )
+
+    cpu = build("cpu")
+    gpu = build("cuda")
+    gpu.load_state_dict({k: v for k, v in cpu.state_dict().items()}, strict=False)
+    # align dense params too
+    gsd = gpu.state_dict()
+    for k, v in cpu.state_dict().items():
+        if k in gsd and isinstance(v, torch.Tensor):
+            gsd[k].data.copy_(v.to(gsd[k].device))
+
+    batches = bench.make_host_batches(3, 32, 1e-4, seed=9, pin=False)
+    opt_c = torch.optim.SGD([p for p in cpu.parameters() if p.requires_grad], lr=0.05)
+    opt_g = torch.optim.SGD([p for p in gpu.parameters() if p.requires_grad], lr=0.05)
+    for b in batches:
+        lc, _ = cpu(b)
+        lg, _ = gpu(b.to(torch.device("cuda")))
+        torch.testing.assert_close(lg.cpu(), lc, atol=1e-4, rtol=1e-4)
+        opt_c.zero_grad(); lc.backward(); opt_c.step()
+        opt_g.zero_grad(); lg.backward(); opt_g.step()
+    torch.cuda.synchronize()
