@@ -164,16 +164,22 @@ class EmbeddingEnumerator:
             # perf: bytes moved / bandwidth
             if opt.sharding_type == ShardingType.DATA_PARALLEL.value:
                 local_B = B
-                fwd_bytes = local_B * pooling * cols * elem
-                comms = 2 * rows * cols * elem / topo.intra_host_bw  # allreduce-ish
+                # dense-kernel autograd path + grad allreduce: heavily penalized
+                # (reference constants.py DP_ELEMENTWISE_KERNELS_PERF_FACTOR)
+                fwd_bytes = (
+                    local_B * pooling * cols * elem
+                ) * constants.DP_ELEMENTWISE_KERNELS_PERF_FACTOR
+                comms = (
+                    2 * rows * cols * elem / topo.intra_host_bw if W > 1 else 0.0
+                )
             elif opt.sharding_type == ShardingType.ROW_WISE.value:
                 local_B = B * W
                 fwd_bytes = local_B * pooling / W * cols * elem
-                comms = local_B * cols * elem / topo.intra_host_bw
+                comms = local_B * cols * elem / topo.intra_host_bw if W > 1 else 0.0
             else:  # TW / CW: global batch through this shard
                 local_B = B * W
                 fwd_bytes = local_B * pooling * cols * elem
-                comms = local_B * cols * elem / topo.intra_host_bw
+                comms = local_B * cols * elem / topo.intra_host_bw if W > 1 else 0.0
             fwd_compute = fwd_bytes / topo.hbm_mem_bw
             shard.perf = Perf(
                 fwd_compute=fwd_compute,
