@@ -144,7 +144,9 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float)
     dense_opt = torch.optim.SGD(
         [p for p in dmp.parameters() if p.requires_grad], lr=LR
     )
-    pipeline = TrainPipelineSparseDist(dmp, dense_opt, device)
+    pipeline = TrainPipelineSparseDist(
+        dmp, dense_opt, device, autocast_dtype=torch.bfloat16
+    )
 
     batches = make_host_batches(8, batch_size, scale, seed=1234 + rank, pin=True)
     it = _CyclingIterator(batches)
@@ -180,7 +182,7 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float)
         "higher_is_better": True,
         "scaling": "weak",
         "vs_baseline": None,
-        "dtype": "fp32",
+        "dtype": "bf16",
         "data": "synthetic (random ids/dense/labels, Criteo-TB shapes)",
         "config": {
             "model": "DLRM (dot interaction, fused rowwise-Adagrad HIP TBE)",
@@ -189,6 +191,8 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float)
             "tables": len(DLRM_EMB_ROWS),
             "embedding_dim": EMB_DIM,
             "row_scale": scale,
+            "emb_dtype": "fp32",
+            "dense_dtype": "bf16-autocast",
             "parallelism": f"planner/dmp x{world} + pipeline",
         },
     }

@@ -14,6 +14,7 @@ consuming stream.
 
 from __future__ import annotations
 
+import contextlib
 import logging
 from typing import Any, Callable, Dict, Iterator, List, Optional, Tuple
 
@@ -43,15 +44,22 @@ class TrainPipelineBase:
         model: torch.nn.Module,
         optimizer: torch.optim.Optimizer,
         device: torch.device,
+        autocast_dtype: Optional[torch.dtype] = None,
     ) -> None:
         self._model = model
         self._optimizer = optimizer
         self._device = device
+        self._autocast_dtype = autocast_dtype
         self._memcpy_stream: Optional[torch.cuda.Stream] = (
             torch.cuda.Stream(priority=-1) if device.type == "cuda" else None
         )
         self._cur_batch: Optional[Pipelineable] = None
         self._connected = False
+
+    def _autocast_ctx(self):
+        if self._autocast_dtype is None:
+            return contextlib.nullcontext()
+        return torch.autocast(device_type=self._device.type, dtype=self._autocast_dtype)
 
     def _connect(self, dataloader_iter: Iterator[Pipelineable]) -> None:
         cur_batch = next(dataloader_iter)
@@ -78,7 +86,7 @@ class TrainPipelineBase:
                 self._cur_batch = None
 
         self._optimizer.zero_grad(set_to_none=True)
-        with torch.autograd.profiler.record_function("## forward ##"):
+        with torch.autograd.profiler.record_function("## forward ##"), self._autocast_ctx():
             losses, output = self._model(cur_batch)
         with torch.autograd.profiler.record_function("## backward ##"):
             torch.sum(losses).backward()
@@ -123,10 +131,12 @@ class TrainPipelineSparseDist:
         device: torch.device,
         execute_all_batches: bool = True,
         extract_kjt: Optional[Callable[[Any], KeyedJaggedTensor]] = None,
+        autocast_dtype: Optional[torch.dtype] = None,
     ) -> None:
         self._model = model
         self._optimizer = optimizer
         self._device = device
+        self._autocast_dtype = autocast_dtype
         self._execute_all_batches = execute_all_batches
         self._extract_kjt = extract_kjt or (lambda batch: batch.sparse_features)
         is_cuda = device.type == "cuda"
@@ -208,7 +218,12 @@ class TrainPipelineSparseDist:
         # batch i+1 sparse dist overlaps batch i compute
         d_next = self._start_sparse_data_dist(next_batch)
 
-        with torch.autograd.profiler.record_function("## forward ##"):
+        ac = (
+            torch.autocast(device_type=self._device.type, dtype=self._autocast_dtype)
+            if self._autocast_dtype is not None
+            else contextlib.nullcontext()
+        )
+        with torch.autograd.profiler.record_function("## forward ##"), ac:
             losses, output = self._model(cur_batch)
         with torch.autograd.profiler.record_function("## backward ##"):
             torch.sum(losses).backward()
