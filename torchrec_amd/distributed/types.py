@@ -98,6 +98,18 @@ class LazyAwaitable(Awaitable[W]):
             raise AttributeError(name)
         return getattr(self._force(), name)
 
+    def __getitem__(self, key):
+        return self._force()[key]
+
+    def __iter__(self):
+        return iter(self._force())
+
+    def __len__(self):
+        return len(self._force())
+
+    def __contains__(self, key):
+        return key in self._force()
+
 
 class LazyNoWait(LazyAwaitable[W]):
     def __init__(self, obj: W) -> None:
