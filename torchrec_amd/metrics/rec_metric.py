@@ -1,0 +1,288 @@
+"""Metrics framework core.
+
+Reference parity: torchrec/metrics/rec_metric.py (RecMetric :393,
+RecMetricComputation :202 — torchmetrics-style local state + windowed buffers
++ all-reduce on compute) and torchrec/metrics/metric_module.py:197
+(RecMetricModule).
+"""
+
+from __future__ import annotations
+
+import abc
+import math
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Any, Deque, Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+
+@dataclass
+class RecTaskInfo:
+    name: str = "DefaultTask"
+    label_name: str = "label"
+    prediction_name: str = "prediction"
+    weight_name: str = "weight"
+
+
+class WindowBuffer:
+    """Sliding window of per-batch states (reference rec_metric.py:162)."""
+
+    def __init__(self, max_size: int) -> None:
+        self._max_size = max_size
+        self._buffers: Deque[List[torch.Tensor]] = deque()
+
+    def append(self, states: List[torch.Tensor]) -> None:
+        self._buffers.append([s.detach().clone() for s in states])
+        while len(self._buffers) > self._max_size:
+            self._buffers.popleft()
+
+    def aggregate(self, n_states: int) -> List[torch.Tensor]:
+        if not self._buffers:
+            return []
+        out = []
+        for i in range(n_states):
+            out.append(torch.stack([b[i] for b in self._buffers]).sum(dim=0))
+        return out
+
+
+class RecMetricComputation(abc.ABC, nn.Module):
+    """Accumulates local state; all-reduces at compute()."""
+
+    STATE_NAMES: List[str] = []
+
+    def __init__(self, window_size: int = 100, process_group=None) -> None:
+        super().__init__()
+        self._pg = process_group
+        self._window = WindowBuffer(window_size)
+        for name in self.STATE_NAMES:
+            self.register_buffer(name, torch.zeros(1, dtype=torch.float64), persistent=False)
+
+    def _states(self) -> List[torch.Tensor]:
+        return [getattr(self, n) for n in self.STATE_NAMES]
+
+    @abc.abstractmethod
+    def update(
+        self, predictions: torch.Tensor, labels: torch.Tensor, weights: Optional[torch.Tensor]
+    ) -> None:
+        ...
+
+    @abc.abstractmethod
+    def _compute_from(self, states: List[torch.Tensor]) -> torch.Tensor:
+        ...
+
+    def _record_window(self, batch_states: List[torch.Tensor]) -> None:
+        self._window.append(batch_states)
+
+    def _reduced(self, states: List[torch.Tensor]) -> List[torch.Tensor]:
+        if self._pg is None and not (dist.is_available() and dist.is_initialized()):
+            return states
+        pg = self._pg
+        out = []
+        for s in states:
+            t = s.clone()
+            dist.all_reduce(t, group=pg)
+            out.append(t)
+        return out
+
+    def compute(self) -> Dict[str, torch.Tensor]:
+        lifetime = self._reduced(self._states())
+        window = self._reduced(self._window.aggregate(len(self.STATE_NAMES)) or self._states())
+        return {
+            "lifetime": self._compute_from(lifetime),
+            "window": self._compute_from(window),
+        }
+
+
+def _weights_or_ones(labels: torch.Tensor, weights: Optional[torch.Tensor]) -> torch.Tensor:
+    if weights is None:
+        return torch.ones_like(labels, dtype=torch.float64)
+    return weights.to(torch.float64)
+
+
+class NEComputation(RecMetricComputation):
+    """Normalized (cross-)entropy (reference metrics/ne.py)."""
+
+    STATE_NAMES = ["cross_entropy_sum", "weighted_num_samples", "pos_labels", "neg_labels"]
+
+    def update(self, predictions, labels, weights=None) -> None:
+        p = predictions.double().clamp(1e-7, 1 - 1e-7)
+        y = labels.double()
+        w = _weights_or_ones(y, weights)
+        ce = -(y * p.log() + (1 - y) * (1 - p).log())
+        batch = [
+            (ce * w).sum().reshape(1),
+            w.sum().reshape(1),
+            (y * w).sum().reshape(1),
+            ((1 - y) * w).sum().reshape(1),
+        ]
+        for name, b in zip(self.STATE_NAMES, batch):
+            getattr(self, name).add_(b)
+        self._record_window(batch)
+
+    def _compute_from(self, states) -> torch.Tensor:
+        ce_sum, n, pos, neg = states
+        total = pos + neg
+        base_p = (pos / total.clamp(min=1e-12)).clamp(1e-7, 1 - 1e-7)
+        base_ce = -(base_p * base_p.log() + (1 - base_p) * (1 - base_p).log())
+        return (ce_sum / n.clamp(min=1e-12)) / base_ce.clamp(min=1e-12)
+
+
+class AUCComputation(RecMetricComputation):
+    """Streaming AUC over a bounded window of raw scores (reference
+    metrics/auc.py keeps full windowed tensors; we cap the buffer)."""
+
+    STATE_NAMES: List[str] = []
+
+    def __init__(self, window_size: int = 100, process_group=None, max_elems: int = 1_000_000):
+        super().__init__(window_size, process_group)
+        self._preds: List[torch.Tensor] = []
+        self._labels: List[torch.Tensor] = []
+        self._weights: List[torch.Tensor] = []
+        self._max_elems = max_elems
+
+    def update(self, predictions, labels, weights=None) -> None:
+        self._preds.append(predictions.detach().double().reshape(-1))
+        self._labels.append(labels.detach().double().reshape(-1))
+        self._weights.append(_weights_or_ones(labels.reshape(-1), weights))
+        total = sum(p.numel() for p in self._preds)
+        while total > self._max_elems and len(self._preds) > 1:
+            total -= self._preds[0].numel()
+            self._preds.pop(0)
+            self._labels.pop(0)
+            self._weights.pop(0)
+
+    def compute(self) -> Dict[str, torch.Tensor]:
+        if not self._preds:
+            return {"lifetime": torch.tensor(0.5), "window": torch.tensor(0.5)}
+        p = torch.cat(self._preds)
+        y = torch.cat(self._labels)
+        w = torch.cat(self._weights)
+        if dist.is_available() and dist.is_initialized():
+            gp = [None] * dist.get_world_size(self._pg)
+            dist.all_gather_object(gp, (p, y, w), group=self._pg)
+            p = torch.cat([t[0] for t in gp])
+            y = torch.cat([t[1] for t in gp])
+            w = torch.cat([t[2] for t in gp])
+        auc = _weighted_auc(p, y, w)
+        return {"lifetime": auc, "window": auc}
+
+    def _compute_from(self, states):  # pragma: no cover - unused
+        raise NotImplementedError
+
+
+def _weighted_auc(preds: torch.Tensor, labels: torch.Tensor, weights: torch.Tensor) -> torch.Tensor:
+    order = torch.argsort(preds, descending=True)
+    y = labels[order]
+    w = weights[order]
+    tp = torch.cumsum(y * w, 0)
+    fp = torch.cumsum((1 - y) * w, 0)
+    total_pos = tp[-1] if tp.numel() else torch.tensor(0.0)
+    total_neg = fp[-1] if fp.numel() else torch.tensor(0.0)
+    if float(total_pos) == 0 or float(total_neg) == 0:
+        return torch.tensor(0.5, dtype=torch.float64)
+    tpr = torch.cat([torch.zeros(1, dtype=tp.dtype), tp / total_pos])
+    fpr = torch.cat([torch.zeros(1, dtype=fp.dtype), fp / total_neg])
+    return torch.trapz(tpr, fpr)
+
+
+class CalibrationComputation(RecMetricComputation):
+    """sum(pred)/sum(label) (reference metrics/calibration.py)."""
+
+    STATE_NAMES = ["calibration_num", "calibration_denom"]
+
+    def update(self, predictions, labels, weights=None) -> None:
+        w = _weights_or_ones(labels.double(), weights)
+        batch = [
+            (predictions.double() * w).sum().reshape(1),
+            (labels.double() * w).sum().reshape(1),
+        ]
+        for name, b in zip(self.STATE_NAMES, batch):
+            getattr(self, name).add_(b)
+        self._record_window(batch)
+
+    def _compute_from(self, states) -> torch.Tensor:
+        num, denom = states
+        return num / denom.clamp(min=1e-12)
+
+
+class MSEComputation(RecMetricComputation):
+    STATE_NAMES = ["error_sum", "weighted_num_samples"]
+
+    def update(self, predictions, labels, weights=None) -> None:
+        w = _weights_or_ones(labels.double(), weights)
+        batch = [
+            (w * (predictions.double() - labels.double()) ** 2).sum().reshape(1),
+            w.sum().reshape(1),
+        ]
+        for name, b in zip(self.STATE_NAMES, batch):
+            getattr(self, name).add_(b)
+        self._record_window(batch)
+
+    def _compute_from(self, states) -> torch.Tensor:
+        err, n = states
+        return err / n.clamp(min=1e-12)
+
+
+class RecMetric(nn.Module):
+    """Multi-task wrapper over a computation class (reference rec_metric.py:393)."""
+
+    COMPUTATION: type = NEComputation
+    NAME = "metric"
+
+    def __init__(
+        self,
+        tasks: List[RecTaskInfo],
+        window_size: int = 100,
+        process_group=None,
+        **kwargs,
+    ) -> None:
+        super().__init__()
+        self._tasks = tasks
+        self._computations = nn.ModuleList(
+            [self.COMPUTATION(window_size=window_size, process_group=process_group) for _ in tasks]
+        )
+
+    def update(
+        self,
+        *,
+        predictions: Dict[str, torch.Tensor],
+        labels: Dict[str, torch.Tensor],
+        weights: Optional[Dict[str, torch.Tensor]] = None,
+    ) -> None:
+        for task, comp in zip(self._tasks, self._computations):
+            comp.update(
+                predictions[task.name],
+                labels[task.name],
+                weights.get(task.name) if weights else None,
+            )
+
+    def compute(self) -> Dict[str, torch.Tensor]:
+        out = {}
+        for task, comp in zip(self._tasks, self._computations):
+            res = comp.compute()
+            out[f"{self.NAME}-{task.name}|lifetime_{self.NAME}"] = res["lifetime"]
+            out[f"{self.NAME}-{task.name}|window_{self.NAME}"] = res["window"]
+        return out
+
+
+class NEMetric(RecMetric):
+    COMPUTATION = NEComputation
+    NAME = "ne"
+
+
+class AUCMetric(RecMetric):
+    COMPUTATION = AUCComputation
+    NAME = "auc"
+
+
+class CalibrationMetric(RecMetric):
+    COMPUTATION = CalibrationComputation
+    NAME = "calibration"
+
+
+class MSEMetric(RecMetric):
+    COMPUTATION = MSEComputation
+    NAME = "mse"
