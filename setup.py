@@ -20,6 +20,7 @@ ext = CUDAExtension(
     sources=[
         os.path.join(CSRC, "jagged_ops.hip"),
         os.path.join(CSRC, "tbe.hip"),
+        os.path.join(CSRC, "quant_tbe.hip"),
         os.path.join(CSRC, "bindings.hip"),
     ],
     extra_compile_args={

@@ -52,6 +52,20 @@ at::Tensor tbe_grad_per_sample_weights(const at::Tensor& weights,
 at::Tensor bounds_check_indices(at::Tensor indices, const at::Tensor& feat_val_offsets,
                                 const at::Tensor& rows, const at::Tensor& feat_table);
 
+// quant_tbe.hip
+at::Tensor quantize_rowwise_int8(const at::Tensor& weights);
+at::Tensor tbe_forward_pooled_int8(const at::Tensor& qweights,
+                                   const at::Tensor& table_byte_offsets, const at::Tensor& dims,
+                                   const at::Tensor& feat_table, const at::Tensor& d_out_offsets,
+                                   const at::Tensor& indices, const at::Tensor& offsets,
+                                   const at::Tensor& per_sample_weights, int64_t B,
+                                   int64_t total_D, int64_t max_D, bool mean_pool);
+at::Tensor tbe_forward_seq_int8(const at::Tensor& qweights,
+                                const at::Tensor& table_byte_offsets, const at::Tensor& dims,
+                                const at::Tensor& feat_table,
+                                const at::Tensor& feat_val_offsets, const at::Tensor& indices,
+                                int64_t D_out, int64_t max_D);
+
 }  // namespace trec_amd
 
 TORCH_LIBRARY(trec_amd, m) {
@@ -95,6 +109,15 @@ TORCH_LIBRARY(trec_amd, m) {
   m.def(
       "bounds_check_indices(Tensor(a!) indices, Tensor feat_val_offsets, Tensor rows,"
       " Tensor feat_table) -> Tensor");
+  m.def("quantize_rowwise_int8(Tensor weights) -> Tensor");
+  m.def(
+      "tbe_forward_pooled_int8(Tensor qweights, Tensor table_byte_offsets, Tensor dims,"
+      " Tensor feat_table, Tensor d_out_offsets, Tensor indices, Tensor offsets,"
+      " Tensor per_sample_weights, int B, int total_D, int max_D, bool mean_pool) -> Tensor");
+  m.def(
+      "tbe_forward_seq_int8(Tensor qweights, Tensor table_byte_offsets, Tensor dims,"
+      " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"
+      " -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
@@ -113,6 +136,9 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("tbe_backward_fused", trec_amd::tbe_backward_fused);
   m.impl("tbe_grad_per_sample_weights", trec_amd::tbe_grad_per_sample_weights);
   m.impl("bounds_check_indices", trec_amd::bounds_check_indices);
+  m.impl("quantize_rowwise_int8", trec_amd::quantize_rowwise_int8);
+  m.impl("tbe_forward_pooled_int8", trec_amd::tbe_forward_pooled_int8);
+  m.impl("tbe_forward_seq_int8", trec_amd::tbe_forward_seq_int8);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
