@@ -245,3 +245,36 @@ class TestTBEGpu:
             atol=1e-4,
             rtol=1e-3,
         )
+
+
+@pytest.mark.gpu
+class TestTBEUvm:
+    def test_uvm_matches_device(self):
+        """MANAGED (pinned-host) weights produce identical results to
+        HBM-resident weights (config #4 host-spill path)."""
+        from torchrec_amd.ops.tbe import EmbeddingLocation
+
+        specs = [("t0", 200, 128), ("t1", 64, 64)]
+        torch.manual_seed(0)
+        dev = TableBatchedEmbeddingBags(
+            specs, optimizer="rowwise_adagrad", learning_rate=0.05,
+            device=torch.device("cuda"),
+        )
+        uvm = TableBatchedEmbeddingBags(
+            specs, optimizer="rowwise_adagrad", learning_rate=0.05,
+            device=torch.device("cuda"), location=EmbeddingLocation.MANAGED,
+        )
+        assert not uvm.weights.is_cuda and uvm.weights.is_pinned()
+        uvm.weights.copy_(dev.weights.cpu())
+        for step in range(3):
+            indices, offsets = make_inputs(specs, B=8, L=4, seed=step, device="cuda")
+            out_d = dev(indices, offsets)
+            out_u = uvm(indices, offsets)
+            grad = torch.randn_like(out_d)
+            out_d.backward(grad)
+            out_u.backward(grad)
+        torch.cuda.synchronize()
+        assert torch.allclose(out_u, out_d, atol=1e-5, rtol=1e-5)
+        for wd, wu in zip(dev.split_embedding_weights(), uvm.split_embedding_weights()):
+            assert torch.allclose(wu, wd.cpu(), atol=1e-5, rtol=1e-4)
+        assert torch.allclose(uvm.momentum, dev.momentum.cpu(), atol=1e-6)

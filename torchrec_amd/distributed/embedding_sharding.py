@@ -116,6 +116,13 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
             )
             specs = [(t.name, t.local_rows, t.local_dim) for t in group]
             feature_table_map = [i for i, t in enumerate(group) for _ in t.feature_names]
+            from torchrec_amd.ops.tbe import EmbeddingLocation
+
+            location = (
+                EmbeddingLocation.MANAGED
+                if group[0].kernel == EmbeddingComputeKernel.FUSED_UVM.value
+                else EmbeddingLocation.DEVICE
+            )
             tbe = TableBatchedEmbeddingBags(
                 specs,
                 feature_table_map=feature_table_map,
@@ -124,6 +131,7 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                 learning_rate=fused_params.get("learning_rate", 0.01),
                 eps=fused_params.get("eps", 1.0e-8),
                 device=device,
+                location=location,
             )
             self._emb_modules.append(tbe)
             nf = sum(len(t.feature_names) for t in group)

@@ -118,6 +118,14 @@ class EmbeddingEnumerator:
                         if cons and cons.compute_kernels
                         else sharder.compute_kernels(st, self._topology.compute_device)
                     )
+                    if (
+                        self._topology.compute_device == "cuda"
+                        and st != ShardingType.DATA_PARALLEL.value
+                        and EmbeddingComputeKernel.FUSED.value in kernels
+                        and EmbeddingComputeKernel.FUSED_UVM.value not in kernels
+                    ):
+                        # host-DRAM spill variant for tables beyond HBM
+                        kernels = kernels + [EmbeddingComputeKernel.FUSED_UVM.value]
                     for kernel in kernels:
                         shards = [
                             Shard(size=list(sz), offset=list(off))
@@ -155,12 +163,16 @@ class EmbeddingEnumerator:
             if opt.compute_kernel == EmbeddingComputeKernel.DENSE.value
             else 1.0 / D  # rowwise adagrad: one momentum scalar per row
         )
+        is_uvm = opt.compute_kernel == EmbeddingComputeKernel.FUSED_UVM.value
         for shard in opt.shards:
             rows, cols = shard.size
             # storage: weights + optimizer + a slice of activation/grad buffers
             weight_bytes = int(rows * cols * elem * (1 + opt_mult))
             act_bytes = int(B * W * cols * elem * 4)
-            shard.storage = Storage(hbm=weight_bytes + act_bytes, ddr=0)
+            if is_uvm:
+                shard.storage = Storage(hbm=act_bytes, ddr=weight_bytes)
+            else:
+                shard.storage = Storage(hbm=weight_bytes + act_bytes, ddr=0)
             # perf: bytes moved / bandwidth
             if opt.sharding_type == ShardingType.DATA_PARALLEL.value:
                 local_B = B
@@ -180,7 +192,8 @@ class EmbeddingEnumerator:
                 local_B = B * W
                 fwd_bytes = local_B * pooling * cols * elem
                 comms = local_B * cols * elem / topo.intra_host_bw if W > 1 else 0.0
-            fwd_compute = fwd_bytes / topo.hbm_mem_bw
+            mem_bw = topo.ddr_mem_bw if is_uvm else topo.hbm_mem_bw
+            fwd_compute = fwd_bytes / mem_bw
             shard.perf = Perf(
                 fwd_compute=fwd_compute,
                 fwd_comms=comms,
@@ -292,6 +305,45 @@ class UniformProposer(GreedyProposer):
         return proposals
 
 
+class UVMSpillProposer(GreedyProposer):
+    """Best DEVICE-kernel plan, spilling the k largest tables to host DRAM
+    (FUSED_UVM) — covers models beyond 288 GB HBM (reference
+    EmbeddingOffloadScaleupProposer role, proposers.py:471)."""
+
+    def propose(self, options: List[ShardingOption]) -> List[List[ShardingOption]]:
+        by_table: Dict[Tuple[str, str], List[ShardingOption]] = {}
+        for o in options:
+            by_table.setdefault((o.module_fqn, o.name), []).append(o)
+        # order tables by size desc
+        sizes = {
+            k: max(o.config.num_embeddings * o.config.embedding_dim for o in v)
+            for k, v in by_table.items()
+        }
+        ordered = sorted(by_table.keys(), key=lambda k: -sizes[k])
+        proposals = []
+        k = 1
+        while k <= len(ordered):
+            spill = set(ordered[:k])
+            prop = []
+            ok = True
+            for key, opts in by_table.items():
+                want_uvm = key in spill
+                cands = [
+                    o
+                    for o in opts
+                    if (o.compute_kernel == EmbeddingComputeKernel.FUSED_UVM.value)
+                    == want_uvm
+                ]
+                if not cands:
+                    ok = False
+                    break
+                prop.append(min(cands, key=lambda o: o.total_perf))
+            if ok:
+                proposals.append([self._clone(o) for o in prop])
+            k *= 2
+        return proposals
+
+
 class EmbeddingShardingPlanner(ShardingPlanner):
     """Reference parity: planners.py:668."""
 
@@ -311,7 +363,7 @@ class EmbeddingShardingPlanner(ShardingPlanner):
         self._topology = topology
         self._enumerator = EmbeddingEnumerator(topology, constraints)
         self._partitioner = GreedyPerfPartitioner()
-        self._proposers = [GreedyProposer(), UniformProposer()]
+        self._proposers = [GreedyProposer(), UniformProposer(), UVMSpillProposer()]
 
     def plan(
         self, module: nn.Module, sharders: List[ModuleSharder[nn.Module]]

@@ -29,6 +29,22 @@
 
 namespace trec_amd {
 
+
+// ---------------------------------------------------------------------------
+// UVM support: weights may live in pinned host memory (EmbeddingLocation::
+// MANAGED) — kernels read/write it over PCIe via the device-visible alias.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+static T* uvm_ptr(const at::Tensor& t) {
+  if (t.numel() == 0) return nullptr;
+  if (t.is_cuda()) return t.data_ptr<T>();
+  TORCH_CHECK(t.is_pinned(), "TBE host-resident tensors must be pinned (UVM/MANAGED)");
+  void* dp = nullptr;
+  TREC_HIP_CHECK(hipHostGetDevicePointer(&dp, t.data_ptr(), 0));
+  return static_cast<T*>(dp);
+}
+
 static inline hipStream_t tbe_stream() {
   return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
@@ -98,7 +114,7 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
 
 #define TBE_FWD_LAUNCH(LPS, CHUNKS)                                                      \
   hipLaunchKernelGGL((tbe_fwd_pooled_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads), \
-                     0, stream, weights.data_ptr<float>(),                                \
+                     0, stream, uvm_ptr<float>(weights),                                  \
                      table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),    \
                      feat_table.data_ptr<int32_t>(), d_out_offsets.data_ptr<int64_t>(),   \
                      indices.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(),            \
@@ -109,10 +125,10 @@ at::Tensor tbe_forward_pooled(
     const at::Tensor& feat_table, const at::Tensor& d_out_offsets, const at::Tensor& indices,
     const at::Tensor& offsets, const at::Tensor& per_sample_weights, int64_t B,
     int64_t total_D, int64_t max_D, bool mean_pool) {
-  TORCH_CHECK(weights.is_cuda() && weights.scalar_type() == at::kFloat);
+  TORCH_CHECK(weights.scalar_type() == at::kFloat);
   TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048, "TBE: dims must be %4==0 and <=2048");
   int F = feat_table.numel();
-  auto out = at::empty({B, total_D}, weights.options());
+  auto out = at::empty({B, total_D}, indices.options().dtype(at::kFloat));
   if (B == 0 || F == 0) return out;
   const float* psw_ptr =
       per_sample_weights.numel() > 0 ? per_sample_weights.data_ptr<float>() : nullptr;
@@ -172,9 +188,9 @@ at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_el
                            const at::Tensor& dims, const at::Tensor& feat_table,
                            const at::Tensor& feat_val_offsets, const at::Tensor& indices,
                            int64_t D_out, int64_t max_D) {
-  TORCH_CHECK(weights.is_cuda() && max_D % 4 == 0 && max_D <= 2048);
+  TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048);
   int64_t N = indices.numel();
-  auto out = at::empty({N, D_out}, weights.options());
+  auto out = at::empty({N, D_out}, indices.options().dtype(at::kFloat));
   if (N == 0) return out;
   int F = feat_table.numel();
   auto stream = tbe_stream();
@@ -183,7 +199,7 @@ at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_el
   int grid = grid_for(N * lps, kBlockThreads);
 #define TBE_SEQ_LAUNCH(LPS, CHUNKS)                                                        \
   hipLaunchKernelGGL((tbe_fwd_seq_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads), 0, \
-                     stream, weights.data_ptr<float>(),                                    \
+                     stream, uvm_ptr<float>(weights),                                      \
                      table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),     \
                      feat_table.data_ptr<int32_t>(), feat_val_offsets.data_ptr<int64_t>(), \
                      indices.data_ptr<int64_t>(), F, N, D_out, out.data_ptr<float>())
@@ -519,7 +535,7 @@ void tbe_backward_fused(
     const at::Tensor& table_row_offsets, const at::Tensor& table_elem_offsets,
     const at::Tensor& dims, int64_t max_D, double lr, double eps, int64_t mode,
     at::Tensor grad_weights) {
-  TORCH_CHECK(weights.is_cuda() && grad.scalar_type() == at::kFloat);
+  TORCH_CHECK(grad.scalar_type() == at::kFloat);
   int64_t n = sorted_linear.numel();
   if (n == 0) return;
   int T = table_elem_offsets.numel();
@@ -534,7 +550,7 @@ void tbe_backward_fused(
   constexpr int kChunkSize = 32;
   auto [chunk_offsets, total_chunks] = tbe_backward_chunk_prep(seg_offsets, num_runs, kChunkSize);
   int64_t max_chunks = 2 * ((n + kChunkSize - 1) / kChunkSize) + 2;
-  auto scratch = at::empty({max_chunks * max_D}, weights.options());
+  auto scratch = at::empty({max_chunks * max_D}, grad.options());
   int grid_long = grid_for(max_chunks * lps, kBlockThreads);
 
 #define TBE_BWD_LAUNCH(LPS, CHUNKS)                                                          \
@@ -548,8 +564,8 @@ void tbe_backward_fused(
                        table_row_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,   \
                        kChunkSize, max_D, scratch.data_ptr<float>());                        \
     hipLaunchKernelGGL((tbe_bwd_fused_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads), \
-                       0, stream, weights.data_ptr<float>(),                                 \
-                       momentum.numel() ? momentum.data_ptr<float>() : nullptr,              \
+                       0, stream, uvm_ptr<float>(weights),                                   \
+                       uvm_ptr<float>(momentum),                                             \
                        grad.data_ptr<float>(), grad.size(1),                                 \
                        sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
                        seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
@@ -601,12 +617,12 @@ at::Tensor tbe_grad_per_sample_weights(
     const at::Tensor& grad, const at::Tensor& indices, const at::Tensor& pos_row,
     const at::Tensor& pos_col, const at::Tensor& pos_table, int64_t max_D) {
   int64_t N = indices.numel();
-  auto out = at::empty({N}, weights.options());
+  auto out = at::empty({N}, indices.options().dtype(at::kFloat));
   if (N == 0) return out;
   auto stream = tbe_stream();
   int grid = grid_for(N * 16, kBlockThreads);
   hipLaunchKernelGGL((tbe_grad_psw_kernel<16>), dim3(grid), dim3(kBlockThreads), 0, stream,
-                     weights.data_ptr<float>(), table_elem_offsets.data_ptr<int64_t>(),
+                     uvm_ptr<float>(weights), table_elem_offsets.data_ptr<int64_t>(),
                      dims.data_ptr<int32_t>(), grad.data_ptr<float>(), grad.size(1),
                      indices.data_ptr<int64_t>(), pos_row.data_ptr<int32_t>(),
                      pos_col.data_ptr<int64_t>(), pos_table.data_ptr<int32_t>(), N,

@@ -138,9 +138,15 @@ class TableBatchedEmbeddingBags(nn.Module):
         device: Optional[torch.device] = None,
         init_min: float = -0.01,
         init_max: float = 0.01,
+        location: EmbeddingLocation = EmbeddingLocation.DEVICE,
     ) -> None:
         super().__init__()
         device = device or torch.device("cpu")
+        self.location = location
+        # MANAGED: weights live in pinned host DRAM, addressed by the HIP
+        # kernels over PCIe (reference: FBGEMM EmbeddingLocation.MANAGED)
+        self._uvm = location == EmbeddingLocation.MANAGED and device.type == "cuda"
+        weights_device = torch.device("cpu") if self._uvm else device
         self.pooling_mode = pooling_mode
         self.optimizer = _OPT_NAMES[optimizer]
         self.learning_rate = learning_rate
@@ -172,19 +178,22 @@ class TableBatchedEmbeddingBags(nn.Module):
         self._total_D = d_out[-1]
         self._max_D = max(dims) if dims else 0
 
-        weights = torch.empty(self._total_elems, dtype=torch.float32, device=device)
-        if device.type != "meta":
+        weights = torch.empty(self._total_elems, dtype=torch.float32, device=weights_device)
+        if weights_device.type != "meta":
             weights.uniform_(init_min, init_max)
+        if self._uvm:
+            weights = weights.pin_memory()
         if self.optimizer == OPT_DENSE:
             self.weights = nn.Parameter(weights)
         else:
             self.register_buffer("weights", weights)
         if self.optimizer == OPT_ROWWISE_ADAGRAD:
-            self.register_buffer(
-                "momentum", torch.zeros(self._total_rows, dtype=torch.float32, device=device)
-            )
+            mom = torch.zeros(self._total_rows, dtype=torch.float32, device=weights_device)
+            if self._uvm:
+                mom = mom.pin_memory()
+            self.register_buffer("momentum", mom)
         else:
-            self.register_buffer("momentum", torch.empty(0, device=device))
+            self.register_buffer("momentum", torch.empty(0, device=weights_device))
 
         def reg(name: str, t: torch.Tensor) -> None:
             self.register_buffer(name, t.to(device), persistent=False)
@@ -220,7 +229,7 @@ class TableBatchedEmbeddingBags(nn.Module):
         per_sample_weights: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         B = (offsets.numel() - 1) // self._num_features
-        if not self.weights.is_cuda:
+        if not indices.is_cuda:
             return self._forward_cpu(indices, offsets, per_sample_weights, B)
         ops.hip_ops()  # fail loudly if the extension is missing on GPU
         if self.optimizer == OPT_DENSE:
@@ -544,7 +553,7 @@ class TableBatchedEmbeddings(nn.Module):
     def forward(self, indices: torch.Tensor, offsets: torch.Tensor) -> torch.Tensor:
         host = self._bags
         B = (offsets.numel() - 1) // host._num_features
-        if not host.weights.is_cuda:
+        if not indices.is_cuda:
             return _TBESeqCpuFunction.apply(host._dummy, host, indices, offsets, B)
         ops.hip_ops()
         # feature value-range offsets: offsets at bag boundaries f*B
