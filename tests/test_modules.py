@@ -165,3 +165,39 @@ class TestDLRM:
         )
         logits = model(batch.dense_features, batch.sparse_features)
         assert logits.shape == (4, 1)
+
+
+class TestFeatureProcessors:
+    def test_position_weighted_module(self):
+        from torchrec_amd.modules.feature_processor import PositionWeightedModule
+        from torchrec_amd.sparse.jagged_tensor import JaggedTensor
+
+        pw = PositionWeightedModule(max_feature_length=4)
+        with torch.no_grad():
+            pw.position_weight.copy_(torch.tensor([1.0, 2.0, 3.0, 4.0]))
+        jt = JaggedTensor(values=torch.tensor([10, 11, 12]), lengths=torch.tensor([2, 1]))
+        out = pw(jt)
+        assert out.weights().tolist() == [1.0, 2.0, 1.0]
+
+    def test_fp_ebc(self):
+        from torchrec_amd.modules.feature_processor import (
+            FeatureProcessedEmbeddingBagCollection,
+            PositionWeightedModuleCollection,
+        )
+
+        tables = [
+            EmbeddingBagConfig(num_embeddings=10, embedding_dim=4, name="t1", feature_names=["f1"]),
+        ]
+        ebc = EmbeddingBagCollection(tables=tables, is_weighted=True)
+        fp = PositionWeightedModuleCollection({"f1": 8})
+        fp_ebc = FeatureProcessedEmbeddingBagCollection(ebc, fp)
+        kjt = KeyedJaggedTensor(
+            keys=["f1"], values=torch.tensor([1, 2]), lengths=torch.tensor([2]), stride=1
+        )
+        kt = fp_ebc(kjt)
+        w = ebc.embedding_bags["t1"].weight
+        # default position weights are all 1 -> plain sum
+        assert torch.allclose(kt["f1"][0], w[1] + w[2])
+        # gradient flows to position weights
+        kt.values().sum().backward()
+        assert fp.position_weights["f1"].grad is not None
