@@ -109,3 +109,37 @@ def test_permute_pooled_embs():
     ref.backward(g)
     got.backward(g.cuda())
     assert torch.allclose(vals.grad, vals_g.grad.cpu())
+
+
+def test_fused_interaction_matches_eager():
+    torch.manual_seed(0)
+    B, F, D = 32, 5, 16
+    dense = torch.randn(B, D, requires_grad=True)
+    sparse = torch.randn(B, F, D, requires_grad=True)
+    combined = torch.cat([dense.unsqueeze(1), sparse], dim=1)
+    inter = torch.bmm(combined, combined.transpose(1, 2))
+    tri = torch.triu_indices(F + 1, F + 1, offset=1)
+    ref = torch.cat([dense, inter[:, tri[0], tri[1]]], dim=1)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+
+    d_g = dense.detach().clone().cuda().requires_grad_(True)
+    s_g = sparse.detach().clone().cuda().requires_grad_(True)
+    out = ops.fused_interaction(d_g, s_g)
+    torch.cuda.synchronize()
+    assert torch.allclose(out.cpu(), ref.detach(), atol=1e-4, rtol=1e-4)
+    out.backward(g.cuda())
+    torch.cuda.synchronize()
+    assert torch.allclose(d_g.grad.cpu(), dense.grad, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(s_g.grad.cpu(), sparse.grad, atol=1e-4, rtol=1e-4)
+
+
+def test_fused_interaction_bf16_inputs():
+    torch.manual_seed(1)
+    dense = torch.randn(8, 8, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+    sparse = torch.randn(8, 3, 8, device="cuda", requires_grad=True)
+    out = ops.fused_interaction(dense, sparse)
+    out.sum().backward()
+    torch.cuda.synchronize()
+    assert dense.grad.dtype == torch.bfloat16
+    assert sparse.grad.dtype == torch.float32

@@ -73,7 +73,11 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
         )
         lengths2d = self._out_lengths.view(W * F, self._B)
         pl, pv, pw = ops.permute_2d_sparse_data(
-            perm, lengths2d, self._out_values, self._out_weights
+            perm,
+            lengths2d,
+            self._out_values,
+            self._out_weights,
+            permuted_lengths_sum=int(self._out_values.numel()),
         )
         kjt = KeyedJaggedTensor(
             keys=self._keys,

@@ -292,7 +292,10 @@ class ShardedEmbeddingCollection(nn.Module):
             )
             positions = torch.arange(rows.shape[0], device=rows.device)
             _, perm_positions, _ = ops.permute_2d_sparse_data(
-                perm, seg_counts.reshape(-1, 1), positions
+                perm,
+                seg_counts.reshape(-1, 1),
+                positions,
+                permuted_lengths_sum=int(rows.shape[0]),
             )
             rows_rank_major = rows.index_select(0, perm_positions)
             in_splits, out_splits = kjt._dist_value_splits

@@ -250,7 +250,10 @@ class ShardedEmbeddingBagCollection(nn.Module):
             features = features.permute(
                 self._features_order, self._features_order_tensor
             )
-        feature_kjts = features.split(self._feature_splits)
+        if len(self._shardings) == 1:
+            feature_kjts = [features]  # avoids split()'s host sync on offsets
+        else:
+            feature_kjts = features.split(self._feature_splits)
         # mean divisors from pre-dist lengths
         for si, mean_feats in enumerate(self._mean_features_per_sharding):
             if mean_feats:

@@ -90,9 +90,12 @@ class InteractionArch(nn.Module):
     def forward(self, dense_features: torch.Tensor, sparse_features: torch.Tensor) -> torch.Tensor:
         if self.F <= 0:
             return dense_features
+        if dense_features.is_cuda:
+            from torchrec_amd import ops as _ops
+
+            return _ops.fused_interaction(dense_features, sparse_features)
         B = dense_features.shape[0]
         combined = torch.cat([dense_features.unsqueeze(1), sparse_features], dim=1)
-        # [B, F+1, F+1] pairwise dots — bmm maps to hipBLASLt batched GEMM
         inter = torch.bmm(combined, combined.transpose(1, 2))
         flat = inter[:, self.triu_indices[0], self.triu_indices[1]]
         return torch.cat([dense_features, flat], dim=1)

@@ -125,7 +125,11 @@ def permute_2d_sparse_data(
     if values.is_cuda:
         hip_ops()
         pl, pv, pw = torch.ops.trec_amd.permute_2d_sparse_data(
-            permute, lengths, values, weights if weights is not None else torch.empty(0)
+            permute,
+            lengths,
+            values,
+            weights if weights is not None else torch.empty(0),
+            -1 if permuted_lengths_sum is None else permuted_lengths_sum,
         )
         return pl, pv, (pw if weights is not None else None)
     K, B = lengths.shape
@@ -373,3 +377,56 @@ def permute_pooled_embs(
     out_dims = [group_dims[int(order[i])] for i in range(order.numel())]
     out_offsets = complete_cumsum(torch.tensor(out_dims, dtype=torch.int64)).to(device)
     return _PermutePooledEmbs.apply(values, in_offsets, out_offsets, order.to(device))
+
+
+# ------------------------------------------------------------------------
+# fused DLRM interaction (pairwise dot + triu + cat), CDNA4 kernel
+# ------------------------------------------------------------------------
+
+_PAIR_TABLES: dict = {}
+
+
+def _pair_tables(F1: int, device: torch.device):
+    key = (F1, device)
+    if key not in _PAIR_TABLES:
+        tri = torch.triu_indices(F1, F1, offset=1)
+        P = tri.shape[1]
+        pi = tri[0].to(torch.int8)
+        pj = tri[1].to(torch.int8)
+        pair_col = torch.full((F1 * F1,), -1, dtype=torch.int32)
+        for p in range(P):
+            i, j = int(tri[0, p]), int(tri[1, p])
+            pair_col[i * F1 + j] = p
+            pair_col[j * F1 + i] = p
+        _PAIR_TABLES[key] = (pi.to(device), pj.to(device), pair_col.to(device))
+    return _PAIR_TABLES[key]
+
+
+class _FusedInteraction(torch.autograd.Function):
+    """out = [dense, triu(T @ T^T)] with T = [dense; sparse] — single kernel
+    each way (csrc/interaction.hip); replaces cat+bmm+gather."""
+
+    @staticmethod
+    def forward(ctx, dense, sparse):  # type: ignore[override]
+        F1 = sparse.shape[1] + 1
+        pi, pj, pair_col = _pair_tables(F1, dense.device)
+        d32 = dense.float()
+        s32 = sparse.float()
+        ctx.save_for_backward(d32, s32, pair_col)
+        ctx.dtypes = (dense.dtype, sparse.dtype)
+        return torch.ops.trec_amd.interaction_forward(d32, s32, pi, pj)
+
+    @staticmethod
+    def backward(ctx, grad_out):  # type: ignore[override]
+        d32, s32, pair_col = ctx.saved_tensors
+        dd, ds = torch.ops.trec_amd.interaction_backward(
+            grad_out.float().contiguous(), d32, s32, pair_col
+        )
+        ddt, dst = ctx.dtypes
+        return dd.to(ddt), ds.to(dst)
+
+
+def fused_interaction(dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+    """GPU: fused kernel; CPU callers should use the eager path."""
+    hip_ops()
+    return _FusedInteraction.apply(dense, sparse)

@@ -11,7 +11,7 @@ at::Tensor complete_cumsum(const at::Tensor& lengths);
 at::Tensor lengths_range(const at::Tensor& offsets);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> permute_2d_sparse_data(
     const at::Tensor& permute, const at::Tensor& lengths, const at::Tensor& values,
-    const at::Tensor& weights);
+    const at::Tensor& weights, int64_t out_size_hint);
 at::Tensor jagged_to_padded_dense(const at::Tensor& values, const at::Tensor& offsets,
                                   int64_t max_length, double padding_value);
 at::Tensor dense_to_jagged(const at::Tensor& dense, const at::Tensor& offsets);
@@ -66,14 +66,22 @@ at::Tensor tbe_forward_seq_int8(const at::Tensor& qweights,
                                 const at::Tensor& feat_val_offsets, const at::Tensor& indices,
                                 int64_t D_out, int64_t max_D);
 
+// interaction.hip
+at::Tensor interaction_forward(const at::Tensor& dense, const at::Tensor& sparse,
+                               const at::Tensor& pi, const at::Tensor& pj);
+std::tuple<at::Tensor, at::Tensor> interaction_backward(const at::Tensor& grad_out,
+                                                        const at::Tensor& dense,
+                                                        const at::Tensor& sparse,
+                                                        const at::Tensor& pair_col);
+
 }  // namespace trec_amd
 
 TORCH_LIBRARY(trec_amd, m) {
   m.def("complete_cumsum(Tensor lengths) -> Tensor");
   m.def("lengths_range(Tensor offsets) -> Tensor");
   m.def(
-      "permute_2d_sparse_data(Tensor permute, Tensor lengths, Tensor values, Tensor weights)"
-      " -> (Tensor, Tensor, Tensor)");
+      "permute_2d_sparse_data(Tensor permute, Tensor lengths, Tensor values, Tensor weights,"
+      " int out_size_hint) -> (Tensor, Tensor, Tensor)");
   m.def(
       "jagged_to_padded_dense(Tensor values, Tensor offsets, int max_length,"
       " float padding_value) -> Tensor");
@@ -118,6 +126,10 @@ TORCH_LIBRARY(trec_amd, m) {
       "tbe_forward_seq_int8(Tensor qweights, Tensor table_byte_offsets, Tensor dims,"
       " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"
       " -> Tensor");
+  m.def("interaction_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
+  m.def(
+      "interaction_backward(Tensor grad_out, Tensor dense, Tensor sparse, Tensor pair_col)"
+      " -> (Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
@@ -139,6 +151,8 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("quantize_rowwise_int8", trec_amd::quantize_rowwise_int8);
   m.impl("tbe_forward_pooled_int8", trec_amd::tbe_forward_pooled_int8);
   m.impl("tbe_forward_seq_int8", trec_amd::tbe_forward_seq_int8);
+  m.impl("interaction_forward", trec_amd::interaction_forward);
+  m.impl("interaction_backward", trec_amd::interaction_backward);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

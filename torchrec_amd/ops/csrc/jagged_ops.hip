@@ -113,7 +113,7 @@ __global__ void permute_bags_kernel(
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> permute_2d_sparse_data(
     const at::Tensor& permute, const at::Tensor& lengths, const at::Tensor& values,
-    const at::Tensor& weights) {
+    const at::Tensor& weights, int64_t out_size_hint) {
   TORCH_CHECK(values.is_cuda() && lengths.dim() == 2);
   int64_t K_in = lengths.size(0), B = lengths.size(1);
   auto perm = permute.to(at::kLong).contiguous();
@@ -123,8 +123,8 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> permute_2d_sparse_data(
   auto out_offsets = complete_cumsum(perm_lengths.reshape({-1}).to(at::kLong).contiguous());
   bool has_w = weights.numel() > 0;
   int64_t n_bags = K_out * B;
-  // output size: device-side tail of out_offsets — one small D2H read
-  int64_t out_n = out_offsets[-1].item<int64_t>();
+  // output size: caller-provided hint avoids a D2H sync; -1 -> read tail
+  int64_t out_n = out_size_hint >= 0 ? out_size_hint : out_offsets[-1].item<int64_t>();
   auto out_values = at::empty({out_n}, values.options());
   auto out_weights = has_w ? at::empty({out_n}, weights.options()) : at::empty({0}, values.options().dtype(at::kFloat));
   if (n_bags > 0 && out_n > 0) {

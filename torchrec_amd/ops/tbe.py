@@ -273,7 +273,9 @@ class TableBatchedEmbeddingBags(nn.Module):
         lengths = offsets[1:] - offsets[:-1]
         FB = lengths.numel()
         bag_ids = torch.repeat_interleave(
-            torch.arange(FB, device=indices.device, dtype=torch.int64), lengths
+            torch.arange(FB, device=indices.device, dtype=torch.int64),
+            lengths,
+            output_size=indices.numel(),  # known host-side: avoids a D2H sync
         )
         f = torch.div(bag_ids, B, rounding_mode="floor")
         b = bag_ids - f * B
@@ -348,7 +350,9 @@ class TableBatchedEmbeddingBags(nn.Module):
         N = indices.numel()
         counts = feat_val_offsets[1:] - feat_val_offsets[:-1]
         f = torch.repeat_interleave(
-            torch.arange(counts.numel(), device=indices.device, dtype=torch.int64), counts
+            torch.arange(counts.numel(), device=indices.device, dtype=torch.int64),
+            counts,
+            output_size=N,
         )
         pos_row = torch.arange(N, device=indices.device, dtype=torch.int32)
         pos_col = torch.zeros(N, device=indices.device, dtype=torch.int64)
