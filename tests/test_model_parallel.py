@@ -192,3 +192,48 @@ def test_sharded_ebc_weighted_tw():
         _run_sharding_test, 2, "gloo", ShardingType.TABLE_WISE.value,
         PoolingType.SUM.value, True,
     )
+
+
+def _run_qcomm_test(rank, world_size):
+    """TW sharding with bf16-compressed pooled a2a stays close to fp32."""
+    from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
+
+    B = 4
+    tables = make_tables()
+    torch.manual_seed(42)
+    model = SparseModel(make_tables())
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(sharding_types=[ShardingType.TABLE_WISE.value])
+            for cfg in tables
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={
+            "optimizer": "rowwise_adagrad",
+            "learning_rate": LR,
+            "qcomms_config": QCommsConfig(forward_precision=CommType.BF16),
+        }
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    golden = _golden(tables, None, world_size)
+    dmp.load_state_dict(
+        {
+            f"sparse.embedding_bags.{cfg.name}.weight": w
+            for cfg, w in zip(tables, golden.split_embedding_weights())
+        },
+        strict=False,
+    )
+    kjt_global = make_global_kjt(tables, B * world_size)
+    kjt_local = kjt_local_slice(kjt_global, rank * B, (rank + 1) * B)
+    vals = dmp(kjt_local).values()
+    expected = golden(kjt_global).values()[rank * B : (rank + 1) * B]
+    torch.testing.assert_close(vals, expected, atol=0.05, rtol=0.05)  # bf16 wire
+
+
+def test_qcomm_bf16_tw():
+    run_multi_process(_run_qcomm_test, 2, "gloo")

@@ -171,6 +171,12 @@ def alltoall_pooled(
     pg: dist.ProcessGroup,
     codec: Any = None,
 ) -> Awaitable[torch.Tensor]:
+    if codec is not None and getattr(codec, "precision", None) is not None:
+        from torchrec_amd.distributed.qcomm_codecs import CommType
+
+        assert codec.precision != CommType.INT8, (
+            "INT8 comm needs per-block scales; use FP16/BF16/FP8 for a2a"
+        )
     """Async pooled-embedding a2a (reference comm_ops.py:508).
 
     ``pooled``: [W*B_local, D_local] laid out source-rank-major.
