@@ -43,6 +43,7 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
         B_local: int,
         in_value_splits: Optional[List[int]] = None,
         out_value_splits: Optional[List[int]] = None,
+        rank_order: Optional[List[int]] = None,
     ) -> None:
         super().__init__()
         self._pg = pg
@@ -56,6 +57,9 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
         # host-side a2a splits; sequence output_dist mirrors them
         self.in_value_splits = in_value_splits or []
         self.out_value_splits = out_value_splits or []
+        # order in which source-rank blocks concatenate into the global batch
+        # (TWRW staggers this so intra-node RS blocks align with cross groups)
+        self._rank_order = rank_order or list(range(self._W))
 
     def _wait_impl(self) -> KeyedJaggedTensor:
         for w in self._works:
@@ -65,9 +69,9 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
         W = self._W
         if F == 0:
             return KeyedJaggedTensor.empty(device=self._out_values.device)
-        # recat (r, f) -> (f, r)
+        # recat (r, f) -> (f, r') with r' in the (possibly staggered) order
         perm = torch.tensor(
-            [r * F + f for f in range(F) for r in range(W)],
+            [r * F + f for f in range(F) for r in self._rank_order],
             dtype=torch.int64,
             device=self._out_values.device,
         )
@@ -104,12 +108,14 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
         splits: List[int],  # features per destination rank (input already ordered)
         keys: List[str],  # my features post-exchange
         stagger: int = 1,
+        rank_order: Optional[List[int]] = None,
     ) -> None:
         super().__init__()
         self._pg = pg
         self._input = input
         self._splits = splits
         self._keys = keys
+        self._rank_order = rank_order
         self._W = dist.get_world_size(pg)
         B = input.stride()
         self._B = B
@@ -187,6 +193,7 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
             B,
             in_value_splits=in_value_splits,
             out_value_splits=out_value_splits,
+            rank_order=self._rank_order,
         )
 
 
@@ -197,11 +204,18 @@ class KJTAllToAll(nn.Module):
     forward(kjt) -> Awaitable[Awaitable[KJT]] (splits phase, tensors phase).
     """
 
-    def __init__(self, pg: dist.ProcessGroup, splits: List[int], stagger: int = 1) -> None:
+    def __init__(
+        self,
+        pg: dist.ProcessGroup,
+        splits: List[int],
+        stagger: int = 1,
+        rank_order: Optional[List[int]] = None,
+    ) -> None:
         super().__init__()
         self._pg = pg
         self._splits = splits
         self._stagger = stagger
+        self._rank_order = rank_order
         self._splits_cumsum = [0]
         for s in splits:
             self._splits_cumsum.append(self._splits_cumsum[-1] + s)
@@ -212,7 +226,7 @@ class KJTAllToAll(nn.Module):
             self._splits_cumsum[rank] : self._splits_cumsum[rank + 1]
         ]
         return KJTAllToAllSplitsAwaitable(
-            self._pg, input, self._splits, local_keys, self._stagger
+            self._pg, input, self._splits, local_keys, self._stagger, self._rank_order
         )
 
 

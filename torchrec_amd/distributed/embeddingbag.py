@@ -54,6 +54,18 @@ def create_sharding(
         return CwPooledEmbeddingSharding(infos, env, device)
     if sharding_type == ShardingType.DATA_PARALLEL.value:
         return DpPooledEmbeddingSharding(infos, env, device)
+    if sharding_type == ShardingType.TABLE_ROW_WISE.value:
+        from torchrec_amd.distributed.sharding.twrw_sharding import (
+            TwRwPooledEmbeddingSharding,
+        )
+
+        return TwRwPooledEmbeddingSharding(infos, env, device)
+    if sharding_type == ShardingType.GRID_SHARD.value:
+        from torchrec_amd.distributed.sharding.grid_sharding import (
+            GridPooledEmbeddingSharding,
+        )
+
+        return GridPooledEmbeddingSharding(infos, env, device)
     raise ValueError(f"unsupported sharding type {sharding_type}")
 
 

@@ -48,12 +48,41 @@ def _shard_bytes(rows: int, cols: int, elem: int = 4, opt_mult: float = 1.0 / 12
 
 
 def calculate_shards(
-    rows: int, cols: int, sharding_type: str, world_size: int, min_partition: int = 64
+    rows: int,
+    cols: int,
+    sharding_type: str,
+    world_size: int,
+    min_partition: int = 64,
+    local_size: int = 8,
 ) -> List[Tuple[List[int], List[int]]]:
     """(sizes, offsets) per shard (reference sharding_plan.py:94)."""
     if sharding_type in (ShardingType.TABLE_WISE.value, ShardingType.DATA_PARALLEL.value):
         n = 1 if sharding_type == ShardingType.TABLE_WISE.value else world_size
         return [([rows, cols], [0, 0]) for _ in range(n)]
+    if sharding_type == ShardingType.GRID_SHARD.value:
+        from torchrec_amd.distributed.sharding.cw_sharding import cw_shard_dims
+
+        L = min(local_size, world_size)
+        NN = max(1, world_size // L)
+        col_dims = cw_shard_dims(cols, NN)
+        out = []
+        for n in range(NN):
+            col_off = sum(col_dims[:n])
+            block = (rows + L - 1) // L
+            for r in range(L):
+                lo = min(r * block, rows)
+                hi = min((r + 1) * block, rows)
+                out.append(([hi - lo, col_dims[n]], [lo, col_off]))
+        return out
+    if sharding_type == ShardingType.TABLE_ROW_WISE.value:
+        L = min(local_size, world_size)
+        block = (rows + L - 1) // L
+        out = []
+        for r in range(L):
+            lo = min(r * block, rows)
+            hi = min((r + 1) * block, rows)
+            out.append(([hi - lo, cols], [lo, 0]))
+        return out
     if sharding_type == ShardingType.ROW_WISE.value:
         block = (rows + world_size - 1) // world_size
         out = []
@@ -135,6 +164,7 @@ class EmbeddingEnumerator:
                                 st,
                                 W,
                                 min_partition=(cons.min_partition if cons and cons.min_partition else 64),
+                                local_size=self._topology.local_world_size,
                             )
                         ]
                         opt = ShardingOption(
@@ -217,6 +247,7 @@ class GreedyPerfPartitioner:
             if opt.sharding_type in (
                 ShardingType.ROW_WISE.value,
                 ShardingType.DATA_PARALLEL.value,
+                ShardingType.GRID_SHARD.value,
             ):
                 for r, shard in enumerate(opt.shards):
                     shard.rank = r
@@ -225,6 +256,26 @@ class GreedyPerfPartitioner:
                     dev.perf = dev.perf + shard.perf
                     if dev.storage.hbm < 0:
                         raise PlannerError(f"OOM on rank {r} for {opt.name}")
+        # TWRW: place the whole row-shard group on one node (greedy by load)
+        L = topology.local_world_size
+        for opt in proposal:
+            if opt.sharding_type != ShardingType.TABLE_ROW_WISE.value:
+                continue
+            n_nodes = max(1, topology.world_size // L)
+            best_node, best_load = None, None
+            for node in range(n_nodes):
+                devs = devices[node * L : node * L + len(opt.shards)]
+                if all(s.storage.fits_in(d.storage) for s, d in zip(opt.shards, devs)):
+                    load = max(d.perf.total for d in devs)
+                    if best_load is None or load < best_load:
+                        best_node, best_load = node, load
+            if best_node is None:
+                raise PlannerError(f"no node fits TWRW table {opt.name}")
+            for i, shard in enumerate(opt.shards):
+                dev = devices[best_node * L + i]
+                shard.rank = dev.rank
+                dev.storage = dev.storage - shard.storage
+                dev.perf = dev.perf + shard.perf
         # greedy for TW/CW: biggest perf first onto least-loaded feasible device
         movable = [
             (shard, opt)
