@@ -1,0 +1,78 @@
+"""DMPCollection (2D: MP groups x DP replicas) tests (reference:
+torchrec/distributed/tests/test_dmp_collection.py pattern). Gloo world 4 =
+2 sharding groups x 2 replicas."""
+
+import torch
+import torch.distributed as dist
+
+from tests.dist_utils import run_multi_process
+from tests.test_model_parallel import (
+    SparseModel,
+    _golden,
+    kjt_local_slice,
+    make_global_kjt,
+    make_tables,
+    LR,
+)
+from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+from torchrec_amd.distributed.model_parallel import DMPCollection
+from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+from torchrec_amd.distributed.planner.types import ParameterConstraints, Topology
+from torchrec_amd.distributed.types import ShardingType
+
+
+def _run_2d_test(rank, world_size):
+    S = 2  # sharding group size
+    B = 4
+    tables = make_tables()
+    torch.manual_seed(42)
+    model = SparseModel(make_tables())
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=S, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(sharding_types=[ShardingType.TABLE_WISE.value])
+            for cfg in tables
+        },
+    )
+    plan = planner.plan(model, [sharder])  # identical on every rank (seeded)
+    dmp2d = DMPCollection(
+        model, sharding_group_size=S, plan=plan, sharders=[sharder], sync_interval=1
+    )
+
+    golden = _golden(tables, None, world_size)
+    dmp2d.load_state_dict(
+        {
+            f"sparse.embedding_bags.{cfg.name}.weight": w
+            for cfg, w in zip(tables, golden.split_embedding_weights())
+        },
+        strict=False,
+    )
+
+    # per-replica-group global batch: group g sees batches [g] (same within group)
+    group = rank // S
+    rank_in_group = rank % S
+    kjt_group = make_global_kjt(tables, B * S, seed=100 + group)
+    kjt_local = kjt_local_slice(kjt_group, rank_in_group * B, (rank_in_group + 1) * B)
+    kt = dmp2d(kjt_local)
+    vals = kt.values()
+    expected = golden(kjt_group).values()[rank_in_group * B : (rank_in_group + 1) * B]
+    torch.testing.assert_close(vals, expected, atol=1e-5, rtol=1e-5)
+
+    # train one step with group-specific data, then sync replicas
+    vals.sum().backward()
+    dmp2d.sync()
+
+    # after sync, replica peers hold identical shard contents
+    for sharded in dmp2d.sharded_modules().values():
+        for tbe in sharded.tbes():
+            w = tbe.weights
+            peers = [torch.empty_like(w) for _ in range(2)]
+            dist.all_gather(peers, w.detach(), group=dmp2d._replica_pg)
+            torch.testing.assert_close(peers[0], peers[1], atol=1e-6, rtol=1e-6)
+
+
+def test_dmp_collection_2d():
+    run_multi_process(_run_2d_test, 4, "gloo")

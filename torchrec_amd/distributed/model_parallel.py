@@ -44,6 +44,7 @@ class DistributedModelParallel(nn.Module):
         sharders: Optional[List[ModuleSharder[nn.Module]]] = None,
         init_data_parallel: bool = True,
         init_parameters: bool = True,
+        data_parallel_pg: Optional[dist.ProcessGroup] = None,
     ) -> None:
         super().__init__()
         torch._C._log_api_usage_once("torchrec_amd.DistributedModelParallel")
@@ -73,8 +74,14 @@ class DistributedModelParallel(nn.Module):
 
         self._optim = self._init_optim()
 
+        self._data_parallel_pg = data_parallel_pg
         self._ddp_wrapped = False
-        if init_data_parallel and env.world_size > 1:
+        dp_world = (
+            dist.get_world_size(data_parallel_pg)
+            if data_parallel_pg is not None
+            else env.world_size
+        )
+        if init_data_parallel and dp_world > 1:
             self.init_data_parallel()
 
     # -- sharding ----------------------------------------------------------
@@ -111,7 +118,8 @@ class DistributedModelParallel(nn.Module):
     def init_data_parallel(self) -> None:
         """DDP over dense params; sharded-module params/buffers ignored
         (reference model_parallel.py:142-254 DefaultDataParallelWrapper)."""
-        if self._ddp_wrapped or self._env.process_group is None:
+        ddp_pg = self._data_parallel_pg or self._env.process_group
+        if self._ddp_wrapped or ddp_pg is None:
             return
         ignore: List[str] = []
         for fqn, sharded in self._sharded_modules.items():
@@ -137,7 +145,7 @@ class DistributedModelParallel(nn.Module):
         self._dmp_wrapped_module = nn.parallel.DistributedDataParallel(
             self._dmp_wrapped_module,
             device_ids=[self.device] if self.device.type == "cuda" else None,
-            process_group=self._env.process_group,
+            process_group=ddp_pg,
             gradient_as_bucket_view=True,
             static_graph=False,
         )
@@ -195,3 +203,115 @@ def _cpu_topology(env: ShardingEnv):
     return Topology(
         world_size=env.world_size, compute_device="cpu", hbm_cap=constants.DDR_CAP
     )
+
+
+class DMPCollection(nn.Module):
+    """2D parallelism: model-parallel sharding groups x data-parallel replicas.
+
+    Reference parity: model_parallel.py:1028 — shards within
+    ``sharding_group_size`` consecutive ranks, replicates across groups;
+    ``sync()`` = allreduce(AVG) of sharded weights + optimizer states over the
+    replica group (reference :1402); dense params ride global DDP.
+    """
+
+    def __init__(
+        self,
+        module: nn.Module,
+        sharding_group_size: int,
+        device: Optional[torch.device] = None,
+        plan: Optional[ShardingPlan] = None,
+        sharders: Optional[List[ModuleSharder[nn.Module]]] = None,
+        sync_interval: int = 1,
+    ) -> None:
+        super().__init__()
+        assert dist.is_initialized(), "DMPCollection needs torch.distributed"
+        world = dist.get_world_size()
+        rank = dist.get_rank()
+        S = sharding_group_size
+        assert world % S == 0, "world_size must be a multiple of sharding_group_size"
+        self._num_groups = world // S
+        my_group = rank // S
+        my_replica_slot = rank % S
+        # build both families of groups on every rank (collective requirement)
+        self._sharding_pg = None
+        self._replica_pg = None
+        for g in range(self._num_groups):
+            ranks = list(range(g * S, (g + 1) * S))
+            pg = dist.new_group(ranks=ranks)
+            if g == my_group:
+                self._sharding_pg = pg
+        for slot in range(S):
+            ranks = list(range(slot, world, S))
+            pg = dist.new_group(ranks=ranks)
+            if slot == my_replica_slot:
+                self._replica_pg = pg
+        env = ShardingEnv(S, rank % S, self._sharding_pg)
+        if plan is None and sharders is None:
+            sharders = get_default_sharders()
+        if plan is None:
+            planner = EmbeddingShardingPlanner(
+                topology=None if (device or torch.device("cpu")).type == "cuda" else _cpu_topology(env)
+            )
+            plan = planner.collective_plan(module, sharders, self._sharding_pg)
+        self._dmp = DistributedModelParallel(
+            module,
+            env=env,
+            device=device,
+            plan=plan,
+            sharders=sharders,
+            data_parallel_pg=dist.group.WORLD,
+        )
+        self._sync_interval = sync_interval
+        self._step = 0
+
+    @property
+    def fused_optimizer(self) -> KeyedOptimizer:
+        return self._dmp.fused_optimizer
+
+    @property
+    def plan(self) -> ShardingPlan:
+        return self._dmp.plan
+
+    @property
+    def module(self) -> nn.Module:
+        return self._dmp.module
+
+    def sharded_modules(self):
+        return self._dmp.sharded_modules()
+
+    def state_dict(self, *args, **kwargs):
+        return self._dmp.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, *args, **kwargs):
+        return self._dmp.load_state_dict(*args, **kwargs)
+
+    def forward(self, *args, **kwargs):
+        return self._dmp(*args, **kwargs)
+
+    def maybe_sync(self) -> None:
+        """Call once per step; syncs replicas every ``sync_interval`` steps."""
+        self._step += 1
+        if self._step % self._sync_interval == 0:
+            self.sync()
+
+    @torch.no_grad()
+    def sync(self, include_optimizer_state: bool = True) -> None:
+        """Average sharded weights (and fused-optimizer state) across the
+        replica group (reference model_parallel.py:1402)."""
+        R = dist.get_world_size(self._replica_pg)
+        if R <= 1:
+            return
+        tensors: List[torch.Tensor] = []
+        for sharded in self._dmp.sharded_modules().values():
+            for tbe in getattr(sharded, "tbes", lambda: [])():
+                inner = getattr(tbe, "_bags", tbe)
+                w = inner.weights
+                tensors.append(w.data if isinstance(w, nn.Parameter) else w)
+                if include_optimizer_state and inner.momentum.numel():
+                    tensors.append(inner.momentum)
+        works = []
+        for t in tensors:
+            works.append(dist.all_reduce(t, group=self._replica_pg, async_op=True))
+        for w, t in zip(works, tensors):
+            w.wait()
+            t.div_(R)
