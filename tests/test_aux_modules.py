@@ -1,0 +1,116 @@
+"""Pools, towers, delta tracker tests."""
+
+import torch
+
+from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+from torchrec_amd.modules.embedding_tower import EmbeddingTower, EmbeddingTowerCollection
+from torchrec_amd.modules.object_pools import KeyedJaggedTensorPool, TensorPool
+from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+
+class TestTensorPool:
+    def test_roundtrip(self):
+        pool = TensorPool(pool_size=16, dim=4)
+        ids = torch.tensor([3, 7, 3])
+        vals = torch.randn(3, 4)
+        pool.update(ids, vals)
+        out = pool(ids)
+        torch.testing.assert_close(out[1], vals[1])
+        assert pool(torch.tensor([0])).abs().sum() == 0
+
+
+class TestKJTPool:
+    def test_roundtrip(self):
+        pool = KeyedJaggedTensorPool(pool_size=8, feature_max_lengths={"a": 3, "b": 2})
+        kjt = KeyedJaggedTensor(
+            keys=["a", "b"],
+            values=torch.tensor([1, 2, 3, 9, 8]),
+            lengths=torch.tensor([2, 1, 1, 1]),
+            stride=2,
+        )
+        ids = torch.tensor([5, 2])
+        pool.update(ids, kjt)
+        out = pool.lookup(ids)
+        assert out.keys() == ["a", "b"]
+        assert out["a"].values().tolist() == [1, 2, 3]
+        assert out["a"].lengths().tolist() == [2, 1]
+        assert out["b"].values().tolist() == [9, 8]
+
+
+class TestTower:
+    def test_forward(self):
+        tables = [
+            EmbeddingBagConfig(num_embeddings=10, embedding_dim=4, name="t", feature_names=["f"])
+        ]
+        tower = EmbeddingTower(
+            EmbeddingBagCollection(tables=tables),
+            torch.nn.Identity(),
+        )
+
+        class ValuesOf(torch.nn.Module):
+            def __init__(self, inner):
+                super().__init__()
+                self.inner = inner
+
+            def forward(self, kjt):
+                return self.inner(kjt).values()
+
+        tc = EmbeddingTowerCollection([EmbeddingTower(ValuesOf(EmbeddingBagCollection(tables=tables)), torch.nn.Identity())])
+        kjt = KeyedJaggedTensor(
+            keys=["f"], values=torch.tensor([1, 2]), lengths=torch.tensor([2]), stride=1
+        )
+        out = tc(kjt)
+        assert out.shape == (1, 4)
+
+
+class TestDeltaTracker:
+    def test_tracks_touched_ids(self):
+        import torch.distributed as dist
+        import os
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        if not dist.is_initialized():
+            dist.init_process_group("gloo", rank=0, world_size=1)
+        try:
+            from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+            from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+            from torchrec_amd.distributed.model_tracker import ModelDeltaTracker
+            from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+            from torchrec_amd.distributed.planner.types import Topology
+
+            tables = [
+                EmbeddingBagConfig(num_embeddings=50, embedding_dim=8, name="t0", feature_names=["f0"])
+            ]
+
+            class M(torch.nn.Module):
+                def __init__(self):
+                    super().__init__()
+                    self.sparse = EmbeddingBagCollection(tables=tables)
+
+                def forward(self, kjt):
+                    return self.sparse(kjt)
+
+            model = M()
+            planner = EmbeddingShardingPlanner(
+                topology=Topology(world_size=1, compute_device="cpu", hbm_cap=1 << 40)
+            )
+            sharder = EmbeddingBagCollectionSharder()
+            plan = planner.plan(model, [sharder])
+            dmp = DistributedModelParallel(
+                model, plan=plan, sharders=[sharder], init_data_parallel=False
+            )
+            tracker = ModelDeltaTracker(dmp)
+            kjt = KeyedJaggedTensor(
+                keys=["f0"], values=torch.tensor([7, 3, 7]), lengths=torch.tensor([3]), stride=1
+            )
+            dmp(kjt).values()
+            ids = tracker.get_delta_ids()
+            assert ids["t0"].tolist() == [3, 7]
+            delta = tracker.get_delta(list(dmp.sharded_modules().values())[0])
+            assert delta["t0"].shape == (2, 8)
+            tracker.clear()
+            assert tracker.get_delta_ids() == {}
+        finally:
+            dist.destroy_process_group()
