@@ -11,7 +11,7 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension  # noqa: E402
 
 CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)), "torchrec_amd", "ops", "csrc")
 
@@ -30,9 +30,23 @@ ext = CUDAExtension(
     },
 )
 
+infer_ext = CppExtension(
+    name="torchrec_amd.inference._batching",
+    sources=[
+        os.path.join(
+            os.path.dirname(os.path.abspath(__file__)),
+            "torchrec_amd",
+            "inference",
+            "csrc",
+            "batching_queue.cpp",
+        )
+    ],
+    extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
+)
+
 setup(
     name="torchrec_amd_ext",
     version="0.1.0",
-    ext_modules=[ext],
+    ext_modules=[ext, infer_ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=True)},
 )
