@@ -177,6 +177,13 @@ class DistributedModelParallel(nn.Module):
     def sharded_modules(self) -> Dict[str, nn.Module]:
         return self._sharded_modules
 
+    def reshard(self, module_fqn: str, new_module_plan) -> nn.Module:
+        """Live plan change with P2P shard movement (reference
+        model_parallel.py:813)."""
+        from torchrec_amd.distributed.dynamic_sharding import reshard_ebc
+
+        return reshard_ebc(self, module_fqn, new_module_plan)
+
     # -- nn.Module ---------------------------------------------------------
 
     def forward(self, *args, **kwargs) -> Any:
