@@ -334,3 +334,18 @@ def _run_grid_test(rank, world_size):
 
 def test_grid_sharding():
     run_multi_process(_run_grid_test, 4, "gloo")
+
+
+def _run_collective_validation(rank, world_size):
+    from torchrec_amd.distributed import collective_utils
+
+    collective_utils.set_collective_validation(True)
+    try:
+        _run_sharding_test(rank, world_size, ShardingType.TABLE_WISE.value,
+                           PoolingType.SUM.value, False)
+    finally:
+        collective_utils.set_collective_validation(False)
+
+
+def test_collective_validation_tw():
+    run_multi_process(_run_collective_validation, 2, "gloo")

@@ -129,6 +129,18 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
         send_counts = torch.stack(
             [feat_sums[boundaries[i] : boundaries[i + 1]].sum() for i in range(self._W)]
         )
+        from torchrec_amd.distributed.collective_utils import (
+            collective_tag,
+            collective_validation_enabled,
+        )
+
+        self._validate = collective_validation_enabled()
+        if self._validate:
+            # append the 31-bit signature tag per peer (reference
+            # dist_data.py:443-476 collective mismatch detection)
+            self._tag = collective_tag("kjt_a2a", input.keys(), splits)
+            tags = torch.full((self._W, 1), self._tag, dtype=send_counts.dtype, device=device)
+            send_counts = torch.cat([send_counts.view(-1, 1), tags], dim=1).reshape(-1)
         self._in_splits_t = send_counts  # device
         self._out_splits_t = torch.empty_like(send_counts)
         self._splits_work = dist.all_to_all_single(
@@ -137,8 +149,16 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
 
     def _wait_impl(self) -> KJTAllToAllTensorsAwaitable:
         self._splits_work.wait()
-        in_value_splits = self._in_splits_t.cpu().tolist()  # sync (small)
-        out_value_splits = self._out_splits_t.cpu().tolist()
+        if self._validate:
+            from torchrec_amd.distributed.collective_utils import verify_tags
+
+            recv = self._out_splits_t.view(self._W, 2).cpu()
+            verify_tags(recv[:, 1], self._tag, "kjt_a2a", self._pg)
+            in_value_splits = self._in_splits_t.view(self._W, 2)[:, 0].cpu().tolist()
+            out_value_splits = recv[:, 0].tolist()
+        else:
+            in_value_splits = self._in_splits_t.cpu().tolist()  # sync (small)
+            out_value_splits = self._out_splits_t.cpu().tolist()
         kjt = self._input
         B = self._B
         W = self._W
