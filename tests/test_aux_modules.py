@@ -114,3 +114,29 @@ class TestDeltaTracker:
             assert tracker.get_delta_ids() == {}
         finally:
             dist.destroy_process_group()
+
+
+class TestITEP:
+    def test_remap_and_prune(self):
+        from torchrec_amd.modules.itep_modules import GenericITEPModule
+
+        itep = GenericITEPModule(
+            {"t": 1000}, pruning_interval=1, pruned_hash_sizes={"t": 8}
+        )
+        itep.train()
+        kjt = KeyedJaggedTensor(
+            keys=["f"],
+            values=torch.tensor([0, 3, 500, 500, 500]),
+            lengths=torch.tensor([5]),
+            stride=1,
+        )
+        out = itep.remap(kjt, {"f": "t"})
+        assert out.values().max() < 8
+        # first pass: 500 unmapped -> sacrificial row 7; prune ran
+        out2 = itep.remap(kjt, {"f": "t"})
+        # after pruning, hot id 500 owns a real row != sacrificial
+        v = out2.values()
+        assert v[2] == v[3] == v[4]
+        assert int(v[2]) != 7
+        # mapped ids stay stable
+        assert int(v[0]) == 0 and int(v[1]) == 3

@@ -1,0 +1,120 @@
+"""In-Training Embedding Pruning (ITEP).
+
+Reference parity: torchrec/modules/itep_modules.py:78 (GenericITEPModule —
+fbgemm init_address_lookup :282, prune_embedding_tables :556,
+remap_indices_update_utils :619). Each pruned table keeps an address-lookup
+map from the full id space to a smaller physical row space; row utilization
+is tracked and tables are re-pruned every ``pruning_interval`` iterations,
+evicting cold physical rows for hot unmapped ids.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from torchrec_amd.sparse.jagged_tensor import JaggedTensor, KeyedJaggedTensor
+
+
+class GenericITEPModule(nn.Module):
+    def __init__(
+        self,
+        table_name_to_unpruned_hash_sizes: Dict[str, int],
+        lookups: Optional[List[nn.Module]] = None,
+        pruning_interval: int = 1001,
+        enable_pruning: bool = True,
+        pruned_hash_sizes: Optional[Dict[str, int]] = None,
+    ) -> None:
+        super().__init__()
+        self._unpruned = dict(table_name_to_unpruned_hash_sizes)
+        self._pruning_interval = pruning_interval
+        self._enable_pruning = enable_pruning
+        self._iter = 0
+        self._address_lookup: Dict[str, torch.Tensor] = {}
+        self._row_util: Dict[str, torch.Tensor] = {}
+        self._pruned_sizes: Dict[str, int] = {}
+        self._misses: Dict[str, List[torch.Tensor]] = {}
+        for name, unpruned in self._unpruned.items():
+            pruned = (pruned_hash_sizes or {}).get(name, max(1, unpruned // 4))
+            self._pruned_sizes[name] = pruned
+            # init: first `pruned` raw ids own the physical rows
+            lookup = torch.full((unpruned,), -1, dtype=torch.int64)
+            lookup[:pruned] = torch.arange(pruned)
+            self.register_buffer(f"_lookup_{name}", lookup, persistent=True)
+            self.register_buffer(
+                f"_util_{name}", torch.zeros(pruned, dtype=torch.int64), persistent=True
+            )
+            owner = torch.full((pruned,), -1, dtype=torch.int64)
+            owner[: min(pruned, unpruned)] = torch.arange(min(pruned, unpruned))
+            self.register_buffer(f"_owner_{name}", owner, persistent=True)
+            self._address_lookup[name] = getattr(self, f"_lookup_{name}")
+            self._row_util[name] = getattr(self, f"_util_{name}")
+
+    def pruned_size(self, table: str) -> int:
+        return self._pruned_sizes[table]
+
+    def remap_table(self, table: str, values: torch.Tensor) -> torch.Tensor:
+        lookup = getattr(self, f"_lookup_{table}")
+        util = getattr(self, f"_util_{table}")
+        mapped = lookup[values]
+        miss = mapped < 0
+        # unmapped ids share the last physical row (sacrificial bucket)
+        out = torch.where(miss, torch.full_like(mapped, self._pruned_sizes[table] - 1), mapped)
+        if self.training:
+            util.scatter_add_(0, out.clamp(min=0), torch.ones_like(out))
+            if bool(miss.any()):
+                self._misses.setdefault(table, []).append(values[miss])
+        return out
+
+    def remap(self, features: KeyedJaggedTensor, table_by_feature: Dict[str, str]) -> KeyedJaggedTensor:
+        jts = features.to_dict()
+        out = {}
+        for f, jt in jts.items():
+            t = table_by_feature.get(f)
+            if t is None or t not in self._unpruned:
+                out[f] = jt
+                continue
+            out[f] = JaggedTensor(
+                values=self.remap_table(t, jt.values()),
+                lengths=jt.lengths(),
+                weights=jt.weights_or_none(),
+            )
+        if self.training:
+            self._iter += 1
+            if self._enable_pruning and self._iter % self._pruning_interval == 0:
+                self.prune()
+        return KeyedJaggedTensor.from_jt_dict({k: out[k] for k in features.keys()})
+
+    @torch.no_grad()
+    def prune(self) -> Dict[str, torch.Tensor]:
+        """Re-prune: hot unmapped ids displace cold physical rows. Returns
+        per-table evicted physical rows (to be reset by the caller)."""
+        evicted: Dict[str, torch.Tensor] = {}
+        for name in self._unpruned:
+            lookup = getattr(self, f"_lookup_{name}")
+            util = getattr(self, f"_util_{name}")
+            misses = self._misses.pop(name, [])
+            if not misses:
+                continue
+            cand = torch.cat(misses)
+            uniq, cnt = torch.unique(cand, return_counts=True)
+            k = min(uniq.numel(), util.numel() - 1)
+            top_cnt, top_idx = torch.topk(cnt, k)
+            cold_cnt, cold_rows = torch.sort(util[:-1])  # keep sacrificial row
+            promote = top_cnt > cold_cnt[:k]
+            n = int(promote.sum())
+            if n == 0:
+                continue
+            rows = cold_rows[:k][promote]
+            new_ids = uniq[top_idx][promote]
+            owner = getattr(self, f"_owner_{name}")
+            old_owner = owner[rows]
+            valid = old_owner >= 0
+            lookup[old_owner[valid]] = -1
+            lookup[new_ids] = rows
+            owner[rows] = new_ids
+            util[rows] = top_cnt[promote]
+            evicted[name] = rows
+        return evicted
