@@ -102,3 +102,37 @@ def test_pipeline_base_cpu_single():
             assert torch.isfinite(out[0])
     finally:
         dist.destroy_process_group()
+
+
+def test_staged_pipeline_cpu():
+    from torchrec_amd.distributed.train_pipeline import PipelineStage, StagedTrainPipeline
+
+    log = []
+    pipe = StagedTrainPipeline(
+        [
+            PipelineStage("double", lambda x: x * 2),
+            PipelineStage("inc", lambda x: x + 1),
+        ]
+    )
+    it = iter(range(5))
+    outs = []
+    for _ in range(10):
+        o = pipe.progress(it)
+        if o is not None:
+            outs.append(o)
+    assert outs[:5] == [1, 3, 5, 7, 9]
+
+
+def _run_eval_pipeline(rank, world_size):
+    from torchrec_amd.distributed.train_pipeline import EvalPipelineSparseDist
+
+    dmp, opt, tables = _build_dmp(world_size)
+    pipe = EvalPipelineSparseDist(dmp, opt, torch.device("cpu"))
+    it = iter(_batches(rank, tables, n=4))
+    for _ in range(4):
+        out = pipe.progress(it)
+        assert torch.isfinite(out[0])
+
+
+def test_eval_pipeline_gloo():
+    run_multi_process(_run_eval_pipeline, 2, "gloo")

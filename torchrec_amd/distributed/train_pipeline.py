@@ -233,3 +233,75 @@ class TrainPipelineSparseDist:
         self._batches = [next_batch, batch_ip2]
         self._dist_contexts = [d_next]
         return output
+
+
+class PipelineStage:
+    """User-defined pre-forward stage (reference pipeline_stage.py:74)."""
+
+    def __init__(self, name: str, runnable, stream: Optional[torch.cuda.Stream] = None,
+                 fill_callback=None) -> None:
+        self.name = name
+        self.runnable = runnable
+        self.stream = stream
+        self.fill_callback = fill_callback
+
+
+class StagedTrainPipeline:
+    """Generic K-stage pipeline: each batch flows through user stages on their
+    own streams before the caller's fwd/bwd (reference train_pipelines.py:2579).
+
+    progress() returns the oldest fully-staged batch (or None while filling).
+    """
+
+    def __init__(self, pipeline_stages: List[PipelineStage], device: Optional[torch.device] = None) -> None:
+        self._stages = pipeline_stages
+        self._device = device
+        # slot i holds the output of stage i for the batch currently there
+        self._slots: List[Optional[Any]] = [None] * (len(pipeline_stages) + 1)
+        self._filled = False
+
+    def _run_stage(self, i: int, item: Any) -> Any:
+        stage = self._stages[i]
+        ctx = (
+            torch.cuda.stream(stage.stream)
+            if stage.stream is not None
+            else contextlib.nullcontext()
+        )
+        with torch.autograd.profiler.record_function(f"## stage {stage.name} ##"), ctx:
+            return stage.runnable(item)
+
+    def progress(self, dataloader_iter: Iterator[Any]) -> Optional[Any]:
+        # advance the pipeline one tick: shift each slot forward
+        out = self._slots[-1]
+        for i in reversed(range(len(self._stages))):
+            item = self._slots[i]
+            self._slots[i + 1] = self._run_stage(i, item) if item is not None else None
+        try:
+            self._slots[0] = next(dataloader_iter)
+        except StopIteration:
+            self._slots[0] = None
+        if out is None and any(s is not None for s in self._slots):
+            # pipeline still filling
+            return self.progress(dataloader_iter) if not self._filled else None
+        self._filled = True
+        return out
+
+
+class EvalPipelineSparseDist(TrainPipelineSparseDist):
+    """Inference-only variant: no backward/optimizer (reference :2259)."""
+
+    def progress(self, dataloader_iter: Iterator[Pipelineable]) -> Any:
+        if not self._batches:
+            self._fill_pipeline(dataloader_iter)
+        cur_batch = self._batches[0]
+        next_batch = self._batches[1]
+        if cur_batch is None:
+            raise StopIteration
+        batch_ip2 = self._copy_batch_to_gpu(dataloader_iter)
+        _wait_for_batch(cur_batch, self._memcpy_stream)
+        d_next = self._start_sparse_data_dist(next_batch)
+        with torch.no_grad(), torch.autograd.profiler.record_function("## forward ##"):
+            losses, output = self._model(cur_batch)
+        self._batches = [next_batch, batch_ip2]
+        self._dist_contexts = [d_next]
+        return output
