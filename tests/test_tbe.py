@@ -278,3 +278,38 @@ class TestTBEUvm:
         for wd, wu in zip(dev.split_embedding_weights(), uvm.split_embedding_weights()):
             assert torch.allclose(wu, wd.cpu(), atol=1e-5, rtol=1e-4)
         assert torch.allclose(uvm.momentum, dev.momentum.cpu(), atol=1e-6)
+
+
+@pytest.mark.gpu
+class TestTBECached:
+    def test_uvm_caching_matches_device(self):
+        """MANAGED_CACHING (host table + HBM lxu cache) matches HBM-resident
+        results across steps, including evictions (tiny cache)."""
+        from torchrec_amd.ops.tbe import EmbeddingLocation
+
+        specs = [("t0", 500, 64), ("t1", 300, 128)]
+        torch.manual_seed(0)
+        dev = TableBatchedEmbeddingBags(
+            specs, optimizer="rowwise_adagrad", learning_rate=0.05,
+            device=torch.device("cuda"),
+        )
+        cached = TableBatchedEmbeddingBags(
+            specs, optimizer="rowwise_adagrad", learning_rate=0.05,
+            device=torch.device("cuda"), location=EmbeddingLocation.MANAGED_CACHING,
+            cache_load_factor=0.15,  # tiny: forces eviction traffic
+        )
+        assert cached._uvm_caching and cached.weights.is_pinned()
+        cached.weights.copy_(dev.weights.cpu())
+        for step in range(5):
+            indices, offsets = make_inputs(specs, B=16, L=6, seed=step, device="cuda")
+            out_d = dev(indices, offsets)
+            out_c = cached(indices, offsets)
+            grad = torch.randn_like(out_d)
+            out_d.backward(grad)
+            out_c.backward(grad)
+            torch.cuda.synchronize()
+            assert torch.allclose(out_c, out_d, atol=1e-5, rtol=1e-5), f"fwd step {step}"
+        # flush + compare full tables
+        for wd, wc in zip(dev.split_embedding_weights(), cached.split_embedding_weights()):
+            assert torch.allclose(wc, wd.cpu(), atol=1e-5, rtol=1e-4)
+        assert torch.allclose(cached.momentum, dev.momentum.cpu(), atol=1e-6)

@@ -28,7 +28,8 @@ at::Tensor tbe_forward_pooled(const at::Tensor& weights, const at::Tensor& table
                               const at::Tensor& dims, const at::Tensor& feat_table,
                               const at::Tensor& d_out_offsets, const at::Tensor& indices,
                               const at::Tensor& offsets, const at::Tensor& per_sample_weights,
-                              int64_t B, int64_t total_D, int64_t max_D, bool mean_pool);
+                              int64_t B, int64_t total_D, int64_t max_D, bool mean_pool,
+                              const at::Tensor& cache_weights, const at::Tensor& cache_loc);
 at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_elem_offsets,
                            const at::Tensor& dims, const at::Tensor& feat_table,
                            const at::Tensor& feat_val_offsets, const at::Tensor& indices,
@@ -42,7 +43,10 @@ void tbe_backward_fused(at::Tensor weights, at::Tensor momentum, const at::Tenso
                         const at::Tensor& pos_scale, const at::Tensor& table_row_offsets,
                         const at::Tensor& table_elem_offsets, const at::Tensor& dims,
                         int64_t max_D, double lr, double eps, int64_t mode,
-                        at::Tensor grad_weights);
+                        at::Tensor grad_weights, at::Tensor cache_weights,
+                        const at::Tensor& cache_loc);
+at::Tensor gather_run_heads(const at::Tensor& sorted_linear, const at::Tensor& seg_offsets,
+                            const at::Tensor& num_runs);
 at::Tensor tbe_grad_per_sample_weights(const at::Tensor& weights,
                                        const at::Tensor& table_elem_offsets,
                                        const at::Tensor& dims, const at::Tensor& grad,
@@ -51,6 +55,19 @@ at::Tensor tbe_grad_per_sample_weights(const at::Tensor& weights,
                                        int64_t max_D);
 at::Tensor bounds_check_indices(at::Tensor indices, const at::Tensor& feat_val_offsets,
                                 const at::Tensor& rows, const at::Tensor& feat_table);
+
+// cache.hip
+void lxu_cache_populate(at::Tensor host_weights, const at::Tensor& table_row_offsets,
+                        const at::Tensor& table_elem_offsets, const at::Tensor& dims,
+                        const at::Tensor& sorted_uniq_ids, const at::Tensor& seg_offsets,
+                        const at::Tensor& num_runs, at::Tensor cache_weights,
+                        at::Tensor cache_tags, at::Tensor cache_lru, int64_t max_D,
+                        int64_t timestamp);
+at::Tensor lxu_cache_lookup(const at::Tensor& ids, const at::Tensor& cache_tags);
+void lxu_cache_flush(at::Tensor host_weights, const at::Tensor& table_row_offsets,
+                     const at::Tensor& table_elem_offsets, const at::Tensor& dims,
+                     const at::Tensor& cache_weights, const at::Tensor& cache_tags,
+                     int64_t max_D);
 
 // quant_tbe.hip
 at::Tensor quantize_rowwise_int8(const at::Tensor& weights);
@@ -97,7 +114,8 @@ TORCH_LIBRARY(trec_amd, m) {
   m.def(
       "tbe_forward_pooled(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
       " Tensor feat_table, Tensor d_out_offsets, Tensor indices, Tensor offsets,"
-      " Tensor per_sample_weights, int B, int total_D, int max_D, bool mean_pool) -> Tensor");
+      " Tensor per_sample_weights, int B, int total_D, int max_D, bool mean_pool,"
+      " Tensor cache_weights, Tensor cache_loc) -> Tensor");
   m.def(
       "tbe_forward_seq(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
       " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"
@@ -109,7 +127,18 @@ TORCH_LIBRARY(trec_amd, m) {
       " Tensor sorted_linear, Tensor sort_perm, Tensor seg_offsets, Tensor num_runs,"
       " Tensor pos_row, Tensor pos_col, Tensor pos_scale, Tensor table_row_offsets,"
       " Tensor table_elem_offsets, Tensor dims, int max_D, float lr, float eps, int mode,"
-      " Tensor(c!) grad_weights) -> ()");
+      " Tensor(c!) grad_weights, Tensor(d!) cache_weights, Tensor cache_loc) -> ()");
+  m.def("gather_run_heads(Tensor sorted_linear, Tensor seg_offsets, Tensor num_runs) -> Tensor");
+  m.def(
+      "lxu_cache_populate(Tensor(a!) host_weights, Tensor table_row_offsets,"
+      " Tensor table_elem_offsets, Tensor dims, Tensor sorted_uniq_ids, Tensor seg_offsets,"
+      " Tensor num_runs, Tensor(b!) cache_weights, Tensor(c!) cache_tags,"
+      " Tensor(d!) cache_lru, int max_D, int timestamp) -> ()");
+  m.def("lxu_cache_lookup(Tensor ids, Tensor cache_tags) -> Tensor");
+  m.def(
+      "lxu_cache_flush(Tensor(a!) host_weights, Tensor table_row_offsets,"
+      " Tensor table_elem_offsets, Tensor dims, Tensor cache_weights, Tensor cache_tags,"
+      " int max_D) -> ()");
   m.def(
       "tbe_grad_per_sample_weights(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
       " Tensor grad, Tensor indices, Tensor pos_row, Tensor pos_col, Tensor pos_table,"
@@ -147,6 +176,10 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("tbe_backward_prep", trec_amd::tbe_backward_prep);
   m.impl("tbe_backward_fused", trec_amd::tbe_backward_fused);
   m.impl("tbe_grad_per_sample_weights", trec_amd::tbe_grad_per_sample_weights);
+  m.impl("gather_run_heads", trec_amd::gather_run_heads);
+  m.impl("lxu_cache_populate", trec_amd::lxu_cache_populate);
+  m.impl("lxu_cache_lookup", trec_amd::lxu_cache_lookup);
+  m.impl("lxu_cache_flush", trec_amd::lxu_cache_flush);
   m.impl("bounds_check_indices", trec_amd::bounds_check_indices);
   m.impl("quantize_rowwise_int8", trec_amd::quantize_rowwise_int8);
   m.impl("tbe_forward_pooled_int8", trec_amd::tbe_forward_pooled_int8);

@@ -63,6 +63,9 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
     const int64_t* __restrict__ indices,
     const int64_t* __restrict__ offsets,  // [F*B+1] feature-major bags
     const float* __restrict__ psw,        // nullable per-sample weights
+    const float* __restrict__ cache_weights,   // nullable lxu cache rows
+    const int32_t* __restrict__ cache_loc,     // per-position slot or -1
+    int64_t cache_stride,
     int F, int B, int64_t total_D, bool mean_pool,
     float* __restrict__ out /* [B, total_D] */) {
   constexpr int SLOTS = kWaveSize / LPS;
@@ -83,7 +86,13 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
     int64_t i0 = offsets[bag], i1 = offsets[bag + 1];
     for (int64_t i = i0; i < i1; ++i) {
       int64_t idx = indices[i];
-      const float4* row = reinterpret_cast<const float4*>(tab + idx * D);
+      const float4* row;
+      if (cache_loc && cache_loc[i] >= 0) {
+        row = reinterpret_cast<const float4*>(cache_weights +
+                                              static_cast<int64_t>(cache_loc[i]) * cache_stride);
+      } else {
+        row = reinterpret_cast<const float4*>(tab + idx * D);
+      }
       float w = psw ? psw[i] : 1.f;
 #pragma unroll
       for (int c = 0; c < CHUNKS; ++c) {
@@ -118,13 +127,15 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
                      table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),    \
                      feat_table.data_ptr<int32_t>(), d_out_offsets.data_ptr<int64_t>(),   \
                      indices.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(),            \
-                     psw_ptr, F, B, total_D, mean_pool, out.data_ptr<float>())
+                     psw_ptr, cache_w_ptr, cache_loc_ptr, max_D, F, B, total_D, mean_pool, \
+                     out.data_ptr<float>())
 
 at::Tensor tbe_forward_pooled(
     const at::Tensor& weights, const at::Tensor& table_elem_offsets, const at::Tensor& dims,
     const at::Tensor& feat_table, const at::Tensor& d_out_offsets, const at::Tensor& indices,
     const at::Tensor& offsets, const at::Tensor& per_sample_weights, int64_t B,
-    int64_t total_D, int64_t max_D, bool mean_pool) {
+    int64_t total_D, int64_t max_D, bool mean_pool, const at::Tensor& cache_weights,
+    const at::Tensor& cache_loc) {
   TORCH_CHECK(weights.scalar_type() == at::kFloat);
   TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048, "TBE: dims must be %4==0 and <=2048");
   int F = feat_table.numel();
@@ -132,6 +143,10 @@ at::Tensor tbe_forward_pooled(
   if (B == 0 || F == 0) return out;
   const float* psw_ptr =
       per_sample_weights.numel() > 0 ? per_sample_weights.data_ptr<float>() : nullptr;
+  const float* cache_w_ptr =
+      cache_weights.numel() > 0 ? cache_weights.data_ptr<float>() : nullptr;
+  const int32_t* cache_loc_ptr =
+      cache_loc.numel() > 0 ? cache_loc.data_ptr<int32_t>() : nullptr;
   auto stream = tbe_stream();
   int64_t n_bags = static_cast<int64_t>(F) * B;
   // pick lanes-per-slot and register chunks from max_D
@@ -424,7 +439,8 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     const int64_t* __restrict__ table_row_offsets,   // [T+1] rows
     const int64_t* __restrict__ table_elem_offsets,  // [T]
     const int32_t* __restrict__ dims, int T, float lr, float eps, int mode,
-    float* __restrict__ grad_weights) {
+    float* __restrict__ grad_weights, float* __restrict__ cache_weights,
+    const int32_t* __restrict__ cache_loc, int64_t cache_stride) {
   int sl = threadIdx.x % LPS;
   int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
   int64_t n_slots = (static_cast<int64_t>(gridDim.x) * blockDim.x) / LPS;
@@ -474,8 +490,15 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
         }
       }
     }
-    float4* wrow = reinterpret_cast<float4*>(weights + table_elem_offsets[t] +
-                                             local * static_cast<int64_t>(D));
+    float4* wrow;
+    int32_t cloc = cache_loc ? cache_loc[sort_perm[k0]] : -1;
+    if (cloc >= 0) {
+      wrow = reinterpret_cast<float4*>(cache_weights +
+                                       static_cast<int64_t>(cloc) * cache_stride);
+    } else {
+      wrow = reinterpret_cast<float4*>(weights + table_elem_offsets[t] +
+                                       local * static_cast<int64_t>(D));
+    }
     if (mode == 1) {
       // rowwise Adagrad: m += mean(g^2); w -= lr * g / (sqrt(m) + eps)
       float gsq = 0.f;
@@ -534,7 +557,7 @@ void tbe_backward_fused(
     const at::Tensor& pos_col, const at::Tensor& pos_scale,
     const at::Tensor& table_row_offsets, const at::Tensor& table_elem_offsets,
     const at::Tensor& dims, int64_t max_D, double lr, double eps, int64_t mode,
-    at::Tensor grad_weights) {
+    at::Tensor grad_weights, at::Tensor cache_weights, const at::Tensor& cache_loc) {
   TORCH_CHECK(grad.scalar_type() == at::kFloat);
   int64_t n = sorted_linear.numel();
   if (n == 0) return;
@@ -545,6 +568,9 @@ void tbe_backward_fused(
   int grid = grid_for(n * lps, kBlockThreads);  // upper bound: runs <= n
   const float* scale_ptr = pos_scale.numel() > 0 ? pos_scale.data_ptr<float>() : nullptr;
   float* gw_ptr = grad_weights.numel() > 0 ? grad_weights.data_ptr<float>() : nullptr;
+  float* cache_w_ptr = cache_weights.numel() > 0 ? cache_weights.data_ptr<float>() : nullptr;
+  const int32_t* cache_loc_ptr =
+      cache_loc.numel() > 0 ? cache_loc.data_ptr<int32_t>() : nullptr;
 
   // chunk long duplicate-runs: partials scratch sized by the host upper bound
   constexpr int kChunkSize = 32;
@@ -573,7 +599,8 @@ void tbe_backward_fused(
                        pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
                        table_row_offsets.data_ptr<int64_t>(),                                \
                        table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,  \
-                       (float)lr, (float)eps, (int)mode, gw_ptr);                            \
+                       (float)lr, (float)eps, (int)mode, gw_ptr, cache_w_ptr,                \
+                       cache_loc_ptr, max_D);                                                \
   } while (0)
   if (lps == 16) TBE_BWD_LAUNCH(16, 1);
   else if (lps == 32) TBE_BWD_LAUNCH(32, 1);
@@ -584,6 +611,29 @@ void tbe_backward_fused(
     default: TBE_BWD_LAUNCH(64, 8); break;
   }
 #undef TBE_BWD_LAUNCH
+}
+
+__global__ void gather_run_heads_kernel(const int64_t* __restrict__ sorted_lin,
+                                        const int32_t* __restrict__ seg_offsets,
+                                        const int32_t* __restrict__ num_runs_ptr, int64_t n,
+                                        int64_t* __restrict__ out /* [n], -1 padded */) {
+  int32_t num_runs = *num_runs_ptr;
+  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; i < n;
+       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    out[i] = (i < num_runs) ? sorted_lin[seg_offsets[i]] : -1;
+  }
+}
+
+at::Tensor gather_run_heads(const at::Tensor& sorted_linear, const at::Tensor& seg_offsets,
+                            const at::Tensor& num_runs) {
+  int64_t n = sorted_linear.numel();
+  auto out = at::empty({n}, sorted_linear.options());
+  if (n == 0) return out;
+  hipLaunchKernelGGL(gather_run_heads_kernel, dim3(grid_for(n, kBlockThreads)),
+                     dim3(kBlockThreads), 0, tbe_stream(), sorted_linear.data_ptr<int64_t>(),
+                     seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(), n,
+                     out.data_ptr<int64_t>());
+  return out;
 }
 
 // ---------------------------------------------------------------------------

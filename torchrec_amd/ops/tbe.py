@@ -33,7 +33,8 @@ class PoolingMode(Enum):
 class EmbeddingLocation(Enum):
     DEVICE = 0  # HBM-resident
     MANAGED = 1  # host-resident, GPU-addressable (UVM-style spill)
-    HOST = 2
+    MANAGED_CACHING = 2  # host-resident + software LRU cache in HBM
+    HOST = 3
 
 
 class EmbeddingSpec(NamedTuple):
@@ -64,6 +65,7 @@ class _TBEPooledFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, dummy, host, indices, offsets, psw):  # type: ignore[override]
+        cache_loc = host._prefetch_cache(indices, offsets)
         out = torch.ops.trec_amd.tbe_forward_pooled(
             host.weights,
             host._table_elem_offsets,
@@ -77,18 +79,24 @@ class _TBEPooledFunction(torch.autograd.Function):
             host._total_D,
             host._max_D,
             host.pooling_mode == PoolingMode.MEAN,
+            host.cache_weights,
+            cache_loc,
         )
         ctx.host = host
-        ctx.save_for_backward(indices, offsets, psw if psw is not None else host._empty_f)
+        ctx.save_for_backward(
+            indices, offsets, psw if psw is not None else host._empty_f, cache_loc
+        )
         ctx.has_psw = psw is not None
         return out
 
     @staticmethod
     def backward(ctx, grad):  # type: ignore[override]
         host = ctx.host
-        indices, offsets, psw = ctx.saved_tensors
+        indices, offsets, psw, cache_loc = ctx.saved_tensors
         psw_t = psw if ctx.has_psw else None
-        grad_psw = host._backward_pooled(grad.contiguous(), indices, offsets, psw_t)
+        grad_psw = host._backward_pooled(
+            grad.contiguous(), indices, offsets, psw_t, cache_loc=cache_loc
+        )
         return None, None, None, None, grad_psw
 
 
@@ -139,13 +147,21 @@ class TableBatchedEmbeddingBags(nn.Module):
         init_min: float = -0.01,
         init_max: float = 0.01,
         location: EmbeddingLocation = EmbeddingLocation.DEVICE,
+        cache_load_factor: float = 0.2,
     ) -> None:
         super().__init__()
         device = device or torch.device("cpu")
         self.location = location
         # MANAGED: weights live in pinned host DRAM, addressed by the HIP
-        # kernels over PCIe (reference: FBGEMM EmbeddingLocation.MANAGED)
-        self._uvm = location == EmbeddingLocation.MANAGED and device.type == "cuda"
+        # kernels over PCIe (reference: FBGEMM EmbeddingLocation.MANAGED);
+        # MANAGED_CACHING adds a set-associative LRU cache in HBM (lxu cache)
+        self._uvm = (
+            location in (EmbeddingLocation.MANAGED, EmbeddingLocation.MANAGED_CACHING)
+            and device.type == "cuda"
+        )
+        self._uvm_caching = (
+            location == EmbeddingLocation.MANAGED_CACHING and device.type == "cuda"
+        )
         weights_device = torch.device("cpu") if self._uvm else device
         self.pooling_mode = pooling_mode
         self.optimizer = _OPT_NAMES[optimizer]
@@ -213,6 +229,26 @@ class TableBatchedEmbeddingBags(nn.Module):
             torch.tensor(d_out[:-1], dtype=torch.int64),
         )
         reg("_empty_f", torch.empty(0, dtype=torch.float32))
+        reg("_empty_i", torch.empty(0, dtype=torch.int32))
+        if self._uvm_caching:
+            # cache sizing: cache_load_factor of total rows, 32 ways per set
+            ways = 32
+            cache_rows = max(ways, int(self._total_rows * cache_load_factor))
+            sets = max(1, cache_rows // ways)
+            self.register_buffer(
+                "cache_weights",
+                torch.zeros(sets * ways, max(self._max_D, 4), dtype=torch.float32, device=device),
+            )
+            self.register_buffer(
+                "cache_tags", torch.full((sets * ways,), -1, dtype=torch.int64, device=device)
+            )
+            self.register_buffer(
+                "cache_lru", torch.zeros(sets * ways, dtype=torch.int64, device=device)
+            )
+            self._cache_sets = sets
+            self._cache_timestamp = 0
+        else:
+            self.register_buffer("cache_weights", torch.empty(0, device=device))
         # autograd anchor for the fused path: requires-grad, non-persistent
         # (not an nn.Parameter so it stays out of checkpoints / optimizers)
         dummy = torch.zeros(1, device=device)
@@ -241,6 +277,8 @@ class TableBatchedEmbeddingBags(nn.Module):
         )
 
     def split_embedding_weights(self) -> List[torch.Tensor]:
+        if getattr(self, "_uvm_caching", False):
+            self.flush_cache()
         out = []
         for i, s in enumerate(self._specs):
             start = int(self._table_elem_offsets[i])
@@ -292,6 +330,7 @@ class TableBatchedEmbeddingBags(nn.Module):
         psw: Optional[torch.Tensor],
         mode: Optional[int] = None,
         grad_weights: Optional[torch.Tensor] = None,
+        cache_loc: Optional[torch.Tensor] = None,
     ) -> Optional[torch.Tensor]:
         B = (offsets.numel() - 1) // self._num_features
         pos_row, pos_col, linear, bag_ids = self._bag_metadata(indices, offsets, B)
@@ -326,6 +365,8 @@ class TableBatchedEmbeddingBags(nn.Module):
             self.eps,
             self.optimizer if mode is None else mode,
             grad_weights if grad_weights is not None else self._empty_f,
+            self.cache_weights,
+            cache_loc if cache_loc is not None else self._empty_i,
         )
         if psw is not None and psw.requires_grad:
             f = torch.div(bag_ids, B, rounding_mode="floor")
@@ -342,6 +383,62 @@ class TableBatchedEmbeddingBags(nn.Module):
                 self._max_D,
             )
         return None
+
+    def _prefetch_cache(
+        self, indices: torch.Tensor, offsets: torch.Tensor
+    ) -> torch.Tensor:
+        """Populate the lxu cache for this batch; return per-position slots.
+
+        Returns the empty tensor when caching is off (kernels then read the
+        weights buffer directly)."""
+        if not self._uvm_caching:
+            return self._empty_i
+        B = (offsets.numel() - 1) // self._num_features
+        _, _, linear, _ = self._bag_metadata(indices, offsets, B)
+        sorted_lin, _ = torch.ops.trec_amd.sort_pairs(linear, _bits_needed(self._total_rows))
+        seg_offsets, num_runs = torch.ops.trec_amd.tbe_backward_prep(sorted_lin)
+        uniq = torch.ops.trec_amd.gather_run_heads(sorted_lin, seg_offsets, num_runs)
+        # order unique ids by cache set; sentinel (-1) pads sort to the end
+        big = torch.where(
+            uniq >= 0, uniq % self._cache_sets, torch.full_like(uniq, self._cache_sets + 1)
+        )
+        sorted_sets, perm2 = torch.ops.trec_amd.sort_pairs(
+            big, _bits_needed(self._cache_sets + 2)
+        )
+        ids_by_set = uniq[perm2.to(torch.int64)]
+        set_seg, set_runs = torch.ops.trec_amd.tbe_backward_prep(
+            sorted_sets
+        )
+        self._cache_timestamp += 1
+        torch.ops.trec_amd.lxu_cache_populate(
+            self.weights,
+            self._table_row_offsets,
+            self._table_elem_offsets,
+            self._dims_t,
+            ids_by_set,
+            set_seg,
+            set_runs,
+            self.cache_weights,
+            self.cache_tags,
+            self.cache_lru,
+            self.cache_weights.shape[1],
+            self._cache_timestamp,
+        )
+        return torch.ops.trec_amd.lxu_cache_lookup(linear, self.cache_tags)
+
+    def flush_cache(self) -> None:
+        """Write cached rows back to the host table (before reading weights)."""
+        if not self._uvm_caching:
+            return
+        torch.ops.trec_amd.lxu_cache_flush(
+            self.weights,
+            self._table_row_offsets,
+            self._table_elem_offsets,
+            self._dims_t,
+            self.cache_weights,
+            self.cache_tags,
+            self.cache_weights.shape[1],
+        )
 
     def _backward_seq(
         self, grad: torch.Tensor, indices: torch.Tensor, feat_val_offsets: torch.Tensor
@@ -378,6 +475,8 @@ class TableBatchedEmbeddingBags(nn.Module):
             self.eps,
             self.optimizer,
             self._empty_f,
+            self.cache_weights,
+            self._empty_i,
         )
 
     # -- CPU oracle path -----------------------------------------------------
@@ -480,6 +579,8 @@ class _TBEDenseFunction(torch.autograd.Function):
             host._total_D,
             host._max_D,
             host.pooling_mode == PoolingMode.MEAN,
+            host._empty_f,
+            host._empty_i,
         )
         ctx.host = host
         ctx.save_for_backward(indices, offsets, psw if psw is not None else host._empty_f)
