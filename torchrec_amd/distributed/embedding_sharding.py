@@ -118,11 +118,11 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
             feature_table_map = [i for i, t in enumerate(group) for _ in t.feature_names]
             from torchrec_amd.ops.tbe import EmbeddingLocation
 
-            location = (
-                EmbeddingLocation.MANAGED
-                if group[0].kernel == EmbeddingComputeKernel.FUSED_UVM.value
-                else EmbeddingLocation.DEVICE
-            )
+            location = EmbeddingLocation.DEVICE
+            if group[0].kernel == EmbeddingComputeKernel.FUSED_UVM.value:
+                location = EmbeddingLocation.MANAGED
+            elif group[0].kernel == EmbeddingComputeKernel.FUSED_UVM_CACHING.value:
+                location = EmbeddingLocation.MANAGED_CACHING
             tbe = TableBatchedEmbeddingBags(
                 specs,
                 feature_table_map=feature_table_map,
@@ -132,6 +132,7 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                 eps=fused_params.get("eps", 1.0e-8),
                 device=device,
                 location=location,
+                cache_load_factor=fused_params.get("cache_load_factor", 0.2),
             )
             self._emb_modules.append(tbe)
             nf = sum(len(t.feature_names) for t in group)

@@ -153,8 +153,11 @@ class EmbeddingEnumerator:
                         and EmbeddingComputeKernel.FUSED.value in kernels
                         and EmbeddingComputeKernel.FUSED_UVM.value not in kernels
                     ):
-                        # host-DRAM spill variant for tables beyond HBM
-                        kernels = kernels + [EmbeddingComputeKernel.FUSED_UVM.value]
+                        # host-DRAM spill variants for tables beyond HBM
+                        kernels = kernels + [
+                            EmbeddingComputeKernel.FUSED_UVM_CACHING.value,
+                            EmbeddingComputeKernel.FUSED_UVM.value,
+                        ]
                     for kernel in kernels:
                         shards = [
                             Shard(size=list(sz), offset=list(off))
@@ -194,6 +197,9 @@ class EmbeddingEnumerator:
             else 1.0 / D  # rowwise adagrad: one momentum scalar per row
         )
         is_uvm = opt.compute_kernel == EmbeddingComputeKernel.FUSED_UVM.value
+        is_cached = opt.compute_kernel == EmbeddingComputeKernel.FUSED_UVM_CACHING.value
+        CACHE_LOAD = 0.2
+        CACHE_HIT = 0.8  # assumed; refine with table stats
         for shard in opt.shards:
             rows, cols = shard.size
             # storage: weights + optimizer + a slice of activation/grad buffers
@@ -201,6 +207,9 @@ class EmbeddingEnumerator:
             act_bytes = int(B * W * cols * elem * 4)
             if is_uvm:
                 shard.storage = Storage(hbm=act_bytes, ddr=weight_bytes)
+            elif is_cached:
+                cache_bytes = int(weight_bytes * CACHE_LOAD)
+                shard.storage = Storage(hbm=act_bytes + cache_bytes, ddr=weight_bytes)
             else:
                 shard.storage = Storage(hbm=weight_bytes + act_bytes, ddr=0)
             # perf: bytes moved / bandwidth
@@ -222,7 +231,12 @@ class EmbeddingEnumerator:
                 local_B = B * W
                 fwd_bytes = local_B * pooling * cols * elem
                 comms = local_B * cols * elem / topo.intra_host_bw if W > 1 else 0.0
-            mem_bw = topo.ddr_mem_bw if is_uvm else topo.hbm_mem_bw
+            if is_uvm:
+                mem_bw = topo.ddr_mem_bw
+            elif is_cached:
+                mem_bw = 1.0 / (CACHE_HIT / topo.hbm_mem_bw + (1 - CACHE_HIT) / topo.ddr_mem_bw)
+            else:
+                mem_bw = topo.hbm_mem_bw
             fwd_compute = fwd_bytes / mem_bw
             shard.perf = Perf(
                 fwd_compute=fwd_compute,
