@@ -32,26 +32,35 @@ __global__ void __launch_bounds__(64) interaction_fwd_kernel(
     int B, int F1, int D, int P, float* __restrict__ out /* [B, D+P] */) {
   // ONE wave per block: __syncthreads() is a cheap wave-local fence, no
   // cross-wave iteration-count hazards in the grid-stride loop.
-  extern __shared__ float lds[];  // [F1 * (D+1)]
+  // LDS tile is float4-strided with a +1 float4 pad: rows stay 16B-aligned
+  // for ds_read_b128 while the pad staggers banks across rows.
+  extern __shared__ float lds[];  // [F1][D/4+1] float4
   int l = lane_id();
-  float* T = lds;
+  float4* T4 = reinterpret_cast<float4*>(lds);
+  const int d4 = D / 4;
+  const int stride4 = d4 + 1;
   int64_t out_w = D + P;
   for (int64_t b = blockIdx.x; b < B; b += gridDim.x) {
-    // stage T: row 0 = dense, rows 1..F1-1 = sparse (+1 pad kills the
-    // D-stride bank conflict on the dot reads)
-    for (int t = l; t < D; t += kWaveSize) T[t] = dense[b * D + t];
-    for (int t = l; t < (F1 - 1) * D; t += kWaveSize) {
-      int r = t / D, c = t - r * D;
-      T[(r + 1) * (D + 1) + c] = sparse[(b * (F1 - 1) + r) * D + c];
+    // stage T: row 0 = dense, rows 1..F1-1 = sparse
+    const float4* drow = reinterpret_cast<const float4*>(dense + b * D);
+    for (int t = l; t < d4; t += kWaveSize) T4[t] = drow[t];
+    const float4* srow = reinterpret_cast<const float4*>(sparse + b * (int64_t)(F1 - 1) * D);
+    for (int t = l; t < (F1 - 1) * d4; t += kWaveSize) {
+      int r = t / d4, c = t - r * d4;
+      T4[(r + 1) * stride4 + c] = srow[t];
     }
     __syncthreads();
     float* orow = out + b * out_w;
-    for (int t = l; t < D; t += kWaveSize) orow[t] = T[t];
+    float4* orow4 = reinterpret_cast<float4*>(orow);
+    for (int t = l; t < d4; t += kWaveSize) orow4[t] = T4[t];
     for (int p = l; p < P; p += kWaveSize) {
-      const float* Ti = T + pi[p] * (D + 1);
-      const float* Tj = T + pj[p] * (D + 1);
+      const float4* Ti = T4 + pi[p] * stride4;
+      const float4* Tj = T4 + pj[p] * stride4;
       float acc = 0.f;
-      for (int d = 0; d < D; ++d) acc += Ti[d] * Tj[d];
+      for (int k = 0; k < d4; ++k) {
+        float4 a = Ti[k], c = Tj[k];
+        acc += a.x * c.x + a.y * c.y + a.z * c.z + a.w * c.w;
+      }
       orow[D + p] = acc;
     }
     __syncthreads();
@@ -67,31 +76,47 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
     float* __restrict__ d_dense,   // [B, D]
     float* __restrict__ d_sparse   // [B, F, D]
 ) {
-  extern __shared__ float lds[];  // [F1 * (D+1)]
-  float* T = lds;
+  extern __shared__ float lds[];  // [F1][D/4+1] float4
+  float4* T4 = reinterpret_cast<float4*>(lds);
   int l = lane_id();
   int wave = wave_id();
+  const int d4 = D / 4;
+  const int stride4 = d4 + 1;
   int64_t out_w = D + P;
   for (int64_t b = blockIdx.x; b < B; b += gridDim.x) {
-    for (int t = threadIdx.x; t < D; t += blockDim.x) T[t] = dense[b * D + t];
-    for (int t = threadIdx.x; t < (F1 - 1) * D; t += blockDim.x) {
-      int r = t / D, c = t - r * D;
-      T[(r + 1) * (D + 1) + c] = sparse[(b * (F1 - 1) + r) * D + c];
+    const float4* drow = reinterpret_cast<const float4*>(dense + b * D);
+    for (int t = threadIdx.x; t < d4; t += blockDim.x) T4[t] = drow[t];
+    const float4* srow = reinterpret_cast<const float4*>(sparse + b * (int64_t)(F1 - 1) * D);
+    for (int t = threadIdx.x; t < (F1 - 1) * d4; t += blockDim.x) {
+      int r = t / d4, c = t - r * d4;
+      T4[(r + 1) * stride4 + c] = srow[t];
     }
     __syncthreads();
     const float* grow = grad_out + b * out_w;
     for (int i = wave; i < F1; i += kBlockThreads / kWaveSize) {
-      // lanes cover D
-      for (int d0 = l; d0 < D; d0 += kWaveSize) {
-        float acc = (i == 0) ? grow[d0] : 0.f;
+      // lanes cover D/4 float4 columns
+      for (int k = l; k < d4; k += kWaveSize) {
+        float4 acc;
+        if (i == 0) {
+          acc = reinterpret_cast<const float4*>(grow)[k];
+        } else {
+          acc = make_float4(0.f, 0.f, 0.f, 0.f);
+        }
         for (int j = 0; j < F1; ++j) {
           int c = pair_col[i * F1 + j];
-          if (c >= 0) acc += grow[D + c] * T[j * (D + 1) + d0];
+          if (c >= 0) {
+            float dz = grow[D + c];
+            float4 t = T4[j * stride4 + k];
+            acc.x += dz * t.x;
+            acc.y += dz * t.y;
+            acc.z += dz * t.z;
+            acc.w += dz * t.w;
+          }
         }
         if (i == 0) {
-          d_dense[b * D + d0] = acc;
+          reinterpret_cast<float4*>(d_dense + b * D)[k] = acc;
         } else {
-          d_sparse[(b * (F1 - 1) + (i - 1)) * D + d0] = acc;
+          reinterpret_cast<float4*>(d_sparse + (b * (int64_t)(F1 - 1) + (i - 1)) * D)[k] = acc;
         }
       }
     }
@@ -109,8 +134,9 @@ at::Tensor interaction_forward(const at::Tensor& dense, const at::Tensor& sparse
   int P = pi.numel();
   auto out = at::empty({B, D + P}, dense.options());
   if (B == 0) return out;
-  int lds_bytes = F1 * (D + 1) * sizeof(float);
-  int grid = std::min<int>(B, kNumCU * 16);
+  TORCH_CHECK(D % 4 == 0, "interaction kernel needs D %% 4 == 0");
+  int lds_bytes = F1 * (D / 4 + 1) * sizeof(float4);
+  int grid = std::min<int>(B, kNumCU * 32);
   hipLaunchKernelGGL(interaction_fwd_kernel, dim3(grid), dim3(64), lds_bytes, ia_stream(),
                      dense.contiguous().data_ptr<float>(),
                      sparse.contiguous().data_ptr<float>(), pi.data_ptr<int8_t>(),
@@ -128,7 +154,8 @@ std::tuple<at::Tensor, at::Tensor> interaction_backward(
   auto d_dense = at::empty_like(dense);
   auto d_sparse = at::empty_like(sparse);
   if (B == 0) return {d_dense, d_sparse};
-  int lds_bytes = F1 * (D + 1) * sizeof(float);
+  TORCH_CHECK(D % 4 == 0, "interaction kernel needs D %% 4 == 0");
+  int lds_bytes = F1 * (D / 4 + 1) * sizeof(float4);
   int grid = std::min<int>(B, kMaxBlocks);
   hipLaunchKernelGGL(interaction_bwd_kernel, dim3(grid), dim3(kBlockThreads), lds_bytes,
                      ia_stream(), grad_out.contiguous().data_ptr<float>(),
