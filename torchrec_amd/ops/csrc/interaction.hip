@@ -93,9 +93,15 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
     }
     __syncthreads();
     const float* grow = grad_out + b * out_w;
-    for (int i = wave; i < F1; i += kBlockThreads / kWaveSize) {
+    // when a row needs <= 32 float4 columns, each half-wave takes its own
+    // row so no lanes idle (D=128 -> rows i and i+ROWS_PER_WAVE/2)
+    const int halves = (d4 <= kWaveSize / 2) ? 2 : 1;
+    const int rows_per_iter = (kBlockThreads / kWaveSize) * halves;
+    const int half = (halves == 2) ? (l >> 5) : 0;
+    const int lk = (halves == 2) ? (l & 31) : l;
+    for (int i = wave * halves + half; i < F1; i += rows_per_iter) {
       // lanes cover D/4 float4 columns
-      for (int k = l; k < d4; k += kWaveSize) {
+      for (int k = lk; k < d4; k += kWaveSize / halves) {
         float4 acc;
         if (i == 0) {
           acc = reinterpret_cast<const float4*>(grow)[k];
