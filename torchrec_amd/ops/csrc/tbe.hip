@@ -238,10 +238,26 @@ std::tuple<at::Tensor, at::Tensor> sort_pairs(const at::Tensor& keys, int64_t en
   TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == at::kLong);
   int64_t n = keys.numel();
   auto vals_in = at::arange(n, keys.options().dtype(at::kInt));
-  auto keys_out = at::empty_like(keys);
   auto vals_out = at::empty_like(vals_in);
-  if (n == 0) return {keys_out, vals_out};
   auto stream = tbe_stream();
+  if (n == 0) return {at::empty_like(keys), vals_out};
+  if (end_bit <= 31) {
+    // id space fits 32-bit: halve the radix-sort key bandwidth
+    auto k32 = keys.to(at::kInt);
+    auto k32_out = at::empty_like(k32);
+    size_t tmp_bytes = 0;
+    hipcub::DeviceRadixSort::SortPairs(
+        nullptr, tmp_bytes, k32.data_ptr<int32_t>(), k32_out.data_ptr<int32_t>(),
+        vals_in.data_ptr<int32_t>(), vals_out.data_ptr<int32_t>(), n, 0, (int)end_bit,
+        stream);
+    auto tmp = at::empty({static_cast<int64_t>(tmp_bytes)}, keys.options().dtype(at::kByte));
+    hipcub::DeviceRadixSort::SortPairs(
+        tmp.data_ptr(), tmp_bytes, k32.data_ptr<int32_t>(), k32_out.data_ptr<int32_t>(),
+        vals_in.data_ptr<int32_t>(), vals_out.data_ptr<int32_t>(), n, 0, (int)end_bit,
+        stream);
+    return {k32_out.to(at::kLong), vals_out};
+  }
+  auto keys_out = at::empty_like(keys);
   size_t tmp_bytes = 0;
   hipcub::DeviceRadixSort::SortPairs(
       nullptr, tmp_bytes, keys.data_ptr<int64_t>(), keys_out.data_ptr<int64_t>(),
