@@ -109,6 +109,27 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                 if group[0].use_sum_kernel
                 else group[0].pooling
             )
+            if group[0].kernel == EmbeddingComputeKernel.QUANT.value:
+                from torchrec_amd.quant.embedding_modules import (
+                    QuantTableBatchedEmbeddingBags,
+                )
+
+                qtbe = QuantTableBatchedEmbeddingBags(
+                    [(t.name, t.local_rows, t.local_dim) for t in group],
+                    feature_table_map=[
+                        i for i, t in enumerate(group) for _ in t.feature_names
+                    ],
+                    pooling=pool,
+                    device=device,
+                )
+                self._emb_modules.append(qtbe)
+                self._feature_splits.append(
+                    sum(len(t.feature_names) for t in group)
+                )
+                self._group_dims.extend(
+                    t.local_dim for t in group for _ in t.feature_names
+                )
+                continue
             optimizer = (
                 "dense"
                 if group[0].kernel == EmbeddingComputeKernel.DENSE.value
@@ -169,6 +190,8 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
         momentum_view) per local shard — the sharded-checkpoint surface."""
         out = []
         for group, tbe in zip(self._grouped_tables, self._emb_modules):
+            if not hasattr(tbe, "split_embedding_weights"):
+                continue  # quant inference group: no float state surface
             weights = tbe.split_embedding_weights()
             states = tbe.split_optimizer_states()
             for t, w, st in zip(group, weights, states):

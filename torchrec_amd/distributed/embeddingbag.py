@@ -422,7 +422,7 @@ class EmbeddingFusedOptimizer(FusedOptimizer):
         state: Dict[torch.Tensor, Any] = {}
         param_groups: List[Dict[str, Any]] = []
         for tbe in sharded_ebc.tbes():
-            if tbe.optimizer == 2:  # dense — external optimizer owns it
+            if getattr(tbe, "optimizer", 2) == 2:  # dense/quant — no fused state
                 continue
             weights = tbe.split_embedding_weights()
             states = tbe.split_optimizer_states()

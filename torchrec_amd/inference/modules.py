@@ -64,7 +64,34 @@ def shard_quant_model(
         model = model.to(device)
         plan = {"world_size": 1, "placement": str(device)}
         return model, plan
-    raise NotImplementedError("multi-GPU quant inference sharding lands next round")
+    # multi-rank serving: shard quant EBCs with the training dists (TW/RW)
+    import torch.distributed as dist
+
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.quant_embeddingbag import (
+        QuantEmbeddingBagCollectionSharder,
+    )
+    from torchrec_amd.distributed.types import ShardingEnv
+
+    assert dist.is_initialized(), "multi-rank quant sharding needs torch.distributed"
+    env = ShardingEnv.from_process_group(dist.group.WORLD)
+    sharder = QuantEmbeddingBagCollectionSharder()
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(
+            world_size=world_size,
+            compute_device=compute_device,
+            hbm_cap=None if compute_device == "cuda" else 1 << 40,
+        ),
+        constraints=constraints,
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, env=env, plan=plan, sharders=[sharder], device=device,
+        init_data_parallel=False,
+    )
+    return dmp, plan
 
 
 class PredictModule(nn.Module):
