@@ -349,3 +349,40 @@ def _run_collective_validation(rank, world_size):
 
 def test_collective_validation_tw():
     run_multi_process(_run_collective_validation, 2, "gloo")
+
+
+def _run_optimizer_state_test(rank, world_size):
+    from torch.distributed._shard.sharded_tensor import ShardedTensor
+
+    _run_sharding_test(rank, world_size, ShardingType.ROW_WISE.value,
+                       PoolingType.SUM.value, False)
+    # rebuild a DMP to inspect optimizer state layout
+    tables = make_tables()
+    torch.manual_seed(42)
+    model = SparseModel(make_tables())
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(sharding_types=[ShardingType.ROW_WISE.value])
+            for cfg in tables
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    sd = dmp.fused_optimizer.state_dict()
+    key = "sparse.embedding_bags.t1.weight"
+    assert key in sd["state"], sd["state"].keys()
+    mom = sd["state"][key]["t1.momentum1"]
+    assert isinstance(mom, ShardedTensor)
+    assert mom.size(0) == 170  # full rows
+    for shard in mom.local_shards():
+        assert shard.tensor.dim() == 1
+
+
+def test_fused_optimizer_sharded_state():
+    run_multi_process(_run_optimizer_state_test, 2, "gloo")

@@ -414,25 +414,71 @@ class ShardedEmbeddingBagCollection(nn.Module):
 
 
 class EmbeddingFusedOptimizer(FusedOptimizer):
-    """Exposes TBE fused state as FQN-keyed optimizer state
+    """Exposes TBE fused state as FQN-keyed optimizer state; momentum
+    surfaces as a 1-D ShardedTensor for TW/RW shards so the optimizer
+    checkpoint has the reference layout
     (reference batched_embedding_kernel.py:1218)."""
 
     def __init__(self, sharded_ebc: ShardedEmbeddingBagCollection) -> None:
+        from torchrec_amd.distributed.sharded_state import build_sharded_tensor
+
         params: Dict[str, torch.Tensor] = {}
         state: Dict[torch.Tensor, Any] = {}
         param_groups: List[Dict[str, Any]] = []
-        for tbe in sharded_ebc.tbes():
-            if getattr(tbe, "optimizer", 2) == 2:  # dense/quant — no fused state
+        pg = sharded_ebc._env.process_group
+        device_type = sharded_ebc._device.type
+        lr = 0.01
+        views = sharded_ebc._shard_views()
+        by_table: Dict[str, List] = {}
+        for (t, ro, co, full, w, m) in views:
+            by_table.setdefault(t, []).append((ro, co, full, w, m))
+        for t, shards in by_table.items():
+            ps = sharded_ebc._plan_by_table.get(t)
+            key = f"embedding_bags.{t}.weight"
+            (ro, co, full, w, m) = shards[0]
+            if m is None:
                 continue
-            weights = tbe.split_embedding_weights()
-            states = tbe.split_optimizer_states()
-            for spec, w, st in zip(tbe.embedding_specs, weights, states):
-                key = f"embedding_bags.{spec.name}.weight"
-                params[key] = w
-                if st:
-                    state[w] = {f"{spec.name}.momentum1": st[0]}
-                param_groups.append({"params": [w], "lr": tbe.learning_rate})
+            params[key] = w
+            param_groups.append({"params": [w], "lr": lr})
+            single_col_shard = len(shards) == 1 and co == 0
+            if (
+                pg is not None
+                and ps is not None
+                and ps.sharding_type
+                in (ShardingType.TABLE_WISE.value, ShardingType.ROW_WISE.value)
+                and single_col_shard
+            ):
+                mom_ps = ParameterShardingView1D(ps)
+                st = build_sharded_tensor(
+                    [(m, [ro, 0])][: 1 if m.numel() else 0] or [],
+                    (full[0],),
+                    mom_ps,
+                    pg,
+                    device_type,
+                )
+                state[w] = {f"{t}.momentum1": st}
+            else:
+                state[w] = {f"{t}.momentum1": m}
         super().__init__(params, state, param_groups)
+
+
+class ParameterShardingView1D:
+    """Project a 2-D table plan to the 1-D momentum row space."""
+
+    def __init__(self, ps) -> None:
+        self.sharding_type = ps.sharding_type
+        self.compute_kernel = ps.compute_kernel
+        self.ranks = ps.ranks
+        from torchrec_amd.distributed.types import ShardMetadata
+
+        self.sharding_spec = [
+            ShardMetadata(
+                shard_offsets=[md.shard_offsets[0]],
+                shard_sizes=[md.shard_sizes[0]],
+                placement_rank=md.placement_rank,
+            )
+            for md in (ps.sharding_spec or [])
+        ]
 
 
 class EmbeddingBagCollectionSharder(ModuleSharder[EmbeddingBagCollection]):

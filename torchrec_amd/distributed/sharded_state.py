@@ -32,8 +32,8 @@ def _placement(rank: int, device_type: str, local_size: int = 8) -> str:
 
 
 def build_sharded_tensor(
-    local_shards: List[Tuple[torch.Tensor, List[int]]],  # (tensor, [row_off, col_off])
-    full_shape: Tuple[int, int],
+    local_shards: List[Tuple[torch.Tensor, List[int]]],  # (tensor, offsets)
+    full_shape,
     ps: ParameterSharding,
     pg: Optional[dist.ProcessGroup],
     device_type: str,
@@ -47,13 +47,21 @@ def build_sharded_tensor(
     if pg is None:
         # single-process: just return the (only) local shard view
         return local_shards[0][0] if local_shards else torch.empty(0)
+    def to_global(r: int) -> int:
+        # plan ranks are group-local; ShardedTensor placements must be the
+        # GLOBAL ranks of the members of `pg`
+        try:
+            return dist.get_global_rank(pg, r)
+        except (RuntimeError, ValueError):
+            return r
+
     shards_md = []
     for md in ps.sharding_spec or []:
         shards_md.append(
             STShardMetadata(
                 shard_offsets=list(md.shard_offsets),
                 shard_sizes=list(md.shard_sizes),
-                placement=_placement(md.placement_rank, device_type),
+                placement=_placement(to_global(md.placement_rank), device_type),
             )
         )
     st_meta = ShardedTensorMetadata(
@@ -67,15 +75,16 @@ def build_sharded_tensor(
             pin_memory=False,
         ),
     )
+    ndim = len(full_shape)
     st_local = []
     for t, off in local_shards:
         st_local.append(
             STShard(
                 tensor=t,
                 metadata=STShardMetadata(
-                    shard_offsets=[off[0], off[1]],
+                    shard_offsets=list(off[:ndim]),
                     shard_sizes=list(t.shape),
-                    placement=_placement(dist.get_rank(pg), device_type),
+                    placement=_placement(to_global(dist.get_rank(pg)), device_type),
                 ),
             )
         )
