@@ -201,3 +201,58 @@ class TestFeatureProcessors:
         # gradient flows to position weights
         kt.values().sum().backward()
         assert fp.position_weights["f1"].grad is not None
+
+
+class TestDeepFMAndCross:
+    def test_simple_deepfm(self):
+        from torchrec_amd.modules.deepfm import SimpleDeepFMNN
+
+        tables = [
+            EmbeddingBagConfig(num_embeddings=50, embedding_dim=8, name=f"t{i}", feature_names=[f"f{i}"])
+            for i in range(2)
+        ]
+        model = SimpleDeepFMNN(
+            num_dense_features=4,
+            embedding_bag_collection=EmbeddingBagCollection(tables=tables),
+            hidden_layer_size=16,
+            deep_fm_dimension=8,
+        )
+        kjt = KeyedJaggedTensor(
+            keys=["f0", "f1"],
+            values=torch.tensor([1, 2, 3, 4]),
+            lengths=torch.tensor([1, 1, 1, 1]),
+            stride=2,
+        )
+        out = model(torch.rand(2, 4), kjt)
+        assert out.shape == (2, 1)
+        out.sum().backward()
+
+    def test_crossnets(self):
+        from torchrec_amd.modules.crossnet import CrossNet, VectorCrossNet
+
+        x = torch.randn(4, 8)
+        assert CrossNet(8, 2)(x).shape == (4, 8)
+        assert VectorCrossNet(8, 2)(x).shape == (4, 8)
+
+    def test_fx_tracer(self):
+        from torchrec_amd.fx.tracer import symbolic_trace
+
+        tables = [
+            EmbeddingBagConfig(num_embeddings=10, embedding_dim=4, name="t", feature_names=["f"])
+        ]
+
+        class M(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.ebc = EmbeddingBagCollection(tables=tables)
+                self.lin = torch.nn.Linear(4, 1)
+
+            def forward(self, kjt):
+                return self.lin(self.ebc(kjt).values())
+
+        gm = symbolic_trace(M())
+        kjt = KeyedJaggedTensor(
+            keys=["f"], values=torch.tensor([1, 2]), lengths=torch.tensor([2]), stride=1
+        )
+        out = gm(kjt)
+        assert out.shape == (1, 1)
