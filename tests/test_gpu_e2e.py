@@ -89,3 +89,46 @@ def test_cuda_vs_cpu_dmp_losses():
         opt_c.zero_grad(); lc.backward(); opt_c.step()
         opt_g.zero_grad(); lg.backward(); opt_g.step()
     torch.cuda.synchronize()
+
+
+def test_prefetch_pipeline_uvm_caching_cuda():
+    """4-stage prefetch pipeline over UVM-cached tables on cuda:0."""
+    import bench
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import ParameterConstraints, Topology
+    from torchrec_amd.distributed.train_pipeline import PrefetchTrainPipelineSparseDist
+    from torchrec_amd.distributed.types import ShardingEnv, ShardingType
+
+    device = torch.device("cuda", 0)
+    model = bench.build_model(1e-4)
+    rows = bench.scaled_rows(1e-4)
+    constraints = {
+        f"t_cat_{i}": ParameterConstraints(
+            sharding_types=[ShardingType.TABLE_WISE.value],
+            compute_kernels=["fused_uvm_caching"],
+        )
+        for i in range(len(rows))
+    }
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.05,
+                      "cache_load_factor": 0.3}
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=1, compute_device="cuda", batch_size=64),
+        constraints=constraints,
+    )
+    plan = planner.plan(model, [sharder])
+    kernels = {ps.compute_kernel for mp in plan.plan.values() for ps in mp.values()}
+    assert kernels == {"fused_uvm_caching"}
+    dmp = DistributedModelParallel(
+        model, env=ShardingEnv.from_local(1, 0), plan=plan, sharders=[sharder], device=device
+    )
+    opt = torch.optim.SGD([p for p in dmp.parameters() if p.requires_grad], lr=0.05)
+    pipe = PrefetchTrainPipelineSparseDist(dmp, opt, device)
+    it = bench._CyclingIterator(bench.make_host_batches(4, 64, 1e-4, seed=5, pin=True))
+    for _ in range(6):
+        out = pipe.progress(it)
+        assert torch.isfinite(out[0])
+    torch.cuda.synchronize()

@@ -136,3 +136,28 @@ def _run_eval_pipeline(rank, world_size):
 
 def test_eval_pipeline_gloo():
     run_multi_process(_run_eval_pipeline, 2, "gloo")
+
+
+def _run_prefetch_pipeline(rank, world_size):
+    from torchrec_amd.distributed.train_pipeline import PrefetchTrainPipelineSparseDist
+
+    dmp_a, opt_a, tables = _build_dmp(world_size)
+    ref_losses = []
+    for batch in _batches(rank, tables):
+        loss, _ = dmp_a(batch)
+        opt_a.zero_grad()
+        loss.backward()
+        opt_a.step()
+        ref_losses.append(float(loss.detach()))
+
+    dmp_b, opt_b, _ = _build_dmp(world_size)
+    pipe = PrefetchTrainPipelineSparseDist(dmp_b, opt_b, torch.device("cpu"))
+    it = iter(_batches(rank, tables))
+    pipe_losses = [float(pipe.progress(it)[0]) for _ in range(6)]
+    torch.testing.assert_close(
+        torch.tensor(pipe_losses), torch.tensor(ref_losses), atol=1e-5, rtol=1e-5
+    )
+
+
+def test_prefetch_pipeline_gloo():
+    run_multi_process(_run_prefetch_pipeline, 2, "gloo")

@@ -305,3 +305,97 @@ class EvalPipelineSparseDist(TrainPipelineSparseDist):
         self._batches = [next_batch, batch_ip2]
         self._dist_contexts = [d_next]
         return output
+
+
+class PrefetchTrainPipelineSparseDist(TrainPipelineSparseDist):
+    """4-stage pipeline, 3 batches in flight (reference train_pipelines.py:1967):
+
+    batch i:   fwd/bwd/opt (default stream)
+    batch i+1: UVM-cache prefetch (prefetch stream)
+    batch i+2: sparse input_dist (data-dist stream)
+    batch i+3: H2D copy (memcpy stream)
+    """
+
+    def __init__(self, *args, **kwargs) -> None:
+        super().__init__(*args, **kwargs)
+        self._prefetch_stream = (
+            torch.cuda.Stream(priority=-1) if self._device.type == "cuda" else None
+        )
+        self._prefetched_dist: List[Dict[str, Tuple[Any, Any]]] = []
+
+    def _prefetch(self, dists: Dict[str, Tuple[Any, Any]]) -> Dict[str, Tuple[Any, Any]]:
+        """Wait the batch's input_dist and run cache prefetch on its stream."""
+        out: Dict[str, Tuple[Any, Any]] = {}
+        ctxmgr = (
+            torch.cuda.stream(self._prefetch_stream)
+            if self._prefetch_stream is not None
+            else contextlib.nullcontext()
+        )
+        with torch.autograd.profiler.record_function("## prefetch ##"), ctxmgr:
+            for fqn, (ctx, tensors_aw) in dists.items():
+                dist_input = tensors_aw.wait()
+                module = self._pipelined[fqn]
+                for lookup in getattr(module, "_lookups", []):
+                    kjts = (
+                        dist_input
+                        if isinstance(dist_input, list)
+                        else [dist_input]
+                    )
+                    for kjt, tbe_list in zip(kjts, [lookup.tbes()]):
+                        for tbe in tbe_list:
+                            if getattr(tbe, "_uvm_caching", False):
+                                tbe.prefetch(kjt.values(), kjt.offsets())
+                out[fqn] = (ctx, _Ready(dist_input))
+        return out
+
+    def _fill_pipeline(self, dataloader_iter) -> None:
+        b0 = self._copy_batch_to_gpu(dataloader_iter)
+        if b0 is None:
+            raise StopIteration
+        if not self._rewritten:
+            self._rewrite_model()
+        d0 = self._prefetch(self._start_sparse_data_dist(b0))
+        b1 = self._copy_batch_to_gpu(dataloader_iter)
+        d1 = self._start_sparse_data_dist(b1)
+        b2 = self._copy_batch_to_gpu(dataloader_iter)
+        self._batches = [b0, b1, b2]
+        self._dist_contexts = [d0, d1]
+
+    def progress(self, dataloader_iter: Iterator[Pipelineable]) -> Any:
+        if not self._batches:
+            self._fill_pipeline(dataloader_iter)
+        cur_batch = self._batches[0]
+        if cur_batch is None:
+            raise StopIteration
+        batch_ip3 = self._copy_batch_to_gpu(dataloader_iter)
+        _wait_for_batch(cur_batch, self._memcpy_stream)
+        self._optimizer.zero_grad(set_to_none=True)
+        # start dist for i+2, prefetch i+1
+        d_ip2 = self._start_sparse_data_dist(self._batches[2])
+        d_ip1 = self._prefetch(self._dist_contexts[1])
+        if self._prefetch_stream is not None:
+            torch.cuda.current_stream().wait_stream(self._prefetch_stream)
+        ac = (
+            torch.autocast(device_type=self._device.type, dtype=self._autocast_dtype)
+            if self._autocast_dtype is not None
+            else contextlib.nullcontext()
+        )
+        with torch.autograd.profiler.record_function("## forward ##"), ac:
+            losses, output = self._model(cur_batch)
+        with torch.autograd.profiler.record_function("## backward ##"):
+            torch.sum(losses).backward()
+        with torch.autograd.profiler.record_function("## optimizer ##"):
+            self._optimizer.step()
+        self._batches = [self._batches[1], self._batches[2], batch_ip3]
+        self._dist_contexts = [d_ip1, d_ip2]
+        return output
+
+
+class _Ready:
+    """Pre-waited awaitable."""
+
+    def __init__(self, value) -> None:
+        self._value = value
+
+    def wait(self):
+        return self._value

@@ -247,6 +247,9 @@ class TableBatchedEmbeddingBags(nn.Module):
             )
             self._cache_sets = sets
             self._cache_timestamp = 0
+            from collections import deque
+
+            self._prefetched = deque()
         else:
             self.register_buffer("cache_weights", torch.empty(0, device=device))
         # autograd anchor for the fused path: requires-grad, non-persistent
@@ -384,15 +387,32 @@ class TableBatchedEmbeddingBags(nn.Module):
             )
         return None
 
+    def prefetch(self, indices: torch.Tensor, offsets: torch.Tensor) -> None:
+        """Explicit cache prefetch (reference SplitTBE.prefetch()): populate
+        the lxu cache ahead of time; the next forward pops the locations.
+        Used by PrefetchTrainPipelineSparseDist's prefetch stream."""
+        if not self._uvm_caching:
+            return
+        loc = self._populate_and_lookup(indices, offsets)
+        self._prefetched.append(loc)
+
     def _prefetch_cache(
         self, indices: torch.Tensor, offsets: torch.Tensor
     ) -> torch.Tensor:
-        """Populate the lxu cache for this batch; return per-position slots.
+        """Per-position cache slots for this batch: pop a prefetched batch if
+        the pipeline ran prefetch(), else populate inline.
 
         Returns the empty tensor when caching is off (kernels then read the
         weights buffer directly)."""
         if not self._uvm_caching:
             return self._empty_i
+        if self._prefetched:
+            return self._prefetched.popleft()
+        return self._populate_and_lookup(indices, offsets)
+
+    def _populate_and_lookup(
+        self, indices: torch.Tensor, offsets: torch.Tensor
+    ) -> torch.Tensor:
         B = (offsets.numel() - 1) // self._num_features
         _, _, linear, _ = self._bag_metadata(indices, offsets, B)
         sorted_lin, _ = torch.ops.trec_amd.sort_pairs(linear, _bits_needed(self._total_rows))
