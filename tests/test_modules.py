@@ -256,3 +256,27 @@ class TestDeepFMAndCross:
         )
         out = gm(kjt)
         assert out.shape == (1, 1)
+
+
+class TestVbeFusedEBC:
+    def test_vbe_forward(self):
+        from torchrec_amd.modules.fused_embedding_modules import FusedEmbeddingBagCollection
+
+        torch.manual_seed(0)
+        tables = [
+            EmbeddingBagConfig(num_embeddings=30, embedding_dim=8, name="t0", feature_names=["f0"]),
+            EmbeddingBagConfig(num_embeddings=40, embedding_dim=4, name="t1", feature_names=["f1"]),
+        ]
+        ebc = FusedEmbeddingBagCollection(tables, optimizer="sgd")
+        # f0 batch 2, f1 batch 3 (variable stride)
+        kjt = KeyedJaggedTensor(
+            keys=["f0", "f1"],
+            values=torch.tensor([1, 2, 3, 4, 5]),
+            lengths=torch.tensor([2, 1, 1, 0, 1]),
+            stride_per_key_per_rank=[[2], [3]],
+        )
+        kt = ebc(kjt)
+        assert kt.key_dim() == 0
+        assert kt.length_per_key() == [2 * 8, 3 * 4]
+        w0 = ebc.split_embedding_weights()[0]
+        torch.testing.assert_close(kt["f0"][:8], w0[1] + w0[2])

@@ -313,3 +313,76 @@ class TestTBECached:
         for wd, wc in zip(dev.split_embedding_weights(), cached.split_embedding_weights()):
             assert torch.allclose(wc, wd.cpu(), atol=1e-5, rtol=1e-4)
         assert torch.allclose(cached.momentum, dev.momentum.cpu(), atol=1e-6)
+
+
+class TestTBEVbe:
+    def _vbe_ref(self, tbe, indices, offsets, bpf):
+        outs = []
+        bag = 0
+        off0 = 0
+        ws = tbe.split_embedding_weights()
+        for f, (t, bf) in enumerate(zip(tbe._feature_table_map, bpf)):
+            w = ws[t]
+            for b in range(bf):
+                lo, hi = int(offsets[bag]), int(offsets[bag + 1])
+                rows = w[indices[lo:hi]]
+                outs.append(rows.sum(0) if hi > lo else torch.zeros(w.shape[1]))
+                bag += 1
+        return torch.cat(outs)
+
+    def test_vbe_forward_cpu(self):
+        torch.manual_seed(0)
+        specs = [("t0", 30, 8), ("t1", 40, 4)]
+        tbe = TableBatchedEmbeddingBags(specs, optimizer="sgd", learning_rate=0.5)
+        bpf = [3, 5]
+        lengths = torch.randint(0, 4, (sum(bpf),))
+        indices = torch.cat([
+            torch.randint(0, specs[0][1], (int(lengths[:3].sum()),)),
+            torch.randint(0, specs[1][1], (int(lengths[3:].sum()),)),
+        ])
+        offsets = torch.zeros(sum(bpf) + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        out = tbe.forward_vbe(indices, offsets, bpf)
+        ref = self._vbe_ref(tbe, indices, offsets, bpf)
+        torch.testing.assert_close(out, ref, atol=1e-6, rtol=1e-6)
+
+    @pytest.mark.gpu
+    def test_vbe_gpu_matches_cpu_with_update(self):
+        torch.manual_seed(1)
+        specs = [("t0", 50, 64), ("t1", 40, 128)]
+        cpu = TableBatchedEmbeddingBags(specs, optimizer="rowwise_adagrad", learning_rate=0.1)
+        gpu = TableBatchedEmbeddingBags(
+            specs, optimizer="rowwise_adagrad", learning_rate=0.1, device=torch.device("cuda")
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        bpf = [4, 7]
+        lengths = torch.randint(0, 5, (sum(bpf),))
+        indices = torch.cat([
+            torch.randint(0, specs[0][1], (int(lengths[:4].sum()),)),
+            torch.randint(0, specs[1][1], (int(lengths[4:].sum()),)),
+        ])
+        offsets = torch.zeros(sum(bpf) + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        out_c = cpu.forward_vbe(indices, offsets, bpf)
+        out_g = gpu.forward_vbe(indices.cuda(), offsets.cuda(), bpf)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(out_g.cpu(), out_c, atol=1e-5, rtol=1e-5)
+        grad = torch.randn_like(out_c)
+        out_g.backward(grad.cuda())
+        # eager update oracle on cpu
+        w = cpu.weights.detach().requires_grad_(True)
+        import torch.nn.functional as Fn
+        outs = []
+        bag_off = [0, 4, 11]
+        for f, t in enumerate(cpu._feature_table_map):
+            sspec = cpu._specs[t]
+            e0 = int(cpu._table_elem_offsets[t])
+            wt = w[e0 : e0 + sspec.rows * sspec.dim].view(sspec.rows, sspec.dim)
+            off = offsets[bag_off[f] : bag_off[f + 1] + 1] - offsets[bag_off[f]]
+            idx = indices[int(offsets[bag_off[f]]) : int(offsets[bag_off[f + 1]])]
+            outs.append(Fn.embedding_bag(idx, wt, off, mode="sum", include_last_offset=True).reshape(-1))
+        torch.cat(outs).backward(grad)
+        cpu._cpu_apply_update(w.grad)
+        torch.cuda.synchronize()
+        for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+            assert torch.allclose(wg.cpu(), wc, atol=1e-5, rtol=1e-4)

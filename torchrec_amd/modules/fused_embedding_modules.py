@@ -83,6 +83,23 @@ class FusedEmbeddingBagCollection(nn.Module):
         if features.keys() != self._feature_names:
             order = [features.keys().index(f) for f in self._feature_names]
             features = features.permute(order)
+        if features.variable_stride_per_key():
+            # VBE: 1-D packed output [sum_f B_f * D_f], keyed along dim 0
+            bpf = [sum(s) for s in features.stride_per_key_per_rank()]
+            values = self._tbe.forward_vbe(
+                features.values(),
+                features.offsets(),
+                bpf,
+                features.weights_or_none() if self._is_weighted else None,
+            )
+            return KeyedTensor(
+                keys=self._feature_names,
+                values=values,
+                length_per_key=[
+                    b * d for b, d in zip(bpf, self._lengths_per_embedding)
+                ],
+                key_dim=0,
+            )
         values = self._tbe(
             features.values(),
             features.offsets(),

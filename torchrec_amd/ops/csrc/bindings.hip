@@ -34,6 +34,13 @@ at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_el
                            const at::Tensor& dims, const at::Tensor& feat_table,
                            const at::Tensor& feat_val_offsets, const at::Tensor& indices,
                            int64_t D_out, int64_t max_D);
+at::Tensor tbe_forward_pooled_vbe(const at::Tensor& weights,
+                                  const at::Tensor& table_elem_offsets, const at::Tensor& dims,
+                                  const at::Tensor& feat_table, const at::Tensor& bag_offsets,
+                                  const at::Tensor& out_offsets, const at::Tensor& indices,
+                                  const at::Tensor& offsets,
+                                  const at::Tensor& per_sample_weights, int64_t n_bags,
+                                  int64_t out_numel, int64_t max_D, bool mean_pool);
 std::tuple<at::Tensor, at::Tensor> sort_pairs(const at::Tensor& keys, int64_t end_bit);
 std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_linear);
 void tbe_backward_fused(at::Tensor weights, at::Tensor momentum, const at::Tensor& grad,
@@ -120,6 +127,11 @@ TORCH_LIBRARY(trec_amd, m) {
       "tbe_forward_seq(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
       " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"
       " -> Tensor");
+  m.def(
+      "tbe_forward_pooled_vbe(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
+      " Tensor feat_table, Tensor bag_offsets, Tensor out_offsets, Tensor indices,"
+      " Tensor offsets, Tensor per_sample_weights, int n_bags, int out_numel, int max_D,"
+      " bool mean_pool) -> Tensor");
   m.def("sort_pairs(Tensor keys, int end_bit) -> (Tensor, Tensor)");
   m.def("tbe_backward_prep(Tensor sorted_linear) -> (Tensor, Tensor)");
   m.def(
@@ -172,6 +184,7 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("permute_pooled_embs", trec_amd::permute_pooled_embs);
   m.impl("tbe_forward_pooled", trec_amd::tbe_forward_pooled);
   m.impl("tbe_forward_seq", trec_amd::tbe_forward_seq);
+  m.impl("tbe_forward_pooled_vbe", trec_amd::tbe_forward_pooled_vbe);
   m.impl("sort_pairs", trec_amd::sort_pairs);
   m.impl("tbe_backward_prep", trec_amd::tbe_backward_prep);
   m.impl("tbe_backward_fused", trec_amd::tbe_backward_fused);

@@ -171,6 +171,99 @@ at::Tensor tbe_forward_pooled(
 }
 
 // ---------------------------------------------------------------------------
+// forward (pooled, VBE): per-feature batch sizes; output is 1-D packed
+// [sum_f B_f * D_f] feature-major (reference: VBE TBE via generate_vbe_metadata,
+// triton TBE b_t_map path :367). The backward needs no VBE variant: the
+// generic (row, col) grad addressing treats the packed output as one row.
+// ---------------------------------------------------------------------------
+
+template <int LPS, int CHUNKS>
+__global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_vbe_kernel(
+    const float* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
+    const int32_t* __restrict__ dims, const int32_t* __restrict__ feat_table,
+    const int64_t* __restrict__ bag_offsets,   // [F+1] cumsum of B_f
+    const int64_t* __restrict__ out_offsets,   // [F+1] cumsum of B_f * D_f
+    const int64_t* __restrict__ indices, const int64_t* __restrict__ offsets,
+    const float* __restrict__ psw, int F, int64_t n_bags, bool mean_pool,
+    float* __restrict__ out) {
+  int sl = threadIdx.x % LPS;
+  int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
+  int64_t n_slots = (static_cast<int64_t>(gridDim.x) * blockDim.x) / LPS;
+  for (int64_t bag = slot; bag < n_bags; bag += n_slots) {
+    int f = upper_bound_segment(bag_offsets, F, bag);
+    int64_t b = bag - bag_offsets[f];
+    int t = feat_table[f];
+    int D = dims[t];
+    const float* tab = weights + table_elem_offsets[t];
+    float4 acc[CHUNKS];
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) acc[c] = make_float4(0.f, 0.f, 0.f, 0.f);
+    int64_t i0 = offsets[bag], i1 = offsets[bag + 1];
+    for (int64_t i = i0; i < i1; ++i) {
+      const float4* row = reinterpret_cast<const float4*>(tab + indices[i] * D);
+      float w = psw ? psw[i] : 1.f;
+#pragma unroll
+      for (int c = 0; c < CHUNKS; ++c) {
+        int col4 = c * LPS + sl;
+        if (col4 * 4 < D) {
+          float4 v = row[col4];
+          acc[c].x += w * v.x;
+          acc[c].y += w * v.y;
+          acc[c].z += w * v.z;
+          acc[c].w += w * v.w;
+        }
+      }
+    }
+    float scale = 1.f;
+    if (mean_pool && i1 > i0) scale = 1.f / static_cast<float>(i1 - i0);
+    float4* orow = reinterpret_cast<float4*>(out + out_offsets[f] + b * D);
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      int col4 = c * LPS + sl;
+      if (col4 * 4 < D)
+        orow[col4] = make_float4(acc[c].x * scale, acc[c].y * scale, acc[c].z * scale,
+                                 acc[c].w * scale);
+    }
+  }
+}
+
+at::Tensor tbe_forward_pooled_vbe(
+    const at::Tensor& weights, const at::Tensor& table_elem_offsets, const at::Tensor& dims,
+    const at::Tensor& feat_table, const at::Tensor& bag_offsets,
+    const at::Tensor& out_offsets, const at::Tensor& indices, const at::Tensor& offsets,
+    const at::Tensor& per_sample_weights, int64_t n_bags, int64_t out_numel, int64_t max_D,
+    bool mean_pool) {
+  TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048);
+  int F = feat_table.numel();
+  auto out = at::empty({out_numel}, indices.options().dtype(at::kFloat));
+  if (n_bags == 0 || F == 0) return out;
+  const float* psw_ptr =
+      per_sample_weights.numel() > 0 ? per_sample_weights.data_ptr<float>() : nullptr;
+  auto stream = tbe_stream();
+  int lps = (max_D <= 64) ? 16 : (max_D <= 128 ? 32 : 64);
+  int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
+  int grid = grid_for(n_bags * lps, kBlockThreads);
+#define TBE_VBE_LAUNCH(LPS, CHUNKS)                                                      \
+  hipLaunchKernelGGL((tbe_fwd_pooled_vbe_kernel<LPS, CHUNKS>), dim3(grid),               \
+                     dim3(kBlockThreads), 0, stream, uvm_ptr<float>(weights),            \
+                     table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),   \
+                     feat_table.data_ptr<int32_t>(), bag_offsets.data_ptr<int64_t>(),    \
+                     out_offsets.data_ptr<int64_t>(), indices.data_ptr<int64_t>(),       \
+                     offsets.data_ptr<int64_t>(), psw_ptr, F, n_bags, mean_pool,         \
+                     out.data_ptr<float>())
+  if (lps == 16) TBE_VBE_LAUNCH(16, 1);
+  else if (lps == 32) TBE_VBE_LAUNCH(32, 1);
+  else switch (chunks) {
+    case 1: TBE_VBE_LAUNCH(64, 1); break;
+    case 2: TBE_VBE_LAUNCH(64, 2); break;
+    case 3: case 4: TBE_VBE_LAUNCH(64, 4); break;
+    default: TBE_VBE_LAUNCH(64, 8); break;
+  }
+#undef TBE_VBE_LAUNCH
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // forward (sequence / non-pooled): out[n, :] = W[table(f(n))][idx[n]]
 // ---------------------------------------------------------------------------
 

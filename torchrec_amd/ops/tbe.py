@@ -100,6 +100,47 @@ class _TBEPooledFunction(torch.autograd.Function):
         return None, None, None, None, grad_psw
 
 
+class _TBEVbeFunction(torch.autograd.Function):
+    """VBE (variable batch per feature) pooled path: 1-D packed output
+    [sum_f B_f * D_f]; backward reuses the generic fused kernel with the
+    packed output treated as a single grad row."""
+
+    @staticmethod
+    def forward(ctx, dummy, host, indices, offsets, psw, bag_offsets, out_offsets, out_numel):  # type: ignore[override]
+        out = torch.ops.trec_amd.tbe_forward_pooled_vbe(
+            host.weights,
+            host._table_elem_offsets,
+            host._dims_t,
+            host._feat_table_t,
+            bag_offsets,
+            out_offsets,
+            indices,
+            offsets,
+            psw if psw is not None else host._empty_f,
+            offsets.numel() - 1,
+            out_numel,
+            host._max_D,
+            host.pooling_mode == PoolingMode.MEAN,
+        )
+        ctx.host = host
+        ctx.save_for_backward(
+            indices, offsets, psw if psw is not None else host._empty_f,
+            bag_offsets, out_offsets,
+        )
+        ctx.has_psw = psw is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):  # type: ignore[override]
+        host = ctx.host
+        indices, offsets, psw, bag_offsets, out_offsets = ctx.saved_tensors
+        host._backward_vbe(
+            grad.contiguous(), indices, offsets,
+            psw if ctx.has_psw else None, bag_offsets, out_offsets,
+        )
+        return None, None, None, None, None, None, None, None
+
+
 class _TBESeqFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, dummy, host, indices, feat_val_offsets):  # type: ignore[override]
@@ -458,6 +499,100 @@ class TableBatchedEmbeddingBags(nn.Module):
             self.cache_weights,
             self.cache_tags,
             self.cache_weights.shape[1],
+        )
+
+    def forward_vbe(
+        self,
+        indices: torch.Tensor,
+        offsets: torch.Tensor,
+        batch_size_per_feature: List[int],
+        per_sample_weights: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """Variable-batch pooled forward -> 1-D packed [sum_f B_f * D_f]."""
+        F = self._num_features
+        assert len(batch_size_per_feature) == F
+        dims = [self._specs[t].dim for t in self._feature_table_map]
+        bag_off = [0]
+        out_off = [0]
+        for bf, d in zip(batch_size_per_feature, dims):
+            bag_off.append(bag_off[-1] + bf)
+            out_off.append(out_off[-1] + bf * d)
+        if not indices.is_cuda:
+            outs = []
+            for f in range(F):
+                t = self._feature_table_map[f]
+                sspec = self._specs[t]
+                e0 = int(self._table_elem_offsets[t])
+                w = self.weights[e0 : e0 + sspec.rows * sspec.dim].view(sspec.rows, sspec.dim)
+                off = offsets[bag_off[f] : bag_off[f + 1] + 1] - offsets[bag_off[f]]
+                idx = indices[int(offsets[bag_off[f]]) : int(offsets[bag_off[f + 1]])]
+                mode = {PoolingMode.SUM: "sum", PoolingMode.MEAN: "mean"}[self.pooling_mode]
+                outs.append(
+                    torch.nn.functional.embedding_bag(
+                        idx, w, off, mode=mode, include_last_offset=True
+                    ).reshape(-1)
+                )
+            return torch.cat(outs) if outs else self.weights.new_empty(0)
+        ops.hip_ops()
+        device = indices.device
+        bag_offsets = torch.tensor(bag_off, dtype=torch.int64, device=device)
+        out_offsets = torch.tensor(out_off, dtype=torch.int64, device=device)
+        return _TBEVbeFunction.apply(
+            self._dummy, self, indices, offsets, per_sample_weights,
+            bag_offsets, out_offsets, out_off[-1],
+        )
+
+    def _backward_vbe(
+        self,
+        grad: torch.Tensor,
+        indices: torch.Tensor,
+        offsets: torch.Tensor,
+        psw: Optional[torch.Tensor],
+        bag_offsets: torch.Tensor,
+        out_offsets: torch.Tensor,
+    ) -> None:
+        lengths = offsets[1:] - offsets[:-1]
+        n_bags = lengths.numel()
+        bag_ids = torch.repeat_interleave(
+            torch.arange(n_bags, device=indices.device, dtype=torch.int64),
+            lengths,
+            output_size=indices.numel(),
+        )
+        f = torch.searchsorted(bag_offsets, bag_ids, right=True) - 1
+        b = bag_ids - bag_offsets[f]
+        dims64 = self._dims_t.to(torch.int64)[self._feat_table_t.to(torch.int64)[f]]
+        pos_row = torch.zeros_like(bag_ids, dtype=torch.int32)
+        pos_col = out_offsets[f] + b * dims64
+        linear = indices + self._feat_row_offset[f]
+        scale = self._empty_f
+        if self.pooling_mode == PoolingMode.MEAN:
+            inv = 1.0 / lengths.clamp(min=1).to(torch.float32)
+            scale = inv[bag_ids]
+        elif psw is not None:
+            scale = psw
+        sorted_lin, perm = torch.ops.trec_amd.sort_pairs(linear, _bits_needed(self._total_rows))
+        seg_offsets, num_runs = torch.ops.trec_amd.tbe_backward_prep(sorted_lin)
+        torch.ops.trec_amd.tbe_backward_fused(
+            self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
+            self.momentum,
+            grad.view(1, -1),
+            sorted_lin,
+            perm,
+            seg_offsets,
+            num_runs,
+            pos_row,
+            pos_col,
+            scale,
+            self._table_row_offsets,
+            self._table_elem_offsets,
+            self._dims_t,
+            self._max_D,
+            self.learning_rate,
+            self.eps,
+            self.optimizer,
+            self._empty_f,
+            self.cache_weights,
+            self._empty_i,
         )
 
     def _backward_seq(
