@@ -109,7 +109,8 @@ def _dense_to_bf16(dmp) -> None:
         m.to(torch.bfloat16)
 
 
-def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float) -> None:
+def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
+              qcomm: str = "none") -> None:
     import torch.distributed as dist
 
     from torchrec_amd.distributed.model_parallel import DistributedModelParallel
@@ -133,9 +134,14 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float)
         pg = None
 
     model = build_model(scale)
-    sharder = EmbeddingBagCollectionSharder(
-        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
-    )
+    fused_params = {"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    if qcomm != "none":
+        from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
+
+        fused_params["qcomms_config"] = QCommsConfig(
+            forward_precision=CommType(qcomm), backward_precision=CommType(qcomm)
+        )
+    sharder = EmbeddingBagCollectionSharder(fused_params=fused_params)
     planner = EmbeddingShardingPlanner(
         topology=Topology(world_size=world, compute_device="cuda", batch_size=batch_size)
     )
@@ -206,6 +212,7 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float)
             "emb_dtype": "fp32",
             "dense_dtype": "bf16-autocast",
             "parallelism": f"planner/dmp x{world} + pipeline",
+            "qcomm": qcomm,
         },
     }
     if rank == 0:
@@ -228,9 +235,14 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
     device = torch.device("cuda", 0)
     torch.cuda.set_device(device)
     model = build_model(scale)
-    sharder = EmbeddingBagCollectionSharder(
-        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
-    )
+    fused_params = {"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    if qcomm != "none":
+        from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
+
+        fused_params["qcomms_config"] = QCommsConfig(
+            forward_precision=CommType(qcomm), backward_precision=CommType(qcomm)
+        )
+    sharder = EmbeddingBagCollectionSharder(fused_params=fused_params)
     planner = EmbeddingShardingPlanner(
         topology=Topology(world_size=1, compute_device="cuda", batch_size=batch_size)
     )
@@ -363,10 +375,17 @@ if __name__ == "__main__":
         action="store_true",
         help="capture the train step in a hipGraph (single GPU, static shapes)",
     )
+    p.add_argument(
+        "--qcomm",
+        default="none",
+        choices=["none", "fp16", "bf16", "fp8"],
+        help="compress the pooled a2a wire (fwd+bwd)",
+    )
     args = p.parse_args()
     if args.smoke:
         run_smoke()
     elif args.hipgraph and int(os.environ.get("WORLD_SIZE", "1")) == 1:
         run_graph_bench(args.steps, args.warmup, args.batch_size, args.scale)
     else:
-        run_bench(args.gpus, args.steps, args.warmup, args.batch_size, args.scale)
+        run_bench(args.gpus, args.steps, args.warmup, args.batch_size, args.scale,
+                  qcomm=args.qcomm)
