@@ -140,3 +140,36 @@ class TestKeyedTensor:
         kt = KeyedTensor.from_tensor_list(["x", "y"], [torch.ones(2, 2), torch.zeros(2, 1)])
         assert kt.length_per_key() == [2, 1]
         assert kt.values().shape == (2, 3)
+
+
+class TestValidatorsAndInterop:
+    def test_validate_kjt(self):
+        from torchrec_amd.sparse.jagged_tensor_validator import (
+            validate_jagged_tensor,
+            validate_keyed_jagged_tensor,
+        )
+
+        kjt = make_kjt()
+        validate_keyed_jagged_tensor(kjt)
+        validate_jagged_tensor(kjt["f1"])
+        bad = KeyedJaggedTensor(
+            keys=["a"], values=torch.tensor([1, 2, 3]), lengths=torch.tensor([1]), stride=1
+        )
+        import pytest as _pytest
+
+        with _pytest.raises(AssertionError):
+            validate_keyed_jagged_tensor(bad)
+
+    def test_maybe_td_to_kjt(self):
+        from torchrec_amd.sparse.tensor_dict import maybe_td_to_kjt
+
+        kjt = maybe_td_to_kjt(
+            {
+                "a": torch.tensor([[1, 2], [3, 4]]),
+                "b": JaggedTensor(values=torch.tensor([7]), lengths=torch.tensor([1, 0])),
+            },
+            keys=["a", "b"],
+        )
+        assert kjt.keys() == ["a", "b"]
+        assert kjt["a"].values().tolist() == [1, 2, 3, 4]
+        assert kjt["b"].lengths().tolist() == [1, 0]
