@@ -28,8 +28,7 @@ def make_predictor_app(
 ):
     """Build the FastAPI app around a (dense, values, lengths) -> predictions
     callable (typically a quantized sharded model)."""
-    from fastapi import FastAPI
-    from pydantic import BaseModel
+    from fastapi import FastAPI, Request
 
     from torchrec_amd.inference._batching import BatchingQueue, GPUExecutor
 
@@ -40,35 +39,25 @@ def make_predictor_app(
     )
     executor = GPUExecutor(queue, model_fn, num_threads=num_exec_threads)
 
-    class IdList(BaseModel):
-        values: List[int]
-        lengths: List[int]
-
-    class PredictRequest(BaseModel):
-        float_features: List[List[float]]
-        id_list_features: Dict[str, IdList]
-
-    class PredictResponse(BaseModel):
-        predictions: List[float]
-
     app = FastAPI(title="torchrec_amd predictor")
     app.state.queue = queue
     app.state.executor = executor
 
-    @app.post("/predict", response_model=PredictResponse)
-    def predict(req: PredictRequest) -> PredictResponse:
-        dense = torch.tensor(req.float_features, dtype=torch.float32)
+    @app.post("/predict")
+    async def predict(request: Request) -> Dict[str, List[float]]:
+        req = await request.json()
+        dense = torch.tensor(req["float_features"], dtype=torch.float32)
         values = [
-            torch.tensor(req.id_list_features[f].values, dtype=torch.int64)
+            torch.tensor(req["id_list_features"][f]["values"], dtype=torch.int64)
             for f in feature_names
         ]
         lengths = [
-            torch.tensor(req.id_list_features[f].lengths, dtype=torch.int64)
+            torch.tensor(req["id_list_features"][f]["lengths"], dtype=torch.int64)
             for f in feature_names
         ]
         fut = queue.add(dense, values, lengths)
         out = fut.get()
-        return PredictResponse(predictions=out.reshape(-1).tolist())
+        return {"predictions": out.reshape(-1).tolist()}
 
     @app.get("/health")
     def health() -> Dict[str, str]:
