@@ -12,8 +12,6 @@ Requests are funneled through the C++ BatchingQueue + GPUExecutor
 cross-request batched before hitting the GPU.
 """
 
-from __future__ import annotations
-
 from typing import Callable, Dict, List, Optional
 
 import torch
