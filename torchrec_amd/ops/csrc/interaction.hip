@@ -76,8 +76,9 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
     float* __restrict__ d_dense,   // [B, D]
     float* __restrict__ d_sparse   // [B, F, D]
 ) {
-  extern __shared__ float lds[];  // [F1][D/4+1] float4
+  extern __shared__ float lds[];  // [F1][D/4+1] float4 then [P] dz floats
   float4* T4 = reinterpret_cast<float4*>(lds);
+  float* dzbuf = lds + 4 * F1 * (D / 4 + 1);
   int l = lane_id();
   int wave = wave_id();
   const int d4 = D / 4;
@@ -91,8 +92,10 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
       int r = t / d4, c = t - r * d4;
       T4[(r + 1) * stride4 + c] = srow[t];
     }
-    __syncthreads();
     const float* grow = grad_out + b * out_w;
+    // stage the pair-gradient row once (each i-row re-reads F1-1 of them)
+    for (int t = threadIdx.x; t < P; t += blockDim.x) dzbuf[t] = grow[D + t];
+    __syncthreads();
     // when a row needs <= 32 float4 columns, each half-wave takes its own
     // row so no lanes idle (D=128 -> rows i and i+ROWS_PER_WAVE/2)
     const int halves = (d4 <= kWaveSize / 2) ? 2 : 1;
@@ -111,7 +114,7 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
         for (int j = 0; j < F1; ++j) {
           int c = pair_col[i * F1 + j];
           if (c >= 0) {
-            float dz = grow[D + c];
+            float dz = dzbuf[c];
             float4 t = T4[j * stride4 + k];
             acc.x += dz * t.x;
             acc.y += dz * t.y;
@@ -161,7 +164,7 @@ std::tuple<at::Tensor, at::Tensor> interaction_backward(
   auto d_sparse = at::empty_like(sparse);
   if (B == 0) return {d_dense, d_sparse};
   TORCH_CHECK(D % 4 == 0, "interaction kernel needs D %% 4 == 0");
-  int lds_bytes = F1 * (D / 4 + 1) * sizeof(float4);
+  int lds_bytes = F1 * (D / 4 + 1) * sizeof(float4) + P * sizeof(float);
   int grid = std::min<int>(B, kMaxBlocks);
   hipLaunchKernelGGL(interaction_bwd_kernel, dim3(grid), dim3(kBlockThreads), lds_bytes,
                      ia_stream(), grad_out.contiguous().data_ptr<float>(),
