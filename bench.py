@@ -155,10 +155,12 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
         dmp.init_data_parallel()
     if rank == 0:
         counts = {}
+        kcounts = {}
         for mplan in plan.plan.values():
             for ps in mplan.values():
                 counts[ps.sharding_type] = counts.get(ps.sharding_type, 0) + 1
-        print(f"# plan sharding mix: {counts}", flush=True)
+                kcounts[ps.compute_kernel] = kcounts.get(ps.compute_kernel, 0) + 1
+        print(f"# plan sharding mix: {counts} kernels: {kcounts}", flush=True)
     dense_opt = torch.optim.SGD(
         [p for p in dmp.parameters() if p.requires_grad], lr=LR
     )
