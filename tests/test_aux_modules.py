@@ -140,3 +140,38 @@ class TestITEP:
         assert int(v[2]) != 7
         # mapped ids stay stable
         assert int(v[0]) == 0 and int(v[1]) == 3
+
+
+class TestApplyOptimizerInBackward:
+    def test_sgd_in_backward(self):
+        from torchrec_amd.optim.apply_optimizer_in_backward import (
+            apply_optimizer_in_backward,
+        )
+
+        lin = torch.nn.Linear(4, 2, bias=False)
+        w0 = lin.weight.detach().clone()
+        apply_optimizer_in_backward(torch.optim.SGD, lin.parameters(), {"lr": 1.0})
+        x = torch.ones(1, 4)
+        lin(x).sum().backward()
+        # weight already updated, grad cleared
+        assert lin.weight.grad is None
+        expected = w0 - torch.ones_like(w0)
+        torch.testing.assert_close(lin.weight.detach(), expected)
+
+
+class TestSingleHostP2P:
+    def test_merge_and_reduce_cpu(self):
+        from torchrec_amd.distributed.dist_data import (
+            EmbeddingsAllToOne,
+            EmbeddingsAllToOneReduce,
+            merge_pooled_embeddings,
+        )
+
+        a = torch.ones(2, 3)
+        b = torch.full((2, 2), 2.0)
+        out = merge_pooled_embeddings([a, b], torch.device("cpu"))
+        assert out.shape == (2, 5)
+        gather = EmbeddingsAllToOne(torch.device("cpu"), 2)
+        assert gather([a, b]).shape == (2, 5)
+        red = EmbeddingsAllToOneReduce(torch.device("cpu"), 2)
+        torch.testing.assert_close(red([a, a]), 2 * a)

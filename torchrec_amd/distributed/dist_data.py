@@ -340,3 +340,53 @@ class SequenceEmbeddingsAllToAll(nn.Module):
         return SequenceEmbeddingsAwaitable(
             alltoall_sequence(local_embs, fwd_in_splits, fwd_out_splits, self._pg)
         )
+
+
+class EmbeddingsAllToOne(nn.Module):
+    """Single-host inference: gather per-GPU pooled slices onto one device
+    (reference dist_data.py:1632, fbgemm all_to_one_device). On MI355X the
+    copies ride xGMI peer-to-peer links."""
+
+    def __init__(self, device: torch.device, world_size: int, cat_dim: int = 1) -> None:
+        super().__init__()
+        self._device = device
+        self._cat_dim = cat_dim
+
+    def forward(self, tensors: List[torch.Tensor]) -> torch.Tensor:
+        moved = [t.to(self._device, non_blocking=True) for t in tensors]
+        return torch.cat(moved, dim=self._cat_dim)
+
+
+class EmbeddingsAllToOneReduce(nn.Module):
+    """Sum per-GPU partials onto one device (reference :1588,
+    fbgemm sum_reduce_to_one)."""
+
+    def __init__(self, device: torch.device, world_size: int) -> None:
+        super().__init__()
+        self._device = device
+
+    def forward(self, tensors: List[torch.Tensor]) -> torch.Tensor:
+        out = tensors[0].to(self._device, non_blocking=True).clone()
+        for t in tensors[1:]:
+            out += t.to(self._device, non_blocking=True)
+        return out
+
+
+class KJTOneToAll(nn.Module):
+    """Split one KJT's features across local devices (reference :1244)."""
+
+    def __init__(self, splits: List[int], world_size: int, devices: List[torch.device]) -> None:
+        super().__init__()
+        self._splits = splits
+        self._devices = devices
+
+    def forward(self, kjt: KeyedJaggedTensor) -> List[KeyedJaggedTensor]:
+        parts = kjt.split(self._splits)
+        return [p.to(d, non_blocking=True) for p, d in zip(parts, self._devices)]
+
+
+def merge_pooled_embeddings(
+    tensors: List[torch.Tensor], device: torch.device, cat_dim: int = 1
+) -> torch.Tensor:
+    """Reference parity: fbgemm merge_pooled_embeddings (dist_data.py:378)."""
+    return torch.cat([t.to(device, non_blocking=True) for t in tensors], dim=cat_dim)
