@@ -24,12 +24,10 @@ def apply_optimizer_in_backward(
         opt = optimizer_class([param], **optimizer_kwargs)
         param._in_backward_optimizer = opt  # type: ignore[attr-defined]
 
-        acc = param.view_as(param).grad_fn.next_functions[0][0]
-
-        def hook(*_unused, p=param, o=opt):
+        def hook(p, o=opt):
             if p.grad is not None:
                 o.step()
                 p.grad = None
 
-        handle = acc.register_hook(hook)
+        handle = param.register_post_accumulate_grad_hook(hook)
         param._in_backward_optimizer_handle = handle  # type: ignore[attr-defined]
