@@ -161,3 +161,23 @@ def _run_prefetch_pipeline(rank, world_size):
 
 def test_prefetch_pipeline_gloo():
     run_multi_process(_run_prefetch_pipeline, 2, "gloo")
+
+
+def _run_semisync_pipeline(rank, world_size):
+    from torchrec_amd.distributed.train_pipeline import TrainPipelineSemiSync
+
+    dmp_ref, opt_ref, tables = _build_dmp(world_size)
+    batches = _batches(rank, tables)
+    # reference first step (no staleness possible on step 0)
+    loss0_ref, _ = dmp_ref(batches[0])
+
+    dmp, opt, _ = _build_dmp(world_size)
+    pipe = TrainPipelineSemiSync(dmp, opt, torch.device("cpu"))
+    it = iter(batches)
+    losses = [float(pipe.progress(it)[0]) for _ in range(6)]
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert abs(losses[0] - float(loss0_ref.detach())) < 1e-5
+
+
+def test_semisync_pipeline_gloo():
+    run_multi_process(_run_semisync_pipeline, 2, "gloo")

@@ -399,3 +399,99 @@ class _Ready:
 
     def wait(self):
         return self._value
+
+
+class TrainPipelineSemiSync(TrainPipelineSparseDist):
+    """Semi-synchronous pipeline (reference train_pipelines.py:1637): the
+    embedding lookup + output_dist for batch i+1 is issued on the lookup
+    stream BEFORE batch i's backward/optimizer, so the sparse compute and
+    comms hide under the dense backward. Embeddings are therefore one
+    optimizer step stale (the documented semi-sync relaxation).
+    """
+
+    def __init__(self, *args, stash_gradients: bool = False, **kwargs) -> None:
+        super().__init__(*args, **kwargs)
+        self._lookup_stream = (
+            torch.cuda.Stream(priority=-1) if self._device.type == "cuda" else None
+        )
+        self._emb_awaitables: List[Dict[str, Any]] = []
+
+    def _start_embedding_lookup(self, dists: Dict[str, Tuple[Any, Any]]) -> Dict[str, Any]:
+        out: Dict[str, Any] = {}
+        ctxmgr = (
+            torch.cuda.stream(self._lookup_stream)
+            if self._lookup_stream is not None
+            else contextlib.nullcontext()
+        )
+        with torch.autograd.profiler.record_function("## start_embedding_lookup ##"), ctxmgr:
+            for fqn, (ctx, tensors_aw) in dists.items():
+                dist_input = tensors_aw.wait()
+                module = self._pipelined[fqn]
+                out[fqn] = module.compute_and_output_dist(ctx, dist_input)
+        return out
+
+    def _fetch_dist(self, fqn: str):  # PipelinedForward hook
+        raise RuntimeError("semi-sync consumes precomputed lookups")
+
+    def _rewrite_model(self) -> None:
+        for fqn, module in self._model.named_modules():
+            if hasattr(module, "compute_and_output_dist") and hasattr(module, "input_dist"):
+                self._pipelined[fqn] = module
+                pipeline = self
+
+                def make_forward(fq):
+                    def fwd(*args, **kwargs):
+                        if pipeline._lookup_stream is not None:
+                            torch.cuda.current_stream().wait_stream(pipeline._lookup_stream)
+                        return pipeline._emb_awaitables[0][fq]
+
+                    return fwd
+
+                module.forward = make_forward(fqn)
+        self._rewritten = True
+
+    def _fill_pipeline(self, dataloader_iter) -> None:
+        b0 = self._copy_batch_to_gpu(dataloader_iter)
+        if b0 is None:
+            raise StopIteration
+        if not self._rewritten:
+            self._rewrite_model()
+        e0 = self._start_embedding_lookup(self._start_sparse_data_dist(b0))
+        b1 = self._copy_batch_to_gpu(dataloader_iter)
+        d1 = self._start_sparse_data_dist(b1)
+        self._batches = [b0, b1]
+        self._dist_contexts = [d1]
+        self._emb_awaitables = [e0]
+
+    def progress(self, dataloader_iter: Iterator[Pipelineable]) -> Any:
+        if not self._batches:
+            self._fill_pipeline(dataloader_iter)
+        cur_batch = self._batches[0]
+        next_batch = self._batches[1]
+        if cur_batch is None:
+            raise StopIteration
+        batch_ip2 = self._copy_batch_to_gpu(dataloader_iter)
+        _wait_for_batch(cur_batch, self._memcpy_stream)
+        self._optimizer.zero_grad(set_to_none=True)
+        ac = (
+            torch.autocast(device_type=self._device.type, dtype=self._autocast_dtype)
+            if self._autocast_dtype is not None
+            else contextlib.nullcontext()
+        )
+        with torch.autograd.profiler.record_function("## forward ##"), ac:
+            losses, output = self._model(cur_batch)
+        # semi-sync: issue batch i+1's lookup BEFORE batch i's backward
+        e_next = (
+            self._start_embedding_lookup(self._dist_contexts[0])
+            if next_batch is not None
+            else {}
+        )
+        d_ip2 = self._start_sparse_data_dist(batch_ip2)
+        with torch.autograd.profiler.record_function("## backward ##"):
+            torch.sum(losses).backward()
+        with torch.autograd.profiler.record_function("## optimizer ##"):
+            self._optimizer.step()
+        self._batches = [next_batch, batch_ip2]
+        self._dist_contexts = [d_ip2]
+        self._emb_awaitables = [e_next]
+        return output
