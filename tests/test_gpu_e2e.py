@@ -132,3 +132,61 @@ def test_prefetch_pipeline_uvm_caching_cuda():
         out = pipe.progress(it)
         assert torch.isfinite(out[0])
     torch.cuda.synchronize()
+
+
+def test_sharded_ec_cuda():
+    """Sequence EC under DMP on cuda:0 (native seq TBE path)."""
+    from torchrec_amd.distributed.embedding import EmbeddingCollectionSharder
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.types import ShardingEnv
+    from torchrec_amd.modules.embedding_configs import EmbeddingConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingCollection
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    device = torch.device("cuda", 0)
+    tables = [
+        EmbeddingConfig(num_embeddings=64, embedding_dim=32, name=f"t{i}", feature_names=[f"f{i}"])
+        for i in range(2)
+    ]
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.ec = EmbeddingCollection(tables=tables)
+
+        def forward(self, kjt):
+            return self.ec(kjt)
+
+    torch.manual_seed(0)
+    model = M()
+    golden = EmbeddingCollection(tables=tables)
+    for i, cfg in enumerate(tables):
+        golden.embeddings[cfg.name].weight.data.copy_(
+            model.ec.embeddings[cfg.name].weight.data
+        )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=1, compute_device="cuda", batch_size=4)
+    )
+    sharder = EmbeddingCollectionSharder(fused_params={"optimizer": "sgd", "learning_rate": 0.1})
+    plan = planner.plan(model, [sharder])
+    dmp = DistributedModelParallel(
+        model, env=ShardingEnv.from_local(1, 0), plan=plan, sharders=[sharder], device=device
+    )
+    # load golden weights
+    for tbe in dmp.module.ec.tbes():
+        inner = tbe._bags
+        for spec, w in zip(inner.embedding_specs, inner.split_embedding_weights()):
+            w.copy_(golden.embeddings[spec.name].weight.detach().to(w.device))
+    kjt = KeyedJaggedTensor(
+        keys=["f0", "f1"],
+        values=torch.tensor([1, 2, 3, 9, 8, 7]),
+        lengths=torch.tensor([2, 1, 2, 1]),
+        stride=2,
+    ).to(device)
+    out = dmp(kjt)
+    ref = golden(kjt.to(torch.device("cpu")))
+    for k in ["f0", "f1"]:
+        torch.testing.assert_close(out[k].values().cpu(), ref[k].values(), atol=1e-6, rtol=1e-6)
+    torch.cuda.synchronize()
