@@ -45,9 +45,23 @@ infer_ext = CppExtension(
     extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
 )
 
+de_ext = CppExtension(
+    name="torchrec_amd.dynamic_embedding._id_transformer",
+    sources=[
+        os.path.join(
+            os.path.dirname(os.path.abspath(__file__)),
+            "torchrec_amd",
+            "dynamic_embedding",
+            "csrc",
+            "id_transformer.cpp",
+        )
+    ],
+    extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
+)
+
 setup(
     name="torchrec_amd_ext",
     version="0.1.0",
-    ext_modules=[ext, infer_ext],
+    ext_modules=[ext, infer_ext, de_ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=True)},
 )
