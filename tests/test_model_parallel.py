@@ -386,3 +386,61 @@ def _run_optimizer_state_test(rank, world_size):
 
 def test_fused_optimizer_sharded_state():
     run_multi_process(_run_optimizer_state_test, 2, "gloo")
+
+
+def _run_bf16_tables_test(rank, world_size):
+    """BF16 EmbeddingBagConfig.data_type flows into the sharded TBE storage
+    (reference: group_tables by data_type; weights_precision)."""
+    from torchrec_amd.modules.embedding_configs import DataType
+
+    B = 4
+    tables = [
+        EmbeddingBagConfig(
+            num_embeddings=17, embedding_dim=8, name="t0", feature_names=["f0"],
+            data_type=DataType.BF16,
+        ),
+        EmbeddingBagConfig(
+            num_embeddings=33, embedding_dim=16, name="t1", feature_names=["f1"],
+            data_type=DataType.BF16,
+        ),
+    ]
+    torch.manual_seed(42)
+    model = SparseModel(tables)
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(
+                sharding_types=[ShardingType.TABLE_WISE.value]
+            )
+            for cfg in tables
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], device=torch.device("cpu"),
+        init_data_parallel=False,
+    )
+    # local TBE storage is bf16
+    for lookup in dmp.module.sparse._lookups:
+        for tbe in lookup.tbes():
+            assert tbe.weights.dtype == torch.bfloat16
+    kjt_global = make_global_kjt(tables, B * world_size)
+    kjt_local = kjt_local_slice(kjt_global, rank * B, (rank + 1) * B)
+    kt = dmp(kjt_local)
+    vals = kt.values()
+    assert vals.dtype == torch.float32 and vals.shape == (B, 24)
+    vals.sum().backward()  # fused bf16 update
+    # checkpoint round-trip keeps dtype
+    sd = dmp.state_dict()
+    st = sd["sparse.embedding_bags.t0.weight"]
+    shards = st.local_shards() if hasattr(st, "local_shards") else []
+    for shard in shards:
+        assert shard.tensor.dtype == torch.bfloat16
+    dmp.load_state_dict(sd)
+
+
+def test_bf16_data_type_tables():
+    run_multi_process(_run_bf16_tables_test, 2, "gloo")

@@ -386,3 +386,97 @@ class TestTBEVbe:
         torch.cuda.synchronize()
         for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
             assert torch.allclose(wg.cpu(), wc, atol=1e-5, rtol=1e-4)
+
+
+class TestTBEPrecisionCpu:
+    """weights_precision='bf16'/'fp16' storage (reference: SplitTBE
+    weights_precision / EmbeddingBagConfig.data_type). fp32 accumulate."""
+
+    @pytest.mark.parametrize("prec", ["bf16", "fp16"])
+    def test_forward(self, prec):
+        torch.manual_seed(3)
+        tbe = TableBatchedEmbeddingBags(SPECS, weights_precision=prec)
+        assert tbe.weights.dtype == (torch.bfloat16 if prec == "bf16" else torch.float16)
+        indices, offsets = make_inputs(SPECS)
+        out = tbe(indices, offsets)
+        assert out.dtype == torch.float32
+        ref = reference_forward(tbe, indices, offsets, 4)
+        assert torch.allclose(out, ref, atol=1e-5)
+
+    def test_fused_update_tracks_fp32(self):
+        # the bf16 fused step must match the fp32 step to bf16 resolution
+        torch.manual_seed(5)
+        kw = dict(optimizer="rowwise_adagrad", learning_rate=0.1)
+        lo = TableBatchedEmbeddingBags([("t0", 40, 8)], weights_precision="bf16", **kw)
+        hi = TableBatchedEmbeddingBags([("t0", 40, 8)], **kw)
+        hi.weights.data.copy_(lo.weights.data.float())
+        indices = torch.tensor([1, 2, 1, 7])
+        offsets = torch.tensor([0, 2, 4])
+        for _ in range(3):
+            lo(indices, offsets).sum().backward()
+            hi(indices, offsets).sum().backward()
+        wl = lo.split_embedding_weights()[0].float()
+        wh = hi.split_embedding_weights()[0]
+        assert torch.allclose(wl, wh, atol=3e-2, rtol=3e-2)
+        assert not torch.equal(wl, lo.weights.data.new_zeros(wl.shape).float())
+
+    def test_sequence_bf16(self):
+        tbe = TableBatchedEmbeddings([("t0", 30, 8), ("t1", 20, 8)], weights_precision="bf16")
+        indices, offsets = make_inputs([("t0", 30, 8), ("t1", 20, 8)])
+        out = tbe(indices, offsets)
+        assert out.dtype == torch.float32 and out.shape == (indices.numel(), 8)
+        out.sum().backward()  # fused update applies without error
+
+    def test_caching_requires_fp32(self):
+        from torchrec_amd.ops.tbe import EmbeddingLocation
+
+        with pytest.raises(AssertionError):
+            TableBatchedEmbeddingBags(
+                SPECS, weights_precision="bf16",
+                location=EmbeddingLocation.MANAGED_CACHING,
+            )
+
+
+@pytest.mark.gpu
+class TestTBEPrecisionGpu:
+    @pytest.mark.parametrize("prec", ["bf16", "fp16"])
+    def test_forward_backward_matches_cpu(self, prec):
+        specs = [("t0", 100, 8), ("t1", 50, 128), ("t2", 1000, 64)]
+        torch.manual_seed(0)
+        cpu = TableBatchedEmbeddingBags(
+            specs, optimizer="rowwise_adagrad", learning_rate=0.05, weights_precision=prec
+        )
+        gpu = TableBatchedEmbeddingBags(
+            specs, optimizer="rowwise_adagrad", learning_rate=0.05, weights_precision=prec,
+            device=torch.device("cuda"),
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        indices, offsets = make_inputs(specs, B=16, L=7)
+        out_c = cpu(indices, offsets)
+        out_g = gpu(indices.cuda(), offsets.cuda())
+        torch.cuda.synchronize()
+        # forward: identical storage bits, fp32 accumulate both sides
+        assert torch.allclose(out_g.cpu(), out_c, atol=1e-3, rtol=1e-3)
+        out_c.sum().backward()
+        out_g.sum().backward()
+        torch.cuda.synchronize()
+        for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+            assert wg.dtype == wc.dtype
+            # updates computed in fp32, stored at half precision; oracle rounds
+            # once per step the same way
+            assert torch.allclose(wg.cpu().float(), wc.float(), atol=2e-2, rtol=2e-2)
+        for sc, sg in zip(cpu.split_optimizer_states(), gpu.split_optimizer_states()):
+            assert torch.allclose(sg[0].cpu(), sc[0], atol=1e-4, rtol=1e-4)
+
+    def test_dense_bf16_grad(self):
+        specs = [("t0", 60, 64)]
+        gpu = TableBatchedEmbeddingBags(
+            specs, optimizer="dense", weights_precision="bf16", device=torch.device("cuda")
+        )
+        indices, offsets = make_inputs(specs, B=8)
+        out = gpu(indices.cuda(), offsets.cuda())
+        out.sum().backward()
+        torch.cuda.synchronize()
+        assert gpu.weights.grad is not None
+        assert gpu.weights.grad.dtype == torch.bfloat16
+        assert float(gpu.weights.grad.abs().sum()) > 0

@@ -206,6 +206,7 @@ class ShardedEmbeddingCollection(nn.Module):
             local_dim=cfg.embedding_dim,
             pooling=PoolingType.NONE,
             kernel="fused",
+            data_type=getattr(getattr(cfg, "data_type", None), "name", "FP32"),
             feature_names=list(cfg.feature_names),
             full_dim=cfg.embedding_dim,
         )
@@ -222,6 +223,9 @@ class ShardedEmbeddingCollection(nn.Module):
             learning_rate=self._fused_params.get("learning_rate", 0.01),
             eps=self._fused_params.get("eps", 1.0e-8),
             device=self._device,
+            weights_precision={"FP32": "fp32", "FP16": "fp16", "BF16": "bf16"}[
+                tables[0].data_type if tables else "FP32"
+            ],
         )
 
     # -- forward ------------------------------------------------------------

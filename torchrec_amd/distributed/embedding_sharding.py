@@ -54,6 +54,7 @@ class ShardedTableLocal:
     full_dim: int = 0
     full_rows: int = 0
     use_sum_kernel: bool = False  # RW mean: kernel sums, divisor applied later
+    data_type: str = "FP32"  # weights precision (DataType.name): FP32/FP16/BF16
 
 
 def group_tables_by_kernel(
@@ -61,14 +62,15 @@ def group_tables_by_kernel(
 ) -> List[List[ShardedTableLocal]]:
     """Bucket local shards into TBE groups by (kernel, pooling).
 
-    Reference parity: embedding_sharding.py:556 group_tables (data_type and
-    cache grouping collapse here because one flat fp32 TBE handles mixed dims).
+    Reference parity: embedding_sharding.py:556 group_tables (cache grouping
+    collapses here because one flat TBE handles mixed dims; tables with
+    different weights precisions get separate TBE groups/buffers).
     """
-    groups: Dict[Tuple[str, str], List[ShardedTableLocal]] = {}
-    order: List[Tuple[str, str]] = []
+    groups: Dict[Tuple[str, str, str], List[ShardedTableLocal]] = {}
+    order: List[Tuple[str, str, str]] = []
     for t in tables:
         pool = PoolingType.SUM if t.use_sum_kernel else t.pooling
-        key = (t.kernel, pool.value)
+        key = (t.kernel, pool.value, t.data_type)
         if key not in groups:
             groups[key] = []
             order.append(key)
@@ -154,6 +156,9 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                 device=device,
                 location=location,
                 cache_load_factor=fused_params.get("cache_load_factor", 0.2),
+                weights_precision={"FP32": "fp32", "FP16": "fp16", "BF16": "bf16"}[
+                    group[0].data_type
+                ],
             )
             self._emb_modules.append(tbe)
             nf = sum(len(t.feature_names) for t in group)

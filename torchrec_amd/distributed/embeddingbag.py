@@ -392,8 +392,12 @@ class ShardedEmbeddingBagCollection(nn.Module):
         # values, so nothing is overwritten and no keys are reported missing
         for name, p in self.named_parameters():
             state_dict.setdefault(prefix + name, p.data)
-        for name, b in self.named_buffers():
-            state_dict.setdefault(prefix + name, b)
+        for mod_name, mod in self.named_modules():
+            for bname, b in mod._buffers.items():
+                if b is None or bname in mod._non_persistent_buffers_set:
+                    continue  # non-persistent: never a state_dict key
+                full = prefix + (mod_name + "." if mod_name else "") + bname
+                state_dict.setdefault(full, b)
 
     @property
     def fused_optimizer(self) -> FusedOptimizer:

@@ -63,6 +63,7 @@ class CwPooledEmbeddingSharding(TwPooledEmbeddingSharding):
                         local_dim=width,
                         pooling=cfg.pooling,
                         kernel=info.param_sharding.compute_kernel,
+                        data_type=getattr(getattr(cfg, "data_type", None), "name", "FP32"),
                         feature_names=list(cfg.feature_names),
                         col_offset=col_off,
                         full_dim=cfg.embedding_dim,

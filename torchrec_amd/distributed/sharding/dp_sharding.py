@@ -41,6 +41,7 @@ class DpPooledEmbeddingSharding(EmbeddingSharding):
                     pooling=cfg.pooling,
                     # DP must surface grads for allreduce -> dense kernel
                     kernel=EmbeddingComputeKernel.DENSE.value,
+                    data_type=getattr(getattr(cfg, "data_type", None), "name", "FP32"),
                     feature_names=list(cfg.feature_names),
                     full_dim=cfg.embedding_dim,
                 )

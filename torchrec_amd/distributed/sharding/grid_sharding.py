@@ -76,6 +76,7 @@ class GridPooledEmbeddingSharding(TwRwPooledEmbeddingSharding):
                     local_dim=width,
                     pooling=cfg.pooling,
                     kernel=info.param_sharding.compute_kernel,
+                    data_type=getattr(getattr(cfg, "data_type", None), "name", "FP32"),
                     feature_names=list(cfg.feature_names),
                     col_offset=col_off,
                     row_offset=min(my_local * block, cfg.num_embeddings),

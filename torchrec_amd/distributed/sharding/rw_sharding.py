@@ -96,6 +96,7 @@ class RwPooledEmbeddingSharding(EmbeddingSharding):
                     local_dim=cfg.embedding_dim,
                     pooling=cfg.pooling,
                     kernel=info.param_sharding.compute_kernel,
+                    data_type=getattr(getattr(cfg, "data_type", None), "name", "FP32"),
                     feature_names=list(cfg.feature_names),
                     row_offset=min(rank * block, cfg.num_embeddings),
                     full_dim=cfg.embedding_dim,

@@ -82,6 +82,7 @@ class TwPooledEmbeddingSharding(EmbeddingSharding):
                     local_dim=cfg.embedding_dim,
                     pooling=cfg.pooling,
                     kernel=info.param_sharding.compute_kernel,
+                    data_type=getattr(getattr(cfg, "data_type", None), "name", "FP32"),
                     feature_names=list(cfg.feature_names),
                     full_dim=cfg.embedding_dim,
                     full_rows=cfg.num_embeddings,

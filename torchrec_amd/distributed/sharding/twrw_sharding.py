@@ -186,6 +186,7 @@ class TwRwPooledEmbeddingSharding(EmbeddingSharding):
                     local_dim=cfg.embedding_dim,
                     pooling=cfg.pooling,
                     kernel=info.param_sharding.compute_kernel,
+                    data_type=getattr(getattr(cfg, "data_type", None), "name", "FP32"),
                     feature_names=list(cfg.feature_names),
                     row_offset=min(my_local * block, cfg.num_embeddings),
                     full_dim=cfg.embedding_dim,

@@ -53,6 +53,8 @@ class FusedEmbeddingBagCollection(nn.Module):
         ]
         specs = [(t.name, t.num_embeddings, t.embedding_dim) for t in tables]
         feature_table_map = [i for i, t in enumerate(tables) for _ in t.feature_names]
+        precisions = {getattr(getattr(t, "data_type", None), "name", "FP32") for t in tables}
+        assert len(precisions) == 1, "one TBE group per data_type (group before fusing)"
         self._tbe = TableBatchedEmbeddingBags(
             specs,
             feature_table_map=feature_table_map,
@@ -61,6 +63,9 @@ class FusedEmbeddingBagCollection(nn.Module):
             learning_rate=learning_rate,
             eps=eps,
             device=device,
+            weights_precision={"FP32": "fp32", "FP16": "fp16", "BF16": "bf16"}[
+                next(iter(precisions))
+            ],
         )
         for cfg, w in zip(tables, self._tbe.split_embedding_weights()):
             with torch.no_grad():

@@ -61,3 +61,76 @@ __device__ __forceinline__ int upper_bound_segment(const int64_t* offs, int n, i
 }
 
 }  // namespace trec_amd
+
+// -----------------------------------------------------------------------
+// dtype-generic 4-element row accessor for embedding tables.
+// fp32 rows move as float4 (16 B/lane); bf16/fp16 rows as 8 B/lane with
+// fp32 accumulate and round-to-nearest store (reference weights_precision).
+// -----------------------------------------------------------------------
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+namespace trec_amd {
+
+template <typename emb_t>
+struct Vec4;
+
+template <>
+struct Vec4<float> {
+  __device__ static float4 load(const float* row, int col4) {
+    return reinterpret_cast<const float4*>(row)[col4];
+  }
+  __device__ static void store(float* row, int col4, float4 v) {
+    reinterpret_cast<float4*>(row)[col4] = v;
+  }
+};
+
+template <>
+struct Vec4<__hip_bfloat16> {
+  __device__ static float4 load(const __hip_bfloat16* row, int col4) {
+    const ushort4 q = reinterpret_cast<const ushort4*>(row)[col4];
+    __hip_bfloat16 a, b, c, d;
+    *reinterpret_cast<unsigned short*>(&a) = q.x;
+    *reinterpret_cast<unsigned short*>(&b) = q.y;
+    *reinterpret_cast<unsigned short*>(&c) = q.z;
+    *reinterpret_cast<unsigned short*>(&d) = q.w;
+    return make_float4(__bfloat162float(a), __bfloat162float(b), __bfloat162float(c),
+                       __bfloat162float(d));
+  }
+  __device__ static void store(__hip_bfloat16* row, int col4, float4 v) {
+    ushort4 q;
+    __hip_bfloat16 a = __float2bfloat16(v.x), b = __float2bfloat16(v.y),
+                   c = __float2bfloat16(v.z), d = __float2bfloat16(v.w);
+    q.x = *reinterpret_cast<unsigned short*>(&a);
+    q.y = *reinterpret_cast<unsigned short*>(&b);
+    q.z = *reinterpret_cast<unsigned short*>(&c);
+    q.w = *reinterpret_cast<unsigned short*>(&d);
+    reinterpret_cast<ushort4*>(row)[col4] = q;
+  }
+};
+
+template <>
+struct Vec4<__half> {
+  __device__ static float4 load(const __half* row, int col4) {
+    const ushort4 q = reinterpret_cast<const ushort4*>(row)[col4];
+    __half a, b, c, d;
+    *reinterpret_cast<unsigned short*>(&a) = q.x;
+    *reinterpret_cast<unsigned short*>(&b) = q.y;
+    *reinterpret_cast<unsigned short*>(&c) = q.z;
+    *reinterpret_cast<unsigned short*>(&d) = q.w;
+    return make_float4(__half2float(a), __half2float(b), __half2float(c), __half2float(d));
+  }
+  __device__ static void store(__half* row, int col4, float4 v) {
+    ushort4 q;
+    __half a = __float2half(v.x), b = __float2half(v.y), c = __float2half(v.z),
+           d = __float2half(v.w);
+    q.x = *reinterpret_cast<unsigned short*>(&a);
+    q.y = *reinterpret_cast<unsigned short*>(&b);
+    q.z = *reinterpret_cast<unsigned short*>(&c);
+    q.w = *reinterpret_cast<unsigned short*>(&d);
+    reinterpret_cast<ushort4*>(row)[col4] = q;
+  }
+};
+
+}  // namespace trec_amd

@@ -25,6 +25,8 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hipcub/hipcub.hpp>
 
+#include <type_traits>
+
 #include "common.h"
 
 namespace trec_amd {
@@ -34,6 +36,11 @@ namespace trec_amd {
 // UVM support: weights may live in pinned host memory (EmbeddingLocation::
 // MANAGED) — kernels read/write it over PCIe via the device-visible alias.
 // ---------------------------------------------------------------------------
+
+template <typename scalar_t>
+struct DevType { using type = scalar_t; };
+template <> struct DevType<at::Half> { using type = __half; };
+template <> struct DevType<at::BFloat16> { using type = __hip_bfloat16; };
 
 template <typename T>
 static T* uvm_ptr(const at::Tensor& t) {
@@ -53,9 +60,9 @@ static inline hipStream_t tbe_stream() {
 // forward (pooled)
 // ---------------------------------------------------------------------------
 
-template <int LPS, int CHUNKS>
+template <typename emb_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
-    const float* __restrict__ weights,
+    const emb_t* __restrict__ weights,
     const int64_t* __restrict__ table_elem_offsets,  // [T]
     const int32_t* __restrict__ dims,                // [T]
     const int32_t* __restrict__ feat_table,          // [F]
@@ -79,26 +86,24 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
     int b = bag - static_cast<int64_t>(f) * B;
     int t = feat_table[f];
     int D = dims[t];
-    const float* tab = weights + table_elem_offsets[t];
+    const emb_t* tab = weights + table_elem_offsets[t];
     float4 acc[CHUNKS];
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) acc[c] = make_float4(0.f, 0.f, 0.f, 0.f);
     int64_t i0 = offsets[bag], i1 = offsets[bag + 1];
     for (int64_t i = i0; i < i1; ++i) {
       int64_t idx = indices[i];
-      const float4* row;
+      const emb_t* row = tab + idx * D;
+      const float* crow = nullptr;
       if (cache_loc && cache_loc[i] >= 0) {
-        row = reinterpret_cast<const float4*>(cache_weights +
-                                              static_cast<int64_t>(cache_loc[i]) * cache_stride);
-      } else {
-        row = reinterpret_cast<const float4*>(tab + idx * D);
+        crow = cache_weights + static_cast<int64_t>(cache_loc[i]) * cache_stride;
       }
       float w = psw ? psw[i] : 1.f;
 #pragma unroll
       for (int c = 0; c < CHUNKS; ++c) {
         int col4 = c * LPS + sl;
         if (col4 * 4 < D) {
-          float4 v = row[col4];
+          float4 v = crow ? Vec4<float>::load(crow, col4) : Vec4<emb_t>::load(row, col4);
           acc[c].x += w * v.x;
           acc[c].y += w * v.y;
           acc[c].z += w * v.z;
@@ -122,8 +127,9 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
 }
 
 #define TBE_FWD_LAUNCH(LPS, CHUNKS)                                                      \
-  hipLaunchKernelGGL((tbe_fwd_pooled_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads), \
-                     0, stream, uvm_ptr<float>(weights),                                  \
+  hipLaunchKernelGGL((tbe_fwd_pooled_kernel<dev_t, LPS, CHUNKS>), dim3(grid),             \
+                     dim3(kBlockThreads),                                                 \
+                     0, stream, reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights)), \
                      table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),    \
                      feat_table.data_ptr<int32_t>(), d_out_offsets.data_ptr<int64_t>(),   \
                      indices.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(),            \
@@ -136,7 +142,6 @@ at::Tensor tbe_forward_pooled(
     const at::Tensor& offsets, const at::Tensor& per_sample_weights, int64_t B,
     int64_t total_D, int64_t max_D, bool mean_pool, const at::Tensor& cache_weights,
     const at::Tensor& cache_loc) {
-  TORCH_CHECK(weights.scalar_type() == at::kFloat);
   TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048, "TBE: dims must be %4==0 and <=2048");
   int F = feat_table.numel();
   auto out = at::empty({B, total_D}, indices.options().dtype(at::kFloat));
@@ -153,20 +158,28 @@ at::Tensor tbe_forward_pooled(
   int lps = (max_D <= 64) ? 16 : (max_D <= 128 ? 32 : 64);
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
   int grid = grid_for(n_bags * lps, kBlockThreads);
-  if (lps == 16) {
-    TORCH_CHECK(chunks == 1);
-    TBE_FWD_LAUNCH(16, 1);
-  } else if (lps == 32) {
-    TORCH_CHECK(chunks == 1);
-    TBE_FWD_LAUNCH(32, 1);
-  } else {
-    switch (chunks) {
-      case 1: TBE_FWD_LAUNCH(64, 1); break;
-      case 2: TBE_FWD_LAUNCH(64, 2); break;
-      case 3: case 4: TBE_FWD_LAUNCH(64, 4); break;
-      default: TBE_FWD_LAUNCH(64, 8); break;
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
+                                  "tbe_fwd_pooled", [&] {
+    using dev_t = typename DevType<scalar_t>::type;
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 embedding tables unsupported");
+    } else {
+    if (lps == 16) {
+      TORCH_CHECK(chunks == 1);
+      TBE_FWD_LAUNCH(16, 1);
+    } else if (lps == 32) {
+      TORCH_CHECK(chunks == 1);
+      TBE_FWD_LAUNCH(32, 1);
+    } else {
+      switch (chunks) {
+        case 1: TBE_FWD_LAUNCH(64, 1); break;
+        case 2: TBE_FWD_LAUNCH(64, 2); break;
+        case 3: case 4: TBE_FWD_LAUNCH(64, 4); break;
+        default: TBE_FWD_LAUNCH(64, 8); break;
+      }
     }
-  }
+    }
+  });
   return out;
 }
 
@@ -177,9 +190,9 @@ at::Tensor tbe_forward_pooled(
 // generic (row, col) grad addressing treats the packed output as one row.
 // ---------------------------------------------------------------------------
 
-template <int LPS, int CHUNKS>
+template <typename emb_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_vbe_kernel(
-    const float* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
+    const emb_t* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
     const int32_t* __restrict__ dims, const int32_t* __restrict__ feat_table,
     const int64_t* __restrict__ bag_offsets,   // [F+1] cumsum of B_f
     const int64_t* __restrict__ out_offsets,   // [F+1] cumsum of B_f * D_f
@@ -194,19 +207,19 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_vbe_kernel(
     int64_t b = bag - bag_offsets[f];
     int t = feat_table[f];
     int D = dims[t];
-    const float* tab = weights + table_elem_offsets[t];
+    const emb_t* tab = weights + table_elem_offsets[t];
     float4 acc[CHUNKS];
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) acc[c] = make_float4(0.f, 0.f, 0.f, 0.f);
     int64_t i0 = offsets[bag], i1 = offsets[bag + 1];
     for (int64_t i = i0; i < i1; ++i) {
-      const float4* row = reinterpret_cast<const float4*>(tab + indices[i] * D);
+      const emb_t* row = tab + indices[i] * D;
       float w = psw ? psw[i] : 1.f;
 #pragma unroll
       for (int c = 0; c < CHUNKS; ++c) {
         int col4 = c * LPS + sl;
         if (col4 * 4 < D) {
-          float4 v = row[col4];
+          float4 v = Vec4<emb_t>::load(row, col4);
           acc[c].x += w * v.x;
           acc[c].y += w * v.y;
           acc[c].z += w * v.z;
@@ -244,21 +257,30 @@ at::Tensor tbe_forward_pooled_vbe(
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
   int grid = grid_for(n_bags * lps, kBlockThreads);
 #define TBE_VBE_LAUNCH(LPS, CHUNKS)                                                      \
-  hipLaunchKernelGGL((tbe_fwd_pooled_vbe_kernel<LPS, CHUNKS>), dim3(grid),               \
-                     dim3(kBlockThreads), 0, stream, uvm_ptr<float>(weights),            \
+  hipLaunchKernelGGL((tbe_fwd_pooled_vbe_kernel<dev_t, LPS, CHUNKS>), dim3(grid),        \
+                     dim3(kBlockThreads), 0, stream,                                     \
+                     reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights)),         \
                      table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),   \
                      feat_table.data_ptr<int32_t>(), bag_offsets.data_ptr<int64_t>(),    \
                      out_offsets.data_ptr<int64_t>(), indices.data_ptr<int64_t>(),       \
                      offsets.data_ptr<int64_t>(), psw_ptr, F, n_bags, mean_pool,         \
                      out.data_ptr<float>())
-  if (lps == 16) TBE_VBE_LAUNCH(16, 1);
-  else if (lps == 32) TBE_VBE_LAUNCH(32, 1);
-  else switch (chunks) {
-    case 1: TBE_VBE_LAUNCH(64, 1); break;
-    case 2: TBE_VBE_LAUNCH(64, 2); break;
-    case 3: case 4: TBE_VBE_LAUNCH(64, 4); break;
-    default: TBE_VBE_LAUNCH(64, 8); break;
-  }
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
+                                  "tbe_fwd_vbe", [&] {
+    using dev_t = typename DevType<scalar_t>::type;
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 embedding tables unsupported");
+    } else {
+    if (lps == 16) TBE_VBE_LAUNCH(16, 1);
+    else if (lps == 32) TBE_VBE_LAUNCH(32, 1);
+    else switch (chunks) {
+      case 1: TBE_VBE_LAUNCH(64, 1); break;
+      case 2: TBE_VBE_LAUNCH(64, 2); break;
+      case 3: case 4: TBE_VBE_LAUNCH(64, 4); break;
+      default: TBE_VBE_LAUNCH(64, 8); break;
+    }
+    }
+  });
 #undef TBE_VBE_LAUNCH
   return out;
 }
@@ -267,9 +289,9 @@ at::Tensor tbe_forward_pooled_vbe(
 // forward (sequence / non-pooled): out[n, :] = W[table(f(n))][idx[n]]
 // ---------------------------------------------------------------------------
 
-template <int LPS, int CHUNKS>
+template <typename emb_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_seq_kernel(
-    const float* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
+    const emb_t* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
     const int32_t* __restrict__ dims, const int32_t* __restrict__ feat_table,
     const int64_t* __restrict__ feat_val_offsets,  // [F+1] value range per feature
     const int64_t* __restrict__ indices, int F, int64_t N, int64_t D_out,
@@ -281,13 +303,12 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_seq_kernel(
     int f = upper_bound_segment(feat_val_offsets, F, n);
     int t = feat_table[f];
     int D = dims[t];
-    const float4* row =
-        reinterpret_cast<const float4*>(weights + table_elem_offsets[t] + indices[n] * D);
+    const emb_t* row = weights + table_elem_offsets[t] + indices[n] * D;
     float4* orow = reinterpret_cast<float4*>(out + n * D_out);
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
       int col4 = c * LPS + sl;
-      if (col4 * 4 < D) orow[col4] = row[col4];
+      if (col4 * 4 < D) orow[col4] = Vec4<emb_t>::load(row, col4);
     }
   }
 }
@@ -306,19 +327,28 @@ at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_el
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
   int grid = grid_for(N * lps, kBlockThreads);
 #define TBE_SEQ_LAUNCH(LPS, CHUNKS)                                                        \
-  hipLaunchKernelGGL((tbe_fwd_seq_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads), 0, \
-                     stream, uvm_ptr<float>(weights),                                      \
+  hipLaunchKernelGGL((tbe_fwd_seq_kernel<dev_t, LPS, CHUNKS>), dim3(grid),                  \
+                     dim3(kBlockThreads), 0,                                                \
+                     stream, reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights)),    \
                      table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),     \
                      feat_table.data_ptr<int32_t>(), feat_val_offsets.data_ptr<int64_t>(), \
                      indices.data_ptr<int64_t>(), F, N, D_out, out.data_ptr<float>())
-  if (lps == 16) TBE_SEQ_LAUNCH(16, 1);
-  else if (lps == 32) TBE_SEQ_LAUNCH(32, 1);
-  else switch (chunks) {
-    case 1: TBE_SEQ_LAUNCH(64, 1); break;
-    case 2: TBE_SEQ_LAUNCH(64, 2); break;
-    case 3: case 4: TBE_SEQ_LAUNCH(64, 4); break;
-    default: TBE_SEQ_LAUNCH(64, 8); break;
-  }
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
+                                  "tbe_fwd_seq", [&] {
+    using dev_t = typename DevType<scalar_t>::type;
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 embedding tables unsupported");
+    } else {
+    if (lps == 16) TBE_SEQ_LAUNCH(16, 1);
+    else if (lps == 32) TBE_SEQ_LAUNCH(32, 1);
+    else switch (chunks) {
+      case 1: TBE_SEQ_LAUNCH(64, 1); break;
+      case 2: TBE_SEQ_LAUNCH(64, 2); break;
+      case 3: case 4: TBE_SEQ_LAUNCH(64, 4); break;
+      default: TBE_SEQ_LAUNCH(64, 8); break;
+    }
+    }
+  });
 #undef TBE_SEQ_LAUNCH
   return out;
 }
@@ -535,9 +565,9 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
 // mode: 0 = SGD, 1 = rowwise Adagrad, 2 = dense grad (write grad_weights).
 // ---------------------------------------------------------------------------
 
-template <int LPS, int CHUNKS>
+template <typename emb_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
-    float* __restrict__ weights, float* __restrict__ momentum,
+    emb_t* __restrict__ weights, float* __restrict__ momentum,
     const float* __restrict__ grad, int64_t grad_stride,
     const int64_t* __restrict__ sorted_linear, const int32_t* __restrict__ sort_perm,
     const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
@@ -548,7 +578,7 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     const int64_t* __restrict__ table_row_offsets,   // [T+1] rows
     const int64_t* __restrict__ table_elem_offsets,  // [T]
     const int32_t* __restrict__ dims, int T, float lr, float eps, int mode,
-    float* __restrict__ grad_weights, float* __restrict__ cache_weights,
+    emb_t* __restrict__ grad_weights, float* __restrict__ cache_weights,
     const int32_t* __restrict__ cache_loc, int64_t cache_stride) {
   int sl = threadIdx.x % LPS;
   int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
@@ -599,15 +629,11 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
         }
       }
     }
-    float4* wrow;
+    // weight row: emb_t in the table, fp32 if the row sits in the lxu cache
+    emb_t* wrow = weights + table_elem_offsets[t] + local * static_cast<int64_t>(D);
+    float* crow = nullptr;
     int32_t cloc = cache_loc ? cache_loc[sort_perm[k0]] : -1;
-    if (cloc >= 0) {
-      wrow = reinterpret_cast<float4*>(cache_weights +
-                                       static_cast<int64_t>(cloc) * cache_stride);
-    } else {
-      wrow = reinterpret_cast<float4*>(weights + table_elem_offsets[t] +
-                                       local * static_cast<int64_t>(D));
-    }
+    if (cloc >= 0) crow = cache_weights + static_cast<int64_t>(cloc) * cache_stride;
     if (mode == 1) {
       // rowwise Adagrad: m += mean(g^2); w -= lr * g / (sqrt(m) + eps)
       float gsq = 0.f;
@@ -626,12 +652,13 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
       for (int c = 0; c < CHUNKS; ++c) {
         int col4 = c * LPS + sl;
         if (col4 * 4 < D) {
-          float4 w = wrow[col4];
+          float4 w = crow ? Vec4<float>::load(crow, col4) : Vec4<emb_t>::load(wrow, col4);
           w.x -= step * acc[c].x;
           w.y -= step * acc[c].y;
           w.z -= step * acc[c].z;
           w.w -= step * acc[c].w;
-          wrow[col4] = w;
+          if (crow) Vec4<float>::store(crow, col4, w);
+          else Vec4<emb_t>::store(wrow, col4, w);
         }
       }
     } else if (mode == 0) {
@@ -639,21 +666,21 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
       for (int c = 0; c < CHUNKS; ++c) {
         int col4 = c * LPS + sl;
         if (col4 * 4 < D) {
-          float4 w = wrow[col4];
+          float4 w = crow ? Vec4<float>::load(crow, col4) : Vec4<emb_t>::load(wrow, col4);
           w.x -= lr * acc[c].x;
           w.y -= lr * acc[c].y;
           w.z -= lr * acc[c].z;
           w.w -= lr * acc[c].w;
-          wrow[col4] = w;
+          if (crow) Vec4<float>::store(crow, col4, w);
+          else Vec4<emb_t>::store(wrow, col4, w);
         }
       }
     } else {
-      float4* gw = reinterpret_cast<float4*>(grad_weights + table_elem_offsets[t] +
-                                             local * static_cast<int64_t>(D));
+      emb_t* gw = grad_weights + table_elem_offsets[t] + local * static_cast<int64_t>(D);
 #pragma unroll
       for (int c = 0; c < CHUNKS; ++c) {
         int col4 = c * LPS + sl;
-        if (col4 * 4 < D) gw[col4] = acc[c];
+        if (col4 * 4 < D) Vec4<emb_t>::store(gw, col4, acc[c]);
       }
     }
   }
@@ -676,7 +703,6 @@ void tbe_backward_fused(
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
   int grid = grid_for(n * lps, kBlockThreads);  // upper bound: runs <= n
   const float* scale_ptr = pos_scale.numel() > 0 ? pos_scale.data_ptr<float>() : nullptr;
-  float* gw_ptr = grad_weights.numel() > 0 ? grad_weights.data_ptr<float>() : nullptr;
   float* cache_w_ptr = cache_weights.numel() > 0 ? cache_weights.data_ptr<float>() : nullptr;
   const int32_t* cache_loc_ptr =
       cache_loc.numel() > 0 ? cache_loc.data_ptr<int32_t>() : nullptr;
@@ -698,8 +724,9 @@ void tbe_backward_fused(
                        pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
                        table_row_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,   \
                        kChunkSize, max_D, scratch.data_ptr<float>());                        \
-    hipLaunchKernelGGL((tbe_bwd_fused_kernel<LPS, CHUNKS>), dim3(grid), dim3(kBlockThreads), \
-                       0, stream, uvm_ptr<float>(weights),                                   \
+    hipLaunchKernelGGL((tbe_bwd_fused_kernel<dev_t, LPS, CHUNKS>), dim3(grid),               \
+                       dim3(kBlockThreads),                                                  \
+                       0, stream, reinterpret_cast<dev_t*>(uvm_ptr<scalar_t>(weights)),      \
                        uvm_ptr<float>(momentum),                                             \
                        grad.data_ptr<float>(), grad.size(1),                                 \
                        sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
@@ -711,14 +738,24 @@ void tbe_backward_fused(
                        (float)lr, (float)eps, (int)mode, gw_ptr, cache_w_ptr,                \
                        cache_loc_ptr, max_D);                                                \
   } while (0)
-  if (lps == 16) TBE_BWD_LAUNCH(16, 1);
-  else if (lps == 32) TBE_BWD_LAUNCH(32, 1);
-  else switch (chunks) {
-    case 1: TBE_BWD_LAUNCH(64, 1); break;
-    case 2: TBE_BWD_LAUNCH(64, 2); break;
-    case 3: case 4: TBE_BWD_LAUNCH(64, 4); break;
-    default: TBE_BWD_LAUNCH(64, 8); break;
-  }
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
+                                  "tbe_bwd_fused", [&] {
+    using dev_t = typename DevType<scalar_t>::type;
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 embedding tables unsupported");
+    } else {
+    dev_t* gw_ptr = grad_weights.numel() > 0
+        ? reinterpret_cast<dev_t*>(grad_weights.data_ptr<scalar_t>()) : nullptr;
+    if (lps == 16) TBE_BWD_LAUNCH(16, 1);
+    else if (lps == 32) TBE_BWD_LAUNCH(32, 1);
+    else switch (chunks) {
+      case 1: TBE_BWD_LAUNCH(64, 1); break;
+      case 2: TBE_BWD_LAUNCH(64, 2); break;
+      case 3: case 4: TBE_BWD_LAUNCH(64, 4); break;
+      default: TBE_BWD_LAUNCH(64, 8); break;
+    }
+    }
+  });
 #undef TBE_BWD_LAUNCH
 }
 
@@ -749,9 +786,9 @@ at::Tensor gather_run_heads(const at::Tensor& sorted_linear, const at::Tensor& s
 // grad wrt per-sample weights: dL/dw_i = dot(grad_out_bag, W[idx_i])
 // ---------------------------------------------------------------------------
 
-template <int LPS>
+template <typename emb_t, int LPS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_grad_psw_kernel(
-    const float* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
+    const emb_t* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
     const int32_t* __restrict__ dims, const float* __restrict__ grad, int64_t grad_stride,
     const int64_t* __restrict__ indices, const int32_t* __restrict__ pos_row,
     const int64_t* __restrict__ pos_col, const int32_t* __restrict__ pos_table, int64_t N,
@@ -762,10 +799,11 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_grad_psw_kernel(
   for (int64_t p = slot; p < N; p += n_slots) {
     int t = pos_table[p];
     int D = dims[t];
-    const float* row = weights + table_elem_offsets[t] + indices[p] * static_cast<int64_t>(D);
+    const emb_t* row = weights + table_elem_offsets[t] + indices[p] * static_cast<int64_t>(D);
     const float* grow = grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p];
     float acc = 0.f;
-    for (int d = sl; d < D; d += LPS) acc += row[d] * grow[d];
+    for (int d = sl; d < D; d += LPS)
+      acc += static_cast<float>(row[d]) * grow[d];
     acc = group_reduce_sum<LPS>(acc);
     if (sl == 0) grad_psw[p] = acc;
   }
@@ -780,12 +818,21 @@ at::Tensor tbe_grad_per_sample_weights(
   if (N == 0) return out;
   auto stream = tbe_stream();
   int grid = grid_for(N * 16, kBlockThreads);
-  hipLaunchKernelGGL((tbe_grad_psw_kernel<16>), dim3(grid), dim3(kBlockThreads), 0, stream,
-                     uvm_ptr<float>(weights), table_elem_offsets.data_ptr<int64_t>(),
-                     dims.data_ptr<int32_t>(), grad.data_ptr<float>(), grad.size(1),
-                     indices.data_ptr<int64_t>(), pos_row.data_ptr<int32_t>(),
-                     pos_col.data_ptr<int64_t>(), pos_table.data_ptr<int32_t>(), N,
-                     out.data_ptr<float>());
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
+                                  "tbe_grad_psw", [&] {
+    using dev_t = typename DevType<scalar_t>::type;
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 embedding tables unsupported");
+    } else {
+    hipLaunchKernelGGL((tbe_grad_psw_kernel<dev_t, 16>), dim3(grid), dim3(kBlockThreads), 0,
+                       stream, reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights)),
+                       table_elem_offsets.data_ptr<int64_t>(),
+                       dims.data_ptr<int32_t>(), grad.data_ptr<float>(), grad.size(1),
+                       indices.data_ptr<int64_t>(), pos_row.data_ptr<int32_t>(),
+                       pos_col.data_ptr<int64_t>(), pos_table.data_ptr<int32_t>(), N,
+                       out.data_ptr<float>());
+    }
+  });
   return out;
 }
 

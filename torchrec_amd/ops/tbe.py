@@ -189,10 +189,23 @@ class TableBatchedEmbeddingBags(nn.Module):
         init_max: float = 0.01,
         location: EmbeddingLocation = EmbeddingLocation.DEVICE,
         cache_load_factor: float = 0.2,
+        weights_precision: str = "fp32",
     ) -> None:
         super().__init__()
         device = device or torch.device("cpu")
         self.location = location
+        # table storage precision (reference: SplitTBE weights_precision /
+        # EmbeddingBagConfig.data_type). Grad accumulate + pooled output stay
+        # fp32; bf16/fp16 rows move as 8 B/lane (Vec4<emb_t> in common.h).
+        self.weights_precision = weights_precision
+        self._weights_dtype = {
+            "fp32": torch.float32,
+            "fp16": torch.float16,
+            "bf16": torch.bfloat16,
+        }[weights_precision]
+        assert weights_precision == "fp32" or location != EmbeddingLocation.MANAGED_CACHING, (
+            "lxu-cached (MANAGED_CACHING) tables must be fp32"
+        )
         # MANAGED: weights live in pinned host DRAM, addressed by the HIP
         # kernels over PCIe (reference: FBGEMM EmbeddingLocation.MANAGED);
         # MANAGED_CACHING adds a set-associative LRU cache in HBM (lxu cache)
@@ -235,7 +248,9 @@ class TableBatchedEmbeddingBags(nn.Module):
         self._total_D = d_out[-1]
         self._max_D = max(dims) if dims else 0
 
-        weights = torch.empty(self._total_elems, dtype=torch.float32, device=weights_device)
+        weights = torch.empty(
+            self._total_elems, dtype=self._weights_dtype, device=weights_device
+        )
         if weights_device.type != "meta":
             weights.uniform_(init_min, init_max)
         if self._uvm:
@@ -648,19 +663,24 @@ class TableBatchedEmbeddingBags(nn.Module):
         return _TBECpuFusedFunction.apply(self._dummy, self, indices, offsets, psw, B)
 
     def _cpu_apply_update(self, grad_flat: torch.Tensor) -> None:
-        """Apply fused optimizer given a dense flat gradient (CPU oracle)."""
+        """Apply fused optimizer given a dense flat gradient (CPU oracle).
+        Math in fp32; weights are written back in their storage dtype."""
         with torch.no_grad():
+            gf = grad_flat.float()
             if self.optimizer == OPT_ROWWISE_ADAGRAD:
                 for i, s in enumerate(self._specs):
                     e0 = int(self._table_elem_offsets[i])
                     r0 = int(self._table_row_offsets[i])
-                    g = grad_flat[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
+                    g = gf[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
                     m = self.momentum[r0 : r0 + s.rows]
                     m += g.pow(2).mean(dim=1)
                     w = self.weights[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
-                    w -= self.learning_rate * g / (m.sqrt() + self.eps).unsqueeze(1)
+                    w.copy_(
+                        w.float()
+                        - self.learning_rate * g / (m.sqrt() + self.eps).unsqueeze(1)
+                    )
             elif self.optimizer == OPT_SGD:
-                self.weights -= self.learning_rate * grad_flat
+                self.weights.copy_(self.weights.float() - self.learning_rate * gf)
 
 
 def _tbe_cpu_forward(weights, host, indices, offsets, psw, B):
@@ -672,6 +692,8 @@ def _tbe_cpu_forward(weights, host, indices, offsets, psw, B):
         s = host._specs[t]
         e0 = int(host._table_elem_offsets[t])
         w = weights[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
+        if w.dtype != torch.float32:
+            w = w.float()  # fp32 oracle math; grads flow back through the cast
         off = offsets[f * B : (f + 1) * B + 1] - offsets[f * B]
         idx = indices[int(offsets[f * B]) : int(offsets[(f + 1) * B])]
         pw = (
@@ -706,7 +728,7 @@ class _TBECpuFusedFunction(torch.autograd.Function):
         B = ctx.B
         indices, offsets, psw = ctx.saved_tensors
         psw_t = psw if ctx.has_psw else None
-        w = host.weights.detach().requires_grad_(True)
+        w = host.weights.detach().float().requires_grad_(True)
         with torch.enable_grad():
             out = _tbe_cpu_forward(w, host, indices, offsets, psw_t, B)
             out.backward(grad)
@@ -772,6 +794,7 @@ class TableBatchedEmbeddings(nn.Module):
         device: Optional[torch.device] = None,
         init_min: float = -0.01,
         init_max: float = 0.01,
+        weights_precision: str = "fp32",
     ) -> None:
         super().__init__()
         dims = {s[2] for s in embedding_specs}
@@ -786,6 +809,7 @@ class TableBatchedEmbeddings(nn.Module):
             device=device,
             init_min=init_min,
             init_max=init_max,
+            weights_precision=weights_precision,
         )
         self._dim = next(iter(dims))
 
@@ -838,7 +862,7 @@ class _TBESeqCpuFunction(torch.autograd.Function):
                 e0 = int(host._table_elem_offsets[t])
                 w = host.weights[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
                 idx = indices[int(offsets[f * B]) : int(offsets[(f + 1) * B])]
-                outs.append(w[idx])
+                outs.append(w[idx].float())
             return torch.cat(outs, dim=0)
 
     @staticmethod
@@ -846,7 +870,9 @@ class _TBESeqCpuFunction(torch.autograd.Function):
         host = ctx.host
         B = ctx.B
         indices, offsets = ctx.saved_tensors
-        grad_flat = torch.zeros_like(host.weights)
+        grad_flat = torch.zeros(
+            host.weights.shape, dtype=torch.float32, device=host.weights.device
+        )
         for f in range(host._num_features):
             t = host._feature_table_map[f]
             s = host._specs[t]
