@@ -73,6 +73,12 @@ __device__ __forceinline__ int upper_bound_segment(const int64_t* offs, int n, i
 
 namespace trec_amd {
 
+// scalar emb_t -> fp32 (PyTorch extension builds define
+// __HIP_NO_HALF_CONVERSIONS__, so implicit casts are unavailable)
+__device__ __forceinline__ float emb2float(float v) { return v; }
+__device__ __forceinline__ float emb2float(__half v) { return __half2float(v); }
+__device__ __forceinline__ float emb2float(__hip_bfloat16 v) { return __bfloat162float(v); }
+
 template <typename emb_t>
 struct Vec4;
 

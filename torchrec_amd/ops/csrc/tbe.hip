@@ -803,7 +803,7 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_grad_psw_kernel(
     const float* grow = grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p];
     float acc = 0.f;
     for (int d = sl; d < D; d += LPS)
-      acc += static_cast<float>(row[d]) * grow[d];
+      acc += emb2float(row[d]) * grow[d];
     acc = group_reduce_sum<LPS>(acc);
     if (sl == 0) grad_psw[p] = acc;
   }
