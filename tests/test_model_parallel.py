@@ -444,3 +444,72 @@ def _run_bf16_tables_test(rank, world_size):
 
 def test_bf16_data_type_tables():
     run_multi_process(_run_bf16_tables_test, 2, "gloo")
+
+
+def _run_twcw_test(rank, world_size):
+    """TWCW on 4 ranks as 2 nodes x 2 local: every table's column shards land
+    on ONE node (reference sharding/twcw_sharding.py)."""
+    import os
+
+    os.environ["LOCAL_WORLD_SIZE"] = "2"
+    os.environ["LOCAL_RANK"] = str(rank % 2)
+    B = 4
+    tables = make_tables()
+    torch.manual_seed(42)
+    model = SparseModel(make_tables())
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(
+            world_size=world_size, compute_device="cpu", hbm_cap=1 << 40,
+            local_world_size=2,
+        ),
+        constraints={
+            cfg.name: ParameterConstraints(
+                sharding_types=[ShardingType.TABLE_COLUMN_WISE.value],
+                min_partition=4,
+            )
+            for cfg in tables
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    # node locality: all shards of a table on one 2-rank node
+    for ps in plan.plan["sparse"].values():
+        assert ps.sharding_type == ShardingType.TABLE_COLUMN_WISE.value
+        nodes = {r // 2 for r in ps.ranks}
+        assert len(nodes) == 1, f"TWCW shards crossed nodes: {ps.ranks}"
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    golden = _golden(tables, None, world_size)
+    dmp.load_state_dict(
+        {
+            f"sparse.embedding_bags.{cfg.name}.weight": w
+            for cfg, w in zip(tables, golden.split_embedding_weights())
+        },
+        strict=False,
+    )
+    kjt_global = make_global_kjt(tables, B * world_size)
+    kjt_local = kjt_local_slice(kjt_global, rank * B, (rank + 1) * B)
+    kt = dmp(kjt_local)
+    vals = kt.values()
+    golden_out = golden(kjt_global).values()
+    expected = golden_out[rank * B : (rank + 1) * B]
+    torch.testing.assert_close(vals, expected, atol=1e-5, rtol=1e-5)
+    # fused update parity
+    vals.sum().backward()
+    (golden_out.sum() / world_size).backward()
+    sharded_sd = dmp.state_dict()
+    for cfg, gw in zip(tables, golden.split_embedding_weights()):
+        st = sharded_sd[f"sparse.embedding_bags.{cfg.name}.weight"]
+        for shard in st.local_shards():
+            ro, co = shard.metadata.shard_offsets
+            h, w = shard.metadata.shard_sizes
+            torch.testing.assert_close(
+                shard.tensor, gw[ro : ro + h, co : co + w], atol=1e-4, rtol=1e-4
+            )
+
+
+def test_twcw_sharding():
+    run_multi_process(_run_twcw_test, 4, "gloo")

@@ -52,6 +52,12 @@ def create_sharding(
         return RwPooledEmbeddingSharding(infos, env, device)
     if sharding_type == ShardingType.COLUMN_WISE.value:
         return CwPooledEmbeddingSharding(infos, env, device)
+    if sharding_type == ShardingType.TABLE_COLUMN_WISE.value:
+        from torchrec_amd.distributed.sharding.twcw_sharding import (
+            TwCwPooledEmbeddingSharding,
+        )
+
+        return TwCwPooledEmbeddingSharding(infos, env, device)
     if sharding_type == ShardingType.DATA_PARALLEL.value:
         return DpPooledEmbeddingSharding(infos, env, device)
     if sharding_type == ShardingType.TABLE_ROW_WISE.value:

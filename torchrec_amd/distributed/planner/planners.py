@@ -91,9 +91,16 @@ def calculate_shards(
             hi = min((r + 1) * block, rows)
             out.append(([hi - lo, cols], [lo, 0]))
         return out
-    if sharding_type == ShardingType.COLUMN_WISE.value:
+    if sharding_type in (
+        ShardingType.COLUMN_WISE.value,
+        ShardingType.TABLE_COLUMN_WISE.value,
+    ):
         # split into shards of >= min_partition cols, at most world_size shards
-        n_shards = min(world_size, max(1, cols // max(4, min_partition)))
+        # (TWCW: at most local_size shards so the whole table fits one node)
+        cap = min(local_size, world_size) if (
+            sharding_type == ShardingType.TABLE_COLUMN_WISE.value
+        ) else world_size
+        n_shards = min(cap, max(1, cols // max(4, min_partition)))
         base = cols // n_shards
         base -= base % 4
         out = []
@@ -270,10 +277,13 @@ class GreedyPerfPartitioner:
                     dev.perf = dev.perf + shard.perf
                     if dev.storage.hbm < 0:
                         raise PlannerError(f"OOM on rank {r} for {opt.name}")
-        # TWRW: place the whole row-shard group on one node (greedy by load)
+        # TWRW/TWCW: place the whole shard group on one node (greedy by load)
         L = topology.local_world_size
         for opt in proposal:
-            if opt.sharding_type != ShardingType.TABLE_ROW_WISE.value:
+            if opt.sharding_type not in (
+                ShardingType.TABLE_ROW_WISE.value,
+                ShardingType.TABLE_COLUMN_WISE.value,
+            ):
                 continue
             n_nodes = max(1, topology.world_size // L)
             best_node, best_load = None, None
