@@ -480,3 +480,34 @@ class TestTBEPrecisionGpu:
         assert gpu.weights.grad is not None
         assert gpu.weights.grad.dtype == torch.bfloat16
         assert float(gpu.weights.grad.abs().sum()) > 0
+
+
+class TestTBESeqDedup:
+    def test_dedup_matches_plain_cpu_flag(self):
+        # CPU path ignores dedup (oracle); flag accepted end-to-end
+        tbe = TableBatchedEmbeddings([("t0", 30, 8)], use_index_dedup=True)
+        out = tbe(torch.tensor([1, 2, 1]), torch.tensor([0, 2, 3]))
+        assert out.shape == (3, 8)
+
+    @pytest.mark.gpu
+    def test_dedup_matches_plain_gpu(self):
+        """Dedup forward must be bit-identical to the plain gather, and the
+        backward (shared) must leave identical weights."""
+        specs = [("t0", 50, 64), ("t1", 200, 64)]
+        torch.manual_seed(0)
+        a = TableBatchedEmbeddings(specs, device=torch.device("cuda"))
+        b = TableBatchedEmbeddings(
+            specs, device=torch.device("cuda"), use_index_dedup=True
+        )
+        b.weights.data.copy_(a.weights.data)
+        indices, offsets = make_inputs(specs, B=16, L=9, seed=11)
+        ic, oc = indices.cuda(), offsets.cuda()
+        out_a = a(ic, oc)
+        out_b = b(ic, oc)
+        torch.cuda.synchronize()
+        assert torch.equal(out_a, out_b)
+        out_a.sum().backward()
+        out_b.sum().backward()
+        torch.cuda.synchronize()
+        assert torch.equal(a.weights, b.weights)
+        assert torch.equal(a.momentum, b.momentum)

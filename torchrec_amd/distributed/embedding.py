@@ -107,11 +107,13 @@ class ShardedEmbeddingCollection(nn.Module):
         env: ShardingEnv,
         fused_params: Optional[Dict[str, Any]] = None,
         device: Optional[torch.device] = None,
+        use_index_dedup: bool = False,
     ) -> None:
         super().__init__()
         self._env = env
         self._device = device or torch.device("cpu")
         self._fused_params = dict(fused_params or {})
+        self._use_index_dedup = use_index_dedup
         self._need_indices = module.need_indices()
         W = env.world_size
         rank = env.rank
@@ -226,6 +228,7 @@ class ShardedEmbeddingCollection(nn.Module):
             weights_precision={"FP32": "fp32", "FP16": "fp16", "BF16": "bf16"}[
                 tables[0].data_type if tables else "FP32"
             ],
+            use_index_dedup=self._use_index_dedup,
         )
 
     # -- forward ------------------------------------------------------------
@@ -353,8 +356,13 @@ class _ECFusedOptimizer(FusedOptimizer):
 
 
 class EmbeddingCollectionSharder(ModuleSharder[EmbeddingCollection]):
-    def __init__(self, fused_params: Optional[Dict[str, Any]] = None) -> None:
+    def __init__(
+        self,
+        fused_params: Optional[Dict[str, Any]] = None,
+        use_index_dedup: bool = False,
+    ) -> None:
         self._fused_params = fused_params or {}
+        self._use_index_dedup = use_index_dedup
 
     def shard(
         self,
@@ -364,7 +372,8 @@ class EmbeddingCollectionSharder(ModuleSharder[EmbeddingCollection]):
         device: Optional[torch.device] = None,
     ) -> ShardedEmbeddingCollection:
         return ShardedEmbeddingCollection(
-            module, params, env, fused_params=self._fused_params, device=device
+            module, params, env, fused_params=self._fused_params, device=device,
+            use_index_dedup=self._use_index_dedup,
         )
 
     @property

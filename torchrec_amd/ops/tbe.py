@@ -286,6 +286,7 @@ class TableBatchedEmbeddingBags(nn.Module):
         )
         reg("_empty_f", torch.empty(0, dtype=torch.float32))
         reg("_empty_i", torch.empty(0, dtype=torch.int32))
+        reg("_table_identity", torch.arange(T, dtype=torch.int32))
         if self._uvm_caching:
             # cache sizing: cache_load_factor of total rows, 32 ways per set
             ways = 32
@@ -795,8 +796,10 @@ class TableBatchedEmbeddings(nn.Module):
         init_min: float = -0.01,
         init_max: float = 0.01,
         weights_precision: str = "fp32",
+        use_index_dedup: bool = False,
     ) -> None:
         super().__init__()
+        self._use_index_dedup = use_index_dedup
         dims = {s[2] for s in embedding_specs}
         assert len(dims) == 1, "sequence TBE requires a uniform embedding dim"
         self._bags = TableBatchedEmbeddingBags(
@@ -845,7 +848,57 @@ class TableBatchedEmbeddings(nn.Module):
         feat_val_offsets = offsets[:: B][: F + 1].contiguous()
         if feat_val_offsets.numel() < F + 1:
             feat_val_offsets = torch.cat([feat_val_offsets, offsets[-1:]])
+        if self._use_index_dedup:
+            return _TBESeqDedupFunction.apply(host._dummy, host, indices, feat_val_offsets)
         return _TBESeqFunction.apply(host._dummy, host, indices, feat_val_offsets)
+
+
+class _TBESeqDedupFunction(torch.autograd.Function):
+    """Sequence forward with index dedup (reference: ShardedEmbeddingCollection
+    use_index_dedup / fbgemm jagged_unique_indices, distributed/embedding.py:1439).
+
+    Ids are linearized into the group's table-row space and deduped with
+    ``torch.unique`` so each hot row is read from HBM once; the output is
+    expanded back with an index_select. The backward is the standard
+    sort+segment fused update over the ORIGINAL ids — deterministic, and
+    already duplicate-efficient, so dedup only changes the forward."""
+
+    @staticmethod
+    def forward(ctx, dummy, host, indices, feat_val_offsets):  # type: ignore[override]
+        N = indices.numel()
+        counts = feat_val_offsets[1:] - feat_val_offsets[:-1]
+        f = torch.repeat_interleave(
+            torch.arange(counts.numel(), device=indices.device, dtype=torch.int64),
+            counts,
+            output_size=N,
+        )
+        linear = indices + host._feat_row_offset[f]
+        uniq, inv = torch.unique(linear, sorted=True, return_inverse=True)
+        row_offs = host._table_row_offsets  # [T+1]
+        t_per = torch.searchsorted(row_offs[1:], uniq, right=True)
+        local = uniq - row_offs[t_per]
+        table_ranges = torch.searchsorted(uniq, row_offs)  # [T+1] in unique space
+        rows_u = torch.ops.trec_amd.tbe_forward_seq(
+            host.weights,
+            host._table_elem_offsets,
+            host._dims_t,
+            host._table_identity,
+            table_ranges,
+            local,
+            host._max_D,
+            host._max_D,
+        )
+        out = rows_u.index_select(0, inv)
+        ctx.host = host
+        ctx.save_for_backward(indices, feat_val_offsets)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):  # type: ignore[override]
+        host = ctx.host
+        indices, feat_val_offsets = ctx.saved_tensors
+        host._backward_seq(grad.contiguous(), indices, feat_val_offsets)
+        return None, None, None, None
 
 
 class _TBESeqCpuFunction(torch.autograd.Function):
