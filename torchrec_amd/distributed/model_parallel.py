@@ -91,6 +91,8 @@ class DistributedModelParallel(nn.Module):
             fqn = f"{path}.{name}" if path else name
             sharder = self._sharder_map.get(type(child))
             mplan = self._plan.get_plan_for_module(fqn) if sharder else None
+            if mplan is None and sharder is not None and getattr(sharder, "plan_optional", False):
+                mplan = {}  # e.g. tower sharders place modules themselves
             if sharder is not None and mplan is not None:
                 sharded = sharder.shard(child, mplan, self._env, self.device)
                 setattr(module, name, sharded)

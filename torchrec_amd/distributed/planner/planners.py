@@ -134,6 +134,8 @@ class EmbeddingEnumerator:
             sharder = sharder_by_type.get(type(child))
             if sharder is None:
                 continue
+            if getattr(sharder, "plan_optional", False):
+                continue  # e.g. tower sharders place modules themselves
             configs = child.embedding_bag_configs() if hasattr(child, "embedding_bag_configs") else child.embedding_configs()
             is_weighted = child.is_weighted() if hasattr(child, "is_weighted") else False
             for cfg in configs:
