@@ -134,7 +134,13 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
         pg = None
 
     model = build_model(scale)
-    fused_params = {"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    fused_params = {
+        "optimizer": "rowwise_adagrad",
+        "learning_rate": LR,
+        # Criteo categoricals are one-hot: enables the single-launch
+        # segmented backward sort
+        "fixed_bag_length": IDS_PER_FEATURE,
+    }
     if qcomm != "none":
         from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
 
@@ -237,7 +243,13 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
     device = torch.device("cuda", 0)
     torch.cuda.set_device(device)
     model = build_model(scale)
-    fused_params = {"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    fused_params = {
+        "optimizer": "rowwise_adagrad",
+        "learning_rate": LR,
+        # Criteo categoricals are one-hot: enables the single-launch
+        # segmented backward sort
+        "fixed_bag_length": IDS_PER_FEATURE,
+    }
     if qcomm != "none":
         from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
 

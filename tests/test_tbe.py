@@ -511,3 +511,60 @@ class TestTBESeqDedup:
         torch.cuda.synchronize()
         assert torch.equal(a.weights, b.weights)
         assert torch.equal(a.momentum, b.momentum)
+
+
+@pytest.mark.gpu
+class TestSegSort:
+    def test_seg_sort_backward_matches_device_sort(self):
+        """fixed_bag_length enables the single-launch segmented sort; the
+        resulting fused update must match the hipCUB path exactly."""
+        specs = [("t0", 300, 64), ("t1", 500, 64), ("t2", 64, 64)]
+        torch.manual_seed(0)
+        a = TableBatchedEmbeddingBags(specs, device=torch.device("cuda"), learning_rate=0.05)
+        b = TableBatchedEmbeddingBags(
+            specs, device=torch.device("cuda"), learning_rate=0.05, fixed_bag_length=2
+        )
+        b.weights.data.copy_(a.weights.data)
+        B = 32
+        g = torch.Generator().manual_seed(4)
+        # exactly 2 ids per bag per feature (one-hot-ish fixed length)
+        lengths = torch.full((3 * B,), 2, dtype=torch.int64)
+        indices = torch.cat(
+            [torch.randint(0, specs[f][1], (2 * B,), generator=g) for f in range(3)]
+        )
+        offsets = torch.zeros(3 * B + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        ic, oc = indices.cuda(), offsets.cuda()
+        assert b._seg_sort_ok
+        out_a = a(ic, oc)
+        out_b = b(ic, oc)
+        torch.cuda.synchronize()
+        assert torch.equal(out_a, out_b)
+        out_a.sum().backward()
+        out_b.sum().backward()
+        torch.cuda.synchronize()
+        assert torch.equal(a.weights, b.weights)
+        assert torch.equal(a.momentum, b.momentum)
+
+    def test_seg_sort_direct(self):
+        from torchrec_amd import ops as O
+        O.hip_ops()
+        B, F = 16, 2
+        g = torch.Generator().manual_seed(0)
+        lengths = torch.randint(0, 3, (F * B,), generator=g)
+        n = int(lengths.sum())
+        offsets = torch.zeros(F * B + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        # per-feature disjoint id ranges (table-ordered linear space)
+        linear = torch.empty(n, dtype=torch.int64)
+        for f in range(F):
+            lo, hi = int(offsets[f * B]), int(offsets[(f + 1) * B])
+            linear[lo:hi] = torch.randint(f * 100, f * 100 + 100, (hi - lo,), generator=g)
+        sorted_l, perm, ovf = torch.ops.trec_amd.seg_sort_pairs(
+            linear.cuda(), offsets.cuda(), B, F, 9, B * 4
+        )
+        torch.cuda.synchronize()
+        assert int(ovf.item()) == 0
+        ref = torch.sort(linear).values
+        assert torch.equal(sorted_l.cpu(), ref)
+        assert torch.equal(linear[perm.cpu().long()], sorted_l.cpu())

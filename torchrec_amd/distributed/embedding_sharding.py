@@ -159,6 +159,7 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                 weights_precision={"FP32": "fp32", "FP16": "fp16", "BF16": "bf16"}[
                     group[0].data_type
                 ],
+                fixed_bag_length=fused_params.get("fixed_bag_length"),
             )
             self._emb_modules.append(tbe)
             nf = sum(len(t.feature_names) for t in group)

@@ -24,6 +24,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hipcub/hipcub.hpp>
+#include <rocprim/block/block_radix_sort.hpp>
 
 #include <type_traits>
 
@@ -390,6 +391,80 @@ std::tuple<at::Tensor, at::Tensor> sort_pairs(const at::Tensor& keys, int64_t en
       tmp.data_ptr(), tmp_bytes, keys.data_ptr<int64_t>(), keys_out.data_ptr<int64_t>(),
       vals_in.data_ptr<int32_t>(), vals_out.data_ptr<int32_t>(), n, 0, (int)end_bit, stream);
   return {keys_out, vals_out};
+}
+
+// ---------------------------------------------------------------------------
+// segmented block radix sort: ONE launch replaces hipCUB's ~20-kernel device
+// sort when every feature segment fits a workgroup (fixed-length bags, e.g.
+// one-hot Criteo). Segment f of the feature-major layout is
+// [offsets[f*B], offsets[(f+1)*B]); strictly-increasing feature_table_map
+// keeps the concatenation globally grouped (caller-checked).
+// ---------------------------------------------------------------------------
+
+template <int ITEMS>
+__global__ void __launch_bounds__(kBlockThreads) seg_block_sort_kernel(
+    const int64_t* __restrict__ linear, const int64_t* __restrict__ offsets, int B,
+    int end_bit, int64_t* __restrict__ sorted_out, int32_t* __restrict__ perm_out,
+    int32_t* __restrict__ overflow) {
+  using sorter = rocprim::block_radix_sort<uint32_t, kBlockThreads, ITEMS, uint32_t>;
+  __shared__ typename sorter::storage_type storage;
+  int f = blockIdx.x;
+  int64_t lo = offsets[static_cast<int64_t>(f) * B];
+  int64_t hi = offsets[static_cast<int64_t>(f + 1) * B];
+  int count = static_cast<int>(hi - lo);
+  if (count > kBlockThreads * ITEMS) {
+    if (threadIdx.x == 0) atomicOr(overflow, 1);
+    count = kBlockThreads * ITEMS;
+  }
+  uint32_t keys[ITEMS];
+  uint32_t vals[ITEMS];
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    int k = static_cast<int>(threadIdx.x) * ITEMS + i;  // blocked arrangement
+    if (k < count) {
+      keys[i] = static_cast<uint32_t>(linear[lo + k]);
+      vals[i] = static_cast<uint32_t>(lo + k);
+    } else {
+      keys[i] = 0xFFFFFFFFu;
+      vals[i] = 0xFFFFFFFFu;
+    }
+  }
+  sorter().sort(keys, vals, storage, 0u, static_cast<unsigned>(end_bit));
+  // blocked arrangement out: thread t holds ranks [t*ITEMS, t*ITEMS+ITEMS)
+#pragma unroll
+  for (int i = 0; i < ITEMS; ++i) {
+    int k = static_cast<int>(threadIdx.x) * ITEMS + i;
+    if (k < count) {
+      sorted_out[lo + k] = static_cast<int64_t>(keys[i]);
+      perm_out[lo + k] = static_cast<int32_t>(vals[i]);
+    }
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs(
+    const at::Tensor& linear, const at::Tensor& offsets, int64_t B, int64_t F,
+    int64_t end_bit, int64_t capacity) {
+  TORCH_CHECK(linear.is_cuda() && linear.scalar_type() == at::kLong);
+  TORCH_CHECK(end_bit <= 32, "segmented sort requires 32-bit ids");
+  auto sorted = at::empty_like(linear);
+  auto perm = at::empty({linear.numel()}, linear.options().dtype(at::kInt));
+  auto overflow = at::zeros({1}, linear.options().dtype(at::kInt));
+  if (linear.numel() == 0 || F == 0) return {sorted, perm, overflow};
+  auto stream = tbe_stream();
+#define SEG_SORT_LAUNCH(ITEMS)                                                         \
+  hipLaunchKernelGGL((seg_block_sort_kernel<ITEMS>), dim3((int)F), dim3(kBlockThreads), \
+                     0, stream, linear.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(), \
+                     (int)B, (int)end_bit, sorted.data_ptr<int64_t>(),                  \
+                     perm.data_ptr<int32_t>(), overflow.data_ptr<int32_t>())
+  if (capacity <= kBlockThreads * 8) SEG_SORT_LAUNCH(8);
+  else if (capacity <= kBlockThreads * 16) SEG_SORT_LAUNCH(16);
+  else if (capacity <= kBlockThreads * 32) SEG_SORT_LAUNCH(32);
+  else {
+    TORCH_CHECK(capacity <= kBlockThreads * 64, "segment too large for block sort");
+    SEG_SORT_LAUNCH(64);
+  }
+#undef SEG_SORT_LAUNCH
+  return {sorted, perm, overflow};
 }
 
 __global__ void mark_runs_kernel(const int64_t* __restrict__ sorted, int64_t n,

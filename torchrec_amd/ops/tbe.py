@@ -190,6 +190,7 @@ class TableBatchedEmbeddingBags(nn.Module):
         location: EmbeddingLocation = EmbeddingLocation.DEVICE,
         cache_load_factor: float = 0.2,
         weights_precision: str = "fp32",
+        fixed_bag_length: Optional[int] = None,
     ) -> None:
         super().__init__()
         device = device or torch.device("cpu")
@@ -287,6 +288,17 @@ class TableBatchedEmbeddingBags(nn.Module):
         reg("_empty_f", torch.empty(0, dtype=torch.float32))
         reg("_empty_i", torch.empty(0, dtype=torch.int32))
         reg("_table_identity", torch.arange(T, dtype=torch.int32))
+        # single-launch segmented sort in the backward: legal when the
+        # feature segments are table-ordered and disjoint (global grouping
+        # preserved) and each segment fits one workgroup's LDS tile
+        self.fixed_bag_length = fixed_bag_length
+        self._seg_sort_ok = (
+            all(
+                feature_table_map[i] < feature_table_map[i + 1]
+                for i in range(len(feature_table_map) - 1)
+            )
+            and self._total_rows < (1 << 31)
+        )
         if self._uvm_caching:
             # cache sizing: cache_load_factor of total rows, 32 ways per set
             ways = 32
@@ -402,9 +414,16 @@ class TableBatchedEmbeddingBags(nn.Module):
             assert psw is None, "mean pooling with per-sample weights unsupported"
         elif psw is not None:
             scale = psw
-        sorted_lin, perm = torch.ops.trec_amd.sort_pairs(
-            linear, _bits_needed(self._total_rows)
-        )
+        cap = (self.fixed_bag_length or 0) * B
+        if self._seg_sort_ok and 0 < cap <= 16384:
+            sorted_lin, perm, _overflow = torch.ops.trec_amd.seg_sort_pairs(
+                linear, offsets, B, self._num_features,
+                _bits_needed(self._total_rows), cap,
+            )
+        else:
+            sorted_lin, perm = torch.ops.trec_amd.sort_pairs(
+                linear, _bits_needed(self._total_rows)
+            )
         seg_offsets, num_runs = torch.ops.trec_amd.tbe_backward_prep(sorted_lin)
         torch.ops.trec_amd.tbe_backward_fused(
             self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
