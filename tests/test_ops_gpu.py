@@ -143,3 +143,49 @@ def test_fused_interaction_bf16_inputs():
     torch.cuda.synchronize()
     assert dense.grad.dtype == torch.bfloat16
     assert sparse.grad.dtype == torch.float32
+
+
+@pytest.mark.gpu
+class TestColSum:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16, torch.float16])
+    def test_matches_torch_sum(self, dtype):
+        from torchrec_amd import ops as O
+        O.hip_ops()
+        torch.manual_seed(0)
+        x = torch.randn(1000, 513, device="cuda").to(dtype)
+        got = torch.ops.trec_amd.col_sum(x)
+        ref = x.float().sum(0)
+        torch.cuda.synchronize()
+        assert got.dtype == dtype
+        tol = 1e-4 if dtype == torch.float32 else 3e-2
+        torch.testing.assert_close(got.float(), ref, atol=tol, rtol=tol)
+
+    def test_deterministic(self):
+        from torchrec_amd import ops as O
+        O.hip_ops()
+        x = torch.randn(8192, 1024, device="cuda", dtype=torch.bfloat16)
+        a = torch.ops.trec_amd.col_sum(x)
+        b = torch.ops.trec_amd.col_sum(x)
+        torch.cuda.synchronize()
+        assert torch.equal(a, b)
+
+    def test_perceptron_grads_match_linear(self):
+        from torchrec_amd.modules.mlp import Perceptron
+
+        torch.manual_seed(0)
+        p = Perceptron(64, 32, device=torch.device("cuda"))
+        ref = torch.nn.Linear(64, 32, device="cuda")
+        with torch.no_grad():
+            ref.weight.copy_(p._linear.weight)
+            ref.bias.copy_(p._linear.bias)
+        x = torch.randn(128, 64, device="cuda", requires_grad=True)
+        x2 = x.detach().clone().requires_grad_(True)
+        out = p(x)
+        out_ref = torch.relu(ref(x2))
+        out.sum().backward()
+        out_ref.sum().backward()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(out, out_ref)
+        torch.testing.assert_close(p._linear.weight.grad, ref.weight.grad, atol=1e-5, rtol=1e-5)
+        torch.testing.assert_close(p._linear.bias.grad, ref.bias.grad, atol=1e-5, rtol=1e-5)
+        torch.testing.assert_close(x.grad, x2.grad, atol=1e-5, rtol=1e-5)

@@ -1,4 +1,9 @@
-"""Dense building blocks: Perceptron / MLP (reference: torchrec/modules/mlp.py)."""
+"""Dense building blocks: Perceptron / MLP (reference: torchrec/modules/mlp.py).
+
+On CUDA the Perceptron's linear runs through a custom autograd Function whose
+backward computes the bias gradient with the deterministic two-phase
+``col_sum`` HIP kernel — torch's bf16 column ``reduce_kernel`` is the slowest
+non-GEMM kernel in the DLRM step (~17 us/layer at [8192, 1024])."""
 
 from __future__ import annotations
 
@@ -6,6 +11,37 @@ from typing import Callable, List, Optional, Union
 
 import torch
 import torch.nn as nn
+
+
+class _LinearColSumBias(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):  # type: ignore[override]
+        ctx.save_for_backward(x, w)
+        return torch.addmm(b, x, w.t())
+
+    @staticmethod
+    def backward(ctx, g):  # type: ignore[override]
+        from torchrec_amd import ops
+
+        x, w = ctx.saved_tensors
+        g = g.contiguous()
+        gx = g @ w
+        gw = g.t() @ x
+        ops.hip_ops()
+        gb = torch.ops.trec_amd.col_sum(g)
+        return gx, gw, gb
+
+
+def _linear_fwd(linear: nn.Linear, input: torch.Tensor) -> torch.Tensor:
+    w, b = linear.weight, linear.bias
+    if input.is_cuda and b is not None and input.dim() == 2 and torch.is_grad_enabled():
+        if torch.is_autocast_enabled("cuda"):
+            dt = torch.get_autocast_dtype("cuda")
+            with torch.autocast("cuda", enabled=False):
+                return _LinearColSumBias.apply(input.to(dt), w.to(dt), b.to(dt))
+        if input.dtype == w.dtype == b.dtype:
+            return _LinearColSumBias.apply(input, w, b)
+    return linear(input)
 
 
 class Perceptron(nn.Module):
@@ -27,7 +63,7 @@ class Perceptron(nn.Module):
         self._activation_fn = activation
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
-        return self._activation_fn(self._linear(input))
+        return self._activation_fn(_linear_fwd(self._linear, input))
 
 
 class MLP(nn.Module):

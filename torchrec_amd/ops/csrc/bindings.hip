@@ -94,6 +94,7 @@ at::Tensor tbe_forward_seq_int8(const at::Tensor& qweights,
                                 int64_t D_out, int64_t max_D);
 
 // interaction.hip
+at::Tensor col_sum(const at::Tensor& input);
 at::Tensor interaction_forward(const at::Tensor& dense, const at::Tensor& sparse,
                                const at::Tensor& pi, const at::Tensor& pj);
 std::tuple<at::Tensor, at::Tensor> interaction_backward(const at::Tensor& grad_out,
@@ -173,6 +174,7 @@ TORCH_LIBRARY(trec_amd, m) {
       "tbe_forward_seq_int8(Tensor qweights, Tensor table_byte_offsets, Tensor dims,"
       " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"
       " -> Tensor");
+  m.def("col_sum(Tensor input) -> Tensor");
   m.def("interaction_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
   m.def(
       "interaction_backward(Tensor grad_out, Tensor dense, Tensor sparse, Tensor pair_col)"
@@ -204,6 +206,7 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("quantize_rowwise_int8", trec_amd::quantize_rowwise_int8);
   m.impl("tbe_forward_pooled_int8", trec_amd::tbe_forward_pooled_int8);
   m.impl("tbe_forward_seq_int8", trec_amd::tbe_forward_seq_int8);
+  m.impl("col_sum", trec_amd::col_sum);
   m.impl("interaction_forward", trec_amd::interaction_forward);
   m.impl("interaction_backward", trec_amd::interaction_backward);
 }
