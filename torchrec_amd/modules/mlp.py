@@ -32,9 +32,20 @@ class _LinearColSumBias(torch.autograd.Function):
         return gx, gw, gb
 
 
+import os
+
+_USE_COLSUM = os.environ.get("TREC_COLSUM_LINEAR", "1") != "0"
+
+
 def _linear_fwd(linear: nn.Linear, input: torch.Tensor) -> torch.Tensor:
     w, b = linear.weight, linear.bias
-    if input.is_cuda and b is not None and input.dim() == 2 and torch.is_grad_enabled():
+    if (
+        _USE_COLSUM
+        and input.is_cuda
+        and b is not None
+        and input.dim() == 2
+        and torch.is_grad_enabled()
+    ):
         if torch.is_autocast_enabled("cuda"):
             dt = torch.get_autocast_dtype("cuda")
             with torch.autocast("cuda", enabled=False):
