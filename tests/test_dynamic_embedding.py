@@ -57,3 +57,67 @@ class TestDataloaderWrap:
             assert int(vals.max()) < 64
         tr = wrapped.transformer
         assert tr.transformer("f0").size() <= 64
+
+
+class TestKeyValueEmbedding:
+    def test_virtual_ids_beyond_capacity(self):
+        from torchrec_amd.ops.kv_embedding import KeyValueEmbeddingBags
+
+        torch.manual_seed(0)
+        kv = KeyValueEmbeddingBags(
+            [("t0", 10**12, 8)], capacity=16, learning_rate=0.1
+        )
+        # raw ids far beyond the physical table
+        indices = torch.tensor([10**11, 5, 10**11, 999_999_999])
+        offsets = torch.tensor([0, 2, 4])
+        out = kv(indices, offsets)
+        assert out.shape == (2, 8)
+        out.sum().backward()  # fused update on translated slots
+        # same raw ids hit the same slots -> embeddings persist across calls
+        out2 = kv(indices, offsets)
+        assert not torch.equal(out, out2)  # weights moved by the update
+        ids_map = kv.save_ids()[0]
+        assert (ids_map >= 0).sum() <= 16
+        assert 5 in ids_map.tolist()
+
+    def test_eviction_reinitializes_slots(self):
+        from torchrec_amd.ops.kv_embedding import KeyValueEmbeddingBags
+
+        torch.manual_seed(1)
+        kv = KeyValueEmbeddingBags([("t0", 10**9, 4)], capacity=4)
+        # fill capacity then overflow: evictions must not leak old rows
+        for start in (0, 4):
+            idx = torch.arange(start, start + 4) * 1000
+            out = kv(idx, torch.tensor([0, 1, 2, 3, 4]))
+            assert out.shape == (4, 4)
+        ids_map = kv.save_ids()[0]
+        assert (ids_map >= 0).sum() <= 4
+
+    def test_sharded_key_value_kernel(self):
+        """KEY_VALUE compute kernel through the sharded lookup layer."""
+        from torchrec_amd.distributed.embedding_sharding import (
+            GroupedPooledEmbeddingsLookup,
+            ShardedTableLocal,
+        )
+        from torchrec_amd.modules.embedding_configs import PoolingType
+
+        tables = [
+            ShardedTableLocal(
+                name="t0", local_rows=10**9, local_dim=8,
+                pooling=PoolingType.SUM, kernel="key_value",
+                feature_names=["f0"], full_dim=8, full_rows=10**9,
+            )
+        ]
+        lookup = GroupedPooledEmbeddingsLookup(
+            [tables], fused_params={"kv_capacity": 32, "optimizer": "rowwise_adagrad"}
+        )
+        from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+        kjt = KeyedJaggedTensor(
+            keys=["f0"],
+            values=torch.tensor([123456789, 42]),
+            lengths=torch.tensor([1, 1]),
+            stride=2,
+        )
+        out = lookup(kjt)
+        assert out.shape == (2, 8)

@@ -139,6 +139,25 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
             )
             specs = [(t.name, t.local_rows, t.local_dim) for t in group]
             feature_table_map = [i for i, t in enumerate(group) for _ in t.feature_names]
+            if group[0].kernel == EmbeddingComputeKernel.KEY_VALUE.value:
+                from torchrec_amd.ops.kv_embedding import KeyValueEmbeddingBags
+
+                kv = KeyValueEmbeddingBags(
+                    specs,
+                    capacity=fused_params.get("kv_capacity", 1 << 20),
+                    feature_table_map=feature_table_map,
+                    pooling_mode=_POOL_TO_MODE[pool],
+                    optimizer=optimizer,
+                    learning_rate=fused_params.get("learning_rate", 0.01),
+                    eps=fused_params.get("eps", 1.0e-8),
+                    device=device,
+                )
+                self._emb_modules.append(kv)
+                self._feature_splits.append(sum(len(t.feature_names) for t in group))
+                self._group_dims.extend(
+                    t.local_dim for t in group for _ in t.feature_names
+                )
+                continue
             from torchrec_amd.ops.tbe import EmbeddingLocation
 
             location = EmbeddingLocation.DEVICE

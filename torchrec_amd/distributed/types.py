@@ -42,6 +42,7 @@ class EmbeddingComputeKernel(Enum):
     FUSED = "fused"
     FUSED_UVM = "fused_uvm"
     FUSED_UVM_CACHING = "fused_uvm_caching"
+    KEY_VALUE = "key_value"  # bounded DRAM virtual table + id translation
     QUANT = "quant"
 
 

@@ -1,0 +1,129 @@
+"""Key-value (virtual-table) embedding: unbounded id spaces over bounded
+DRAM-resident tables.
+
+Reference parity: torchrec/distributed/batched_embedding_kernel.py:3153
+(KeyValueEmbeddingBag -> SSDTableBatchedEmbeddingBags) and
+RFC-0002 (collision-free virtual tables, DRAM_VIRTUAL_TABLE kernel).
+
+MI355X design: instead of an SSD/RocksDB tier, the bounded physical table
+lives in pinned host DRAM (`EmbeddingLocation.MANAGED`) addressed directly by
+the TBE kernels over PCIe; the C++ ``IdTransformer``
+(dynamic_embedding/csrc/id_transformer.cpp) maps raw ids to dense slots with
+mixed LFU/LRU eviction. Evicted slots are re-initialized (weights uniform,
+momentum zero) so reused slots never leak old state.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from torchrec_amd.dynamic_embedding._id_transformer import IdTransformer
+from torchrec_amd.ops.tbe import (
+    EmbeddingLocation,
+    PoolingMode,
+    TableBatchedEmbeddingBags,
+)
+
+
+class KeyValueEmbeddingBags(nn.Module):
+    """Pooled TBE over virtual id spaces: per-table bounded capacity +
+    host-side id->slot translation.
+
+    ``embedding_specs`` entries are (name, virtual_rows, dim); ``capacity``
+    bounds the physical rows actually stored per table.
+    """
+
+    def __init__(
+        self,
+        embedding_specs: List[Tuple[str, int, int]],
+        capacity: int,
+        feature_table_map: Optional[List[int]] = None,
+        pooling_mode: PoolingMode = PoolingMode.SUM,
+        optimizer: str = "rowwise_adagrad",
+        learning_rate: float = 0.01,
+        eps: float = 1.0e-8,
+        device: Optional[torch.device] = None,
+        init_min: float = -0.01,
+        init_max: float = 0.01,
+        location: EmbeddingLocation = EmbeddingLocation.MANAGED,
+    ) -> None:
+        super().__init__()
+        self._virtual_rows = [int(s[1]) for s in embedding_specs]
+        self._capacity = capacity
+        physical = [(s[0], min(int(s[1]), capacity), s[2]) for s in embedding_specs]
+        self._tbe = TableBatchedEmbeddingBags(
+            physical,
+            feature_table_map=feature_table_map,
+            pooling_mode=pooling_mode,
+            optimizer=optimizer,
+            learning_rate=learning_rate,
+            eps=eps,
+            device=device,
+            init_min=init_min,
+            init_max=init_max,
+            location=location if (device or torch.device("cpu")).type == "cuda"
+            else EmbeddingLocation.DEVICE,
+        )
+        self._init_min = init_min
+        self._init_max = init_max
+        # one transformer per physical table (shared across its features)
+        self._transformers = [
+            IdTransformer(min(int(s[1]), capacity)) for s in embedding_specs
+        ]
+        self._ftm = self._tbe._feature_table_map
+
+    @property
+    def embedding_specs(self):
+        return self._tbe.embedding_specs
+
+    def split_embedding_weights(self) -> List[torch.Tensor]:
+        return self._tbe.split_embedding_weights()
+
+    def split_optimizer_states(self):
+        return self._tbe.split_optimizer_states()
+
+    def set_learning_rate(self, lr: float) -> None:
+        self._tbe.set_learning_rate(lr)
+
+    def save_ids(self) -> List[torch.Tensor]:
+        """Per-table (slot -> raw id) map for checkpointing the virtual
+        space (reference: SSD TBE's id snapshot)."""
+        return [torch.tensor(t.save_ids(), dtype=torch.int64) for t in self._transformers]
+
+    def _reinit_slots(self, table: int, slots: torch.Tensor) -> None:
+        if slots.numel() == 0:
+            return
+        with torch.no_grad():
+            w = self._tbe.split_embedding_weights()[table]
+            dev_slots = slots.to(w.device)
+            fresh = torch.empty(
+                (slots.numel(), w.shape[1]), dtype=torch.float32, device=w.device
+            ).uniform_(self._init_min, self._init_max)
+            w[dev_slots] = fresh.to(w.dtype)
+            states = self._tbe.split_optimizer_states()[table]
+            if states:
+                states[0][dev_slots.to(states[0].device)] = 0.0
+
+    def forward(
+        self,
+        indices: torch.Tensor,
+        offsets: torch.Tensor,
+        per_sample_weights: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        F = len(self._ftm)
+        B = (offsets.numel() - 1) // F
+        idx_cpu = indices.cpu()
+        off_cpu = offsets.cpu()
+        out_slots = torch.empty_like(idx_cpu)
+        for f in range(F):
+            t = self._ftm[f]
+            lo, hi = int(off_cpu[f * B]), int(off_cpu[(f + 1) * B])
+            if hi == lo:
+                continue
+            slots, ev_slots, _ev_ids = self._transformers[t].transform(idx_cpu[lo:hi])
+            self._reinit_slots(t, ev_slots)
+            out_slots[lo:hi] = slots
+        return self._tbe(out_slots.to(indices.device), offsets, per_sample_weights)
