@@ -52,8 +52,13 @@ def generate_batch(
     device: Optional[torch.device] = None,
     generator: Optional[torch.Generator] = None,
     pinned: bool = False,
+    learnable_labels: bool = False,
 ) -> Batch:
-    """One synthetic Criteo-shaped batch (random ids / dense / labels)."""
+    """One synthetic Criteo-shaped batch (random ids / dense / labels).
+
+    ``learnable_labels`` makes the label a deterministic function of the ids
+    (parity of the first feature's id) so a training loop has signal to fit —
+    used by the convergence tests."""
     device = device or torch.device("cpu")
     lengths_list = []
     values_list = []
@@ -74,10 +79,17 @@ def generate_batch(
         lengths=torch.cat(lengths_list) if lengths_list else torch.empty(0, dtype=torch.int64),
         stride=batch_size,
     )
+    if learnable_labels and values_list and values_list[0].numel() >= batch_size:
+        # label = parity of the sample's first id of feature 0 (lengths must
+        # be >=1 for f0; callers pass fixed lengths for convergence tests)
+        first = values_list[0][: batch_size]
+        labels = (first % 2).to(torch.int64)
+    else:
+        labels = torch.randint(0, 2, (batch_size,), device=device, generator=generator)
     batch = Batch(
         dense_features=torch.rand(batch_size, num_dense, device=device, generator=generator),
         sparse_features=kjt,
-        labels=torch.randint(0, 2, (batch_size,), device=device, generator=generator),
+        labels=labels,
     )
     if pinned and device.type == "cpu":
         batch = batch.pin_memory()
