@@ -184,19 +184,43 @@ std::tuple<at::Tensor, at::Tensor> interaction_backward(
 // ~17 us per layer at [8192, 1024]; this is a plain HBM-bound sweep.
 // ---------------------------------------------------------------------------
 
-template <typename scalar_t>
+template <typename scalar_t, int VPT>
 __global__ void __launch_bounds__(kBlockThreads) col_sum_partial_kernel(
     const scalar_t* __restrict__ in, int64_t M, int64_t N, int rows_per_group,
     float* __restrict__ partial /* [G, N] */) {
-  int tile = blockIdx.x;   // column tile of kBlockThreads columns
-  int g = blockIdx.y;      // row group
-  int64_t c = static_cast<int64_t>(tile) * kBlockThreads + threadIdx.x;
-  if (c >= N) return;
+  // VPT consecutive columns per thread -> one 16 B load per row for 16-bit
+  // dtypes (wave reads 1 KB contiguous), fp32 accumulate
+  int tile = blockIdx.x;
+  int g = blockIdx.y;
+  int64_t c0 = (static_cast<int64_t>(tile) * kBlockThreads + threadIdx.x) * VPT;
+  if (c0 >= N) return;
   int64_t r0 = static_cast<int64_t>(g) * rows_per_group;
   int64_t r1 = min(M, r0 + rows_per_group);
-  float acc = 0.f;
-  for (int64_t r = r0; r < r1; ++r) acc += emb2float(in[r * N + c]);
-  partial[static_cast<int64_t>(g) * N + c] = acc;
+  float acc[VPT];
+#pragma unroll
+  for (int v = 0; v < VPT; ++v) acc[v] = 0.f;
+  if (c0 + VPT <= N) {
+    for (int64_t r = r0; r < r1; ++r) {
+      const scalar_t* row = in + r * N + c0;
+      if constexpr (VPT == 8 && sizeof(scalar_t) == 2) {
+        uint4 q = *reinterpret_cast<const uint4*>(row);  // 8 halfs, 16 B
+        const scalar_t* h = reinterpret_cast<const scalar_t*>(&q);
+#pragma unroll
+        for (int v = 0; v < 8; ++v) acc[v] += emb2float(h[v]);
+      } else {
+#pragma unroll
+        for (int v = 0; v < VPT; ++v) acc[v] += emb2float(row[v]);
+      }
+    }
+  } else {
+    for (int64_t r = r0; r < r1; ++r)
+      for (int v = 0; v < VPT && c0 + v < N; ++v)
+        acc[v] += emb2float(in[r * N + c0 + v]);
+  }
+  float* prow = partial + static_cast<int64_t>(g) * N + c0;
+#pragma unroll
+  for (int v = 0; v < VPT; ++v)
+    if (c0 + v < N) prow[v] = acc[v];
 }
 
 template <typename scalar_t>
@@ -222,7 +246,8 @@ at::Tensor col_sum(const at::Tensor& input) {
   auto out = at::empty({N}, input.options());
   if (N == 0) return out;
   auto in = input.contiguous();
-  int ntiles = (int)((N + kBlockThreads - 1) / kBlockThreads);
+  constexpr int kVPT = 8;  // 16 B per thread-row for 16-bit dtypes
+  int ntiles = (int)((N + (int64_t)kBlockThreads * kVPT - 1) / (kBlockThreads * kVPT));
   int G = std::max(1, std::min<int>(kMaxBlocks / std::max(ntiles, 1), (int)((M + 31) / 32)));
   int rows_per_group = (int)((M + G - 1) / G);
   auto partial = at::empty({(int64_t)G * N}, input.options().dtype(at::kFloat));
@@ -236,7 +261,7 @@ at::Tensor col_sum(const at::Tensor& input) {
           std::is_same_v<scalar_t, at::Half>, __half,
           std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16,
                              float>>;
-      hipLaunchKernelGGL((col_sum_partial_kernel<dev_t>), dim3(ntiles, G),
+      hipLaunchKernelGGL((col_sum_partial_kernel<dev_t, kVPT>), dim3(ntiles, G),
                          dim3(kBlockThreads), 0, stream,
                          reinterpret_cast<const dev_t*>(in.data_ptr<scalar_t>()), M, N,
                          rows_per_group, partial.data_ptr<float>());

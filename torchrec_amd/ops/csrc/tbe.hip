@@ -401,20 +401,22 @@ std::tuple<at::Tensor, at::Tensor> sort_pairs(const at::Tensor& keys, int64_t en
 // keeps the concatenation globally grouped (caller-checked).
 // ---------------------------------------------------------------------------
 
+constexpr int kSortThreads = 1024;  // wide blocks: few segments must fill CUs
+
 template <int ITEMS>
-__global__ void __launch_bounds__(kBlockThreads) seg_block_sort_kernel(
+__global__ void __launch_bounds__(kSortThreads) seg_block_sort_kernel(
     const int64_t* __restrict__ linear, const int64_t* __restrict__ offsets, int B,
     int end_bit, int64_t* __restrict__ sorted_out, int32_t* __restrict__ perm_out,
     int32_t* __restrict__ overflow) {
-  using sorter = rocprim::block_radix_sort<uint32_t, kBlockThreads, ITEMS, uint32_t>;
+  using sorter = rocprim::block_radix_sort<uint32_t, kSortThreads, ITEMS, uint32_t>;
   __shared__ typename sorter::storage_type storage;
   int f = blockIdx.x;
   int64_t lo = offsets[static_cast<int64_t>(f) * B];
   int64_t hi = offsets[static_cast<int64_t>(f + 1) * B];
   int count = static_cast<int>(hi - lo);
-  if (count > kBlockThreads * ITEMS) {
+  if (count > kSortThreads * ITEMS) {
     if (threadIdx.x == 0) atomicOr(overflow, 1);
-    count = kBlockThreads * ITEMS;
+    count = kSortThreads * ITEMS;
   }
   uint32_t keys[ITEMS];
   uint32_t vals[ITEMS];
@@ -452,16 +454,16 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs(
   if (linear.numel() == 0 || F == 0) return {sorted, perm, overflow};
   auto stream = tbe_stream();
 #define SEG_SORT_LAUNCH(ITEMS)                                                         \
-  hipLaunchKernelGGL((seg_block_sort_kernel<ITEMS>), dim3((int)F), dim3(kBlockThreads), \
+  hipLaunchKernelGGL((seg_block_sort_kernel<ITEMS>), dim3((int)F), dim3(kSortThreads),  \
                      0, stream, linear.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(), \
                      (int)B, (int)end_bit, sorted.data_ptr<int64_t>(),                  \
                      perm.data_ptr<int32_t>(), overflow.data_ptr<int32_t>())
-  if (capacity <= kBlockThreads * 8) SEG_SORT_LAUNCH(8);
-  else if (capacity <= kBlockThreads * 16) SEG_SORT_LAUNCH(16);
-  else if (capacity <= kBlockThreads * 32) SEG_SORT_LAUNCH(32);
+  if (capacity <= kSortThreads * 2) SEG_SORT_LAUNCH(2);
+  else if (capacity <= kSortThreads * 4) SEG_SORT_LAUNCH(4);
+  else if (capacity <= kSortThreads * 8) SEG_SORT_LAUNCH(8);
   else {
-    TORCH_CHECK(capacity <= kBlockThreads * 64, "segment too large for block sort");
-    SEG_SORT_LAUNCH(64);
+    TORCH_CHECK(capacity <= kSortThreads * 16, "segment too large for block sort");
+    SEG_SORT_LAUNCH(16);
   }
 #undef SEG_SORT_LAUNCH
   return {sorted, perm, overflow};
