@@ -202,11 +202,16 @@ __global__ void __launch_bounds__(kBlockThreads) col_sum_partial_kernel(
   if (c0 + VPT <= N) {
     for (int64_t r = r0; r < r1; ++r) {
       const scalar_t* row = in + r * N + c0;
-      if constexpr (VPT == 8 && sizeof(scalar_t) == 2) {
-        uint4 q = *reinterpret_cast<const uint4*>(row);  // 8 halfs, 16 B
+      if constexpr (VPT == 4 && sizeof(scalar_t) == 2) {
+        uint2 q = *reinterpret_cast<const uint2*>(row);  // 4 halfs, 8 B
         const scalar_t* h = reinterpret_cast<const scalar_t*>(&q);
 #pragma unroll
-        for (int v = 0; v < 8; ++v) acc[v] += emb2float(h[v]);
+        for (int v = 0; v < 4; ++v) acc[v] += emb2float(h[v]);
+      } else if constexpr (VPT == 4 && sizeof(scalar_t) == 4) {
+        uint4 q = *reinterpret_cast<const uint4*>(row);  // 4 floats, 16 B
+        const scalar_t* h = reinterpret_cast<const scalar_t*>(&q);
+#pragma unroll
+        for (int v = 0; v < 4; ++v) acc[v] += emb2float(h[v]);
       } else {
 #pragma unroll
         for (int v = 0; v < VPT; ++v) acc[v] += emb2float(row[v]);
@@ -246,10 +251,11 @@ at::Tensor col_sum(const at::Tensor& input) {
   auto out = at::empty({N}, input.options());
   if (N == 0) return out;
   auto in = input.contiguous();
-  constexpr int kVPT = 8;  // 16 B per thread-row for 16-bit dtypes
+  constexpr int kVPT = 4;  // 8 B per thread-row (full 512 B line per wave)
   int ntiles = (int)((N + (int64_t)kBlockThreads * kVPT - 1) / (kBlockThreads * kVPT));
-  int G = std::max(1, std::min<int>(kMaxBlocks / std::max(ntiles, 1), (int)((M + 31) / 32)));
+  int G = std::max(1, std::min<int>(kMaxBlocks / std::max(ntiles, 1), (int)((M + 7) / 8)));
   int rows_per_group = (int)((M + G - 1) / G);
+  int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);  // final pass
   auto partial = at::empty({(int64_t)G * N}, input.options().dtype(at::kFloat));
   auto stream = ia_stream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, input.scalar_type(),
@@ -265,7 +271,7 @@ at::Tensor col_sum(const at::Tensor& input) {
                          dim3(kBlockThreads), 0, stream,
                          reinterpret_cast<const dev_t*>(in.data_ptr<scalar_t>()), M, N,
                          rows_per_group, partial.data_ptr<float>());
-      hipLaunchKernelGGL((col_sum_final_kernel<dev_t>), dim3(ntiles),
+      hipLaunchKernelGGL((col_sum_final_kernel<dev_t>), dim3(ftiles),
                          dim3(kBlockThreads), 0, stream, partial.data_ptr<float>(), G, N,
                          reinterpret_cast<dev_t*>(out.data_ptr<scalar_t>()));
     }
