@@ -14,6 +14,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
+#include <algorithm>
 #include <type_traits>
 
 #include "common.h"
@@ -253,7 +254,10 @@ at::Tensor col_sum(const at::Tensor& input) {
   auto in = input.contiguous();
   constexpr int kVPT = 4;  // 8 B per thread-row (full 512 B line per wave)
   int ntiles = (int)((N + (int64_t)kBlockThreads * kVPT - 1) / (kBlockThreads * kVPT));
-  int G = std::max(1, std::min<int>(kMaxBlocks / std::max(ntiles, 1), (int)((M + 7) / 8)));
+  // small G keeps the strided final pass negligible; 128 blocks of >=64 rows
+  // saturate HBM for the bandwidth-bound partial pass
+  int G = std::max(1, std::min({128, kMaxBlocks / std::max(ntiles, 1),
+                                (int)((M + 63) / 64)}));
   int rows_per_group = (int)((M + G - 1) / G);
   int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);  // final pass
   auto partial = at::empty({(int64_t)G * N}, input.options().dtype(at::kFloat));
