@@ -69,6 +69,8 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
         W = self._W
         if F == 0:
             return KeyedJaggedTensor.empty(device=self._out_values.device)
+        if getattr(self, "_vbe_strides_rf", None) is not None:
+            return self._wait_impl_vbe()
         # recat (r, f) -> (f, r') with r' in the (possibly staggered) order
         perm = torch.tensor(
             [r * F + f for f in range(F) for r in self._rank_order],
@@ -89,6 +91,50 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
             weights=pw,
             lengths=pl.reshape(-1),
             stride=W * self._B,
+        )
+        kjt._dist_value_splits = (self.in_value_splits, self.out_value_splits)
+        return kjt
+
+    def _wait_impl_vbe(self) -> KeyedJaggedTensor:
+        """VBE recat: variable bag counts per (rank, feature). Segment moves
+        are host-sized split+cat for lengths, device jagged permute for
+        values (no host sync on value counts)."""
+        F = len(self._keys)
+        W = self._W
+        strides_rf = self._vbe_strides_rf  # [W][F]
+        device = self._out_values.device
+        # source-major bag segment sizes
+        seg_bags = [strides_rf[r][f] for r in range(W) for f in range(F)]
+        # lengths: reorder (r, f) -> (f, r') host-side sizes
+        pieces = list(self._out_lengths.split(seg_bags))
+        order = [r * F + f for f in range(F) for r in self._rank_order]
+        lengths = torch.cat([pieces[i] for i in order]) if pieces else self._out_lengths
+        # values: jagged permute over (r, f) segments; per-segment value
+        # counts stay on device
+        loffs = torch.zeros(
+            self._out_lengths.numel() + 1, dtype=torch.int64, device=device
+        )
+        torch.cumsum(self._out_lengths, 0, out=loffs[1:])
+        bounds = torch.tensor(
+            [0] + list(torch.tensor(seg_bags).cumsum(0)), device=device
+        )
+        seg_vals = loffs[bounds[1:]] - loffs[bounds[:-1]]
+        perm = torch.tensor(order, dtype=torch.int64, device=device)
+        pv_lengths, pv, pw = ops.permute_2d_sparse_data(
+            perm,
+            seg_vals.view(-1, 1),
+            self._out_values,
+            self._out_weights,
+            permuted_lengths_sum=int(self._out_values.numel()),
+        )
+        kjt = KeyedJaggedTensor(
+            keys=self._keys,
+            values=pv,
+            weights=pw,
+            lengths=lengths,
+            stride_per_key_per_rank=[
+                [strides_rf[r][f] for r in self._rank_order] for f in range(F)
+            ],
         )
         kjt._dist_value_splits = (self.in_value_splits, self.out_value_splits)
         return kjt
@@ -115,6 +161,7 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
         self._input = input
         self._splits = splits
         self._keys = keys
+        keys = keys  # alias used by the VBE payload branch below
         self._rank_order = rank_order
         self._W = dist.get_world_size(pg)
         B = input.stride()
@@ -122,13 +169,57 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
         device = input.device()
         # per-dest-rank value counts (device, async)
         lengths = input.lengths()
-        feat_sums = lengths.view(len(input.keys()), B).sum(dim=1)
+        self._vbe = input.variable_stride_per_key()
+        if self._vbe:
+            # VBE: per-key strides differ; key k occupies stride_per_key[k]
+            # lengths entries (reference: variable-batch KJT a2a)
+            spk = [sum(sp) for sp in input.stride_per_key_per_rank()]
+            self._stride_per_key = spk
+            len_bounds = [0]
+            for ks in spk:
+                len_bounds.append(len_bounds[-1] + ks)
+            loffs = torch.zeros(lengths.numel() + 1, dtype=torch.int64, device=device)
+            torch.cumsum(lengths, 0, out=loffs[1:])
+            kb = torch.tensor(len_bounds, device=device)
+            feat_sums = loffs[kb[1:]] - loffs[kb[:-1]]
+        else:
+            feat_sums = lengths.view(len(input.keys()), B).sum(dim=1)
         boundaries = torch.tensor(
             [sum(splits[:i]) for i in range(len(splits) + 1)], device=device
         )
         send_counts = torch.stack(
             [feat_sums[boundaries[i] : boundaries[i + 1]].sum() for i in range(self._W)]
         )
+        if self._vbe:
+            # append each destination's feature strides to the splits payload
+            # so the receiver learns B_f per (source rank, feature)
+            b_host = [sum(splits[:i]) for i in range(len(splits) + 1)]
+            pieces = []
+            for r in range(self._W):
+                pieces.append(send_counts[r : r + 1])
+                pieces.append(
+                    torch.tensor(
+                        spk[b_host[r] : b_host[r + 1]],
+                        dtype=send_counts.dtype,
+                        device=device,
+                    )
+                )
+            payload = torch.cat(pieces)
+            self._vbe_in_payload_splits = [1 + splits[r] for r in range(self._W)]
+            F_mine = len(keys)
+            self._vbe_out_payload_splits = [1 + F_mine] * self._W
+            self._in_splits_t = payload
+            self._out_splits_t = payload.new_empty(sum(self._vbe_out_payload_splits))
+            self._splits_work = dist.all_to_all_single(
+                self._out_splits_t,
+                payload,
+                output_split_sizes=self._vbe_out_payload_splits,
+                input_split_sizes=self._vbe_in_payload_splits,
+                group=pg,
+                async_op=True,
+            )
+            self._validate = False
+            return
         from torchrec_amd.distributed.collective_utils import (
             collective_tag,
             collective_validation_enabled,
@@ -149,6 +240,8 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
 
     def _wait_impl(self) -> KJTAllToAllTensorsAwaitable:
         self._splits_work.wait()
+        if getattr(self, "_vbe", False):
+            return self._wait_impl_vbe()
         if self._validate:
             from torchrec_amd.distributed.collective_utils import verify_tags
 
@@ -215,6 +308,66 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
             out_value_splits=out_value_splits,
             rank_order=self._rank_order,
         )
+
+
+    def _wait_impl_vbe(self) -> "KJTAllToAllTensorsAwaitable":
+        W = self._W
+        F_mine = len(self._keys)
+        kjt = self._input
+        recv = self._out_splits_t.cpu()
+        sent = self._in_splits_t.cpu()
+        # parse [count, strides...] per rank
+        out_value_splits: List[int] = []
+        strides_rf: List[List[int]] = []
+        pos = 0
+        for r in range(W):
+            out_value_splits.append(int(recv[pos]))
+            strides_rf.append([int(x) for x in recv[pos + 1 : pos + 1 + F_mine]])
+            pos += 1 + F_mine
+        in_value_splits: List[int] = []
+        pos = 0
+        for r in range(W):
+            in_value_splits.append(int(sent[pos]))
+            pos += self._vbe_in_payload_splits[r]
+        device = kjt.device()
+        works = []
+        b_host = [sum(self._splits[:i]) for i in range(len(self._splits) + 1)]
+        len_in_splits = [
+            sum(self._stride_per_key[b_host[r] : b_host[r + 1]]) for r in range(W)
+        ]
+        len_out_splits = [sum(strides_rf[r]) for r in range(W)]
+        out_lengths = kjt.lengths().new_empty(sum(len_out_splits))
+        works.append(
+            dist.all_to_all_single(
+                out_lengths, kjt.lengths().contiguous(),
+                len_out_splits, len_in_splits, group=self._pg, async_op=True,
+            )
+        )
+        out_values = kjt.values().new_empty(sum(out_value_splits))
+        works.append(
+            dist.all_to_all_single(
+                out_values, kjt.values().contiguous(),
+                out_value_splits, in_value_splits, group=self._pg, async_op=True,
+            )
+        )
+        out_weights = None
+        if kjt.weights_or_none() is not None:
+            out_weights = kjt.weights().new_empty(sum(out_value_splits))
+            works.append(
+                dist.all_to_all_single(
+                    out_weights, kjt.weights().contiguous(),
+                    out_value_splits, in_value_splits, group=self._pg, async_op=True,
+                )
+            )
+        aw = KJTAllToAllTensorsAwaitable(
+            self._pg, self._keys, works, out_lengths, out_values, out_weights,
+            self._B,
+            in_value_splits=in_value_splits,
+            out_value_splits=out_value_splits,
+            rank_order=self._rank_order,
+        )
+        aw._vbe_strides_rf = strides_rf
+        return aw
 
 
 class KJTAllToAll(nn.Module):

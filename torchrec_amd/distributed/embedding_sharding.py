@@ -192,6 +192,22 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
             return torch.zeros(
                 B, 0, dtype=torch.float32, device=sparse_features.device()
             )
+        if sparse_features.variable_stride_per_key():
+            # VBE: 1-D packed [sum_f B_f * D_f] per group, feature-major
+            kjts = (
+                [sparse_features]
+                if len(self._emb_modules) == 1
+                else sparse_features.split(self._feature_splits)
+            )
+            outs_1d: List[torch.Tensor] = []
+            for kjt, tbe in zip(kjts, self._emb_modules):
+                bpf = [sum(sp) for sp in kjt.stride_per_key_per_rank()]
+                outs_1d.append(
+                    tbe.forward_vbe(
+                        kjt.values(), kjt.offsets(), bpf, kjt.weights_or_none()
+                    )
+                )
+            return torch.cat(outs_1d) if len(outs_1d) > 1 else outs_1d[0]
         if len(self._emb_modules) == 1:
             kjt = sparse_features
             tbe = self._emb_modules[0]

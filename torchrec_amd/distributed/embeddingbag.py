@@ -80,6 +80,8 @@ class EmbeddingBagCollectionContext:
     """Per-forward state (reference embeddingbag.py EmbeddingBagCollectionContext)."""
 
     mean_divisors: List[Optional[torch.Tensor]] = field(default_factory=list)
+    variable_batch_per_feature: bool = False
+    vbe_local_strides: Optional[Dict[str, int]] = None
 
 
 class KJTListSplitsAwaitable(Awaitable[Awaitable[List[KeyedJaggedTensor]]]):
@@ -147,6 +149,49 @@ class EmbeddingBagCollectionAwaitable(LazyAwaitable[KeyedTensor]):
             keys=self._embedding_names,
             length_per_key=self._embedding_dims,
             values=values,
+        )
+
+
+class _VbeEmbeddingBagCollectionAwaitable(LazyAwaitable[KeyedTensor]):
+    """Assembles the canonical VBE KeyedTensor (key_dim=0, 1-D packed values
+    [sum_f B_f * D_f] in EBC feature order) from per-sharding a2a results."""
+
+    def __init__(
+        self,
+        awaitables,
+        local_parts,
+        local_names,
+        owner_major_names,
+        embedding_names,
+        dims_by_name,
+        local_strides,
+    ) -> None:
+        super().__init__()
+        self._awaitables = awaitables
+        self._local_parts = local_parts
+        self._local_names = local_names
+        self._owner_major_names = owner_major_names
+        self._embedding_names = embedding_names
+        self._dims = dims_by_name
+        self._strides = local_strides
+
+    def _wait_impl(self) -> KeyedTensor:
+        by_name: Dict[str, torch.Tensor] = dict(zip(self._local_names, self._local_parts))
+        for aw in self._awaitables:
+            flat = aw.wait().view(-1)
+            sizes = [
+                self._strides[f] * self._dims[f] for f in self._owner_major_names
+            ]
+            for f, part in zip(self._owner_major_names, flat.split(sizes)):
+                by_name[f] = part
+        values = torch.cat([by_name[f] for f in self._embedding_names])
+        return KeyedTensor(
+            keys=self._embedding_names,
+            values=values,
+            length_per_key=[
+                self._strides[f] * self._dims[f] for f in self._embedding_names
+            ],
+            key_dim=0,
         )
 
 
@@ -305,11 +350,92 @@ class ShardedEmbeddingBagCollection(nn.Module):
 
     def compute_and_output_dist(
         self, ctx: EmbeddingBagCollectionContext, dist_input: List[KeyedJaggedTensor]
-    ) -> EmbeddingBagCollectionAwaitable:
+    ):
+        if ctx.variable_batch_per_feature:
+            return self._compute_and_output_dist_vbe(ctx, dist_input)
         awaitables = []
         for lookup, dist_mod, kjt in zip(self._lookups, self._output_dists, dist_input):
             awaitables.append(dist_mod(lookup(kjt)))
         return self._make_output_awaitable(ctx, awaitables)
+
+    def _compute_and_output_dist_vbe(self, ctx, dist_input):
+        """VBE output path: the lookup emits a 1-D packed [sum_f B_f_total *
+        D_f] feature-major vector; each (feature, source-rank) block is routed
+        back to its source with one variable-split a2a (split+cat keeps the
+        reorders autograd-transparent; element splits ride the host-side
+        stride table exchanged in the splits phase)."""
+        from torchrec_amd.distributed.dist_data import SequenceEmbeddingsAllToAll
+
+        W = self._env.world_size
+        dims_by_name = dict(zip(self._embedding_names, self._embedding_dims))
+        pieces_per_rank: List[List[torch.Tensor]] = [[] for _ in range(max(W, 1))]
+        name_per_rank: List[List[str]] = [[] for _ in range(max(W, 1))]
+        local_parts: List[torch.Tensor] = []
+        local_names: List[str] = []
+        for si, (st, lookup, kjt) in enumerate(
+            zip(self._sharding_types, self._lookups, dist_input)
+        ):
+            packed = lookup(kjt)
+            sharding = self._shardings[si]
+            if (
+                st == ShardingType.DATA_PARALLEL.value
+                or self._env.process_group is None
+                or W == 1
+            ):
+                # local only: my own bags, feature-major already
+                names = kjt.keys()
+                sizes = [
+                    sum(sp) * dims_by_name[f]
+                    for f, sp in zip(names, kjt.stride_per_key_per_rank())
+                ]
+                local_parts.extend(packed.split(sizes))
+                local_names.extend(names)
+                continue
+            # TW: reorder (f, r) element blocks to rank-major and a2a back
+            spr = kjt.stride_per_key_per_rank()  # [F_mine][W]
+            names_mine = kjt.keys()
+            F_mine = len(names_mine)
+            sizes_fmaj = [
+                spr[f][r] * dims_by_name[names_mine[f]]
+                for f in range(F_mine)
+                for r in range(W)
+            ]
+            blocks = list(packed.split(sizes_fmaj)) if F_mine else []
+            send = (
+                torch.cat(
+                    [blocks[f * W + r] for r in range(W) for f in range(F_mine)]
+                )
+                if blocks
+                else packed
+            )
+            in_splits = [
+                sum(spr[f][r] * dims_by_name[names_mine[f]] for f in range(F_mine))
+                for r in range(W)
+            ]
+            # recv: my local bags for each owner's features
+            fpr = sharding._features_per_rank
+            out_splits = [
+                sum(
+                    ctx.vbe_local_strides[f] * dims_by_name[f] for f in fpr[r]
+                )
+                for r in range(W)
+            ]
+            aw = SequenceEmbeddingsAllToAll(self._env.process_group)(
+                send.view(-1, 1), in_splits, out_splits
+            )
+            # remember the owner-major feature layout for reassembly
+            for r in range(W):
+                name_per_rank[r].extend(fpr[r])
+            pieces_per_rank[0].append(aw)  # store awaitable (one per sharding)
+        return _VbeEmbeddingBagCollectionAwaitable(
+            awaitables=[a for a in pieces_per_rank[0]],
+            local_parts=local_parts,
+            local_names=local_names,
+            owner_major_names=[n for r in range(max(W, 1)) for n in name_per_rank[r]],
+            embedding_names=self._embedding_names,
+            dims_by_name=dims_by_name,
+            local_strides=ctx.vbe_local_strides or {},
+        )
 
     def _make_output_awaitable(self, ctx, awaitables) -> EmbeddingBagCollectionAwaitable:
         # resolve per-feature mean divisors into column ranges lazily
@@ -334,6 +460,19 @@ class ShardedEmbeddingBagCollection(nn.Module):
 
     def forward(self, features: KeyedJaggedTensor) -> LazyAwaitable[KeyedTensor]:
         ctx = self.create_context()
+        if features.variable_stride_per_key():
+            ctx.variable_batch_per_feature = True
+            ctx.vbe_local_strides = {
+                k: sum(sp)
+                for k, sp in zip(features.keys(), features.stride_per_key_per_rank())
+            }
+            assert not self._is_weighted, "VBE v1: unweighted only"
+            assert all(
+                t in (ShardingType.TABLE_WISE.value, ShardingType.DATA_PARALLEL.value)
+                for t in self._sharding_types
+            ) or self._env.world_size == 1, (
+                "VBE through the sharded path supports TW/DP shardings"
+            )
         dist_input = self.input_dist(ctx, features).wait().wait()
         return self.compute_and_output_dist(ctx, dist_input)
 

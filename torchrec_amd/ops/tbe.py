@@ -553,21 +553,9 @@ class TableBatchedEmbeddingBags(nn.Module):
             bag_off.append(bag_off[-1] + bf)
             out_off.append(out_off[-1] + bf * d)
         if not indices.is_cuda:
-            outs = []
-            for f in range(F):
-                t = self._feature_table_map[f]
-                sspec = self._specs[t]
-                e0 = int(self._table_elem_offsets[t])
-                w = self.weights[e0 : e0 + sspec.rows * sspec.dim].view(sspec.rows, sspec.dim)
-                off = offsets[bag_off[f] : bag_off[f + 1] + 1] - offsets[bag_off[f]]
-                idx = indices[int(offsets[bag_off[f]]) : int(offsets[bag_off[f + 1]])]
-                mode = {PoolingMode.SUM: "sum", PoolingMode.MEAN: "mean"}[self.pooling_mode]
-                outs.append(
-                    torch.nn.functional.embedding_bag(
-                        idx, w, off, mode=mode, include_last_offset=True
-                    ).reshape(-1)
-                )
-            return torch.cat(outs) if outs else self.weights.new_empty(0)
+            return _TBEVbeCpuFunction.apply(
+                self._dummy, self, indices, offsets, bag_off
+            )
         ops.hip_ops()
         device = indices.device
         bag_offsets = torch.tensor(bag_off, dtype=torch.int64, device=device)
@@ -918,6 +906,49 @@ class _TBESeqDedupFunction(torch.autograd.Function):
         indices, feat_val_offsets = ctx.saved_tensors
         host._backward_seq(grad.contiguous(), indices, feat_val_offsets)
         return None, None, None, None
+
+
+def _tbe_cpu_vbe_forward(weights, host, indices, offsets, bag_off):
+    outs = []
+    for f in range(host._num_features):
+        t = host._feature_table_map[f]
+        sspec = host._specs[t]
+        e0 = int(host._table_elem_offsets[t])
+        w = weights[e0 : e0 + sspec.rows * sspec.dim].view(sspec.rows, sspec.dim)
+        if w.dtype != torch.float32:
+            w = w.float()
+        off = offsets[bag_off[f] : bag_off[f + 1] + 1] - offsets[bag_off[f]]
+        idx = indices[int(offsets[bag_off[f]]) : int(offsets[bag_off[f + 1]])]
+        mode = {PoolingMode.SUM: "sum", PoolingMode.MEAN: "mean"}[host.pooling_mode]
+        outs.append(
+            torch.nn.functional.embedding_bag(
+                idx, w, off, mode=mode, include_last_offset=True
+            ).reshape(-1)
+        )
+    return torch.cat(outs) if outs else weights.new_empty(0)
+
+
+class _TBEVbeCpuFunction(torch.autograd.Function):
+    """CPU oracle of the VBE fused path (mirrors _TBECpuFusedFunction)."""
+
+    @staticmethod
+    def forward(ctx, dummy, host, indices, offsets, bag_off):  # type: ignore[override]
+        ctx.host = host
+        ctx.bag_off = bag_off
+        ctx.save_for_backward(indices, offsets)
+        with torch.no_grad():
+            return _tbe_cpu_vbe_forward(host.weights, host, indices, offsets, bag_off)
+
+    @staticmethod
+    def backward(ctx, grad):  # type: ignore[override]
+        host = ctx.host
+        indices, offsets = ctx.saved_tensors
+        w = host.weights.detach().float().requires_grad_(True)
+        with torch.enable_grad():
+            out = _tbe_cpu_vbe_forward(w, host, indices, offsets, ctx.bag_off)
+            out.backward(grad)
+        host._cpu_apply_update(w.grad)
+        return None, None, None, None, None
 
 
 class _TBESeqCpuFunction(torch.autograd.Function):

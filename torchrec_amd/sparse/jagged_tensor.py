@@ -399,6 +399,33 @@ class KeyedJaggedTensor(Pipelineable):
         out: List[KeyedJaggedTensor] = []
         start = 0
         opk = self.offset_per_key()
+        if self._variable_stride_per_key:
+            # per-key lengths segments have per-key sizes (VBE)
+            spk = self.stride_per_key_per_rank()
+            key_strides = [sum(s) for s in spk]
+            len_bounds = [0]
+            for ks in key_strides:
+                len_bounds.append(len_bounds[-1] + ks)
+            lengths = self.lengths()
+            for seg in segments:
+                keys = self._keys[start : start + seg]
+                vstart, vend = opk[start], opk[start + seg]
+                out.append(
+                    KeyedJaggedTensor(
+                        keys=keys,
+                        values=self._values[vstart:vend],
+                        weights=self._weights[vstart:vend]
+                        if self._weights is not None
+                        else None,
+                        lengths=lengths[len_bounds[start] : len_bounds[start + seg]],
+                        stride_per_key_per_rank=spk[start : start + seg],
+                        length_per_key=self._length_per_key[start : start + seg]
+                        if self._length_per_key is not None
+                        else None,
+                    )
+                )
+                start += seg
+            return out
         B = self._stride
         lengths = self.lengths()
         for seg in segments:
@@ -429,6 +456,36 @@ class KeyedJaggedTensor(Pipelineable):
         """
         if indices_tensor is None:
             indices_tensor = torch.tensor(indices, dtype=torch.int64, device=self.device())
+        if self._variable_stride_per_key:
+            # VBE: per-key jagged segments; value counts stay on device
+            spk = self.stride_per_key_per_rank()
+            key_strides = [sum(sp) for sp in spk]
+            lengths = self.lengths()
+            loffs = torch.zeros(
+                lengths.numel() + 1, dtype=torch.int64, device=lengths.device
+            )
+            torch.cumsum(lengths, 0, out=loffs[1:])
+            bounds = [0]
+            for ks in key_strides:
+                bounds.append(bounds[-1] + ks)
+            bt = torch.tensor(bounds, device=lengths.device)
+            seg_vals = loffs[bt[1:]] - loffs[bt[:-1]]
+            _pl, pv, pw = ops.permute_2d_sparse_data(
+                indices_tensor,
+                seg_vals.view(-1, 1),
+                self._values,
+                self._weights,
+                permuted_lengths_sum=int(self._values.numel()),
+            )
+            len_pieces = list(lengths.split(key_strides))
+            new_lengths = torch.cat([len_pieces[i] for i in indices])
+            return KeyedJaggedTensor(
+                keys=[self._keys[i] for i in indices],
+                values=pv,
+                weights=pw,
+                lengths=new_lengths,
+                stride_per_key_per_rank=[spk[i] for i in indices],
+            )
         B = self._stride
         lengths2d = self.lengths().view(len(self._keys), B)
         pl, pv, pw = ops.permute_2d_sparse_data(
