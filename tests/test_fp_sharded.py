@@ -1,0 +1,98 @@
+"""Sharded feature-processed EBC (reference: distributed/fp_embeddingbag.py)."""
+
+import torch
+import torch.distributed as dist
+
+from tests.dist_utils import run_multi_process
+from tests.test_model_parallel import kjt_local_slice, make_global_kjt
+from torchrec_amd.distributed.fp_embeddingbag import (
+    FeatureProcessedEmbeddingBagCollectionSharder,
+    ShardedFeatureProcessedEmbeddingBagCollection,
+)
+from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+from torchrec_amd.distributed.planner.types import ParameterConstraints, Topology
+from torchrec_amd.distributed.types import ShardingType
+from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+from torchrec_amd.modules.feature_processor import (
+    FeatureProcessedEmbeddingBagCollection,
+    PositionWeightedModuleCollection,
+)
+
+LR = 0.05
+
+
+def _make_fp_ebc(seed=42):
+    torch.manual_seed(seed)
+    tables = [
+        EmbeddingBagConfig(num_embeddings=20, embedding_dim=8, name="t0", feature_names=["f0"]),
+        EmbeddingBagConfig(num_embeddings=30, embedding_dim=16, name="t1", feature_names=["f1"]),
+    ]
+    ebc = EmbeddingBagCollection(tables=tables, is_weighted=True)
+    fps = PositionWeightedModuleCollection({"f0": 8, "f1": 8})
+    with torch.no_grad():
+        fps.position_weights["f0"].copy_(torch.linspace(0.5, 2.0, 8))
+        fps.position_weights["f1"].copy_(torch.linspace(2.0, 0.5, 8))
+    return FeatureProcessedEmbeddingBagCollection(ebc, fps), tables
+
+
+class _Model(torch.nn.Module):
+    def __init__(self, fp_ebc):
+        super().__init__()
+        self.sparse = fp_ebc
+
+    def forward(self, kjt):
+        return self.sparse(kjt)
+
+
+def _run_fp_sharded(rank, world_size):
+    B = 4
+    fp_ebc, tables = _make_fp_ebc()
+    model = _Model(fp_ebc)
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(sharding_types=[ShardingType.TABLE_WISE.value])
+            for cfg in tables
+        },
+    )
+    sharder = FeatureProcessedEmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    assert isinstance(dmp.module.sparse, ShardedFeatureProcessedEmbeddingBagCollection)
+
+    golden_fp, _ = _make_fp_ebc()
+    # sync table weights golden -> sharded
+    gsd = {
+        f"sparse.embedding_bags.{cfg.name}.weight": w.detach()
+        for cfg, w in zip(
+            tables,
+            (
+                golden_fp._embedding_bag_collection.embedding_bags[cfg.name].weight
+                for cfg in tables
+            ),
+        )
+    }
+    dmp.load_state_dict(gsd, strict=False)
+
+    kjt_global = make_global_kjt(tables, B * world_size)
+    kjt_local = kjt_local_slice(kjt_global, rank * B, (rank + 1) * B)
+    kt = dmp(kjt_local)
+    vals = kt.values()
+    golden_out = golden_fp(kjt_global).values()
+    torch.testing.assert_close(
+        vals, golden_out[rank * B : (rank + 1) * B], atol=1e-5, rtol=1e-5
+    )
+    # position-weight grads flow on the owning rank
+    vals.sum().backward()
+    for k, p in dmp.module.sparse.feature_processors.position_weights.items():
+        assert p.grad is not None and float(p.grad.abs().sum()) > 0
+
+
+def test_fp_ebc_sharded():
+    run_multi_process(_run_fp_sharded, 2, "gloo")

@@ -98,6 +98,12 @@ class FeatureProcessedEmbeddingBagCollection(nn.Module):
         self._embedding_bag_collection = embedding_bag_collection
         self._feature_processors = feature_processors
 
+    def embedding_bag_configs(self):
+        return self._embedding_bag_collection.embedding_bag_configs()
+
+    def is_weighted(self) -> bool:
+        return True
+
     def forward(self, features: KeyedJaggedTensor) -> KeyedTensor:
         if isinstance(self._feature_processors, PositionWeightedModuleCollection):
             processed = self._feature_processors(features)

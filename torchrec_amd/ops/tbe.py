@@ -725,7 +725,12 @@ class _TBECpuFusedFunction(torch.autograd.Function):
     def forward(ctx, dummy, host, indices, offsets, psw, B):  # type: ignore[override]
         ctx.host = host
         ctx.B = B
-        ctx.save_for_backward(indices, offsets, psw if psw is not None else torch.empty(0))
+        # psw is saved DETACHED: the backward replays the forward on fresh
+        # leaves and returns grad_psw, so the outer engine traverses the
+        # feature-processor graph exactly once (no shared-buffer free)
+        ctx.save_for_backward(
+            indices, offsets, psw.detach() if psw is not None else torch.empty(0)
+        )
         ctx.has_psw = psw is not None
         with torch.no_grad():
             return _tbe_cpu_forward(host.weights, host, indices, offsets, psw, B)
@@ -735,13 +740,14 @@ class _TBECpuFusedFunction(torch.autograd.Function):
         host = ctx.host
         B = ctx.B
         indices, offsets, psw = ctx.saved_tensors
-        psw_t = psw if ctx.has_psw else None
+        need_psw_grad = ctx.has_psw and ctx.needs_input_grad[4]
+        psw_t = psw.requires_grad_(True) if need_psw_grad else (psw if ctx.has_psw else None)
         w = host.weights.detach().float().requires_grad_(True)
         with torch.enable_grad():
             out = _tbe_cpu_forward(w, host, indices, offsets, psw_t, B)
             out.backward(grad)
         host._cpu_apply_update(w.grad)
-        grad_psw = None
+        grad_psw = psw_t.grad if need_psw_grad else None
         return None, None, None, None, grad_psw, None
 
 
