@@ -175,3 +175,88 @@ class TestSingleHostP2P:
         assert gather([a, b]).shape == (2, 5)
         red = EmbeddingsAllToOneReduce(torch.device("cpu"), 2)
         torch.testing.assert_close(red([a, a]), 2 * a)
+
+
+def test_sharded_itep_ebc():
+    """ITEP sharded wrapper (reference distributed/itep_embeddingbag.py)."""
+    from tests.dist_utils import run_multi_process
+
+    run_multi_process(_run_sharded_itep, 2, "gloo")
+
+
+def _run_sharded_itep(rank, world_size):
+    import torch.distributed as dist
+
+    from torchrec_amd.distributed.itep_embeddingbag import (
+        ITEPEmbeddingBagCollectionSharder,
+        ShardedITEPEmbeddingBagCollection,
+    )
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+    from torchrec_amd.modules.itep_modules import (
+        GenericITEPModule,
+        ITEPEmbeddingBagCollection,
+    )
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    torch.manual_seed(0)
+    # physical table is the PRUNED size; id space is 400 raw ids
+    ebc = EmbeddingBagCollection(
+        tables=[
+            EmbeddingBagConfig(
+                num_embeddings=100, embedding_dim=8, name="t0", feature_names=["f0"]
+            )
+        ]
+    )
+    itep = GenericITEPModule(
+        {"t0": 400}, pruning_interval=4, pruned_hash_sizes={"t0": 100}
+    )
+    model = torch.nn.Module()
+    model.sparse = ITEPEmbeddingBagCollection(ebc, itep)
+    model.forward = lambda kjt: model.sparse(kjt)
+    dmp = DistributedModelParallel(
+        torch.nn.Sequential(),  # placeholder; shard the module directly
+        sharders=[],
+        init_data_parallel=False,
+    )
+    from torchrec_amd.distributed.types import ShardingEnv
+
+    env = ShardingEnv.from_process_group(dist.group.WORLD)
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import ParameterConstraints, Topology
+    from torchrec_amd.distributed.types import ShardingType
+
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={"t0": ParameterConstraints(sharding_types=[ShardingType.ROW_WISE.value])},
+    )
+    sharder = ITEPEmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.05}
+    )
+
+    class M(torch.nn.Module):
+        def __init__(self, s):
+            super().__init__()
+            self.sparse = s
+
+        def forward(self, kjt):
+            return self.sparse(kjt)
+
+    m = M(ITEPEmbeddingBagCollection(ebc, itep))
+    plan = planner.collective_plan(m, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        m, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    assert isinstance(dmp.module.sparse, ShardedITEPEmbeddingBagCollection)
+    g = torch.Generator().manual_seed(3 + rank)
+    for step in range(6):
+        vals = torch.randint(0, 400, (8,), generator=g)  # raw (unpruned) ids
+        kjt = KeyedJaggedTensor(
+            keys=["f0"], values=vals, lengths=torch.full((4,), 2, dtype=torch.int64),
+            stride=4,
+        )
+        out = dmp(kjt)
+        v = out.values()
+        assert v.shape == (4, 8)
+        v.sum().backward()

@@ -118,3 +118,33 @@ class GenericITEPModule(nn.Module):
             util[rows] = top_cnt[promote]
             evicted[name] = rows
         return evicted
+
+
+class ITEPEmbeddingBagCollection(nn.Module):
+    """ITEP-wrapped EBC: raw unpruned-space ids are remapped to the pruned
+    physical tables before lookup (reference
+    torchrec/modules/itep_embedding_modules.py ITEPEmbeddingBagCollection)."""
+
+    def __init__(
+        self,
+        embedding_bag_collection: nn.Module,
+        itep_module: GenericITEPModule,
+    ) -> None:
+        super().__init__()
+        self._embedding_bag_collection = embedding_bag_collection
+        self._itep_module = itep_module
+        self._table_by_feature: Dict[str, str] = {
+            f: cfg.name
+            for cfg in embedding_bag_collection.embedding_bag_configs()
+            for f in cfg.feature_names
+        }
+
+    def embedding_bag_configs(self):
+        return self._embedding_bag_collection.embedding_bag_configs()
+
+    def is_weighted(self) -> bool:
+        return self._embedding_bag_collection.is_weighted()
+
+    def forward(self, features: KeyedJaggedTensor):
+        remapped = self._itep_module.remap(features, self._table_by_feature)
+        return self._embedding_bag_collection(remapped)
