@@ -116,3 +116,38 @@ class TestCriteo:
         assert batch.sparse_features.stride() == 8
         assert len(batch.sparse_features.keys()) == CAT_FEATURE_COUNT
         assert int(batch.sparse_features.values().max()) < 1000
+
+
+def test_accuracy_precision_recall_gauc():
+    import torch
+
+    from torchrec_amd.metrics.rec_metric import (
+        AccuracyMetric,
+        PrecisionMetric,
+        RecallMetric,
+        RecTaskInfo,
+        grouped_auc,
+    )
+
+    preds = torch.tensor([0.9, 0.8, 0.3, 0.2, 0.6, 0.1])
+    labels = torch.tensor([1.0, 1.0, 0.0, 1.0, 0.0, 0.0])
+    tasks = [RecTaskInfo(name="t")]
+    acc = AccuracyMetric(tasks)
+    acc.update(predictions={"t": preds}, labels={"t": labels})
+    a = acc.compute()["accuracy-t|lifetime_accuracy"]
+    # correct: 0.9->1, 0.8->1, 0.3->0, 0.2 vs 1 wrong, 0.6 vs 0 wrong, 0.1->0
+    assert abs(float(a) - 4 / 6) < 1e-6
+    prec = PrecisionMetric(tasks)
+    prec.update(predictions={"t": preds}, labels={"t": labels})
+    p = prec.compute()["precision-t|lifetime_precision"]
+    assert abs(float(p) - 2 / 3) < 1e-6  # predicted pos: 0.9,0.8,0.6 -> 2 tp
+    rec = RecallMetric(tasks)
+    rec.update(predictions={"t": preds}, labels={"t": labels})
+    r = rec.compute()["recall-t|lifetime_recall"]
+    assert abs(float(r) - 2 / 3) < 1e-6  # 3 positives, 2 recovered
+    # GAUC: group 0 perfectly ordered, group 1 inverted
+    g = torch.tensor([0, 0, 0, 1, 1, 1])
+    preds2 = torch.tensor([0.9, 0.8, 0.1, 0.2, 0.9, 0.8])
+    labels2 = torch.tensor([1.0, 1.0, 0.0, 1.0, 0.0, 0.0])
+    gauc = grouped_auc(preds2, labels2, g)
+    assert abs(float(gauc) - 0.5) < 1e-6  # (1.0 + 0.0) / 2

@@ -286,3 +286,102 @@ class CalibrationMetric(RecMetric):
 class MSEMetric(RecMetric):
     COMPUTATION = MSEComputation
     NAME = "mse"
+
+
+class AccuracyComputation(RecMetricComputation):
+    """Thresholded accuracy (reference metrics/accuracy.py)."""
+
+    STATE_NAMES = ["accuracy_sum", "weighted_num_samples"]
+
+    def update(self, predictions, labels, weights=None) -> None:
+        y = labels.double()
+        w = _weights_or_ones(y, weights)
+        hit = ((predictions.double() >= 0.5) == (y >= 0.5)).double()
+        batch = [(hit * w).sum().reshape(1), w.sum().reshape(1)]
+        for name, b in zip(self.STATE_NAMES, batch):
+            getattr(self, name).add_(b)
+        self._record_window(batch)
+
+    def _compute_from(self, states) -> torch.Tensor:
+        acc, n = states
+        return acc / n.clamp(min=1e-12)
+
+
+class PrecisionComputation(RecMetricComputation):
+    """Precision at threshold 0.5 (reference metrics/precision.py)."""
+
+    STATE_NAMES = ["true_pos_sum", "false_pos_sum"]
+
+    def update(self, predictions, labels, weights=None) -> None:
+        y = labels.double()
+        w = _weights_or_ones(y, weights)
+        pred_pos = (predictions.double() >= 0.5).double()
+        batch = [
+            (pred_pos * y * w).sum().reshape(1),
+            (pred_pos * (1 - y) * w).sum().reshape(1),
+        ]
+        for name, b in zip(self.STATE_NAMES, batch):
+            getattr(self, name).add_(b)
+        self._record_window(batch)
+
+    def _compute_from(self, states) -> torch.Tensor:
+        tp, fp = states
+        return tp / (tp + fp).clamp(min=1e-12)
+
+
+class RecallComputation(RecMetricComputation):
+    """Recall at threshold 0.5 (reference metrics/recall.py)."""
+
+    STATE_NAMES = ["true_pos_sum", "false_neg_sum"]
+
+    def update(self, predictions, labels, weights=None) -> None:
+        y = labels.double()
+        w = _weights_or_ones(y, weights)
+        pred_pos = (predictions.double() >= 0.5).double()
+        batch = [
+            (pred_pos * y * w).sum().reshape(1),
+            ((1 - pred_pos) * y * w).sum().reshape(1),
+        ]
+        for name, b in zip(self.STATE_NAMES, batch):
+            getattr(self, name).add_(b)
+        self._record_window(batch)
+
+    def _compute_from(self, states) -> torch.Tensor:
+        tp, fn = states
+        return tp / (tp + fn).clamp(min=1e-12)
+
+
+def grouped_auc(
+    predictions: torch.Tensor,
+    labels: torch.Tensor,
+    group_ids: torch.Tensor,
+    weights: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """GAUC: mean per-group AUC over groups with both classes (reference
+    metrics/gauc.py compute_gauc_3d — jagged per-session AUC)."""
+    out = []
+    for g in torch.unique(group_ids):
+        m = group_ids == g
+        y = labels[m]
+        if y.numel() < 2 or y.min() == y.max():
+            continue
+        w = _weights_or_ones(y.double(), weights[m] if weights is not None else None)
+        out.append(_weighted_auc(predictions[m].double(), y.double(), w))
+    if not out:
+        return torch.tensor(0.5, dtype=torch.float64)
+    return torch.stack(out).mean()
+
+
+class AccuracyMetric(RecMetric):
+    COMPUTATION = AccuracyComputation
+    NAME = "accuracy"
+
+
+class PrecisionMetric(RecMetric):
+    COMPUTATION = PrecisionComputation
+    NAME = "precision"
+
+
+class RecallMetric(RecMetric):
+    COMPUTATION = RecallComputation
+    NAME = "recall"
