@@ -129,3 +129,37 @@ class TestQuantGpu:
             torch.testing.assert_close(
                 out_g[k].values().cpu(), out_c[k].values(), atol=1e-4, rtol=1e-4
             )
+
+
+@pytest.mark.gpu
+def test_quant_uvm_matches_device():
+    """QUANT_UVM: pinned-host packed tables read over PCIe match HBM tables."""
+    import torch
+
+    from torchrec_amd.quant.embedding_modules import QuantTableBatchedEmbeddingBags
+
+    torch.manual_seed(0)
+    specs = [("t0", 50, 64), ("t1", 30, 128)]
+    dev = QuantTableBatchedEmbeddingBags(specs, device=torch.device("cuda"))
+    uvm = QuantTableBatchedEmbeddingBags(
+        specs, device=torch.device("cuda"), location="managed"
+    )
+    assert not uvm.qweights.is_cuda and uvm.qweights.is_pinned()
+    for i, (n, rows, dim) in enumerate(specs):
+        w = torch.randn(rows, dim)
+        dev.load_float_table(i, w)
+        # identical packed bytes (CPU/GPU quantize may round the last bit apart)
+        uvm.packed_table(i).copy_(dev.packed_table(i).cpu())
+    B = 8
+    g = torch.Generator().manual_seed(1)
+    lengths = torch.randint(0, 4, (2 * B,), generator=g)
+    indices = torch.cat([
+        torch.randint(0, specs[i // B][1], (int(l),), generator=g)
+        for i, l in enumerate(lengths)
+    ])
+    offsets = torch.zeros(2 * B + 1, dtype=torch.int64)
+    torch.cumsum(lengths, 0, out=offsets[1:])
+    out_dev = dev(indices.cuda(), offsets.cuda())
+    out_uvm = uvm(indices.cuda(), offsets.cuda())
+    torch.cuda.synchronize()
+    assert torch.equal(out_dev, out_uvm)

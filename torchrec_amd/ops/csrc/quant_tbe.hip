@@ -16,6 +16,19 @@
 
 namespace trec_amd {
 
+// UVM-capable pointer: host-pinned packed tables are addressed by the
+// kernels over PCIe (QUANT_UVM compute kernel)
+template <typename T>
+static T* uvm_ptr(const at::Tensor& t) {
+  if (t.numel() == 0) return nullptr;
+  if (t.is_cuda()) return t.data_ptr<T>();
+  TORCH_CHECK(t.is_pinned(), "quant TBE host-resident tensors must be pinned");
+  void* dp = nullptr;
+  TREC_HIP_CHECK(hipHostGetDevicePointer(&dp, t.data_ptr(), 0));
+  return static_cast<T*>(dp);
+}
+
+
 static inline hipStream_t q_stream() {
   return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
@@ -156,7 +169,7 @@ at::Tensor tbe_forward_pooled_int8(
   int grid = grid_for(static_cast<int64_t>(F) * B * lps, kBlockThreads);
 #define TBE_Q_LAUNCH(LPS, CHUNKS)                                                         \
   hipLaunchKernelGGL((tbe_fwd_pooled_int8_kernel<LPS, CHUNKS>), dim3(grid),               \
-                     dim3(kBlockThreads), 0, stream, qweights.data_ptr<uint8_t>(),        \
+                     dim3(kBlockThreads), 0, stream, uvm_ptr<uint8_t>(qweights),           \
                      table_byte_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),    \
                      feat_table.data_ptr<int32_t>(), d_out_offsets.data_ptr<int64_t>(),   \
                      indices.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(), psw_ptr,   \
@@ -221,7 +234,7 @@ at::Tensor tbe_forward_seq_int8(
   int grid = grid_for(N * lps, kBlockThreads);
 #define TBE_QS_LAUNCH(LPS, CHUNKS)                                                        \
   hipLaunchKernelGGL((tbe_fwd_seq_int8_kernel<LPS, CHUNKS>), dim3(grid),                  \
-                     dim3(kBlockThreads), 0, stream, qweights.data_ptr<uint8_t>(),        \
+                     dim3(kBlockThreads), 0, stream, uvm_ptr<uint8_t>(qweights),           \
                      table_byte_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),    \
                      feat_table.data_ptr<int32_t>(), feat_val_offsets.data_ptr<int64_t>(),\
                      indices.data_ptr<int64_t>(), F, N, D_out, out.data_ptr<float>())

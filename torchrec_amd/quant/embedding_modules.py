@@ -70,9 +70,12 @@ class QuantTableBatchedEmbeddingBags(nn.Module):
         feature_table_map: Optional[List[int]] = None,
         pooling: PoolingType = PoolingType.SUM,
         device: Optional[torch.device] = None,
+        location: str = "device",  # "device" (HBM) | "managed" (pinned host, QUANT_UVM)
     ) -> None:
         super().__init__()
         device = device or torch.device("cpu")
+        self._uvm = location == "managed" and device.type == "cuda"
+        qdevice = torch.device("cpu") if self._uvm else device
         self._specs = specs
         T = len(specs)
         self._feature_table_map = feature_table_map or list(range(T))
@@ -89,9 +92,10 @@ class QuantTableBatchedEmbeddingBags(nn.Module):
             d_out.append(d_out[-1] + d)
         self._total_D = d_out[-1]
         self._max_D = max(dims) if dims else 0
-        self.register_buffer(
-            "qweights", torch.zeros(byte_offsets[-1], dtype=torch.uint8, device=device)
-        )
+        qw = torch.zeros(byte_offsets[-1], dtype=torch.uint8, device=qdevice)
+        if getattr(self, "_uvm", False):
+            qw = qw.pin_memory()
+        self.register_buffer("qweights", qw)
         reg = lambda n, t: self.register_buffer(n, t.to(device), persistent=False)
         reg("_table_byte_offsets", torch.tensor(byte_offsets[:-1], dtype=torch.int64))
         reg("_dims_t", torch.tensor(dims, dtype=torch.int32))
