@@ -55,3 +55,69 @@ def _run_quant_shard_test(rank, world_size, sharding_type):
 )
 def test_sharded_quant_inference(sharding_type):
     run_multi_process(_run_quant_shard_test, 2, "gloo", sharding_type)
+
+
+def test_sharded_quant_ec():
+    """Sequence quant inference sharding (reference distributed/quant_embedding.py)."""
+    from tests.dist_utils import run_multi_process
+
+    run_multi_process(_run_quant_ec, 2, "gloo")
+
+
+def _run_quant_ec(rank, world_size):
+    import torch
+    import torch.distributed as dist
+
+    from torchrec_amd.distributed.quant_embedding import (
+        QuantEmbeddingCollectionSharder,
+        ShardedQuantEmbeddingCollection,
+    )
+    from torchrec_amd.distributed.types import (
+        EmbeddingModuleShardingPlan,
+        ParameterSharding,
+        ShardingEnv,
+        ShardingType,
+    )
+    from torchrec_amd.modules.embedding_configs import EmbeddingConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingCollection
+    from torchrec_amd.quant.embedding_modules import (
+        EmbeddingCollection as QuantEC,
+    )
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    torch.manual_seed(0)
+    tables = [
+        EmbeddingConfig(num_embeddings=40, embedding_dim=8, name="t0", feature_names=["f0"]),
+        EmbeddingConfig(num_embeddings=24, embedding_dim=8, name="t1", feature_names=["f1"]),
+    ]
+    float_ec = EmbeddingCollection(tables=tables)
+    qec = QuantEC.from_float(float_ec)
+    plan = EmbeddingModuleShardingPlan(
+        {
+            "t0": ParameterSharding(
+                sharding_type=ShardingType.TABLE_WISE.value, compute_kernel="quant",
+                ranks=[0],
+            ),
+            "t1": ParameterSharding(
+                sharding_type=ShardingType.ROW_WISE.value, compute_kernel="quant",
+                ranks=list(range(world_size)),
+            ),
+        }
+    )
+    env = ShardingEnv.from_process_group(dist.group.WORLD)
+    sharded = ShardedQuantEmbeddingCollection(qec, plan, env)
+    B = 3
+    g = torch.Generator().manual_seed(5 + rank)
+    lengths = torch.randint(0, 3, (2 * B,), generator=g)
+    values = torch.cat([
+        torch.randint(0, tables[i // B].num_embeddings, (int(l),), generator=g)
+        for i, l in enumerate(lengths)
+    ])
+    kjt = KeyedJaggedTensor(keys=["f0", "f1"], values=values, lengths=lengths, stride=B)
+    out = sharded(kjt)
+    ref = qec(kjt)
+    for k in ("f0", "f1"):
+        jt_s = out[k]
+        jt_r = ref[k]
+        torch.testing.assert_close(jt_s.values(), jt_r.values(), atol=1e-6, rtol=1e-6)
+        assert torch.equal(jt_s.lengths(), jt_r.lengths())

@@ -68,7 +68,9 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
         F = len(self._keys)
         W = self._W
         if F == 0:
-            return KeyedJaggedTensor.empty(device=self._out_values.device)
+            kjt = KeyedJaggedTensor.empty(device=self._out_values.device)
+            kjt._dist_value_splits = (self.in_value_splits, self.out_value_splits)
+            return kjt
         if getattr(self, "_vbe_strides_rf", None) is not None:
             return self._wait_impl_vbe()
         # recat (r, f) -> (f, r') with r' in the (possibly staggered) order

@@ -114,7 +114,9 @@ class ShardedEmbeddingCollection(nn.Module):
         self._device = device or torch.device("cpu")
         self._fused_params = dict(fused_params or {})
         self._use_index_dedup = use_index_dedup
-        self._need_indices = module.need_indices()
+        self._need_indices = (
+            module.need_indices() if hasattr(module, "need_indices") else False
+        )
         W = env.world_size
         rank = env.rank
 
@@ -339,9 +341,9 @@ class _ECFusedOptimizer(FusedOptimizer):
         state: Dict[torch.Tensor, Any] = {}
         param_groups: List[Dict[str, Any]] = []
         for tbe in sharded_ec.tbes():
-            inner = tbe._bags
-            if inner.optimizer == 2:
-                continue
+            inner = getattr(tbe, "_bags", None)
+            if inner is None or inner.optimizer == 2:
+                continue  # quant / dense lookups expose no fused state
             for spec, w, st in zip(
                 inner.embedding_specs,
                 inner.split_embedding_weights(),
