@@ -252,14 +252,11 @@ at::Tensor col_sum(const at::Tensor& input) {
   auto out = at::empty({N}, input.options());
   if (N == 0) return out;
   auto in = input.contiguous();
-  constexpr int kVPT = 4;  // 8 B per thread-row (full 512 B line per wave)
-  int ntiles = (int)((N + (int64_t)kBlockThreads * kVPT - 1) / (kBlockThreads * kVPT));
-  // small G keeps the strided final pass negligible; 128 blocks of >=64 rows
-  // saturate HBM for the bandwidth-bound partial pass
-  int G = std::max(1, std::min({128, kMaxBlocks / std::max(ntiles, 1),
-                                (int)((M + 63) / 64)}));
+  constexpr int kVPT = 1;  // scalar columns: measured best at [8192, 1024] bf16
+  int ntiles = (int)((N + kBlockThreads - 1) / kBlockThreads);
+  int G = std::max(1, std::min<int>(kMaxBlocks / std::max(ntiles, 1), (int)((M + 31) / 32)));
   int rows_per_group = (int)((M + G - 1) / G);
-  int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);  // final pass
+  int ftiles = ntiles;  // final pass covers kBlockThreads columns per block
   auto partial = at::empty({(int64_t)G * N}, input.options().dtype(at::kFloat));
   auto stream = ia_stream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, input.scalar_type(),

@@ -124,7 +124,7 @@ class QuantTableBatchedEmbeddingBags(nn.Module):
         per_sample_weights: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         B = (offsets.numel() - 1) // self._num_features
-        if self.qweights.is_cuda:
+        if self.qweights.is_cuda or (self.qweights.is_pinned() and indices.is_cuda):
             ops.hip_ops()
             return torch.ops.trec_amd.tbe_forward_pooled_int8(
                 self.qweights,
@@ -275,7 +275,9 @@ class EmbeddingCollection(nn.Module):
         F = tbe._num_features
         B = features.stride()
         offsets = features.offsets()
-        if tbe.qweights.is_cuda:
+        if tbe.qweights.is_cuda or (
+            tbe.qweights.is_pinned() and features.values().is_cuda
+        ):
             ops.hip_ops()
             feat_val_offsets = offsets[:: B][: F + 1].contiguous()
             rows = torch.ops.trec_amd.tbe_forward_seq_int8(
