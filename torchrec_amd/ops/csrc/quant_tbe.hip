@@ -156,7 +156,7 @@ at::Tensor tbe_forward_pooled_int8(
     const at::Tensor& feat_table, const at::Tensor& d_out_offsets, const at::Tensor& indices,
     const at::Tensor& offsets, const at::Tensor& per_sample_weights, int64_t B,
     int64_t total_D, int64_t max_D, bool mean_pool) {
-  TORCH_CHECK(qweights.is_cuda() && qweights.scalar_type() == at::kByte);
+  TORCH_CHECK((qweights.is_cuda() || qweights.is_pinned()) && qweights.scalar_type() == at::kByte);
   TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048);
   int F = feat_table.numel();
   auto out = at::empty({B, total_D}, qweights.options().dtype(at::kFloat));
@@ -223,7 +223,7 @@ at::Tensor tbe_forward_seq_int8(
     const at::Tensor& qweights, const at::Tensor& table_byte_offsets, const at::Tensor& dims,
     const at::Tensor& feat_table, const at::Tensor& feat_val_offsets, const at::Tensor& indices,
     int64_t D_out, int64_t max_D) {
-  TORCH_CHECK(qweights.is_cuda() && max_D % 4 == 0 && max_D <= 2048);
+  TORCH_CHECK((qweights.is_cuda() || qweights.is_pinned()) && max_D % 4 == 0 && max_D <= 2048);
   int64_t N = indices.numel();
   auto out = at::empty({N, D_out}, qweights.options().dtype(at::kFloat));
   if (N == 0) return out;
