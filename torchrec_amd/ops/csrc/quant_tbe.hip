@@ -159,7 +159,7 @@ at::Tensor tbe_forward_pooled_int8(
   TORCH_CHECK((qweights.is_cuda() || qweights.is_pinned()) && qweights.scalar_type() == at::kByte);
   TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048);
   int F = feat_table.numel();
-  auto out = at::empty({B, total_D}, qweights.options().dtype(at::kFloat));
+  auto out = at::empty({B, total_D}, indices.options().dtype(at::kFloat));
   if (B == 0 || F == 0) return out;
   const float* psw_ptr =
       per_sample_weights.numel() > 0 ? per_sample_weights.data_ptr<float>() : nullptr;
@@ -225,7 +225,7 @@ at::Tensor tbe_forward_seq_int8(
     int64_t D_out, int64_t max_D) {
   TORCH_CHECK((qweights.is_cuda() || qweights.is_pinned()) && max_D % 4 == 0 && max_D <= 2048);
   int64_t N = indices.numel();
-  auto out = at::empty({N, D_out}, qweights.options().dtype(at::kFloat));
+  auto out = at::empty({N, D_out}, indices.options().dtype(at::kFloat));
   if (N == 0) return out;
   int F = feat_table.numel();
   auto stream = q_stream();
