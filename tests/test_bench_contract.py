@@ -6,7 +6,6 @@ import sys
 
 import pytest
 
-pytestmark = pytest.mark.gpu
 
 REQUIRED = [
     "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
@@ -14,6 +13,7 @@ REQUIRED = [
 ]
 
 
+@pytest.mark.gpu
 def test_bench_json_contract():
     out = subprocess.run(
         [sys.executable, "bench.py", "--steps", "3", "--warmup", "1", "--scale", "1e-4"],
@@ -29,3 +29,31 @@ def test_bench_json_contract():
     assert res["higher_is_better"] is True
     assert res["value"] > 0 and res["ms_per_step"] > 0
     assert "global_batch" in res["config"]
+
+
+def test_bench_plan_world8_cpu():
+    """The 8-GPU plan the driver's SCALE run will request must be buildable
+    (no OOM/partition errors) — planner is pure CPU."""
+    import bench
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+
+    model = bench.build_model(1.0)
+    for world in (2, 4, 8):
+        planner = EmbeddingShardingPlanner(
+            topology=Topology(
+                world_size=world, compute_device="cuda", batch_size=8192
+            )
+        )
+        sharder = EmbeddingBagCollectionSharder(
+            fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.01}
+        )
+        plan = planner.plan(model, [sharder])
+        mplan = plan.get_plan_for_module("model.sparse_arch.ebc") or next(
+            iter(plan.plan.values())
+        )
+        entries = dict(mplan.items()) if hasattr(mplan, "items") else dict(mplan)
+        assert len(entries) == len(bench.DLRM_EMB_ROWS)
+        for name, ps in entries.items():
+            assert ps.compute_kernel in ("fused", "fused_uvm", "fused_uvm_caching")
