@@ -260,3 +260,28 @@ def _run_sharded_itep(rank, world_size):
         v = out.values()
         assert v.shape == (4, 8)
         v.sum().backward()
+
+
+def test_logging_handlers():
+    import logging
+
+    from torchrec_amd.distributed.logging_handlers import (
+        EventLoggingHandler,
+        get_logging_handler,
+        register_logging_handler,
+        torchrec_method_logger,
+    )
+
+    h = EventLoggingHandler()
+    register_logging_handler("test", h)
+    assert get_logging_handler("test") is h
+    log = logging.getLogger("trec_amd_test")
+    log.addHandler(h)
+    log.setLevel(logging.DEBUG)
+
+    @torchrec_method_logger(log)
+    def f(x):
+        return x + 1
+
+    assert f(1) == 2
+    assert any("f" in (r.getMessage() or "") for r in h.events)

@@ -513,3 +513,29 @@ def _run_twcw_test(rank, world_size):
 
 def test_twcw_sharding():
     run_multi_process(_run_twcw_test, 4, "gloo")
+
+
+def _run_rs_v(rank, world_size):
+    from torchrec_amd.distributed.comm_ops import reduce_scatter_v_pooled
+
+    splits = [2, 3]  # uneven rows per rank
+    D = 4
+    torch.manual_seed(7)
+    full = torch.arange(sum(splits) * D, dtype=torch.float32).view(sum(splits), D)
+    inp = (full + rank).requires_grad_(True)
+    out = reduce_scatter_v_pooled(inp, splits, dist.group.WORLD).wait()
+    lo = sum(splits[:rank])
+    expected = sum(full + r for r in range(world_size))[lo : lo + splits[rank]]
+    torch.testing.assert_close(out, expected)
+    out.sum().backward()
+    assert inp.grad is not None and inp.grad.shape == inp.shape
+    # backward of rs is allgather: every row's grad is 1 (scaled by 1/W
+    # under gradient division)
+    from torchrec_amd.distributed.comm_ops import get_gradient_division
+
+    scale = 1.0 / world_size if get_gradient_division() else 1.0
+    torch.testing.assert_close(inp.grad, torch.full_like(inp, scale))
+
+
+def test_reduce_scatter_v():
+    run_multi_process(_run_rs_v, 2, "gloo")

@@ -89,9 +89,10 @@ class _All2AllPooledReq(torch.autograd.Function):
             out = flat_in.new_empty(B * sum_D)
             in_splits = [B * D_local] * W
             out_splits = [B * d for d in meta.dim_sum_per_rank]
-        work = dist.all_to_all_single(
-            out, flat_in.contiguous(), out_splits, in_splits, group=pg, async_op=True
-        )
+        with torch.autograd.profiler.record_function("## alltoall_pooled ##"):
+            work = dist.all_to_all_single(
+                out, flat_in.contiguous(), out_splits, in_splits, group=pg, async_op=True
+            )
         myreq.work = work
         myreq.tensor = out
         myreq.meta = meta
@@ -208,9 +209,10 @@ class _All2AllSeqReq(torch.autograd.Function):
     def forward(ctx, pg, myreq, meta: A2ASeqMeta, input: torch.Tensor):  # type: ignore[override]
         D = meta.D
         out = input.new_empty(sum(meta.fwd_out_splits) * D)
-        work = dist.all_to_all_single(
-            out,
-            input.reshape(-1).contiguous(),
+        with torch.autograd.profiler.record_function("## alltoall_pooled ##"):
+            work = dist.all_to_all_single(
+                out,
+                input.reshape(-1).contiguous(),
             [n * D for n in meta.fwd_out_splits],
             [n * D for n in meta.fwd_in_splits],
             group=pg,
@@ -415,3 +417,93 @@ def all_gather_base_pooled(
     myreq = Request(pg=pg)
     dummy = _AllGatherBaseReq.apply(pg, myreq, pooled)
     return TensorAwaitable(lambda: _AllGatherBaseWait.apply(pg, myreq, dummy))
+
+
+# ---------------------------------------------------------------------------
+# reduce-scatter-v: uneven per-rank row splits. RCCL has no native v-variant
+# (comm backend) — decompose into one all_to_all_single of row
+# partials plus a local sum: identical wire bytes to an ideal rs-v, and each
+# of the 7 xGMI links carries its pair's chunk concurrently.
+# ---------------------------------------------------------------------------
+
+
+class _ReduceScatterVReq(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, pg, myreq, splits, input: torch.Tensor):  # type: ignore[override]
+        W = dist.get_world_size(pg)
+        rank = dist.get_rank(pg)
+        D = input.shape[1] if input.dim() == 2 else 1
+        my_rows = splits[rank]
+        out = input.new_empty(W * my_rows * D)
+        in_splits = [s * D for s in splits]
+        out_splits = [my_rows * D] * W
+        with torch.autograd.profiler.record_function("## reduce_scatter_v ##"):
+            work = dist.all_to_all_single(
+                out, input.reshape(-1).contiguous(), out_splits, in_splits,
+                group=pg, async_op=True,
+            )
+        myreq.work = work
+        myreq.tensor = out
+        myreq.meta = (list(splits), D, tuple(input.shape))
+        ctx.myreq = myreq
+        ctx.pg = pg
+        dummy = input.new_empty(0, requires_grad=True)
+        myreq.dummy = dummy
+        return dummy
+
+    @staticmethod
+    def backward(ctx, _grad):  # type: ignore[override]
+        myreq = ctx.myreq
+        myreq.wait_work()
+        splits, D, shape = myreq.meta
+        grad_input = myreq.tensor.view(shape)
+        if GRADIENT_DIVISION:
+            grad_input = grad_input / dist.get_world_size(ctx.pg)
+        myreq.tensor = None
+        return None, None, None, grad_input
+
+
+class _ReduceScatterVWait(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, pg, myreq, _dummy):  # type: ignore[override]
+        myreq.wait_work()
+        splits, D, shape = myreq.meta
+        W = dist.get_world_size(pg)
+        rank = dist.get_rank(pg)
+        my_rows = splits[rank]
+        out = myreq.tensor.view(W, my_rows, D).sum(dim=0)
+        myreq.tensor = None
+        ctx.myreq = myreq
+        ctx.pg = pg
+        return out if len(shape) == 2 else out.reshape(-1)
+
+    @staticmethod
+    def backward(ctx, grad_output):  # type: ignore[override]
+        myreq = ctx.myreq
+        pg = ctx.pg
+        W = dist.get_world_size(pg)
+        rank = dist.get_rank(pg)
+        splits, D, shape = myreq.meta
+        # mirror: every rank gets my grad block (allgather-v via a2a)
+        g = grad_output.reshape(-1).contiguous()
+        send = g.repeat(W)
+        out = g.new_empty(sum(s * D for s in splits))
+        in_splits = [g.numel()] * W
+        out_splits = [s * D for s in splits]
+        work = dist.all_to_all_single(
+            out, send, out_splits, in_splits, group=pg, async_op=True
+        )
+        myreq.work = work
+        myreq.tensor = out
+        return None, None, myreq.dummy
+
+
+def reduce_scatter_v_pooled(
+    pooled: torch.Tensor, splits: List[int], pg: dist.ProcessGroup
+) -> Awaitable[torch.Tensor]:
+    """Uneven reduce-scatter over dim 0 (reference comm_ops.py:1230
+    reduce_scatter_v_pooled): rank r receives the sum over ranks of each
+    rank's rows [sum(splits[:r]) : sum(splits[:r+1]))."""
+    myreq = Request(pg=pg)
+    dummy = _ReduceScatterVReq.apply(pg, myreq, splits, pooled)
+    return TensorAwaitable(lambda: _ReduceScatterVWait.apply(pg, myreq, dummy))
