@@ -257,3 +257,37 @@ def test_dlrm_convergence_learnable_labels():
     first = sum(losses[:5]) / 5
     last = sum(losses[-5:]) / 5
     assert last < first * 0.55, f"no convergence: first={first:.4f} last={last:.4f}"
+
+
+def test_fused_sparse_dist_pipeline_cuda():
+    """Dedicated emb_lookup stream variant on cuda:0."""
+    import bench
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.train_pipeline import TrainPipelineFusedSparseDist
+    from torchrec_amd.distributed.types import ShardingEnv
+
+    device = torch.device("cuda", 0)
+    model = bench.build_model(1e-4)
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.01}
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=1, compute_device="cuda", batch_size=64)
+    )
+    plan = planner.plan(model, [sharder])
+    dmp = DistributedModelParallel(
+        model, env=ShardingEnv.from_local(1, 0), plan=plan, sharders=[sharder],
+        device=device,
+    )
+    opt = torch.optim.SGD([p for p in dmp.parameters() if p.requires_grad], lr=0.01)
+    pipe = TrainPipelineFusedSparseDist(
+        dmp, opt, device, autocast_dtype=torch.bfloat16
+    )
+    batches = bench.make_host_batches(4, 64, 1e-4, seed=3, pin=True)
+    it = bench._CyclingIterator(batches)
+    losses = [float(pipe.progress(it)[0]) for _ in range(6)]
+    torch.cuda.synchronize()
+    assert all(l == l for l in losses)  # no NaNs; stream discipline held

@@ -20,6 +20,7 @@ from typing import Any, Callable, Dict, Iterator, List, Optional, Tuple
 
 import torch
 
+from torchrec_amd.distributed.types import LazyAwaitable
 from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
 from torchrec_amd.streamable import Pipelineable
 
@@ -233,6 +234,109 @@ class TrainPipelineSparseDist:
         self._batches = [next_batch, batch_ip2]
         self._dist_contexts = [d_next]
         return output
+
+
+class _StreamJoinedAwaitable(LazyAwaitable):
+    """Joins the embedding-lookup stream into the consumer stream at wait."""
+
+    def __init__(self, inner, stream) -> None:
+        super().__init__()
+        self._inner = inner
+        self._stream = stream
+
+    def _wait_impl(self):
+        out = self._inner.wait()
+        if self._stream is not None:
+            torch.cuda.current_stream().wait_stream(self._stream)
+            vals = out.values() if hasattr(out, "values") else out
+            if isinstance(vals, torch.Tensor):
+                vals.record_stream(torch.cuda.current_stream())
+        return out
+
+
+class FusedPipelinedForward(PipelinedForward):
+    """compute_and_output_dist on a dedicated emb_lookup stream
+    (reference TrainPipelineFusedSparseDist, train_pipelines.py:1424)."""
+
+    def __call__(self, *args, **kwargs):
+        ctx, tensors_awaitable = self._pipeline._fetch_dist(self._fqn)
+        with torch.autograd.profiler.record_function(
+            f"## wait_sparse_data_dist {self._fqn} ##"
+        ):
+            dist_input = tensors_awaitable.wait()
+        ls = getattr(self._pipeline, "_emb_lookup_stream", None)
+        if ls is None:
+            if self._pipeline._data_dist_stream is not None:
+                torch.cuda.current_stream().wait_stream(self._pipeline._data_dist_stream)
+            return self._module.compute_and_output_dist(ctx, dist_input)
+        ls.wait_stream(self._pipeline._data_dist_stream)
+        with torch.cuda.stream(ls):
+            for kjt in dist_input if isinstance(dist_input, list) else [dist_input]:
+                if hasattr(kjt, "record_stream"):
+                    kjt.record_stream(ls)
+            with torch.autograd.profiler.record_function("## emb_lookup ##"):
+                aw = self._module.compute_and_output_dist(ctx, dist_input)
+        return _StreamJoinedAwaitable(aw, ls)
+
+
+class TrainPipelineFusedSparseDist(TrainPipelineSparseDist):
+    """SparseDist + a dedicated embedding-lookup stream: the TBE gather and
+    pooled output a2a of batch i run concurrent with the dense layers that
+    do not consume them yet (reference train_pipelines.py:1424)."""
+
+    def __init__(self, *args, **kwargs) -> None:
+        super().__init__(*args, **kwargs)
+        self._emb_lookup_stream = (
+            torch.cuda.Stream(priority=-1) if self._device.type == "cuda" else None
+        )
+
+    def _rewrite_model(self) -> None:
+        for fqn, module in self._model.named_modules():
+            if hasattr(module, "compute_and_output_dist") and hasattr(module, "input_dist"):
+                self._pipelined[fqn] = module
+                module.forward = FusedPipelinedForward(fqn, module, self)
+        self._rewritten = True
+
+
+class GradientAccumulationPipeline:
+    """Gradient-accumulation wrapper (reference gradient_accumulation.py):
+    dense optimizer steps every ``accumulation_steps`` batches with loss
+    scaled by 1/N. NOTE (reference semantics): FUSED sparse optimizers apply
+    their update inside every backward — accumulation only defers the dense
+    step, exactly as in the reference wrapper."""
+
+    def __init__(self, pipeline, accumulation_steps: int) -> None:
+        assert accumulation_steps >= 1
+        self._pipeline = pipeline
+        self._n = accumulation_steps
+        self._i = 0
+        self._opt = pipeline._optimizer
+
+        class _NoStep:
+            def __init__(self, opt):
+                self._opt = opt
+
+            def zero_grad(self, set_to_none=True):
+                pass  # keep accumulating
+
+            def step(self):
+                pass
+
+        self._nostep = _NoStep(self._opt)
+
+    def progress(self, dataloader_iter):
+        first_of_window = self._i % self._n == 0
+        last_of_window = (self._i + 1) % self._n == 0
+        if first_of_window:
+            self._opt.zero_grad(set_to_none=True)
+        # swap a no-op optimizer in for non-boundary steps
+        self._pipeline._optimizer = self._opt if last_of_window else self._nostep
+        try:
+            out = self._pipeline.progress(dataloader_iter)
+        finally:
+            self._pipeline._optimizer = self._opt
+        self._i += 1
+        return out
 
 
 class PipelineStage:
