@@ -170,7 +170,12 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
     dense_opt = torch.optim.SGD(
         [p for p in dmp.parameters() if p.requires_grad], lr=LR
     )
-    pipeline = TrainPipelineSparseDist(
+    pipeline_cls = TrainPipelineSparseDist
+    if os.environ.get("TREC_PIPELINE") == "fused":
+        from torchrec_amd.distributed.train_pipeline import TrainPipelineFusedSparseDist
+
+        pipeline_cls = TrainPipelineFusedSparseDist
+    pipeline = pipeline_cls(
         dmp, dense_opt, device, autocast_dtype=torch.bfloat16
     )
 
