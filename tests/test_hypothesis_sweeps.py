@@ -122,3 +122,63 @@ class TestTBEProperties:
             torch.testing.assert_close(
                 tbe.split_embedding_weights()[i], expected, atol=1e-5, rtol=1e-5
             )
+
+
+@given(
+    n_tables=st.integers(1, 4),
+    B=st.integers(1, 9),
+    L=st.integers(0, 4),
+    prec=st.sampled_from(["fp32", "bf16", "fp16"]),
+)
+@settings(max_examples=25, deadline=None)
+def test_tbe_precision_sweep(n_tables, B, L, prec):
+    """weights_precision across random shapes vs fp32-cast reference."""
+    torch.manual_seed(0)
+    specs = [(f"t{i}", 10 + 7 * i, 4 * (i + 1)) for i in range(n_tables)]
+    tbe = TableBatchedEmbeddingBags(specs, weights_precision=prec)
+    g = torch.Generator().manual_seed(B * 10 + L)
+    lengths = torch.randint(0, L + 1, (n_tables * B,), generator=g)
+    vals = [
+        torch.randint(0, specs[i // B][1], (int(l),), generator=g)
+        for i, l in enumerate(lengths)
+    ]
+    indices = torch.cat(vals) if vals and int(lengths.sum()) else torch.empty(0, dtype=torch.int64)
+    offsets = torch.zeros(n_tables * B + 1, dtype=torch.int64)
+    torch.cumsum(lengths, 0, out=offsets[1:])
+    out = tbe(indices, offsets)
+    assert out.shape == (B, sum(s[2] for s in specs))
+    # reference: fp32 gather over the cast weights
+    ws = [w.float() for w in tbe.split_embedding_weights()]
+    for f in range(n_tables):
+        for b in range(B):
+            lo, hi = int(offsets[f * B + b]), int(offsets[f * B + b + 1])
+            ref = ws[f][indices[lo:hi]].sum(0) if hi > lo else torch.zeros(specs[f][2])
+            col0 = sum(s[2] for s in specs[:f])
+            torch.testing.assert_close(
+                out[b, col0 : col0 + specs[f][2]], ref, atol=1e-5, rtol=1e-5
+            )
+
+
+@given(
+    strides=st.lists(st.integers(0, 5), min_size=1, max_size=4),
+    dim_i=st.integers(1, 3),
+)
+@settings(max_examples=25, deadline=None)
+def test_vbe_module_sweep(strides, dim_i):
+    """VBE forward across random per-feature batch sizes."""
+    torch.manual_seed(1)
+    D = 4 * dim_i
+    F = len(strides)
+    specs = [(f"t{i}", 20, D) for i in range(F)]
+    tbe = TableBatchedEmbeddingBags(specs)
+    g = torch.Generator().manual_seed(7)
+    lengths = torch.cat([torch.randint(0, 3, (s,), generator=g) for s in strides]) \
+        if sum(strides) else torch.empty(0, dtype=torch.int64)
+    indices = torch.randint(0, 20, (int(lengths.sum()),), generator=g) \
+        if lengths.numel() else torch.empty(0, dtype=torch.int64)
+    offsets = torch.zeros(int(sum(strides)) + 1, dtype=torch.int64)
+    if lengths.numel():
+        torch.cumsum(lengths, 0, out=offsets[1:])
+    out = tbe.forward_vbe(indices, offsets, list(strides))
+    assert out.numel() == sum(s * D for s in strides)
+    out.sum().backward() if out.numel() else None

@@ -568,3 +568,52 @@ class TestSegSort:
         ref = torch.sort(linear).values
         assert torch.equal(sorted_l.cpu(), ref)
         assert torch.equal(linear[perm.cpu().long()], sorted_l.cpu())
+
+
+class TestSegSortGating:
+    def test_shared_table_disables_seg_sort(self):
+        """Shared tables (non-injective feature_table_map) must not take the
+        segmented-sort fast path — the concatenation would not be globally
+        grouped and two wave-slots could race on one row."""
+        tbe = TableBatchedEmbeddingBags(
+            [("t0", 50, 8)], feature_table_map=[0, 0], fixed_bag_length=1
+        )
+        assert not tbe._seg_sort_ok
+
+    def test_out_of_order_tables_disable_seg_sort(self):
+        tbe = TableBatchedEmbeddingBags(
+            [("t0", 50, 8), ("t1", 40, 8)], feature_table_map=[1, 0],
+            fixed_bag_length=1,
+        )
+        assert not tbe._seg_sort_ok
+
+    def test_ordered_tables_enable_seg_sort(self):
+        tbe = TableBatchedEmbeddingBags(
+            [("t0", 50, 8), ("t1", 40, 8)], fixed_bag_length=1
+        )
+        assert tbe._seg_sort_ok
+
+    @pytest.mark.gpu
+    def test_shared_table_fallback_correct_on_gpu(self):
+        """With a shared table the hipCUB path runs and stays correct."""
+        specs = [("t0", 64, 32)]
+        torch.manual_seed(0)
+        cpu = TableBatchedEmbeddingBags(specs, feature_table_map=[0, 0], learning_rate=0.05)
+        gpu = TableBatchedEmbeddingBags(
+            specs, feature_table_map=[0, 0], learning_rate=0.05,
+            device=torch.device("cuda"), fixed_bag_length=2,
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        B = 8
+        lengths = torch.full((2 * B,), 2, dtype=torch.int64)
+        g = torch.Generator().manual_seed(3)
+        indices = torch.randint(0, 64, (int(lengths.sum()),), generator=g)
+        offsets = torch.zeros(2 * B + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        out_c = cpu(indices, offsets)
+        out_g = gpu(indices.cuda(), offsets.cuda())
+        out_c.sum().backward()
+        out_g.sum().backward()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(out_g.cpu(), out_c, atol=1e-5, rtol=1e-5)
+        torch.testing.assert_close(gpu.weights.cpu(), cpu.weights, atol=1e-5, rtol=1e-5)
