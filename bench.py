@@ -255,12 +255,6 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
         # segmented backward sort
         "fixed_bag_length": IDS_PER_FEATURE,
     }
-    if qcomm != "none":
-        from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
-
-        fused_params["qcomms_config"] = QCommsConfig(
-            forward_precision=CommType(qcomm), backward_precision=CommType(qcomm)
-        )
     sharder = EmbeddingBagCollectionSharder(fused_params=fused_params)
     planner = EmbeddingShardingPlanner(
         topology=Topology(world_size=1, compute_device="cuda", batch_size=batch_size)
