@@ -173,3 +173,55 @@ class TestValidatorsAndInterop:
         assert kjt.keys() == ["a", "b"]
         assert kjt["a"].values().tolist() == [1, 2, 3, 4]
         assert kjt["b"].lengths().tolist() == [1, 0]
+
+
+class TestOpSurface:
+    def test_jagged_index_select_2d(self):
+        from torchrec_amd import ops
+
+        values = torch.arange(10)
+        lengths = torch.tensor([2, 3, 1, 4])
+        idx = torch.tensor([2, 0, 2])
+        v, l = ops.jagged_index_select_2d(values, lengths, idx)
+        assert l.tolist() == [1, 2, 1]
+        assert v.tolist() == [5, 0, 1, 5]
+
+    def test_jagged_unique_indices(self):
+        from torchrec_amd import ops
+
+        # two buckets over the lengths array: features [0,1) and [1,2)
+        hash_off = torch.tensor([0, 1, 2])
+        offsets = torch.tensor([0, 3, 6])
+        indices = torch.tensor([4, 1, 4, 1, 1, 2])
+        ol, oo, uniq, rev = ops.jagged_unique_indices(hash_off, offsets, indices)
+        assert ol.tolist() == [2, 2]  # {1,4} and {1,2}
+        assert oo.tolist() == [0, 2, 4]
+        # reverse index reconstructs the original stream bucket-locally
+        recon = uniq[rev]
+        assert recon.tolist() == indices.tolist()
+
+    def test_group_and_batch_index_select(self):
+        from torchrec_amd import ops
+
+        t0 = torch.arange(12.0).view(4, 3)
+        t1 = torch.arange(10.0).view(5, 2)
+        g = ops.group_index_select_dim0([t0, t1], [torch.tensor([1, 3]), torch.tensor([0])])
+        assert torch.equal(g[0], t0[[1, 3]])
+        assert torch.equal(g[1], t1[[0]])
+        flat = torch.cat([t0.reshape(-1), t1.reshape(-1)])
+        out = ops.batch_index_select_dim0(
+            flat, torch.tensor([1, 3, 0]), [2, 1], [4, 5], [3, 2]
+        )
+        assert torch.equal(out, torch.cat([t0[[1, 3]].reshape(-1), t1[[0]].reshape(-1)]))
+
+    def test_expand_into_jagged_permute(self):
+        from torchrec_amd import ops
+
+        # segments sized [2,1,3] permuted [2,0,1]
+        in_off = torch.tensor([0, 2, 3, 6])
+        perm = torch.tensor([2, 0, 1])
+        out_sizes = torch.tensor([3, 2, 1])
+        out_off = torch.zeros(4, dtype=torch.int64)
+        torch.cumsum(out_sizes, 0, out=out_off[1:])
+        pp = ops.expand_into_jagged_permute(perm, in_off, out_off, 6)
+        assert pp.tolist() == [3, 4, 5, 0, 1, 2]

@@ -430,3 +430,97 @@ def fused_interaction(dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor
     """GPU: fused kernel; CPU callers should use the eager path."""
     hip_ops()
     return _FusedInteraction.apply(dense, sparse)
+
+
+def jagged_index_select_2d(
+    values: torch.Tensor, lengths: torch.Tensor, indices: torch.Tensor
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Select (and possibly duplicate) jagged segments by segment index.
+
+    Semantics of ``fbgemm.jagged_index_select_2d_forward_v2`` (reference:
+    torchrec/modules/utils.py:427). Returns (selected_values,
+    selected_lengths)."""
+    pl, pv, _ = permute_1d_sparse_data(indices, lengths, values)
+    return pv, pl
+
+
+def jagged_unique_indices(
+    hash_size_offsets: torch.Tensor,
+    offsets: torch.Tensor,
+    indices: torch.Tensor,
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Dedup ids within hash-size buckets (reference fbgemm
+    jagged_unique_indices, used for sequence input dedup at
+    torchrec/distributed/embedding.py:1439).
+
+    Returns (output_lengths, output_offsets, unique_indices, reverse_index):
+    ids are linearized by their bucket, deduped (sorted), and every original
+    position maps to its unique slot via ``reverse_index``."""
+    n_buckets = hash_size_offsets.numel() - 1
+    lengths = offsets[1:] - offsets[:-1]
+    # bucket of each position (buckets delimit ranges of the lengths array)
+    pos_bucket = torch.repeat_interleave(
+        torch.arange(n_buckets, device=indices.device),
+        (
+            offsets[hash_size_offsets[1:].to(torch.int64)]
+            - offsets[hash_size_offsets[:-1].to(torch.int64)]
+        ),
+    ) if n_buckets else torch.zeros_like(indices)
+    lin = indices + pos_bucket * (int(indices.max()) + 1 if indices.numel() else 1)
+    uniq, inverse = torch.unique(lin, sorted=True, return_inverse=True)
+    u_bucket = torch.div(
+        uniq, (int(indices.max()) + 1 if indices.numel() else 1), rounding_mode="floor"
+    )
+    out_lengths = torch.bincount(u_bucket, minlength=n_buckets)
+    out_offsets = torch.zeros(n_buckets + 1, dtype=offsets.dtype, device=offsets.device)
+    torch.cumsum(out_lengths, 0, out=out_offsets[1:])
+    uniq_local = uniq - u_bucket * (int(indices.max()) + 1 if indices.numel() else 1)
+    return out_lengths, out_offsets, uniq_local, inverse
+
+
+def group_index_select_dim0(
+    tensors: List[torch.Tensor], indices: List[torch.Tensor]
+) -> List[torch.Tensor]:
+    """Grouped row select (reference fbgemm group_index_select_dim0, used at
+    torchrec/modules/embedding_modules.py:78)."""
+    return [t.index_select(0, i.to(t.device)) for t, i in zip(tensors, indices)]
+
+
+def batch_index_select_dim0(
+    flat: torch.Tensor,
+    indices: torch.Tensor,
+    input_num_indices: List[int],
+    input_rows: List[int],
+    input_columns: List[int],
+) -> torch.Tensor:
+    """Batched row select over a flat buffer of stacked tables (reference
+    fbgemm batch_index_select_dim0, torchrec/distributed/embeddingbag.py:411)."""
+    outs = []
+    row_off = 0
+    idx_off = 0
+    for n_idx, rows, cols in zip(input_num_indices, input_rows, input_columns):
+        table = flat[row_off : row_off + rows * cols].view(rows, cols)
+        idx = indices[idx_off : idx_off + n_idx]
+        outs.append(table.index_select(0, idx).reshape(-1))
+        row_off += rows * cols
+        idx_off += n_idx
+    return torch.cat(outs) if outs else flat.new_empty(0)
+
+
+def expand_into_jagged_permute(
+    permute: torch.Tensor,
+    input_offsets: torch.Tensor,
+    output_offsets: torch.Tensor,
+    output_size: int,
+) -> torch.Tensor:
+    """Expand a segment-level permute into a position-level permute
+    (reference fbgemm expand_into_jagged_permute, VBE output permute at
+    torchrec/distributed/dist_data.py:340): output position j of segment i
+    maps to input position input_offsets[permute[i]] + j."""
+    n = permute.numel()
+    out = torch.empty(output_size, dtype=torch.int64, device=permute.device)
+    for i in range(n):
+        o0, o1 = int(output_offsets[i]), int(output_offsets[i + 1])
+        src = int(input_offsets[int(permute[i])])
+        out[o0:o1] = torch.arange(src, src + (o1 - o0), device=permute.device)
+    return out
