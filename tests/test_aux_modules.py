@@ -285,3 +285,30 @@ def test_logging_handlers():
 
     assert f(1) == 2
     assert any("f" in (r.getMessage() or "") for r in h.events)
+
+
+def test_local_shards_wrapper_and_stashing():
+    from torchrec_amd.distributed.memory_stashing import MemoryStashingManager
+    from torchrec_amd.distributed.shards_wrapper import LocalShardsWrapper
+
+    w = LocalShardsWrapper([torch.ones(2, 3), torch.zeros(1, 3)], [(0, 0), (2, 0)])
+    assert w.numel() == 9 and w.local_offsets() == [(0, 0), (2, 0)]
+    w2 = w.to(torch.float64)
+    assert w2.dtype == torch.float64
+
+    m = MemoryStashingManager()
+    t = torch.arange(6.0)
+    m.stash("a", t)
+    assert "a" in m.stashed()
+    back = m.unstash("a")
+    torch.testing.assert_close(back, t)
+
+
+def test_maybe_td_passthrough():
+    from torchrec_amd.sparse.tensor_dict import maybe_td_to_kjt
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    kjt = KeyedJaggedTensor(
+        keys=["f"], values=torch.tensor([1, 2]), lengths=torch.tensor([2]), stride=1
+    )
+    assert maybe_td_to_kjt(kjt) is kjt  # tensordict absent or not a TD
