@@ -79,6 +79,9 @@ class RecMetricComputation(abc.ABC, nn.Module):
     def _reduced(self, states: List[torch.Tensor]) -> List[torch.Tensor]:
         if self._pg is None and not (dist.is_available() and dist.is_initialized()):
             return states
+        if dist.get_world_size(self._pg) <= 1:
+            return states  # single rank: nothing to reduce (and the default
+            # pg's backend may not cover CPU state tensors)
         pg = self._pg
         out = []
         for s in states:
@@ -160,7 +163,11 @@ class AUCComputation(RecMetricComputation):
         p = torch.cat(self._preds)
         y = torch.cat(self._labels)
         w = torch.cat(self._weights)
-        if dist.is_available() and dist.is_initialized():
+        if (
+            dist.is_available()
+            and dist.is_initialized()
+            and dist.get_world_size(self._pg) > 1
+        ):
             gp = [None] * dist.get_world_size(self._pg)
             dist.all_gather_object(gp, (p, y, w), group=self._pg)
             p = torch.cat([t[0] for t in gp])
