@@ -124,7 +124,7 @@ def bench_col_sum(M=8192, N=1024):
 if __name__ == "__main__":
     p = argparse.ArgumentParser()
     p.add_argument("--suite", default="all",
-                   choices=["all", "tbe", "interaction", "sort", "col_sum"])
+                   choices=["all", "tbe", "interaction", "sort", "col_sum", "quant"])
     a = p.parse_args()
     assert torch.cuda.is_available(), "run on a GPU box"
     if a.suite in ("all", "tbe"):
@@ -135,3 +135,28 @@ if __name__ == "__main__":
         bench_sort()
     if a.suite in ("all", "col_sum"):
         bench_col_sum()
+    if a.suite in ("all", "quant"):
+        bench_quant_tbe()
+
+
+def bench_quant_tbe(B=8192, D=128, tables=26, rows=100_000):
+    """int8 inference TBE vs fp32 training TBE forward."""
+    from torchrec_amd.quant.embedding_modules import QuantTableBatchedEmbeddingBags
+    from torchrec_amd.ops.tbe import TableBatchedEmbeddingBags
+
+    torch.manual_seed(0)
+    specs = [(f"t{i}", rows, D) for i in range(tables)]
+    q = QuantTableBatchedEmbeddingBags(specs, device=torch.device("cuda"))
+    for i in range(tables):
+        q.load_float_table(i, torch.randn(rows, D, device="cuda"))
+    f32 = TableBatchedEmbeddingBags(specs, device=torch.device("cuda"))
+    F = tables
+    indices = torch.randint(0, rows, (F * B,)).cuda()
+    offsets = torch.arange(F * B + 1, dtype=torch.int64).cuda()
+    qt = _time_kernel(lambda: q(indices, offsets))
+    ft = _time_kernel(lambda: f32(indices, offsets))
+    print(json.dumps({
+        "bench": "quant_tbe_fwd", "B": B, "tables": tables, "D": D,
+        "int8_us": round(qt, 1), "fp32_us": round(ft, 1),
+        "int8_qps_m": round(B / qt, 1),
+    }))
