@@ -121,24 +121,6 @@ def bench_col_sum(M=8192, N=1024):
     }))
 
 
-if __name__ == "__main__":
-    p = argparse.ArgumentParser()
-    p.add_argument("--suite", default="all",
-                   choices=["all", "tbe", "interaction", "sort", "col_sum", "quant"])
-    a = p.parse_args()
-    assert torch.cuda.is_available(), "run on a GPU box"
-    if a.suite in ("all", "tbe"):
-        bench_tbe()
-    if a.suite in ("all", "interaction"):
-        bench_interaction()
-    if a.suite in ("all", "sort"):
-        bench_sort()
-    if a.suite in ("all", "col_sum"):
-        bench_col_sum()
-    if a.suite in ("all", "quant"):
-        bench_quant_tbe()
-
-
 def bench_quant_tbe(B=8192, D=128, tables=26, rows=100_000):
     """int8 inference TBE vs fp32 training TBE forward."""
     from torchrec_amd.quant.embedding_modules import QuantTableBatchedEmbeddingBags
@@ -160,3 +142,21 @@ def bench_quant_tbe(B=8192, D=128, tables=26, rows=100_000):
         "int8_us": round(qt, 1), "fp32_us": round(ft, 1),
         "int8_qps_m": round(B / qt, 1),
     }))
+
+
+if __name__ == "__main__":
+    p = argparse.ArgumentParser()
+    p.add_argument("--suite", default="all",
+                   choices=["all", "tbe", "interaction", "sort", "col_sum", "quant"])
+    a = p.parse_args()
+    assert torch.cuda.is_available(), "run on a GPU box"
+    if a.suite in ("all", "tbe"):
+        bench_tbe()
+    if a.suite in ("all", "interaction"):
+        bench_interaction()
+    if a.suite in ("all", "sort"):
+        bench_sort()
+    if a.suite in ("all", "col_sum"):
+        bench_col_sum()
+    if a.suite in ("all", "quant"):
+        bench_quant_tbe()
