@@ -31,13 +31,14 @@ def _time_kernel(fn, iters=50, warmup=10) -> float:
     return (time.perf_counter() - t0) / iters * 1e6  # us
 
 
-def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1):
+def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1, precision="fp32"):
     from torchrec_amd.ops.tbe import TableBatchedEmbeddingBags
 
     torch.manual_seed(0)
     specs = [(f"t{i}", rows, D) for i in range(tables)]
     tbe = TableBatchedEmbeddingBags(
-        specs, device=torch.device("cuda"), fixed_bag_length=L
+        specs, device=torch.device("cuda"), fixed_bag_length=L,
+        weights_precision=precision,
     )
     F = tables
     lengths = torch.full((F * B,), L, dtype=torch.int64)
@@ -56,8 +57,11 @@ def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1):
 
     full_us = _time_kernel(step)
     bytes_moved = F * B * L * D * 4 * 2 + B * F * D * 4  # read rows + write out
+    elem = 4 if precision == "fp32" else 2
+    bytes_moved = F * B * L * D * elem + B * F * D * 4
     print(json.dumps({
         "bench": "tbe", "B": B, "D": D, "tables": tables, "rows_per_table": rows,
+        "precision": precision,
         "fwd_us": round(fwd_us, 1), "fwd_bwd_opt_us": round(full_us, 1),
         "fwd_gb_s": round(bytes_moved / fwd_us / 1e3, 1),
     }))
@@ -152,6 +156,7 @@ if __name__ == "__main__":
     assert torch.cuda.is_available(), "run on a GPU box"
     if a.suite in ("all", "tbe"):
         bench_tbe()
+        bench_tbe(precision="bf16")
     if a.suite in ("all", "interaction"):
         bench_interaction()
     if a.suite in ("all", "sort"):
