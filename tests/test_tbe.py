@@ -617,3 +617,33 @@ class TestSegSortGating:
         torch.cuda.synchronize()
         torch.testing.assert_close(out_g.cpu(), out_c, atol=1e-5, rtol=1e-5)
         torch.testing.assert_close(gpu.weights.cpu(), cpu.weights, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.gpu
+def test_seg_sort_large_matches_device_sort():
+    """B large enough to exceed the block-sort tile: the rocPRIM segmented
+    path must match the plain hipCUB path exactly."""
+    specs = [("t0", 3000, 64), ("t1", 5000, 64)]
+    torch.manual_seed(0)
+    a = TableBatchedEmbeddingBags(specs, device=torch.device("cuda"), learning_rate=0.05)
+    b = TableBatchedEmbeddingBags(
+        specs, device=torch.device("cuda"), learning_rate=0.05, fixed_bag_length=1
+    )
+    b.weights.data.copy_(a.weights.data)
+    B = 20000  # cap = 20000 > 16384 -> large segmented path
+    g = torch.Generator().manual_seed(4)
+    lengths = torch.ones(2 * B, dtype=torch.int64)
+    indices = torch.cat(
+        [torch.randint(0, specs[f][1], (B,), generator=g) for f in range(2)]
+    )
+    offsets = torch.zeros(2 * B + 1, dtype=torch.int64)
+    torch.cumsum(lengths, 0, out=offsets[1:])
+    ic, oc = indices.cuda(), offsets.cuda()
+    out_a = a(ic, oc)
+    out_b = b(ic, oc)
+    out_a.sum().backward()
+    out_b.sum().backward()
+    torch.cuda.synchronize()
+    assert torch.equal(out_a, out_b)
+    assert torch.equal(a.weights, b.weights)
+    assert torch.equal(a.momentum, b.momentum)

@@ -45,6 +45,8 @@ std::tuple<at::Tensor, at::Tensor> sort_pairs(const at::Tensor& keys, int64_t en
 std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs(
     const at::Tensor& linear, const at::Tensor& offsets, int64_t B, int64_t F,
     int64_t end_bit, int64_t capacity);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_large(
+    const at::Tensor& linear, const at::Tensor& feat_bounds, int64_t F, int64_t end_bit);
 std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_linear);
 void tbe_backward_fused(at::Tensor weights, at::Tensor momentum, const at::Tensor& grad,
                         const at::Tensor& sorted_linear, const at::Tensor& sort_perm,
@@ -140,6 +142,9 @@ TORCH_LIBRARY(trec_amd, m) {
   m.def(
       "seg_sort_pairs(Tensor linear, Tensor offsets, int B, int F, int end_bit, "
       "int capacity) -> (Tensor, Tensor, Tensor)");
+  m.def(
+      "seg_sort_pairs_large(Tensor linear, Tensor feat_bounds, int F, int end_bit)"
+      " -> (Tensor, Tensor, Tensor)");
   m.def("tbe_backward_prep(Tensor sorted_linear) -> (Tensor, Tensor)");
   m.def(
       "tbe_backward_fused(Tensor(a!) weights, Tensor(b!) momentum, Tensor grad,"
@@ -195,6 +200,7 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("tbe_forward_pooled_vbe", trec_amd::tbe_forward_pooled_vbe);
   m.impl("sort_pairs", trec_amd::sort_pairs);
   m.impl("seg_sort_pairs", trec_amd::seg_sort_pairs);
+  m.impl("seg_sort_pairs_large", trec_amd::seg_sort_pairs_large);
   m.impl("tbe_backward_prep", trec_amd::tbe_backward_prep);
   m.impl("tbe_backward_fused", trec_amd::tbe_backward_fused);
   m.impl("tbe_grad_per_sample_weights", trec_amd::tbe_grad_per_sample_weights);

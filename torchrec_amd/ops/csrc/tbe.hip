@@ -25,6 +25,7 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hipcub/hipcub.hpp>
 #include <rocprim/block/block_radix_sort.hpp>
+#include <rocprim/device/device_segmented_radix_sort.hpp>
 
 #include <type_traits>
 
@@ -467,6 +468,40 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs(
   }
 #undef SEG_SORT_LAUNCH
   return {sorted, perm, overflow};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_large(
+    const at::Tensor& linear, const at::Tensor& feat_bounds /* [F+1] */, int64_t F,
+    int64_t end_bit) {
+  // device segmented radix sort (rocPRIM): segments of any size, ONE public
+  // call — the large-batch (B > block-sort LDS tile) backward path
+  TORCH_CHECK(linear.is_cuda() && linear.scalar_type() == at::kLong);
+  TORCH_CHECK(end_bit <= 31, "segmented sort requires 32-bit ids");
+  auto stream = tbe_stream();
+  int64_t n = linear.numel();
+  auto perm_in = at::arange(n, linear.options().dtype(at::kInt));
+  auto perm = at::empty_like(perm_in);
+  auto overflow = at::zeros({1}, linear.options().dtype(at::kInt));
+  if (n == 0 || F == 0) return {at::empty_like(linear), perm, overflow};
+  auto k32 = linear.to(at::kInt);
+  auto k32_out = at::empty_like(k32);
+  auto bounds32 = feat_bounds.to(at::kInt);
+  size_t tmp_bytes = 0;
+  auto err = rocprim::segmented_radix_sort_pairs(
+      nullptr, tmp_bytes, k32.data_ptr<int32_t>(), k32_out.data_ptr<int32_t>(),
+      perm_in.data_ptr<int32_t>(), perm.data_ptr<int32_t>(), (unsigned)n, (unsigned)F,
+      bounds32.data_ptr<int32_t>(), bounds32.data_ptr<int32_t>() + 1, 0,
+      (unsigned)end_bit, stream);
+  TORCH_CHECK(err == hipSuccess, "segmented_radix_sort sizing failed");
+  auto tmp = at::empty({static_cast<int64_t>(tmp_bytes)},
+                       linear.options().dtype(at::kByte));
+  err = rocprim::segmented_radix_sort_pairs(
+      tmp.data_ptr(), tmp_bytes, k32.data_ptr<int32_t>(), k32_out.data_ptr<int32_t>(),
+      perm_in.data_ptr<int32_t>(), perm.data_ptr<int32_t>(), (unsigned)n, (unsigned)F,
+      bounds32.data_ptr<int32_t>(), bounds32.data_ptr<int32_t>() + 1, 0,
+      (unsigned)end_bit, stream);
+  TORCH_CHECK(err == hipSuccess, "segmented_radix_sort failed");
+  return {k32_out.to(at::kLong), perm, overflow};
 }
 
 __global__ void mark_runs_kernel(const int64_t* __restrict__ sorted, int64_t n,
