@@ -420,16 +420,11 @@ class TableBatchedEmbeddingBags(nn.Module):
                 linear, offsets, B, self._num_features,
                 _bits_needed(self._total_rows), cap,
             )
-        elif self._seg_sort_ok and cap > 16384:
-            # large batches: rocPRIM device segmented sort (any segment size)
-            fb = offsets[:: B][: self._num_features + 1]
-            if fb.numel() < self._num_features + 1:
-                fb = torch.cat([fb, offsets[-1:]])
-            sorted_lin, perm, _overflow = torch.ops.trec_amd.seg_sort_pairs_large(
-                linear, fb.contiguous(), self._num_features,
-                _bits_needed(self._total_rows),
-            )
         else:
+            # NOTE: rocPRIM's device segmented sort (seg_sort_pairs_large)
+            # was measured 7x SLOWER than hipCUB's device radix for few large
+            # segments (1169 vs 168 us at 26 x 65536 keys) — per-segment
+            # parallelism collapses — so large batches stay on hipCUB
             sorted_lin, perm = torch.ops.trec_amd.sort_pairs(
                 linear, _bits_needed(self._total_rows)
             )
