@@ -45,12 +45,18 @@ def build_model(scale: float):
     from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
     from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
 
+    from torchrec_amd.modules.embedding_configs import DataType
+
+    dt = {"fp32": DataType.FP32, "bf16": DataType.BF16, "fp16": DataType.FP16}[
+        emb_precision
+    ]
     tables = [
         EmbeddingBagConfig(
             num_embeddings=r,
             embedding_dim=EMB_DIM,
             name=f"t_cat_{i}",
             feature_names=[f"cat_{i}"],
+            data_type=dt,
         )
         for i, r in enumerate(scaled_rows(scale))
     ]
@@ -160,7 +166,8 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
         env = ShardingEnv.from_local(1, 0)
         pg = None
 
-    model = build_model(scale)
+    emb_precision = os.environ.get("TREC_EMB_PRECISION", "fp32")
+    model = build_model(scale, emb_precision)
     fused_params = {
         "optimizer": "rowwise_adagrad",
         "learning_rate": LR,
@@ -249,7 +256,7 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
             "tables": len(DLRM_EMB_ROWS),
             "embedding_dim": EMB_DIM,
             "row_scale": scale,
-            "emb_dtype": "fp32",
+            "emb_dtype": emb_precision,
             "dense_dtype": "bf16-autocast",
             "parallelism": f"planner/dmp x{world} + pipeline",
             "qcomm": qcomm,
@@ -275,7 +282,8 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
     device = torch.device("cuda", 0)
     torch.cuda.set_device(device)
     enable_tuned_gemms()
-    model = build_model(scale)
+    emb_precision = os.environ.get("TREC_EMB_PRECISION", "fp32")
+    model = build_model(scale, emb_precision)
     fused_params = {
         "optimizer": "rowwise_adagrad",
         "learning_rate": LR,
