@@ -39,7 +39,7 @@ def scaled_rows(scale: float) -> List[int]:
     return [max(10, int(r * scale)) for r in DLRM_EMB_ROWS]
 
 
-def build_model(scale: float):
+def build_model(scale: float, emb_precision: str = "fp32"):
     """DLRM over a meta-device EBC: only local shards materialize on GPU."""
     from torchrec_amd.models.dlrm import DLRM, DLRMTrain
     from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
