@@ -151,3 +151,32 @@ def test_accuracy_precision_recall_gauc():
     labels2 = torch.tensor([1.0, 1.0, 0.0, 1.0, 0.0, 0.0])
     gauc = grouped_auc(preds2, labels2, g)
     assert abs(float(gauc) - 0.5) < 1e-6  # (1.0 + 0.0) / 2
+
+
+def test_cpu_offloaded_metric_module():
+    import torch
+
+    from torchrec_amd.metrics.cpu_offloaded_metric_module import (
+        CPUOffloadedRecMetricModule,
+    )
+    from torchrec_amd.metrics.metric_module import RecMetricModule
+    from torchrec_amd.metrics.rec_metric import NEMetric, RecTaskInfo
+
+    tasks = [RecTaskInfo(name="t")]
+    off = CPUOffloadedRecMetricModule(
+        batch_size=4, world_size=1, rec_tasks=tasks, rec_metrics=[NEMetric(tasks)]
+    )
+    sync = RecMetricModule(
+        batch_size=4, world_size=1, rec_tasks=tasks, rec_metrics=[NEMetric(tasks)]
+    )
+    g = torch.Generator().manual_seed(0)
+    for _ in range(5):
+        p = torch.rand(16, generator=g)
+        y = (torch.rand(16, generator=g) > 0.5).float()
+        off.update(predictions={"t": p}, labels={"t": y})
+        sync.update(predictions={"t": p}, labels={"t": y})
+    a = off.compute()
+    b = sync.compute()
+    for k in b:
+        torch.testing.assert_close(a[k], b[k])
+    off.shutdown()
