@@ -539,3 +539,57 @@ def _run_rs_v(rank, world_size):
 
 def test_reduce_scatter_v():
     run_multi_process(_run_rs_v, 2, "gloo")
+
+
+def _run_mixed_sharding(rank, world_size):
+    """One EBC with tables simultaneously TW, RW, CW and DP: the per-type
+    shardings compose (split input dist, concat output, canonical permute)."""
+    B = 4
+    tables = make_tables()
+    mix = [
+        ShardingType.TABLE_WISE.value,
+        ShardingType.ROW_WISE.value,
+        ShardingType.COLUMN_WISE.value,
+        ShardingType.DATA_PARALLEL.value,
+    ]
+    torch.manual_seed(42)
+    model = SparseModel(make_tables())
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(sharding_types=[mix[i]], min_partition=4)
+            for i, cfg in enumerate(tables)
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    st_by_table = {n: ps.sharding_type for n, ps in plan.plan["sparse"].items()}
+    assert st_by_table == {f"t{i}": mix[i] for i in range(4)}
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], device=torch.device("cpu"),
+        init_data_parallel=False,
+    )
+    golden = _golden(tables, None, world_size)
+    dmp.load_state_dict(
+        {
+            f"sparse.embedding_bags.{cfg.name}.weight": w
+            for cfg, w in zip(tables, golden.split_embedding_weights())
+        },
+        strict=False,
+    )
+    kjt_global = make_global_kjt(tables, B * world_size)
+    kjt_local = kjt_local_slice(kjt_global, rank * B, (rank + 1) * B)
+    kt = dmp(kjt_local)
+    vals = kt.values()
+    golden_out = golden(kjt_global).values()
+    torch.testing.assert_close(
+        vals, golden_out[rank * B : (rank + 1) * B], atol=1e-5, rtol=1e-5
+    )
+    assert kt.keys() == [c.feature_names[0] for c in tables]
+    vals.sum().backward()  # all four backward paths coexist
+
+
+def test_mixed_sharding_types_one_ebc():
+    run_multi_process(_run_mixed_sharding, 2, "gloo")

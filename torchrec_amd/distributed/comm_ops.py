@@ -113,6 +113,10 @@ class _All2AllPooledReq(torch.autograd.Function):
         if GRADIENT_DIVISION:
             grad_input = grad_input / dist.get_world_size(ctx.pg)
         myreq.tensor = None
+        if meta.D_local == 0:
+            # featureless rank: zero-width grad with the right row count
+            W = dist.get_world_size(ctx.pg)
+            return None, None, None, grad_input.new_zeros(W * meta.B, 0)
         return None, None, None, grad_input.view(-1, meta.D_local)
 
 

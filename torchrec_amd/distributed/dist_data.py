@@ -68,7 +68,11 @@ class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
         F = len(self._keys)
         W = self._W
         if F == 0:
-            kjt = KeyedJaggedTensor.empty(device=self._out_values.device)
+            # a feature-less rank still participates in the pooled output a2a:
+            # keep the GLOBAL batch stride so [W*B, 0] lookups line up
+            kjt = KeyedJaggedTensor.empty(
+                device=self._out_values.device, stride=W * self._B
+            )
             kjt._dist_value_splits = (self.in_value_splits, self.out_value_splits)
             return kjt
         if getattr(self, "_vbe_strides_rf", None) is not None:

@@ -188,9 +188,12 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
     def forward(self, sparse_features: KeyedJaggedTensor) -> torch.Tensor:
         B = sparse_features.stride()
         if len(self._emb_modules) == 0:
-            # this rank holds no shards of this sharding — contributes 0 cols
+            # this rank holds no shards of this sharding — contributes 0 cols.
+            # requires_grad anchors the output a2a in the autograd graph so
+            # this rank still issues the BACKWARD collective its peers expect
             return torch.zeros(
-                B, 0, dtype=torch.float32, device=sparse_features.device()
+                B, 0, dtype=torch.float32, device=sparse_features.device(),
+                requires_grad=torch.is_grad_enabled(),
             )
         if sparse_features.variable_stride_per_key():
             # VBE: 1-D packed [sum_f B_f * D_f] per group, feature-major
@@ -279,6 +282,17 @@ class EmbeddingSharding(abc.ABC):
 
     Reference parity: embedding_sharding.py:1183.
     """
+
+    def out_pg(self):
+        """Process group for the OUTPUT (differentiable) collectives.
+
+        When one module mixes sharding types, each sharding's backward
+        collective fires when autograd reaches it — an order that differs
+        across ranks (each rank's graph differs). Giving every sharding its
+        own communicator removes the cross-sharding ordering requirement
+        (set by ShardedEBC/EC via ``_pg_out``); NCCL/RCCL and gloo both only
+        order collectives WITHIN a group."""
+        return getattr(self, "_pg_out", None) or self._pg
 
     @abc.abstractmethod
     def create_input_dist(self, device: torch.device) -> BaseSparseFeaturesDist:

@@ -227,6 +227,15 @@ class ShardedEmbeddingBagCollection(nn.Module):
         self._shardings: List[EmbeddingSharding] = [
             create_sharding(t, infos, env, self._device) for t, infos in by_type.items()
         ]
+        # mixed sharding types: one communicator per sharding so their
+        # backward collectives never need a cross-rank issue order
+        if env.process_group is not None and len(self._shardings) > 1:
+            import torch.distributed as dist_mod
+
+            for sh in self._shardings[1:]:
+                sh._pg_out = dist_mod.new_group(
+                    backend=dist_mod.get_backend(env.process_group)
+                )
 
         self._input_dists = nn.ModuleList(
             [s.create_input_dist(self._device) for s in self._shardings]

@@ -155,4 +155,4 @@ class RwPooledEmbeddingSharding(EmbeddingSharding):
             from torchrec_amd.distributed.sharding.tw_sharding import _NoOpEmbeddingDist
 
             return _NoOpEmbeddingDist()
-        return RwPooledEmbeddingDist(self._pg)
+        return RwPooledEmbeddingDist(self.out_pg())

@@ -131,4 +131,4 @@ class TwPooledEmbeddingSharding(EmbeddingSharding):
             from torchrec_amd.distributed.qcomm_codecs import get_qcomm_codecs
 
             codec, _ = get_qcomm_codecs(qc)
-        return TwPooledEmbeddingDist(self._pg, self._dim_sum_per_rank, codec=codec)
+        return TwPooledEmbeddingDist(self.out_pg(), self._dim_sum_per_rank, codec=codec)

@@ -272,13 +272,14 @@ class KeyedJaggedTensor(Pipelineable):
         values_dtype: Optional[torch.dtype] = None,
         weights_dtype: Optional[torch.dtype] = None,
         lengths_dtype: torch.dtype = torch.int64,
+        stride: int = 0,
     ) -> "KeyedJaggedTensor":
         return KeyedJaggedTensor(
             keys=[],
             values=torch.empty(0, dtype=values_dtype, device=device),
             weights=torch.empty(0, dtype=weights_dtype, device=device) if is_weighted else None,
             lengths=torch.empty(0, dtype=lengths_dtype, device=device),
-            stride=0,
+            stride=stride,
         )
 
     @staticmethod
