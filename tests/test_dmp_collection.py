@@ -76,3 +76,55 @@ def _run_2d_test(rank, world_size):
 
 def test_dmp_collection_2d():
     run_multi_process(_run_2d_test, 4, "gloo")
+
+
+def _run_2d_mixed(rank, world_size):
+    """2D + MIXED sharding types per group: exercises the globally-coordinated
+    per-sharding communicator creation (env.all_group_ranks)."""
+    S = 2
+    B = 4
+    tables = make_tables()
+    mix = [
+        ShardingType.TABLE_WISE.value,
+        ShardingType.ROW_WISE.value,
+        ShardingType.TABLE_WISE.value,
+        ShardingType.ROW_WISE.value,
+    ]
+    torch.manual_seed(42)
+    model = SparseModel(make_tables())
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=S, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(sharding_types=[mix[i]])
+            for i, cfg in enumerate(tables)
+        },
+    )
+    plan = planner.plan(model, [sharder])
+    dmp2d = DMPCollection(
+        model, sharding_group_size=S, plan=plan, sharders=[sharder], sync_interval=1
+    )
+    golden = _golden(tables, None, world_size)
+    dmp2d.load_state_dict(
+        {
+            f"sparse.embedding_bags.{cfg.name}.weight": w
+            for cfg, w in zip(tables, golden.split_embedding_weights())
+        },
+        strict=False,
+    )
+    group = rank // S
+    rank_in_group = rank % S
+    kjt_group = make_global_kjt(tables, B * S, seed=300 + group)
+    kjt_local = kjt_local_slice(kjt_group, rank_in_group * B, (rank_in_group + 1) * B)
+    kt = dmp2d(kjt_local)
+    vals = kt.values()
+    expected = golden(kjt_group).values()[rank_in_group * B : (rank_in_group + 1) * B]
+    torch.testing.assert_close(vals, expected, atol=1e-5, rtol=1e-5)
+    vals.sum().backward()  # mixed backward collectives across 2 groups
+    dmp2d.sync()
+
+
+def test_dmp_collection_2d_mixed_shardings():
+    run_multi_process(_run_2d_mixed, 4, "gloo")

@@ -232,10 +232,19 @@ class ShardedEmbeddingBagCollection(nn.Module):
         if env.process_group is not None and len(self._shardings) > 1:
             import torch.distributed as dist_mod
 
+            # new_group is collective over the DEFAULT group with identical
+            # arguments everywhere: loop over EVERY sharding group's rank
+            # list (2D sets env.all_group_ranks; 1D = just this group)
+            groups = env.all_group_ranks or [
+                dist_mod.get_process_group_ranks(env.process_group)
+            ]
+            my_ranks = dist_mod.get_process_group_ranks(env.process_group)
+            backend = dist_mod.get_backend(env.process_group)
             for sh in self._shardings[1:]:
-                sh._pg_out = dist_mod.new_group(
-                    backend=dist_mod.get_backend(env.process_group)
-                )
+                for ranks in groups:
+                    pg = dist_mod.new_group(ranks=ranks, backend=backend)
+                    if ranks == my_ranks:
+                        sh._pg_out = pg
 
         self._input_dists = nn.ModuleList(
             [s.create_input_dist(self._device) for s in self._shardings]

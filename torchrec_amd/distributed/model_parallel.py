@@ -255,6 +255,9 @@ class DMPCollection(nn.Module):
             if slot == my_replica_slot:
                 self._replica_pg = pg
         env = ShardingEnv(S, rank % S, self._sharding_pg)
+        env.all_group_ranks = [
+            list(range(g * S, (g + 1) * S)) for g in range(self._num_groups)
+        ]
         if plan is None and sharders is None:
             sharders = get_default_sharders()
         if plan is None:

@@ -193,6 +193,10 @@ class ShardingEnv:
         self.world_size = world_size
         self.rank = rank
         self.process_group = pg
+        # 2D: rank lists of EVERY sharding group (set by DMPCollection) so
+        # per-sharding communicator creation can satisfy new_group's
+        # all-ranks-same-arguments collective contract
+        self.all_group_ranks: Optional[List[List[int]]] = None
 
     @classmethod
     def from_process_group(cls, pg: dist.ProcessGroup) -> "ShardingEnv":
