@@ -200,6 +200,35 @@ class ShardedEmbeddingCollection(nn.Module):
                 self._a2a_modules.append(nn.Identity())
                 self._seq_a2a.append(nn.Identity())
 
+        # mixed sharding types: the differentiable output a2as get their own
+        # communicators so backward issue order needs no cross-rank agreement
+        # (same treatment as ShardedEBC; new_group satisfied globally via
+        # env.all_group_ranks under 2D)
+        n_comm = sum(
+            1
+            for st in self._sharding_types
+            if st != ShardingType.DATA_PARALLEL.value and W > 1
+        )
+        if env.process_group is not None and n_comm > 1:
+            import torch.distributed as dist_mod
+
+            groups = getattr(env, "all_group_ranks", None) or [
+                dist_mod.get_process_group_ranks(env.process_group)
+            ]
+            my_ranks = dist_mod.get_process_group_ranks(env.process_group)
+            backend = dist_mod.get_backend(env.process_group)
+            first = True
+            for si, st in enumerate(self._sharding_types):
+                if st == ShardingType.DATA_PARALLEL.value or W <= 1:
+                    continue
+                if first:
+                    first = False  # first sharding keeps the shared pg
+                    continue
+                for ranks in groups:
+                    pg = dist_mod.new_group(ranks=ranks, backend=backend)
+                    if ranks == my_ranks:
+                        self._seq_a2a[si] = SequenceEmbeddingsAllToAll(pg)
+
         self._fused_optimizer = _ECFusedOptimizer(self)
 
     @staticmethod
@@ -220,7 +249,7 @@ class ShardedEmbeddingCollection(nn.Module):
     ) -> nn.Module:
         specs = [(t.name, max(t.local_rows, 1), t.local_dim) for t in tables]
         ftm = [i for i, t in enumerate(tables) for _ in t.feature_names]
-        return TableBatchedEmbeddings(
+        m = TableBatchedEmbeddings(
             specs,
             feature_table_map=ftm,
             optimizer="dense" if dense else self._fused_params.get("optimizer", "rowwise_adagrad"),
@@ -232,6 +261,9 @@ class ShardedEmbeddingCollection(nn.Module):
             ],
             use_index_dedup=self._use_index_dedup,
         )
+        if not tables:
+            m._dim = D  # featureless rank still answers [0, D] for the a2a
+        return m
 
     # -- forward ------------------------------------------------------------
 

@@ -818,7 +818,7 @@ class TableBatchedEmbeddings(nn.Module):
         super().__init__()
         self._use_index_dedup = use_index_dedup
         dims = {s[2] for s in embedding_specs}
-        assert len(dims) == 1, "sequence TBE requires a uniform embedding dim"
+        assert len(dims) <= 1, "sequence TBE requires a uniform embedding dim"
         self._bags = TableBatchedEmbeddingBags(
             embedding_specs,
             feature_table_map,
@@ -831,7 +831,7 @@ class TableBatchedEmbeddings(nn.Module):
             init_max=init_max,
             weights_precision=weights_precision,
         )
-        self._dim = next(iter(dims))
+        self._dim = next(iter(dims)) if dims else 0
 
     @property
     def weights(self) -> torch.Tensor:
@@ -856,6 +856,13 @@ class TableBatchedEmbeddings(nn.Module):
 
     def forward(self, indices: torch.Tensor, offsets: torch.Tensor) -> torch.Tensor:
         host = self._bags
+        if host._num_features == 0:
+            # featureless rank of a sharding: zero rows, but keep the output
+            # in the autograd graph so backward collectives still fire
+            return torch.zeros(
+                0, self._dim, device=indices.device,
+                requires_grad=torch.is_grad_enabled(),
+            )
         B = (offsets.numel() - 1) // host._num_features
         if not indices.is_cuda:
             return _TBESeqCpuFunction.apply(host._dummy, host, indices, offsets, B)
