@@ -178,3 +178,40 @@ class ShardedManagedCollisionEmbeddingBagCollection(nn.Module):
             w = views[name]
             valid = slots[slots < w.shape[0]]
             w[valid] = 0.0
+
+
+class ShardedManagedCollisionEmbeddingCollection(nn.Module):
+    """Sharded MC + sequence EC (reference distributed/mc_embedding.py
+    ShardedManagedCollisionEmbeddingCollection): remap raw ids to the bounded
+    slot space, run the sharded sequence lookup, reset evicted rows."""
+
+    def __init__(
+        self,
+        sharded_mcc: ShardedManagedCollisionCollection,
+        sharded_ec: nn.Module,
+    ) -> None:
+        super().__init__()
+        self._mcc = sharded_mcc
+        self._ec = sharded_ec
+
+    def forward(self, features: KeyedJaggedTensor):
+        remapped = self._mcc(features)
+        out = self._ec(remapped)
+        self._reset_evicted()
+        return out, remapped
+
+    @torch.no_grad()
+    def _reset_evicted(self) -> None:
+        evictions = self._mcc.evict()
+        tbes = self._ec.tbes() if hasattr(self._ec, "tbes") else []
+        views = {}
+        for tbe in tbes:
+            inner = getattr(tbe, "_bags", tbe)
+            for spec, w in zip(inner.embedding_specs, inner.split_embedding_weights()):
+                views[spec.name] = w
+        for name, slots in evictions.items():
+            if slots is None or slots.numel() == 0 or name not in views:
+                continue
+            w = views[name]
+            valid = slots[slots < w.shape[0]]
+            w[valid] = 0.0
