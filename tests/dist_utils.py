@@ -41,11 +41,26 @@ def _entry(rank: int, world_size: int, port: int, backend: str, fn, args, kwargs
 def run_multi_process(
     fn: Callable, world_size: int = 2, backend: str = "gloo", *args: Any, **kwargs: Any
 ) -> None:
-    port = _free_port()
-    mp.start_processes(
-        _entry,
-        args=(world_size, port, backend, fn, args, kwargs),
-        nprocs=world_size,
-        start_method="spawn",  # HIP requires spawn (reference multi_process.py:147)
-        join=True,
-    )
+    last: Exception = RuntimeError("unreachable")
+    for attempt in range(2):
+        port = _free_port()
+        try:
+            mp.start_processes(
+                _entry,
+                args=(world_size, port, backend, fn, args, kwargs),
+                nprocs=world_size,
+                start_method="spawn",  # HIP requires spawn (reference multi_process.py:147)
+                join=True,
+            )
+            return
+        except Exception as e:  # retry ONLY rendezvous port races
+            msg = str(e)
+            rendezvous_race = any(
+                t in msg
+                for t in ("Address already in use", "EADDRINUSE", "Connection refused")
+            )
+            if attempt == 0 and rendezvous_race:
+                last = e
+                continue
+            raise
+    raise last
