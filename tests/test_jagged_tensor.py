@@ -225,3 +225,19 @@ class TestOpSurface:
         torch.cumsum(out_sizes, 0, out=out_off[1:])
         pp = ops.expand_into_jagged_permute(perm, in_off, out_off, 6)
         assert pp.tolist() == [3, 4, 5, 0, 1, 2]
+
+
+def test_vbe_to_dict_unequal_strides():
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    kjt = KeyedJaggedTensor(
+        keys=["a", "b"], values=torch.arange(7),
+        lengths=torch.tensor([1, 2, 1, 2, 1]),
+        stride_per_key_per_rank=[[2, 1], [1, 1]],
+    )
+    d = kjt.to_dict()
+    assert d["a"].values().tolist() == [0, 1, 2, 3]
+    assert d["a"].lengths().tolist() == [1, 2, 1]
+    assert d["b"].values().tolist() == [4, 5, 6]
+    assert d["b"].lengths().tolist() == [2, 1]
+    assert kjt.length_per_key() == [4, 3]

@@ -365,6 +365,13 @@ class KeyedJaggedTensor(Pipelineable):
         if self._length_per_key is None:
             if len(self._keys) == 0:
                 self._length_per_key = []
+            elif self._variable_stride_per_key:
+                # per-key lengths segments have per-key sizes (VBE)
+                key_strides = [sum(sp) for sp in self.stride_per_key_per_rank()]
+                self._length_per_key = [
+                    int(seg.sum())
+                    for seg in self.lengths().split(key_strides)
+                ]
             else:
                 lpk = self.lengths().view(len(self._keys), -1).sum(dim=1)
                 self._length_per_key = lpk.cpu().tolist()
@@ -508,12 +515,18 @@ class KeyedJaggedTensor(Pipelineable):
 
     def __getitem__(self, key: str) -> JaggedTensor:
         i = self._key_indices()[key]
-        B = self._stride
         opk = self.offset_per_key()
+        if self._variable_stride_per_key:
+            key_strides = [sum(sp) for sp in self.stride_per_key_per_rank()]
+            lo = sum(key_strides[:i])
+            lengths_i = self.lengths()[lo : lo + key_strides[i]]
+        else:
+            B = self._stride
+            lengths_i = self.lengths()[i * B : (i + 1) * B]
         return JaggedTensor(
             values=self._values[opk[i] : opk[i + 1]],
             weights=self._weights[opk[i] : opk[i + 1]] if self._weights is not None else None,
-            lengths=self.lengths()[i * B : (i + 1) * B],
+            lengths=lengths_i,
         )
 
     def to_dict(self) -> Dict[str, JaggedTensor]:
