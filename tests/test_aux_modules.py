@@ -312,3 +312,21 @@ def test_maybe_td_passthrough():
         keys=["f"], values=torch.tensor([1, 2]), lengths=torch.tensor([2]), stride=1
     )
     assert maybe_td_to_kjt(kjt) is kjt  # tensordict absent or not a TD
+
+
+def test_planner_stats_table():
+    import bench
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+
+    model = bench.build_model(1e-4)
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=2, compute_device="cuda", batch_size=128)
+    )
+    sharder = EmbeddingBagCollectionSharder()
+    planner.plan(model, [sharder])
+    table = planner.last_stats
+    assert "sharding" in table and "kernel" in table
+    assert "per-device HBM" in table
+    assert table.count("\n") >= 26  # one row per table

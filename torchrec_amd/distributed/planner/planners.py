@@ -471,6 +471,9 @@ class EmbeddingShardingPlanner(ShardingPlanner):
             raise PlannerError(
                 f"no feasible sharding plan found; partition errors: {errors[:3]}"
             )
+        from torchrec_amd.distributed.planner.stats import EmbeddingStats
+
+        self.last_stats = EmbeddingStats().log(best, self._topology)
         return self._to_sharding_plan(best)
 
     def collective_plan(
