@@ -330,3 +330,40 @@ def test_planner_stats_table():
     assert "sharding" in table and "kernel" in table
     assert "per-device HBM" in table
     assert table.count("\n") >= 26  # one row per table
+
+
+def test_dp_and_gridsearch_proposers():
+    from torchrec_amd.distributed.planner.planners import (
+        DynamicProgrammingProposer,
+        EmbeddingShardingPlanner,
+        GridSearchProposer,
+    )
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.sparse = EmbeddingBagCollection(
+                tables=[
+                    EmbeddingBagConfig(
+                        num_embeddings=100 * (i + 1), embedding_dim=8,
+                        name=f"t{i}", feature_names=[f"f{i}"],
+                    )
+                    for i in range(3)
+                ]
+            )
+
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=2, compute_device="cpu", hbm_cap=1 << 40)
+    )
+    sharder = EmbeddingBagCollectionSharder()
+    options = planner._enumerator.enumerate(M(), [sharder])
+    gs = GridSearchProposer().propose(options)
+    assert gs and all(len(p) == 3 for p in gs)
+    dp = DynamicProgrammingProposer().propose(options)
+    assert len(dp) == 1 and len(dp[0]) == 3
+    names = sorted(o.name for o in dp[0])
+    assert names == ["t0", "t1", "t2"]

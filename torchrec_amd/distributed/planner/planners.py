@@ -382,6 +382,85 @@ class UniformProposer(GreedyProposer):
         return proposals
 
 
+class GridSearchProposer(GreedyProposer):
+    """Exhaustive cross product of per-table options — only viable for small
+    models (reference proposers.py:207 caps the same way)."""
+
+    MAX_PROPOSALS = 256
+
+    def propose(self, options: List[ShardingOption]) -> List[List[ShardingOption]]:
+        import itertools
+
+        by_table: Dict[Tuple[str, str], List[ShardingOption]] = {}
+        for o in options:
+            by_table.setdefault((o.module_fqn, o.name), []).append(o)
+        total = 1
+        for opts in by_table.values():
+            total *= len(opts)
+            if total > self.MAX_PROPOSALS:
+                return []  # defer to the other proposers
+        return [
+            [self._clone(o) for o in combo]
+            for combo in itertools.product(*by_table.values())
+        ]
+
+
+class DynamicProgrammingProposer(GreedyProposer):
+    """HBM-budgeted DP over tables (reference proposers.py:287): tables are
+    stages, HBM is discretized into bins, and dp[t][h] holds the min total
+    estimated perf using <= h HBM for the first t tables. Finds plans the
+    per-table greedy misses when the best-perf option of a big table starves
+    smaller ones out of HBM."""
+
+    BINS = 64
+
+    def propose(self, options: List[ShardingOption]) -> List[List[ShardingOption]]:
+        by_table: Dict[Tuple[str, str], List[ShardingOption]] = {}
+        for o in options:
+            by_table.setdefault((o.module_fqn, o.name), []).append(o)
+        if not by_table:
+            return []
+        tables = list(by_table.values())
+        total_hbm = sum(
+            max(sum(s.storage.hbm for s in o.shards) for o in opts)
+            for opts in tables
+        )
+        if total_hbm <= 0:
+            return []
+        bin_sz = max(1, total_hbm // self.BINS)
+        B = self.BINS + 1
+        INF = float("inf")
+        dp = [[INF] * B for _ in range(len(tables) + 1)]
+        choice: List[List[Optional[int]]] = [[None] * B for _ in range(len(tables))]
+        for h in range(B):
+            dp[0][h] = 0.0
+        for t, opts in enumerate(tables):
+            for h in range(B):
+                if dp[t][h] == INF:
+                    continue
+                for oi, o in enumerate(opts):
+                    hbm = sum(s.storage.hbm for s in o.shards)
+                    nh = min(B - 1, h + max(1, int(hbm // bin_sz)))
+                    cost = dp[t][h] + o.total_perf
+                    if cost < dp[t + 1][nh]:
+                        dp[t + 1][nh] = cost
+                        choice[t][nh] = (h, oi)
+        # backtrack the best terminal bin
+        best_h = min(range(B), key=lambda h: dp[len(tables)][h])
+        if dp[len(tables)][best_h] == INF:
+            return []
+        prop: List[ShardingOption] = []
+        h = best_h
+        for t in range(len(tables) - 1, -1, -1):
+            back = choice[t][h]
+            if back is None:
+                return []
+            h, oi = back
+            prop.append(self._clone(tables[t][oi]))
+        prop.reverse()
+        return [prop]
+
+
 class UVMSpillProposer(GreedyProposer):
     """Best DEVICE-kernel plan, spilling the k largest tables to host DRAM
     (FUSED_UVM) — covers models beyond 288 GB HBM (reference
@@ -440,7 +519,13 @@ class EmbeddingShardingPlanner(ShardingPlanner):
         self._topology = topology
         self._enumerator = EmbeddingEnumerator(topology, constraints)
         self._partitioner = GreedyPerfPartitioner()
-        self._proposers = [GreedyProposer(), UniformProposer(), UVMSpillProposer()]
+        self._proposers = [
+            GreedyProposer(),
+            UniformProposer(),
+            GridSearchProposer(),
+            DynamicProgrammingProposer(),
+            UVMSpillProposer(),
+        ]
 
     def plan(
         self, module: nn.Module, sharders: List[ModuleSharder[nn.Module]]
