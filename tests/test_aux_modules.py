@@ -367,3 +367,43 @@ def test_dp_and_gridsearch_proposers():
     assert len(dp) == 1 and len(dp[0]) == 3
     names = sorted(o.name for o in dp[0])
     assert names == ["t0", "t1", "t2"]
+
+
+def test_memory_balanced_partitioner():
+    from torchrec_amd.distributed.planner.planners import (
+        EmbeddingShardingPlanner,
+        GreedyProposer,
+        MemoryBalancedPartitioner,
+    )
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.sparse = EmbeddingBagCollection(
+                tables=[
+                    EmbeddingBagConfig(
+                        num_embeddings=1000 * (i + 1), embedding_dim=8,
+                        name=f"t{i}", feature_names=[f"f{i}"],
+                    )
+                    for i in range(4)
+                ]
+            )
+
+    topo = Topology(world_size=2, compute_device="cpu", hbm_cap=1 << 40)
+    planner = EmbeddingShardingPlanner(topology=topo)
+    planner._partitioner = MemoryBalancedPartitioner()
+    sharder = EmbeddingBagCollectionSharder()
+    plan = planner.plan(M(), [sharder])
+    mplan = next(iter(plan.plan.values()))
+    ranks = [ps.ranks for _n, ps in mplan.items()]
+    used = [0, 0]
+    for n, ps in mplan.items():
+        if ps.sharding_type == "table_wise":
+            used[ps.ranks[0]] += 1
+    # storage-balanced: both devices hold shards
+    flat = [r for rs in ranks for r in rs]
+    assert set(flat) == {0, 1}

@@ -322,6 +322,43 @@ class GreedyPerfPartitioner:
         return proposal
 
 
+class MemoryBalancedPartitioner(GreedyPerfPartitioner):
+    """Balance HBM instead of perf for the movable (TW/CW) shards
+    (reference partitioners.py:694): picks the feasible device with the most
+    free HBM. Used when capacity, not step time, is the binding constraint."""
+
+    def partition(self, proposal, topology):
+        # fixed-rank and node-local placements follow the perf partitioner
+        placed = super().partition(proposal, topology)
+        # rebalance movable shards by storage: biggest first onto emptiest
+        devices = {d.rank: [topology.devices[d.rank].storage.hbm, 0.0] for d in topology.devices}
+        for opt in placed:
+            for shard in opt.shards:
+                if opt.sharding_type in (
+                    ShardingType.TABLE_WISE.value,
+                    ShardingType.COLUMN_WISE.value,
+                ):
+                    continue
+                devices[shard.rank][1] += shard.storage.hbm if shard.storage else 0
+        movable = [
+            (shard, opt)
+            for opt in placed
+            if opt.sharding_type
+            in (ShardingType.TABLE_WISE.value, ShardingType.COLUMN_WISE.value)
+            for shard in opt.shards
+        ]
+        movable.sort(key=lambda x: x[0].storage.hbm if x[0].storage else 0, reverse=True)
+        for shard, opt in movable:
+            need = shard.storage.hbm if shard.storage else 0
+            feasible = [r for r, (cap, used) in devices.items() if cap - used >= need]
+            if not feasible:
+                raise PlannerError(f"no device fits shard of {opt.name}")
+            r = max(feasible, key=lambda r: devices[r][0] - devices[r][1])
+            shard.rank = r
+            devices[r][1] += need
+        return placed
+
+
 class GreedyProposer:
     """Per-table best-perf choice, tables in size-desc order
     (reference proposers.py:34)."""
