@@ -407,3 +407,26 @@ def test_memory_balanced_partitioner():
     # storage-balanced: both devices hold shards
     flat = [r for rs in ranks for r in rs]
     assert set(flat) == {0, 1}
+
+
+def test_storage_reservations():
+    from torchrec_amd.distributed.planner.storage_reservations import (
+        FixedPercentageStorageReservation,
+        HeuristicalStorageReservation,
+        InferenceStorageReservation,
+    )
+    from torchrec_amd.distributed.planner.types import Topology
+
+    t = Topology(world_size=2, compute_device="cuda", hbm_cap=100_000_000)
+    FixedPercentageStorageReservation(0.2).reserve(t)
+    assert all(d.storage.hbm == 80_000_000 for d in t.devices)
+
+    lin = torch.nn.Linear(100, 100)
+    t2 = Topology(world_size=1, compute_device="cuda", hbm_cap=100_000_000)
+    HeuristicalStorageReservation(0.1, kjt_bytes_per_sample=0).reserve(t2, lin, 0)
+    dense = sum(p.numel() * p.element_size() for p in lin.parameters())
+    assert t2.devices[0].storage.hbm == int(100_000_000 * 0.9) - dense * 4
+
+    t3 = Topology(world_size=1, compute_device="cuda", hbm_cap=100_000_000)
+    InferenceStorageReservation(0.0).reserve(t3, lin)
+    assert t3.devices[0].storage.hbm == 100_000_000 - dense

@@ -545,7 +545,9 @@ class EmbeddingShardingPlanner(ShardingPlanner):
         topology: Optional[Topology] = None,
         constraints: Optional[Dict[str, ParameterConstraints]] = None,
         batch_size: Optional[int] = None,
+        storage_reservation=None,
     ) -> None:
+        self._storage_reservation = storage_reservation
         if topology is None:
             topology = Topology(
                 world_size=dist.get_world_size() if dist.is_initialized() else 1,
@@ -567,6 +569,10 @@ class EmbeddingShardingPlanner(ShardingPlanner):
     def plan(
         self, module: nn.Module, sharders: List[ModuleSharder[nn.Module]]
     ) -> ShardingPlan:
+        if self._storage_reservation is not None:
+            self._topology = self._storage_reservation.reserve(
+                self._topology, module, self._topology.batch_size
+            )
         options = self._enumerator.enumerate(module, sharders)
         if not options:
             return ShardingPlan({})
