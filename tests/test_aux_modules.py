@@ -430,3 +430,23 @@ def test_storage_reservations():
     t3 = Topology(world_size=1, compute_device="cuda", hbm_cap=100_000_000)
     InferenceStorageReservation(0.0).reserve(t3, lin)
     assert t3.devices[0].storage.hbm == 100_000_000 - dense
+
+
+def test_fx_trace_dlrm_sparse_arch():
+    """fx tracer treats KJT-consuming modules as leaves (reference fx/tracer)."""
+    from torchrec_amd.fx.tracer import Tracer, symbolic_trace
+    from torchrec_amd.models.dlrm import DLRM
+    from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+
+    ebc = EmbeddingBagCollection(
+        tables=[
+            EmbeddingBagConfig(num_embeddings=10, embedding_dim=8, name="t0", feature_names=["f0"]),
+        ]
+    )
+    model = DLRM(
+        embedding_bag_collection=ebc, dense_in_features=4,
+        dense_arch_layer_sizes=[8, 8], over_arch_layer_sizes=[8, 1],
+    )
+    gm = symbolic_trace(model)
+    assert any("sparse" in n.name or "ebc" in str(n.target) for n in gm.graph.nodes)

@@ -16,9 +16,16 @@ from torchrec_amd.distributed.types import LazyAwaitable
 
 
 class Tracer(torch.fx.Tracer):
+    DEFAULT_LEAVES = [
+        # branch on tensor properties (device/dtype) -> must stay opaque
+        "InteractionArch",
+        "InteractionDCNArch",
+        "SparseArch",
+    ]
+
     def __init__(self, leaf_modules: Optional[list] = None) -> None:
         super().__init__()
-        self._leaf_modules = leaf_modules or []
+        self._leaf_modules = list(self.DEFAULT_LEAVES) + (leaf_modules or [])
 
     def is_leaf_module(self, m: torch.nn.Module, module_qualified_name: str) -> bool:
         if type(m).__name__ in self._leaf_modules:
