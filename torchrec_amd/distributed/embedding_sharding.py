@@ -144,7 +144,10 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
 
                 kv = KeyValueEmbeddingBags(
                     specs,
-                    capacity=fused_params.get("kv_capacity", 1 << 20),
+                    capacity=(
+                        getattr(fused_params.get("kv_params"), "capacity", None)
+                        or fused_params.get("kv_capacity", 1 << 20)
+                    ),
                     feature_table_map=feature_table_map,
                     pooling_mode=_POOL_TO_MODE[pool],
                     optimizer=optimizer,
@@ -174,7 +177,10 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                 eps=fused_params.get("eps", 1.0e-8),
                 device=device,
                 location=location,
-                cache_load_factor=fused_params.get("cache_load_factor", 0.2),
+                cache_load_factor=(
+                    getattr(fused_params.get("cache_params"), "load_factor", None)
+                    or fused_params.get("cache_load_factor", 0.2)
+                ),
                 weights_precision={"FP32": "fp32", "FP16": "fp16", "BF16": "bf16"}[
                     group[0].data_type
                 ],

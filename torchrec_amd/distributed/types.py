@@ -46,6 +46,30 @@ class EmbeddingComputeKernel(Enum):
     QUANT = "quant"
 
 
+@dataclass
+class CacheParams:
+    """UVM-cache tuning knobs riding fused_params (reference types.py:643).
+
+    ``load_factor`` sizes the HBM lxu cache relative to the table;
+    ``reserved_memory`` is subtracted from the planner's HBM budget;
+    ``precision`` is the cache line dtype (fp32 in v1)."""
+
+    algorithm: str = "lru"
+    load_factor: Optional[float] = None
+    reserved_memory: Optional[float] = None
+    precision: Optional[str] = None
+    prefetch_pipeline: Optional[bool] = None
+
+
+@dataclass
+class KeyValueParams:
+    """KEY_VALUE (virtual table) kernel knobs (reference types.py:685)."""
+
+    capacity: Optional[int] = None
+    eviction: str = "lfu_lru"
+    bulk_init_chunk_size: Optional[int] = None
+
+
 class Awaitable(abc.ABC, Generic[W]):
     """Handle for an async (collective) result (reference types.py:367)."""
 
