@@ -231,3 +231,43 @@ def _run_grad_accum(rank, world_size):
 
 def test_gradient_accumulation_pipeline():
     run_multi_process(_run_grad_accum, 2, "gloo")
+
+
+def test_micro_batch_pipeline_cpu():
+    from torchrec_amd.datasets.random import generate_batch
+    from torchrec_amd.distributed.train_pipeline import MicroBatchPipeline
+    from torchrec_amd.models.dlrm import DLRM
+    from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+
+    torch.manual_seed(0)
+    keys = ["f0", "f1"]
+    rows = [30, 40]
+    model = DLRM(
+        embedding_bag_collection=EmbeddingBagCollection(
+            tables=[
+                EmbeddingBagConfig(num_embeddings=r, embedding_dim=8, name=f"t{i}", feature_names=[keys[i]])
+                for i, r in enumerate(rows)
+            ]
+        ),
+        dense_in_features=4,
+        dense_arch_layer_sizes=[8, 8],
+        over_arch_layer_sizes=[8, 1],
+    )
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.m = model
+
+        def forward(self, batch):
+            return self.m(batch.dense_features, batch.sparse_features)
+
+    batch = generate_batch(keys, 8, rows, ids_per_feature=2, num_dense=4,
+                           generator=torch.Generator().manual_seed(1))
+    wrapped = M()
+    pipe = MicroBatchPipeline(wrapped, torch.device("cpu"), num_micro=3)
+    out = pipe.progress(batch)
+    with torch.no_grad():
+        full = wrapped(batch)
+    torch.testing.assert_close(out, full, atol=1e-5, rtol=1e-5)

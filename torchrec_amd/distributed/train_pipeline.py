@@ -599,3 +599,74 @@ class TrainPipelineSemiSync(TrainPipelineSparseDist):
         self._dist_contexts = [d_ip2]
         self._emb_awaitables = [e_next]
         return output
+
+
+class MicroBatchPipeline:
+    """Micro-batch serving/eval pipeline (reference maglev/pipeline.py:20):
+    splits each incoming batch into ``num_micro`` slices and runs them
+    back-to-back, overlapping each slice's H2D copy (memcpy stream) with the
+    previous slice's compute — bounds serving latency without giving up
+    copy/compute overlap."""
+
+    def __init__(self, model, device: torch.device, num_micro: int = 2) -> None:
+        self._model = model
+        self._device = device
+        self._n = max(1, num_micro)
+        self._memcpy_stream = (
+            torch.cuda.Stream(priority=-1) if device.type == "cuda" else None
+        )
+
+    def _split(self, batch):
+        if self._n == 1 or not hasattr(batch, "dense_features"):
+            return [batch]
+        B = batch.dense_features.shape[0]
+        step = (B + self._n - 1) // self._n
+        out = []
+        for lo in range(0, B, step):
+            hi = min(B, lo + step)
+            kjt = batch.sparse_features
+            K = len(kjt.keys())
+            lengths2d = kjt.lengths().view(K, B)
+            offs = torch.zeros(K * B + 1, dtype=torch.int64)
+            torch.cumsum(kjt.lengths(), 0, out=offs[1:])
+            vals, lens = [], []
+            for k in range(K):
+                l0, h0 = int(offs[k * B + lo]), int(offs[k * B + hi])
+                vals.append(kjt.values()[l0:h0])
+                lens.append(lengths2d[k, lo:hi])
+            out.append(
+                type(batch)(
+                    dense_features=batch.dense_features[lo:hi],
+                    sparse_features=KeyedJaggedTensor(
+                        keys=kjt.keys(),
+                        values=torch.cat(vals),
+                        lengths=torch.cat(lens),
+                        stride=hi - lo,
+                    ),
+                    labels=batch.labels[lo:hi],
+                )
+            )
+        return out
+
+    @torch.no_grad()
+    def progress(self, batch):
+        micro = self._split(batch)
+        outs = []
+        staged = None
+        for i, m in enumerate(micro):
+            if self._memcpy_stream is not None:
+                with torch.cuda.stream(self._memcpy_stream):
+                    nxt = m.to(self._device, non_blocking=True)
+            else:
+                nxt = m.to(self._device) if self._device.type != "cpu" else m
+            if staged is not None:
+                outs.append(self._model(staged))
+            if self._memcpy_stream is not None:
+                torch.cuda.current_stream().wait_stream(self._memcpy_stream)
+                nxt.record_stream(torch.cuda.current_stream())
+            staged = nxt
+        outs.append(self._model(staged))
+        first = outs[0]
+        if isinstance(first, torch.Tensor):
+            return torch.cat(outs)
+        return outs
