@@ -291,3 +291,71 @@ def test_fused_sparse_dist_pipeline_cuda():
     losses = [float(pipe.progress(it)[0]) for _ in range(6)]
     torch.cuda.synchronize()
     assert all(l == l for l in losses)  # no NaNs; stream discipline held
+
+
+def test_dlrm_dcn_and_two_tower_cuda():
+    """Model-family smoke on cuda:0: DLRM-DCN (low-rank crossnet interaction)
+    and two-tower retrieval both step through fwd+bwd."""
+    from torchrec_amd.models.dlrm import DLRM_DCN
+    from torchrec_amd.models.two_tower import TwoTower
+    from torchrec_amd.modules.embedding_configs import EmbeddingBagConfig
+    from torchrec_amd.modules.embedding_modules import EmbeddingBagCollection
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    device = torch.device("cuda:0")
+    torch.manual_seed(0)
+    keys = ["f0", "f1"]
+    rows = [50, 70]
+    ebc = EmbeddingBagCollection(
+        tables=[
+            EmbeddingBagConfig(num_embeddings=r, embedding_dim=16, name=f"t{i}",
+                               feature_names=[keys[i]])
+            for i, r in enumerate(rows)
+        ]
+    ).to(device)
+    dcn = DLRM_DCN(
+        embedding_bag_collection=ebc,
+        dense_in_features=4,
+        dense_arch_layer_sizes=[16, 16],
+        over_arch_layer_sizes=[16, 1],
+        dcn_num_layers=2,
+        dcn_low_rank_dim=8,
+    ).to(device)
+    B = 16
+    g = torch.Generator().manual_seed(1)
+    kjt = KeyedJaggedTensor(
+        keys=keys,
+        values=torch.cat([torch.randint(0, r, (B,), generator=g) for r in rows]),
+        lengths=torch.ones(2 * B, dtype=torch.int64),
+        stride=B,
+    ).to(device)
+    dense = torch.rand(B, 4, device=device)
+    out = dcn(dense, kjt)
+    assert out.shape == (B, 1)
+    out.sum().backward()
+
+    tt_ebc = EmbeddingBagCollection(
+        tables=[
+            EmbeddingBagConfig(num_embeddings=40, embedding_dim=16, name="q", feature_names=["fq"]),
+            EmbeddingBagConfig(num_embeddings=60, embedding_dim=16, name="c", feature_names=["fc"]),
+        ]
+    )
+    tt = TwoTower(
+        embedding_bag_collection=tt_ebc,
+        query_features=["fq"],
+        candidate_features=["fc"],
+        layer_sizes=[16, 8],
+    ).to(device)
+    kjt2 = KeyedJaggedTensor(
+        keys=["fq", "fc"],
+        values=torch.cat([
+            torch.randint(0, 40, (B,), generator=g),
+            torch.randint(0, 60, (B,), generator=g),
+        ]),
+        lengths=torch.ones(2 * B, dtype=torch.int64),
+        stride=B,
+    ).to(device)
+    q, c = tt(kjt2)
+    assert q.shape == (B, 8) and c.shape == (B, 8)
+    (q * c).sum().backward()
+    torch.cuda.synchronize()

@@ -156,7 +156,6 @@ class ShardedPECEmbeddingCollection(ShardedEmbeddingCollection):
             prio_out, defer_out = [], []
             prio_parts, defer_parts = [], []
             prio_pos_parts, defer_pos_parts = [], []
-            dest_pos = torch.empty_like(positions)
             # positions within each destination's block (what the receiver
             # scatters by)
             for r in range(W):
