@@ -64,20 +64,23 @@ class EmbeddingCollectionAwaitable(LazyAwaitable[Dict[str, JaggedTensor]]):
         ctx: EmbeddingCollectionContext,
         features_per_sharding: List[List[str]],
         need_indices: bool = False,
+        cw_pos_meta: Optional[List[Optional[List[tuple]]]] = None,
     ) -> None:
         super().__init__()
         self._awaitables = awaitables
         self._ctx = ctx
         self._features_per_sharding = features_per_sharding
         self._need_indices = need_indices
+        self._cw_pos_meta = cw_pos_meta or [None] * len(awaitables)
 
     def _wait_impl(self) -> Dict[str, JaggedTensor]:
         out: Dict[str, JaggedTensor] = {}
-        for aw, feats, local_kjt, unbucketize in zip(
+        for aw, feats, local_kjt, unbucketize, cw_meta in zip(
             self._awaitables,
             self._features_per_sharding,
             self._ctx.local_features,
             self._ctx.unbucketize_permutes,
+            self._cw_pos_meta,
         ):
             rows = aw.wait()
             if unbucketize is not None:
@@ -86,6 +89,27 @@ class EmbeddingCollectionAwaitable(LazyAwaitable[Dict[str, JaggedTensor]]):
             lengths = local_kjt.lengths()
             opk = local_kjt.offset_per_key()
             B = local_kjt.stride()
+            if cw_meta is not None:
+                # CW: positions are (feature, shard_idx) duplicates; collect
+                # each feature's [N, D/k] slices and concat column-wise
+                segs: Dict[str, Dict[int, torch.Tensor]] = {}
+                first_pos: Dict[str, int] = {}
+                for i, (f, j) in enumerate(cw_meta):
+                    segs.setdefault(f, {})[j] = rows[opk[i] : opk[i + 1]]
+                    first_pos.setdefault(f, i)
+                for f, by_shard in segs.items():
+                    i0 = first_pos[f]
+                    vals = torch.cat(
+                        [by_shard[j] for j in sorted(by_shard)], dim=1
+                    )
+                    out[f] = JaggedTensor(
+                        values=vals,
+                        lengths=lengths[i0 * B : (i0 + 1) * B],
+                        weights=local_kjt.values()[opk[i0] : opk[i0 + 1]]
+                        if self._need_indices
+                        else None,
+                    )
+                continue
             for i, f in enumerate(feats):
                 out[f] = JaggedTensor(
                     values=rows[opk[i] : opk[i + 1]],
@@ -141,12 +165,14 @@ class ShardedEmbeddingCollection(nn.Module):
         self._a2a_modules = nn.ModuleList()
         self._seq_a2a = nn.ModuleList()
         self._dims_per_sharding: List[int] = []
+        self._cw_pos_meta: List[Optional[List[tuple]]] = []
 
         for st, cfgs in by_type.items():
             dims = {c.embedding_dim for c in cfgs}
             assert len(dims) == 1, "sequence sharding requires uniform dim per group"
             D = next(iter(dims))
             self._dims_per_sharding.append(D)
+            self._cw_pos_meta.append(None)
             feats = [f for c in cfgs for f in c.feature_names]
             self._features_per_sharding.append(feats)
             self._emb_names_per_sharding.append(
@@ -167,6 +193,66 @@ class ShardedEmbeddingCollection(nn.Module):
                 ]
                 local = tables_per_rank[rank]
                 self._lookups.append(self._make_lookup(local, D))
+                self._input_splits_per_sharding.append(
+                    [len([f for t in tables_per_rank[r] for f in t.feature_names]) for r in range(W)]
+                )
+                self._block_sizes.append(None)
+            elif st == ShardingType.COLUMN_WISE.value:
+                # sequence CW (reference sharding/cw_sequence_sharding.py):
+                # each column shard is a TW-placed virtual table sharing the
+                # feature name; ids fan out to every shard rank and the
+                # returned [N, D/k] slices re-concatenate column-wise
+                from torchrec_amd.distributed.sharding.cw_sharding import (
+                    cw_shard_dims,
+                )
+
+                tables_per_rank = [[] for _ in range(W)]
+                pos_meta_per_rank: List[List[tuple]] = [[] for _ in range(W)]
+                for c in cfgs:
+                    ps = self._ps[c.name]
+                    ranks = ps.ranks or [0]
+                    dims = cw_shard_dims(c.embedding_dim, len(ranks))
+                    assert len(set(dims)) == 1, (
+                        "sequence CW v1 needs equal column shards "
+                        "(choose min_partition dividing dim)"
+                    )
+                    off = 0
+                    for j, (r, dj) in enumerate(zip(ranks, dims)):
+                        tables_per_rank[r].append(
+                            ShardedTableLocal(
+                                name=c.name,
+                                local_rows=c.num_embeddings,
+                                local_dim=dj,
+                                pooling=PoolingType.NONE,
+                                kernel="fused",
+                                feature_names=list(c.feature_names),
+                                col_offset=off,
+                                full_dim=c.embedding_dim,
+                                full_rows=c.num_embeddings,
+                            )
+                        )
+                        for f in c.feature_names:
+                            pos_meta_per_rank[r].append((f, j))
+                        off += dj
+                feats_cw = [
+                    f for r in range(W) for t in tables_per_rank[r] for f in t.feature_names
+                ]
+                self._features_per_sharding[-1] = feats_cw
+                self._emb_names_per_sharding[-1] = [
+                    self._emb_name_per_feature[f] for f in feats_cw
+                ]
+                self._cw_pos_meta[-1] = [m for r in range(W) for m in pos_meta_per_rank[r]]
+                local = tables_per_rank[rank]
+                D_shard = local[0].local_dim if local else (
+                    cw_shard_dims(
+                        cfgs[0].embedding_dim,
+                        len(self._ps[cfgs[0].name].ranks or [0]),
+                    )[0]
+                )
+                assert all(t.local_dim == D_shard for t in local), (
+                    "sequence CW v1: uniform shard dim per rank"
+                )
+                self._lookups.append(self._make_lookup(local, D_shard))
                 self._input_splits_per_sharding.append(
                     [len([f for t in tables_per_rank[r] for f in t.feature_names]) for r in range(W)]
                 )
@@ -345,7 +431,8 @@ class ShardedEmbeddingCollection(nn.Module):
                 self._seq_a2a[si](rows_rank_major, out_splits, in_splits)
             )
         return EmbeddingCollectionAwaitable(
-            awaitables, ctx, self._emb_names_per_sharding, self._need_indices
+            awaitables, ctx, self._emb_names_per_sharding, self._need_indices,
+            cw_pos_meta=self._cw_pos_meta,
         )
 
     def compute(self, ctx, dist_input):
@@ -419,4 +506,5 @@ class EmbeddingCollectionSharder(ModuleSharder[EmbeddingCollection]):
             ShardingType.DATA_PARALLEL.value,
             ShardingType.TABLE_WISE.value,
             ShardingType.ROW_WISE.value,
+            ShardingType.COLUMN_WISE.value,
         ]
