@@ -97,12 +97,26 @@ def main(reduction: int, B: int) -> None:
         out.sum().backward()
 
     t_fused = time_fn(fused_step)
+    # hipGraph-captured fused step: the small-batch fused step is
+    # launch-bound (~25 kernels); one graph replay removes that
+    t_graph = None
+    try:
+        fused_step()
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            fused_step()
+        t_graph = time_fn(lambda: g.replay())
+    except Exception as exc:
+        print(f"# graph capture skipped: {exc}", file=sys.stderr)
     print(json.dumps({
         "bench": "ebc_vs_fused", "reduction": reduction, "B": B,
         "tables": len(rows), "dim": EMB_DIM,
         "naive_ms": round(t_naive * 1e3, 3),
         "fused_ms": round(t_fused * 1e3, 3),
         "speedup": round(t_naive / t_fused, 1),
+        "fused_hipgraph_ms": round(t_graph * 1e3, 3) if t_graph else None,
+        "speedup_hipgraph": round(t_naive / t_graph, 1) if t_graph else None,
         "reference_bar_8xV100": {128: "13x", 64: "18x", 32: "23x"}.get(reduction),
     }))
 
