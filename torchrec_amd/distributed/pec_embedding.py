@@ -178,13 +178,15 @@ class ShardedPECEmbeddingCollection(ShardedEmbeddingCollection):
             split_payload = torch.tensor(
                 [v for pair in zip(prio_out, defer_out) for v in pair],
                 dtype=torch.int64,
+                device=rows.device,  # RCCL needs device tensors
             )
             recv_payload = torch.empty_like(split_payload)
             dist.all_to_all_single(
                 recv_payload, split_payload, group=self._env.process_group
             )
-            prio_in = recv_payload[0::2].tolist()
-            defer_in = recv_payload[1::2].tolist()
+            recv_cpu = recv_payload.cpu()
+            prio_in = recv_cpu[0::2].tolist()
+            defer_in = recv_cpu[1::2].tolist()
             # my-row positions of each leg on the RECEIVING side: exchange the
             # within-block positions alongside (int64, 8 B/row)
             prio_pos = torch.cat(prio_pos_parts) if prio_pos_parts else positions[:0]
