@@ -486,10 +486,9 @@ class ShardedEmbeddingBagCollection(nn.Module):
             }
             assert not self._is_weighted, "VBE v1: unweighted only"
             assert all(
-                t in (ShardingType.TABLE_WISE.value, ShardingType.DATA_PARALLEL.value)
-                for t in self._sharding_types
+                t == ShardingType.TABLE_WISE.value for t in self._sharding_types
             ) or self._env.world_size == 1, (
-                "VBE through the sharded path supports TW/DP shardings"
+                "VBE through the sharded path supports TW shardings (v1)"
             )
         dist_input = self.input_dist(ctx, features).wait().wait()
         return self.compute_and_output_dist(ctx, dist_input)

@@ -548,6 +548,11 @@ class TableBatchedEmbeddingBags(nn.Module):
         per_sample_weights: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         """Variable-batch pooled forward -> 1-D packed [sum_f B_f * D_f]."""
+        assert self.optimizer != OPT_DENSE, (
+            "VBE v1 requires a fused optimizer (rowwise_adagrad/sgd); the "
+            "DENSE kernel's autograd grad surface is not wired for the "
+            "packed layout — shard VBE tables TW/RW instead of DP"
+        )
         F = self._num_features
         assert len(batch_size_per_feature) == F
         dims = [self._specs[t].dim for t in self._feature_table_map]
