@@ -637,3 +637,16 @@ def test_seg_sort_large_op_correct():
     ref = torch.sort(lin).values
     assert torch.equal(sorted_l, ref)
     assert torch.equal(lin[perm.long()], sorted_l)
+
+
+def test_vbe_weighted_cpu():
+    """Weighted VBE on the CPU oracle honors per-sample weights."""
+    torch.manual_seed(0)
+    tbe = TableBatchedEmbeddingBags([("t0", 20, 8)])
+    indices = torch.tensor([1, 2, 3])
+    offsets = torch.tensor([0, 2, 3])
+    psw = torch.tensor([2.0, 0.5, 3.0])
+    out = tbe.forward_vbe(indices, offsets, [2], psw)
+    w = tbe.split_embedding_weights()[0]
+    exp = torch.cat([2.0 * w[1] + 0.5 * w[2], 3.0 * w[3]])
+    torch.testing.assert_close(out, exp, atol=1e-6, rtol=1e-6)

@@ -563,7 +563,7 @@ class TableBatchedEmbeddingBags(nn.Module):
             out_off.append(out_off[-1] + bf * d)
         if not indices.is_cuda:
             return _TBEVbeCpuFunction.apply(
-                self._dummy, self, indices, offsets, bag_off
+                self._dummy, self, indices, offsets, per_sample_weights, bag_off
             )
         ops.hip_ops()
         device = indices.device
@@ -930,7 +930,7 @@ class _TBESeqDedupFunction(torch.autograd.Function):
         return None, None, None, None
 
 
-def _tbe_cpu_vbe_forward(weights, host, indices, offsets, bag_off):
+def _tbe_cpu_vbe_forward(weights, host, indices, offsets, bag_off, psw=None):
     outs = []
     for f in range(host._num_features):
         t = host._feature_table_map[f]
@@ -940,11 +940,14 @@ def _tbe_cpu_vbe_forward(weights, host, indices, offsets, bag_off):
         if w.dtype != torch.float32:
             w = w.float()
         off = offsets[bag_off[f] : bag_off[f + 1] + 1] - offsets[bag_off[f]]
-        idx = indices[int(offsets[bag_off[f]]) : int(offsets[bag_off[f + 1]])]
+        lo, hi = int(offsets[bag_off[f]]), int(offsets[bag_off[f + 1]])
+        idx = indices[lo:hi]
+        pw = psw[lo:hi] if psw is not None else None
         mode = {PoolingMode.SUM: "sum", PoolingMode.MEAN: "mean"}[host.pooling_mode]
         outs.append(
             torch.nn.functional.embedding_bag(
-                idx, w, off, mode=mode, include_last_offset=True
+                idx, w, off, mode=mode, include_last_offset=True,
+                per_sample_weights=pw,
             ).reshape(-1)
         )
     return torch.cat(outs) if outs else weights.new_empty(0)
@@ -954,23 +957,27 @@ class _TBEVbeCpuFunction(torch.autograd.Function):
     """CPU oracle of the VBE fused path (mirrors _TBECpuFusedFunction)."""
 
     @staticmethod
-    def forward(ctx, dummy, host, indices, offsets, bag_off):  # type: ignore[override]
+    def forward(ctx, dummy, host, indices, offsets, psw, bag_off):  # type: ignore[override]
         ctx.host = host
         ctx.bag_off = bag_off
-        ctx.save_for_backward(indices, offsets)
+        ctx.save_for_backward(
+            indices, offsets, psw.detach() if psw is not None else torch.empty(0)
+        )
+        ctx.has_psw = psw is not None
         with torch.no_grad():
-            return _tbe_cpu_vbe_forward(host.weights, host, indices, offsets, bag_off)
+            return _tbe_cpu_vbe_forward(host.weights, host, indices, offsets, bag_off, psw)
 
     @staticmethod
     def backward(ctx, grad):  # type: ignore[override]
         host = ctx.host
-        indices, offsets = ctx.saved_tensors
+        indices, offsets, psw = ctx.saved_tensors
+        psw_t = psw if ctx.has_psw else None
         w = host.weights.detach().float().requires_grad_(True)
         with torch.enable_grad():
-            out = _tbe_cpu_vbe_forward(w, host, indices, offsets, ctx.bag_off)
+            out = _tbe_cpu_vbe_forward(w, host, indices, offsets, ctx.bag_off, psw_t)
             out.backward(grad)
         host._cpu_apply_update(w.grad)
-        return None, None, None, None, None
+        return None, None, None, None, None, None
 
 
 class _TBESeqCpuFunction(torch.autograd.Function):
