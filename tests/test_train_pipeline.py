@@ -220,13 +220,19 @@ def _run_grad_accum(rank, world_size):
     dmp, opt, tables = _build_dmp(world_size)
     pipe = TrainPipelineBase(dmp, opt, torch.device("cpu"))
     gp = GradientAccumulationPipeline(pipe, accumulation_steps=2)
-    it = iter(_batches(rank, tables))
+    batches = _batches(rank, tables)
+    it = iter(list(batches))
     dense = [p for p in dmp.parameters() if p.requires_grad][0]
     w0 = dense.detach().clone()
     gp.progress(it)  # accumulate only: dense params unchanged
     torch.testing.assert_close(dense.detach(), w0)
-    gp.progress(it)  # boundary: dense step applied
+    g1 = dense.grad.detach().clone()
+    assert float(g1.abs().sum()) > 0
+    gp.progress(it)  # boundary: step over BOTH batches' grads
     assert not torch.equal(dense.detach(), w0)
+    # the boundary step must have seen accumulated (not just last-batch) grads
+    g2 = dense.grad.detach()
+    assert float((g2 - g1).abs().sum()) > 0  # grew past batch-1's grads
 
 
 def test_gradient_accumulation_pipeline():

@@ -312,25 +312,29 @@ class GradientAccumulationPipeline:
         self._i = 0
         self._opt = pipeline._optimizer
 
-        class _NoStep:
+        class _Ctl:
+            """zero_grad is ALWAYS a no-op inside the pipeline (it would
+            wipe the accumulation right before the boundary backward);
+            step fires only on window boundaries."""
+
             def __init__(self, opt):
                 self._opt = opt
+                self.do_step = False
 
             def zero_grad(self, set_to_none=True):
-                pass  # keep accumulating
-
-            def step(self):
                 pass
 
-        self._nostep = _NoStep(self._opt)
+            def step(self):
+                if self.do_step:
+                    self._opt.step()
+
+        self._ctl = _Ctl(self._opt)
 
     def progress(self, dataloader_iter):
-        first_of_window = self._i % self._n == 0
-        last_of_window = (self._i + 1) % self._n == 0
-        if first_of_window:
-            self._opt.zero_grad(set_to_none=True)
-        # swap a no-op optimizer in for non-boundary steps
-        self._pipeline._optimizer = self._opt if last_of_window else self._nostep
+        if self._i % self._n == 0:
+            self._opt.zero_grad(set_to_none=True)  # window start
+        self._ctl.do_step = (self._i + 1) % self._n == 0
+        self._pipeline._optimizer = self._ctl
         try:
             out = self._pipeline.progress(dataloader_iter)
         finally:
