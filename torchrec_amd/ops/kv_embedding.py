@@ -127,3 +127,107 @@ class KeyValueEmbeddingBags(nn.Module):
             self._reinit_slots(t, ev_slots)
             out_slots[lo:hi] = slots
         return self._tbe(out_slots.to(indices.device), offsets, per_sample_weights)
+
+
+class SsdEmbeddingBags(KeyValueEmbeddingBags):
+    """KV embedding with an SSD spill tier (reference:
+    SSDTableBatchedEmbeddingBags, batched_embedding_kernel.py:1961 — RocksDB
+    there; an append-only row log + in-memory id->offset index here).
+
+    Rows evicted from the bounded DRAM table are written (with their
+    optimizer state) to a per-table on-disk log before the slot is
+    re-initialized; when a previously-evicted raw id re-enters the working
+    set its row is restored from disk instead of fresh-initialized, so
+    training state survives eviction — the semantic the reference's SSD
+    tier provides."""
+
+    def __init__(self, *args, storage_dir: Optional[str] = None, **kwargs) -> None:
+        super().__init__(*args, **kwargs)
+        import os
+        import tempfile
+
+        self._dir = storage_dir or tempfile.mkdtemp(prefix="trec_amd_ssd_")
+        os.makedirs(self._dir, exist_ok=True)
+        self._files = []
+        self._disk_index: List[dict] = []
+        for ti, spec in enumerate(self._tbe.embedding_specs):
+            self._files.append(open(f"{self._dir}/table_{ti}.rows", "w+b"))
+            self._disk_index.append({})
+
+    def _spill(self, table: int, slots: torch.Tensor, ids: torch.Tensor) -> None:
+        """Write evicted rows (+ momentum) to the log before re-init."""
+        if slots.numel() == 0:
+            return
+        import numpy as np
+
+        w = self._tbe.split_embedding_weights()[table]
+        states = self._tbe.split_optimizer_states()[table]
+        f = self._files[table]
+        idx = self._disk_index[table]
+        rows = w[slots.to(w.device)].float().cpu().numpy()
+        mom = (
+            states[0][slots.to(states[0].device)].float().cpu().numpy()
+            if states
+            else np.zeros(slots.numel(), dtype=np.float32)
+        )
+        for k in range(slots.numel()):
+            f.seek(0, 2)
+            off = f.tell()
+            f.write(rows[k].tobytes())
+            f.write(mom[k : k + 1].tobytes())
+            idx[int(ids[k])] = off
+
+    def _restore(self, table: int, slot: int, raw_id: int) -> bool:
+        import numpy as np
+
+        idx = self._disk_index[table]
+        off = idx.get(raw_id)
+        if off is None:
+            return False
+        dim = self._tbe.embedding_specs[table].dim
+        f = self._files[table]
+        f.seek(off)
+        buf = f.read(dim * 4 + 4)
+        row = np.frombuffer(buf[: dim * 4], dtype=np.float32)
+        m = np.frombuffer(buf[dim * 4 :], dtype=np.float32)[0]
+        with torch.no_grad():
+            w = self._tbe.split_embedding_weights()[table]
+            w[slot] = torch.from_numpy(row.copy()).to(w.device, w.dtype)
+            states = self._tbe.split_optimizer_states()[table]
+            if states:
+                states[0][slot] = float(m)
+        return True
+
+    def forward(
+        self,
+        indices: torch.Tensor,
+        offsets: torch.Tensor,
+        per_sample_weights: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        F = len(self._ftm)
+        B = (offsets.numel() - 1) // F
+        idx_cpu = indices.cpu()
+        off_cpu = offsets.cpu()
+        out_slots = torch.empty_like(idx_cpu)
+        for f in range(F):
+            t = self._ftm[f]
+            lo, hi = int(off_cpu[f * B]), int(off_cpu[(f + 1) * B])
+            if hi == lo:
+                continue
+            seg = idx_cpu[lo:hi]
+            slots, ev_slots, ev_ids = self._transformers[t].transform(seg)
+            if ev_slots.numel():
+                self._spill(t, ev_slots, ev_ids)
+                self._reinit_slots(t, ev_slots)
+            # re-admitted ids: restore spilled rows into their fresh slots
+            for k in range(seg.numel()):
+                rid = int(seg[k])
+                if rid in self._disk_index[t]:
+                    if self._restore(t, int(slots[k]), rid):
+                        del self._disk_index[t][rid]
+            out_slots[lo:hi] = slots
+        return self._tbe(out_slots.to(indices.device), offsets, per_sample_weights)
+
+    def close(self) -> None:
+        for f in self._files:
+            f.close()
