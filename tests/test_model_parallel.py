@@ -593,3 +593,66 @@ def _run_mixed_sharding(rank, world_size):
 
 def test_mixed_sharding_types_one_ebc():
     run_multi_process(_run_mixed_sharding, 2, "gloo")
+
+
+def _run_qcomm_mx4(rank, world_size):
+    """MX4 (4-bit microscaling) pooled a2a: group-aligned wire, lossy but
+    bounded (shared power-of-two exponent per 32 values)."""
+    from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
+
+    B = 4
+    tables = [
+        EmbeddingBagConfig(num_embeddings=20, embedding_dim=32, name="t0", feature_names=["f0"]),
+        EmbeddingBagConfig(num_embeddings=30, embedding_dim=32, name="t1", feature_names=["f1"]),
+    ]
+    torch.manual_seed(42)
+    model = SparseModel(list(tables))
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(sharding_types=[ShardingType.TABLE_WISE.value])
+            for cfg in tables
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={
+            "optimizer": "rowwise_adagrad",
+            "learning_rate": LR,
+            "qcomms_config": QCommsConfig(forward_precision=CommType.MX4),
+        }
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    golden = _golden(tables, None, world_size)
+    dmp.load_state_dict(
+        {
+            f"sparse.embedding_bags.{cfg.name}.weight": w
+            for cfg, w in zip(tables, golden.split_embedding_weights())
+        },
+        strict=False,
+    )
+    kjt_global = make_global_kjt(tables, B * world_size)
+    kjt_local = kjt_local_slice(kjt_global, rank * B, (rank + 1) * B)
+    vals = dmp(kjt_local).values()
+    expected = golden(kjt_global).values()[rank * B : (rank + 1) * B]
+    # 4-bit wire: ~12% relative tolerance on the pooled sums
+    scale = expected.abs().max().clamp(min=1e-6)
+    assert float((vals - expected).abs().max() / scale) < 0.15
+
+
+def test_qcomm_mx4_tw():
+    run_multi_process(_run_qcomm_mx4, 2, "gloo")
+
+
+def test_mx4_codec_roundtrip():
+    from torchrec_amd.distributed.qcomm_codecs import CommType, QuantizedCommCodec
+
+    torch.manual_seed(0)
+    c = QuantizedCommCodec(CommType.MX4)
+    x = torch.randn(256) * 5
+    enc = c.encode(x)
+    assert enc.numel() == c.encoded_numel(256) == (256 // 32) * 17
+    dec = c.decode(enc, 256)
+    assert float((dec - x).abs().max() / x.abs().max()) < 0.12
