@@ -1,4 +1,5 @@
 """Managed-collision (ZCH) tests (reference: torchrec/modules/tests/test_mc_modules.py)."""
+import pytest
 
 import torch
 
@@ -90,3 +91,48 @@ class TestMCEmbeddingModules:
         )
         out, _ = mc_ebc(kjt)
         assert out.values().shape == (2, 8)
+
+
+class TestHashZch:
+    def test_cpu_zero_collision(self):
+        from torchrec_amd.modules.hash_mc_modules import HashZchManagedCollisionModule
+
+        m = HashZchManagedCollisionModule(zch_size=64, max_probe=64)
+        ids = torch.tensor([10**12, 55, 10**12, 999999, 55])
+        s1 = m.remap(ids)
+        # identical raw ids -> identical slots; distinct ids -> distinct slots
+        assert s1[0] == s1[2] and s1[1] == s1[4]
+        assert len({int(s1[0]), int(s1[1]), int(s1[3])}) == 3
+        # stable across calls
+        s2 = m.remap(ids)
+        assert torch.equal(s1, s2)
+
+    def test_cpu_eviction_frees_slots(self):
+        from torchrec_amd.modules.hash_mc_modules import HashZchManagedCollisionModule
+
+        m = HashZchManagedCollisionModule(zch_size=16, max_probe=16, eviction_interval=4)
+        m.remap(torch.tensor([1, 2, 3]))
+        for _ in range(3):
+            m.remap(torch.tensor([100]))  # ids 1..3 go stale
+        freed = m.evict()
+        assert freed is not None and freed.numel() >= 3
+
+    @pytest.mark.gpu
+    def test_gpu_matches_semantics(self):
+        from torchrec_amd.modules.hash_mc_modules import HashZchManagedCollisionModule
+
+        m = HashZchManagedCollisionModule(
+            zch_size=128, device=torch.device("cuda"), max_probe=128
+        )
+        ids = torch.tensor([10**12, 55, 10**12, 999999, 55, 10**12]).cuda()
+        s1 = m.remap(ids)
+        torch.cuda.synchronize()
+        assert int(s1[0]) == int(s1[2]) == int(s1[5])
+        assert int(s1[1]) == int(s1[4])
+        assert len({int(s1[0]), int(s1[1]), int(s1[3])}) == 3
+        s2 = m.remap(ids)
+        assert torch.equal(s1, s2)
+        # capacity-many distinct ids all get distinct slots (zero collision)
+        many = torch.arange(100).cuda() * 7919
+        slots = m.remap(many)
+        assert slots.unique().numel() == 100

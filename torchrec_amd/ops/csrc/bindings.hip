@@ -76,6 +76,10 @@ void lxu_cache_populate(at::Tensor host_weights, const at::Tensor& table_row_off
                         at::Tensor cache_tags, at::Tensor cache_lru, int64_t max_D,
                         int64_t timestamp);
 at::Tensor lxu_cache_lookup(const at::Tensor& ids, const at::Tensor& cache_tags);
+at::Tensor hash_zch_remap(const at::Tensor& ids, at::Tensor identity,
+                          at::Tensor metadata, int64_t max_probe, int64_t stamp,
+                          bool train);
+at::Tensor hash_zch_evict(at::Tensor identity, at::Tensor metadata, int64_t older_than);
 void lxu_cache_flush(at::Tensor host_weights, const at::Tensor& table_row_offsets,
                      const at::Tensor& table_elem_offsets, const at::Tensor& dims,
                      const at::Tensor& cache_weights, const at::Tensor& cache_tags,
@@ -160,6 +164,10 @@ TORCH_LIBRARY(trec_amd, m) {
       " Tensor(d!) cache_lru, int max_D, int timestamp) -> ()");
   m.def("lxu_cache_lookup(Tensor ids, Tensor cache_tags) -> Tensor");
   m.def(
+      "hash_zch_remap(Tensor ids, Tensor(a!) identity, Tensor(b!) metadata, "
+      "int max_probe, int stamp, bool train) -> Tensor");
+  m.def("hash_zch_evict(Tensor(a!) identity, Tensor(b!) metadata, int older_than) -> Tensor");
+  m.def(
       "lxu_cache_flush(Tensor(a!) host_weights, Tensor table_row_offsets,"
       " Tensor table_elem_offsets, Tensor dims, Tensor cache_weights, Tensor cache_tags,"
       " int max_D) -> ()");
@@ -207,6 +215,8 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("gather_run_heads", trec_amd::gather_run_heads);
   m.impl("lxu_cache_populate", trec_amd::lxu_cache_populate);
   m.impl("lxu_cache_lookup", trec_amd::lxu_cache_lookup);
+  m.impl("hash_zch_remap", trec_amd::hash_zch_remap);
+  m.impl("hash_zch_evict", trec_amd::hash_zch_evict);
   m.impl("lxu_cache_flush", trec_amd::lxu_cache_flush);
   m.impl("bounds_check_indices", trec_amd::bounds_check_indices);
   m.impl("quantize_rowwise_int8", trec_amd::quantize_rowwise_int8);

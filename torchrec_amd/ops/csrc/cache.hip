@@ -198,4 +198,104 @@ void lxu_cache_flush(at::Tensor host_weights, const at::Tensor& table_row_offset
                      max_D);
 }
 
+
+// ---------------------------------------------------------------------------
+// MPZCH: GPU zero-collision hash remapping (reference:
+// torchrec/modules/hash_mc_modules.py HashZchManagedCollisionModule :196,
+// fbgemm zero_collision_hash :451).
+//
+// identity[z] holds the raw id owning slot z (-1 = free). Each input id
+// linear-probes from hash(id) % Z; in training mode a free slot is claimed
+// with a 64-bit atomicCAS (first-claimer wins — the id->slot map stays
+// consistent because every probe sequence re-reads the claimed value).
+// metadata[z] is a last-seen counter for eviction.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint64_t zch_mix(uint64_t x) {
+  // splitmix64 finalizer
+  x += 0x9e3779b97f4a7c15ull;
+  x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ull;
+  x = (x ^ (x >> 27)) * 0x94d049bb133111ebull;
+  return x ^ (x >> 31);
+}
+
+__global__ void hash_zch_kernel(const int64_t* __restrict__ ids, int64_t n,
+                                int64_t* __restrict__ identity,
+                                int32_t* __restrict__ metadata, int64_t Z,
+                                int max_probe, int32_t stamp, bool train,
+                                int64_t* __restrict__ out) {
+  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; i < n;
+       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    int64_t id = ids[i];
+    uint64_t h = zch_mix(static_cast<uint64_t>(id));
+    int64_t slot = -1;
+    for (int p = 0; p < max_probe; ++p) {
+      int64_t z = static_cast<int64_t>((h + p) % static_cast<uint64_t>(Z));
+      int64_t cur = identity[z];
+      if (cur == id) { slot = z; break; }
+      if (cur == -1 && train) {
+        int64_t prev = atomicCAS(
+            reinterpret_cast<unsigned long long*>(identity + z),
+            static_cast<unsigned long long>(-1ll),
+            static_cast<unsigned long long>(id));
+        if (prev == static_cast<unsigned long long>(-1ll) ||
+            static_cast<int64_t>(prev) == id) {
+          slot = z;
+          break;
+        }
+        cur = static_cast<int64_t>(prev);
+        if (cur == id) { slot = z; break; }
+      }
+    }
+    if (slot < 0) {
+      // probe budget exhausted: shared fallback bucket (hash slot)
+      slot = static_cast<int64_t>(h % static_cast<uint64_t>(Z));
+    }
+    if (train) metadata[slot] = stamp;
+    out[i] = slot;
+  }
+}
+
+__global__ void hash_zch_evict_kernel(int64_t* __restrict__ identity,
+                                      int32_t* __restrict__ metadata, int64_t Z,
+                                      int32_t older_than,
+                                      int64_t* __restrict__ evicted /* [Z] -1 pad */) {
+  for (int64_t z = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; z < Z;
+       z += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    if (identity[z] != -1 && metadata[z] < older_than) {
+      evicted[z] = z;
+      identity[z] = -1;
+      metadata[z] = 0;
+    } else {
+      evicted[z] = -1;
+    }
+  }
+}
+
+at::Tensor hash_zch_remap(const at::Tensor& ids, at::Tensor identity,
+                          at::Tensor metadata, int64_t max_probe, int64_t stamp,
+                          bool train) {
+  TORCH_CHECK(ids.is_cuda() && identity.is_cuda());
+  int64_t n = ids.numel();
+  int64_t Z = identity.numel();
+  auto out = at::empty({n}, ids.options());
+  if (n == 0) return out;
+  hipLaunchKernelGGL(hash_zch_kernel, dim3(grid_for(n, kBlockThreads)),
+                     dim3(kBlockThreads), 0, c_stream(), ids.data_ptr<int64_t>(), n,
+                     identity.data_ptr<int64_t>(), metadata.data_ptr<int32_t>(), Z,
+                     (int)max_probe, (int32_t)stamp, train, out.data_ptr<int64_t>());
+  return out;
+}
+
+at::Tensor hash_zch_evict(at::Tensor identity, at::Tensor metadata, int64_t older_than) {
+  int64_t Z = identity.numel();
+  auto evicted = at::empty({Z}, identity.options());
+  if (Z == 0) return evicted;
+  hipLaunchKernelGGL(hash_zch_evict_kernel, dim3(grid_for(Z, kBlockThreads)),
+                     dim3(kBlockThreads), 0, c_stream(), identity.data_ptr<int64_t>(),
+                     metadata.data_ptr<int32_t>(), Z, (int32_t)older_than,
+                     evicted.data_ptr<int64_t>());
+  return evicted;
+}
+
 }  // namespace trec_amd
