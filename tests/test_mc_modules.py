@@ -110,11 +110,11 @@ class TestHashZch:
     def test_cpu_eviction_frees_slots(self):
         from torchrec_amd.modules.hash_mc_modules import HashZchManagedCollisionModule
 
-        m = HashZchManagedCollisionModule(zch_size=16, max_probe=16, eviction_interval=4)
-        m.remap(torch.tensor([1, 2, 3]))
+        m = HashZchManagedCollisionModule(zch_size=16, max_probe=16, eviction_interval=2)
+        m.remap(torch.tensor([1, 2, 3]))  # step 1
         for _ in range(3):
-            m.remap(torch.tensor([100]))  # ids 1..3 go stale
-        freed = m.evict()
+            m.remap(torch.tensor([100]))  # steps 2-4: ids 1..3 leave the window
+        freed = m.evict()  # step 4, window [3, 4]
         assert freed is not None and freed.numel() >= 3
 
     @pytest.mark.gpu

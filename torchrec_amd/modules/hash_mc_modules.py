@@ -102,7 +102,8 @@ class HashZchManagedCollisionModule(nn.Module):
         """Free slots unseen for an eviction interval; returns freed slots."""
         if self._step == 0 or self._step % self._eviction_interval != 0:
             return None
-        older = self._step - self._eviction_interval
+        # stale = not seen within the last `eviction_interval` steps
+        older = self._step - self._eviction_interval + 1
         if self.identity.is_cuda:
             ops.hip_ops()
             ev = torch.ops.trec_amd.hash_zch_evict(self.identity, self.metadata, older)
