@@ -241,3 +241,18 @@ def test_vbe_to_dict_unequal_strides():
     assert d["b"].values().tolist() == [4, 5, 6]
     assert d["b"].lengths().tolist() == [2, 1]
     assert kjt.length_per_key() == [4, 3]
+
+
+def test_keyed_jagged_index_select_dim1():
+    from torchrec_amd import ops
+
+    # K=2 keys, B=3: select batch positions [2, 0]
+    lengths = torch.tensor([1, 2, 1, 2, 0, 1])
+    values = torch.arange(7)
+    offsets = torch.zeros(7, dtype=torch.int64)
+    torch.cumsum(lengths, 0, out=offsets[1:])
+    v, l, _ = ops.keyed_jagged_index_select_dim1(
+        values, lengths, offsets, torch.tensor([2, 0]), batch_size=3
+    )
+    assert l.tolist() == [1, 1, 1, 2]  # k0:[b2,b0], k1:[b2,b0]
+    assert v.tolist() == [3, 0, 6, 4, 5]

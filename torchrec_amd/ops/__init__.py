@@ -524,3 +524,24 @@ def expand_into_jagged_permute(
         src = int(input_offsets[int(permute[i])])
         out[o0:o1] = torch.arange(src, src + (o1 - o0), device=permute.device)
     return out
+
+
+def keyed_jagged_index_select_dim1(
+    values: torch.Tensor,
+    lengths: torch.Tensor,
+    offsets: torch.Tensor,
+    indices: torch.Tensor,
+    batch_size: int,
+    weights: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
+    """Select batch positions across every key of a feature-major KJT
+    (semantics of ``fbgemm.keyed_jagged_index_select_dim1``, reference
+    torchrec/sparse/jagged_tensor.py:531). ``lengths`` is [K*B]; the same
+    ``indices`` (into the batch dim) apply to each key. Returns
+    (values, lengths[, weights]) of the selected KJT."""
+    K = lengths.numel() // batch_size
+    sel = torch.cat(
+        [indices + k * batch_size for k in range(K)]
+    )
+    pl, pv, pw = permute_1d_sparse_data(sel, lengths, values, weights)
+    return pv, pl, pw
