@@ -359,3 +359,51 @@ def test_dlrm_dcn_and_two_tower_cuda():
     assert q.shape == (B, 8) and c.shape == (B, 8)
     (q * c).sum().backward()
     torch.cuda.synchronize()
+
+
+def test_e2e_determinism_bitwise():
+    """Two identical runs produce BITWISE-equal table weights after 5 steps —
+    the no-atomics backward design's core claim."""
+    import bench
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.types import ShardingEnv
+
+    device = torch.device("cuda", 0)
+
+    def run_once():
+        torch.manual_seed(0)
+        model = bench.build_model(1e-4)
+        sharder = EmbeddingBagCollectionSharder(
+            fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.05,
+                          "fixed_bag_length": 1}
+        )
+        planner = EmbeddingShardingPlanner(
+            topology=Topology(world_size=1, compute_device="cuda", batch_size=64)
+        )
+        plan = planner.plan(model, [sharder])
+        dmp = DistributedModelParallel(
+            model, env=ShardingEnv.from_local(1, 0), plan=plan,
+            sharders=[sharder], device=device,
+        )
+        opt = torch.optim.SGD([p for p in dmp.parameters() if p.requires_grad], lr=0.05)
+        batches = bench.make_host_batches(5, 64, 1e-4, seed=9, pin=True)
+        for b in batches:
+            bd = b.to(device)
+            loss, _ = dmp(bd)
+            opt.zero_grad(set_to_none=True)
+            loss.backward()
+            opt.step()
+        torch.cuda.synchronize()
+        return [
+            t.weights.detach().clone()
+            for sm in dmp.sharded_modules().values()
+            for t in sm.tbes()
+        ]
+
+    w1 = run_once()
+    w2 = run_once()
+    for a, b in zip(w1, w2):
+        assert torch.equal(a, b)
