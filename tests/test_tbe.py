@@ -650,3 +650,25 @@ def test_vbe_weighted_cpu():
     w = tbe.split_embedding_weights()[0]
     exp = torch.cat([2.0 * w[1] + 0.5 * w[2], 3.0 * w[3]])
     torch.testing.assert_close(out, exp, atol=1e-6, rtol=1e-6)
+
+
+@pytest.mark.gpu
+def test_precision_wide_dims_gpu():
+    """bf16 tables on the CHUNKS>1 kernel path (D=512, LPS=64)."""
+    specs = [("t0", 40, 512), ("t1", 30, 256)]
+    torch.manual_seed(0)
+    cpu = TableBatchedEmbeddingBags(specs, weights_precision="bf16", learning_rate=0.05)
+    gpu = TableBatchedEmbeddingBags(
+        specs, weights_precision="bf16", learning_rate=0.05, device=torch.device("cuda")
+    )
+    gpu.weights.data.copy_(cpu.weights.data)
+    indices, offsets = make_inputs(specs, B=8, L=4, seed=2)
+    out_c = cpu(indices, offsets)
+    out_g = gpu(indices.cuda(), offsets.cuda())
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out_g.cpu(), out_c, atol=1e-3, rtol=1e-3)
+    out_c.sum().backward()
+    out_g.sum().backward()
+    torch.cuda.synchronize()
+    for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+        torch.testing.assert_close(wg.cpu().float(), wc.float(), atol=2e-2, rtol=2e-2)
