@@ -141,58 +141,47 @@ class SsdEmbeddingBags(KeyValueEmbeddingBags):
     training state survives eviction — the semantic the reference's SSD
     tier provides."""
 
-    def __init__(self, *args, storage_dir: Optional[str] = None, **kwargs) -> None:
+    def __init__(
+        self,
+        *args,
+        storage_dir: Optional[str] = None,
+        io: str = "file",
+        **kwargs,
+    ) -> None:
         super().__init__(*args, **kwargs)
-        import os
-        import tempfile
+        from torchrec_amd.dynamic_embedding.ps import ParameterServer
 
-        self._dir = storage_dir or tempfile.mkdtemp(prefix="trec_amd_ssd_")
-        os.makedirs(self._dir, exist_ok=True)
-        self._files = []
-        self._disk_index: List[dict] = []
-        for ti, spec in enumerate(self._tbe.embedding_specs):
-            self._files.append(open(f"{self._dir}/table_{ti}.rows", "w+b"))
-            self._disk_index.append({})
+        io_kwargs = {}
+        if io == "file" and storage_dir is not None:
+            io_kwargs["path"] = None  # FilePSIO manages per-table paths below
+        self._ps = ParameterServer(
+            [s.dim for s in self._tbe.embedding_specs], io=io
+        )
 
     def _spill(self, table: int, slots: torch.Tensor, ids: torch.Tensor) -> None:
-        """Write evicted rows (+ momentum) to the log before re-init."""
+        """Push evicted rows (+ momentum) to the parameter server."""
         if slots.numel() == 0:
             return
-        import numpy as np
-
         w = self._tbe.split_embedding_weights()[table]
         states = self._tbe.split_optimizer_states()[table]
-        f = self._files[table]
-        idx = self._disk_index[table]
-        rows = w[slots.to(w.device)].float().cpu().numpy()
+        rows = w[slots.to(w.device)]
         mom = (
-            states[0][slots.to(states[0].device)].float().cpu().numpy()
+            states[0][slots.to(states[0].device)]
             if states
-            else np.zeros(slots.numel(), dtype=np.float32)
+            else torch.zeros(slots.numel())
         )
-        for k in range(slots.numel()):
-            f.seek(0, 2)
-            off = f.tell()
-            f.write(rows[k].tobytes())
-            f.write(mom[k : k + 1].tobytes())
-            idx[int(ids[k])] = off
+        self._ps.evict(table, ids, rows, mom)
 
     def _restore(self, table: int, slot: int, raw_id: int) -> bool:
         import numpy as np
 
-        idx = self._disk_index[table]
-        off = idx.get(raw_id)
-        if off is None:
+        got = self._ps.fetch(table, raw_id)
+        if got is None:
             return False
-        dim = self._tbe.embedding_specs[table].dim
-        f = self._files[table]
-        f.seek(off)
-        buf = f.read(dim * 4 + 4)
-        row = np.frombuffer(buf[: dim * 4], dtype=np.float32)
-        m = np.frombuffer(buf[dim * 4 :], dtype=np.float32)[0]
+        row, m = got
         with torch.no_grad():
             w = self._tbe.split_embedding_weights()[table]
-            w[slot] = torch.from_numpy(row.copy()).to(w.device, w.dtype)
+            w[slot] = torch.from_numpy(np.asarray(row)).to(w.device, w.dtype)
             states = self._tbe.split_optimizer_states()[table]
             if states:
                 states[0][slot] = float(m)
@@ -221,13 +210,9 @@ class SsdEmbeddingBags(KeyValueEmbeddingBags):
                 self._reinit_slots(t, ev_slots)
             # re-admitted ids: restore spilled rows into their fresh slots
             for k in range(seg.numel()):
-                rid = int(seg[k])
-                if rid in self._disk_index[t]:
-                    if self._restore(t, int(slots[k]), rid):
-                        del self._disk_index[t][rid]
+                self._restore(t, int(slots[k]), int(seg[k]))
             out_slots[lo:hi] = slots
         return self._tbe(out_slots.to(indices.device), offsets, per_sample_weights)
 
     def close(self) -> None:
-        for f in self._files:
-            f.close()
+        self._ps.close()
