@@ -201,9 +201,26 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
                 counts[ps.sharding_type] = counts.get(ps.sharding_type, 0) + 1
                 kcounts[ps.compute_kernel] = kcounts.get(ps.compute_kernel, 0) + 1
         print(f"# plan sharding mix: {counts} kernels: {kcounts}", flush=True)
-    dense_opt = torch.optim.SGD(
-        [p for p in dmp.parameters() if p.requires_grad], lr=LR
-    )
+    dense_params = [p for p in dmp.parameters() if p.requires_grad]
+    if os.environ.get("TREC_DENSE_OIB") == "1":
+        # dense optimizer applied inside backward (per-param post-accumulate
+        # hooks): drops the separate foreach-SGD launch at step end
+        from torchrec_amd.optim.apply_optimizer_in_backward import (
+            apply_optimizer_in_backward,
+        )
+
+        apply_optimizer_in_backward(torch.optim.SGD, dense_params, {"lr": LR})
+
+        class _NoOpOpt:
+            def zero_grad(self, set_to_none=True):
+                pass
+
+            def step(self):
+                pass
+
+        dense_opt = _NoOpOpt()
+    else:
+        dense_opt = torch.optim.SGD(dense_params, lr=LR)
     pipeline_cls = TrainPipelineSparseDist
     if os.environ.get("TREC_PIPELINE") == "fused":
         from torchrec_amd.distributed.train_pipeline import TrainPipelineFusedSparseDist
