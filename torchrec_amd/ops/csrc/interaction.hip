@@ -27,18 +27,20 @@ static inline hipStream_t ia_stream() {
 
 // pair p (i<j) column = D + p; host precomputes i_of_pair / j_of_pair.
 
-__global__ void __launch_bounds__(64) interaction_fwd_kernel(
+__global__ void __launch_bounds__(kBlockThreads) interaction_fwd_kernel(
     const float* __restrict__ dense,   // [B, D]
     const float* __restrict__ sparse,  // [B, F, D]
     const int8_t* __restrict__ pi,     // [P] row i of pair p
     const int8_t* __restrict__ pj,     // [P] row j of pair p
     int B, int F1, int D, int P, float* __restrict__ out /* [B, D+P] */) {
-  // ONE wave per block: __syncthreads() is a cheap wave-local fence, no
-  // cross-wave iteration-count hazards in the grid-stride loop.
+  // 256 threads (4 waves) per sample: staging and the P pair-dots spread
+  // across the whole block (351 pairs -> ~1.4 per lane instead of 5.5 on a
+  // single wave). The grid-stride trip count depends only on blockIdx, so
+  // __syncthreads() inside the loop is uniform per block.
   // LDS tile is float4-strided with a +1 float4 pad: rows stay 16B-aligned
   // for ds_read_b128 while the pad staggers banks across rows.
   extern __shared__ float lds[];  // [F1][D/4+1] float4
-  int l = lane_id();
+  int tid = threadIdx.x;
   float4* T4 = reinterpret_cast<float4*>(lds);
   const int d4 = D / 4;
   const int stride4 = d4 + 1;
@@ -46,17 +48,17 @@ __global__ void __launch_bounds__(64) interaction_fwd_kernel(
   for (int64_t b = blockIdx.x; b < B; b += gridDim.x) {
     // stage T: row 0 = dense, rows 1..F1-1 = sparse
     const float4* drow = reinterpret_cast<const float4*>(dense + b * D);
-    for (int t = l; t < d4; t += kWaveSize) T4[t] = drow[t];
+    for (int t = tid; t < d4; t += blockDim.x) T4[t] = drow[t];
     const float4* srow = reinterpret_cast<const float4*>(sparse + b * (int64_t)(F1 - 1) * D);
-    for (int t = l; t < (F1 - 1) * d4; t += kWaveSize) {
+    for (int t = tid; t < (F1 - 1) * d4; t += blockDim.x) {
       int r = t / d4, c = t - r * d4;
       T4[(r + 1) * stride4 + c] = srow[t];
     }
     __syncthreads();
     float* orow = out + b * out_w;
     float4* orow4 = reinterpret_cast<float4*>(orow);
-    for (int t = l; t < d4; t += kWaveSize) orow4[t] = T4[t];
-    for (int p = l; p < P; p += kWaveSize) {
+    for (int t = tid; t < d4; t += blockDim.x) orow4[t] = T4[t];
+    for (int p = tid; p < P; p += blockDim.x) {
       const float4* Ti = T4 + pi[p] * stride4;
       const float4* Tj = T4 + pj[p] * stride4;
       float acc = 0.f;
@@ -149,7 +151,7 @@ at::Tensor interaction_forward(const at::Tensor& dense, const at::Tensor& sparse
   TORCH_CHECK(D % 4 == 0, "interaction kernel needs D %% 4 == 0");
   int lds_bytes = F1 * (D / 4 + 1) * sizeof(float4);
   int grid = std::min<int>(B, kNumCU * 32);
-  hipLaunchKernelGGL(interaction_fwd_kernel, dim3(grid), dim3(64), lds_bytes, ia_stream(),
+  hipLaunchKernelGGL(interaction_fwd_kernel, dim3(grid), dim3(kBlockThreads), lds_bytes, ia_stream(),
                      dense.contiguous().data_ptr<float>(),
                      sparse.contiguous().data_ptr<float>(), pi.data_ptr<int8_t>(),
                      pj.data_ptr<int8_t>(), B, F1, D, P, out.data_ptr<float>());
