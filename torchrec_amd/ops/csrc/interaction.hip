@@ -81,14 +81,18 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
     float* __restrict__ d_dense,   // [B, D]
     float* __restrict__ d_sparse   // [B, F, D]
 ) {
-  extern __shared__ float lds[];  // [F1][D/4+1] float4 then [P] dz floats
+  extern __shared__ float lds[];  // [F1][D/4+1] float4, [P] dz, [F1*F1] cols
   float4* T4 = reinterpret_cast<float4*>(lds);
   float* dzbuf = lds + 4 * F1 * (D / 4 + 1);
+  int32_t* pc = reinterpret_cast<int32_t*>(dzbuf + P);
   int l = lane_id();
   int wave = wave_id();
   const int d4 = D / 4;
   const int stride4 = d4 + 1;
   int64_t out_w = D + P;
+  // pair_col is sample-invariant: stage it once per block
+  for (int t = threadIdx.x; t < F1 * F1; t += blockDim.x) pc[t] = pair_col[t];
+  __syncthreads();
   for (int64_t b = blockIdx.x; b < B; b += gridDim.x) {
     const float4* drow = reinterpret_cast<const float4*>(dense + b * D);
     for (int t = threadIdx.x; t < d4; t += blockDim.x) T4[t] = drow[t];
@@ -117,7 +121,7 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
           acc = make_float4(0.f, 0.f, 0.f, 0.f);
         }
         for (int j = 0; j < F1; ++j) {
-          int c = pair_col[i * F1 + j];
+          int c = pc[i * F1 + j];
           if (c >= 0) {
             float dz = dzbuf[c];
             float4 t = T4[j * stride4 + k];
@@ -169,7 +173,8 @@ std::tuple<at::Tensor, at::Tensor> interaction_backward(
   auto d_sparse = at::empty_like(sparse);
   if (B == 0) return {d_dense, d_sparse};
   TORCH_CHECK(D % 4 == 0, "interaction kernel needs D %% 4 == 0");
-  int lds_bytes = F1 * (D / 4 + 1) * sizeof(float4) + P * sizeof(float);
+  int lds_bytes = F1 * (D / 4 + 1) * sizeof(float4) + P * sizeof(float)
+                  + F1 * F1 * sizeof(int32_t);
   int grid = std::min<int>(B, kMaxBlocks);
   hipLaunchKernelGGL(interaction_bwd_kernel, dim3(grid), dim3(kBlockThreads), lds_bytes,
                      ia_stream(), grad_out.contiguous().data_ptr<float>(),
