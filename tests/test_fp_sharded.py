@@ -92,6 +92,10 @@ def _run_fp_sharded(rank, world_size):
     vals.sum().backward()
     for k, p in dmp.module.sparse.feature_processors.position_weights.items():
         assert p.grad is not None and float(p.grad.abs().sum()) > 0
+    # processor params are part of the checkpoint
+    sd = dmp.state_dict()
+    fp_keys = [k for k in sd if "feature_processors" in k]
+    assert fp_keys, sorted(sd)[:5]
 
 
 def test_fp_ebc_sharded():

@@ -88,11 +88,28 @@ class ShardedFeatureProcessedEmbeddingBagCollection(nn.Module):
     def fused_optimizer(self):
         return self._ebc.fused_optimizer
 
-    def state_dict(self, *args, **kwargs):
-        return self._ebc.state_dict(*args, **kwargs)
+    def state_dict(self, destination=None, prefix: str = "", keep_vars: bool = False):
+        destination = self._ebc.state_dict(destination, prefix, keep_vars)
+        # the owning rank checkpoints its position-weight parameters too
+        for n, p in self.feature_processors.named_parameters():
+            destination[f"{prefix}feature_processors.{n}"] = (
+                p if keep_vars else p.detach()
+            )
+        return destination
 
-    def _load_from_state_dict(self, *args, **kwargs):
-        return self._ebc._load_from_state_dict(*args, **kwargs)
+    def _load_from_state_dict(
+        self, state_dict, prefix, local_metadata, strict, missing_keys,
+        unexpected_keys, error_msgs,
+    ):
+        for n, p in self.feature_processors.named_parameters():
+            key = f"{prefix}feature_processors.{n}"
+            if key in state_dict:
+                with torch.no_grad():
+                    p.copy_(state_dict[key])
+        return self._ebc._load_from_state_dict(
+            state_dict, prefix, local_metadata, strict, missing_keys,
+            unexpected_keys, error_msgs,
+        )
 
 
 class FeatureProcessedEmbeddingBagCollectionSharder(
