@@ -59,11 +59,28 @@ class ShardedITEPEmbeddingBagCollection(nn.Module):
     def fused_optimizer(self):
         return self._ebc.fused_optimizer
 
-    def state_dict(self, *args, **kwargs):
-        return self._ebc.state_dict(*args, **kwargs)
+    def state_dict(self, destination=None, prefix: str = "", keep_vars: bool = False):
+        destination = self._ebc.state_dict(destination, prefix, keep_vars)
+        # pruning state (address lookup / row utilisation) checkpoints too
+        for n, b in self._itep.state_dict().items():
+            destination[f"{prefix}itep.{n}"] = b
+        return destination
 
-    def _load_from_state_dict(self, *args, **kwargs):
-        return self._ebc._load_from_state_dict(*args, **kwargs)
+    def _load_from_state_dict(
+        self, state_dict, prefix, local_metadata, strict, missing_keys,
+        unexpected_keys, error_msgs,
+    ):
+        itep_sd = {
+            n[len(prefix) + 5 :]: v
+            for n, v in state_dict.items()
+            if n.startswith(f"{prefix}itep.")
+        }
+        if itep_sd:
+            self._itep.load_state_dict(itep_sd, strict=False)
+        return self._ebc._load_from_state_dict(
+            state_dict, prefix, local_metadata, strict, missing_keys,
+            unexpected_keys, error_msgs,
+        )
 
 
 class ITEPEmbeddingBagCollectionSharder(ModuleSharder[ITEPEmbeddingBagCollection]):
