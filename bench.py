@@ -155,11 +155,18 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
-    torch.cuda.set_device(local_rank)
-    enable_tuned_gemms()
-    device = torch.device("cuda", local_rank)
+    # TREC_BENCH_CPU=1: rehearse the exact multi-rank torchrun path (planner,
+    # mixed shardings, per-sharding communicators, JSON aggregation) on
+    # gloo/CPU — the driver's real runs use cuda+RCCL
+    cpu_mode = os.environ.get("TREC_BENCH_CPU") == "1"
+    if cpu_mode:
+        device = torch.device("cpu")
+    else:
+        torch.cuda.set_device(local_rank)
+        enable_tuned_gemms()
+        device = torch.device("cuda", local_rank)
     if world > 1:
-        dist.init_process_group("nccl")
+        dist.init_process_group("gloo" if cpu_mode else "nccl")
         env = ShardingEnv.from_process_group(dist.group.WORLD)
         pg = dist.group.WORLD
     else:
@@ -183,14 +190,20 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
         )
     sharder = EmbeddingBagCollectionSharder(fused_params=fused_params)
     planner = EmbeddingShardingPlanner(
-        topology=Topology(world_size=world, compute_device="cuda", batch_size=batch_size)
+        topology=Topology(
+            world_size=world,
+            compute_device="cpu" if cpu_mode else "cuda",
+            batch_size=batch_size,
+            hbm_cap=(1 << 42) if cpu_mode else None,
+        )
     )
     plan = planner.collective_plan(model, [sharder], pg)
     dmp = DistributedModelParallel(
         model, env=env, plan=plan, sharders=[sharder], device=device,
         init_data_parallel=False,
     )
-    _dense_to_bf16(dmp)  # before DDP wrap: buckets must see the bf16 params
+    if not cpu_mode:
+        _dense_to_bf16(dmp)  # before DDP wrap: buckets must see the bf16 params
     if world > 1:
         dmp.init_data_parallel()
     if rank == 0:
@@ -227,21 +240,26 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
 
         pipeline_cls = TrainPipelineFusedSparseDist
     pipeline = pipeline_cls(
-        dmp, dense_opt, device, autocast_dtype=torch.bfloat16
+        dmp, dense_opt, device,
+        autocast_dtype=None if cpu_mode else torch.bfloat16,
     )
 
-    batches = make_host_batches(8, batch_size, scale, seed=1234 + rank, pin=True)
+    batches = make_host_batches(
+        8, batch_size, scale, seed=1234 + rank, pin=not cpu_mode
+    )
     it = _CyclingIterator(batches)
 
     for _ in range(warmup):
         pipeline.progress(it)
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    if not cpu_mode:
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(steps):
         pipeline.progress(it)
-    torch.cuda.synchronize()
+    if not cpu_mode:
+        torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
     dt = time.perf_counter() - t0
