@@ -296,24 +296,21 @@ class ShardedEmbeddingCollection(nn.Module):
             if st != ShardingType.DATA_PARALLEL.value and W > 1
         )
         if env.process_group is not None and n_comm > 1:
-            import torch.distributed as dist_mod
+            from torchrec_amd.distributed.embeddingbag import _sharding_out_pg
 
-            groups = getattr(env, "all_group_ranks", None) or [
-                dist_mod.get_process_group_ranks(env.process_group)
-            ]
-            my_ranks = dist_mod.get_process_group_ranks(env.process_group)
-            backend = dist_mod.get_backend(env.process_group)
             first = True
+            extra = 0
             for si, st in enumerate(self._sharding_types):
                 if st == ShardingType.DATA_PARALLEL.value or W <= 1:
                     continue
                 if first:
                     first = False  # first sharding keeps the shared pg
                     continue
-                for ranks in groups:
-                    pg = dist_mod.new_group(ranks=ranks, backend=backend)
-                    if ranks == my_ranks:
-                        self._seq_a2a[si] = SequenceEmbeddingsAllToAll(pg)
+                # offset the cache index so EC groups never collide with the
+                # pooled-EBC groups of the same module tree
+                pg = _sharding_out_pg(env, 1000 + extra)
+                extra += 1
+                self._seq_a2a[si] = SequenceEmbeddingsAllToAll(pg)
 
         self._fused_optimizer = _ECFusedOptimizer(self)
 

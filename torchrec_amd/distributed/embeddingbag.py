@@ -40,6 +40,31 @@ from torchrec_amd.optim.keyed import FusedOptimizer
 from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor, KeyedTensor
 
 
+_OUT_PG_CACHE: Dict[Any, Any] = {}
+
+
+def _sharding_out_pg(env: ShardingEnv, index: int):
+    """Cached per-sharding output communicator (see EmbeddingSharding.out_pg).
+
+    new_group is collective over the DEFAULT group with identical arguments
+    everywhere, so every rank loops over EVERY sharding group's rank list
+    (2D sets env.all_group_ranks). The cache keys on (member ranks, index):
+    repeated sharding (resharding, multiple sharded modules) REUSES the
+    communicators instead of growing an unbounded RCCL comm set."""
+    import torch.distributed as dist_mod
+
+    my_ranks = tuple(dist_mod.get_process_group_ranks(env.process_group))
+    key = (my_ranks, index)
+    if key in _OUT_PG_CACHE:
+        return _OUT_PG_CACHE[key]
+    groups = env.all_group_ranks or [list(my_ranks)]
+    backend = dist_mod.get_backend(env.process_group)
+    for ranks in groups:
+        pg = dist_mod.new_group(ranks=ranks, backend=backend)
+        _OUT_PG_CACHE[(tuple(ranks), index)] = pg
+    return _OUT_PG_CACHE[key]
+
+
 def create_sharding(
     sharding_type: str,
     infos: List[EmbeddingShardingInfo],
@@ -230,21 +255,8 @@ class ShardedEmbeddingBagCollection(nn.Module):
         # mixed sharding types: one communicator per sharding so their
         # backward collectives never need a cross-rank issue order
         if env.process_group is not None and len(self._shardings) > 1:
-            import torch.distributed as dist_mod
-
-            # new_group is collective over the DEFAULT group with identical
-            # arguments everywhere: loop over EVERY sharding group's rank
-            # list (2D sets env.all_group_ranks; 1D = just this group)
-            groups = env.all_group_ranks or [
-                dist_mod.get_process_group_ranks(env.process_group)
-            ]
-            my_ranks = dist_mod.get_process_group_ranks(env.process_group)
-            backend = dist_mod.get_backend(env.process_group)
-            for sh in self._shardings[1:]:
-                for ranks in groups:
-                    pg = dist_mod.new_group(ranks=ranks, backend=backend)
-                    if ranks == my_ranks:
-                        sh._pg_out = pg
+            for i, sh in enumerate(self._shardings[1:]):
+                sh._pg_out = _sharding_out_pg(env, i)
 
         self._input_dists = nn.ModuleList(
             [s.create_input_dist(self._device) for s in self._shardings]
