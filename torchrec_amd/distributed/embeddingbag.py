@@ -50,7 +50,13 @@ def _sharding_out_pg(env: ShardingEnv, index: int):
     everywhere, so every rank loops over EVERY sharding group's rank list
     (2D sets env.all_group_ranks). The cache keys on (member ranks, index):
     repeated sharding (resharding, multiple sharded modules) REUSES the
-    communicators instead of growing an unbounded RCCL comm set."""
+    communicators instead of growing an unbounded RCCL comm set.
+
+    KNOWN EDGE: two sharded modules reuse the same per-index groups, so a
+    model with MULTIPLE multi-sharding-type sparse modules still shares
+    communicators across modules (as all single-sharding modules share the
+    default group); module-granular backward ordering is deterministic for
+    the supported pipelines."""
     import torch.distributed as dist_mod
 
     my_ranks = tuple(dist_mod.get_process_group_ranks(env.process_group))
