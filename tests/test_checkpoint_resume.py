@@ -86,3 +86,7 @@ def test_checkpoint_resume_tw():
 
 def test_checkpoint_resume_rw():
     run_multi_process(_run_resume, 2, "gloo", ShardingType.ROW_WISE.value)
+
+
+def test_checkpoint_resume_cw():
+    run_multi_process(_run_resume, 2, "gloo", ShardingType.COLUMN_WISE.value)
