@@ -672,3 +672,30 @@ def test_precision_wide_dims_gpu():
     torch.cuda.synchronize()
     for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
         torch.testing.assert_close(wg.cpu().float(), wc.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_vbe_with_bf16_tables_cpu():
+    """VBE forward over bf16 storage (combined feature coverage)."""
+    torch.manual_seed(0)
+    tbe = TableBatchedEmbeddingBags([("t0", 20, 8)], weights_precision="bf16")
+    indices = torch.tensor([1, 2, 3])
+    offsets = torch.tensor([0, 2, 3])
+    out = tbe.forward_vbe(indices, offsets, [2])
+    w = tbe.split_embedding_weights()[0].float()
+    exp = torch.cat([w[1] + w[2], w[3]])
+    torch.testing.assert_close(out, exp, atol=1e-5, rtol=1e-5)
+    out.sum().backward()
+
+
+def test_seg_sort_gate_respects_dedup_and_cache():
+    """seg-sort gating composes with other TBE modes without breaking them."""
+    from torchrec_amd.ops.tbe import EmbeddingLocation
+
+    # caching + fixed_bag_length: constructor must not fight the cache assert
+    tbe = TableBatchedEmbeddingBags(
+        [("t0", 64, 8)], fixed_bag_length=1,
+        location=EmbeddingLocation.MANAGED_CACHING,  # CPU device: caching off
+    )
+    assert tbe._seg_sort_ok
+    out = tbe(torch.tensor([1, 2]), torch.tensor([0, 1, 2]))
+    out.sum().backward()
