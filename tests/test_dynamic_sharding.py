@@ -71,3 +71,54 @@ def _run_reshard_test(rank, world_size):
 
 def test_dynamic_resharding_tw():
     run_multi_process(_run_reshard_test, 2, "gloo")
+
+
+def _run_cw_reshard(rank, world_size):
+    """CW column shards move between ranks; values survive the move."""
+    from copy import deepcopy
+
+    from torchrec_amd.distributed.types import (
+        EmbeddingModuleShardingPlan,
+        ShardingType,
+    )
+
+    B = 4
+    tables = make_tables()
+    torch.manual_seed(42)
+    model = SparseModel(make_tables())
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(
+                sharding_types=[ShardingType.COLUMN_WISE.value], min_partition=4
+            )
+            for cfg in tables
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    kjt_global = make_global_kjt(tables, B * world_size)
+    kjt_local = kjt_local_slice(kjt_global, rank * B, (rank + 1) * B)
+    before = dmp(kjt_local).values().detach().clone()
+
+    # swap every CW shard's placement to the other rank
+    old_plan = dmp.plan.get_plan_for_module("sparse")
+    new_plan_dict = {}
+    for name, ps in old_plan.items():
+        ps2 = deepcopy(ps)
+        for md in ps2.sharding_spec or []:
+            md.placement_rank = (md.placement_rank + 1) % world_size
+        ps2.ranks = [(r + 1) % world_size for r in (ps.ranks or [])]
+        new_plan_dict[name] = ps2
+    dmp.reshard("sparse", EmbeddingModuleShardingPlan(plan=new_plan_dict))
+    after = dmp(kjt_local).values()
+    torch.testing.assert_close(after, before, atol=1e-6, rtol=1e-6)
+
+
+def test_cw_dynamic_resharding():
+    run_multi_process(_run_cw_reshard, 2, "gloo")
