@@ -11,6 +11,11 @@ Single process (CPU or one GPU) also works directly.
 
 import os
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 import torch.distributed as dist
 
