@@ -95,6 +95,7 @@ def main(num_steps: int = 100, batch_size: int = 256) -> None:
     )
 
     dataset = RandomRecDataset(
+        learnable_labels=True,  # label = f(ids): the loop can actually converge
         keys=keys, batch_size=batch_size, hash_sizes=hash_sizes,
         ids_per_feature=10, num_dense=13, seed=100 + rank,
     )

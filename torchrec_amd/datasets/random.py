@@ -109,7 +109,9 @@ class RandomRecDataset(IterableDataset):
         pooling_avg: Optional[int] = None,
         num_batches: Optional[int] = None,
         seed: int = 0,
+        learnable_labels: bool = False,
     ) -> None:
+        self.learnable_labels = learnable_labels
         super().__init__()
         self.keys = keys
         self.batch_size = batch_size
@@ -132,5 +134,6 @@ class RandomRecDataset(IterableDataset):
                 num_dense=self.num_dense,
                 pooling_avg=self.pooling_avg,
                 generator=gen,
+                learnable_labels=self.learnable_labels,
             )
             i += 1
