@@ -49,7 +49,7 @@ def main(num_steps: int = 100, batch_size: int = 256) -> None:
         torch.cuda.set_device(device)
 
     num_features = 8
-    hash_sizes = [100_000] * num_features
+    hash_sizes = [2_000] * num_features  # small tables: ids repeat, the demo converges in ~100 steps
     keys = [f"cat_{i}" for i in range(num_features)]
     tables = [
         EmbeddingBagConfig(
@@ -97,7 +97,7 @@ def main(num_steps: int = 100, batch_size: int = 256) -> None:
     dataset = RandomRecDataset(
         learnable_labels=True,  # label = f(ids): the loop can actually converge
         keys=keys, batch_size=batch_size, hash_sizes=hash_sizes,
-        ids_per_feature=10, num_dense=13, seed=100 + rank,
+        ids_per_feature=1, num_dense=13, seed=100 + rank,
     )
     it = iter(dataset)
     for step in range(num_steps):
