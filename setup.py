@@ -59,9 +59,17 @@ de_ext = CppExtension(
     extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
 )
 
+from setuptools import find_packages
+
 setup(
-    name="torchrec_amd_ext",
+    name="torchrec_amd",
     version="0.1.0",
+    description=(
+        "MI355X-native sparse/recommender-systems training framework "
+        "(TorchRec capabilities; CDNA4 HIP kernels; RCCL over xGMI)"
+    ),
+    packages=find_packages(include=["torchrec_amd", "torchrec_amd.*"]),
+    python_requires=">=3.10",
     ext_modules=[ext, infer_ext, de_ext],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=True)},
 )
