@@ -70,6 +70,9 @@ def shard_quant_model(
     from torchrec_amd.distributed.model_parallel import DistributedModelParallel
     from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
     from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.quant_embedding import (
+        QuantEmbeddingCollectionSharder,
+    )
     from torchrec_amd.distributed.quant_embeddingbag import (
         QuantEmbeddingBagCollectionSharder,
     )
@@ -77,7 +80,7 @@ def shard_quant_model(
 
     assert dist.is_initialized(), "multi-rank quant sharding needs torch.distributed"
     env = ShardingEnv.from_process_group(dist.group.WORLD)
-    sharder = QuantEmbeddingBagCollectionSharder()
+    sharders = [QuantEmbeddingBagCollectionSharder(), QuantEmbeddingCollectionSharder()]
     planner = EmbeddingShardingPlanner(
         topology=Topology(
             world_size=world_size,
@@ -86,9 +89,9 @@ def shard_quant_model(
         ),
         constraints=constraints,
     )
-    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    plan = planner.collective_plan(model, sharders, dist.group.WORLD)
     dmp = DistributedModelParallel(
-        model, env=env, plan=plan, sharders=[sharder], device=device,
+        model, env=env, plan=plan, sharders=sharders, device=device,
         init_data_parallel=False,
     )
     return dmp, plan
