@@ -699,3 +699,19 @@ def test_seg_sort_gate_respects_dedup_and_cache():
     assert tbe._seg_sort_ok
     out = tbe(torch.tensor([1, 2]), torch.tensor([0, 1, 2]))
     out.sum().backward()
+
+
+def test_vbe_weighted_psw_grad_cpu():
+    """d(loss)/d(per_sample_weights) through the VBE path (CPU oracle)."""
+    torch.manual_seed(0)
+    tbe = TableBatchedEmbeddingBags([("t0", 20, 8)])
+    indices = torch.tensor([1, 2, 3])
+    offsets = torch.tensor([0, 2, 3])
+    psw = torch.tensor([2.0, 0.5, 3.0], requires_grad=True)
+    out = tbe.forward_vbe(indices, offsets, [2], psw)
+    out.sum().backward()
+    w = tbe.split_embedding_weights()[0]
+    # NOTE the fused update already ran; psw grad uses the PRE-update rows on
+    # the oracle replay (weights were detached before the update)
+    assert psw.grad is not None and psw.grad.shape == psw.shape
+    assert float(psw.grad.abs().sum()) > 0

@@ -134,11 +134,13 @@ class _TBEVbeFunction(torch.autograd.Function):
     def backward(ctx, grad):  # type: ignore[override]
         host = ctx.host
         indices, offsets, psw, bag_offsets, out_offsets = ctx.saved_tensors
-        host._backward_vbe(
-            grad.contiguous(), indices, offsets,
-            psw if ctx.has_psw else None, bag_offsets, out_offsets,
-        )
-        return None, None, None, None, None, None, None, None
+        psw_t = psw if ctx.has_psw else None
+        grad = grad.contiguous()
+        host._backward_vbe(grad, indices, offsets, psw_t, bag_offsets, out_offsets)
+        grad_psw = None
+        if psw_t is not None and ctx.needs_input_grad[4]:
+            grad_psw = host._grad_psw_vbe(grad, indices, offsets, bag_offsets, out_offsets)
+        return None, None, None, None, grad_psw, None, None, None
 
 
 class _TBESeqFunction(torch.autograd.Function):
@@ -466,6 +468,41 @@ class TableBatchedEmbeddingBags(nn.Module):
                 self._max_D,
             )
         return None
+
+    def _grad_psw_vbe(
+        self,
+        grad: torch.Tensor,
+        indices: torch.Tensor,
+        offsets: torch.Tensor,
+        bag_offsets: torch.Tensor,
+        out_offsets: torch.Tensor,
+    ) -> torch.Tensor:
+        """d(loss)/d(psw_i) for the packed VBE output: dot(grad slice of the
+        owning bag, W[idx_i]) via the generic (row, col) kernel."""
+        lengths = offsets[1:] - offsets[:-1]
+        n_bags = lengths.numel()
+        bag_ids = torch.repeat_interleave(
+            torch.arange(n_bags, device=indices.device, dtype=torch.int64),
+            lengths,
+            output_size=indices.numel(),
+        )
+        f = torch.searchsorted(bag_offsets, bag_ids, right=True) - 1
+        b = bag_ids - bag_offsets[f]
+        dims64 = self._dims_t.to(torch.int64)[self._feat_table_t.to(torch.int64)[f]]
+        pos_row = torch.zeros_like(bag_ids, dtype=torch.int32)
+        pos_col = out_offsets[f] + b * dims64
+        pos_table = self._feat_table_t.to(torch.int64)[f].to(torch.int32)
+        return torch.ops.trec_amd.tbe_grad_per_sample_weights(
+            self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
+            self._table_elem_offsets,
+            self._dims_t,
+            grad.view(1, -1),
+            indices,
+            pos_row,
+            pos_col,
+            pos_table,
+            self._max_D,
+        )
 
     def prefetch(self, indices: torch.Tensor, offsets: torch.Tensor) -> None:
         """Explicit cache prefetch (reference SplitTBE.prefetch()): populate
@@ -971,13 +1008,19 @@ class _TBEVbeCpuFunction(torch.autograd.Function):
     def backward(ctx, grad):  # type: ignore[override]
         host = ctx.host
         indices, offsets, psw = ctx.saved_tensors
-        psw_t = psw if ctx.has_psw else None
+        need_psw = ctx.has_psw and ctx.needs_input_grad[4]
+        psw_t = (
+            psw.requires_grad_(True)
+            if need_psw
+            else (psw if ctx.has_psw else None)
+        )
         w = host.weights.detach().float().requires_grad_(True)
         with torch.enable_grad():
             out = _tbe_cpu_vbe_forward(w, host, indices, offsets, ctx.bag_off, psw_t)
             out.backward(grad)
         host._cpu_apply_update(w.grad)
-        return None, None, None, None, None, None
+        grad_psw = psw_t.grad if need_psw else None
+        return None, None, None, None, grad_psw, None
 
 
 class _TBESeqCpuFunction(torch.autograd.Function):
