@@ -136,10 +136,11 @@ class _TBEVbeFunction(torch.autograd.Function):
         indices, offsets, psw, bag_offsets, out_offsets = ctx.saved_tensors
         psw_t = psw if ctx.has_psw else None
         grad = grad.contiguous()
-        host._backward_vbe(grad, indices, offsets, psw_t, bag_offsets, out_offsets)
         grad_psw = None
         if psw_t is not None and ctx.needs_input_grad[4]:
+            # read rows BEFORE the fused update rewrites them
             grad_psw = host._grad_psw_vbe(grad, indices, offsets, bag_offsets, out_offsets)
+        host._backward_vbe(grad, indices, offsets, psw_t, bag_offsets, out_offsets)
         return None, None, None, None, grad_psw, None, None, None
 
 
@@ -416,6 +417,22 @@ class TableBatchedEmbeddingBags(nn.Module):
             assert psw is None, "mean pooling with per-sample weights unsupported"
         elif psw is not None:
             scale = psw
+        grad_psw_out = None
+        if psw is not None and psw.requires_grad:
+            # read rows BEFORE the fused update rewrites them
+            fpsw = torch.div(bag_ids, B, rounding_mode="floor")
+            pos_table = self._feat_table_t.to(torch.int64)[fpsw].to(torch.int32)
+            grad_psw_out = torch.ops.trec_amd.tbe_grad_per_sample_weights(
+                self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
+                self._table_elem_offsets,
+                self._dims_t,
+                grad,
+                indices,
+                pos_row,
+                pos_col,
+                pos_table,
+                self._max_D,
+            )
         cap = (self.fixed_bag_length or 0) * B
         if self._seg_sort_ok and 0 < cap <= 16384:
             sorted_lin, perm, _overflow = torch.ops.trec_amd.seg_sort_pairs(
@@ -453,21 +470,7 @@ class TableBatchedEmbeddingBags(nn.Module):
             self.cache_weights,
             cache_loc if cache_loc is not None else self._empty_i,
         )
-        if psw is not None and psw.requires_grad:
-            f = torch.div(bag_ids, B, rounding_mode="floor")
-            pos_table = self._feat_table_t.to(torch.int64)[f].to(torch.int32)
-            return torch.ops.trec_amd.tbe_grad_per_sample_weights(
-                self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
-                self._table_elem_offsets,
-                self._dims_t,
-                grad,
-                indices,
-                pos_row,
-                pos_col,
-                pos_table,
-                self._max_D,
-            )
-        return None
+        return grad_psw_out
 
     def _grad_psw_vbe(
         self,
