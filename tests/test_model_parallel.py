@@ -656,3 +656,25 @@ def test_mx4_codec_roundtrip():
     assert enc.numel() == c.encoded_numel(256) == (256 // 32) * 17
     dec = c.decode(enc, 256)
     assert float((dec - x).abs().max() / x.abs().max()) < 0.12
+
+
+def test_qcomm_codec_roundtrips():
+    """Every wire precision encodes/decodes within its format's error."""
+    from torchrec_amd.distributed.qcomm_codecs import CommType, QuantizedCommCodec
+
+    torch.manual_seed(0)
+    x = torch.randn(512) * 3
+    tolerances = {
+        CommType.FP16: 2e-3,
+        CommType.BF16: 2e-2,
+        CommType.FP8: 8e-2,
+        CommType.INT8: 3e-2,
+        CommType.MX4: 1.5e-1,
+    }
+    for ct, tol in tolerances.items():
+        c = QuantizedCommCodec(ct)
+        enc = c.encode(x)
+        assert enc.numel() == c.encoded_numel(x.numel()), ct
+        dec = c.decode(enc, x.numel())
+        rel = float((dec - x).abs().max() / x.abs().max())
+        assert rel < tol, (ct, rel)
