@@ -246,7 +246,13 @@ class EmbeddingEnumerator:
                 mem_bw = 1.0 / (CACHE_HIT / topo.hbm_mem_bw + (1 - CACHE_HIT) / topo.ddr_mem_bw)
             else:
                 mem_bw = topo.hbm_mem_bw
-            fwd_compute = fwd_bytes / mem_bw
+            # narrow-shard penalty (CW/TWCW/GRID): the TBE assigns 64/LPS
+            # bags per 64-lane wavefront with LPS lanes covering D/4 float4
+            # columns — shards narrower than 64 columns leave lanes idle
+            # (D=16 -> 4 of 16 lanes active), so their effective bandwidth
+            # shrinks proportionally
+            lane_eff = min(1.0, cols / 64.0) if topo.compute_device == "cuda" else 1.0
+            fwd_compute = fwd_bytes / (mem_bw * max(lane_eff, 1e-3))
             shard.perf = Perf(
                 fwd_compute=fwd_compute,
                 fwd_comms=comms,
