@@ -450,3 +450,15 @@ def test_fx_trace_dlrm_sparse_arch():
     )
     gm = symbolic_trace(model)
     assert any("sparse" in n.name or "ebc" in str(n.target) for n in gm.graph.nodes)
+
+
+def test_percentile_logger():
+    from torchrec_amd.utils.percentile_logger import PercentileLogger
+
+    pl = PercentileLogger(name="step_ms")
+    for v in range(1, 101):
+        pl.add(float(v))
+    stats = pl.summary()
+    assert abs(stats["p50"] - 50) <= 1.5
+    assert abs(stats["p99"] - 99) <= 1.5
+    assert stats["count"] == 100
