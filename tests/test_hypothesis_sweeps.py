@@ -182,3 +182,36 @@ def test_vbe_module_sweep(strides, dim_i):
     out = tbe.forward_vbe(indices, offsets, list(strides))
     assert out.numel() == sum(s * D for s in strides)
     out.sum().backward() if out.numel() else None
+
+
+@given(
+    spk=st.lists(
+        st.lists(st.integers(0, 4), min_size=1, max_size=3), min_size=1, max_size=5
+    ),
+    seed=st.integers(0, 10_000),
+)
+@settings(max_examples=40, deadline=None)
+def test_vbe_kjt_invariants(spk, seed):
+    """VBE KJT: split/permute/to_dict preserve values and key structure."""
+    import random
+
+    random.seed(seed)
+    K = len(spk)
+    strides = [sum(s) for s in spk]
+    lengths = torch.tensor([random.randint(0, 3) for _ in range(sum(strides))])
+    vals = torch.arange(int(lengths.sum()))
+    kjt = KeyedJaggedTensor(
+        keys=[f"k{i}" for i in range(K)], values=vals, lengths=lengths,
+        stride_per_key_per_rank=spk,
+    )
+    parts = kjt.split([1] * K)
+    recon = torch.cat([p.values() for p in parts]) if K else vals
+    assert torch.equal(recon, vals)
+    order = list(range(K))
+    random.shuffle(order)
+    inv = [order.index(i) for i in range(K)]
+    p2 = kjt.permute(order).permute(inv)
+    assert torch.equal(p2.values(), vals)
+    assert p2.keys() == kjt.keys()
+    d = kjt.to_dict()
+    assert sum(v.values().numel() for v in d.values()) == vals.numel()
