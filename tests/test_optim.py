@@ -73,3 +73,30 @@ def test_warmup_schedule():
     assert abs(lrs[0] - 0.1) < 1e-6
     assert lrs[0] <= lrs[1] <= lrs[2] <= lrs[3] <= 1.0 + 1e-6
     assert abs(lrs[-1] - 0.5) < 1e-6
+
+
+def _run_dist_norm_clip(rank, world_size):
+    import torch.distributed as dist
+
+    w = torch.nn.Parameter(torch.zeros(4))
+    opt = GradientClippingOptimizer(
+        _keyed_sgd(w), clipping=GradientClipping.NORM, max_gradient=1.0,
+        process_group=dist.group.WORLD,
+    )
+    # each rank holds HALF the global gradient mass: per-rank norm 10,
+    # global norm sqrt(2)*10 — the sharded clip must use the GLOBAL norm
+    w.grad = torch.full((4,), 5.0)
+    opt.step()
+    import math
+
+    global_norm = math.sqrt(world_size * (4 * 25.0))
+    coef = 1.0 / global_norm
+    torch.testing.assert_close(
+        w.detach(), torch.full((4,), -5.0 * coef), atol=1e-4, rtol=1e-4
+    )
+
+
+def test_distributed_norm_clipping():
+    from tests.dist_utils import run_multi_process
+
+    run_multi_process(_run_dist_norm_clip, 2, "gloo")
