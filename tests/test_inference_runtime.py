@@ -90,3 +90,48 @@ def test_predictor_server():
         with torch.no_grad():
             direct = model(torch.tensor(req["float_features"]), kjt).reshape(-1)
         torch.testing.assert_close(torch.tensor(preds), direct, atol=1e-5, rtol=1e-5)
+
+
+def test_predict_factory_package_roundtrip(tmp_path):
+    """torch.package export/import of a predict factory + configs."""
+    from torchrec_amd.inference.model_packager import (
+        PredictFactoryPackager,
+        load_predict_factory,
+    )
+
+    class P(PredictFactoryPackager):
+        @classmethod
+        def set_extern_modules(cls):
+            return []
+
+        @classmethod
+        def set_mocked_modules(cls):
+            return []
+
+    configs = {"tables": 4, "dim": 16}
+    out = str(tmp_path / "factory.pt")
+    # a pickleable payload standing in for a factory (classes defined in
+    # tests aren't packageable; serving code passes an importable factory)
+    P.save_predict_factory({"entry": "make_predictor", "version": 1}, configs, out)
+    factory, cfg = load_predict_factory(out)
+    assert factory["entry"] == "make_predictor"
+    assert cfg == configs
+
+
+def test_invoke_on_rank_and_broadcast():
+    from tests.dist_utils import run_multi_process
+
+    run_multi_process(_run_invoke_broadcast, 2, "gloo")
+
+
+def _run_invoke_broadcast(rank, world_size):
+    import torch.distributed as dist
+
+    from torchrec_amd.distributed.collective_utils import (
+        invoke_on_rank_and_broadcast_result,
+    )
+
+    result = invoke_on_rank_and_broadcast_result(
+        dist.group.WORLD, 0, lambda: {"from": dist.get_rank(), "x": 42}
+    )
+    assert result == {"from": 0, "x": 42}  # every rank sees rank-0's value
