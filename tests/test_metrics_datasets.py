@@ -180,3 +180,21 @@ def test_cpu_offloaded_metric_module():
     for k in b:
         torch.testing.assert_close(a[k], b[k])
     off.shutdown()
+
+
+def test_criteo_shard_math():
+    from torchrec_amd.datasets.criteo import BinaryCriteoUtils
+
+    lengths = [100, 50, 75]  # three files, 225 rows
+    seen = []
+    for rank in range(4):
+        ranges, rem = BinaryCriteoUtils.get_file_row_ranges_and_remainder(
+            lengths, rank, 4
+        )
+        n = sum(hi - lo for lo, hi in ranges.values())
+        assert n == 225 // 4
+        for f, (lo, hi) in ranges.items():
+            assert 0 <= lo <= hi <= lengths[f]
+            seen.extend((f, r) for r in range(lo, hi))
+    # ranks cover disjoint rows
+    assert len(seen) == len(set(seen)) == (225 // 4) * 4
