@@ -81,7 +81,7 @@ def _run_dist_norm_clip(rank, world_size):
     w = torch.nn.Parameter(torch.zeros(4))
     opt = GradientClippingOptimizer(
         _keyed_sgd(w), clipping=GradientClipping.NORM, max_gradient=1.0,
-        process_group=dist.group.WORLD,
+        process_group=dist.group.WORLD, sharded_params={w},
     )
     # each rank holds HALF the global gradient mass: per-rank norm 10,
     # global norm sqrt(2)*10 — the sharded clip must use the GLOBAL norm
@@ -94,6 +94,43 @@ def _run_dist_norm_clip(rank, world_size):
     torch.testing.assert_close(
         w.detach(), torch.full((4,), -5.0 * coef), atol=1e-4, rtol=1e-4
     )
+
+
+def _run_dist_norm_clip_mixed(rank, world_size):
+    import math
+
+    import torch.distributed as dist
+
+    # sharded param: each rank a distinct shard; replicated param: identical
+    # grads on all ranks and must be counted ONCE in the global norm
+    ws = torch.nn.Parameter(torch.zeros(4))
+    wr = torch.nn.Parameter(torch.zeros(4))
+    from torchrec_amd.optim.keyed import KeyedOptimizerWrapper
+
+    inner = KeyedOptimizerWrapper(
+        {"s": ws, "r": wr}, lambda params: torch.optim.SGD(params, lr=1.0)
+    )
+    opt = GradientClippingOptimizer(
+        inner, clipping=GradientClipping.NORM, max_gradient=1.0,
+        process_group=dist.group.WORLD, sharded_params={ws},
+    )
+    ws.grad = torch.full((4,), 3.0)
+    wr.grad = torch.full((4,), 4.0)
+    opt.step()
+    global_norm = math.sqrt(world_size * 4 * 9.0 + 4 * 16.0)
+    coef = 1.0 / global_norm
+    torch.testing.assert_close(
+        ws.detach(), torch.full((4,), -3.0 * coef), atol=1e-4, rtol=1e-4
+    )
+    torch.testing.assert_close(
+        wr.detach(), torch.full((4,), -4.0 * coef), atol=1e-4, rtol=1e-4
+    )
+
+
+def test_distributed_norm_clipping_mixed_replicated():
+    from tests.dist_utils import run_multi_process
+
+    run_multi_process(_run_dist_norm_clip_mixed, 2, "gloo")
 
 
 def test_distributed_norm_clipping():

@@ -242,6 +242,7 @@ class ShardedEmbeddingBagCollection(nn.Module):
         self._device = device or torch.device("cpu")
         self._is_weighted = module.is_weighted()
         fused_params = dict(fused_params or {})
+        self._fused_params = fused_params
 
         # group tables by sharding type (stable order)
         by_type: Dict[str, List[EmbeddingShardingInfo]] = {}
@@ -611,7 +612,13 @@ class EmbeddingFusedOptimizer(FusedOptimizer):
         param_groups: List[Dict[str, Any]] = []
         pg = sharded_ebc._env.process_group
         device_type = sharded_ebc._device.type
-        lr = 0.01
+        self._tbes = sharded_ebc.tbes()
+        lr = float(
+            (sharded_ebc._fused_params or {}).get("learning_rate", 0.01)
+            if hasattr(sharded_ebc, "_fused_params")
+            else 0.01
+        )
+        self._pushed_lr = lr
         views = sharded_ebc._shard_views()
         by_table: Dict[str, List] = {}
         for (t, ro, co, full, w, m) in views:
@@ -619,50 +626,80 @@ class EmbeddingFusedOptimizer(FusedOptimizer):
         for t, shards in by_table.items():
             ps = sharded_ebc._plan_by_table.get(t)
             key = f"embedding_bags.{t}.weight"
-            (ro, co, full, w, m) = shards[0]
-            if m is None:
+            # every local shard of the table contributes its momentum (a rank
+            # may hold >1 shard under CW/TWCW when col-shards > ranks)
+            shards_m = [s for s in shards if s[4] is not None]
+            if not shards_m:
                 continue
-            params[key] = w
-            param_groups.append({"params": [w], "lr": lr})
-            single_col_shard = len(shards) == 1 and co == 0
-            if (
-                pg is not None
-                and ps is not None
-                and ps.sharding_type
-                in (ShardingType.TABLE_WISE.value, ShardingType.ROW_WISE.value)
-                and single_col_shard
-            ):
-                mom_ps = ParameterShardingView1D(ps)
+            (ro0, co0, full0, w0, m0) = shards_m[0]
+            params[key] = w0
+            param_groups.append({"params": [w0], "lr": lr})
+            if pg is not None and ps is not None and ps.sharding_spec:
+                # 1-D momentum ShardedTensor. Column shards concatenate row
+                # spaces (rowwise-state convention: global shape =
+                # rows * n_col_shards, col-shard c's rows at offset c*rows —
+                # reference batched_embedding_kernel.py:1218 rowwise metadata).
+                rows_total = full0[0]
+                col_offsets = sorted({md.shard_offsets[1] for md in ps.sharding_spec})
+                ci_of = {c: i for i, c in enumerate(col_offsets)}
+                mom_ps = ParameterShardingView1D(ps, rows_total, ci_of)
+                local = [
+                    (m, [ci_of[co] * rows_total + ro])
+                    for (ro, co, full, w, m) in shards_m
+                    if m.numel()
+                ]
                 st = build_sharded_tensor(
-                    [(m, [ro, 0])][: 1 if m.numel() else 0] or [],
-                    (full[0],),
+                    local,
+                    (rows_total * len(col_offsets),),
                     mom_ps,
                     pg,
                     device_type,
                 )
-                state[w] = {f"{t}.momentum1": st}
+                state[w0] = {f"{t}.momentum1": st}
             else:
-                state[w] = {f"{t}.momentum1": m}
+                mom_state = {f"{t}.momentum1": m0}
+                for i, (ro, co, full, w, m) in enumerate(shards_m[1:], start=1):
+                    mom_state[f"{t}.momentum1.shard{i}"] = m
+                state[w0] = mom_state
         super().__init__(params, state, param_groups)
+
+    def step(self, closure: Any = None) -> None:
+        # the update itself runs inside the TBE backward; step() only
+        # propagates LR-schedule changes (WarmupOptimizer mutates the shared
+        # param_group dicts) down to the fused kernels
+        if self.param_groups:
+            lr = self.param_groups[0].get("lr", self._pushed_lr)
+            if lr != self._pushed_lr:
+                for tbe in self._tbes:
+                    tbe.set_learning_rate(lr)
+                self._pushed_lr = lr
 
 
 class ParameterShardingView1D:
-    """Project a 2-D table plan to the 1-D momentum row space."""
+    """Project a 2-D table plan to the 1-D momentum row space.
 
-    def __init__(self, ps) -> None:
+    Column shards map into a concatenated row space: col-shard index ci's
+    rows live at [ci * rows_total, ci * rows_total + rows)."""
+
+    def __init__(self, ps, rows_total: Optional[int] = None,
+                 ci_of: Optional[Dict[int, int]] = None) -> None:
         self.sharding_type = ps.sharding_type
         self.compute_kernel = ps.compute_kernel
         self.ranks = ps.ranks
         from torchrec_amd.distributed.types import ShardMetadata
 
-        self.sharding_spec = [
-            ShardMetadata(
-                shard_offsets=[md.shard_offsets[0]],
-                shard_sizes=[md.shard_sizes[0]],
-                placement_rank=md.placement_rank,
+        self.sharding_spec = []
+        for md in ps.sharding_spec or []:
+            base = 0
+            if rows_total is not None and ci_of is not None and len(md.shard_offsets) > 1:
+                base = ci_of.get(md.shard_offsets[1], 0) * rows_total
+            self.sharding_spec.append(
+                ShardMetadata(
+                    shard_offsets=[base + md.shard_offsets[0]],
+                    shard_sizes=[md.shard_sizes[0]],
+                    placement_rank=md.placement_rank,
+                )
             )
-            for md in (ps.sharding_spec or [])
-        ]
 
 
 class EmbeddingBagCollectionSharder(ModuleSharder[EmbeddingBagCollection]):

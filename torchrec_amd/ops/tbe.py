@@ -13,6 +13,7 @@ numerics oracle and keeps CPU (gloo) tests runnable.
 from __future__ import annotations
 
 import math
+import os
 from enum import Enum, unique
 from typing import List, NamedTuple, Optional, Tuple
 
@@ -237,6 +238,15 @@ class TableBatchedEmbeddingBags(nn.Module):
 
         rows = [s.rows for s in self._specs]
         dims = [s.dim for s in self._specs]
+        for d in dims:
+            # the kernels move rows as float4 (16 B/lane); a dim that is not a
+            # multiple of 4 would read/write past the row end (silent
+            # corruption on GPU while the CPU oracle works)
+            if d % 4 != 0:
+                raise ValueError(
+                    f"TableBatchedEmbeddingBags: embedding_dim {d} is not a "
+                    "multiple of 4 (float4 row layout); pad the table dim"
+                )
         elem_offsets = [0]
         for s in self._specs:
             elem_offsets.append(elem_offsets[-1] + s.rows * s.dim)
@@ -434,11 +444,23 @@ class TableBatchedEmbeddingBags(nn.Module):
                 self._max_D,
             )
         cap = (self.fixed_bag_length or 0) * B
+        # fixed_bag_length is a caller promise; a lying caller would overflow
+        # the block-sort tile and silently truncate tail gradients. The
+        # aggregate host-side check is free (numel is host-known); the exact
+        # per-segment device flag costs a sync so it is debug-gated.
+        if cap > 0 and linear.numel() > cap * self._num_features:
+            cap = 0  # provably overflowing: fall through to the device sort
         if self._seg_sort_ok and 0 < cap <= 16384:
-            sorted_lin, perm, _overflow = torch.ops.trec_amd.seg_sort_pairs(
+            sorted_lin, perm, overflow = torch.ops.trec_amd.seg_sort_pairs(
                 linear, offsets, B, self._num_features,
                 _bits_needed(self._total_rows), cap,
             )
+            if os.environ.get("TREC_DEBUG") == "1" and bool(overflow.item()):
+                raise RuntimeError(
+                    "TBE segmented sort overflow: a bag exceeded "
+                    f"fixed_bag_length={self.fixed_bag_length}; tail gradients "
+                    "would be dropped. Remove fixed_bag_length or raise it."
+                )
         else:
             # NOTE: rocPRIM's device segmented sort (seg_sort_pairs_large)
             # was measured 7x SLOWER than hipCUB's device radix for few large

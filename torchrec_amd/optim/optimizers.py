@@ -71,6 +71,7 @@ class GradientClippingOptimizer(KeyedOptimizer):
         max_gradient: float = 0.1,
         norm_type: float = 2.0,
         process_group=None,
+        sharded_params: Optional[set] = None,
     ) -> None:
         super().__init__(optimizer.params, optimizer.state, optimizer.param_groups)
         self._optimizer = optimizer
@@ -78,6 +79,12 @@ class GradientClippingOptimizer(KeyedOptimizer):
         self._max_gradient = max_gradient
         self._norm_type = norm_type
         self._pg = process_group
+        # tensors whose grads are rank-local shards of a global tensor: only
+        # their partial norms are all-reduced. Replicated (DP/dense) params
+        # hold identical grads on every rank — counting them once locally
+        # gives the exact global norm (reference clipping.py handles the two
+        # classes separately).
+        self._sharded_params = sharded_params or set()
 
     def step(self, closure: Any = None) -> None:
         if self._clipping == GradientClipping.NORM:
@@ -91,10 +98,15 @@ class GradientClippingOptimizer(KeyedOptimizer):
                 if self._pg is not None:
                     import torch.distributed as dist
 
-                    total = torch.zeros(1, device=params[0].device)
-                    for p in params:
+                    device = params[0].device
+                    sharded = [p for p in params if p in self._sharded_params]
+                    replicated = [p for p in params if p not in self._sharded_params]
+                    total = torch.zeros(1, device=device)
+                    for p in sharded:
                         total += p.grad.norm(self._norm_type) ** self._norm_type
                     dist.all_reduce(total, group=self._pg)
+                    for p in replicated:
+                        total += p.grad.norm(self._norm_type) ** self._norm_type
                     total_norm = total.pow(1.0 / self._norm_type)
                     coef = (self._max_gradient / (float(total_norm) + 1e-6)).__float__()
                     if coef < 1.0:
