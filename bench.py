@@ -162,9 +162,12 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
     if cpu_mode:
         device = torch.device("cpu")
     else:
-        torch.cuda.set_device(local_rank)
+        # clamp: a world-2 rehearsal on a 1-GPU box runs both ranks on cuda:0
+        # (RCCL over loopback) — exercises the full a2a/RS/DDP paths on metal
+        dev_idx = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(dev_idx)
         enable_tuned_gemms()
-        device = torch.device("cuda", local_rank)
+        device = torch.device("cuda", dev_idx)
     if world > 1:
         dist.init_process_group("gloo" if cpu_mode else "nccl")
         env = ShardingEnv.from_process_group(dist.group.WORLD)

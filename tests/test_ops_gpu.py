@@ -145,6 +145,65 @@ def test_fused_interaction_bf16_inputs():
     assert sparse.grad.dtype == torch.float32
 
 
+class TestInteractionMFMA:
+    """MFMA bf16 interaction vs the fp32 eager oracle (BASELINE mandate:
+    MFMA for the dense interaction). Tolerances are the bf16-input regime:
+    products round to 8-bit mantissa, accumulate fp32."""
+
+    @staticmethod
+    def _oracle(dense, sparse):
+        F = sparse.shape[1]
+        combined = torch.cat([dense.unsqueeze(1), sparse], dim=1)
+        inter = torch.bmm(combined, combined.transpose(1, 2))
+        tri = torch.triu_indices(F + 1, F + 1, offset=1)
+        return torch.cat([dense, inter[:, tri[0], tri[1]]], dim=1)
+
+    @pytest.mark.parametrize("B,F,D", [(33, 26, 128), (16, 10, 64), (8, 31, 96)])
+    def test_matches_fp32_oracle(self, B, F, D):
+        torch.manual_seed(0)
+        dense = torch.randn(B, D, requires_grad=True)
+        sparse = torch.randn(B, F, D, requires_grad=True)
+        ref = self._oracle(dense, sparse)
+        g = torch.randn_like(ref)
+        ref.backward(g)
+
+        d_g = dense.detach().clone().cuda().requires_grad_(True)
+        s_g = sparse.detach().clone().cuda().requires_grad_(True)
+        out = ops._FusedInteractionMFMA.apply(d_g, s_g)
+        torch.cuda.synchronize()
+        scale = ref.detach().abs().max().item()
+        assert torch.allclose(out.cpu(), ref.detach(), atol=0.02 * scale, rtol=0.02)
+        out.backward(g.cuda())
+        torch.cuda.synchronize()
+        gscale = sparse.grad.abs().max().item()
+        assert torch.allclose(d_g.grad.cpu(), dense.grad, atol=0.02 * gscale, rtol=0.02)
+        assert torch.allclose(s_g.grad.cpu(), sparse.grad, atol=0.02 * gscale, rtol=0.02)
+
+    def test_bf16_io(self):
+        torch.manual_seed(1)
+        B, F, D = 64, 12, 128
+        dense = torch.randn(B, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+        sparse = torch.randn(B, F, D, dtype=torch.bfloat16, device="cuda", requires_grad=True)
+        out = ops.fused_interaction(dense, sparse)
+        assert out.dtype == torch.bfloat16
+        out.float().sum().backward()
+        torch.cuda.synchronize()
+        ref = self._oracle(dense.detach().float().cpu(), sparse.detach().float().cpu())
+        scale = ref.abs().max().item()
+        assert torch.allclose(out.detach().float().cpu(), ref, atol=0.03 * scale, rtol=0.03)
+        assert dense.grad is not None and sparse.grad.dtype == torch.bfloat16
+
+    def test_dispatch_uses_mfma_for_dlrm_shape(self):
+        # D=128, F1=27 (the flagship shape) must route to the MFMA kernel
+        dense = torch.randn(4, 128, device="cuda")
+        sparse = torch.randn(4, 26, 128, device="cuda")
+        assert ops._mfma_interaction_ok(dense, sparse)
+        # tiny dims fall back to the fp32 VALU kernel
+        assert not ops._mfma_interaction_ok(
+            torch.randn(4, 16, device="cuda"), torch.randn(4, 5, 16, device="cuda")
+        )
+
+
 @pytest.mark.gpu
 class TestColSum:
     @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16, torch.float16])

@@ -426,9 +426,48 @@ class _FusedInteraction(torch.autograd.Function):
         return dd.to(ddt), ds.to(dst)
 
 
+class _FusedInteractionMFMA(torch.autograd.Function):
+    """MFMA (matrix-core) variant: per-sample Z = T @ T^T / dT = G @ T run on
+    v_mfma_f32_16x16x32_bf16 tiles with fp32 accumulate
+    (csrc/interaction_mfma.hip). Inputs round to bf16 inside the kernel —
+    the standard autocast regime of the surrounding dense arch."""
+
+    @staticmethod
+    def forward(ctx, dense, sparse):  # type: ignore[override]
+        F1 = sparse.shape[1] + 1
+        pi, pj, pair_col = _pair_tables(F1, dense.device)
+        if dense.dtype != sparse.dtype:
+            dense = dense.to(sparse.dtype)
+        ctx.save_for_backward(dense, sparse, pair_col)
+        return torch.ops.trec_amd.interaction_mfma_forward(dense, sparse, pi, pj)
+
+    @staticmethod
+    def backward(ctx, grad_out):  # type: ignore[override]
+        d, s, pair_col = ctx.saved_tensors
+        dd, ds = torch.ops.trec_amd.interaction_mfma_backward(
+            grad_out.to(d.dtype).contiguous(), d, s, pair_col
+        )
+        return dd, ds
+
+
+def _mfma_interaction_ok(dense: torch.Tensor, sparse: torch.Tensor) -> bool:
+    D = dense.shape[1]
+    F1 = sparse.shape[1] + 1
+    return (
+        D % 32 == 0
+        and 64 <= D <= 256
+        and F1 <= 32
+        and os.environ.get("TREC_INTERACTION_FP32") != "1"
+    )
+
+
 def fused_interaction(dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
-    """GPU: fused kernel; CPU callers should use the eager path."""
+    """GPU: fused kernel (MFMA bf16 tiles where the shape allows, fp32 VALU
+    otherwise or under TREC_INTERACTION_FP32=1); CPU callers should use the
+    eager path."""
     hip_ops()
+    if _mfma_interaction_ok(dense, sparse):
+        return _FusedInteractionMFMA.apply(dense, sparse)
     return _FusedInteraction.apply(dense, sparse)
 
 

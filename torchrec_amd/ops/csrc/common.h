@@ -79,6 +79,21 @@ __device__ __forceinline__ float emb2float(float v) { return v; }
 __device__ __forceinline__ float emb2float(__half v) { return __half2float(v); }
 __device__ __forceinline__ float emb2float(__hip_bfloat16 v) { return __bfloat162float(v); }
 
+
+// host scalar -> device scalar for AT_DISPATCH'd kernels (requires ATen,
+// which every .hip TU in this tree includes via torch/extension.h)
+template <typename scalar_t>
+struct DevType { using type = scalar_t; };
+template <> struct DevType<c10::Half> { using type = __half; };
+template <> struct DevType<c10::BFloat16> { using type = __hip_bfloat16; };
+
+// fp32 -> scalar emb_t (tag-dispatched: the second arg selects the overload)
+__device__ __forceinline__ float float2emb(float v, float) { return v; }
+__device__ __forceinline__ __half float2emb(float v, __half) { return __float2half(v); }
+__device__ __forceinline__ __hip_bfloat16 float2emb(float v, __hip_bfloat16) {
+  return __float2bfloat16(v);
+}
+
 template <typename emb_t>
 struct Vec4;
 

@@ -39,11 +39,6 @@ namespace trec_amd {
 // MANAGED) — kernels read/write it over PCIe via the device-visible alias.
 // ---------------------------------------------------------------------------
 
-template <typename scalar_t>
-struct DevType { using type = scalar_t; };
-template <> struct DevType<at::Half> { using type = __half; };
-template <> struct DevType<at::BFloat16> { using type = __hip_bfloat16; };
-
 template <typename T>
 static T* uvm_ptr(const at::Tensor& t) {
   if (t.numel() == 0) return nullptr;
