@@ -168,7 +168,9 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
         torch.cuda.set_device(dev_idx)
         enable_tuned_gemms()
         device = torch.device("cuda", dev_idx)
-    if world > 1:
+    # TREC_FORCE_DIST=1: initialize the process group even at world 1 (the
+    # driver-launch rehearsal exercises RCCL init + dist code paths on metal)
+    if world > 1 or os.environ.get("TREC_FORCE_DIST") == "1":
         dist.init_process_group("gloo" if cpu_mode else "nccl")
         env = ShardingEnv.from_process_group(dist.group.WORLD)
         pg = dist.group.WORLD

@@ -31,6 +31,27 @@ def test_bench_json_contract():
     assert "global_batch" in res["config"]
 
 
+def test_bench_world2_gloo_cpu():
+    """Rehearse the driver's multi-rank launch end-to-end on gloo/CPU:
+    torch.distributed.run -> bench.py world 2 -> planner + DMP + pipeline +
+    collectives + MAX-over-ranks JSON aggregation."""
+    import os
+
+    env = dict(os.environ, TREC_BENCH_CPU="1", MASTER_ADDR="127.0.0.1")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr=127.0.0.1",
+         "--master-port=29621", "bench.py", "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--batch-size", "64", "--scale", "1e-4"],
+        capture_output=True, text=True, timeout=600, env=env,
+    )
+    assert out.returncode == 0, (out.stdout + out.stderr)[-3000:]
+    line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["n_gpus"] == 2
+    assert res["config"]["global_batch"] == 128
+
+
 def test_bench_plan_world8_cpu():
     """The 8-GPU plan the driver's SCALE run will request must be buildable
     (no OOM/partition errors) — planner is pure CPU."""

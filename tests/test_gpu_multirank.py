@@ -54,6 +54,23 @@ def _duplicate_gpu_refusal(output: str) -> bool:
 
 
 @pytest.mark.skipif(torch.cuda.is_available() is False, reason="needs GPU")
+def test_world1_torchrun_rccl_bench():
+    """The driver's exact launch shape (torch.distributed.run + RCCL init)
+    at world 1 — must pass on any box. Validates init_process_group("nccl"),
+    ShardingEnv.from_process_group, and the distributed bench flow on metal."""
+    proc = _run_torchrun(
+        1,
+        ["--gpus", "1", "--steps", "3", "--warmup", "1",
+         "--batch-size", "256", "--scale", "1e-4"],
+        extra_env={"TREC_FORCE_DIST": "1"},
+    )
+    out = proc.stdout + "\n" + proc.stderr
+    assert proc.returncode == 0, f"world-1 torchrun bench failed:\n{out[-4000:]}"
+    res = _parse_result(proc.stdout)
+    assert res is not None and res["n_gpus"] == 1
+
+
+@pytest.mark.skipif(torch.cuda.is_available() is False, reason="needs GPU")
 def test_world2_rccl_bench_on_one_gpu():
     """2 RCCL ranks sharing one MI355X step the full DMP+pipeline bench."""
     proc = _run_torchrun(
