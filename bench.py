@@ -179,6 +179,10 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
         pg = None
 
     emb_precision = os.environ.get("TREC_EMB_PRECISION", "fp32")
+    # pooled output in bf16 by default: halves the TBE-output/a2a/interaction
+    # traffic and routes the interaction onto its bf16 MFMA path; the pool
+    # still accumulates fp32 in-kernel (TREC_EMB_OUT=fp32 to disable)
+    emb_out = os.environ.get("TREC_EMB_OUT", "bf16" if not cpu_mode else "fp32")
     model = build_model(scale, emb_precision)
     fused_params = {
         "optimizer": "rowwise_adagrad",
@@ -186,6 +190,7 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
         # Criteo categoricals are one-hot: enables the single-launch
         # segmented backward sort
         "fixed_bag_length": IDS_PER_FEATURE,
+        "output_dtype": emb_out,
     }
     if qcomm != "none":
         from torchrec_amd.distributed.qcomm_codecs import CommType, QCommsConfig
@@ -302,6 +307,7 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
             "embedding_dim": EMB_DIM,
             "row_scale": scale,
             "emb_dtype": emb_precision,
+            "emb_out_dtype": emb_out,
             "dense_dtype": "bf16-autocast",
             "parallelism": f"planner/dmp x{world} + pipeline",
             "qcomm": qcomm,

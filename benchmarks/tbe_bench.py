@@ -70,21 +70,29 @@ def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1, precision="fp32"):
 def bench_interaction(B=8192, F=27, D=128):
     from torchrec_amd import ops
 
-    dense = torch.randn(B, D, device="cuda", requires_grad=True)
-    sparse = torch.randn(B, F - 1, D, device="cuda", requires_grad=True)
-    fwd_us = _time_kernel(lambda: ops.fused_interaction(dense, sparse))
-    out = ops.fused_interaction(dense, sparse)
-    g = torch.ones_like(out)
+    ops.hip_ops()
+    for variant, dtype in (("fp32-valu", torch.float32),
+                           ("mfma-f32io", torch.float32),
+                           ("mfma-bf16io", torch.bfloat16)):
+        dense = torch.randn(B, D, device="cuda").to(dtype).requires_grad_(True)
+        sparse = torch.randn(B, F - 1, D, device="cuda").to(dtype).requires_grad_(True)
+        if variant == "fp32-valu":
+            fn = ops._FusedInteraction.apply
+        else:
+            fn = ops._FusedInteractionMFMA.apply
+        fwd_us = _time_kernel(lambda: fn(dense, sparse))
+        out = fn(dense, sparse)
+        g = torch.ones_like(out)
 
-    def bwd():
-        o = ops.fused_interaction(dense, sparse)
-        o.backward(g)
+        def bwd():
+            o = fn(dense, sparse)
+            o.backward(g)
 
-    both_us = _time_kernel(bwd)
-    print(json.dumps({
-        "bench": "interaction", "B": B, "F": F, "D": D,
-        "fwd_us": round(fwd_us, 1), "fwd_bwd_us": round(both_us, 1),
-    }))
+        both_us = _time_kernel(bwd)
+        print(json.dumps({
+            "bench": "interaction", "variant": variant, "B": B, "F": F, "D": D,
+            "fwd_us": round(fwd_us, 1), "fwd_bwd_us": round(both_us, 1),
+        }))
 
 
 def bench_sort(N=213_000, segments=26):

@@ -247,6 +247,50 @@ class TestTBEGpu:
         )
 
 
+class TestTBEOutputDtype:
+    """output_dtype (reference SplitTBE): fp32 accumulate, one bf16 round on
+    store; backward consumes the matching-precision gradient directly."""
+
+    def test_cpu_bf16_output(self):
+        torch.manual_seed(0)
+        tbe = TableBatchedEmbeddingBags(SPECS, output_dtype="bf16", learning_rate=0.05)
+        ref = TableBatchedEmbeddingBags(SPECS, learning_rate=0.05)
+        ref.weights.data.copy_(tbe.weights.data)
+        indices, offsets = make_inputs(SPECS, B=8)
+        out = tbe(indices, offsets)
+        assert out.dtype == torch.bfloat16
+        out32 = ref(indices, offsets)
+        assert torch.allclose(out.float(), out32, atol=0.02, rtol=0.02)
+        # bf16 grads drive the fused update
+        g = torch.randn_like(out32)
+        out.backward(g.to(torch.bfloat16))
+        out32.backward(g)
+        for wa, wb in zip(tbe.split_embedding_weights(), ref.split_embedding_weights()):
+            assert torch.allclose(wa, wb, atol=0.02, rtol=0.05)
+
+    @pytest.mark.gpu
+    def test_gpu_bf16_output_matches_cpu(self):
+        torch.manual_seed(0)
+        specs = [("t0", 100, 8), ("t1", 50, 128), ("t2", 1000, 64)]
+        cpu = TableBatchedEmbeddingBags(specs, output_dtype="bf16", learning_rate=0.05)
+        gpu = TableBatchedEmbeddingBags(
+            specs, output_dtype="bf16", learning_rate=0.05, device=torch.device("cuda")
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        for step in range(3):
+            indices, offsets = make_inputs(specs, B=16, L=7, seed=step)
+            out_c = cpu(indices, offsets)
+            out_g = gpu(indices.cuda(), offsets.cuda())
+            assert out_g.dtype == torch.bfloat16
+            assert torch.allclose(out_g.float().cpu(), out_c.float(), atol=0.02, rtol=0.02)
+            grad = torch.randn_like(out_c, dtype=torch.float32).to(torch.bfloat16)
+            out_c.backward(grad)
+            out_g.backward(grad.cuda())
+        torch.cuda.synchronize()
+        for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+            assert torch.allclose(wg.cpu(), wc, atol=0.01, rtol=0.02)
+
+
 @pytest.mark.gpu
 class TestTBEUvm:
     def test_uvm_matches_device(self):

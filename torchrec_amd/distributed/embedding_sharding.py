@@ -185,6 +185,7 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                     group[0].data_type
                 ],
                 fixed_bag_length=fused_params.get("fixed_bag_length"),
+                output_dtype=fused_params.get("output_dtype", "fp32"),
             )
             self._emb_modules.append(tbe)
             nf = sum(len(t.feature_names) for t in group)

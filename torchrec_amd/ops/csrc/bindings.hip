@@ -29,7 +29,8 @@ at::Tensor tbe_forward_pooled(const at::Tensor& weights, const at::Tensor& table
                               const at::Tensor& d_out_offsets, const at::Tensor& indices,
                               const at::Tensor& offsets, const at::Tensor& per_sample_weights,
                               int64_t B, int64_t total_D, int64_t max_D, bool mean_pool,
-                              const at::Tensor& cache_weights, const at::Tensor& cache_loc);
+                              const at::Tensor& cache_weights, const at::Tensor& cache_loc,
+                              int64_t out_dtype);
 at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_elem_offsets,
                            const at::Tensor& dims, const at::Tensor& feat_table,
                            const at::Tensor& feat_val_offsets, const at::Tensor& indices,
@@ -139,7 +140,7 @@ TORCH_LIBRARY(trec_amd, m) {
       "tbe_forward_pooled(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
       " Tensor feat_table, Tensor d_out_offsets, Tensor indices, Tensor offsets,"
       " Tensor per_sample_weights, int B, int total_D, int max_D, bool mean_pool,"
-      " Tensor cache_weights, Tensor cache_loc) -> Tensor");
+      " Tensor cache_weights, Tensor cache_loc, int out_dtype) -> Tensor");
   m.def(
       "tbe_forward_seq(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
       " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"

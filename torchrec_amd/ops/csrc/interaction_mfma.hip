@@ -144,32 +144,29 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kern
     const int32_t* __restrict__ pair_col,  // [F1*F1], -1 or pair index
     int B, int F1, int D, int P,
     io_t* __restrict__ d_dense, io_t* __restrict__ d_sparse) {
+  // ONE sample per block (4 waves cooperate): ~16.5 KB LDS keeps 8 blocks
+  // resident per CU (32 waves of latency hiding vs 8 for a wave-per-sample
+  // split — this kernel is HBM-latency/-bandwidth bound, not MFMA bound).
   extern __shared__ char lds[];
   const int l = threadIdx.x & 63;
   const int w = threadIdx.x >> 6;
-  // per-block: pair-col table; per-wave: G [32][40] bf16, Tt [D][40] bf16,
-  // dz [P] fp32 (stride 40 elements = 80 B keeps ds_read_b128 ~2-way)
-  constexpr int kStrideE = 40;
+  constexpr int kStrideE = 40;  // 80 B rows: ds_read_b128 lands ~2-way
   int32_t* pc = reinterpret_cast<int32_t*>(lds);
-  const int wave_bytes = 32 * kStrideE * 2 + D * kStrideE * 2 + 2048;
-  char* wbase = lds + 32 * 32 * 4 + w * wave_bytes;
-  char* g_tile = wbase;
-  char* tt_tile = wbase + 32 * kStrideE * 2;
-  float* dz = reinterpret_cast<float*>(wbase + 32 * kStrideE * 2 + D * kStrideE * 2);
+  char* g_tile = lds + 32 * 32 * 4;
+  char* tt_tile = g_tile + 32 * kStrideE * 2;
   const int64_t out_w = D + P;
   const int n_mt = (F1 + 15) / 16;
+  const int ntiles = n_mt * (D / 16);
 
   // sample-invariant init: pc table (padded -1) and Tt pad columns (zeros —
-  // K-pad products are 0 * Tt, which must not be NaN)
+  // the K-pad products are G_pad (=0) * Tt_pad, which must not be NaN)
   for (int e = threadIdx.x; e < 32 * 32; e += blockDim.x) {
     int i = e >> 5, j = e & 31;
     pc[e] = (i < F1 && j < F1) ? pair_col[i * F1 + j] : -1;
   }
-  // each wave zeroes the pad columns (F1..31) of its own Tt tile: the K-pad
-  // products are G_pad (=0) * Tt_pad, which must not be NaN bit patterns
   if (F1 < 32) {
     const int padw = 32 - F1;
-    for (int e = l; e < D * padw; e += 64) {
+    for (int e = threadIdx.x; e < D * padw; e += blockDim.x) {
       int r = e / padw, c = F1 + e % padw;
       *reinterpret_cast<__bf16*>(tt_tile + r * (kStrideE * 2) + c * 2) =
           static_cast<__bf16>(0.f);
@@ -177,76 +174,54 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kern
   }
   __syncthreads();
 
-  const int64_t iters = (B + kIWaves - 1) / kIWaves;
-  for (int64_t it = blockIdx.x; it < iters; it += gridDim.x) {
-    const int64_t b = it * kIWaves + w;
-    const bool active = b < B;
-    if (active) {
-      const io_t* grow = grad_out + b * out_w;
-      for (int p = l; p < P; p += 64) dz[p] = emb2float(grow[D + p]);
+  for (int64_t b = blockIdx.x; b < B; b += gridDim.x) {
+    const io_t* grow = grad_out + b * out_w;
+    // G[i][j] = dOut[pair(i,j)] (symmetric, zero diagonal / pads): the 1.9 KB
+    // grad row is L1-resident, so the scattered rereads are cache-served
+    for (int e = threadIdx.x; e < 32 * 32; e += blockDim.x) {
+      int i = e >> 5, j = e & 31;
+      int32_t p = pc[e];
+      *reinterpret_cast<__bf16*>(g_tile + i * (kStrideE * 2) + j * 2) =
+          static_cast<__bf16>(p >= 0 ? emb2float(grow[D + p]) : 0.f);
     }
-    __syncthreads();  // dz visible across lanes
-    if (active) {
-      // G[i][j] = dz[pair(i,j)] (symmetric, zero diagonal / pads)
-      for (int e = l; e < 32 * 32; e += 64) {
-        int i = e >> 5, j = e & 31;
-        int32_t p = pc[e];
-        *reinterpret_cast<__bf16*>(g_tile + i * (kStrideE * 2) + j * 2) =
-            static_cast<__bf16>(p >= 0 ? dz[p] : 0.f);
-      }
-      // Tt[d][f] = T[f][d] (bf16): read T rows vectorized, scatter-transpose
-      const io_t* drow = dense + b * D;
-      const io_t* srow = sparse + b * static_cast<int64_t>(F1 - 1) * D;
-      const int chunks_per_row = D / 4;
-      for (int cidx = l; cidx < F1 * chunks_per_row; cidx += 64) {
-        int f = cidx / chunks_per_row, c4 = cidx - f * chunks_per_row;
-        const io_t* src = (f == 0) ? drow + c4 * 4 : srow + (f - 1) * D + c4 * 4;
+    // Tt[d][f] = T[f][d] (bf16): read T rows vectorized, scatter-transpose
+    const io_t* drow = dense + b * D;
+    const io_t* srow = sparse + b * static_cast<int64_t>(F1 - 1) * D;
+    const int chunks_per_row = D / 4;
+    for (int cidx = threadIdx.x; cidx < F1 * chunks_per_row; cidx += blockDim.x) {
+      int f = cidx / chunks_per_row, c4 = cidx - f * chunks_per_row;
+      const io_t* src = (f == 0) ? drow + c4 * 4 : srow + (f - 1) * D + c4 * 4;
 #pragma unroll
-        for (int e = 0; e < 4; ++e)
-          *reinterpret_cast<__bf16*>(tt_tile + (c4 * 4 + e) * (kStrideE * 2) + f * 2) =
-              static_cast<__bf16>(emb2float(src[e]));
-      }
+      for (int e = 0; e < 4; ++e)
+        *reinterpret_cast<__bf16*>(tt_tile + (c4 * 4 + e) * (kStrideE * 2) + f * 2) =
+            static_cast<__bf16>(emb2float(src[e]));
     }
     __syncthreads();
-    if (active) {
-      const io_t* grow = grad_out + b * out_w;
-      io_t* ddrow = d_dense + b * D;
-      io_t* dsrow = d_sparse + b * static_cast<int64_t>(F1 - 1) * D;
-      const int zr = (l >> 4) * 4, zc = l & 15;
-      for (int mi = 0; mi < n_mt; ++mi) {
-        // A fragment from G rows [mi*16, mi*16+16)
-        bf16x8 a;
-        {
-          int row = mi * 16 + (l & 15);
-          a = *reinterpret_cast<const bf16x8*>(g_tile + row * (kStrideE * 2) +
-                                               ((l >> 4) << 4));
-        }
-        for (int ni = 0; ni < D / 16; ++ni) {
-          // B fragment: B[k][n] = T[k][n] = Tt[n][k]
-          bf16x8 bfr;
-          {
-            int row = ni * 16 + (l & 15);
-            bfr = *reinterpret_cast<const bf16x8*>(tt_tile + row * (kStrideE * 2) +
-                                                   ((l >> 4) << 4));
-          }
-          f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
+    io_t* ddrow = d_dense + b * D;
+    io_t* dsrow = d_sparse + b * static_cast<int64_t>(F1 - 1) * D;
+    const int zr = (l >> 4) * 4, zc = l & 15;
+    for (int t = w; t < ntiles; t += kIWaves) {
+      const int mi = t / (D / 16), ni = t - mi * (D / 16);
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          g_tile + (mi * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
+      bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+          tt_tile + (ni * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            int i = mi * 16 + zr + r;     // T row (0 = dense)
-            int dcol = ni * 16 + zc;      // dim column
-            if (i >= F1) continue;
-            float v = acc[r];
-            if (i == 0) {
-              ddrow[dcol] = float2emb(v + emb2float(grow[dcol]), io_t{});
-            } else {
-              dsrow[static_cast<int64_t>(i - 1) * D + dcol] = float2emb(v, io_t{});
-            }
-          }
+      for (int r = 0; r < 4; ++r) {
+        int i = mi * 16 + zr + r;     // T row (0 = dense)
+        int dcol = ni * 16 + zc;      // dim column
+        if (i >= F1) continue;
+        float v = acc[r];
+        if (i == 0) {
+          ddrow[dcol] = float2emb(v + emb2float(grow[dcol]), io_t{});
+        } else {
+          dsrow[static_cast<int64_t>(i - 1) * D + dcol] = float2emb(v, io_t{});
         }
       }
     }
-    __syncthreads();
+    __syncthreads();  // next sample's staging overwrites G/Tt
   }
 }
 
@@ -296,9 +271,8 @@ std::tuple<at::Tensor, at::Tensor> interaction_mfma_backward(
   auto d_sparse = at::empty_like(sparse);
   if (B == 0) return {d_dense, d_sparse};
   constexpr int kStrideE = 40;
-  int lds_bytes = 32 * 32 * 4 +
-                  kIWaves * (32 * kStrideE * 2 + D * kStrideE * 2 + 2048);
-  int grid = std::min<int64_t>((B + kIWaves - 1) / kIWaves, kNumCU * 8);
+  int lds_bytes = 32 * 32 * 4 + 32 * kStrideE * 2 + D * kStrideE * 2;
+  int grid = std::min<int64_t>(B, kNumCU * 8);
   auto stream = imf_stream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, dense.scalar_type(),
                                   "interaction_mfma_bwd", [&] {

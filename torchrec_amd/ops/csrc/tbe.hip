@@ -57,7 +57,7 @@ static inline hipStream_t tbe_stream() {
 // forward (pooled)
 // ---------------------------------------------------------------------------
 
-template <typename emb_t, int LPS, int CHUNKS>
+template <typename emb_t, typename o_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
     const emb_t* __restrict__ weights,
     const int64_t* __restrict__ table_elem_offsets,  // [T]
@@ -71,7 +71,7 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
     const int32_t* __restrict__ cache_loc,     // per-position slot or -1
     int64_t cache_stride,
     int F, int B, int64_t total_D, bool mean_pool,
-    float* __restrict__ out /* [B, total_D] */) {
+    o_t* __restrict__ out /* [B, total_D] */) {
   constexpr int SLOTS = kWaveSize / LPS;
   int sl = threadIdx.x % LPS;                    // lane within slot
   int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
@@ -110,38 +110,65 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_pooled_kernel(
     }
     float scale = 1.f;
     if (mean_pool && i1 > i0) scale = 1.f / static_cast<float>(i1 - i0);
-    float4* orow = reinterpret_cast<float4*>(out + static_cast<int64_t>(b) * total_D +
-                                             d_out_offsets[f]);
+    o_t* orow = out + static_cast<int64_t>(b) * total_D + d_out_offsets[f];
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
       int col4 = c * LPS + sl;
       if (col4 * 4 < D) {
-        orow[col4] = make_float4(acc[c].x * scale, acc[c].y * scale, acc[c].z * scale,
-                                 acc[c].w * scale);
+        Vec4<o_t>::store(orow, col4,
+                         make_float4(acc[c].x * scale, acc[c].y * scale,
+                                     acc[c].z * scale, acc[c].w * scale));
       }
     }
   }
 }
 
-#define TBE_FWD_LAUNCH(LPS, CHUNKS)                                                      \
-  hipLaunchKernelGGL((tbe_fwd_pooled_kernel<dev_t, LPS, CHUNKS>), dim3(grid),             \
-                     dim3(kBlockThreads),                                                 \
-                     0, stream, reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights)), \
+template <typename emb_t, typename o_t, typename host_o_t>
+static void launch_tbe_fwd_pooled(
+    const emb_t* weights, const at::Tensor& table_elem_offsets, const at::Tensor& dims,
+    const at::Tensor& feat_table, const at::Tensor& d_out_offsets,
+    const at::Tensor& indices, const at::Tensor& offsets, const float* psw_ptr,
+    const float* cache_w_ptr, const int32_t* cache_loc_ptr, int64_t max_D, int F,
+    int64_t B, int64_t total_D, bool mean_pool, at::Tensor& out, int lps, int chunks,
+    int grid, hipStream_t stream) {
+  o_t* out_ptr = reinterpret_cast<o_t*>(out.data_ptr<host_o_t>());
+#define TBE_FWD_LAUNCH(LPS, CHUNKS)                                                       \
+  hipLaunchKernelGGL((tbe_fwd_pooled_kernel<emb_t, o_t, LPS, CHUNKS>), dim3(grid),        \
+                     dim3(kBlockThreads), 0, stream, weights,                             \
                      table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),    \
                      feat_table.data_ptr<int32_t>(), d_out_offsets.data_ptr<int64_t>(),   \
                      indices.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(),            \
-                     psw_ptr, cache_w_ptr, cache_loc_ptr, max_D, F, B, total_D, mean_pool, \
-                     out.data_ptr<float>())
+                     psw_ptr, cache_w_ptr, cache_loc_ptr, max_D, F, (int)B, total_D,      \
+                     mean_pool, out_ptr)
+  if (lps == 16) {
+    TORCH_CHECK(chunks == 1);
+    TBE_FWD_LAUNCH(16, 1);
+  } else if (lps == 32) {
+    TORCH_CHECK(chunks == 1);
+    TBE_FWD_LAUNCH(32, 1);
+  } else {
+    switch (chunks) {
+      case 1: TBE_FWD_LAUNCH(64, 1); break;
+      case 2: TBE_FWD_LAUNCH(64, 2); break;
+      case 3: case 4: TBE_FWD_LAUNCH(64, 4); break;
+      default: TBE_FWD_LAUNCH(64, 8); break;
+    }
+  }
+#undef TBE_FWD_LAUNCH
+}
 
 at::Tensor tbe_forward_pooled(
     const at::Tensor& weights, const at::Tensor& table_elem_offsets, const at::Tensor& dims,
     const at::Tensor& feat_table, const at::Tensor& d_out_offsets, const at::Tensor& indices,
     const at::Tensor& offsets, const at::Tensor& per_sample_weights, int64_t B,
     int64_t total_D, int64_t max_D, bool mean_pool, const at::Tensor& cache_weights,
-    const at::Tensor& cache_loc) {
+    const at::Tensor& cache_loc, int64_t out_dtype) {
   TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048, "TBE: dims must be %4==0 and <=2048");
   int F = feat_table.numel();
-  auto out = at::empty({B, total_D}, indices.options().dtype(at::kFloat));
+  // out_dtype: 0 = fp32, 1 = bf16, 2 = fp16 (reference SplitTBE output_dtype;
+  // the pool still accumulates fp32 and rounds once on store)
+  auto out_st = out_dtype == 1 ? at::kBFloat16 : (out_dtype == 2 ? at::kHalf : at::kFloat);
+  auto out = at::empty({B, total_D}, indices.options().dtype(out_st));
   if (B == 0 || F == 0) return out;
   const float* psw_ptr =
       per_sample_weights.numel() > 0 ? per_sample_weights.data_ptr<float>() : nullptr;
@@ -161,20 +188,23 @@ at::Tensor tbe_forward_pooled(
     if constexpr (std::is_same_v<scalar_t, double>) {
       TORCH_CHECK(false, "fp64 embedding tables unsupported");
     } else {
-    if (lps == 16) {
-      TORCH_CHECK(chunks == 1);
-      TBE_FWD_LAUNCH(16, 1);
-    } else if (lps == 32) {
-      TORCH_CHECK(chunks == 1);
-      TBE_FWD_LAUNCH(32, 1);
-    } else {
-      switch (chunks) {
-        case 1: TBE_FWD_LAUNCH(64, 1); break;
-        case 2: TBE_FWD_LAUNCH(64, 2); break;
-        case 3: case 4: TBE_FWD_LAUNCH(64, 4); break;
-        default: TBE_FWD_LAUNCH(64, 8); break;
+      const dev_t* wp = reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights));
+      if (out_dtype == 1) {
+        launch_tbe_fwd_pooled<dev_t, __hip_bfloat16, at::BFloat16>(
+            wp, table_elem_offsets, dims, feat_table, d_out_offsets, indices, offsets,
+            psw_ptr, cache_w_ptr, cache_loc_ptr, max_D, F, B, total_D, mean_pool, out,
+            lps, chunks, grid, stream);
+      } else if (out_dtype == 2) {
+        launch_tbe_fwd_pooled<dev_t, __half, at::Half>(
+            wp, table_elem_offsets, dims, feat_table, d_out_offsets, indices, offsets,
+            psw_ptr, cache_w_ptr, cache_loc_ptr, max_D, F, B, total_D, mean_pool, out,
+            lps, chunks, grid, stream);
+      } else {
+        launch_tbe_fwd_pooled<dev_t, float, float>(
+            wp, table_elem_offsets, dims, feat_table, d_out_offsets, indices, offsets,
+            psw_ptr, cache_w_ptr, cache_loc_ptr, max_D, F, B, total_D, mean_pool, out,
+            lps, chunks, grid, stream);
       }
-    }
     }
   });
   return out;
@@ -613,9 +643,9 @@ __device__ __forceinline__ int upper_bound_segment_i32(const int32_t* offs, int 
   return lo;
 }
 
-template <int LPS, int CHUNKS>
+template <typename g_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
-    const float* __restrict__ grad, int64_t grad_stride,
+    const g_t* __restrict__ grad, int64_t grad_stride,
     const int64_t* __restrict__ sorted_linear, const int32_t* __restrict__ sort_perm,
     const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
     const int32_t* __restrict__ chunk_offsets, const int32_t* __restrict__ total_chunks_ptr,
@@ -641,14 +671,13 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
     for (int cc = 0; cc < CHUNKS; ++cc) acc[cc] = make_float4(0.f, 0.f, 0.f, 0.f);
     for (int32_t k = k0; k < k1; ++k) {
       int32_t p = sort_perm[k];
-      const float4* grow = reinterpret_cast<const float4*>(
-          grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p]);
+      const g_t* grow = grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p];
       float s = pos_scale ? pos_scale[p] : 1.f;
 #pragma unroll
       for (int cc = 0; cc < CHUNKS; ++cc) {
         int col4 = cc * LPS + sl;
         if (col4 * 4 < D) {
-          float4 g = grow[col4];
+          float4 g = Vec4<g_t>::load(grow, col4);
           acc[cc].x += s * g.x;
           acc[cc].y += s * g.y;
           acc[cc].z += s * g.z;
@@ -672,10 +701,10 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
 // mode: 0 = SGD, 1 = rowwise Adagrad, 2 = dense grad (write grad_weights).
 // ---------------------------------------------------------------------------
 
-template <typename emb_t, int LPS, int CHUNKS>
+template <typename emb_t, typename g_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     emb_t* __restrict__ weights, float* __restrict__ momentum,
-    const float* __restrict__ grad, int64_t grad_stride,
+    const g_t* __restrict__ grad, int64_t grad_stride,
     const int64_t* __restrict__ sorted_linear, const int32_t* __restrict__ sort_perm,
     const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
     const int32_t* __restrict__ chunk_offsets, const float* __restrict__ scratch,
@@ -720,14 +749,13 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     } else {
       for (int32_t k = k0; k < k1; ++k) {
         int32_t p = sort_perm[k];
-        const float4* grow = reinterpret_cast<const float4*>(
-            grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p]);
+        const g_t* grow = grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p];
         float s = pos_scale ? pos_scale[p] : 1.f;
 #pragma unroll
         for (int c = 0; c < CHUNKS; ++c) {
           int col4 = c * LPS + sl;
           if (col4 * 4 < D) {
-            float4 g = grow[col4];
+            float4 g = Vec4<g_t>::load(grow, col4);
             acc[c].x += s * g.x;
             acc[c].y += s * g.y;
             acc[c].z += s * g.z;
@@ -793,6 +821,78 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
   }
 }
 
+template <typename emb_t, typename g_t, typename host_w_t>
+static void launch_tbe_bwd(
+    at::Tensor& weights, at::Tensor& momentum, const at::Tensor& grad,
+    const at::Tensor& sorted_linear, const at::Tensor& sort_perm,
+    const at::Tensor& seg_offsets, const at::Tensor& num_runs,
+    const at::Tensor& chunk_offsets, const at::Tensor& total_chunks,
+    const at::Tensor& scratch, const at::Tensor& pos_row, const at::Tensor& pos_col,
+    const float* scale_ptr, const at::Tensor& table_row_offsets,
+    const at::Tensor& table_elem_offsets, const at::Tensor& dims, int T,
+    int chunk_size, int64_t max_D, float lr, float eps, int mode,
+    at::Tensor& grad_weights, float* cache_w_ptr, const int32_t* cache_loc_ptr,
+    int lps, int chunks, int grid, int grid_long, hipStream_t stream) {
+  const g_t* grad_ptr = reinterpret_cast<const g_t*>(grad.data_ptr());
+  emb_t* gw_ptr = grad_weights.numel() > 0
+      ? reinterpret_cast<emb_t*>(grad_weights.data_ptr<host_w_t>()) : nullptr;
+#define TBE_BWD_LAUNCH(LPS, CHUNKS)                                                          \
+  do {                                                                                       \
+    hipLaunchKernelGGL((tbe_bwd_long_partial_kernel<g_t, LPS, CHUNKS>), dim3(grid_long),     \
+                       dim3(kBlockThreads), 0, stream, grad_ptr, grad.size(1),               \
+                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
+                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
+                       chunk_offsets.data_ptr<int32_t>(), total_chunks.data_ptr<int32_t>(),  \
+                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
+                       table_row_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,   \
+                       chunk_size, max_D, scratch.data_ptr<float>());                        \
+    hipLaunchKernelGGL((tbe_bwd_fused_kernel<emb_t, g_t, LPS, CHUNKS>), dim3(grid),          \
+                       dim3(kBlockThreads),                                                  \
+                       0, stream, reinterpret_cast<emb_t*>(uvm_ptr<host_w_t>(weights)),      \
+                       uvm_ptr<float>(momentum),                                             \
+                       grad_ptr, grad.size(1),                                               \
+                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
+                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
+                       chunk_offsets.data_ptr<int32_t>(), scratch.data_ptr<float>(), max_D,  \
+                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
+                       table_row_offsets.data_ptr<int64_t>(),                                \
+                       table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,  \
+                       lr, eps, mode, gw_ptr, cache_w_ptr,                                   \
+                       cache_loc_ptr, max_D);                                                \
+  } while (0)
+  if (lps == 16) TBE_BWD_LAUNCH(16, 1);
+  else if (lps == 32) TBE_BWD_LAUNCH(32, 1);
+  else switch (chunks) {
+    case 1: TBE_BWD_LAUNCH(64, 1); break;
+    case 2: TBE_BWD_LAUNCH(64, 2); break;
+    case 3: case 4: TBE_BWD_LAUNCH(64, 4); break;
+    default: TBE_BWD_LAUNCH(64, 8); break;
+  }
+#undef TBE_BWD_LAUNCH
+}
+
+template <typename emb_t, typename host_w_t>
+static void dispatch_tbe_bwd_grad(const at::ScalarType g_st, at::Tensor& weights,
+    at::Tensor& momentum, const at::Tensor& grad, const at::Tensor& sorted_linear,
+    const at::Tensor& sort_perm, const at::Tensor& seg_offsets, const at::Tensor& num_runs,
+    const at::Tensor& chunk_offsets, const at::Tensor& total_chunks,
+    const at::Tensor& scratch, const at::Tensor& pos_row, const at::Tensor& pos_col,
+    const float* scale_ptr, const at::Tensor& table_row_offsets,
+    const at::Tensor& table_elem_offsets, const at::Tensor& dims, int T, int chunk_size,
+    int64_t max_D, float lr, float eps, int mode, at::Tensor& grad_weights,
+    float* cache_w_ptr, const int32_t* cache_loc_ptr, int lps, int chunks, int grid,
+    int grid_long, hipStream_t stream) {
+#define ARGS weights, momentum, grad, sorted_linear, sort_perm, seg_offsets, num_runs,  \
+    chunk_offsets, total_chunks, scratch, pos_row, pos_col, scale_ptr,                  \
+    table_row_offsets, table_elem_offsets, dims, T, chunk_size, max_D, lr, eps, mode,   \
+    grad_weights, cache_w_ptr, cache_loc_ptr, lps, chunks, grid, grid_long, stream
+  if (g_st == at::kFloat) launch_tbe_bwd<emb_t, float, host_w_t>(ARGS);
+  else if (g_st == at::kBFloat16) launch_tbe_bwd<emb_t, __hip_bfloat16, host_w_t>(ARGS);
+  else if (g_st == at::kHalf) launch_tbe_bwd<emb_t, __half, host_w_t>(ARGS);
+  else TORCH_CHECK(false, "unsupported TBE gradient dtype");
+#undef ARGS
+}
+
 void tbe_backward_fused(
     at::Tensor weights, at::Tensor momentum, const at::Tensor& grad,
     const at::Tensor& sorted_linear, const at::Tensor& sort_perm,
@@ -801,7 +901,6 @@ void tbe_backward_fused(
     const at::Tensor& table_row_offsets, const at::Tensor& table_elem_offsets,
     const at::Tensor& dims, int64_t max_D, double lr, double eps, int64_t mode,
     at::Tensor grad_weights, at::Tensor cache_weights, const at::Tensor& cache_loc) {
-  TORCH_CHECK(grad.scalar_type() == at::kFloat);
   int64_t n = sorted_linear.numel();
   if (n == 0) return;
   int T = table_elem_offsets.numel();
@@ -818,52 +917,22 @@ void tbe_backward_fused(
   constexpr int kChunkSize = 32;
   auto [chunk_offsets, total_chunks] = tbe_backward_chunk_prep(seg_offsets, num_runs, kChunkSize);
   int64_t max_chunks = 2 * ((n + kChunkSize - 1) / kChunkSize) + 2;
-  auto scratch = at::empty({max_chunks * max_D}, grad.options());
+  auto scratch = at::empty({max_chunks * max_D}, grad.options().dtype(at::kFloat));
   int grid_long = grid_for(max_chunks * lps, kBlockThreads);
 
-#define TBE_BWD_LAUNCH(LPS, CHUNKS)                                                          \
-  do {                                                                                       \
-    hipLaunchKernelGGL((tbe_bwd_long_partial_kernel<LPS, CHUNKS>), dim3(grid_long),          \
-                       dim3(kBlockThreads), 0, stream, grad.data_ptr<float>(), grad.size(1), \
-                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
-                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
-                       chunk_offsets.data_ptr<int32_t>(), total_chunks.data_ptr<int32_t>(),  \
-                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
-                       table_row_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,   \
-                       kChunkSize, max_D, scratch.data_ptr<float>());                        \
-    hipLaunchKernelGGL((tbe_bwd_fused_kernel<dev_t, LPS, CHUNKS>), dim3(grid),               \
-                       dim3(kBlockThreads),                                                  \
-                       0, stream, reinterpret_cast<dev_t*>(uvm_ptr<scalar_t>(weights)),      \
-                       uvm_ptr<float>(momentum),                                             \
-                       grad.data_ptr<float>(), grad.size(1),                                 \
-                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
-                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
-                       chunk_offsets.data_ptr<int32_t>(), scratch.data_ptr<float>(), max_D,  \
-                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
-                       table_row_offsets.data_ptr<int64_t>(),                                \
-                       table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,  \
-                       (float)lr, (float)eps, (int)mode, gw_ptr, cache_w_ptr,                \
-                       cache_loc_ptr, max_D);                                                \
-  } while (0)
-  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
-                                  "tbe_bwd_fused", [&] {
-    using dev_t = typename DevType<scalar_t>::type;
-    if constexpr (std::is_same_v<scalar_t, double>) {
-      TORCH_CHECK(false, "fp64 embedding tables unsupported");
-    } else {
-    dev_t* gw_ptr = grad_weights.numel() > 0
-        ? reinterpret_cast<dev_t*>(grad_weights.data_ptr<scalar_t>()) : nullptr;
-    if (lps == 16) TBE_BWD_LAUNCH(16, 1);
-    else if (lps == 32) TBE_BWD_LAUNCH(32, 1);
-    else switch (chunks) {
-      case 1: TBE_BWD_LAUNCH(64, 1); break;
-      case 2: TBE_BWD_LAUNCH(64, 2); break;
-      case 3: case 4: TBE_BWD_LAUNCH(64, 4); break;
-      default: TBE_BWD_LAUNCH(64, 8); break;
-    }
-    }
-  });
-#undef TBE_BWD_LAUNCH
+#define BARGS weights, momentum, grad, sorted_linear, sort_perm, seg_offsets, num_runs, \
+    chunk_offsets, total_chunks, scratch, pos_row, pos_col, scale_ptr,                  \
+    table_row_offsets, table_elem_offsets, dims, T, kChunkSize, max_D, (float)lr,       \
+    (float)eps, (int)mode, grad_weights, cache_w_ptr, cache_loc_ptr, lps, chunks, grid, \
+    grid_long, stream
+  auto g_st = grad.scalar_type();
+  auto w_st = weights.scalar_type();
+  if (w_st == at::kFloat) dispatch_tbe_bwd_grad<float, float>(g_st, BARGS);
+  else if (w_st == at::kBFloat16)
+    dispatch_tbe_bwd_grad<__hip_bfloat16, at::BFloat16>(g_st, BARGS);
+  else if (w_st == at::kHalf) dispatch_tbe_bwd_grad<__half, at::Half>(g_st, BARGS);
+  else TORCH_CHECK(false, "unsupported TBE weights dtype");
+#undef BARGS
 }
 
 __global__ void gather_run_heads_kernel(const int64_t* __restrict__ sorted_lin,

@@ -82,6 +82,7 @@ class _TBEPooledFunction(torch.autograd.Function):
             host.pooling_mode == PoolingMode.MEAN,
             host.cache_weights,
             cache_loc,
+            host._out_dtype_code,
         )
         ctx.host = host
         ctx.save_for_backward(
@@ -195,10 +196,19 @@ class TableBatchedEmbeddingBags(nn.Module):
         cache_load_factor: float = 0.2,
         weights_precision: str = "fp32",
         fixed_bag_length: Optional[int] = None,
+        output_dtype: str = "fp32",
     ) -> None:
         super().__init__()
         device = device or torch.device("cpu")
         self.location = location
+        # pooled-output precision (reference SplitTBE output_dtype): the pool
+        # accumulates fp32 in-kernel and rounds once on store; backward
+        # consumes the matching-precision gradient directly (no cast pass)
+        self.output_dtype = output_dtype
+        self._out_dtype_code = {"fp32": 0, "bf16": 1, "fp16": 2}[output_dtype]
+        self._out_torch_dtype = {
+            "fp32": torch.float32, "bf16": torch.bfloat16, "fp16": torch.float16
+        }[output_dtype]
         # table storage precision (reference: SplitTBE weights_precision /
         # EmbeddingBagConfig.data_type). Grad accumulate + pooled output stay
         # fp32; bf16/fp16 rows move as 8 B/lane (Vec4<emb_t> in common.h).
@@ -432,6 +442,8 @@ class TableBatchedEmbeddingBags(nn.Module):
             # read rows BEFORE the fused update rewrites them
             fpsw = torch.div(bag_ids, B, rounding_mode="floor")
             pos_table = self._feat_table_t.to(torch.int64)[fpsw].to(torch.int32)
+            if grad.dtype != torch.float32:
+                grad = grad.float()  # psw-grad kernel reads fp32 rows
             grad_psw_out = torch.ops.trec_amd.tbe_grad_per_sample_weights(
                 self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
                 self._table_elem_offsets,
@@ -786,7 +798,10 @@ def _tbe_cpu_forward(weights, host, indices, offsets, psw, B):
                 idx, w, off, mode=mode, per_sample_weights=pw, include_last_offset=True,
             )
         )
-    return torch.cat(outs, dim=1)
+    out = torch.cat(outs, dim=1)
+    if host._out_torch_dtype != torch.float32:
+        out = out.to(host._out_torch_dtype)  # parity with the GPU output_dtype
+    return out
 
 
 class _TBECpuFusedFunction(torch.autograd.Function):
@@ -816,7 +831,7 @@ class _TBECpuFusedFunction(torch.autograd.Function):
         w = host.weights.detach().float().requires_grad_(True)
         with torch.enable_grad():
             out = _tbe_cpu_forward(w, host, indices, offsets, psw_t, B)
-            out.backward(grad)
+            out.backward(grad.to(out.dtype))
         host._cpu_apply_update(w.grad)
         grad_psw = psw_t.grad if need_psw_grad else None
         return None, None, None, None, grad_psw, None
@@ -843,6 +858,7 @@ class _TBEDenseFunction(torch.autograd.Function):
             host.pooling_mode == PoolingMode.MEAN,
             host._empty_f,
             host._empty_i,
+            0,
         )
         ctx.host = host
         ctx.save_for_backward(indices, offsets, psw if psw is not None else host._empty_f)
