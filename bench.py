@@ -344,6 +344,7 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
     torch.cuda.set_device(device)
     enable_tuned_gemms()
     emb_precision = os.environ.get("TREC_EMB_PRECISION", "fp32")
+    emb_out = os.environ.get("TREC_EMB_OUT", "bf16")
     model = build_model(scale, emb_precision)
     fused_params = {
         "optimizer": "rowwise_adagrad",
@@ -351,6 +352,7 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
         # Criteo categoricals are one-hot: enables the single-launch
         # segmented backward sort
         "fixed_bag_length": IDS_PER_FEATURE,
+        "output_dtype": emb_out,
     }
     sharder = EmbeddingBagCollectionSharder(fused_params=fused_params)
     planner = EmbeddingShardingPlanner(
@@ -408,10 +410,16 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
         load(i)
         g.replay()
     torch.cuda.synchronize()
+    assert torch.isfinite(static_loss).all(), "non-finite loss in captured step"
+    step_t = []
     t0 = time.perf_counter()
+    tprev = t0
     for i in range(steps):
         load(warmup + i)
         g.replay()
+        tnow = time.perf_counter()
+        step_t.append(tnow - tprev)
+        tprev = tnow
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     ms_per_step = dt / steps * 1e3
@@ -435,11 +443,19 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
             "tables": len(DLRM_EMB_ROWS),
             "embedding_dim": EMB_DIM,
             "row_scale": scale,
-            "emb_dtype": "fp32",
+            "emb_dtype": emb_precision,
+            "emb_out_dtype": emb_out,
             "dense_dtype": "bf16-autocast",
             "parallelism": "planner/dmp x1 + hipGraph step capture",
         },
     }
+    st = sorted(step_t)
+    pct = lambda q: st[min(len(st) - 1, int(q * len(st)))] * 1e3  # noqa: E731
+    print(
+        f"# step-ms p10={pct(0.10):.3f} p50={pct(0.50):.3f} "
+        f"p90={pct(0.90):.3f} p99={pct(0.99):.3f} mean={ms_per_step:.3f}",
+        flush=True,
+    )
     print(json.dumps(result))
 
 
@@ -494,10 +510,29 @@ if __name__ == "__main__":
         help="compress the pooled a2a wire (fwd+bwd)",
     )
     args = p.parse_args()
+    # single-GPU default is the hipGraph-captured step (static one-hot shapes;
+    # full fwd+bwd+optimizer inside the capture). TREC_NO_HIPGRAPH=1 or any
+    # capture failure falls back to the stream-pipelined eager path.
+    want_graph = (
+        args.hipgraph
+        or (
+            int(os.environ.get("WORLD_SIZE", "1")) == 1
+            and os.environ.get("TREC_BENCH_CPU") != "1"
+            and os.environ.get("TREC_FORCE_DIST") != "1"
+            and os.environ.get("TREC_NO_HIPGRAPH") != "1"
+            and not args.smoke
+        )
+    )
     if args.smoke:
         run_smoke()
-    elif args.hipgraph and int(os.environ.get("WORLD_SIZE", "1")) == 1:
-        run_graph_bench(args.steps, args.warmup, args.batch_size, args.scale)
+    elif want_graph and int(os.environ.get("WORLD_SIZE", "1")) == 1:
+        try:
+            run_graph_bench(args.steps, args.warmup, args.batch_size, args.scale)
+        except Exception as exc:  # pragma: no cover — capture fallback
+            print(f"# hipGraph capture failed ({exc!r}); eager pipeline fallback",
+                  flush=True)
+            run_bench(args.gpus, args.steps, args.warmup, args.batch_size,
+                      args.scale, qcomm=args.qcomm)
     else:
         run_bench(args.gpus, args.steps, args.warmup, args.batch_size, args.scale,
                   qcomm=args.qcomm)
