@@ -362,7 +362,10 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
     dmp = DistributedModelParallel(
         model, env=ShardingEnv.from_local(1, 0), plan=plan, sharders=[sharder], device=device
     )
-    dense_opt = torch.optim.SGD([p for p in dmp.parameters() if p.requires_grad], lr=LR)
+    _dense_to_bf16(dmp)  # bf16 dense params: no per-layer autocast weight casts
+    dense_opt = torch.optim.SGD(
+        [p for p in dmp.parameters() if p.requires_grad], lr=LR, foreach=True
+    )
 
     host_batches = make_host_batches(8, batch_size, scale, seed=1234, pin=True)
     b0 = host_batches[0].to(device)
@@ -384,7 +387,9 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
     def one_step():
         with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
             loss, _ = dmp(static_batch)
-        dense_opt.zero_grad(set_to_none=False)
+        # set_to_none: backward writes fresh grad buffers (replay-stable via
+        # the capture pool) — drops one fill kernel per dense param
+        dense_opt.zero_grad(set_to_none=True)
         loss.backward()
         dense_opt.step()
         return loss

@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "quant_tbe.hip"),
         os.path.join(CSRC, "interaction.hip"),
         os.path.join(CSRC, "interaction_mfma.hip"),
+        os.path.join(CSRC, "mlp_ops.hip"),
         os.path.join(CSRC, "cache.hip"),
         os.path.join(CSRC, "bindings.hip"),
     ],

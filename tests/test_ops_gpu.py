@@ -204,6 +204,45 @@ class TestInteractionMFMA:
         )
 
 
+class TestFusedMLPBackward:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    def test_relu_bwd_col_sum(self, dtype):
+        ops.hip_ops()
+        torch.manual_seed(0)
+        y = torch.randn(500, 300, device="cuda").to(dtype).relu()
+        dy = torch.randn(500, 300, device="cuda").to(dtype)
+        g, db = torch.ops.trec_amd.relu_bwd_col_sum(dy, y)
+        ref_g = dy * (y > 0)
+        torch.cuda.synchronize()
+        assert torch.equal(g, ref_g)
+        assert torch.allclose(db, ref_g.float().sum(0), atol=1e-2, rtol=1e-3)
+
+    def test_linear_relu_fused_matches_eager(self):
+        from torchrec_amd.modules.mlp import _LinearReLUFused
+
+        torch.manual_seed(0)
+        B, K, N = 2048 * 4, 64, 96  # B % 8 == 0 and >= 4096: split-K wgrad path
+        x = torch.randn(B, K, device="cuda", requires_grad=True)
+        w = torch.randn(N, K, device="cuda", requires_grad=True)
+        b = torch.randn(N, device="cuda", requires_grad=True)
+        y = _LinearReLUFused.apply(x, w, b)
+        ref = torch.nn.functional.linear(
+            x.detach().clone().requires_grad_(True), w.detach(), b.detach()
+        )
+        g = torch.randn_like(y)
+        y.backward(g)
+        xr = x.detach().clone().requires_grad_(True)
+        wr = w.detach().clone().requires_grad_(True)
+        br = b.detach().clone().requires_grad_(True)
+        yr = torch.relu(torch.nn.functional.linear(xr, wr, br))
+        yr.backward(g)
+        torch.cuda.synchronize()
+        assert torch.allclose(y, yr, atol=1e-5, rtol=1e-5)
+        assert torch.allclose(x.grad, xr.grad, atol=1e-4, rtol=1e-4)
+        assert torch.allclose(w.grad, wr.grad, atol=1e-2, rtol=1e-3)
+        assert torch.allclose(b.grad, br.grad, atol=1e-2, rtol=1e-3)
+
+
 @pytest.mark.gpu
 class TestColSum:
     @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16, torch.float16])

@@ -36,6 +36,43 @@ import os
 
 # opt-in: A/B on MI355X showed parity with torch's fused linear backward
 _USE_COLSUM = os.environ.get("TREC_COLSUM_LINEAR", "0") == "1"
+# default-on GPU path: fused relu-mask + bias-grad kernel and split-K wgrad
+# (hipBLASLt's TN heuristic never split-Ks the B=8192-deep weight gradients:
+# measured 52-65 us per layer; chunked bmm halves it)
+_USE_FUSED_MLP = os.environ.get("TREC_FUSED_MLP", "1") == "1"
+
+
+def _splitk_wgrad(g: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
+    """dW = g^T @ x with K = batch split across a bmm (split-K that
+    hipBLASLt's heuristic refuses to pick for TN deep-K shapes)."""
+    B = g.shape[0]
+    if B >= 4096 and B % 8 == 0:
+        gv = g.view(8, B // 8, g.shape[1])
+        xv = x.view(8, B // 8, x.shape[1])
+        return torch.bmm(gv.transpose(1, 2), xv).sum(0)
+    return g.t() @ x
+
+
+class _LinearReLUFused(torch.autograd.Function):
+    """y = relu(x @ W^T + b); backward fuses the relu mask with the bias
+    column-sum (one dY read) and split-Ks the weight gradient."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):  # type: ignore[override]
+        y = torch.addmm(b, x, w.t()).relu_()
+        ctx.save_for_backward(x, w, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):  # type: ignore[override]
+        from torchrec_amd import ops
+
+        x, w, y = ctx.saved_tensors
+        ops.hip_ops()
+        g, db = torch.ops.trec_amd.relu_bwd_col_sum(dy.contiguous(), y)
+        dx = g @ w
+        dw = _splitk_wgrad(g, x)
+        return dx, dw, db.to(w.dtype)
 
 
 def _linear_fwd(linear: nn.Linear, input: torch.Tensor) -> torch.Tensor:
@@ -75,6 +112,21 @@ class Perceptron(nn.Module):
         self._activation_fn = activation
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
+        if (
+            _USE_FUSED_MLP
+            and self._activation_fn is torch.relu
+            and input.is_cuda
+            and input.dim() == 2
+            and self._linear.bias is not None
+            and torch.is_grad_enabled()
+        ):
+            w, b = self._linear.weight, self._linear.bias
+            if torch.is_autocast_enabled("cuda"):
+                dt = torch.get_autocast_dtype("cuda")
+                with torch.autocast("cuda", enabled=False):
+                    return _LinearReLUFused.apply(input.to(dt), w.to(dt), b.to(dt))
+            if input.dtype == w.dtype == b.dtype:
+                return _LinearReLUFused.apply(input, w, b)
         return self._activation_fn(_linear_fwd(self._linear, input))
 
 

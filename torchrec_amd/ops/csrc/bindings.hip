@@ -102,6 +102,9 @@ at::Tensor tbe_forward_seq_int8(const at::Tensor& qweights,
 
 // interaction.hip
 at::Tensor col_sum(const at::Tensor& input);
+// mlp_ops.hip
+std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
+                                                    const at::Tensor& y);
 // interaction_mfma.hip
 at::Tensor interaction_mfma_forward(const at::Tensor& dense, const at::Tensor& sparse,
                                     const at::Tensor& pi, const at::Tensor& pj);
@@ -197,6 +200,7 @@ TORCH_LIBRARY(trec_amd, m) {
       " -> Tensor");
   m.def("col_sum(Tensor input) -> Tensor");
   m.def("interaction_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
+  m.def("relu_bwd_col_sum(Tensor grad_out, Tensor y) -> (Tensor, Tensor)");
   m.def("interaction_mfma_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
   m.def(
       "interaction_mfma_backward(Tensor grad_out, Tensor dense, Tensor sparse, Tensor pair_col)"
@@ -236,6 +240,7 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("tbe_forward_seq_int8", trec_amd::tbe_forward_seq_int8);
   m.impl("col_sum", trec_amd::col_sum);
   m.impl("interaction_forward", trec_amd::interaction_forward);
+  m.impl("relu_bwd_col_sum", trec_amd::relu_bwd_col_sum);
   m.impl("interaction_mfma_forward", trec_amd::interaction_mfma_forward);
   m.impl("interaction_mfma_backward", trec_amd::interaction_mfma_backward);
   m.impl("interaction_backward", trec_amd::interaction_backward);
