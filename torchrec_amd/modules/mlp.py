@@ -36,17 +36,20 @@ import os
 
 # opt-in: A/B on MI355X showed parity with torch's fused linear backward
 _USE_COLSUM = os.environ.get("TREC_COLSUM_LINEAR", "0") == "1"
-# default-on GPU path: fused relu-mask + bias-grad kernel and split-K wgrad
-# (hipBLASLt's TN heuristic never split-Ks the B=8192-deep weight gradients:
-# measured 52-65 us per layer; chunked bmm halves it)
+# default-on GPU path: split-K weight gradients (hipBLASLt's TN heuristic
+# never split-Ks the B=8192-deep wgrads: measured 52-65 us per layer; chunked
+# bmm halves it — within-box A/B 1.342 vs 1.385 ms/step)
 _USE_FUSED_MLP = os.environ.get("TREC_FUSED_MLP", "1") == "1"
+
+
+_USE_SPLITK = os.environ.get("TREC_SPLITK_WGRAD", "1") == "1"
 
 
 def _splitk_wgrad(g: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     """dW = g^T @ x with K = batch split across a bmm (split-K that
     hipBLASLt's heuristic refuses to pick for TN deep-K shapes)."""
     B = g.shape[0]
-    if B >= 4096 and B % 8 == 0:
+    if _USE_SPLITK and B >= 4096 and B % 8 == 0:
         gv = g.view(8, B // 8, g.shape[1])
         xv = x.view(8, B // 8, x.shape[1])
         return torch.bmm(gv.transpose(1, 2), xv).sum(0)
@@ -68,11 +71,19 @@ class _LinearReLUFused(torch.autograd.Function):
         from torchrec_amd import ops
 
         x, w, y = ctx.saved_tensors
-        ops.hip_ops()
-        g, db = torch.ops.trec_amd.relu_bwd_col_sum(dy.contiguous(), y)
+        # relu_bwd_col_sum kernel: opt-in — the scalar-load version measured
+        # SLOWER than torch's vectorized threshold-backward + reduce pair in
+        # the full step (1.88 vs 1.39 ms/step A/B); split-K wgrad is the win
+        if os.environ.get("TREC_RELU_COLSUM", "0") == "1":
+            ops.hip_ops()
+            g, db = torch.ops.trec_amd.relu_bwd_col_sum(dy.contiguous(), y)
+            db = db.to(w.dtype)
+        else:
+            g = dy * (y > 0)
+            db = g.sum(0)
         dx = g @ w
         dw = _splitk_wgrad(g, x)
-        return dx, dw, db.to(w.dtype)
+        return dx, dw, db
 
 
 def _linear_fwd(linear: nn.Linear, input: torch.Tensor) -> torch.Tensor:
