@@ -247,6 +247,98 @@ class TestTBEGpu:
         )
 
 
+class TestTBEAdam:
+    """Fused Adam / partial-rowwise Adam vs torch.optim.Adam on the same
+    embedding (reference: TBE fused optimizer zoo)."""
+
+    def test_cpu_adam_matches_torch(self):
+        torch.manual_seed(0)
+        specs = [("t0", 40, 8)]
+        tbe = TableBatchedEmbeddingBags(specs, optimizer="adam", learning_rate=0.01)
+        w0 = tbe.split_embedding_weights()[0].clone()
+        ref_w = torch.nn.Parameter(w0.clone())
+        opt = torch.optim.Adam([ref_w], lr=0.01, betas=(0.9, 0.999), eps=tbe.eps)
+        for step in range(4):
+            indices, offsets = make_inputs(specs, B=6, L=3, seed=step)
+            out = tbe(indices, offsets)
+            g = torch.randn_like(out)
+            out.backward(g)
+            # dense reference: same pooled-sum graph over ref_w
+            opt.zero_grad()
+            outs = []
+            B = 6
+            for b in range(B):
+                lo, hi = int(offsets[b]), int(offsets[b + 1])
+                rows = ref_w[indices[lo:hi]]
+                outs.append(rows.sum(0) if hi > lo else torch.zeros(8))
+            torch.stack(outs).backward(g)
+            opt.step()
+        assert torch.allclose(
+            tbe.split_embedding_weights()[0], ref_w.detach(), atol=1e-5, rtol=1e-4
+        )
+
+    @pytest.mark.gpu
+    @pytest.mark.parametrize("optim", ["adam", "partial_rowwise_adam"])
+    def test_gpu_matches_cpu(self, optim):
+        torch.manual_seed(0)
+        specs = [("t0", 100, 8), ("t1", 50, 128), ("t2", 1000, 64)]
+        cpu = TableBatchedEmbeddingBags(specs, optimizer=optim, learning_rate=0.01)
+        gpu = TableBatchedEmbeddingBags(
+            specs, optimizer=optim, learning_rate=0.01, device=torch.device("cuda")
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        for step in range(3):
+            indices, offsets = make_inputs(specs, B=16, L=7, seed=step)
+            out_c = cpu(indices, offsets)
+            out_g = gpu(indices.cuda(), offsets.cuda())
+            grad = torch.randn_like(out_c)
+            out_c.backward(grad)
+            out_g.backward(grad.cuda())
+        torch.cuda.synchronize()
+        for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+            assert torch.allclose(wg.cpu(), wc, atol=1e-5, rtol=1e-4)
+        for sc, sg in zip(cpu.split_optimizer_states(), gpu.split_optimizer_states()):
+            for mc, mg in zip(sc, sg):
+                assert torch.allclose(mg.cpu(), mc, atol=1e-5, rtol=1e-4)
+
+
+class TestStochasticRounding:
+    @pytest.mark.gpu
+    def test_bf16_updates_are_unbiased(self):
+        """Many tiny updates below bf16 resolution: round-to-nearest freezes
+        the weight, stochastic rounding tracks the fp32 trajectory in
+        expectation (reference TBE stochastic_rounding)."""
+        torch.manual_seed(0)
+        specs = [("t0", 64, 64)]
+        tbe = TableBatchedEmbeddingBags(
+            specs, optimizer="sgd", learning_rate=1.0,
+            weights_precision="bf16", stochastic_rounding=True,
+            device=torch.device("cuda"),
+        )
+        tbe.weights.data.fill_(1.0)
+        rtn = TableBatchedEmbeddingBags(
+            specs, optimizer="sgd", learning_rate=1.0,
+            weights_precision="bf16", stochastic_rounding=False,
+            device=torch.device("cuda"),
+        )
+        rtn.weights.data.fill_(1.0)
+        # per-update delta 1e-3: far below the bf16 ulp at 1.0 (2^-8)
+        indices = torch.arange(64, device="cuda")
+        offsets = torch.arange(65, device="cuda")
+        g = torch.full((64, 64), 1e-3, device="cuda")
+        for _ in range(200):
+            out = tbe(indices, offsets)
+            out.backward(g)
+            out2 = rtn(indices, offsets)
+            out2.backward(g)
+        torch.cuda.synchronize()
+        # fp32 trajectory: 1.0 - 200 * 1e-3 = 0.8
+        got = tbe.weights.data.float().mean().item()
+        frozen = rtn.weights.data.float().mean().item()
+        assert abs(frozen - 1.0) < 1e-3, "round-to-nearest should freeze"
+        assert abs(got - 0.8) < 0.02, f"stochastic mean {got} should track 0.8"
+
+
 class TestTBEOutputDtype:
     """output_dtype (reference SplitTBE): fp32 accumulate, one bf16 round on
     store; backward consumes the matching-precision gradient directly."""

@@ -186,6 +186,9 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
                 ],
                 fixed_bag_length=fused_params.get("fixed_bag_length"),
                 output_dtype=fused_params.get("output_dtype", "fp32"),
+                beta1=fused_params.get("beta1", 0.9),
+                beta2=fused_params.get("beta2", 0.999),
+                stochastic_rounding=fused_params.get("stochastic_rounding"),
             )
             self._emb_modules.append(tbe)
             nf = sum(len(t.feature_names) for t in group)

@@ -57,7 +57,10 @@ void tbe_backward_fused(at::Tensor weights, at::Tensor momentum, const at::Tenso
                         const at::Tensor& table_elem_offsets, const at::Tensor& dims,
                         int64_t max_D, double lr, double eps, int64_t mode,
                         at::Tensor grad_weights, at::Tensor cache_weights,
-                        const at::Tensor& cache_loc);
+                        const at::Tensor& cache_loc,
+                        at::Tensor m1, at::Tensor m2, double beta1, double beta2,
+                        const at::Tensor& iter_t, const at::Tensor& rng_state,
+                        bool stochastic);
 at::Tensor gather_run_heads(const at::Tensor& sorted_linear, const at::Tensor& seg_offsets,
                             const at::Tensor& num_runs);
 at::Tensor tbe_grad_per_sample_weights(const at::Tensor& weights,
@@ -166,7 +169,9 @@ TORCH_LIBRARY(trec_amd, m) {
       " Tensor sorted_linear, Tensor sort_perm, Tensor seg_offsets, Tensor num_runs,"
       " Tensor pos_row, Tensor pos_col, Tensor pos_scale, Tensor table_row_offsets,"
       " Tensor table_elem_offsets, Tensor dims, int max_D, float lr, float eps, int mode,"
-      " Tensor(c!) grad_weights, Tensor(d!) cache_weights, Tensor cache_loc) -> ()");
+      " Tensor(c!) grad_weights, Tensor(d!) cache_weights, Tensor cache_loc,"
+      " Tensor(e!) m1, Tensor(f!) m2, float beta1, float beta2, Tensor iter_t,"
+      " Tensor rng_state, bool stochastic) -> ()");
   m.def("gather_run_heads(Tensor sorted_linear, Tensor seg_offsets, Tensor num_runs) -> Tensor");
   m.def(
       "lxu_cache_populate(Tensor(a!) host_weights, Tensor table_row_offsets,"

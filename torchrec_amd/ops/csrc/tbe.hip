@@ -49,6 +49,50 @@ static T* uvm_ptr(const at::Tensor& t) {
   return static_cast<T*>(dp);
 }
 
+// ---------------------------------------------------------------------------
+// stochastic rounding (bf16): add 16 random bits below the kept mantissa and
+// truncate — unbiased, carries across the rounding boundary (reference: TBE
+// stochastic_rounding for low-precision weights). fp16/fp32 stores stay
+// round-to-nearest (fp16 is not a bit-truncation of fp32).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint64_t mix64(uint64_t k) {
+  k ^= k >> 33;
+  k *= 0xff51afd7ed558ccdULL;
+  k ^= k >> 33;
+  k *= 0xc4ceb9fe1a85ec53ULL;
+  k ^= k >> 33;
+  return k;
+}
+
+__device__ __forceinline__ unsigned short stoch_bf16_bits(float v, uint32_t r16) {
+  uint32_t u = __float_as_uint(v);
+  if ((u & 0x7F800000u) != 0x7F800000u) u += (r16 & 0xFFFFu);
+  return (unsigned short)(u >> 16);
+}
+
+template <typename emb_t>
+__device__ __forceinline__ void store4_maybe_stoch(emb_t* row, int col4, float4 v,
+                                                   uint64_t rk, bool stoch) {
+  Vec4<emb_t>::store(row, col4, v);
+}
+
+template <>
+__device__ __forceinline__ void store4_maybe_stoch<__hip_bfloat16>(
+    __hip_bfloat16* row, int col4, float4 v, uint64_t rk, bool stoch) {
+  if (!stoch) {
+    Vec4<__hip_bfloat16>::store(row, col4, v);
+    return;
+  }
+  uint64_t r = mix64(rk);
+  ushort4 q;
+  q.x = stoch_bf16_bits(v.x, (uint32_t)(r));
+  q.y = stoch_bf16_bits(v.y, (uint32_t)(r >> 16));
+  q.z = stoch_bf16_bits(v.z, (uint32_t)(r >> 32));
+  q.w = stoch_bf16_bits(v.w, (uint32_t)(r >> 48));
+  reinterpret_cast<ushort4*>(row)[col4] = q;
+}
+
 static inline hipStream_t tbe_stream() {
   return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
@@ -715,11 +759,15 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     const int64_t* __restrict__ table_elem_offsets,  // [T]
     const int32_t* __restrict__ dims, int T, float lr, float eps, int mode,
     emb_t* __restrict__ grad_weights, float* __restrict__ cache_weights,
-    const int32_t* __restrict__ cache_loc, int64_t cache_stride) {
+    const int32_t* __restrict__ cache_loc, int64_t cache_stride,
+    float* __restrict__ m1, float* __restrict__ m2, float beta1, float beta2,
+    const float* __restrict__ iter_ptr, const int64_t* __restrict__ rng_state,
+    int stochastic) {
   int sl = threadIdx.x % LPS;
   int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
   int64_t n_slots = (static_cast<int64_t>(gridDim.x) * blockDim.x) / LPS;
   int32_t num_runs = *num_runs_ptr;
+  const uint64_t rng_base = rng_state ? (uint64_t)*rng_state : 0;
   for (int64_t r = slot; r < num_runs; r += n_slots) {
     int32_t k0 = seg_offsets[r], k1 = seg_offsets[r + 1];
     int64_t lin = sorted_linear[k0];
@@ -793,7 +841,8 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
           w.z -= step * acc[c].z;
           w.w -= step * acc[c].w;
           if (crow) Vec4<float>::store(crow, col4, w);
-          else Vec4<emb_t>::store(wrow, col4, w);
+          else store4_maybe_stoch(wrow, col4, w, rng_base ^ (uint64_t)(lin * 1024 + col4),
+                                  stochastic != 0);
         }
       }
     } else if (mode == 0) {
@@ -807,7 +856,71 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
           w.z -= lr * acc[c].z;
           w.w -= lr * acc[c].w;
           if (crow) Vec4<float>::store(crow, col4, w);
-          else Vec4<emb_t>::store(wrow, col4, w);
+          else store4_maybe_stoch(wrow, col4, w, rng_base ^ (uint64_t)(lin * 1024 + col4),
+                                  stochastic != 0);
+        }
+      }
+    } else if (mode == 3 || mode == 4) {
+      // Adam (mode 3: elementwise m2) / partial-rowwise Adam (mode 4: one m2
+      // scalar per row = beta2-EMA of mean(g^2)). Reference semantics: TBE
+      // fused optimizers (batched_embedding_kernel.py:36-53 imports).
+      float t_iter = iter_ptr ? *iter_ptr : 1.f;
+      float bc1 = 1.f / (1.f - __powf(beta1, t_iter));
+      float bc2 = 1.f / (1.f - __powf(beta2, t_iter));
+      int64_t ebase = table_elem_offsets[t] + local * static_cast<int64_t>(D);
+      float inv_sqrt_row = 0.f;
+      if (mode == 4) {
+        float gsq = 0.f;
+#pragma unroll
+        for (int c = 0; c < CHUNKS; ++c) {
+          int col4 = c * LPS + sl;
+          if (col4 * 4 < D)
+            gsq += acc[c].x * acc[c].x + acc[c].y * acc[c].y + acc[c].z * acc[c].z +
+                   acc[c].w * acc[c].w;
+        }
+        gsq = group_reduce_sum<LPS>(gsq);
+        float v = beta2 * m2[lin] + (1.f - beta2) * gsq / static_cast<float>(D);
+        if (sl == 0) m2[lin] = v;
+        inv_sqrt_row = 1.f / (sqrtf(v * bc2) + eps);
+      }
+#pragma unroll
+      for (int c = 0; c < CHUNKS; ++c) {
+        int col4 = c * LPS + sl;
+        if (col4 * 4 < D) {
+          float4* m1p = reinterpret_cast<float4*>(m1 + ebase);
+          float4 mv = m1p[col4];
+          mv.x = beta1 * mv.x + (1.f - beta1) * acc[c].x;
+          mv.y = beta1 * mv.y + (1.f - beta1) * acc[c].y;
+          mv.z = beta1 * mv.z + (1.f - beta1) * acc[c].z;
+          mv.w = beta1 * mv.w + (1.f - beta1) * acc[c].w;
+          m1p[col4] = mv;
+          float4 upd;
+          if (mode == 3) {
+            float4* m2p = reinterpret_cast<float4*>(m2 + ebase);
+            float4 vv = m2p[col4];
+            vv.x = beta2 * vv.x + (1.f - beta2) * acc[c].x * acc[c].x;
+            vv.y = beta2 * vv.y + (1.f - beta2) * acc[c].y * acc[c].y;
+            vv.z = beta2 * vv.z + (1.f - beta2) * acc[c].z * acc[c].z;
+            vv.w = beta2 * vv.w + (1.f - beta2) * acc[c].w * acc[c].w;
+            m2p[col4] = vv;
+            upd.x = lr * mv.x * bc1 / (sqrtf(vv.x * bc2) + eps);
+            upd.y = lr * mv.y * bc1 / (sqrtf(vv.y * bc2) + eps);
+            upd.z = lr * mv.z * bc1 / (sqrtf(vv.z * bc2) + eps);
+            upd.w = lr * mv.w * bc1 / (sqrtf(vv.w * bc2) + eps);
+          } else {
+            upd.x = lr * mv.x * bc1 * inv_sqrt_row;
+            upd.y = lr * mv.y * bc1 * inv_sqrt_row;
+            upd.z = lr * mv.z * bc1 * inv_sqrt_row;
+            upd.w = lr * mv.w * bc1 * inv_sqrt_row;
+          }
+          float4 w = crow ? Vec4<float>::load(crow, col4) : Vec4<emb_t>::load(wrow, col4);
+          w.x -= upd.x;
+          w.y -= upd.y;
+          w.z -= upd.z;
+          w.w -= upd.w;
+          if (crow) Vec4<float>::store(crow, col4, w);
+          else store4_maybe_stoch(wrow, col4, w, rng_base ^ (uint64_t)(lin * 1024 + col4),
+                                  stochastic != 0);
         }
       }
     } else {
@@ -832,6 +945,8 @@ static void launch_tbe_bwd(
     const at::Tensor& table_elem_offsets, const at::Tensor& dims, int T,
     int chunk_size, int64_t max_D, float lr, float eps, int mode,
     at::Tensor& grad_weights, float* cache_w_ptr, const int32_t* cache_loc_ptr,
+    float* m1_ptr, float* m2_ptr, float beta1, float beta2,
+    const float* iter_ptr, const int64_t* rng_ptr, int stochastic,
     int lps, int chunks, int grid, int grid_long, hipStream_t stream) {
   const g_t* grad_ptr = reinterpret_cast<const g_t*>(grad.data_ptr());
   emb_t* gw_ptr = grad_weights.numel() > 0
@@ -858,7 +973,8 @@ static void launch_tbe_bwd(
                        table_row_offsets.data_ptr<int64_t>(),                                \
                        table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,  \
                        lr, eps, mode, gw_ptr, cache_w_ptr,                                   \
-                       cache_loc_ptr, max_D);                                                \
+                       cache_loc_ptr, max_D, m1_ptr, m2_ptr, beta1, beta2, iter_ptr,         \
+                       rng_ptr, stochastic);                                                 \
   } while (0)
   if (lps == 16) TBE_BWD_LAUNCH(16, 1);
   else if (lps == 32) TBE_BWD_LAUNCH(32, 1);
@@ -880,12 +996,15 @@ static void dispatch_tbe_bwd_grad(const at::ScalarType g_st, at::Tensor& weights
     const float* scale_ptr, const at::Tensor& table_row_offsets,
     const at::Tensor& table_elem_offsets, const at::Tensor& dims, int T, int chunk_size,
     int64_t max_D, float lr, float eps, int mode, at::Tensor& grad_weights,
-    float* cache_w_ptr, const int32_t* cache_loc_ptr, int lps, int chunks, int grid,
+    float* cache_w_ptr, const int32_t* cache_loc_ptr,
+    float* m1_ptr, float* m2_ptr, float beta1, float beta2, const float* iter_ptr,
+    const int64_t* rng_ptr, int stochastic, int lps, int chunks, int grid,
     int grid_long, hipStream_t stream) {
 #define ARGS weights, momentum, grad, sorted_linear, sort_perm, seg_offsets, num_runs,  \
     chunk_offsets, total_chunks, scratch, pos_row, pos_col, scale_ptr,                  \
     table_row_offsets, table_elem_offsets, dims, T, chunk_size, max_D, lr, eps, mode,   \
-    grad_weights, cache_w_ptr, cache_loc_ptr, lps, chunks, grid, grid_long, stream
+    grad_weights, cache_w_ptr, cache_loc_ptr, m1_ptr, m2_ptr, beta1, beta2, iter_ptr,   \
+    rng_ptr, stochastic, lps, chunks, grid, grid_long, stream
   if (g_st == at::kFloat) launch_tbe_bwd<emb_t, float, host_w_t>(ARGS);
   else if (g_st == at::kBFloat16) launch_tbe_bwd<emb_t, __hip_bfloat16, host_w_t>(ARGS);
   else if (g_st == at::kHalf) launch_tbe_bwd<emb_t, __half, host_w_t>(ARGS);
@@ -900,7 +1019,9 @@ void tbe_backward_fused(
     const at::Tensor& pos_col, const at::Tensor& pos_scale,
     const at::Tensor& table_row_offsets, const at::Tensor& table_elem_offsets,
     const at::Tensor& dims, int64_t max_D, double lr, double eps, int64_t mode,
-    at::Tensor grad_weights, at::Tensor cache_weights, const at::Tensor& cache_loc) {
+    at::Tensor grad_weights, at::Tensor cache_weights, const at::Tensor& cache_loc,
+    at::Tensor m1, at::Tensor m2, double beta1, double beta2,
+    const at::Tensor& iter_t, const at::Tensor& rng_state, bool stochastic) {
   int64_t n = sorted_linear.numel();
   if (n == 0) return;
   int T = table_elem_offsets.numel();
@@ -920,11 +1041,17 @@ void tbe_backward_fused(
   auto scratch = at::empty({max_chunks * max_D}, grad.options().dtype(at::kFloat));
   int grid_long = grid_for(max_chunks * lps, kBlockThreads);
 
+  float* m1_ptr = m1.numel() > 0 ? uvm_ptr<float>(m1) : nullptr;
+  float* m2_ptr = m2.numel() > 0 ? uvm_ptr<float>(m2) : nullptr;
+  const float* iter_ptr = iter_t.numel() > 0 ? iter_t.data_ptr<float>() : nullptr;
+  const int64_t* rng_ptr =
+      rng_state.numel() > 0 ? rng_state.data_ptr<int64_t>() : nullptr;
 #define BARGS weights, momentum, grad, sorted_linear, sort_perm, seg_offsets, num_runs, \
     chunk_offsets, total_chunks, scratch, pos_row, pos_col, scale_ptr,                  \
     table_row_offsets, table_elem_offsets, dims, T, kChunkSize, max_D, (float)lr,       \
-    (float)eps, (int)mode, grad_weights, cache_w_ptr, cache_loc_ptr, lps, chunks, grid, \
-    grid_long, stream
+    (float)eps, (int)mode, grad_weights, cache_w_ptr, cache_loc_ptr, m1_ptr, m2_ptr,    \
+    (float)beta1, (float)beta2, iter_ptr, rng_ptr, stochastic ? 1 : 0, lps, chunks,     \
+    grid, grid_long, stream
   auto g_st = grad.scalar_type();
   auto w_st = weights.scalar_type();
   if (w_st == at::kFloat) dispatch_tbe_bwd_grad<float, float>(g_st, BARGS);

@@ -48,8 +48,16 @@ class EmbeddingSpec(NamedTuple):
 OPT_SGD = 0
 OPT_ROWWISE_ADAGRAD = 1
 OPT_DENSE = 2  # no fused update; gradient surfaces through autograd
+OPT_ADAM = 3  # elementwise Adam (m1 + m2 per element, bias-corrected)
+OPT_PARTIAL_ROWWISE_ADAM = 4  # m1 per element, m2 one scalar per row
 
-_OPT_NAMES = {"sgd": OPT_SGD, "rowwise_adagrad": OPT_ROWWISE_ADAGRAD, "dense": OPT_DENSE}
+_OPT_NAMES = {
+    "sgd": OPT_SGD,
+    "rowwise_adagrad": OPT_ROWWISE_ADAGRAD,
+    "dense": OPT_DENSE,
+    "adam": OPT_ADAM,
+    "partial_rowwise_adam": OPT_PARTIAL_ROWWISE_ADAM,
+}
 
 
 def _bits_needed(n: int) -> int:
@@ -197,6 +205,9 @@ class TableBatchedEmbeddingBags(nn.Module):
         weights_precision: str = "fp32",
         fixed_bag_length: Optional[int] = None,
         output_dtype: str = "fp32",
+        beta1: float = 0.9,
+        beta2: float = 0.999,
+        stochastic_rounding: Optional[bool] = None,
     ) -> None:
         super().__init__()
         device = device or torch.device("cpu")
@@ -290,6 +301,30 @@ class TableBatchedEmbeddingBags(nn.Module):
             self.register_buffer("momentum", mom)
         else:
             self.register_buffer("momentum", torch.empty(0, device=weights_device))
+        # Adam state: m1 per element; m2 per element (adam) or per row
+        # (partial_rowwise_adam). Reference: TBE fused Adam/partial-rowwise.
+        self.beta1, self.beta2 = beta1, beta2
+        if self.optimizer in (OPT_ADAM, OPT_PARTIAL_ROWWISE_ADAM):
+            m1 = torch.zeros(self._total_elems, dtype=torch.float32, device=weights_device)
+            m2n = self._total_elems if self.optimizer == OPT_ADAM else self._total_rows
+            m2 = torch.zeros(m2n, dtype=torch.float32, device=weights_device)
+            if self._uvm:
+                m1, m2 = m1.pin_memory(), m2.pin_memory()
+            self.register_buffer("m1", m1)
+            self.register_buffer("m2", m2)
+        else:
+            self.register_buffer("m1", torch.empty(0, device=weights_device))
+            self.register_buffer("m2", torch.empty(0, device=weights_device))
+        # device-side step counter (bias correction) + RNG state (stochastic
+        # rounding): read by the kernel, bumped by in-graph device ops so
+        # hipGraph replays advance them
+        self.register_buffer("_iter", torch.zeros(1, dtype=torch.float32, device=device))
+        self.register_buffer(
+            "_rng", torch.randint(1, 1 << 62, (1,), dtype=torch.int64, device=device)
+        )
+        if stochastic_rounding is None:
+            stochastic_rounding = weights_precision != "fp32"
+        self.stochastic_rounding = bool(stochastic_rounding)
 
         def reg(name: str, t: torch.Tensor) -> None:
             self.register_buffer(name, t.to(device), persistent=False)
@@ -382,13 +417,25 @@ class TableBatchedEmbeddingBags(nn.Module):
         return out
 
     def split_optimizer_states(self) -> List[List[torch.Tensor]]:
-        if self.optimizer != OPT_ROWWISE_ADAGRAD:
-            return [[] for _ in self._specs]
-        out = []
-        for i, s in enumerate(self._specs):
-            start = int(self._table_row_offsets[i])
-            out.append([self.momentum[start : start + s.rows]])
-        return out
+        if self.optimizer == OPT_ROWWISE_ADAGRAD:
+            out = []
+            for i, s in enumerate(self._specs):
+                start = int(self._table_row_offsets[i])
+                out.append([self.momentum[start : start + s.rows]])
+            return out
+        if self.optimizer in (OPT_ADAM, OPT_PARTIAL_ROWWISE_ADAM):
+            out = []
+            for i, s in enumerate(self._specs):
+                e0 = int(self._table_elem_offsets[i])
+                r0 = int(self._table_row_offsets[i])
+                m1 = self.m1[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
+                if self.optimizer == OPT_ADAM:
+                    m2 = self.m2[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
+                else:
+                    m2 = self.m2[r0 : r0 + s.rows]
+                out.append([m1, m2])
+            return out
+        return [[] for _ in self._specs]
 
     def set_learning_rate(self, lr: float) -> None:
         self.learning_rate = lr
@@ -398,6 +445,14 @@ class TableBatchedEmbeddingBags(nn.Module):
         return self._specs
 
     # -- backward (GPU fused path) ------------------------------------------
+
+    def _pre_update(self) -> None:
+        """Advance the device-side step counter / RNG state (in-graph ops —
+        hipGraph replays advance them too)."""
+        if self.optimizer in (OPT_ADAM, OPT_PARTIAL_ROWWISE_ADAM):
+            self._iter.add_(1.0)
+        if self.stochastic_rounding:
+            self._rng.add_(0x9E3779B97F4A7C15 & ((1 << 62) - 1))
 
     def _bag_metadata(
         self, indices: torch.Tensor, offsets: torch.Tensor, B: int
@@ -482,6 +537,7 @@ class TableBatchedEmbeddingBags(nn.Module):
                 linear, _bits_needed(self._total_rows)
             )
         seg_offsets, num_runs = torch.ops.trec_amd.tbe_backward_prep(sorted_lin)
+        self._pre_update()
         torch.ops.trec_amd.tbe_backward_fused(
             self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
             self.momentum,
@@ -503,6 +559,13 @@ class TableBatchedEmbeddingBags(nn.Module):
             grad_weights if grad_weights is not None else self._empty_f,
             self.cache_weights,
             cache_loc if cache_loc is not None else self._empty_i,
+            self.m1,
+            self.m2,
+            self.beta1,
+            self.beta2,
+            self._iter,
+            self._rng,
+            self.stochastic_rounding,
         )
         return grad_psw_out
 
@@ -678,6 +741,7 @@ class TableBatchedEmbeddingBags(nn.Module):
             scale = psw
         sorted_lin, perm = torch.ops.trec_amd.sort_pairs(linear, _bits_needed(self._total_rows))
         seg_offsets, num_runs = torch.ops.trec_amd.tbe_backward_prep(sorted_lin)
+        self._pre_update()
         torch.ops.trec_amd.tbe_backward_fused(
             self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
             self.momentum,
@@ -699,6 +763,13 @@ class TableBatchedEmbeddingBags(nn.Module):
             self._empty_f,
             self.cache_weights,
             self._empty_i,
+            self.m1,
+            self.m2,
+            self.beta1,
+            self.beta2,
+            self._iter,
+            self._rng,
+            self.stochastic_rounding,
         )
 
     def _backward_seq(
@@ -717,6 +788,7 @@ class TableBatchedEmbeddingBags(nn.Module):
         linear = indices + self._feat_row_offset[f]
         sorted_lin, perm = torch.ops.trec_amd.sort_pairs(linear, _bits_needed(self._total_rows))
         seg_offsets, num_runs = torch.ops.trec_amd.tbe_backward_prep(sorted_lin)
+        self._pre_update()
         torch.ops.trec_amd.tbe_backward_fused(
             self.weights if not isinstance(self.weights, nn.Parameter) else self.weights.data,
             self.momentum,
@@ -738,6 +810,13 @@ class TableBatchedEmbeddingBags(nn.Module):
             self._empty_f,
             self.cache_weights,
             self._empty_i,
+            self.m1,
+            self.m2,
+            self.beta1,
+            self.beta2,
+            self._iter,
+            self._rng,
+            self.stochastic_rounding,
         )
 
     # -- CPU oracle path -----------------------------------------------------
@@ -770,6 +849,27 @@ class TableBatchedEmbeddingBags(nn.Module):
                         w.float()
                         - self.learning_rate * g / (m.sqrt() + self.eps).unsqueeze(1)
                     )
+            elif self.optimizer in (OPT_ADAM, OPT_PARTIAL_ROWWISE_ADAM):
+                self._iter += 1.0
+                t = float(self._iter)
+                bc1 = 1.0 / (1.0 - self.beta1 ** t)
+                bc2 = 1.0 / (1.0 - self.beta2 ** t)
+                for i, s_ in enumerate(self._specs):
+                    e0 = int(self._table_elem_offsets[i])
+                    r0 = int(self._table_row_offsets[i])
+                    g = gf[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
+                    m1 = self.m1[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
+                    m1.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+                    if self.optimizer == OPT_ADAM:
+                        m2 = self.m2[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
+                        m2.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+                        denom = (m2 * bc2).sqrt() + self.eps
+                    else:
+                        m2 = self.m2[r0 : r0 + s_.rows]
+                        m2.mul_(self.beta2).add_(g.pow(2).mean(dim=1), alpha=1 - self.beta2)
+                        denom = ((m2 * bc2).sqrt() + self.eps).unsqueeze(1)
+                    w = self.weights[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
+                    w.copy_(w.float() - self.learning_rate * (m1 * bc1) / denom)
             elif self.optimizer == OPT_SGD:
                 self.weights.copy_(self.weights.float() - self.learning_rate * gf)
 
