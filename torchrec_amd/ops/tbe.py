@@ -858,18 +858,28 @@ class TableBatchedEmbeddingBags(nn.Module):
                     e0 = int(self._table_elem_offsets[i])
                     r0 = int(self._table_row_offsets[i])
                     g = gf[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
+                    # SPARSE Adam (reference TBE): only rows that received a
+                    # gradient this step update their moments/weights
+                    touched = g.abs().amax(dim=1) > 0
                     m1 = self.m1[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
-                    m1.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+                    m1[touched] = self.beta1 * m1[touched] + (1 - self.beta1) * g[touched]
                     if self.optimizer == OPT_ADAM:
                         m2 = self.m2[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
-                        m2.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
-                        denom = (m2 * bc2).sqrt() + self.eps
+                        m2[touched] = (
+                            self.beta2 * m2[touched]
+                            + (1 - self.beta2) * g[touched].pow(2)
+                        )
+                        denom = (m2[touched] * bc2).sqrt() + self.eps
                     else:
                         m2 = self.m2[r0 : r0 + s_.rows]
-                        m2.mul_(self.beta2).add_(g.pow(2).mean(dim=1), alpha=1 - self.beta2)
-                        denom = ((m2 * bc2).sqrt() + self.eps).unsqueeze(1)
+                        m2[touched] = (
+                            self.beta2 * m2[touched]
+                            + (1 - self.beta2) * g[touched].pow(2).mean(dim=1)
+                        )
+                        denom = ((m2[touched] * bc2).sqrt() + self.eps).unsqueeze(1)
                     w = self.weights[e0 : e0 + s_.rows * s_.dim].view(s_.rows, s_.dim)
-                    w.copy_(w.float() - self.learning_rate * (m1 * bc1) / denom)
+                    upd = w[touched].float() - self.learning_rate * (m1[touched] * bc1) / denom
+                    w[touched] = upd.to(w.dtype)
             elif self.optimizer == OPT_SGD:
                 self.weights.copy_(self.weights.float() - self.learning_rate * gf)
 
