@@ -78,6 +78,81 @@ def test_dmp_collection_2d():
     run_multi_process(_run_2d_test, 4, "gloo")
 
 
+def _run_2d_fully_sharded(rank, world_size):
+    """FULLY_SHARDED golden test: per-iter RS/AG across replicas must land on
+    the SAME trajectory as replicated + allreduce(AVG) sync every step
+    (reference model_parallel.py:1043, batched_embedding_kernel.py:2674)."""
+    S = 2
+    B = 4
+    tables = make_tables()
+    torch.manual_seed(42)
+
+    def build(strategy):
+        model = SparseModel(make_tables())
+        sharder = EmbeddingBagCollectionSharder(
+            fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+        )
+        planner = EmbeddingShardingPlanner(
+            topology=Topology(world_size=S, compute_device="cpu", hbm_cap=1 << 40),
+            constraints={
+                cfg.name: ParameterConstraints(
+                    sharding_types=[ShardingType.TABLE_WISE.value]
+                )
+                for cfg in tables
+            },
+        )
+        plan = planner.plan(model, [sharder])
+        dmp = DMPCollection(
+            model, sharding_group_size=S, plan=plan, sharders=[sharder],
+            sync_interval=1, sharding_strategy=strategy,
+        )
+        golden = _golden(tables, None, world_size)
+        dmp.load_state_dict(
+            {
+                f"sparse.embedding_bags.{cfg.name}.weight": w
+                for cfg, w in zip(tables, golden.split_embedding_weights())
+            },
+            strict=False,
+        )
+        if strategy == "fully_sharded":
+            dmp._init_fully_sharded()  # re-slice after the state load
+        return dmp
+
+    rep = build("replicated")
+    fs = build("fully_sharded")
+    group = rank // S
+    rank_in_group = rank % S
+    for step in range(3):
+        kjt_group = make_global_kjt(tables, B * S, seed=10 * step + group)
+        kjt_local = kjt_local_slice(kjt_group, rank_in_group * B, (rank_in_group + 1) * B)
+        for dmp in (rep, fs):
+            kt = dmp(kjt_local)
+            kt.values().sum().backward()
+            dmp.maybe_sync()
+    sd_rep = rep.state_dict()
+    sd_fs = fs.state_dict()
+    from torch.distributed._shard.sharded_tensor import ShardedTensor
+
+    for k, v in sd_rep.items():
+        other = sd_fs[k]
+        if isinstance(v, ShardedTensor):
+            for sa, sb in zip(v.local_shards(), other.local_shards()):
+                torch.testing.assert_close(sb.tensor, sa.tensor, atol=1e-5, rtol=1e-5)
+        elif isinstance(v, torch.Tensor) and v.numel():
+            torch.testing.assert_close(other, v, atol=1e-5, rtol=1e-5)
+    # replica peers agree after gather
+    for sharded in fs.sharded_modules().values():
+        for tbe in sharded.tbes():
+            w = tbe.weights
+            peers = [torch.empty_like(w) for _ in range(2)]
+            dist.all_gather(peers, w.detach(), group=fs._replica_pg)
+            torch.testing.assert_close(peers[0], peers[1], atol=1e-6, rtol=1e-6)
+
+
+def test_dmp_collection_2d_fully_sharded():
+    run_multi_process(_run_2d_fully_sharded, 4, "gloo")
+
+
 def _run_2d_mixed(rank, world_size):
     """2D + MIXED sharding types per group: exercises the globally-coordinated
     per-sharding communicator creation (env.all_group_ranks)."""

@@ -231,6 +231,7 @@ class DMPCollection(nn.Module):
         plan: Optional[ShardingPlan] = None,
         sharders: Optional[List[ModuleSharder[nn.Module]]] = None,
         sync_interval: int = 1,
+        sharding_strategy: str = "replicated",
     ) -> None:
         super().__init__()
         assert dist.is_initialized(), "DMPCollection needs torch.distributed"
@@ -275,6 +276,17 @@ class DMPCollection(nn.Module):
         )
         self._sync_interval = sync_interval
         self._step = 0
+        # FULLY_SHARDED (reference model_parallel.py:1043 + ShardedBatched-
+        # FusedEmbeddingBag batched_embedding_kernel.py:2674): each replica
+        # slot persistently owns 1/R of every sharded weight/state tensor;
+        # per sync the full buffers reduce-scatter(AVG) into the owned slices
+        # and all-gather back before the next forward.
+        assert sharding_strategy in ("replicated", "fully_sharded")
+        self._strategy = sharding_strategy
+        self._owned_slices: List[torch.Tensor] = []
+        self._needs_gather = False
+        if self._strategy == "fully_sharded":
+            self._init_fully_sharded()
 
     @property
     def fused_optimizer(self) -> KeyedOptimizer:
@@ -292,12 +304,16 @@ class DMPCollection(nn.Module):
         return self._dmp.sharded_modules()
 
     def state_dict(self, *args, **kwargs):
+        if self._strategy == "fully_sharded" and self._needs_gather:
+            self._gather_weights()  # full buffers are stale after a RS sync
         return self._dmp.state_dict(*args, **kwargs)
 
     def load_state_dict(self, *args, **kwargs):
         return self._dmp.load_state_dict(*args, **kwargs)
 
     def forward(self, *args, **kwargs):
+        if self._strategy == "fully_sharded" and self._needs_gather:
+            self._gather_weights()
         return self._dmp(*args, **kwargs)
 
     def maybe_sync(self) -> None:
@@ -306,21 +322,82 @@ class DMPCollection(nn.Module):
         if self._step % self._sync_interval == 0:
             self.sync()
 
-    @torch.no_grad()
-    def sync(self, include_optimizer_state: bool = True) -> None:
-        """Average sharded weights (and fused-optimizer state) across the
-        replica group (reference model_parallel.py:1402)."""
-        R = dist.get_world_size(self._replica_pg)
-        if R <= 1:
-            return
+    def _sync_tensors(self, include_optimizer_state: bool = True) -> List[torch.Tensor]:
         tensors: List[torch.Tensor] = []
         for sharded in self._dmp.sharded_modules().values():
             for tbe in getattr(sharded, "tbes", lambda: [])():
                 inner = getattr(tbe, "_bags", tbe)
                 w = inner.weights
                 tensors.append(w.data if isinstance(w, nn.Parameter) else w)
-                if include_optimizer_state and inner.momentum.numel():
-                    tensors.append(inner.momentum)
+                if include_optimizer_state:
+                    for st in ("momentum", "m1", "m2"):
+                        t = getattr(inner, st, None)
+                        if t is not None and t.numel():
+                            tensors.append(t)
+        return tensors
+
+    @torch.no_grad()
+    def _init_fully_sharded(self) -> None:
+        """Align replicas, then carve this slot's owned slice of each tensor."""
+        R = dist.get_world_size(self._replica_pg)
+        self._owned_slices = []
+        for t in self._sync_tensors():
+            if R > 1:
+                dist.all_reduce(t, group=self._replica_pg)
+                t.div_(R)
+            n = t.numel()
+            if R > 1 and n % R == 0:
+                slot = dist.get_rank(self._replica_pg)
+                sl = n // R
+                self._owned_slices.append(
+                    t.view(-1)[slot * sl : (slot + 1) * sl].clone()
+                )
+            else:
+                self._owned_slices.append(torch.empty(0, device=t.device))
+        self._needs_gather = False
+
+    @torch.no_grad()
+    def _gather_weights(self) -> None:
+        R = dist.get_world_size(self._replica_pg)
+        works = []
+        for t, own in zip(self._sync_tensors(), self._owned_slices):
+            if own.numel():
+                works.append(
+                    dist.all_gather_into_tensor(
+                        t.view(-1), own, group=self._replica_pg, async_op=True
+                    )
+                )
+        for w in works:
+            w.wait()
+        self._needs_gather = False
+
+    @torch.no_grad()
+    def _fully_sharded_sync(self) -> None:
+        R = dist.get_world_size(self._replica_pg)
+        if R <= 1:
+            return
+        for t, own in zip(self._sync_tensors(), self._owned_slices):
+            if own.numel():
+                dist.reduce_scatter_tensor(own, t.view(-1), group=self._replica_pg)
+                own.div_(R)
+            else:
+                # tensor numel not divisible by R: plain average
+                dist.all_reduce(t, group=self._replica_pg)
+                t.div_(R)
+        self._needs_gather = True
+
+    @torch.no_grad()
+    def sync(self, include_optimizer_state: bool = True) -> None:
+        """Average sharded weights (and fused-optimizer state) across the
+        replica group (reference model_parallel.py:1402). Under FULLY_SHARDED
+        this is the reduce-scatter half of the RS/AG cycle."""
+        if self._strategy == "fully_sharded":
+            self._fully_sharded_sync()
+            return
+        R = dist.get_world_size(self._replica_pg)
+        if R <= 1:
+            return
+        tensors = self._sync_tensors(include_optimizer_state)
         works = []
         for t in tensors:
             works.append(dist.all_reduce(t, group=self._replica_pg, async_op=True))
