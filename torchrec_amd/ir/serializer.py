@@ -68,7 +68,26 @@ class JsonSerializer:
                 "need_indices": module.need_indices(),
             }
         else:
-            raise NotImplementedError(f"cannot serialize {type(module)}")
+            from torchrec_amd.modules.feature_processor import (
+                FeatureProcessedEmbeddingBagCollection,
+                PositionWeightedModuleCollection,
+            )
+
+            if isinstance(module, FeatureProcessedEmbeddingBagCollection):
+                fp = module._feature_processors
+                if not isinstance(fp, PositionWeightedModuleCollection):
+                    raise NotImplementedError(
+                        "only PositionWeightedModuleCollection FP-EBCs serialize"
+                    )
+                payload = {
+                    "type": "FeatureProcessedEmbeddingBagCollection",
+                    "tables": [
+                        _config_to_dict(c) for c in module.embedding_bag_configs()
+                    ],
+                    "max_feature_lengths": dict(fp.max_feature_lengths),
+                }
+            else:
+                raise NotImplementedError(f"cannot serialize {type(module)}")
         return json.dumps(payload).encode(), payload["type"]
 
     @staticmethod
@@ -85,6 +104,24 @@ class JsonSerializer:
                 tables=[_dict_to_config(d) for d in payload["tables"]],
                 need_indices=payload["need_indices"],
                 device=device,
+            )
+        if payload["type"] == "FeatureProcessedEmbeddingBagCollection":
+            from torchrec_amd.modules.feature_processor import (
+                FeatureProcessedEmbeddingBagCollection,
+                PositionWeightedModuleCollection,
+            )
+
+            ebc = EmbeddingBagCollection(
+                tables=[_dict_to_bag_config(d) for d in payload["tables"]],
+                is_weighted=True,
+                device=device,
+            )
+            return FeatureProcessedEmbeddingBagCollection(
+                ebc,
+                PositionWeightedModuleCollection(
+                    {k: int(v) for k, v in payload["max_feature_lengths"].items()},
+                    device=device,
+                ),
             )
         raise NotImplementedError(payload["type"])
 
