@@ -125,6 +125,7 @@ class Perceptron(nn.Module):
     def forward(self, input: torch.Tensor) -> torch.Tensor:
         if (
             _USE_FUSED_MLP
+            and not isinstance(input, torch.fx.Proxy)  # fx-trace: eager path
             and self._activation_fn is torch.relu
             and input.is_cuda
             and input.dim() == 2
