@@ -137,3 +137,108 @@ def test_distributed_norm_clipping():
     from tests.dist_utils import run_multi_process
 
     run_multi_process(_run_dist_norm_clip, 2, "gloo")
+
+
+def test_lars_sgd_trust_scaling():
+    from torchrec_amd.optim.optimizers import LarsSGD
+
+    torch.manual_seed(0)
+    w = torch.nn.Parameter(torch.randn(8, 4))
+    opt = LarsSGD([w], lr=0.1, momentum=0.0, trust_coefficient=0.001)
+    w0 = w.detach().clone()
+    g = torch.randn_like(w)
+    w.grad = g.clone()
+    opt.step()
+    trust = 0.001 * w0.norm() / (g.norm() + 1e-8)
+    torch.testing.assert_close(w.detach(), w0 - 0.1 * trust * g, atol=1e-6, rtol=1e-5)
+
+
+def test_lamb_matches_adam_without_trust():
+    """With trust ratio == 1 (norm-matched update), LAMB == Adam."""
+    from torchrec_amd.optim.optimizers import LAMB
+
+    torch.manual_seed(0)
+    w = torch.nn.Parameter(torch.randn(6, 3))
+    wl = torch.nn.Parameter(w.detach().clone())
+    adam = torch.optim.Adam([w], lr=1e-3, eps=1e-6)
+    lamb = LAMB([wl], lr=1e-3, eps=1e-6)
+    g = torch.randn_like(w)
+    w.grad = g.clone()
+    wl.grad = g.clone()
+    adam.step()
+    lamb.step()
+    # directions agree; magnitudes differ by the trust ratio only
+    da = (w.detach() - wl.detach()).abs().max()
+    assert da < 0.01  # same scale of update
+
+
+def test_partial_rowwise_adam_rowwise_m2():
+    from torchrec_amd.optim.optimizers import PartialRowWiseAdam
+
+    torch.manual_seed(0)
+    w = torch.nn.Parameter(torch.randn(5, 4))
+    opt = PartialRowWiseAdam([w], lr=0.01)
+    w.grad = torch.randn_like(w)
+    opt.step()
+    st = opt.state[w]
+    assert st["exp_avg"].shape == (5, 4)
+    assert st["exp_avg_sq"].shape == (5,)  # ONE scalar per row
+
+
+def test_partial_rowwise_lamb_state_shapes():
+    from torchrec_amd.optim.optimizers import PartialRowWiseLAMB
+
+    w = torch.nn.Parameter(torch.randn(5, 4))
+    opt = PartialRowWiseLAMB([w], lr=0.01)
+    w.grad = torch.randn_like(w)
+    opt.step()
+    assert opt.state[w]["exp_avg_sq"].shape == (5,)
+
+
+def test_semisync_optimizer_converges_locally():
+    from torchrec_amd.optim.keyed import KeyedOptimizerWrapper
+    from torchrec_amd.optim.optimizers import SemisyncOptimizer
+
+    torch.manual_seed(0)
+    w = torch.nn.Parameter(torch.ones(4) * 5.0)
+    inner = KeyedOptimizerWrapper({"w": w}, lambda p: torch.optim.SGD(p, lr=0.1))
+    opt = SemisyncOptimizer(inner, num_local_steps=4, outer_lr=1.0, outer_momentum=0.0)
+    for _ in range(16):
+        loss = (w ** 2).sum()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    # outer steps with lr=1, momentum=0 keep the local trajectory; w -> 0
+    assert w.detach().abs().max() < 5.0 * (0.8 ** 16) + 1e-3
+
+
+def _run_semisync_dist(rank, world_size):
+    import torch.distributed as dist
+
+    from torchrec_amd.optim.keyed import KeyedOptimizerWrapper
+    from torchrec_amd.optim.optimizers import SemisyncOptimizer
+
+    torch.manual_seed(rank)
+    w = torch.nn.Parameter(torch.full((4,), float(rank + 1)))
+    inner = KeyedOptimizerWrapper({"w": w}, lambda p: torch.optim.SGD(p, lr=0.0))
+    opt = SemisyncOptimizer(
+        inner, num_local_steps=2, outer_lr=1.0, outer_momentum=0.0,
+        pg=dist.group.WORLD,
+    )
+    # no local movement (lr 0); shift params by hand so the pseudo-gradient
+    # differs per rank, then check the global step averaged it
+    for step in range(2):
+        w.grad = torch.zeros_like(w)
+        with torch.no_grad():
+            w -= (rank + 1)  # rank r moves by -(r+1)
+        opt.step()
+    # pseudo-grad at sync = (anchor - param) = (r+1)*2 averaged = 3.0
+    # anchor=r+1 -> anchor - 3.0; replicas agree on the SHIFT not the value
+    expected = (rank + 1) - 2 * (1 + 2) / 2.0
+    torch.testing.assert_close(w.detach(), torch.full((4,), expected))
+
+
+def test_semisync_optimizer_distributed():
+    from tests.dist_utils import run_multi_process
+
+    run_multi_process(_run_semisync_dist, 2, "gloo")
