@@ -55,9 +55,18 @@ class MCHManagedCollisionModule(ManagedCollisionModule):
         eviction_interval: int = 1,
         input_hash_size: int = 2**63 - 1,
         total_num_buckets: Optional[int] = None,
+        eviction_policy: str = "lfu",
+        decay_exponent: float = 1.0,
     ) -> None:
+        """eviction_policy: "lfu" (score = counts), "lru"
+        (score = -(age+1)^decay) or "distance_lfu" (counts / (age+1)^decay) —
+        reference mc_modules.py:647-875 policy zoo."""
         super().__init__()
         device = device or torch.device("cpu")
+        assert eviction_policy in ("lfu", "lru", "distance_lfu"), eviction_policy
+        self._eviction_policy = eviction_policy
+        self._decay_exponent = decay_exponent
+        self._iter = 0
         self._zch_size = zch_size
         self._residual = max(1, zch_size // 16)  # residual hash zone size
         self._output_size = zch_size
@@ -74,7 +83,10 @@ class MCHManagedCollisionModule(ManagedCollisionModule):
             "_sorted_slots", torch.arange(self._slot_zone, device=device)
         )
         self.register_buffer("_counts", torch.zeros(self._slot_zone, dtype=torch.int64, device=device))
-        self._pending_ids: List[torch.Tensor] = []
+        self.register_buffer(
+            "_last_seen", torch.zeros(self._slot_zone, dtype=torch.int64, device=device)
+        )
+        self._pending_ids: List[Tuple[torch.Tensor, int]] = []
         self._evicted_slots: Optional[torch.Tensor] = None
 
     def output_size(self) -> int:
@@ -91,7 +103,10 @@ class MCHManagedCollisionModule(ManagedCollisionModule):
             self._counts.scatter_add_(
                 0, pos_c[hit], torch.ones_like(pos_c[hit])
             )
-            self._pending_ids.append(values[~hit])
+            self._last_seen.scatter_(
+                0, pos_c[hit], torch.full_like(pos_c[hit], self._iter)
+            )
+            self._pending_ids.append((values[~hit], self._iter))
         return out
 
     def remap(self, features: Dict[str, JaggedTensor]) -> Dict[str, JaggedTensor]:
@@ -104,43 +119,75 @@ class MCHManagedCollisionModule(ManagedCollisionModule):
             )
         if self.training:
             self._batches += 1
+            self._iter += 1
             if self._batches % self._eviction_interval == 0:
                 self.profile()
         return out
 
+    def _policy_scores(self, counts: torch.Tensor, last_seen: torch.Tensor) -> torch.Tensor:
+        """Higher score = keep. Reference scoring (mc_modules.py:647-1030):
+        LFU = counts; LRU = -(age+1)^d; DistanceLFU = counts/(age+1)^d."""
+        if self._eviction_policy == "lfu":
+            return counts.double()
+        age = (self._iter - last_seen).clamp(min=0).double() + 1.0
+        if self._eviction_policy == "lru":
+            return -age.pow(self._decay_exponent)
+        return counts.double() / age.pow(self._decay_exponent)
+
     @torch.no_grad()
     def profile(self) -> None:
-        """Promote frequent unseen ids into ZCH slots, evicting cold ones."""
+        """Merge {current entries, unseen candidates} by policy score and keep
+        the top slot_zone; the rest are evicted/rejected."""
         if not self._pending_ids:
             return
-        cand = torch.cat(self._pending_ids)
+        cand = torch.cat([v for v, _ in self._pending_ids])
+        # candidate last-seen = last batch the id appeared in
+        stamp = torch.cat(
+            [torch.full_like(v, it) for v, it in self._pending_ids]
+        )
         self._pending_ids = []
         if cand.numel() == 0:
             return
-        uniq, cnt = torch.unique(cand, return_counts=True)
-        # ids already owned are excluded (they were counted as hits)
-        k = min(uniq.numel(), self._slot_zone)
-        top_cnt, top_idx = torch.topk(cnt, k)
-        new_ids = uniq[top_idx]
-        # candidate slots: lowest-count current entries
-        cold_cnt, cold_pos = torch.sort(self._counts)
-        promote = top_cnt > cold_cnt[:k]
-        n = int(promote.sum())
+        uniq, inv, cnt = torch.unique(cand, return_inverse=True, return_counts=True)
+        last = torch.full_like(uniq, 0)
+        last.scatter_reduce_(0, inv, stamp, reduce="amax", include_self=False)
+        occupied = self._sorted_ids != torch.iinfo(torch.int64).max
+        scores_cur = torch.where(
+            occupied,
+            self._policy_scores(self._counts, self._last_seen),
+            torch.full_like(self._counts, -float("inf"), dtype=torch.float64),
+        )
+        scores_new = self._policy_scores(cnt, last)
+        n_cur = scores_cur.numel()
+        merged = torch.cat([scores_cur, scores_new])
+        k = self._slot_zone
+        top = torch.topk(merged, min(k, merged.numel())).indices
+        keep_cur = torch.zeros(n_cur, dtype=torch.bool, device=merged.device)
+        keep_cur[top[top < n_cur]] = True
+        promote_new = top[top >= n_cur] - n_cur
+        evict_pos = (~keep_cur).nonzero().squeeze(1)
+        n = min(evict_pos.numel(), promote_new.numel())
         if n == 0:
             return
-        evict_pos = cold_pos[:k][promote]
+        evict_pos = evict_pos[:n]
+        promote_new = promote_new[:n]
         evicted_slots = self._sorted_slots[evict_pos].clone()
-        self._sorted_ids[evict_pos] = new_ids[promote]
-        self._counts[evict_pos] = top_cnt[promote]
+        had_id = occupied[evict_pos]
+        self._sorted_ids[evict_pos] = uniq[promote_new]
+        self._counts[evict_pos] = cnt[promote_new]
+        self._last_seen[evict_pos] = last[promote_new]
         order = torch.argsort(self._sorted_ids)
         self._sorted_ids = self._sorted_ids[order]
         self._sorted_slots = self._sorted_slots[order]
         self._counts = self._counts[order]
-        self._evicted_slots = (
-            evicted_slots
-            if self._evicted_slots is None
-            else torch.cat([self._evicted_slots, evicted_slots])
-        )
+        self._last_seen = self._last_seen[order]
+        evicted_slots = evicted_slots[had_id]  # empty slots evict nothing
+        if evicted_slots.numel():
+            self._evicted_slots = (
+                evicted_slots
+                if self._evicted_slots is None
+                else torch.cat([self._evicted_slots, evicted_slots])
+            )
 
     def evict(self) -> Optional[torch.Tensor]:
         out = self._evicted_slots
