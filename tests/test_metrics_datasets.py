@@ -198,3 +198,106 @@ def test_criteo_shard_math():
             seen.extend((f, r) for r in range(lo, hi))
     # ranks cover disjoint rows
     assert len(seen) == len(set(seen)) == (225 // 4) * 4
+
+
+class TestExtendedMetrics:
+    def _mk(self, cls, **kw):
+        from torchrec_amd.metrics.rec_metric import RecTaskInfo
+
+        return cls(tasks=[RecTaskInfo(name="t")], **kw)
+
+    def test_mae_ctr_weighted_avg_logloss(self):
+        from torchrec_amd.metrics.extended_metrics import (
+            CTRMetric, LogLossMetric, MAEMetric, WeightedAvgMetric,
+        )
+
+        torch.manual_seed(0)
+        p = torch.rand(100)
+        y = (torch.rand(100) > 0.7).double()
+        for cls, expect in [
+            (MAEMetric, (p.double() - y).abs().mean()),
+            (CTRMetric, y.mean()),
+            (WeightedAvgMetric, p.double().mean()),
+        ]:
+            m = self._mk(cls)
+            m.update(predictions={"t": p}, labels={"t": y})
+            got = m.compute()[f"{cls.NAME}-t|lifetime_{cls.NAME}"]
+            torch.testing.assert_close(got.squeeze(), expect, atol=1e-6, rtol=1e-6)
+        m = self._mk(LogLossMetric)
+        m.update(predictions={"t": p}, labels={"t": y})
+        got = m.compute()["logloss-t|lifetime_logloss"]
+        pc = p.double().clamp(1e-7, 1 - 1e-7)
+        ce = -(y * pc.log() + (1 - y) * (1 - pc).log())
+        torch.testing.assert_close(got.squeeze(), ce.mean(), atol=1e-6, rtol=1e-6)
+
+    def test_ndcg_perfect_and_inverted(self):
+        from torchrec_amd.metrics.extended_metrics import NDCGMetric
+
+        m = self._mk(NDCGMetric)
+        # one session with labels [1, 0]: perfect ranking -> ndcg 1
+        m.update(
+            predictions={"t": torch.tensor([0.9, 0.1])},
+            labels={"t": torch.tensor([1.0, 0.0])},
+            session_ids={"t": torch.tensor([7, 7])},
+        )
+        got = m.compute()["ndcg-t|lifetime_ndcg"]
+        torch.testing.assert_close(got.squeeze(), torch.tensor(1.0).double())
+
+        m2 = self._mk(NDCGMetric)
+        # inverted ranking: dcg = 1/log2(3), idcg = 1 -> ndcg = 0.6309
+        m2.update(
+            predictions={"t": torch.tensor([0.1, 0.9])},
+            labels={"t": torch.tensor([1.0, 0.0])},
+            session_ids={"t": torch.tensor([7, 7])},
+        )
+        got2 = m2.compute()["ndcg-t|lifetime_ndcg"]
+        import math
+
+        torch.testing.assert_close(
+            got2.squeeze(), torch.tensor(1.0 / math.log2(3)).double(),
+            atol=1e-6, rtol=1e-6,
+        )
+
+    def test_recall_session_topk(self):
+        from torchrec_amd.metrics.extended_metrics import RecallSessionMetric
+
+        m = self._mk(RecallSessionMetric, top_k=2)
+        # session A: positives ranked 1st and 3rd -> 1 of 2 in top-2
+        # session B: positive ranked 1st -> 1 of 1
+        m.update(
+            predictions={"t": torch.tensor([0.9, 0.8, 0.7, 0.95, 0.1])},
+            labels={"t": torch.tensor([1.0, 0.0, 1.0, 1.0, 0.0])},
+            session_ids={"t": torch.tensor([1, 1, 1, 2, 2])},
+        )
+        got = m.compute()["recall_session-t|lifetime_recall_session"]
+        torch.testing.assert_close(got.squeeze(), torch.tensor(2.0 / 3.0).double())
+
+    def test_fused_tasks_matches_unfused(self):
+        from torchrec_amd.metrics.rec_metric import (
+            NEMetric, RecComputeMode, RecTaskInfo,
+        )
+
+        torch.manual_seed(0)
+        tasks = [RecTaskInfo(name="a"), RecTaskInfo(name="b")]
+        fused = NEMetric(tasks, compute_mode=RecComputeMode.FUSED_TASKS_COMPUTATION)
+        unfused = NEMetric(tasks)
+        for seed in range(3):
+            g = torch.Generator().manual_seed(seed)
+            preds = {t.name: torch.rand(50, generator=g) for t in tasks}
+            labels = {t.name: (torch.rand(50, generator=g) > 0.5).double() for t in tasks}
+            fused.update(predictions=preds, labels=labels)
+            unfused.update(predictions=preds, labels=labels)
+        rf, ru = fused.compute(), unfused.compute()
+        for k in ru:
+            torch.testing.assert_close(rf[k].squeeze(), ru[k].squeeze(),
+                                       atol=1e-9, rtol=1e-9)
+
+    def test_tower_qps(self):
+        from torchrec_amd.metrics.extended_metrics import TowerQPSMetric
+
+        m = TowerQPSMetric(["over", "user"], warmup_steps=0)
+        for _ in range(5):
+            m.update({"over": 100, "user": 50})
+        out = m.compute()
+        assert out["tower_qps-over"] > 0
+        assert abs(out["tower_qps-over"] / out["tower_qps-user"] - 2.0) < 1e-6

@@ -9,6 +9,7 @@ RecMetricComputation :202 — torchmetrics-style local state + windowed buffers
 from __future__ import annotations
 
 import abc
+from enum import Enum, unique
 import math
 from collections import deque
 from dataclasses import dataclass, field
@@ -48,17 +49,33 @@ class WindowBuffer:
         return out
 
 
+def _bsum(t: torch.Tensor) -> torch.Tensor:
+    """Sum over the sample dim: [N] -> [1]; [T, N] -> [T] (fused tasks)."""
+    if t.dim() > 1:
+        return t.sum(dim=-1)
+    return t.sum().reshape(1)
+
+
 class RecMetricComputation(abc.ABC, nn.Module):
-    """Accumulates local state; all-reduces at compute()."""
+    """Accumulates local state; all-reduces at compute().
+
+    ``n_tasks`` > 1 vectorizes the state over a leading task dim — the
+    FUSED_TASKS_COMPUTATION mode (reference metrics_config.py:88
+    RecComputeMode): one computation updates all tasks from stacked
+    [T, N] inputs."""
 
     STATE_NAMES: List[str] = []
 
-    def __init__(self, window_size: int = 100, process_group=None) -> None:
+    def __init__(self, window_size: int = 100, process_group=None,
+                 n_tasks: int = 1) -> None:
         super().__init__()
         self._pg = process_group
+        self._n_tasks = n_tasks
         self._window = WindowBuffer(window_size)
         for name in self.STATE_NAMES:
-            self.register_buffer(name, torch.zeros(1, dtype=torch.float64), persistent=False)
+            self.register_buffer(
+                name, torch.zeros(n_tasks, dtype=torch.float64), persistent=False
+            )
 
     def _states(self) -> List[torch.Tensor]:
         return [getattr(self, n) for n in self.STATE_NAMES]
@@ -116,10 +133,10 @@ class NEComputation(RecMetricComputation):
         w = _weights_or_ones(y, weights)
         ce = -(y * p.log() + (1 - y) * (1 - p).log())
         batch = [
-            (ce * w).sum().reshape(1),
-            w.sum().reshape(1),
-            (y * w).sum().reshape(1),
-            ((1 - y) * w).sum().reshape(1),
+            _bsum(ce * w),
+            _bsum(w),
+            _bsum(y * w),
+            _bsum((1 - y) * w),
         ]
         for name, b in zip(self.STATE_NAMES, batch):
             getattr(self, name).add_(b)
@@ -203,8 +220,8 @@ class CalibrationComputation(RecMetricComputation):
     def update(self, predictions, labels, weights=None) -> None:
         w = _weights_or_ones(labels.double(), weights)
         batch = [
-            (predictions.double() * w).sum().reshape(1),
-            (labels.double() * w).sum().reshape(1),
+            _bsum(predictions.double() * w),
+            _bsum(labels.double() * w),
         ]
         for name, b in zip(self.STATE_NAMES, batch):
             getattr(self, name).add_(b)
@@ -221,8 +238,8 @@ class MSEComputation(RecMetricComputation):
     def update(self, predictions, labels, weights=None) -> None:
         w = _weights_or_ones(labels.double(), weights)
         batch = [
-            (w * (predictions.double() - labels.double()) ** 2).sum().reshape(1),
-            w.sum().reshape(1),
+            _bsum(w * (predictions.double() - labels.double()) ** 2),
+            _bsum(w),
         ]
         for name, b in zip(self.STATE_NAMES, batch):
             getattr(self, name).add_(b)
@@ -233,24 +250,57 @@ class MSEComputation(RecMetricComputation):
         return err / n.clamp(min=1e-12)
 
 
+@unique
+class RecComputeMode(Enum):
+    """Reference metrics_config.py:88."""
+
+    UNFUSED_TASKS_COMPUTATION = "unfused"
+    FUSED_TASKS_COMPUTATION = "fused"
+
+
 class RecMetric(nn.Module):
-    """Multi-task wrapper over a computation class (reference rec_metric.py:393)."""
+    """Multi-task wrapper over a computation class (reference rec_metric.py:393).
+
+    FUSED_TASKS_COMPUTATION runs ONE vectorized computation over stacked
+    [n_tasks, N] inputs instead of one computation per task."""
 
     COMPUTATION: type = NEComputation
     NAME = "metric"
+    FUSABLE = True
 
     def __init__(
         self,
         tasks: List[RecTaskInfo],
         window_size: int = 100,
         process_group=None,
+        compute_mode: RecComputeMode = RecComputeMode.UNFUSED_TASKS_COMPUTATION,
         **kwargs,
     ) -> None:
         super().__init__()
         self._tasks = tasks
-        self._computations = nn.ModuleList(
-            [self.COMPUTATION(window_size=window_size, process_group=process_group) for _ in tasks]
-        )
+        self._compute_mode = compute_mode
+        if compute_mode == RecComputeMode.FUSED_TASKS_COMPUTATION:
+            if not self.FUSABLE:
+                raise ValueError(f"{self.NAME} does not support fused-task compute")
+            self._computations = nn.ModuleList(
+                [
+                    self.COMPUTATION(
+                        window_size=window_size,
+                        process_group=process_group,
+                        n_tasks=len(tasks),
+                        **kwargs,
+                    )
+                ]
+            )
+        else:
+            self._computations = nn.ModuleList(
+                [
+                    self.COMPUTATION(
+                        window_size=window_size, process_group=process_group, **kwargs
+                    )
+                    for _ in tasks
+                ]
+            )
 
     def update(
         self,
@@ -259,6 +309,21 @@ class RecMetric(nn.Module):
         labels: Dict[str, torch.Tensor],
         weights: Optional[Dict[str, torch.Tensor]] = None,
     ) -> None:
+        if self._compute_mode == RecComputeMode.FUSED_TASKS_COMPUTATION:
+            p = torch.stack([predictions[t.name] for t in self._tasks])
+            y = torch.stack([labels[t.name] for t in self._tasks])
+            w = (
+                torch.stack(
+                    [
+                        weights.get(t.name, torch.ones_like(labels[t.name]))
+                        for t in self._tasks
+                    ]
+                )
+                if weights
+                else None
+            )
+            self._computations[0].update(p, y, w)
+            return
         for task, comp in zip(self._tasks, self._computations):
             comp.update(
                 predictions[task.name],
@@ -268,6 +333,12 @@ class RecMetric(nn.Module):
 
     def compute(self) -> Dict[str, torch.Tensor]:
         out = {}
+        if self._compute_mode == RecComputeMode.FUSED_TASKS_COMPUTATION:
+            res = self._computations[0].compute()
+            for i, task in enumerate(self._tasks):
+                out[f"{self.NAME}-{task.name}|lifetime_{self.NAME}"] = res["lifetime"][i]
+                out[f"{self.NAME}-{task.name}|window_{self.NAME}"] = res["window"][i]
+            return out
         for task, comp in zip(self._tasks, self._computations):
             res = comp.compute()
             out[f"{self.NAME}-{task.name}|lifetime_{self.NAME}"] = res["lifetime"]
@@ -283,6 +354,7 @@ class NEMetric(RecMetric):
 class AUCMetric(RecMetric):
     COMPUTATION = AUCComputation
     NAME = "auc"
+    FUSABLE = False  # keeps raw score buffers, not sum-states
 
 
 class CalibrationMetric(RecMetric):
@@ -304,7 +376,7 @@ class AccuracyComputation(RecMetricComputation):
         y = labels.double()
         w = _weights_or_ones(y, weights)
         hit = ((predictions.double() >= 0.5) == (y >= 0.5)).double()
-        batch = [(hit * w).sum().reshape(1), w.sum().reshape(1)]
+        batch = [_bsum(hit * w), _bsum(w)]
         for name, b in zip(self.STATE_NAMES, batch):
             getattr(self, name).add_(b)
         self._record_window(batch)
@@ -324,8 +396,8 @@ class PrecisionComputation(RecMetricComputation):
         w = _weights_or_ones(y, weights)
         pred_pos = (predictions.double() >= 0.5).double()
         batch = [
-            (pred_pos * y * w).sum().reshape(1),
-            (pred_pos * (1 - y) * w).sum().reshape(1),
+            _bsum(pred_pos * y * w),
+            _bsum(pred_pos * (1 - y) * w),
         ]
         for name, b in zip(self.STATE_NAMES, batch):
             getattr(self, name).add_(b)
@@ -346,8 +418,8 @@ class RecallComputation(RecMetricComputation):
         w = _weights_or_ones(y, weights)
         pred_pos = (predictions.double() >= 0.5).double()
         batch = [
-            (pred_pos * y * w).sum().reshape(1),
-            ((1 - pred_pos) * y * w).sum().reshape(1),
+            _bsum(pred_pos * y * w),
+            _bsum((1 - pred_pos) * y * w),
         ]
         for name, b in zip(self.STATE_NAMES, batch):
             getattr(self, name).add_(b)
