@@ -225,3 +225,38 @@ def test_vbe_sharded_cuda_single():
             t = shards[0].tensor if shards else None
         if t is not None:
             torch.testing.assert_close(t.cpu(), gw, atol=1e-4, rtol=1e-4)
+
+
+def _run_rs_v_per_feature(rank, world_size):
+    """Golden check for reduce_scatter_v_per_feature_pooled: rank r must
+    receive the rank-sum of its own bags, feature-major."""
+    from torchrec_amd.distributed.comm_ops import (
+        reduce_scatter_v_per_feature_pooled,
+    )
+
+    dims = [2, 4]
+    bs = [[2, 1], [1, 3]]  # bs[f][r]
+    torch.manual_seed(7)
+    # every rank builds the same "global" tensor plus a rank-dependent offset
+    packs = []
+    for r in range(world_size):
+        base = torch.arange(
+            sum(bs[f][rr] * dims[f] for f in range(2) for rr in range(world_size)),
+            dtype=torch.float32,
+        )
+        packs.append(base + 100.0 * r)
+    mine = packs[rank].clone().requires_grad_(True)
+    out = reduce_scatter_v_per_feature_pooled(mine, bs, dims, dist.group.WORLD).wait()
+    # expected: sum over ranks of the (f, rank) blocks belonging to me
+    total = sum(packs)  # elementwise rank-sum of the feature-major pack
+    sizes_fmaj = [bs[f][r] * dims[f] for f in range(2) for r in range(world_size)]
+    blocks = list(total.split(sizes_fmaj))
+    expected = torch.cat([blocks[f * world_size + rank] for f in range(2)])
+    torch.testing.assert_close(out.reshape(-1), expected)
+    # backward mirrors with an all-gather-v: grads flow to every source block
+    out.sum().backward()
+    assert mine.grad is not None and mine.grad.shape == mine.shape
+
+
+def test_reduce_scatter_v_per_feature_pooled():
+    run_multi_process(_run_rs_v_per_feature, 2, "gloo")

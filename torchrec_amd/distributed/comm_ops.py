@@ -502,6 +502,38 @@ class _ReduceScatterVWait(torch.autograd.Function):
         return None, None, myreq.dummy
 
 
+def reduce_scatter_v_per_feature_pooled(
+    pooled: torch.Tensor,
+    batch_size_per_rank_per_feature: List[List[int]],
+    embedding_dims: List[int],
+    pg: dist.ProcessGroup,
+) -> Awaitable[torch.Tensor]:
+    """Per-feature uneven reduce-scatter for the VBE RW output dist
+    (reference comm_ops.py:1330 reduce_scatter_v_per_feature_pooled).
+
+    ``pooled`` is the 1-D feature-major VBE pack
+    [f0 | r0 bags, r1 bags, ... | f1 | ...] where feature f's rank-r block is
+    batch_size_per_rank_per_feature[f][r] * embedding_dims[f] elements. The
+    blocks are regrouped rank-major (split+cat keeps autograd transparent)
+    and reduced with ONE uneven reduce-scatter; rank r receives the summed
+    feature-major pack of its own bags.
+    """
+    W = dist.get_world_size(pg)
+    F = len(embedding_dims)
+    sizes_fmaj = [
+        batch_size_per_rank_per_feature[f][r] * embedding_dims[f]
+        for f in range(F)
+        for r in range(W)
+    ]
+    blocks = list(pooled.reshape(-1).split(sizes_fmaj))
+    rank_major = torch.cat([blocks[f * W + r] for r in range(W) for f in range(F)])
+    splits = [
+        sum(batch_size_per_rank_per_feature[f][r] * embedding_dims[f] for f in range(F))
+        for r in range(W)
+    ]
+    return reduce_scatter_v_pooled(rank_major, splits, pg)
+
+
 def reduce_scatter_v_pooled(
     pooled: torch.Tensor, splits: List[int], pg: dist.ProcessGroup
 ) -> Awaitable[torch.Tensor]:

@@ -168,11 +168,23 @@ class AUCComputation(RecMetricComputation):
         self._labels.append(labels.detach().double().reshape(-1))
         self._weights.append(_weights_or_ones(labels.reshape(-1), weights))
         total = sum(p.numel() for p in self._preds)
+        dropped = False
         while total > self._max_elems and len(self._preds) > 1:
             total -= self._preds[0].numel()
             self._preds.pop(0)
             self._labels.pop(0)
             self._weights.pop(0)
+            dropped = True
+        if dropped and not getattr(self, "_warned_cap", False):
+            import warnings
+
+            warnings.warn(
+                f"AUC score buffer exceeded max_elems={self._max_elems}; oldest "
+                "batches dropped — AUC is now computed over a sliding window. "
+                "Raise AUCMetric(..., max_elems=N) for a longer horizon.",
+                stacklevel=2,
+            )
+            self._warned_cap = True
 
     def compute(self) -> Dict[str, torch.Tensor]:
         if not self._preds:
