@@ -198,3 +198,83 @@ class TestPsBridge:
         torch.testing.assert_close(
             kv.split_embedding_weights()[0][slot], w11, atol=1e-6, rtol=1e-6
         )
+
+
+class TestVirtualTableEvictionPolicies:
+    """Policy-driven eviction on the KV (virtual) tables (reference
+    embedding_configs.py:180-352 KV-ZCH eviction policies)."""
+
+    def _kv(self, policy):
+        from torchrec_amd.ops.kv_embedding import KeyValueEmbeddingBags
+
+        return KeyValueEmbeddingBags(
+            [("t0", 1 << 40, 8)], capacity=16, eviction_policy=policy,
+        )
+
+    @staticmethod
+    def _step(kv, ids):
+        idx = torch.tensor(ids, dtype=torch.int64)
+        offsets = torch.arange(len(ids) + 1, dtype=torch.int64)
+        out = kv(idx, offsets)
+        out.sum().backward()
+
+    def test_count_based_evicts_cold_ids(self):
+        from torchrec_amd.modules.embedding_configs import CountBasedEvictionPolicy
+
+        kv = self._kv(CountBasedEvictionPolicy(eviction_threshold=3,
+                                               eviction_interval_batches=0))
+        kv.train()
+        for _ in range(5):
+            self._step(kv, [1, 2, 3])  # hot
+        self._step(kv, [100, 200])  # cold (count 1)
+        n = kv.run_policy_eviction()
+        assert n == 2
+        ids = set(kv.save_ids()[0].tolist())
+        assert {1, 2, 3} <= ids and 100 not in ids and 200 not in ids
+
+    def test_timestamp_based_evicts_stale(self):
+        from torchrec_amd.modules.embedding_configs import TimestampBasedEvictionPolicy
+
+        kv = self._kv(TimestampBasedEvictionPolicy(eviction_ttl_mins=3,
+                                                   eviction_interval_batches=0))
+        kv.train()
+        self._step(kv, [7])
+        for _ in range(6):
+            self._step(kv, [1, 2])
+        n = kv.run_policy_eviction()
+        assert n == 1
+        assert 7 not in set(kv.save_ids()[0].tolist())
+
+    def test_no_eviction_policy(self):
+        from torchrec_amd.modules.embedding_configs import NoEvictionPolicy
+
+        kv = self._kv(NoEvictionPolicy())
+        kv.train()
+        self._step(kv, [5, 6])
+        assert kv.run_policy_eviction() == 0
+
+    def test_l2norm_evicts_unlearned_rows(self):
+        from torchrec_amd.modules.embedding_configs import (
+            FeatureL2NormBasedEvictionPolicy,
+        )
+
+        kv = self._kv(FeatureL2NormBasedEvictionPolicy(eviction_threshold=1e-6,
+                                                       eviction_interval_batches=0))
+        kv.train()
+        self._step(kv, [1, 2])
+        # zero one row by hand: it must be the eviction victim
+        with torch.no_grad():
+            kv.split_embedding_weights()[0][0].zero_()
+        n = kv.run_policy_eviction()
+        assert n == 1
+
+    def test_interval_triggers_inside_forward(self):
+        from torchrec_amd.modules.embedding_configs import CountBasedEvictionPolicy
+
+        kv = self._kv(CountBasedEvictionPolicy(eviction_threshold=100,
+                                               eviction_interval_batches=2))
+        kv.train()
+        self._step(kv, [1, 2])
+        assert len(set(kv.save_ids()[0].tolist()) - {-1}) == 2
+        self._step(kv, [3])  # batch 2: sweep fires, everything is below 100
+        assert len(set(kv.save_ids()[0].tolist()) - {-1}) <= 1

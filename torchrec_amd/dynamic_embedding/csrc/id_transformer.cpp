@@ -55,7 +55,10 @@ class IdTransformer {
         continue;
       }
       int64_t s;
-      if (free_head_ < capacity_) {
+      if (!recycled_.empty()) {
+        s = recycled_.back();
+        recycled_.pop_back();
+      } else if (free_head_ < capacity_) {
         s = free_head_++;
       } else {
         // mixed LFU/LRU eviction: min (freq, last)
@@ -92,6 +95,45 @@ class IdTransformer {
     for (auto& s : slots_) s.freq >>= 1;
   }
 
+  // Policy-driven eviction support: expose per-slot (freq, last-access) so a
+  // VirtualTableEvictionPolicy (count / timestamp / mixed) can pick victims,
+  // and free specific slots the policy chose.
+  std::tuple<at::Tensor, at::Tensor> slot_stats() {
+    std::lock_guard<std::mutex> g(mu_);
+    auto freq = at::empty({capacity_}, at::kLong);
+    auto last = at::empty({capacity_}, at::kLong);
+    int64_t* fp = freq.data_ptr<int64_t>();
+    int64_t* lp = last.data_ptr<int64_t>();
+    for (int64_t i = 0; i < capacity_; ++i) {
+      fp[i] = slots_[i].freq;
+      lp[i] = (int64_t)slots_[i].last;
+    }
+    return {freq, last};
+  }
+
+  int64_t clock() {
+    std::lock_guard<std::mutex> g(mu_);
+    return (int64_t)clock_;
+  }
+
+  // Free the given slots (returns the raw ids that were evicted). Freed
+  // slots go to a recycle list consumed before any LFU/LRU eviction.
+  at::Tensor evict_slots(const at::Tensor& slots) {
+    TORCH_CHECK(slots.scalar_type() == at::kLong && !slots.is_cuda());
+    auto sc = slots.contiguous();
+    std::lock_guard<std::mutex> g(mu_);
+    std::vector<int64_t> ids;
+    for (int64_t i = 0; i < sc.numel(); ++i) {
+      int64_t s = sc.data_ptr<int64_t>()[i];
+      if (s < 0 || s >= capacity_ || slots_[s].id < 0) continue;
+      ids.push_back(slots_[s].id);
+      map_.erase(slots_[s].id);
+      slots_[s] = SlotStats{};
+      recycled_.push_back(s);
+    }
+    return at::tensor(ids, at::kLong);
+  }
+
   std::vector<int64_t> save_ids() {
     std::lock_guard<std::mutex> g(mu_);
     std::vector<int64_t> out(capacity_, -1);
@@ -104,6 +146,7 @@ class IdTransformer {
   uint64_t clock_;
   int64_t free_head_;
   std::vector<SlotStats> slots_;
+  std::vector<int64_t> recycled_;
   std::unordered_map<int64_t, int64_t> map_;
   std::mutex mu_;
 };
@@ -119,5 +162,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("size", &IdTransformer::size)
       .def("capacity", &IdTransformer::capacity)
       .def("decay", &IdTransformer::decay)
+      .def("slot_stats", &IdTransformer::slot_stats)
+      .def("clock", &IdTransformer::clock)
+      .def("evict_slots", &IdTransformer::evict_slots)
       .def("save_ids", &IdTransformer::save_ids);
 }

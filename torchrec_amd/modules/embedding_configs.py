@@ -70,6 +70,72 @@ def pooling_type_to_str(p: PoolingType) -> str:
     return p.value.lower()
 
 
+
+# ---------------------------------------------------------------------------
+# virtual-table (KV / collision-free) eviction-policy configs
+# (reference modules/embedding_configs.py:180-352 — RFC-0002 KV-ZCH)
+# ---------------------------------------------------------------------------
+
+
+@dataclass
+class VirtualTableEvictionPolicy:
+    """Base policy config for virtual (unbounded-id) tables. The metaheader
+    carries per-row bookkeeping the policy needs (count / timestamp /
+    feature-score) alongside the embedding payload."""
+
+    meta_header_lens: List[int] = field(default_factory=list)
+    eviction_interval_batches: int = 100
+
+    def get_meta_header_len(self) -> int:
+        return sum(self.meta_header_lens)
+
+
+@dataclass
+class CountBasedEvictionPolicy(VirtualTableEvictionPolicy):
+    """Evict rows whose access count falls below the threshold."""
+
+    eviction_threshold: int = 1
+    decay_rate: float = 0.99
+
+    def __post_init__(self) -> None:
+        self.meta_header_lens = [4]  # count: uint32
+
+
+@dataclass
+class TimestampBasedEvictionPolicy(VirtualTableEvictionPolicy):
+    """Evict rows not touched within the TTL."""
+
+    eviction_ttl_mins: int = 24 * 60
+
+    def __post_init__(self) -> None:
+        self.meta_header_lens = [4]  # last-access stamp
+
+
+@dataclass
+class CountTimestampMixedEvictionPolicy(VirtualTableEvictionPolicy):
+    eviction_threshold: int = 1
+    eviction_ttl_mins: int = 24 * 60
+
+    def __post_init__(self) -> None:
+        self.meta_header_lens = [4, 4]
+
+
+@dataclass
+class FeatureL2NormBasedEvictionPolicy(VirtualTableEvictionPolicy):
+    """Evict rows whose embedding L2 norm is below the threshold (rows that
+    never learned anything meaningful)."""
+
+    eviction_threshold: float = 1e-3
+
+    def __post_init__(self) -> None:
+        self.meta_header_lens = []
+
+
+@dataclass
+class NoEvictionPolicy(VirtualTableEvictionPolicy):
+    pass
+
+
 @dataclass
 class BaseEmbeddingConfig:
     num_embeddings: int
@@ -80,6 +146,10 @@ class BaseEmbeddingConfig:
     weight_init_max: Optional[float] = None
     weight_init_min: Optional[float] = None
     need_pos: bool = False
+    # virtual (KV / collision-free) tables: num_embeddings is the VIRTUAL id
+    # space; physical capacity and eviction ride the policy config
+    use_virtual_table: bool = False
+    virtual_table_eviction_policy: Optional[VirtualTableEvictionPolicy] = None
 
     def get_weight_init_max(self) -> float:
         if self.weight_init_max is None:

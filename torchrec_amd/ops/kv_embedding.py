@@ -49,7 +49,13 @@ class KeyValueEmbeddingBags(nn.Module):
         init_min: float = -0.01,
         init_max: float = 0.01,
         location: EmbeddingLocation = EmbeddingLocation.MANAGED,
+        eviction_policy=None,
     ) -> None:
+        """``eviction_policy``: a VirtualTableEvictionPolicy config
+        (embedding_configs.py — count / timestamp / mixed / L2-norm / none).
+        ``run_policy_eviction()`` applies it; capacity-pressure eviction via
+        the id transformer's mixed LFU/LRU continues regardless (except
+        NoEvictionPolicy, which disables the policy sweep only)."""
         super().__init__()
         self._virtual_rows = [int(s[1]) for s in embedding_specs]
         self._capacity = capacity
@@ -74,6 +80,8 @@ class KeyValueEmbeddingBags(nn.Module):
             IdTransformer(min(int(s[1]), capacity)) for s in embedding_specs
         ]
         self._ftm = self._tbe._feature_table_map
+        self._eviction_policy = eviction_policy
+        self._batches_seen = 0
 
     @property
     def embedding_specs(self):
@@ -126,7 +134,59 @@ class KeyValueEmbeddingBags(nn.Module):
             slots, ev_slots, _ev_ids = self._transformers[t].transform(idx_cpu[lo:hi])
             self._reinit_slots(t, ev_slots)
             out_slots[lo:hi] = slots
+        if self.training and self._eviction_policy is not None:
+            self._batches_seen += 1
+            interval = getattr(self._eviction_policy, "eviction_interval_batches", 0)
+            if interval and self._batches_seen % interval == 0:
+                self.run_policy_eviction()
         return self._tbe(out_slots.to(indices.device), offsets, per_sample_weights)
+
+    @torch.no_grad()
+    def run_policy_eviction(self) -> int:
+        """Apply the configured VirtualTableEvictionPolicy: pick victim slots
+        per table, free them in the transformer, re-init their rows. Returns
+        the number of rows evicted (reference: KV-ZCH eviction trigger,
+        embedding_configs.py:180-352 policy configs)."""
+        from torchrec_amd.modules.embedding_configs import (
+            CountBasedEvictionPolicy,
+            CountTimestampMixedEvictionPolicy,
+            FeatureL2NormBasedEvictionPolicy,
+            NoEvictionPolicy,
+            TimestampBasedEvictionPolicy,
+        )
+
+        pol = self._eviction_policy
+        if pol is None or isinstance(pol, NoEvictionPolicy):
+            return 0
+        total = 0
+        for t, tr in enumerate(self._transformers):
+            freq, last = tr.slot_stats()
+            occupied = freq > 0
+            if isinstance(pol, CountBasedEvictionPolicy):
+                victims = occupied & (freq < pol.eviction_threshold)
+            elif isinstance(pol, TimestampBasedEvictionPolicy):
+                # the transformer clock ticks once per transform() call;
+                # ttl is interpreted in those ticks
+                age = tr.clock() - last
+                victims = occupied & (age > pol.eviction_ttl_mins)
+            elif isinstance(pol, CountTimestampMixedEvictionPolicy):
+                age = tr.clock() - last
+                victims = occupied & (
+                    (freq < pol.eviction_threshold) | (age > pol.eviction_ttl_mins)
+                )
+            elif isinstance(pol, FeatureL2NormBasedEvictionPolicy):
+                w = self._tbe.split_embedding_weights()[t]
+                norms = w.float().norm(dim=1).cpu()
+                victims = occupied & (norms < pol.eviction_threshold)
+            else:
+                continue
+            slots = victims.nonzero().squeeze(1)
+            if slots.numel() == 0:
+                continue
+            tr.evict_slots(slots)
+            self._reinit_slots(t, slots)
+            total += int(slots.numel())
+        return total
 
 
 class SsdEmbeddingBags(KeyValueEmbeddingBags):
@@ -212,7 +272,59 @@ class SsdEmbeddingBags(KeyValueEmbeddingBags):
             for k in range(seg.numel()):
                 self._restore(t, int(slots[k]), int(seg[k]))
             out_slots[lo:hi] = slots
+        if self.training and self._eviction_policy is not None:
+            self._batches_seen += 1
+            interval = getattr(self._eviction_policy, "eviction_interval_batches", 0)
+            if interval and self._batches_seen % interval == 0:
+                self.run_policy_eviction()
         return self._tbe(out_slots.to(indices.device), offsets, per_sample_weights)
+
+    @torch.no_grad()
+    def run_policy_eviction(self) -> int:
+        """Apply the configured VirtualTableEvictionPolicy: pick victim slots
+        per table, free them in the transformer, re-init their rows. Returns
+        the number of rows evicted (reference: KV-ZCH eviction trigger,
+        embedding_configs.py:180-352 policy configs)."""
+        from torchrec_amd.modules.embedding_configs import (
+            CountBasedEvictionPolicy,
+            CountTimestampMixedEvictionPolicy,
+            FeatureL2NormBasedEvictionPolicy,
+            NoEvictionPolicy,
+            TimestampBasedEvictionPolicy,
+        )
+
+        pol = self._eviction_policy
+        if pol is None or isinstance(pol, NoEvictionPolicy):
+            return 0
+        total = 0
+        for t, tr in enumerate(self._transformers):
+            freq, last = tr.slot_stats()
+            occupied = freq > 0
+            if isinstance(pol, CountBasedEvictionPolicy):
+                victims = occupied & (freq < pol.eviction_threshold)
+            elif isinstance(pol, TimestampBasedEvictionPolicy):
+                # the transformer clock ticks once per transform() call;
+                # ttl is interpreted in those ticks
+                age = tr.clock() - last
+                victims = occupied & (age > pol.eviction_ttl_mins)
+            elif isinstance(pol, CountTimestampMixedEvictionPolicy):
+                age = tr.clock() - last
+                victims = occupied & (
+                    (freq < pol.eviction_threshold) | (age > pol.eviction_ttl_mins)
+                )
+            elif isinstance(pol, FeatureL2NormBasedEvictionPolicy):
+                w = self._tbe.split_embedding_weights()[t]
+                norms = w.float().norm(dim=1).cpu()
+                victims = occupied & (norms < pol.eviction_threshold)
+            else:
+                continue
+            slots = victims.nonzero().squeeze(1)
+            if slots.numel() == 0:
+                continue
+            tr.evict_slots(slots)
+            self._reinit_slots(t, slots)
+            total += int(slots.numel())
+        return total
 
     def close(self) -> None:
         self._ps.close()
