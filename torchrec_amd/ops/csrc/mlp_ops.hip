@@ -85,6 +85,22 @@ __global__ void __launch_bounds__(kBlockThreads) relu_bwd_colsum_partial_kernel(
   }
 }
 
+// two-level deterministic reduce of [G, N] -> [N]: level 1 collapses G into
+// S segments with a 2-D grid (full-chip parallel), level 2 folds S
+__global__ void __launch_bounds__(kBlockThreads) colsum_seg_f32_kernel(
+    const float* __restrict__ partial, int G, int seg_rows, int64_t N,
+    float* __restrict__ seg_out /* [S, N] */) {
+  int64_t c = static_cast<int64_t>(blockIdx.x) * kBlockThreads + threadIdx.x;
+  if (c >= N) return;
+  int s = blockIdx.y;
+  int g0 = s * seg_rows;
+  int g1 = min(G, g0 + seg_rows);
+  float acc = 0.f;
+  for (int grp = g0; grp < g1; ++grp)
+    acc += partial[static_cast<int64_t>(grp) * N + c];
+  seg_out[static_cast<int64_t>(s) * N + c] = acc;
+}
+
 __global__ void __launch_bounds__(kBlockThreads) colsum_final_f32_kernel(
     const float* __restrict__ partial, int G, int64_t N, float* __restrict__ out) {
   int64_t c = static_cast<int64_t>(blockIdx.x) * kBlockThreads + threadIdx.x;
@@ -132,8 +148,21 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
                          reinterpret_cast<dev_t*>(g.data_ptr<scalar_t>()),
                          partial.data_ptr<float>());
       int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);
-      hipLaunchKernelGGL(colsum_final_f32_kernel, dim3(ftiles), dim3(kBlockThreads), 0,
-                         stream, partial.data_ptr<float>(), G, N, db.data_ptr<float>());
+      if (G > 64) {
+        constexpr int S = 32;
+        int seg_rows = (G + S - 1) / S;
+        auto seg = at::empty({(int64_t)S * N}, dy.options().dtype(at::kFloat));
+        hipLaunchKernelGGL(colsum_seg_f32_kernel, dim3(ftiles, S),
+                           dim3(kBlockThreads), 0, stream, partial.data_ptr<float>(),
+                           G, seg_rows, N, seg.data_ptr<float>());
+        hipLaunchKernelGGL(colsum_final_f32_kernel, dim3(ftiles), dim3(kBlockThreads),
+                           0, stream, seg.data_ptr<float>(), S, N,
+                           db.data_ptr<float>());
+      } else {
+        hipLaunchKernelGGL(colsum_final_f32_kernel, dim3(ftiles), dim3(kBlockThreads),
+                           0, stream, partial.data_ptr<float>(), G, N,
+                           db.data_ptr<float>());
+      }
     }
   });
   return {g, db};
