@@ -183,6 +183,9 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
     # traffic and routes the interaction onto its bf16 MFMA path; the pool
     # still accumulates fp32 in-kernel (TREC_EMB_OUT=fp32 to disable)
     emb_out = os.environ.get("TREC_EMB_OUT", "bf16" if not cpu_mode else "fp32")
+    # the eager pipeline is launch-bound: split-K wgrad's extra launches cost
+    # more than the kernels save (A/B: 2.08 vs 2.29 ms/step)
+    os.environ.setdefault("TREC_SPLITK_WGRAD", "0")
     model = build_model(scale, emb_precision)
     fused_params = {
         "optimizer": "rowwise_adagrad",

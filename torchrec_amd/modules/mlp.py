@@ -42,14 +42,18 @@ _USE_COLSUM = os.environ.get("TREC_COLSUM_LINEAR", "0") == "1"
 _USE_FUSED_MLP = os.environ.get("TREC_FUSED_MLP", "1") == "1"
 
 
-_USE_SPLITK = os.environ.get("TREC_SPLITK_WGRAD", "1") == "1"
+def _use_splitk() -> bool:
+    # dynamic: the hipGraph-captured bench (GPU-time-bound) wins ~25 us per
+    # deep-K wgrad; the eager pipeline (launch-bound) loses more to the two
+    # extra launches than the kernels save
+    return os.environ.get("TREC_SPLITK_WGRAD", "1") == "1"
 
 
 def _splitk_wgrad(g: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     """dW = g^T @ x with K = batch split across a bmm (split-K that
     hipBLASLt's heuristic refuses to pick for TN deep-K shapes)."""
     B = g.shape[0]
-    if _USE_SPLITK and B >= 4096 and B % 8 == 0:
+    if _use_splitk() and B >= 4096 and B % 8 == 0:
         gv = g.view(8, B // 8, g.shape[1])
         xv = x.view(8, B // 8, x.shape[1])
         return torch.bmm(gv.transpose(1, 2), xv).sum(0)
