@@ -35,7 +35,12 @@ def main():
     dmp = DistributedModelParallel(
         model,
         sharders=[EmbeddingBagCollectionSharder(
-            fused_params={"optimizer": "rowwise_adagrad", "learning_rate": 0.02})],
+            fused_params={
+            "optimizer": "rowwise_adagrad", "learning_rate": 0.02,
+            # round-2 defaults under test: bf16 pooled output (MFMA
+            # interaction bf16-io path) + stochastic rounding machinery
+            "output_dtype": os.environ.get("TREC_EMB_OUT", "bf16"),
+        })],
         device=device, init_data_parallel=False,
     )
     opt = torch.optim.Adam([p for p in dmp.parameters() if p.requires_grad], lr=1e-3)
