@@ -462,3 +462,61 @@ def test_percentile_logger():
     assert abs(stats["p50"] - 50) <= 1.5
     assert abs(stats["p99"] - 99) <= 1.5
     assert stats["count"] == 100
+
+
+class TestDeltaTrackerDepth:
+    """Round-2 tracker semantics: batch-indexed store, per-consumer cursors,
+    UpdateMode FIRST/LAST, compaction (reference model_delta_tracker.py:139,
+    delta_store.py:144)."""
+
+    def test_delta_store_update_modes(self):
+        from torchrec_amd.distributed.model_tracker import DeltaStore, UpdateMode
+
+        for mode, expect in [(UpdateMode.FIRST, [1.0, 2.0]), (UpdateMode.LAST, [10.0, 2.0])]:
+            st = DeltaStore(mode)
+            st.append(0, "t", torch.tensor([5, 6]),
+                      torch.tensor([[1.0], [2.0]]))
+            st.append(1, "t", torch.tensor([5]), torch.tensor([[10.0]]))
+            u = st.get_unique()["t"]
+            assert u.ids.tolist() == [5, 6]
+            assert u.rows.squeeze(1).tolist() == expect
+
+    def test_delta_store_compact_and_delete(self):
+        from torchrec_amd.distributed.model_tracker import DeltaStore, UpdateMode
+
+        st = DeltaStore(UpdateMode.NONE)
+        for b in range(4):
+            st.append(b, "t", torch.tensor([b, 100]))
+        st.compact(0, 3)
+        assert st.batch_indices() == [0, 3]
+        u = st.get_unique()["t"]
+        assert sorted(u.ids.tolist()) == [0, 1, 2, 3, 100]
+        st.delete(3)
+        assert st.batch_indices() == [3]
+
+    def test_per_consumer_cursors(self):
+        from torchrec_amd.distributed.model_tracker import (
+            DeltaStore, ModelDeltaTracker, UpdateMode,
+        )
+
+        class Dummy(torch.nn.Module):
+            def sharded_modules(self):
+                return {}
+
+        tr = ModelDeltaTracker(Dummy(), consumers=["a", "b"], delete_on_read=True)
+        tr.record_ids("t", torch.tensor([1, 2]))
+        tr.step()
+        got_a = tr.get_unique_ids("a")
+        assert got_a["t"].tolist() == [1, 2]
+        # consumer b still sees the delta (not deleted until ALL consumed)
+        tr.record_ids("t", torch.tensor([3]))
+        tr.step()
+        got_b = tr.get_unique_ids("b")
+        assert sorted(got_b["t"].tolist()) == [1, 2, 3]
+        # a only sees what arrived after its cursor
+        got_a2 = tr.get_unique_ids("a")
+        assert got_a2["t"].tolist() == [3]
+        import pytest as _pytest
+
+        with _pytest.raises(ValueError):
+            tr.get_unique_ids("nope")
