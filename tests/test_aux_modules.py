@@ -520,3 +520,39 @@ class TestDeltaTrackerDepth:
 
         with _pytest.raises(ValueError):
             tr.get_unique_ids("nope")
+
+
+def test_itep_reset_weight_momentum_and_stats():
+    """Evicted physical rows must reset weights + momentum in the attached
+    lookups (reference itep_modules.py:412) and show up in the stats."""
+    from torchrec_amd.modules.itep_modules import GenericITEPModule
+    from torchrec_amd.ops.tbe import TableBatchedEmbeddingBags
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    torch.manual_seed(0)
+    tbe = TableBatchedEmbeddingBags([("t0", 8, 4)], learning_rate=0.1)
+    tbe.weights.data.fill_(5.0)
+    tbe.momentum.fill_(2.0)
+    itep = GenericITEPModule(
+        {"t0": 64}, lookups=[tbe], pruning_interval=1, pruned_hash_sizes={"t0": 8},
+        pruning_warmup_iters=0,
+    )
+    itep.train()
+    # ids 50..55 are unmapped (physical space owns ids 0..7): hot misses
+    for _ in range(3):
+        kjt = KeyedJaggedTensor(
+            keys=["f0"],
+            values=torch.tensor([50, 51, 52, 50, 51, 50]),
+            lengths=torch.tensor([6]),
+            stride=1,
+        )
+        itep.remap(kjt, {"f0": "t0"})
+    stats = itep.eviction_stats()
+    assert stats["t0"]["evicted_rows_total"] >= 1
+    # some physical rows were reset to 0 (weights and momentum)
+    w = tbe.split_embedding_weights()[0]
+    assert bool((w == 0).all(dim=1).any())
+    assert bool((tbe.momentum == 0).any())
+    # and the hot miss ids now own physical rows
+    out = itep.remap_table("t0", torch.tensor([50]))
+    assert int(out) != itep.pruned_size("t0") - 1  # not the sacrificial row

@@ -26,11 +26,19 @@ class GenericITEPModule(nn.Module):
         pruning_interval: int = 1001,
         enable_pruning: bool = True,
         pruned_hash_sizes: Optional[Dict[str, int]] = None,
+        pruning_warmup_iters: int = 0,
     ) -> None:
         super().__init__()
         self._unpruned = dict(table_name_to_unpruned_hash_sizes)
         self._pruning_interval = pruning_interval
         self._enable_pruning = enable_pruning
+        # lookup modules (TBE hosts) so evicted rows get their weights and
+        # optimizer momentum reset (reference itep_modules.py:412
+        # reset_weight_momentum)
+        self._lookups = list(lookups or [])
+        self._pruning_warmup = pruning_warmup_iters
+        self._evicted_total: Dict[str, int] = {}
+        self._prune_rounds = 0
         self._iter = 0
         self._address_lookup: Dict[str, torch.Tensor] = {}
         self._row_util: Dict[str, torch.Tensor] = {}
@@ -83,7 +91,11 @@ class GenericITEPModule(nn.Module):
             )
         if self.training:
             self._iter += 1
-            if self._enable_pruning and self._iter % self._pruning_interval == 0:
+            if (
+                self._enable_pruning
+                and self._iter > self._pruning_warmup
+                and self._iter % self._pruning_interval == 0
+            ):
                 self.prune()
         return KeyedJaggedTensor.from_jt_dict({k: out[k] for k in features.keys()})
 
@@ -117,7 +129,51 @@ class GenericITEPModule(nn.Module):
             owner[rows] = new_ids
             util[rows] = top_cnt[promote]
             evicted[name] = rows
+            self._evicted_total[name] = self._evicted_total.get(name, 0) + int(rows.numel())
+        self._prune_rounds += 1
+        if evicted:
+            self.reset_weight_momentum(evicted)
         return evicted
+
+    @torch.no_grad()
+    def reset_weight_momentum(self, evicted: Dict[str, torch.Tensor]) -> None:
+        """Zero the evicted physical rows' weights and optimizer momentum in
+        the attached lookup modules so re-assigned rows start fresh
+        (reference itep_modules.py:412)."""
+        for lookup in self._lookups:
+            tbes = getattr(lookup, "tbes", None)
+            tbe_list = tbes() if callable(tbes) else [lookup]
+            for tbe in tbe_list:
+                specs = getattr(tbe, "embedding_specs", None)
+                if specs is None:
+                    continue
+                weights = tbe.split_embedding_weights()
+                states = tbe.split_optimizer_states()
+                for i, spec in enumerate(specs):
+                    rows = evicted.get(spec.name)
+                    if rows is None:
+                        continue
+                    dev_rows = rows.to(weights[i].device)
+                    weights[i][dev_rows] = 0
+                    for st in states[i]:
+                        st[dev_rows.to(st.device)] = 0
+
+    def eviction_stats(self) -> Dict[str, Dict[str, float]]:
+        """Per-table pruning telemetry (reference
+        print_itep_eviction_stats :170)."""
+        out: Dict[str, Dict[str, float]] = {}
+        for name, pruned in self._pruned_sizes.items():
+            total = self._evicted_total.get(name, 0)
+            out[name] = {
+                "pruned_rows": float(pruned),
+                "unpruned_rows": float(self._unpruned[name]),
+                "evicted_rows_total": float(total),
+                "eviction_rate_per_round": (
+                    total / pruned / max(1, self._prune_rounds)
+                ),
+                "prune_rounds": float(self._prune_rounds),
+            }
+        return out
 
 
 class ITEPEmbeddingBagCollection(nn.Module):
