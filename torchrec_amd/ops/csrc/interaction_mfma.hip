@@ -161,7 +161,7 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kern
   // sample-invariant init: pc table (padded -1) and Tt pad columns (zeros —
   // the K-pad products are G_pad (=0) * Tt_pad, which must not be NaN)
   for (int e = threadIdx.x; e < 32 * 32; e += blockDim.x) {
-    int i = e >> 5, j = e & 31;
+    int i = e >> 5, j = e & 31;  // 32-wide pc table index, lint: wave-ok
     pc[e] = (i < F1 && j < F1) ? pair_col[i * F1 + j] : -1;
   }
   if (F1 < 32) {
@@ -179,7 +179,7 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kern
     // G[i][j] = dOut[pair(i,j)] (symmetric, zero diagonal / pads): the 1.9 KB
     // grad row is L1-resident, so the scattered rereads are cache-served
     for (int e = threadIdx.x; e < 32 * 32; e += blockDim.x) {
-      int i = e >> 5, j = e & 31;
+      int i = e >> 5, j = e & 31;  // 32-wide G index, lint: wave-ok
       int32_t p = pc[e];
       *reinterpret_cast<__bf16*>(g_tile + i * (kStrideE * 2) + j * 2) =
           static_cast<__bf16>(p >= 0 ? emb2float(grow[D + p]) : 0.f);
