@@ -338,7 +338,16 @@ def bucketize_kjt_before_all2all(
     stride unchanged) ready for KJTAllToAll with F features per rank.
     """
     F = len(kjt.keys())
-    B = kjt.stride()
+    vbe = kjt.variable_stride_per_key()
+    bag_bounds = None
+    if vbe:
+        # VBE: bag -> feature mapping rides explicit per-feature bag bounds
+        spk = [sum(sp) for sp in kjt.stride_per_key_per_rank()]
+        b = [0]
+        for x in spk:
+            b.append(b[-1] + x)
+        bag_bounds = torch.tensor(b, dtype=torch.int64, device=kjt.device())
+    B = kjt.stride() if not vbe else 0
     bl, bi, bw, bp, unbucketize = ops.block_bucketize_sparse_features(
         kjt.lengths(),
         kjt.values(),
@@ -347,13 +356,28 @@ def bucketize_kjt_before_all2all(
         block_sizes=block_sizes,
         num_buckets=num_buckets,
         weights=kjt.weights_or_none(),
+        bag_feature_bounds=bag_bounds,
     )
-    out = KeyedJaggedTensor(
-        # keys repeated per bucket — the a2a treats each bucket as a rank group
-        keys=[f"{k}" for _ in range(num_buckets) for k in kjt.keys()],
-        values=bi,
-        weights=bw,
-        lengths=bl,
-        stride=B,
-    )
+    keys = [f"{k}" for _ in range(num_buckets) for k in kjt.keys()]
+    if vbe:
+        out = KeyedJaggedTensor(
+            keys=keys,
+            values=bi,
+            weights=bw,
+            lengths=bl,
+            stride_per_key_per_rank=[
+                list(sp) for _ in range(num_buckets)
+                for sp in kjt.stride_per_key_per_rank()
+            ],
+        )
+    else:
+        out = KeyedJaggedTensor(
+            # keys repeated per bucket — the a2a treats each bucket as a rank
+            # group
+            keys=keys,
+            values=bi,
+            weights=bw,
+            lengths=bl,
+            stride=B,
+        )
     return out, unbucketize

@@ -208,12 +208,11 @@ class _VbeEmbeddingBagCollectionAwaitable(LazyAwaitable[KeyedTensor]):
 
     def _wait_impl(self) -> KeyedTensor:
         by_name: Dict[str, torch.Tensor] = dict(zip(self._local_names, self._local_parts))
-        for aw in self._awaitables:
+        for entry in self._awaitables:
+            aw, names = entry if isinstance(entry, tuple) else (entry, self._owner_major_names)
             flat = aw.wait().view(-1)
-            sizes = [
-                self._strides[f] * self._dims[f] for f in self._owner_major_names
-            ]
-            for f, part in zip(self._owner_major_names, flat.split(sizes)):
+            sizes = [self._strides[f] * self._dims[f] for f in names]
+            for f, part in zip(names, flat.split(sizes)):
                 by_name[f] = part
         values = torch.cat([by_name[f] for f in self._embedding_names])
         return KeyedTensor(
@@ -397,16 +396,23 @@ class ShardedEmbeddingBagCollection(nn.Module):
 
     def _compute_and_output_dist_vbe(self, ctx, dist_input):
         """VBE output path: the lookup emits a 1-D packed [sum_f B_f_total *
-        D_f] feature-major vector; each (feature, source-rank) block is routed
-        back to its source with one variable-split a2a (split+cat keeps the
-        reorders autograd-transparent; element splits ride the host-side
-        stride table exchanged in the splits phase)."""
+        D_f] feature-major vector per sharding.
+
+        TW: each (feature, source-rank) element block routes back to its
+        source with one variable-split a2a. RW: every rank holds partial
+        pools of ALL bags for its row bucket, merged with ONE per-feature
+        uneven reduce-scatter (reference comm_ops.py:1330
+        reduce_scatter_v_per_feature_pooled). Split+cat reorders keep both
+        autograd-transparent.
+        """
+        from torchrec_amd.distributed.comm_ops import (
+            reduce_scatter_v_per_feature_pooled,
+        )
         from torchrec_amd.distributed.dist_data import SequenceEmbeddingsAllToAll
 
         W = self._env.world_size
         dims_by_name = dict(zip(self._embedding_names, self._embedding_dims))
-        pieces_per_rank: List[List[torch.Tensor]] = [[] for _ in range(max(W, 1))]
-        name_per_rank: List[List[str]] = [[] for _ in range(max(W, 1))]
+        entries: List[Any] = []
         local_parts: List[torch.Tensor] = []
         local_names: List[str] = []
         for si, (st, lookup, kjt) in enumerate(
@@ -427,6 +433,18 @@ class ShardedEmbeddingBagCollection(nn.Module):
                 ]
                 local_parts.extend(packed.split(sizes))
                 local_names.extend(names)
+                continue
+            if st == ShardingType.ROW_WISE.value:
+                # RW: packed holds partial pools for ALL bags (feature-major,
+                # source-rank-major within feature) — one per-feature RS-v
+                # returns my own bags' sums
+                names_mine = list(kjt.keys())
+                spr = [list(sp) for sp in kjt.stride_per_key_per_rank()]
+                dims = [dims_by_name[f] for f in names_mine]
+                aw = reduce_scatter_v_per_feature_pooled(
+                    packed, spr, dims, self._env.process_group
+                )
+                entries.append((aw, names_mine))
                 continue
             # TW: reorder (f, r) element blocks to rank-major and a2a back
             spr = kjt.stride_per_key_per_rank()  # [F_mine][W]
@@ -460,15 +478,12 @@ class ShardedEmbeddingBagCollection(nn.Module):
             aw = SequenceEmbeddingsAllToAll(self._env.process_group)(
                 send.view(-1, 1), in_splits, out_splits
             )
-            # remember the owner-major feature layout for reassembly
-            for r in range(W):
-                name_per_rank[r].extend(fpr[r])
-            pieces_per_rank[0].append(aw)  # store awaitable (one per sharding)
+            entries.append((aw, [f for r in range(W) for f in fpr[r]]))
         return _VbeEmbeddingBagCollectionAwaitable(
-            awaitables=[a for a in pieces_per_rank[0]],
+            awaitables=entries,
             local_parts=local_parts,
             local_names=local_names,
-            owner_major_names=[n for r in range(max(W, 1)) for n in name_per_rank[r]],
+            owner_major_names=[],
             embedding_names=self._embedding_names,
             dims_by_name=dims_by_name,
             local_strides=ctx.vbe_local_strides or {},
@@ -503,12 +518,16 @@ class ShardedEmbeddingBagCollection(nn.Module):
                 k: sum(sp)
                 for k, sp in zip(features.keys(), features.stride_per_key_per_rank())
             }
-            assert not self._is_weighted, "VBE v1: unweighted only"
-            assert all(
-                t == ShardingType.TABLE_WISE.value for t in self._sharding_types
-            ) or self._env.world_size == 1, (
-                "VBE through the sharded path supports TW shardings (v1)"
-            )
+            assert not self._is_weighted, "VBE: unweighted only"
+            allowed = {
+                ShardingType.TABLE_WISE.value,
+                ShardingType.ROW_WISE.value,
+                ShardingType.DATA_PARALLEL.value,
+            }
+            assert (
+                all(t in allowed for t in self._sharding_types)
+                or self._env.world_size == 1
+            ), "VBE through the sharded path supports TW/RW/DP shardings"
         dist_input = self.input_dist(ctx, features).wait().wait()
         return self.compute_and_output_dist(ctx, dist_input)
 

@@ -252,6 +252,7 @@ def block_bucketize_sparse_features(
     num_buckets: int,
     weights: Optional[torch.Tensor] = None,
     total_num_blocks: Optional[torch.Tensor] = None,  # [F]; uniform blocks when set
+    bag_feature_bounds: Optional[torch.Tensor] = None,  # [F+1] VBE bag->feature
 ) -> Tuple[
     torch.Tensor,
     torch.Tensor,
@@ -272,8 +273,13 @@ def block_bucketize_sparse_features(
     if indices.is_cuda:
         hip_ops()
         w = weights if weights is not None else torch.empty(0, device=indices.device)
+        bb = (
+            bag_feature_bounds
+            if bag_feature_bounds is not None
+            else torch.empty(0, dtype=torch.int64, device=indices.device)
+        )
         bl, bi, bw, bp, up = torch.ops.trec_amd.block_bucketize_sparse_features(
-            lengths, indices, block_sizes, num_buckets, bucketize_pos, sequence, w
+            lengths, indices, block_sizes, num_buckets, bucketize_pos, sequence, w, bb
         )
         return (
             bl,
@@ -285,44 +291,53 @@ def block_bucketize_sparse_features(
     FB = lengths.numel()
     F = block_sizes.numel()
     B = FB // F
+    if bag_feature_bounds is not None:
+        bounds = [int(x) for x in bag_feature_bounds]
+        def _feat_of(bag: int) -> int:
+            for fi in range(F):
+                if bounds[fi] <= bag < bounds[fi + 1]:
+                    return fi
+            return F - 1
+    else:
+        def _feat_of(bag: int) -> int:
+            return bag // B
     offsets = complete_cumsum(lengths)
     new_lengths = torch.zeros(num_buckets * FB, dtype=lengths.dtype)
     # bucket of each value
     bucket_of = torch.empty_like(indices)
     local_idx = torch.empty_like(indices)
     pos_list = torch.empty_like(indices) if bucketize_pos else None
-    for f in range(F):
+    for bag in range(FB):
+        f = _feat_of(bag)
         bs = int(block_sizes[f])
-        for b in range(B):
-            start, end = int(offsets[f * B + b]), int(offsets[f * B + b + 1])
-            for p in range(start, end):
-                idx = int(indices[p])
-                bkt = min(idx // bs, num_buckets - 1)
-                bucket_of[p] = bkt
-                local_idx[p] = idx - bkt * bs
-                new_lengths[bkt * FB + f * B + b] += 1
-                if bucketize_pos:
-                    pos_list[p] = p - start
+        start, end = int(offsets[bag]), int(offsets[bag + 1])
+        for p in range(start, end):
+            idx = int(indices[p])
+            bkt = min(idx // bs, num_buckets - 1)
+            bucket_of[p] = bkt
+            local_idx[p] = idx - bkt * bs
+            new_lengths[bkt * FB + bag] += 1
+            if bucketize_pos:
+                pos_list[p] = p - start
     new_offsets = complete_cumsum(new_lengths)
     new_indices = torch.empty_like(indices)
     new_weights = torch.empty_like(weights) if weights is not None else None
     new_pos = torch.empty_like(indices) if bucketize_pos else None
     unbucketize = torch.empty(indices.numel(), dtype=torch.int64) if sequence else None
     cursor = new_offsets[:-1].clone()
-    for f in range(F):
-        for b in range(B):
-            start, end = int(offsets[f * B + b]), int(offsets[f * B + b + 1])
-            for p in range(start, end):
-                bkt = int(bucket_of[p])
-                slot = int(cursor[bkt * FB + f * B + b])
-                cursor[bkt * FB + f * B + b] += 1
-                new_indices[slot] = local_idx[p]
-                if new_weights is not None:
-                    new_weights[slot] = weights[p]
-                if bucketize_pos:
-                    new_pos[slot] = pos_list[p]
-                if sequence:
-                    unbucketize[p] = slot
+    for bag in range(FB):
+        start, end = int(offsets[bag]), int(offsets[bag + 1])
+        for p in range(start, end):
+            bkt = int(bucket_of[p])
+            slot = int(cursor[bkt * FB + bag])
+            cursor[bkt * FB + bag] += 1
+            new_indices[slot] = local_idx[p]
+            if new_weights is not None:
+                new_weights[slot] = weights[p]
+            if bucketize_pos:
+                new_pos[slot] = pos_list[p]
+            if sequence:
+                unbucketize[p] = slot
     return new_lengths, new_indices, new_weights, new_pos, unbucketize
 
 

@@ -19,7 +19,8 @@ at::Tensor segment_sum_csr(const at::Tensor& csr, const at::Tensor& values);
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor>
 block_bucketize_sparse_features(const at::Tensor& lengths, const at::Tensor& indices,
                                 const at::Tensor& block_sizes, int64_t num_buckets,
-                                bool bucketize_pos, bool sequence, const at::Tensor& weights);
+                                bool bucketize_pos, bool sequence, const at::Tensor& weights,
+                                const at::Tensor& bag_feature_bounds);
 at::Tensor permute_pooled_embs(const at::Tensor& values, const at::Tensor& in_offsets,
                                const at::Tensor& out_offsets, const at::Tensor& order);
 
@@ -137,8 +138,8 @@ TORCH_LIBRARY(trec_amd, m) {
   m.def("segment_sum_csr(Tensor csr, Tensor values) -> Tensor");
   m.def(
       "block_bucketize_sparse_features(Tensor lengths, Tensor indices, Tensor block_sizes,"
-      " int num_buckets, bool bucketize_pos, bool sequence, Tensor weights)"
-      " -> (Tensor, Tensor, Tensor, Tensor, Tensor)");
+      " int num_buckets, bool bucketize_pos, bool sequence, Tensor weights,"
+      " Tensor bag_feature_bounds) -> (Tensor, Tensor, Tensor, Tensor, Tensor)");
   m.def(
       "permute_pooled_embs(Tensor values, Tensor in_offsets, Tensor out_offsets,"
       " Tensor order) -> Tensor");

@@ -270,12 +270,12 @@ __global__ void bucketize_count_kernel(
     const int64_t* __restrict__ offsets,  // [FB+1]
     const index_t* __restrict__ indices,
     const int64_t* __restrict__ block_sizes,  // [F]
-    int F, int B, int num_buckets,
+    const int64_t* __restrict__ bag_bounds,   // [F+1] VBE bag->feature, or null
+    int F, int B, int num_buckets, int64_t FB,
     int64_t* __restrict__ new_lengths /* [num_buckets*FB] zeroed */) {
-  int64_t FB = static_cast<int64_t>(F) * B;
   for (int64_t bag = blockIdx.x * blockDim.x + threadIdx.x; bag < FB;
        bag += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    int f = bag / B;
+    int f = bag_bounds ? upper_bound_segment(bag_bounds, F, bag) : (int)(bag / B);
     int64_t bs = block_sizes[f];
     for (int64_t p = offsets[bag]; p < offsets[bag + 1]; ++p) {
       int64_t bkt = indices[p] / bs;
@@ -288,14 +288,14 @@ __global__ void bucketize_count_kernel(
 template <typename index_t, bool HAS_W, bool POS, bool SEQ>
 __global__ void bucketize_scatter_kernel(
     const int64_t* __restrict__ offsets, const index_t* __restrict__ indices,
-    const float* __restrict__ weights, const int64_t* __restrict__ block_sizes, int F, int B,
-    int num_buckets, const int64_t* __restrict__ new_offsets,
+    const float* __restrict__ weights, const int64_t* __restrict__ block_sizes,
+    const int64_t* __restrict__ bag_bounds, int F, int B,
+    int num_buckets, int64_t FB, const int64_t* __restrict__ new_offsets,
     index_t* __restrict__ new_indices, float* __restrict__ new_weights,
     index_t* __restrict__ new_pos, int64_t* __restrict__ unbucketize) {
-  int64_t FB = static_cast<int64_t>(F) * B;
   for (int64_t bag = blockIdx.x * blockDim.x + threadIdx.x; bag < FB;
        bag += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    int f = bag / B;
+    int f = bag_bounds ? upper_bound_segment(bag_bounds, F, bag) : (int)(bag / B);
     int64_t bs = block_sizes[f];
     int64_t start = offsets[bag];
     // per-bucket rank is recomputed by rescanning the bag prefix: O(L^2) per
@@ -324,11 +324,14 @@ __global__ void bucketize_scatter_kernel(
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor>
 block_bucketize_sparse_features(const at::Tensor& lengths, const at::Tensor& indices,
                                 const at::Tensor& block_sizes, int64_t num_buckets,
-                                bool bucketize_pos, bool sequence, const at::Tensor& weights) {
+                                bool bucketize_pos, bool sequence, const at::Tensor& weights,
+                                const at::Tensor& bag_feature_bounds) {
+  // bag_feature_bounds: empty for uniform batch (bag -> feature = bag / B);
+  // [F+1] cumulative per-feature bag counts for VBE inputs
   TORCH_CHECK(indices.is_cuda());
   int F = block_sizes.numel();
   int64_t FB = lengths.numel();
-  int B = FB / F;
+  int B = bag_feature_bounds.numel() > 0 ? 0 : (int)(FB / F);
   auto offsets = complete_cumsum(lengths.to(at::kLong).contiguous());
   auto bs = block_sizes.to(at::kLong).to(indices.device()).contiguous();
   auto new_lengths = at::zeros({num_buckets * FB}, lengths.options().dtype(at::kLong));
@@ -338,10 +341,14 @@ block_bucketize_sparse_features(const at::Tensor& lengths, const at::Tensor& ind
   auto idx = indices.contiguous();
   std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor> result;
 
+  auto bounds = bag_feature_bounds.numel() > 0
+      ? bag_feature_bounds.to(at::kLong).to(indices.device()).contiguous()
+      : at::Tensor();
+  const int64_t* bounds_ptr = bounds.defined() ? bounds.data_ptr<int64_t>() : nullptr;
   AT_DISPATCH_INDEX_TYPES(idx.scalar_type(), "bucketize", [&] {
     hipLaunchKernelGGL(bucketize_count_kernel<index_t>, dim3(grid), dim3(kBlockThreads), 0,
                        stream, offsets.data_ptr<int64_t>(), idx.data_ptr<index_t>(),
-                       bs.data_ptr<int64_t>(), F, B, (int)num_buckets,
+                       bs.data_ptr<int64_t>(), bounds_ptr, F, B, (int)num_buckets, FB,
                        new_lengths.data_ptr<int64_t>());
     auto new_offsets = complete_cumsum(new_lengths);
     auto new_indices = at::empty_like(idx);
@@ -355,7 +362,7 @@ block_bucketize_sparse_features(const at::Tensor& lengths, const at::Tensor& ind
                                                    decltype(pos_c)::value, decltype(seq_c)::value>),
                          dim3(grid), dim3(kBlockThreads), 0, stream, offsets.data_ptr<int64_t>(),
                          idx.data_ptr<index_t>(), has_w ? weights.data_ptr<float>() : nullptr,
-                         bs.data_ptr<int64_t>(), F, B, (int)num_buckets,
+                         bs.data_ptr<int64_t>(), bounds_ptr, F, B, (int)num_buckets, FB,
                          new_offsets.data_ptr<int64_t>(), new_indices.data_ptr<index_t>(),
                          has_w ? new_weights.data_ptr<float>() : nullptr,
                          bucketize_pos ? new_pos.data_ptr<index_t>() : nullptr,
