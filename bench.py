@@ -250,7 +250,10 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
     if not cpu_mode:
         _dense_to_bf16(dmp)  # before DDP wrap: buckets must see the bf16 params
         _graph_dense_modules(dmp, batch_size, device)
-    if world > 1:
+    if pg is not None:
+        # world 1 under TREC_FORCE_DIST also wraps DDP: the single-GPU
+        # torchrun rehearsal then exercises the exact graphed-dense + DDP
+        # hook machinery the N>1 scaling run depends on
         dmp.init_data_parallel()
     if rank == 0:
         counts = {}
