@@ -356,6 +356,10 @@ class TableBatchedEmbeddingBags(nn.Module):
                 for i in range(len(feature_table_map) - 1)
             )
             and self._total_rows < (1 << 31)
+            # TREC_SEG_SORT=0: force the hipCUB device radix path (in a
+            # hipGraph-captured step its ~20 launches are free and it uses
+            # the whole chip, vs one workgroup per feature segment here)
+            and os.environ.get("TREC_SEG_SORT", "1") == "1"
         )
         if self._uvm_caching:
             # cache sizing: cache_load_factor of total rows, 32 ways per set
