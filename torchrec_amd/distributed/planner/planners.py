@@ -193,13 +193,19 @@ class EmbeddingEnumerator:
         return options
 
     def _estimate(self, opt: ShardingOption, cons: Optional[ParameterConstraints]) -> None:
-        """Perf + storage per shard (reference shard_estimators.py:71,126)."""
+        """Perf + storage per shard with per-sharding-type I/O models
+        (reference shard_estimators.py:71 perf, :126 storage, :702-862
+        _calculate_{tw,rw,cw,twrw}_shard_io_sizes — re-derived for MI355X:
+        xGMI collectives are per-link bound, so a2a/RS/AG costs carry the
+        (W-1)/W ring factor against intra_host_bw)."""
         topo = self._topology
         B = topo.batch_size
         W = topo.world_size
+        L = max(1, topo.local_world_size)
         pooling = cons.pooling_factors[0] if cons else constants.POOLING_FACTOR
         D = opt.config.embedding_dim
         elem = 4  # fp32
+        id_bytes = 8  # int64 ids on the input-dist wire
         opt_mult = (
             0.0
             if opt.compute_kernel == EmbeddingComputeKernel.DENSE.value
@@ -209,6 +215,10 @@ class EmbeddingEnumerator:
         is_cached = opt.compute_kernel == EmbeddingComputeKernel.FUSED_UVM_CACHING.value
         CACHE_LOAD = 0.2
         CACHE_HIT = 0.8  # assumed; refine with table stats
+        n_feat = max(1, len(opt.config.feature_names))
+        ring = (W - 1) / W if W > 1 else 0.0
+        ring_node = (max(1, W // L) - 1) / max(1, W // L)
+        st = opt.sharding_type
         for shard in opt.shards:
             rows, cols = shard.size
             # storage: weights + optimizer + a slice of activation/grad buffers
@@ -221,29 +231,70 @@ class EmbeddingEnumerator:
                 shard.storage = Storage(hbm=act_bytes + cache_bytes, ddr=weight_bytes)
             else:
                 shard.storage = Storage(hbm=weight_bytes + act_bytes, ddr=0)
-            # perf: bytes moved / bandwidth
-            if opt.sharding_type == ShardingType.DATA_PARALLEL.value:
-                local_B = B
-                # dense-kernel autograd path + grad allreduce: heavily penalized
-                # (reference constants.py DP_ELEMENTWISE_KERNELS_PERF_FACTOR)
+
+            # ---- per-sharding-type I/O (bytes on this shard's rank) ----
+            prefetch = 0.0
+            if st == ShardingType.DATA_PARALLEL.value:
+                # local batch through a DENSE autograd kernel + grad allreduce
                 fwd_bytes = (
-                    local_B * pooling * cols * elem
+                    B * pooling * n_feat * cols * elem
                 ) * constants.DP_ELEMENTWISE_KERNELS_PERF_FACTOR
-                comms = (
-                    2 * rows * cols * elem / topo.intra_host_bw if W > 1 else 0.0
+                input_comms = 0.0
+                output_comms = 0.0
+                bwd_comms = (
+                    2.0 * ring * rows * cols * elem / topo.intra_host_bw
+                    if W > 1 else 0.0
                 )
-            elif opt.sharding_type == ShardingType.ROW_WISE.value:
-                local_B = B * W
-                fwd_bytes = local_B * pooling / W * cols * elem
-                comms = local_B * cols * elem / topo.intra_host_bw if W > 1 else 0.0
-            else:  # TW / CW: global batch through this shard
-                local_B = B * W
-                fwd_bytes = local_B * pooling * cols * elem
-                comms = local_B * cols * elem / topo.intra_host_bw if W > 1 else 0.0
+            elif st == ShardingType.ROW_WISE.value:
+                # every rank reads ~1/W of the GLOBAL batch's ids; pooled
+                # partials reduce-scatter back (reference
+                # _calculate_rw_shard_io_sizes)
+                gB = B * W
+                fwd_bytes = gB * pooling * n_feat / W * cols * elem
+                input_comms = gB * pooling * n_feat / W * id_bytes / topo.intra_host_bw
+                output_comms = (
+                    ring * gB * n_feat * cols * elem / topo.intra_host_bw
+                )
+                bwd_comms = output_comms  # AG of grads mirrors the RS
+            elif st == ShardingType.TABLE_ROW_WISE.value:
+                # rows split over one node: intra-node RS + cross-node a2a
+                gB = B * W
+                fwd_bytes = gB * pooling * n_feat / L * cols * elem
+                input_comms = gB * pooling * n_feat / L * id_bytes / topo.intra_host_bw
+                rs = (L - 1) / L * gB * n_feat * cols * elem / topo.intra_host_bw
+                a2a = ring_node * gB * n_feat * cols * elem / topo.inter_host_bw
+                output_comms = rs + a2a
+                bwd_comms = output_comms
+            elif st == ShardingType.GRID_SHARD.value:
+                gB = B * W
+                fwd_bytes = gB * pooling * n_feat / L * cols * elem
+                input_comms = gB * pooling * n_feat / L * id_bytes / topo.intra_host_bw
+                rs = (L - 1) / L * gB * n_feat * cols * elem / topo.intra_host_bw
+                a2a = ring_node * gB * n_feat * cols * elem / topo.inter_host_bw
+                output_comms = rs + a2a
+                bwd_comms = output_comms
+            else:
+                # TW / CW / TWCW: the GLOBAL batch's bags flow through this
+                # shard; ids a2a in, pooled slice a2a out (reference
+                # _calculate_tw/cw_shard_io_sizes)
+                gB = B * W
+                fwd_bytes = gB * pooling * n_feat * cols * elem
+                input_comms = (
+                    ring * gB * pooling * n_feat * id_bytes / topo.intra_host_bw
+                )
+                output_comms = (
+                    ring * gB * n_feat * cols * elem / topo.intra_host_bw
+                )
+                bwd_comms = output_comms
+            if W == 1:
+                input_comms = output_comms = bwd_comms = 0.0
+
             if is_uvm:
                 mem_bw = topo.ddr_mem_bw
             elif is_cached:
                 mem_bw = 1.0 / (CACHE_HIT / topo.hbm_mem_bw + (1 - CACHE_HIT) / topo.ddr_mem_bw)
+                # cache maintenance traffic ahead of the forward
+                prefetch = (1 - CACHE_HIT) * fwd_bytes / topo.ddr_mem_bw
             else:
                 mem_bw = topo.hbm_mem_bw
             # narrow-shard penalty (CW/TWCW/GRID): the TBE assigns 64/LPS
@@ -255,9 +306,10 @@ class EmbeddingEnumerator:
             fwd_compute = fwd_bytes / (mem_bw * max(lane_eff, 1e-3))
             shard.perf = Perf(
                 fwd_compute=fwd_compute,
-                fwd_comms=comms,
+                fwd_comms=input_comms + output_comms,
                 bwd_compute=fwd_compute * constants.BWD_COMPUTE_MULTIPLIER,
-                bwd_comms=comms,
+                bwd_comms=bwd_comms,
+                prefetch_compute=prefetch,
             )
 
 

@@ -38,10 +38,14 @@ class Perf:
     fwd_comms: float = 0.0
     bwd_compute: float = 0.0
     bwd_comms: float = 0.0
+    prefetch_compute: float = 0.0
 
     @property
     def total(self) -> float:
-        return self.fwd_compute + self.fwd_comms + self.bwd_compute + self.bwd_comms
+        return (
+            self.fwd_compute + self.fwd_comms + self.bwd_compute + self.bwd_comms
+            + self.prefetch_compute
+        )
 
     def __add__(self, other: "Perf") -> "Perf":
         return Perf(
@@ -49,6 +53,7 @@ class Perf:
             self.fwd_comms + other.fwd_comms,
             self.bwd_compute + other.bwd_compute,
             self.bwd_comms + other.bwd_comms,
+            self.prefetch_compute + other.prefetch_compute,
         )
 
 
