@@ -296,8 +296,12 @@ class ShardedEmbeddingCollection(nn.Module):
             if st != ShardingType.DATA_PARALLEL.value and W > 1
         )
         if env.process_group is not None and n_comm > 1:
-            from torchrec_amd.distributed.embeddingbag import _sharding_out_pg
+            from torchrec_amd.distributed.embeddingbag import (
+                _sharding_out_pg,
+                next_module_ordinal,
+            )
 
+            base = 1_000_000 + next_module_ordinal() * 100
             first = True
             extra = 0
             for si, st in enumerate(self._sharding_types):
@@ -306,9 +310,9 @@ class ShardedEmbeddingCollection(nn.Module):
                 if first:
                     first = False  # first sharding keeps the shared pg
                     continue
-                # offset the cache index so EC groups never collide with the
-                # pooled-EBC groups of the same module tree
-                pg = _sharding_out_pg(env, 1000 + extra)
+                # module-ordinal namespace: EC groups never collide with
+                # pooled-EBC groups or other EC modules
+                pg = _sharding_out_pg(env, base + extra)
                 extra += 1
                 self._seq_a2a[si] = SequenceEmbeddingsAllToAll(pg)
 

@@ -41,6 +41,15 @@ from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor, KeyedTensor
 
 
 _OUT_PG_CACHE: Dict[Any, Any] = {}
+# each sharded module claims a distinct communicator namespace so two
+# multi-sharding-type sparse modules never share output communicators
+# (their backward collectives could otherwise interleave across modules)
+_MODULE_ORDINAL = [0]
+
+
+def next_module_ordinal() -> int:
+    _MODULE_ORDINAL[0] += 1
+    return _MODULE_ORDINAL[0]
 
 
 def _sharding_out_pg(env: ShardingEnv, index: int):
@@ -52,11 +61,10 @@ def _sharding_out_pg(env: ShardingEnv, index: int):
     repeated sharding (resharding, multiple sharded modules) REUSES the
     communicators instead of growing an unbounded RCCL comm set.
 
-    KNOWN EDGE: two sharded modules reuse the same per-index groups, so a
-    model with MULTIPLE multi-sharding-type sparse modules still shares
-    communicators across modules (as all single-sharding modules share the
-    default group); module-granular backward ordering is deterministic for
-    the supported pipelines."""
+    Callers namespace the index with next_module_ordinal() so distinct
+    sharded modules get distinct communicators; module construction order is
+    identical on every rank (DMP shards deterministically), keeping the
+    collective new_group calls aligned."""
     import torch.distributed as dist_mod
 
     my_ranks = tuple(dist_mod.get_process_group_ranks(env.process_group))
@@ -259,10 +267,12 @@ class ShardedEmbeddingBagCollection(nn.Module):
             create_sharding(t, infos, env, self._device) for t, infos in by_type.items()
         ]
         # mixed sharding types: one communicator per sharding so their
-        # backward collectives never need a cross-rank issue order
+        # backward collectives never need a cross-rank issue order; the
+        # module ordinal keeps communicators module-private
         if env.process_group is not None and len(self._shardings) > 1:
+            base = next_module_ordinal() * 100
             for i, sh in enumerate(self._shardings[1:]):
-                sh._pg_out = _sharding_out_pg(env, i)
+                sh._pg_out = _sharding_out_pg(env, base + i)
 
         self._input_dists = nn.ModuleList(
             [s.create_input_dist(self._device) for s in self._shardings]
