@@ -102,6 +102,9 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
         super().__init__()
         fused_params = fused_params or {}
         self._grouped_tables = grouped_tables
+        self._out_torch_dtype = {
+            "fp32": torch.float32, "bf16": torch.bfloat16, "fp16": torch.float16
+        }[fused_params.get("output_dtype", "fp32")]
         self._emb_modules = nn.ModuleList()
         self._feature_splits: List[int] = []
         self._group_dims: List[int] = []
@@ -200,9 +203,11 @@ class GroupedPooledEmbeddingsLookup(nn.Module):
         if len(self._emb_modules) == 0:
             # this rank holds no shards of this sharding — contributes 0 cols.
             # requires_grad anchors the output a2a in the autograd graph so
-            # this rank still issues the BACKWARD collective its peers expect
+            # this rank still issues the BACKWARD collective its peers expect.
+            # dtype must match the peers' pooled output (a2a dtype agreement)
             return torch.zeros(
-                B, 0, dtype=torch.float32, device=sparse_features.device(),
+                B, 0, dtype=self._out_torch_dtype,
+                device=sparse_features.device(),
                 requires_grad=torch.is_grad_enabled(),
             )
         if sparse_features.variable_stride_per_key():

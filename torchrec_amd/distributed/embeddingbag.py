@@ -175,7 +175,9 @@ class EmbeddingBagCollectionAwaitable(LazyAwaitable[KeyedTensor]):
                 if c0 > prev:
                     pieces.append(values[:, prev:c0])
                 div = self._ctx.mean_divisors[si][fname]
-                pieces.append(values[:, c0:c1] / div.unsqueeze(1))
+                pieces.append(
+                    (values[:, c0:c1] / div.unsqueeze(1).to(values.dtype))
+                )
                 prev = c1
             if prev < values.shape[1]:
                 pieces.append(values[:, prev:])
