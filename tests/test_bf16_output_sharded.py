@@ -51,3 +51,38 @@ def _run(rank, world):
 
 def test_bf16_output_tw_world2():
     run_multi_process(_run, 2, "gloo")
+
+
+def _run_cw(rank, world):
+    torch.manual_seed(0)
+    tables = make_tables()
+    model = SparseModel(make_tables())
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={
+            "optimizer": "rowwise_adagrad",
+            "learning_rate": LR,
+            "output_dtype": "bf16",
+        }
+    )
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            cfg.name: ParameterConstraints(
+                sharding_types=[ShardingType.COLUMN_WISE.value]
+            )
+            for cfg in tables
+        },
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    B = 4
+    kjt = kjt_local_slice(make_global_kjt(tables, B * world, seed=3), rank * B, (rank + 1) * B)
+    kt = dmp(kjt)
+    assert kt.values().dtype == torch.bfloat16
+    kt.values().float().sum().backward()
+
+
+def test_bf16_output_cw_world2():
+    run_multi_process(_run_cw, 2, "gloo")
