@@ -287,3 +287,55 @@ class TestColSum:
         torch.testing.assert_close(p._linear.weight.grad, ref.weight.grad, atol=1e-5, rtol=1e-5)
         torch.testing.assert_close(p._linear.bias.grad, ref.bias.grad, atol=1e-5, rtol=1e-5)
         torch.testing.assert_close(x.grad, x2.grad, atol=1e-5, rtol=1e-5)
+
+
+class TestSegSort2Level:
+    """Two-level segmented sort vs torch.sort oracle (stability included)."""
+
+    @pytest.mark.parametrize("B,L,F", [(512, 1, 4), (8192, 1, 26), (2048, 3, 7),
+                                       (65536, 1, 3)])
+    def test_matches_stable_sort(self, B, L, F):
+        ops.hip_ops()
+        torch.manual_seed(0)
+        rows = 1 << 20
+        lengths = torch.full((F * B,), L, dtype=torch.int64)
+        offsets = torch.zeros(F * B + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        vals = torch.cat([
+            torch.randint(0, rows, (B * L,)) + f * rows for f in range(F)
+        ]).cuda()
+        cap = B * L
+        sorted_lin, perm, _ = torch.ops.trec_amd.seg_sort_pairs_2level(
+            vals, offsets.cuda(), B, F, 26, cap,
+        )
+        torch.cuda.synchronize()
+        for f in range(F):
+            seg = vals[f * B * L : (f + 1) * B * L].cpu()
+            ref_vals, ref_idx = torch.sort(seg, stable=True)
+            got = sorted_lin[f * B * L : (f + 1) * B * L].cpu()
+            assert torch.equal(got, ref_vals), f"segment {f} keys mismatch"
+            got_perm = perm[f * B * L : (f + 1) * B * L].cpu() - f * B * L
+            assert torch.equal(got_perm.long(), ref_idx), f"segment {f} not stable"
+
+    def test_ragged_segments(self):
+        ops.hip_ops()
+        torch.manual_seed(1)
+        # bags of varying length: segment sizes differ from capacity
+        F, B = 3, 700
+        lengths = torch.randint(0, 4, (F * B,), dtype=torch.int64)
+        offsets = torch.zeros(F * B + 1, dtype=torch.int64)
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        n = int(lengths.sum())
+        vals = torch.randint(0, 1 << 20, (n,)).cuda()
+        cap = 4 * B
+        sorted_lin, perm, _ = torch.ops.trec_amd.seg_sort_pairs_2level(
+            vals, offsets.cuda(), B, F, 26, cap,
+        )
+        torch.cuda.synchronize()
+        for f in range(F):
+            lo = int(offsets[f * B])
+            hi = int(offsets[(f + 1) * B])
+            seg = vals[lo:hi].cpu()
+            ref_vals, ref_idx = torch.sort(seg, stable=True)
+            assert torch.equal(sorted_lin[lo:hi].cpu(), ref_vals)
+            assert torch.equal(perm[lo:hi].cpu().long() - lo, ref_idx)

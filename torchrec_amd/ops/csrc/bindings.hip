@@ -47,6 +47,9 @@ std::tuple<at::Tensor, at::Tensor> sort_pairs(const at::Tensor& keys, int64_t en
 std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs(
     const at::Tensor& linear, const at::Tensor& offsets, int64_t B, int64_t F,
     int64_t end_bit, int64_t capacity);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_2level(
+    const at::Tensor& linear, const at::Tensor& offsets, int64_t B, int64_t F,
+    int64_t end_bit, int64_t capacity);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_large(
     const at::Tensor& linear, const at::Tensor& feat_bounds, int64_t F, int64_t end_bit);
 std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_linear);
@@ -230,6 +233,7 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("tbe_forward_pooled_vbe", trec_amd::tbe_forward_pooled_vbe);
   m.impl("sort_pairs", trec_amd::sort_pairs);
   m.impl("seg_sort_pairs", trec_amd::seg_sort_pairs);
+  m.impl("seg_sort_pairs_2level", trec_amd::seg_sort_pairs_2level);
   m.impl("seg_sort_pairs_large", trec_amd::seg_sort_pairs_large);
   m.impl("tbe_backward_prep", trec_amd::tbe_backward_prep);
   m.impl("tbe_backward_fused", trec_amd::tbe_backward_fused);

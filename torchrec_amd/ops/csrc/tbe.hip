@@ -539,6 +539,191 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs(
   return {sorted, perm, overflow};
 }
 
+// ---------------------------------------------------------------------------
+// two-level segmented sort: stable tile sorts (512 keys/block — hundreds of
+// workgroups fill the chip, vs ONE workgroup per segment in the single-level
+// block sort) + log2(tiles) rounds of stable co-rank pairwise merges.
+// No per-segment size cap (B = 65536 runs 128 tiles x 7 rounds).
+// ---------------------------------------------------------------------------
+
+constexpr int kTileThreads = 256;
+constexpr int kTileItems = 2;
+constexpr int kTileSize = kTileThreads * kTileItems;  // 512 keys per tile
+
+__global__ void __launch_bounds__(kTileThreads) seg_tile_sort_kernel(
+    const int64_t* __restrict__ linear, const int64_t* __restrict__ offsets, int B,
+    int tiles_per_seg, int end_bit, uint32_t* __restrict__ keys_out,
+    uint32_t* __restrict__ perm_out) {
+  using sorter = rocprim::block_radix_sort<uint32_t, kTileThreads, kTileItems, uint32_t>;
+  __shared__ typename sorter::storage_type storage;
+  int seg = blockIdx.x / tiles_per_seg;
+  int tile = blockIdx.x - seg * tiles_per_seg;
+  int64_t lo = offsets[static_cast<int64_t>(seg) * B];
+  int64_t hi = offsets[static_cast<int64_t>(seg + 1) * B];
+  int64_t t0 = lo + static_cast<int64_t>(tile) * kTileSize;
+  if (t0 >= hi) return;
+  int count = static_cast<int>(min(hi - t0, (int64_t)kTileSize));
+  uint32_t keys[kTileItems];
+  uint32_t vals[kTileItems];
+#pragma unroll
+  for (int i = 0; i < kTileItems; ++i) {
+    int k = static_cast<int>(threadIdx.x) * kTileItems + i;  // blocked
+    if (k < count) {
+      keys[i] = static_cast<uint32_t>(linear[t0 + k]);
+      vals[i] = static_cast<uint32_t>(t0 + k);
+    } else {
+      keys[i] = 0xFFFFFFFFu;
+      vals[i] = 0xFFFFFFFFu;
+    }
+  }
+  sorter().sort(keys, vals, storage, 0u, static_cast<unsigned>(end_bit));
+#pragma unroll
+  for (int i = 0; i < kTileItems; ++i) {
+    int k = static_cast<int>(threadIdx.x) * kTileItems + i;
+    if (k < count) {
+      keys_out[t0 + k] = keys[i];
+      perm_out[t0 + k] = vals[i];
+    }
+  }
+}
+
+// co-rank of out position k in sorted runs A[0..la), B[0..lb): smallest ai
+// such that merging is consistent; stable (ties take A first).
+__device__ __forceinline__ int co_rank(int k, const uint32_t* A, int la,
+                                       const uint32_t* Bp, int lb) {
+  // monotone predicate: ai is too small while A[ai] <= B[bi-1] (stability:
+  // equal keys take ALL of A's before any of B's). In-loop ai < hi <= la and
+  // bi = k - ai >= 1, so both accesses are in range.
+  int lo = max(0, k - lb), hi = min(k, la);
+  while (lo < hi) {
+    int ai = (lo + hi) >> 1;
+    int bi = k - ai;
+    if (A[ai] <= Bp[bi - 1]) {
+      lo = ai + 1;
+    } else {
+      hi = ai;
+    }
+  }
+  return lo;
+}
+
+template <bool OUT64>
+__global__ void __launch_bounds__(kTileThreads) seg_merge_kernel(
+    const uint32_t* __restrict__ keys_in, const uint32_t* __restrict__ vals_in,
+    const int64_t* __restrict__ offsets, int B, int chunks_per_seg, int run_len,
+    uint32_t* __restrict__ keys_out32, int64_t* __restrict__ keys_out64,
+    uint32_t* __restrict__ vals_out32, int32_t* __restrict__ vals_out32_final) {
+  int seg = blockIdx.x / chunks_per_seg;
+  int chunk = blockIdx.x - seg * chunks_per_seg;
+  int64_t lo = offsets[static_cast<int64_t>(seg) * B];
+  int64_t hi = offsets[static_cast<int64_t>(seg + 1) * B];
+  int seg_len = static_cast<int>(hi - lo);
+  int out0 = chunk * kTileSize;
+  if (out0 >= seg_len) return;
+  // the pair of runs this chunk's outputs come from (chunks never straddle
+  // a pair: kTileSize divides run_len)
+  int pair0 = (out0 / (2 * run_len)) * (2 * run_len);
+  int la = min(run_len, seg_len - pair0);
+  int lb = max(0, min(run_len, seg_len - (pair0 + run_len)));
+  const uint32_t* A = keys_in + lo + pair0;
+  const uint32_t* Bp = A + run_len;
+  const uint32_t* Av = vals_in + lo + pair0;
+  const uint32_t* Bv = Av + run_len;
+  int rel0 = out0 - pair0;
+  for (int i = threadIdx.x; i < kTileSize; i += kTileThreads) {
+    int k = rel0 + i;
+    if (out0 + i >= seg_len || k >= la + lb) break;
+    int ai = co_rank(k, A, la, Bp, lb);
+    int bi = k - ai;
+    bool take_a = (bi >= lb) || (ai < la && A[ai] <= Bp[bi]);
+    uint32_t kv = take_a ? A[ai] : Bp[bi];
+    uint32_t vv = take_a ? Av[ai] : Bv[bi];
+    int64_t out_idx = lo + out0 + i;
+    if (OUT64) {
+      keys_out64[out_idx] = static_cast<int64_t>(kv);
+      vals_out32_final[out_idx] = static_cast<int32_t>(vv);
+    } else {
+      keys_out32[out_idx] = kv;
+      vals_out32[out_idx] = vv;
+    }
+  }
+}
+
+__global__ void seg_cast_out_kernel(const uint32_t* __restrict__ keys,
+                                    const uint32_t* __restrict__ vals, int64_t n,
+                                    int64_t* __restrict__ keys64,
+                                    int32_t* __restrict__ perm) {
+  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; i < n;
+       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    keys64[i] = static_cast<int64_t>(keys[i]);
+    perm[i] = static_cast<int32_t>(vals[i]);
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_2level(
+    const at::Tensor& linear, const at::Tensor& offsets, int64_t B, int64_t F,
+    int64_t end_bit, int64_t capacity) {
+  TORCH_CHECK(linear.is_cuda() && linear.scalar_type() == at::kLong);
+  TORCH_CHECK(end_bit <= 32, "segmented sort requires 32-bit ids");
+  int64_t n = linear.numel();
+  auto sorted64 = at::empty_like(linear);
+  auto perm = at::empty({n}, linear.options().dtype(at::kInt));
+  auto overflow = at::zeros({1}, linear.options().dtype(at::kInt));
+  if (n == 0 || F == 0) return {sorted64, perm, overflow};
+  auto stream = tbe_stream();
+  int tiles_per_seg = (int)((capacity + kTileSize - 1) / kTileSize);
+  auto opts32 = linear.options().dtype(at::kInt);
+  auto keysA = at::empty({n}, opts32);
+  auto valsA = at::empty({n}, opts32);
+  hipLaunchKernelGGL(seg_tile_sort_kernel, dim3((int)F * tiles_per_seg),
+                     dim3(kTileThreads), 0, stream, linear.data_ptr<int64_t>(),
+                     offsets.data_ptr<int64_t>(), (int)B, tiles_per_seg, (int)end_bit,
+                     reinterpret_cast<uint32_t*>(keysA.data_ptr<int32_t>()),
+                     reinterpret_cast<uint32_t*>(valsA.data_ptr<int32_t>()));
+  if (tiles_per_seg == 1) {
+    hipLaunchKernelGGL(seg_cast_out_kernel, dim3(grid_for(n, kBlockThreads)),
+                       dim3(kBlockThreads), 0, stream,
+                       reinterpret_cast<const uint32_t*>(keysA.data_ptr<int32_t>()),
+                       reinterpret_cast<const uint32_t*>(valsA.data_ptr<int32_t>()), n,
+                       sorted64.data_ptr<int64_t>(), perm.data_ptr<int32_t>());
+    return {sorted64, perm, overflow};
+  }
+  auto keysB = at::empty({n}, opts32);
+  auto valsB = at::empty({n}, opts32);
+  int chunks_per_seg = tiles_per_seg;
+  int run_len = kTileSize;
+  bool a_is_src = true;
+  while (run_len < tiles_per_seg * kTileSize) {
+    bool last = (run_len * 2) >= tiles_per_seg * kTileSize;
+    auto& src_k = a_is_src ? keysA : keysB;
+    auto& src_v = a_is_src ? valsA : valsB;
+    auto& dst_k = a_is_src ? keysB : keysA;
+    auto& dst_v = a_is_src ? valsB : valsA;
+    if (last) {
+      hipLaunchKernelGGL((seg_merge_kernel<true>), dim3((int)F * chunks_per_seg),
+                         dim3(kTileThreads), 0, stream,
+                         reinterpret_cast<const uint32_t*>(src_k.data_ptr<int32_t>()),
+                         reinterpret_cast<const uint32_t*>(src_v.data_ptr<int32_t>()),
+                         offsets.data_ptr<int64_t>(), (int)B, chunks_per_seg, run_len,
+                         nullptr, sorted64.data_ptr<int64_t>(), nullptr,
+                         perm.data_ptr<int32_t>());
+    } else {
+      hipLaunchKernelGGL((seg_merge_kernel<false>), dim3((int)F * chunks_per_seg),
+                         dim3(kTileThreads), 0, stream,
+                         reinterpret_cast<const uint32_t*>(src_k.data_ptr<int32_t>()),
+                         reinterpret_cast<const uint32_t*>(src_v.data_ptr<int32_t>()),
+                         offsets.data_ptr<int64_t>(), (int)B, chunks_per_seg, run_len,
+                         reinterpret_cast<uint32_t*>(dst_k.data_ptr<int32_t>()),
+                         nullptr,
+                         reinterpret_cast<uint32_t*>(dst_v.data_ptr<int32_t>()),
+                         nullptr);
+    }
+    run_len *= 2;
+    a_is_src = !a_is_src;
+  }
+  return {sorted64, perm, overflow};
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_large(
     const at::Tensor& linear, const at::Tensor& feat_bounds /* [F+1] */, int64_t F,
     int64_t end_bit) {

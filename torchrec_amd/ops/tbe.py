@@ -356,10 +356,6 @@ class TableBatchedEmbeddingBags(nn.Module):
                 for i in range(len(feature_table_map) - 1)
             )
             and self._total_rows < (1 << 31)
-            # TREC_SEG_SORT=0: force the hipCUB device radix path (in a
-            # hipGraph-captured step its ~20 launches are free and it uses
-            # the whole chip, vs one workgroup per feature segment here)
-            and os.environ.get("TREC_SEG_SORT", "1") == "1"
         )
         if self._uvm_caching:
             # cache sizing: cache_load_factor of total rows, 32 ways per set
@@ -521,11 +517,21 @@ class TableBatchedEmbeddingBags(nn.Module):
         # per-segment device flag costs a sync so it is debug-gated.
         if cap > 0 and linear.numel() > cap * self._num_features:
             cap = 0  # provably overflowing: fall through to the device sort
-        if self._seg_sort_ok and 0 < cap <= 16384:
-            sorted_lin, perm, overflow = torch.ops.trec_amd.seg_sort_pairs(
-                linear, offsets, B, self._num_features,
-                _bits_needed(self._total_rows), cap,
-            )
+        seg_mode = os.environ.get("TREC_SEG_SORT", "1")
+        if self._seg_sort_ok and cap > 0 and seg_mode != "0":
+            if seg_mode == "block" and cap <= 16384:
+                # legacy one-launch block sort (one workgroup per segment)
+                sorted_lin, perm, overflow = torch.ops.trec_amd.seg_sort_pairs(
+                    linear, offsets, B, self._num_features,
+                    _bits_needed(self._total_rows), cap,
+                )
+            else:
+                # two-level: 512-key stable tile sorts fill the chip, then
+                # log2(tiles) co-rank merge rounds; no per-segment size cap
+                sorted_lin, perm, overflow = torch.ops.trec_amd.seg_sort_pairs_2level(
+                    linear, offsets, B, self._num_features,
+                    _bits_needed(self._total_rows), cap,
+                )
             if os.environ.get("TREC_DEBUG") == "1" and bool(overflow.item()):
                 raise RuntimeError(
                     "TBE segmented sort overflow: a bag exceeded "
