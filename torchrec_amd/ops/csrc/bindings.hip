@@ -165,6 +165,9 @@ TORCH_LIBRARY(trec_amd, m) {
       "seg_sort_pairs(Tensor linear, Tensor offsets, int B, int F, int end_bit, "
       "int capacity) -> (Tensor, Tensor, Tensor)");
   m.def(
+      "seg_sort_pairs_2level(Tensor linear, Tensor offsets, int B, int F, int end_bit, "
+      "int capacity) -> (Tensor, Tensor, Tensor)");
+  m.def(
       "seg_sort_pairs_large(Tensor linear, Tensor feat_bounds, int F, int end_bit)"
       " -> (Tensor, Tensor, Tensor)");
   m.def("tbe_backward_prep(Tensor sorted_linear) -> (Tensor, Tensor)");
