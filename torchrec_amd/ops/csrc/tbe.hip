@@ -613,6 +613,12 @@ __global__ void __launch_bounds__(kTileThreads) seg_merge_kernel(
     const int64_t* __restrict__ offsets, int B, int chunks_per_seg, int run_len,
     uint32_t* __restrict__ keys_out32, int64_t* __restrict__ keys_out64,
     uint32_t* __restrict__ vals_out32, int32_t* __restrict__ vals_out32_final) {
+  // chunk boundaries are found with TWO global co-rank searches (thread 0),
+  // then the <=512-element A/B windows stage through LDS so the per-element
+  // co-ranks probe LDS, not HBM
+  __shared__ uint32_t sA[kTileSize + 1], sBk[kTileSize + 1];
+  __shared__ uint32_t sAv[kTileSize + 1], sBv[kTileSize + 1];
+  __shared__ int bounds[4];
   int seg = blockIdx.x / chunks_per_seg;
   int chunk = blockIdx.x - seg * chunks_per_seg;
   int64_t lo = offsets[static_cast<int64_t>(seg) * B];
@@ -620,8 +626,6 @@ __global__ void __launch_bounds__(kTileThreads) seg_merge_kernel(
   int seg_len = static_cast<int>(hi - lo);
   int out0 = chunk * kTileSize;
   if (out0 >= seg_len) return;
-  // the pair of runs this chunk's outputs come from (chunks never straddle
-  // a pair: kTileSize divides run_len)
   int pair0 = (out0 / (2 * run_len)) * (2 * run_len);
   int la = min(run_len, seg_len - pair0);
   int lb = max(0, min(run_len, seg_len - (pair0 + run_len)));
@@ -630,14 +634,31 @@ __global__ void __launch_bounds__(kTileThreads) seg_merge_kernel(
   const uint32_t* Av = vals_in + lo + pair0;
   const uint32_t* Bv = Av + run_len;
   int rel0 = out0 - pair0;
+  int rel1 = min(rel0 + kTileSize, la + lb);
+  if (threadIdx.x == 0) {
+    bounds[0] = co_rank(rel0, A, la, Bp, lb);
+    bounds[1] = co_rank(rel1, A, la, Bp, lb);
+  }
+  __syncthreads();
+  int a_lo = bounds[0], a_hi = bounds[1];
+  int b_lo = rel0 - a_lo, b_hi = rel1 - a_hi;
+  int wa = a_hi - a_lo, wb = b_hi - b_lo;
+  for (int i = threadIdx.x; i < wa; i += kTileThreads) {
+    sA[i] = A[a_lo + i];
+    sAv[i] = Av[a_lo + i];
+  }
+  for (int i = threadIdx.x; i < wb; i += kTileThreads) {
+    sBk[i] = Bp[b_lo + i];
+    sBv[i] = Bv[b_lo + i];
+  }
+  __syncthreads();
   for (int i = threadIdx.x; i < kTileSize; i += kTileThreads) {
-    int k = rel0 + i;
-    if (out0 + i >= seg_len || k >= la + lb) break;
-    int ai = co_rank(k, A, la, Bp, lb);
-    int bi = k - ai;
-    bool take_a = (bi >= lb) || (ai < la && A[ai] <= Bp[bi]);
-    uint32_t kv = take_a ? A[ai] : Bp[bi];
-    uint32_t vv = take_a ? Av[ai] : Bv[bi];
+    if (rel0 + i >= rel1) break;
+    int ai = co_rank(i, sA, wa, sBk, wb);
+    int bi = i - ai;
+    bool take_a = (bi >= wb) || (ai < wa && sA[ai] <= sBk[bi]);
+    uint32_t kv = take_a ? sA[ai] : sBk[bi];
+    uint32_t vv = take_a ? sAv[ai] : sBv[bi];
     int64_t out_idx = lo + out0 + i;
     if (OUT64) {
       keys_out64[out_idx] = static_cast<int64_t>(kv);
