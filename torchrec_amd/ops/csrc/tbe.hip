@@ -548,26 +548,28 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs(
 
 constexpr int kTileThreads = 256;
 constexpr int kTileItems = 2;
-constexpr int kTileSize = kTileThreads * kTileItems;  // 512 keys per tile
+constexpr int kTileSize = kTileThreads * kTileItems;  // merge-chunk width
 
+template <int ITEMS>
 __global__ void __launch_bounds__(kTileThreads) seg_tile_sort_kernel(
     const int64_t* __restrict__ linear, const int64_t* __restrict__ offsets, int B,
     int tiles_per_seg, int end_bit, uint32_t* __restrict__ keys_out,
     uint32_t* __restrict__ perm_out) {
-  using sorter = rocprim::block_radix_sort<uint32_t, kTileThreads, kTileItems, uint32_t>;
+  constexpr int TS = kTileThreads * ITEMS;
+  using sorter = rocprim::block_radix_sort<uint32_t, kTileThreads, ITEMS, uint32_t>;
   __shared__ typename sorter::storage_type storage;
   int seg = blockIdx.x / tiles_per_seg;
   int tile = blockIdx.x - seg * tiles_per_seg;
   int64_t lo = offsets[static_cast<int64_t>(seg) * B];
   int64_t hi = offsets[static_cast<int64_t>(seg + 1) * B];
-  int64_t t0 = lo + static_cast<int64_t>(tile) * kTileSize;
+  int64_t t0 = lo + static_cast<int64_t>(tile) * TS;
   if (t0 >= hi) return;
-  int count = static_cast<int>(min(hi - t0, (int64_t)kTileSize));
-  uint32_t keys[kTileItems];
-  uint32_t vals[kTileItems];
+  int count = static_cast<int>(min(hi - t0, (int64_t)TS));
+  uint32_t keys[ITEMS];
+  uint32_t vals[ITEMS];
 #pragma unroll
-  for (int i = 0; i < kTileItems; ++i) {
-    int k = static_cast<int>(threadIdx.x) * kTileItems + i;  // blocked
+  for (int i = 0; i < ITEMS; ++i) {
+    int k = static_cast<int>(threadIdx.x) * ITEMS + i;  // blocked
     if (k < count) {
       keys[i] = static_cast<uint32_t>(linear[t0 + k]);
       vals[i] = static_cast<uint32_t>(t0 + k);
@@ -578,8 +580,8 @@ __global__ void __launch_bounds__(kTileThreads) seg_tile_sort_kernel(
   }
   sorter().sort(keys, vals, storage, 0u, static_cast<unsigned>(end_bit));
 #pragma unroll
-  for (int i = 0; i < kTileItems; ++i) {
-    int k = static_cast<int>(threadIdx.x) * kTileItems + i;
+  for (int i = 0; i < ITEMS; ++i) {
+    int k = static_cast<int>(threadIdx.x) * ITEMS + i;
     if (k < count) {
       keys_out[t0 + k] = keys[i];
       perm_out[t0 + k] = vals[i];
@@ -692,15 +694,28 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_2level(
   auto overflow = at::zeros({1}, linear.options().dtype(at::kInt));
   if (n == 0 || F == 0) return {sorted64, perm, overflow};
   auto stream = tbe_stream();
-  int tiles_per_seg = (int)((capacity + kTileSize - 1) / kTileSize);
+  // base-tile size: bigger tiles = fewer merge rounds (each merge kernel has
+  // a ~9 us latency floor), smaller tiles = more tile-sort parallelism.
+  // 2048-key tiles balance at the bench shapes.
+  int items = capacity >= 4096 ? 8 : 2;
+  int base_tile = kTileThreads * items;
+  int tiles_per_seg = (int)((capacity + base_tile - 1) / base_tile);
   auto opts32 = linear.options().dtype(at::kInt);
   auto keysA = at::empty({n}, opts32);
   auto valsA = at::empty({n}, opts32);
-  hipLaunchKernelGGL(seg_tile_sort_kernel, dim3((int)F * tiles_per_seg),
-                     dim3(kTileThreads), 0, stream, linear.data_ptr<int64_t>(),
-                     offsets.data_ptr<int64_t>(), (int)B, tiles_per_seg, (int)end_bit,
-                     reinterpret_cast<uint32_t*>(keysA.data_ptr<int32_t>()),
-                     reinterpret_cast<uint32_t*>(valsA.data_ptr<int32_t>()));
+  if (items == 8) {
+    hipLaunchKernelGGL((seg_tile_sort_kernel<8>), dim3((int)F * tiles_per_seg),
+                       dim3(kTileThreads), 0, stream, linear.data_ptr<int64_t>(),
+                       offsets.data_ptr<int64_t>(), (int)B, tiles_per_seg, (int)end_bit,
+                       reinterpret_cast<uint32_t*>(keysA.data_ptr<int32_t>()),
+                       reinterpret_cast<uint32_t*>(valsA.data_ptr<int32_t>()));
+  } else {
+    hipLaunchKernelGGL((seg_tile_sort_kernel<2>), dim3((int)F * tiles_per_seg),
+                       dim3(kTileThreads), 0, stream, linear.data_ptr<int64_t>(),
+                       offsets.data_ptr<int64_t>(), (int)B, tiles_per_seg, (int)end_bit,
+                       reinterpret_cast<uint32_t*>(keysA.data_ptr<int32_t>()),
+                       reinterpret_cast<uint32_t*>(valsA.data_ptr<int32_t>()));
+  }
   if (tiles_per_seg == 1) {
     hipLaunchKernelGGL(seg_cast_out_kernel, dim3(grid_for(n, kBlockThreads)),
                        dim3(kBlockThreads), 0, stream,
@@ -711,11 +726,12 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_2level(
   }
   auto keysB = at::empty({n}, opts32);
   auto valsB = at::empty({n}, opts32);
-  int chunks_per_seg = tiles_per_seg;
-  int run_len = kTileSize;
+  // merge output chunks are kTileSize wide regardless of the base tile
+  int chunks_per_seg = (int)((capacity + kTileSize - 1) / kTileSize);
+  int run_len = base_tile;
   bool a_is_src = true;
-  while (run_len < tiles_per_seg * kTileSize) {
-    bool last = (run_len * 2) >= tiles_per_seg * kTileSize;
+  while (run_len < tiles_per_seg * base_tile) {
+    bool last = (run_len * 2) >= tiles_per_seg * base_tile;
     auto& src_k = a_is_src ? keysA : keysB;
     auto& src_v = a_is_src ? valsA : valsB;
     auto& dst_k = a_is_src ? keysB : keysA;
