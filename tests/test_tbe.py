@@ -357,6 +357,46 @@ class TestTBEOutputDtype:
         for wa, wb in zip(tbe.split_embedding_weights(), ref.split_embedding_weights()):
             assert torch.allclose(wa, wb, atol=0.02, rtol=0.05)
 
+    def test_cpu_seq_bf16_output(self):
+        torch.manual_seed(0)
+        specs = [("t0", 30, 8), ("t1", 40, 8)]
+        tbe = TableBatchedEmbeddings(specs, learning_rate=0.05, output_dtype="bf16")
+        ref = TableBatchedEmbeddings(specs, learning_rate=0.05)
+        ref.weights.data.copy_(tbe.weights.data)
+        indices = torch.tensor([1, 2, 3, 4, 4])
+        offsets = torch.tensor([0, 2, 3, 4, 5])  # B=2, F=2
+        out = tbe(indices, offsets)
+        assert out.dtype == torch.bfloat16
+        out32 = ref(indices, offsets)
+        assert torch.allclose(out.float(), out32, atol=0.02, rtol=0.02)
+        g = torch.randn_like(out32)
+        out.backward(g.to(torch.bfloat16))
+        out32.backward(g)
+        for wa, wb in zip(tbe.split_embedding_weights(), ref.split_embedding_weights()):
+            assert torch.allclose(wa, wb, atol=0.02, rtol=0.05)
+
+    @pytest.mark.gpu
+    def test_gpu_seq_bf16_output_matches_cpu(self):
+        torch.manual_seed(0)
+        specs = [("t0", 100, 64), ("t1", 50, 64)]
+        cpu = TableBatchedEmbeddings(specs, learning_rate=0.05, output_dtype="bf16")
+        gpu = TableBatchedEmbeddings(
+            specs, learning_rate=0.05, output_dtype="bf16", device=torch.device("cuda")
+        )
+        gpu.weights.data.copy_(cpu.weights.data)
+        for step in range(3):
+            indices, offsets = make_inputs(specs, B=16, L=5, seed=step)
+            out_c = cpu(indices, offsets)
+            out_g = gpu(indices.cuda(), offsets.cuda())
+            assert out_g.dtype == torch.bfloat16
+            assert torch.allclose(out_g.float().cpu(), out_c.float(), atol=0.02, rtol=0.02)
+            grad = torch.randn_like(out_c, dtype=torch.float32).to(torch.bfloat16)
+            out_c.backward(grad)
+            out_g.backward(grad.cuda())
+        torch.cuda.synchronize()
+        for wc, wg in zip(cpu.split_embedding_weights(), gpu.split_embedding_weights()):
+            assert torch.allclose(wg.cpu(), wc, atol=0.01, rtol=0.02)
+
     @pytest.mark.gpu
     def test_gpu_bf16_output_matches_cpu(self):
         torch.manual_seed(0)

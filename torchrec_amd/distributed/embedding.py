@@ -347,6 +347,7 @@ class ShardedEmbeddingCollection(nn.Module):
                 tables[0].data_type if tables else "FP32"
             ],
             use_index_dedup=self._use_index_dedup,
+            output_dtype=self._fused_params.get("output_dtype", "fp32"),
         )
         if not tables:
             m._dim = D  # featureless rank still answers [0, D] for the a2a

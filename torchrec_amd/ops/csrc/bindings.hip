@@ -35,7 +35,7 @@ at::Tensor tbe_forward_pooled(const at::Tensor& weights, const at::Tensor& table
 at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_elem_offsets,
                            const at::Tensor& dims, const at::Tensor& feat_table,
                            const at::Tensor& feat_val_offsets, const at::Tensor& indices,
-                           int64_t D_out, int64_t max_D);
+                           int64_t D_out, int64_t max_D, int64_t out_dtype);
 at::Tensor tbe_forward_pooled_vbe(const at::Tensor& weights,
                                   const at::Tensor& table_elem_offsets, const at::Tensor& dims,
                                   const at::Tensor& feat_table, const at::Tensor& bag_offsets,
@@ -153,8 +153,8 @@ TORCH_LIBRARY(trec_amd, m) {
       " Tensor cache_weights, Tensor cache_loc, int out_dtype) -> Tensor");
   m.def(
       "tbe_forward_seq(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
-      " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D)"
-      " -> Tensor");
+      " Tensor feat_table, Tensor feat_val_offsets, Tensor indices, int D_out, int max_D,"
+      " int out_dtype) -> Tensor");
   m.def(
       "tbe_forward_pooled_vbe(Tensor weights, Tensor table_elem_offsets, Tensor dims,"
       " Tensor feat_table, Tensor bag_offsets, Tensor out_offsets, Tensor indices,"

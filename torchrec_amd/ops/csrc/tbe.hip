@@ -360,13 +360,13 @@ at::Tensor tbe_forward_pooled_vbe(
 // forward (sequence / non-pooled): out[n, :] = W[table(f(n))][idx[n]]
 // ---------------------------------------------------------------------------
 
-template <typename emb_t, int LPS, int CHUNKS>
+template <typename emb_t, typename o_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_seq_kernel(
     const emb_t* __restrict__ weights, const int64_t* __restrict__ table_elem_offsets,
     const int32_t* __restrict__ dims, const int32_t* __restrict__ feat_table,
     const int64_t* __restrict__ feat_val_offsets,  // [F+1] value range per feature
     const int64_t* __restrict__ indices, int F, int64_t N, int64_t D_out,
-    float* __restrict__ out /* [N, D_out] */) {
+    o_t* __restrict__ out /* [N, D_out] */) {
   int sl = threadIdx.x % LPS;
   int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
   int64_t n_slots = (static_cast<int64_t>(gridDim.x) * blockDim.x) / LPS;
@@ -375,52 +375,77 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_fwd_seq_kernel(
     int t = feat_table[f];
     int D = dims[t];
     const emb_t* row = weights + table_elem_offsets[t] + indices[n] * D;
-    float4* orow = reinterpret_cast<float4*>(out + n * D_out);
+    o_t* orow = out + n * D_out;
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
       int col4 = c * LPS + sl;
-      if (col4 * 4 < D) orow[col4] = Vec4<emb_t>::load(row, col4);
+      if (col4 * 4 < D) Vec4<o_t>::store(orow, col4, Vec4<emb_t>::load(row, col4));
     }
   }
+}
+
+template <typename emb_t, typename o_t, typename host_o_t>
+static void launch_tbe_fwd_seq(const emb_t* wp,
+                               const at::Tensor& table_elem_offsets, const at::Tensor& dims,
+                               const at::Tensor& feat_table,
+                               const at::Tensor& feat_val_offsets,
+                               const at::Tensor& indices, int F, int64_t N, int64_t D_out,
+                               at::Tensor& out, int lps, int chunks, int grid,
+                               hipStream_t stream) {
+  o_t* out_ptr = reinterpret_cast<o_t*>(out.data_ptr<host_o_t>());
+#define TBE_SEQ_LAUNCH(LPS, CHUNKS)                                                        \
+  hipLaunchKernelGGL((tbe_fwd_seq_kernel<emb_t, o_t, LPS, CHUNKS>), dim3(grid),            \
+                     dim3(kBlockThreads), 0, stream, wp,                                   \
+                     table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),     \
+                     feat_table.data_ptr<int32_t>(), feat_val_offsets.data_ptr<int64_t>(), \
+                     indices.data_ptr<int64_t>(), F, N, D_out, out_ptr)
+  if (lps == 16) TBE_SEQ_LAUNCH(16, 1);
+  else if (lps == 32) TBE_SEQ_LAUNCH(32, 1);
+  else switch (chunks) {
+    case 1: TBE_SEQ_LAUNCH(64, 1); break;
+    case 2: TBE_SEQ_LAUNCH(64, 2); break;
+    case 3: case 4: TBE_SEQ_LAUNCH(64, 4); break;
+    default: TBE_SEQ_LAUNCH(64, 8); break;
+  }
+#undef TBE_SEQ_LAUNCH
 }
 
 at::Tensor tbe_forward_seq(const at::Tensor& weights, const at::Tensor& table_elem_offsets,
                            const at::Tensor& dims, const at::Tensor& feat_table,
                            const at::Tensor& feat_val_offsets, const at::Tensor& indices,
-                           int64_t D_out, int64_t max_D) {
+                           int64_t D_out, int64_t max_D, int64_t out_dtype) {
   TORCH_CHECK(max_D % 4 == 0 && max_D <= 2048);
   int64_t N = indices.numel();
-  auto out = at::empty({N, D_out}, indices.options().dtype(at::kFloat));
+  auto out_st = out_dtype == 1 ? at::kBFloat16 : (out_dtype == 2 ? at::kHalf : at::kFloat);
+  auto out = at::empty({N, D_out}, indices.options().dtype(out_st));
   if (N == 0) return out;
   int F = feat_table.numel();
   auto stream = tbe_stream();
   int lps = (max_D <= 64) ? 16 : (max_D <= 128 ? 32 : 64);
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
   int grid = grid_for(N * lps, kBlockThreads);
-#define TBE_SEQ_LAUNCH(LPS, CHUNKS)                                                        \
-  hipLaunchKernelGGL((tbe_fwd_seq_kernel<dev_t, LPS, CHUNKS>), dim3(grid),                  \
-                     dim3(kBlockThreads), 0,                                                \
-                     stream, reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights)),    \
-                     table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(),     \
-                     feat_table.data_ptr<int32_t>(), feat_val_offsets.data_ptr<int64_t>(), \
-                     indices.data_ptr<int64_t>(), F, N, D_out, out.data_ptr<float>())
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
                                   "tbe_fwd_seq", [&] {
     using dev_t = typename DevType<scalar_t>::type;
     if constexpr (std::is_same_v<scalar_t, double>) {
       TORCH_CHECK(false, "fp64 embedding tables unsupported");
     } else {
-    if (lps == 16) TBE_SEQ_LAUNCH(16, 1);
-    else if (lps == 32) TBE_SEQ_LAUNCH(32, 1);
-    else switch (chunks) {
-      case 1: TBE_SEQ_LAUNCH(64, 1); break;
-      case 2: TBE_SEQ_LAUNCH(64, 2); break;
-      case 3: case 4: TBE_SEQ_LAUNCH(64, 4); break;
-      default: TBE_SEQ_LAUNCH(64, 8); break;
-    }
+      const dev_t* wp = reinterpret_cast<const dev_t*>(uvm_ptr<scalar_t>(weights));
+      if (out_dtype == 1) {
+        launch_tbe_fwd_seq<dev_t, __hip_bfloat16, at::BFloat16>(
+            wp, table_elem_offsets, dims, feat_table, feat_val_offsets, indices, F, N,
+            D_out, out, lps, chunks, grid, stream);
+      } else if (out_dtype == 2) {
+        launch_tbe_fwd_seq<dev_t, __half, at::Half>(
+            wp, table_elem_offsets, dims, feat_table, feat_val_offsets, indices, F, N,
+            D_out, out, lps, chunks, grid, stream);
+      } else {
+        launch_tbe_fwd_seq<dev_t, float, float>(
+            wp, table_elem_offsets, dims, feat_table, feat_val_offsets, indices, F, N,
+            D_out, out, lps, chunks, grid, stream);
+      }
     }
   });
-#undef TBE_SEQ_LAUNCH
   return out;
 }
 

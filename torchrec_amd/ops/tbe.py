@@ -166,6 +166,7 @@ class _TBESeqFunction(torch.autograd.Function):
             indices,
             host._max_D,
             host._max_D,
+            host._out_dtype_code,
         )
         ctx.host = host
         ctx.save_for_backward(indices, feat_val_offsets)
@@ -1017,6 +1018,7 @@ class TableBatchedEmbeddings(nn.Module):
         init_max: float = 0.01,
         weights_precision: str = "fp32",
         use_index_dedup: bool = False,
+        output_dtype: str = "fp32",
     ) -> None:
         super().__init__()
         self._use_index_dedup = use_index_dedup
@@ -1033,6 +1035,7 @@ class TableBatchedEmbeddings(nn.Module):
             init_min=init_min,
             init_max=init_max,
             weights_precision=weights_precision,
+            output_dtype=output_dtype,
         )
         self._dim = next(iter(dims)) if dims else 0
 
@@ -1063,7 +1066,7 @@ class TableBatchedEmbeddings(nn.Module):
             # featureless rank of a sharding: zero rows, but keep the output
             # in the autograd graph so backward collectives still fire
             return torch.zeros(
-                0, self._dim, device=indices.device,
+                0, self._dim, device=indices.device, dtype=host._out_torch_dtype,
                 requires_grad=torch.is_grad_enabled(),
             )
         B = (offsets.numel() - 1) // host._num_features
@@ -1114,6 +1117,7 @@ class _TBESeqDedupFunction(torch.autograd.Function):
             local,
             host._max_D,
             host._max_D,
+            host._out_dtype_code,
         )
         out = rows_u.index_select(0, inv)
         ctx.host = host
@@ -1199,7 +1203,7 @@ class _TBESeqCpuFunction(torch.autograd.Function):
                 w = host.weights[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
                 idx = indices[int(offsets[f * B]) : int(offsets[(f + 1) * B])]
                 outs.append(w[idx].float())
-            return torch.cat(outs, dim=0)
+            return torch.cat(outs, dim=0).to(host._out_torch_dtype)
 
     @staticmethod
     def backward(ctx, grad):  # type: ignore[override]
@@ -1215,6 +1219,6 @@ class _TBESeqCpuFunction(torch.autograd.Function):
             e0 = int(host._table_elem_offsets[t])
             gw = grad_flat[e0 : e0 + s.rows * s.dim].view(s.rows, s.dim)
             lo, hi = int(offsets[f * B]), int(offsets[(f + 1) * B])
-            gw.index_add_(0, indices[lo:hi], grad[lo:hi])
+            gw.index_add_(0, indices[lo:hi], grad[lo:hi].float())
         host._cpu_apply_update(grad_flat)
         return None, None, None, None, None
