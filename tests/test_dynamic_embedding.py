@@ -199,6 +199,59 @@ class TestPsBridge:
             kv.split_embedding_weights()[0][slot], w11, atol=1e-6, rtol=1e-6
         )
 
+    def test_tcp_network_transport(self):
+        """Network PS (reference redis_io.cpp analogue): push/pull over a live
+        TCP row-store server; pull pops and tables are namespaced."""
+        import numpy as np
+
+        from torchrec_amd.dynamic_embedding.ps import ParameterServer
+        from torchrec_amd.dynamic_embedding.ps_net import PSNetServer
+
+        srv = PSNetServer()
+        try:
+            ps = ParameterServer([4, 8], io="tcp", address=srv.address)
+            rows = torch.arange(8.0).reshape(2, 4)
+            ps.evict(0, torch.tensor([7, 9]), rows, torch.tensor([0.5, 0.25]))
+            assert len(srv) == 2
+            row, st = ps.fetch(0, 9)
+            assert np.allclose(row, [4, 5, 6, 7]) and st == 0.25
+            assert ps.fetch(0, 9) is None  # pull pops server-side
+            assert ps.fetch(1, 7) is None  # per-table namespaces
+            # table 1 has dim 8
+            ps.evict(1, torch.tensor([7]), torch.ones(1, 8) * 3.0, torch.tensor([1.5]))
+            row, st = ps.fetch(1, 7)
+            assert np.allclose(row, np.full(8, 3.0)) and st == 1.5
+            ps.close()
+        finally:
+            srv.close()
+
+    def test_ssd_tier_tcp_transport(self):
+        """SSD tier spill/restore round-trips through the network PS."""
+        from torchrec_amd.dynamic_embedding.ps_net import PSNetServer
+        from torchrec_amd.ops.kv_embedding import SsdEmbeddingBags
+
+        srv = PSNetServer()
+        try:
+            torch.manual_seed(0)
+            kv = SsdEmbeddingBags(
+                [("t0", 10**9, 4)], capacity=4, io="tcp",
+                io_kwargs={"address": srv.address},
+            )
+            kv(torch.tensor([11, 22, 33, 44]), torch.tensor([0, 1, 2, 3, 4])).sum().backward()
+            w11 = kv.split_embedding_weights()[0][
+                int(kv._transformers[0].transform(torch.tensor([11]))[0])
+            ].clone()
+            kv(torch.tensor([55, 66, 77, 88]), torch.tensor([0, 1, 2, 3, 4]))
+            kv(torch.tensor([91, 92, 93, 94]), torch.tensor([0, 1, 2, 3, 4]))
+            kv(torch.tensor([11]), torch.tensor([0, 1]))
+            slot = int(kv._transformers[0].transform(torch.tensor([11]))[0])
+            torch.testing.assert_close(
+                kv.split_embedding_weights()[0][slot], w11, atol=1e-6, rtol=1e-6
+            )
+            kv.close()
+        finally:
+            srv.close()
+
 
 class TestVirtualTableEvictionPolicies:
     """Policy-driven eviction on the KV (virtual) tables (reference

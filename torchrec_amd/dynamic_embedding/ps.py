@@ -35,7 +35,7 @@ class PSIO(abc.ABC):
 
 
 class MemoryPSIO(PSIO):
-    def __init__(self, dim: int) -> None:
+    def __init__(self, dim: int, table: int = 0) -> None:
         self._store: Dict[int, tuple] = {}
 
     def push(self, ids, rows, state) -> None:
@@ -49,9 +49,10 @@ class MemoryPSIO(PSIO):
 class FilePSIO(PSIO):
     """Append-only on-disk row log + in-memory offset index (the SSD tier)."""
 
-    def __init__(self, dim: int, path: Optional[str] = None) -> None:
+    def __init__(self, dim: int, path: Optional[str] = None, table: int = 0) -> None:
         self._dim = dim
-        self._path = path or os.path.join(
+        # a shared explicit path is suffixed per table so logs don't clobber
+        self._path = (f"{path}.t{table}" if path else None) or os.path.join(
             tempfile.mkdtemp(prefix="trec_amd_ps_"), "rows.log"
         )
         os.makedirs(os.path.dirname(self._path), exist_ok=True)
@@ -92,6 +93,8 @@ def register_ps_io(name: str, factory: Callable[..., PSIO]) -> None:
 
 
 def get_ps_io(name: str, dim: int, **kwargs) -> PSIO:
+    if name == "tcp" and name not in _IO_REGISTRY:
+        import torchrec_amd.dynamic_embedding.ps_net  # noqa: F401  registers "tcp"
     return _IO_REGISTRY[name](dim, **kwargs)
 
 
@@ -100,7 +103,7 @@ class ParameterServer:
     fetch on admission, evict on displacement)."""
 
     def __init__(self, dims: List[int], io: str = "memory", **io_kwargs) -> None:
-        self._ios = [get_ps_io(io, d, **io_kwargs) for d in dims]
+        self._ios = [get_ps_io(io, d, table=i, **io_kwargs) for i, d in enumerate(dims)]
 
     def evict(
         self, table: int, ids: torch.Tensor, rows: torch.Tensor, state: torch.Tensor

@@ -15,6 +15,7 @@ momentum zero) so reused slots never leak old state.
 
 from __future__ import annotations
 
+import os
 from typing import List, Optional, Tuple
 
 import torch
@@ -206,16 +207,17 @@ class SsdEmbeddingBags(KeyValueEmbeddingBags):
         *args,
         storage_dir: Optional[str] = None,
         io: str = "file",
+        io_kwargs: Optional[dict] = None,
         **kwargs,
     ) -> None:
         super().__init__(*args, **kwargs)
         from torchrec_amd.dynamic_embedding.ps import ParameterServer
 
-        io_kwargs = {}
+        io_kwargs = dict(io_kwargs or {})
         if io == "file" and storage_dir is not None:
-            io_kwargs["path"] = None  # FilePSIO manages per-table paths below
+            io_kwargs["path"] = os.path.join(storage_dir, "rows.log")
         self._ps = ParameterServer(
-            [s.dim for s in self._tbe.embedding_specs], io=io
+            [s.dim for s in self._tbe.embedding_specs], io=io, **io_kwargs
         )
 
     def _spill(self, table: int, slots: torch.Tensor, ids: torch.Tensor) -> None:
