@@ -217,14 +217,37 @@ class _VbeEmbeddingBagCollectionAwaitable(LazyAwaitable[KeyedTensor]):
         self._strides = local_strides
 
     def _wait_impl(self) -> KeyedTensor:
-        by_name: Dict[str, torch.Tensor] = dict(zip(self._local_names, self._local_parts))
+        by_name: Dict[str, Any] = dict(zip(self._local_names, self._local_parts))
         for entry in self._awaitables:
-            aw, names = entry if isinstance(entry, tuple) else (entry, self._owner_major_names)
+            widths = cols = None
+            if isinstance(entry, tuple) and len(entry) == 4:
+                aw, names, widths, cols = entry
+            elif isinstance(entry, tuple):
+                aw, names = entry
+            else:
+                aw, names = entry, self._owner_major_names
+            if widths is None:
+                widths = [self._dims[f] for f in names]
             flat = aw.wait().view(-1)
-            sizes = [self._strides[f] * self._dims[f] for f in names]
-            for f, part in zip(names, flat.split(sizes)):
-                by_name[f] = part
-        values = torch.cat([by_name[f] for f in self._embedding_names])
+            sizes = [self._strides[f] * w for f, w in zip(names, widths)]
+            for i, (f, part) in enumerate(zip(names, flat.split(sizes))):
+                if cols is not None and widths[i] != self._dims[f]:
+                    # GRID column slice: collect parts, paste at assembly
+                    slot = by_name.setdefault(f, [])
+                    slot.append((cols[i], widths[i], part))
+                else:
+                    by_name[f] = part
+        parts: List[torch.Tensor] = []
+        for f in self._embedding_names:
+            v = by_name[f]
+            if isinstance(v, list):
+                B_f = self._strides[f]
+                v = torch.cat(
+                    [p.view(B_f, w) for _, w, p in sorted(v, key=lambda x: x[0])],
+                    dim=1,
+                ).reshape(-1)
+            parts.append(v)
+        values = torch.cat(parts)
         return KeyedTensor(
             keys=self._embedding_names,
             values=values,
@@ -458,6 +481,14 @@ class ShardedEmbeddingBagCollection(nn.Module):
                 )
                 entries.append((aw, names_mine))
                 continue
+            if st in (
+                ShardingType.TABLE_ROW_WISE.value,
+                ShardingType.GRID_SHARD.value,
+            ):
+                entries.append(
+                    self._vbe_twrw_entry(ctx, sharding, kjt, packed, dims_by_name)
+                )
+                continue
             # TW: reorder (f, r) element blocks to rank-major and a2a back
             spr = kjt.stride_per_key_per_rank()  # [F_mine][W]
             names_mine = kjt.keys()
@@ -501,6 +532,94 @@ class ShardedEmbeddingBagCollection(nn.Module):
             local_strides=ctx.vbe_local_strides or {},
         )
 
+    def _vbe_twrw_entry(self, ctx, sharding, kjt, packed, dims_by_name):
+        """VBE output for the two-level shardings (TWRW, GRID).
+
+        Stage 1 rides xGMI: per-feature uneven reduce-scatter over the
+        intra-node group collapses the L partial pools; the stagger recat of
+        the input a2a already grouped each feature's bags by destination
+        local rank, so the RS-v blocks are contiguous. Stage 2 (NN > 1) is a
+        variable cross-node a2a of the reduced bags back to their source
+        nodes (reference twrw_sharding.py:460 two-stage output, VBE form).
+        GRID entries carry (widths, col offsets) so the assembler can paste
+        column slices (reference grid_sharding.py:558).
+        """
+        from torchrec_amd.distributed.comm import intra_and_cross_node_pg
+        from torchrec_amd.distributed.comm_ops import (
+            reduce_scatter_v_per_feature_pooled,
+        )
+        from torchrec_amd.distributed.dist_data import SequenceEmbeddingsAllToAll
+
+        L = sharding._L
+        NN = sharding._NN
+        my_node = self._env.rank // L
+        my_local = self._env.rank % L
+        # node-major names + this sharding's per-feature widths/column offsets
+        names_per_node: List[List[str]] = []
+        widths_per_node: List[List[int]] = []
+        cols_per_node: List[List[int]] = []
+        for grouped in sharding._grouped_per_node:
+            ns: List[str] = []
+            ws: List[int] = []
+            cs: List[int] = []
+            for g in grouped:
+                for t in g:
+                    for f in t.feature_names:
+                        ns.append(f)
+                        ws.append(t.local_dim)
+                        cs.append(getattr(t, "col_offset", 0) or 0)
+            names_per_node.append(ns)
+            widths_per_node.append(ws)
+            cols_per_node.append(cs)
+        names_mine = list(kjt.keys())
+        width_mine = dict(zip(names_per_node[my_node], widths_per_node[my_node]))
+        dims = [width_mine[f] for f in names_mine]
+        F = len(names_mine)
+        # recv spr is in stagger order: position l'*NN + n' = source rank n'*L+l'
+        spr = [list(sp) for sp in kjt.stride_per_key_per_rank()]
+        spr_intra = [
+            [sum(spr[f][l * NN : (l + 1) * NN]) for l in range(L)] for f in range(F)
+        ]
+        intra_pg, cross_pg = intra_and_cross_node_pg()
+        aw1 = reduce_scatter_v_per_feature_pooled(packed, spr_intra, dims, intra_pg)
+        if NN == 1:
+            return (aw1, names_mine, dims, [0] * F)
+        # stage 2: regroup the reduced feature-major pack node-major, a2a back
+        strides = ctx.vbe_local_strides or {}
+        sizes_fmaj = [
+            spr[f][my_local * NN + n] * dims[f] for f in range(F) for n in range(NN)
+        ]
+        in_splits = [
+            sum(spr[f][my_local * NN + n] * dims[f] for f in range(F))
+            for n in range(NN)
+        ]
+        out_splits = [
+            sum(
+                strides[f] * w
+                for f, w in zip(names_per_node[n], widths_per_node[n])
+            )
+            for n in range(NN)
+        ]
+        names_after = [f for n in range(NN) for f in names_per_node[n]]
+        widths_after = [w for n in range(NN) for w in widths_per_node[n]]
+        cols_after = [c for n in range(NN) for c in cols_per_node[n]]
+
+        class _TwoStageVbe:
+            def wait(self) -> torch.Tensor:
+                reduced = aw1.wait().view(-1)
+                if sizes_fmaj:
+                    blocks = list(reduced.split(sizes_fmaj))
+                    send = torch.cat(
+                        [blocks[f * NN + n] for n in range(NN) for f in range(F)]
+                    )
+                else:
+                    send = reduced
+                return SequenceEmbeddingsAllToAll(cross_pg)(
+                    send.view(-1, 1), in_splits, out_splits
+                ).wait()
+
+        return (_TwoStageVbe(), names_after, widths_after, cols_after)
+
     def _make_output_awaitable(self, ctx, awaitables) -> EmbeddingBagCollectionAwaitable:
         # resolve per-feature mean divisors into column ranges lazily
         mean_cols: List[Any] = []
@@ -535,11 +654,13 @@ class ShardedEmbeddingBagCollection(nn.Module):
                 ShardingType.TABLE_WISE.value,
                 ShardingType.ROW_WISE.value,
                 ShardingType.DATA_PARALLEL.value,
+                ShardingType.TABLE_ROW_WISE.value,
+                ShardingType.GRID_SHARD.value,
             }
             assert (
                 all(t in allowed for t in self._sharding_types)
                 or self._env.world_size == 1
-            ), "VBE through the sharded path supports TW/RW/DP shardings"
+            ), "VBE through the sharded path supports TW/RW/DP/TWRW/GRID shardings"
         dist_input = self.input_dist(ctx, features).wait().wait()
         return self.compute_and_output_dist(ctx, dist_input)
 
