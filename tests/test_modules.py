@@ -166,6 +166,31 @@ class TestDLRM:
         logits = model(batch.dense_features, batch.sparse_features)
         assert logits.shape == (4, 1)
 
+    def test_dlrm_projection(self):
+        from torchrec_amd.models.dlrm import DLRM_Projection
+
+        ebc = EmbeddingBagCollection(tables=self._tables())
+        model = DLRM_Projection(
+            embedding_bag_collection=ebc,
+            dense_in_features=13,
+            dense_arch_layer_sizes=[16, 8],
+            over_arch_layer_sizes=[16, 1],
+            interaction_branch1_layer_sizes=[24, 16],  # I1 = 16/8 = 2
+            interaction_branch2_layer_sizes=[24, 24],  # I2 = 24/8 = 3
+        )
+        batch = generate_batch(
+            keys=["f0", "f1", "f2"], batch_size=4, hash_sizes=[100] * 3, ids_per_feature=5
+        )
+        logits = model(batch.dense_features, batch.sparse_features)
+        assert logits.shape == (4, 1)
+        # over-arch input = D + I1*I2 = 8 + 6
+        assert model.over_arch.model[0]._mlp[0]._linear.in_features == 14
+        logits.sum().backward()
+        assert all(
+            p.grad is not None
+            for p in model.inter_arch.parameters()
+        )
+
 
 class TestFeatureProcessors:
     def test_position_weighted_module(self):

@@ -3,10 +3,12 @@
 from torchrec_amd.models.dlrm import (  # noqa: F401
     DLRM,
     DLRM_DCN,
+    DLRM_Projection,
     DLRMTrain,
     DenseArch,
     InteractionArch,
     InteractionDCNArch,
+    InteractionProjectionArch,
     OverArch,
     SparseArch,
 )

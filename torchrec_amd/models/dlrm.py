@@ -144,6 +144,43 @@ class InteractionDCNArch(nn.Module):
         return self.crossnet(combined)
 
 
+class InteractionProjectionArch(nn.Module):
+    """Projected interaction (reference: models/dlrm.py:293).
+
+    Two MLP branches project the combined [B, (F+1)*D] features to I1*D and
+    I2*D; their [B, I1, D] x [B, D, I2] product gives I1*I2 learned
+    interaction terms, concatenated after the dense features.
+    Output: [B, D + I1*I2].
+    """
+
+    def __init__(
+        self,
+        num_sparse_features: int,
+        interaction_branch1: nn.Module,
+        interaction_branch2: nn.Module,
+        embedding_dim: int,
+    ) -> None:
+        super().__init__()
+        self.F = num_sparse_features
+        self.branch1 = interaction_branch1
+        self.branch2 = interaction_branch2
+        self.D = embedding_dim
+
+    def forward(
+        self, dense_features: torch.Tensor, sparse_features: torch.Tensor
+    ) -> torch.Tensor:
+        if self.F <= 0:
+            return dense_features
+        B = dense_features.shape[0]
+        combined = torch.cat(
+            [dense_features.unsqueeze(1), sparse_features], dim=1
+        ).reshape(B, -1)
+        a = self.branch1(combined).reshape(B, -1, self.D)  # [B, I1, D]
+        b = self.branch2(combined).reshape(B, self.D, -1)  # [B, D, I2]
+        inter = torch.bmm(a, b).reshape(B, -1)  # [B, I1*I2]
+        return torch.cat([dense_features, inter], dim=1)
+
+
 class OverArch(nn.Module):
     """Final MLP -> logit (reference: models/dlrm.py:394)."""
 
@@ -215,6 +252,57 @@ class DLRM_DCN(nn.Module):
         self.over_arch = OverArch(cross_in, over_arch_layer_sizes, device=dense_device)
 
     def forward(self, dense_features: torch.Tensor, sparse_features: KeyedJaggedTensor) -> torch.Tensor:
+        embedded_dense = self.dense_arch(dense_features)
+        embedded_sparse = self.sparse_arch(sparse_features)
+        concat = self.inter_arch(embedded_dense, embedded_sparse)
+        return self.over_arch(concat)
+
+
+class DLRM_Projection(nn.Module):
+    """DLRM with projected interactions (reference: models/dlrm.py:624)."""
+
+    def __init__(
+        self,
+        embedding_bag_collection: EmbeddingBagCollection,
+        dense_in_features: int,
+        dense_arch_layer_sizes: List[int],
+        over_arch_layer_sizes: List[int],
+        interaction_branch1_layer_sizes: List[int],
+        interaction_branch2_layer_sizes: List[int],
+        dense_device: Optional[torch.device] = None,
+    ) -> None:
+        super().__init__()
+        self.sparse_arch = SparseArch(embedding_bag_collection)
+        num_sparse = len(self.sparse_arch.sparse_feature_names)
+        D = self.sparse_arch._d
+        assert dense_arch_layer_sizes[-1] == D
+        for name, sizes in (
+            ("branch1", interaction_branch1_layer_sizes),
+            ("branch2", interaction_branch2_layer_sizes),
+        ):
+            assert sizes[-1] % D == 0, (
+                f"interaction {name} must end at a multiple of the embedding "
+                f"dim {D}, got {sizes[-1]}"
+            )
+        self.dense_arch = DenseArch(
+            dense_in_features, dense_arch_layer_sizes, device=dense_device
+        )
+        cross_in = (num_sparse + 1) * D
+        i1 = interaction_branch1_layer_sizes[-1] // D
+        i2 = interaction_branch2_layer_sizes[-1] // D
+        self.inter_arch = InteractionProjectionArch(
+            num_sparse,
+            MLP(cross_in, interaction_branch1_layer_sizes, bias=True,
+                activation="relu", device=dense_device),
+            MLP(cross_in, interaction_branch2_layer_sizes, bias=True,
+                activation="relu", device=dense_device),
+            D,
+        )
+        self.over_arch = OverArch(D + i1 * i2, over_arch_layer_sizes, device=dense_device)
+
+    def forward(
+        self, dense_features: torch.Tensor, sparse_features: KeyedJaggedTensor
+    ) -> torch.Tensor:
         embedded_dense = self.dense_arch(dense_features)
         embedded_sparse = self.sparse_arch(sparse_features)
         concat = self.inter_arch(embedded_dense, embedded_sparse)
