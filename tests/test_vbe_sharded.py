@@ -415,3 +415,111 @@ def test_vbe_grid_sharded_world4():
     run_multi_process(
         _run_vbe_two_level, 4, "gloo", ShardingType.GRID_SHARD.value, STRIDES4
     )
+
+
+def _weighted_vbe_kjt(seed=11):
+    base = _global_vbe_kjt(seed=seed)
+    g = torch.Generator().manual_seed(seed + 1)
+    w = torch.rand(base.values().numel(), generator=g) + 0.5
+    return KeyedJaggedTensor(
+        keys=base.keys(),
+        values=base.values(),
+        weights=w,
+        lengths=base.lengths(),
+        stride_per_key_per_rank=base.stride_per_key_per_rank(),
+    )
+
+
+def _local_slice_weighted(kjt_global, rank):
+    """Rank slice that also carries the per-sample weights."""
+    spk = kjt_global.stride_per_key_per_rank()
+    lengths = kjt_global.lengths()
+    values = kjt_global.values()
+    weights = kjt_global.weights()
+    key_strides = [sum(s) for s in spk]
+    len_bounds = [0]
+    for ks in key_strides:
+        len_bounds.append(len_bounds[-1] + ks)
+    voffs = torch.zeros(lengths.numel() + 1, dtype=torch.int64)
+    torch.cumsum(lengths, 0, out=voffs[1:])
+    out_l, out_v, out_w, out_spk = [], [], [], []
+    for ki in range(len(spk)):
+        r0 = len_bounds[ki] + sum(spk[ki][:rank])
+        r1 = r0 + spk[ki][rank]
+        out_l.append(lengths[r0:r1])
+        out_v.append(values[int(voffs[r0]) : int(voffs[r1])])
+        out_w.append(weights[int(voffs[r0]) : int(voffs[r1])])
+        out_spk.append([spk[ki][rank]])
+    return KeyedJaggedTensor(
+        keys=kjt_global.keys(),
+        values=torch.cat(out_v),
+        weights=torch.cat(out_w),
+        lengths=torch.cat(out_l),
+        stride_per_key_per_rank=out_spk,
+    )
+
+
+def _run_vbe_weighted(rank, world_size, sharding_type):
+    """Weighted VBE golden test: per-sample weights ride the bucketized
+    input dist; pooled sums match a weighted unsharded oracle."""
+    torch.manual_seed(42)
+    model = SparseModel(_make_configs(), is_weighted=True)
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world_size, compute_device="cpu", hbm_cap=1 << 40),
+        constraints={
+            t[0]: ParameterConstraints(sharding_types=[sharding_type])
+            for t in TABLES
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    torch.manual_seed(42)
+    golden = FusedEmbeddingBagCollection(
+        _make_configs(), optimizer="rowwise_adagrad", learning_rate=LR,
+        is_weighted=True,
+    )
+    dmp.load_state_dict(
+        {
+            f"sparse.embedding_bags.{t[0]}.weight": w
+            for t, w in zip(TABLES, golden.split_embedding_weights())
+        },
+        strict=False,
+    )
+    kjt_global = _weighted_vbe_kjt()
+    kjt_local = _local_slice_weighted(kjt_global, rank)
+    kt = dmp(kjt_local)
+    vals = kt.values()
+    golden_vals = golden(kjt_global).values()
+    exp = []
+    goff = 0
+    for (n, rows, d, f) in TABLES:
+        b_tot = sum(STRIDES[f])
+        block = golden_vals[goff : goff + b_tot * d].view(b_tot, d)
+        r0 = sum(STRIDES[f][:rank])
+        exp.append(block[r0 : r0 + STRIDES[f][rank]].reshape(-1))
+        goff += b_tot * d
+    torch.testing.assert_close(vals, torch.cat(exp), atol=1e-5, rtol=1e-5)
+    vals.sum().backward()
+    golden_vals.sum().backward()
+    sd = dmp.state_dict()
+    for (n, rows, d, f), gw in zip(TABLES, golden.split_embedding_weights()):
+        st = sd[f"sparse.embedding_bags.{n}.weight"]
+        for shard in st.local_shards():
+            ro, co = shard.metadata.shard_offsets
+            h, w = shard.metadata.shard_sizes
+            torch.testing.assert_close(
+                shard.tensor, gw[ro : ro + h, co : co + w], atol=1e-4, rtol=1e-4
+            )
+
+
+def test_vbe_weighted_tw():
+    run_multi_process(_run_vbe_weighted, 2, "gloo", ShardingType.TABLE_WISE.value)
+
+
+def test_vbe_weighted_rw():
+    run_multi_process(_run_vbe_weighted, 2, "gloo", ShardingType.ROW_WISE.value)

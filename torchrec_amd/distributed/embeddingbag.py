@@ -649,7 +649,10 @@ class ShardedEmbeddingBagCollection(nn.Module):
                 k: sum(sp)
                 for k, sp in zip(features.keys(), features.stride_per_key_per_rank())
             }
-            assert not self._is_weighted, "VBE: unweighted only"
+            # weighted VBE: per-sample weights ride the bucketize + KJT a2a as
+            # data (the reference's sharded weighted semantics — psw gradients
+            # exist only for post-dist feature processors, which compute them
+            # locally inside the lookup's autograd)
             allowed = {
                 ShardingType.TABLE_WISE.value,
                 ShardingType.ROW_WISE.value,
