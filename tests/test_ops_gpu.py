@@ -339,3 +339,50 @@ class TestSegSort2Level:
             ref_vals, ref_idx = torch.sort(seg, stable=True)
             assert torch.equal(sorted_lin[lo:hi].cpu(), ref_vals)
             assert torch.equal(perm[lo:hi].cpu().long() - lo, ref_idx)
+
+
+@pytest.mark.gpu
+class TestBagMetadata:
+    def test_fused_matches_torch_chain(self):
+        """tbe_bag_metadata (one binary-search launch) must equal the
+        arange/repeat_interleave/index reference for ragged bags."""
+        ops.hip_ops()
+        torch.manual_seed(0)
+        F, B = 5, 64
+        dims = [8, 16, 8, 32, 8]
+        rows = [100, 50, 200, 30, 400]
+        d_out = [0]
+        row_off = [0]
+        for d, r in zip(dims, rows):
+            d_out.append(d_out[-1] + d)
+            row_off.append(row_off[-1] + r)
+        feat_d_out = torch.tensor(d_out[:-1], dtype=torch.int64, device="cuda")
+        feat_row_offset = torch.tensor(row_off[:-1], dtype=torch.int64, device="cuda")
+        lengths = torch.randint(0, 7, (F * B,), device="cuda")
+        offsets = torch.zeros(F * B + 1, dtype=torch.int64, device="cuda")
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        N = int(offsets[-1])
+        indices = torch.cat([
+            torch.randint(0, rows[f], (int(lengths[f * B : (f + 1) * B].sum()),),
+                          device="cuda")
+            for f in range(F)
+        ])
+        pos_row, pos_col, linear = torch.ops.trec_amd.tbe_bag_metadata(
+            offsets, indices, feat_d_out, feat_row_offset, B
+        )
+        # reference chain
+        bag_ids = torch.repeat_interleave(
+            torch.arange(F * B, device="cuda"), lengths, output_size=N
+        )
+        f = torch.div(bag_ids, B, rounding_mode="floor")
+        torch.testing.assert_close(pos_row, (bag_ids - f * B).to(torch.int32))
+        torch.testing.assert_close(pos_col, feat_d_out[f])
+        torch.testing.assert_close(linear, indices + feat_row_offset[f])
+
+    def test_empty(self):
+        ops.hip_ops()
+        z = torch.zeros(1, dtype=torch.int64, device="cuda")
+        pr, pc, ln = torch.ops.trec_amd.tbe_bag_metadata(
+            z, z[:0], z[:0], z[:0], 1
+        )
+        assert pr.numel() == 0 and pc.numel() == 0 and ln.numel() == 0

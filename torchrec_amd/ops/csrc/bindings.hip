@@ -53,6 +53,9 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_2level(
 std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_large(
     const at::Tensor& linear, const at::Tensor& feat_bounds, int64_t F, int64_t end_bit);
 std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_linear);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> tbe_bag_metadata(
+    const at::Tensor& offsets, const at::Tensor& indices,
+    const at::Tensor& feat_d_out, const at::Tensor& feat_row_offset, int64_t B);
 void tbe_backward_fused(at::Tensor weights, at::Tensor momentum, const at::Tensor& grad,
                         const at::Tensor& sorted_linear, const at::Tensor& sort_perm,
                         const at::Tensor& seg_offsets, const at::Tensor& num_runs,
@@ -172,6 +175,9 @@ TORCH_LIBRARY(trec_amd, m) {
       " -> (Tensor, Tensor, Tensor)");
   m.def("tbe_backward_prep(Tensor sorted_linear) -> (Tensor, Tensor)");
   m.def(
+      "tbe_bag_metadata(Tensor offsets, Tensor indices, Tensor feat_d_out,"
+      " Tensor feat_row_offset, int B) -> (Tensor, Tensor, Tensor)");
+  m.def(
       "tbe_backward_fused(Tensor(a!) weights, Tensor(b!) momentum, Tensor grad,"
       " Tensor sorted_linear, Tensor sort_perm, Tensor seg_offsets, Tensor num_runs,"
       " Tensor pos_row, Tensor pos_col, Tensor pos_scale, Tensor table_row_offsets,"
@@ -239,6 +245,7 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("seg_sort_pairs_2level", trec_amd::seg_sort_pairs_2level);
   m.impl("seg_sort_pairs_large", trec_amd::seg_sort_pairs_large);
   m.impl("tbe_backward_prep", trec_amd::tbe_backward_prep);
+  m.impl("tbe_bag_metadata", trec_amd::tbe_bag_metadata);
   m.impl("tbe_backward_fused", trec_amd::tbe_backward_fused);
   m.impl("tbe_grad_per_sample_weights", trec_amd::tbe_grad_per_sample_weights);
   m.impl("gather_run_heads", trec_amd::gather_run_heads);

@@ -357,6 +357,58 @@ at::Tensor tbe_forward_pooled_vbe(
 }
 
 // ---------------------------------------------------------------------------
+// backward metadata: ONE launch for the per-position (row, col, linear-id)
+// arrays, replacing torch's arange + repeat_interleave + div/sub + three
+// index kernels (~12 launch-floor kernels per step in the captured graph).
+// Each position binary-searches its owning bag in the offsets array (the
+// FB+1 prefix sums stay L2-resident across the wave).
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(256) tbe_bag_metadata_kernel(
+    const int64_t* __restrict__ offsets,          // [FB+1]
+    const int64_t* __restrict__ indices,          // [N]
+    const int64_t* __restrict__ feat_d_out,       // [F]
+    const int64_t* __restrict__ feat_row_offset,  // [F]
+    int64_t FB, int64_t N, int B,
+    int32_t* __restrict__ pos_row, int64_t* __restrict__ pos_col,
+    int64_t* __restrict__ linear) {
+  int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  int64_t nthreads = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  for (; i < N; i += nthreads) {
+    // first bag whose exclusive end exceeds position i
+    int64_t lo = 0, hi = FB;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      if (offsets[mid + 1] <= i) lo = mid + 1; else hi = mid;
+    }
+    int64_t f = lo / B;
+    pos_row[i] = static_cast<int32_t>(lo - f * B);
+    pos_col[i] = feat_d_out[f];
+    linear[i] = indices[i] + feat_row_offset[f];
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> tbe_bag_metadata(
+    const at::Tensor& offsets, const at::Tensor& indices,
+    const at::Tensor& feat_d_out, const at::Tensor& feat_row_offset, int64_t B) {
+  int64_t N = indices.numel();
+  int64_t FB = offsets.numel() - 1;
+  auto pos_row = at::empty({N}, indices.options().dtype(at::kInt));
+  auto pos_col = at::empty({N}, indices.options().dtype(at::kLong));
+  auto linear = at::empty({N}, indices.options().dtype(at::kLong));
+  if (N == 0) return {pos_row, pos_col, linear};
+  auto stream = tbe_stream();
+  int grid = grid_for(N, 256);
+  hipLaunchKernelGGL(tbe_bag_metadata_kernel, dim3(grid), dim3(256), 0, stream,
+                     offsets.data_ptr<int64_t>(), indices.data_ptr<int64_t>(),
+                     feat_d_out.data_ptr<int64_t>(),
+                     feat_row_offset.data_ptr<int64_t>(), FB, N, (int)B,
+                     pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(),
+                     linear.data_ptr<int64_t>());
+  return {pos_row, pos_col, linear};
+}
+
+// ---------------------------------------------------------------------------
 // forward (sequence / non-pooled): out[n, :] = W[table(f(n))][idx[n]]
 // ---------------------------------------------------------------------------
 
