@@ -386,3 +386,31 @@ class TestBagMetadata:
             z, z[:0], z[:0], z[:0], 1
         )
         assert pr.numel() == 0 and pc.numel() == 0 and ln.numel() == 0
+
+
+@pytest.mark.gpu
+class TestFusedBce:
+    def test_matches_torch_bce(self):
+        ops.hip_ops()
+        torch.manual_seed(0)
+        for B, dtype in [(8192, torch.float32), (8192, torch.bfloat16), (1000, torch.float32)]:
+            logits = (torch.randn(B, device="cuda") * 3).to(dtype).requires_grad_(True)
+            labels = torch.randint(0, 2, (B,), device="cuda").float()
+            ref_in = logits.detach().float().requires_grad_(True)
+            loss = ops.fused_bce_with_logits(logits, labels)
+            ref = torch.nn.functional.binary_cross_entropy_with_logits(ref_in, labels)
+            torch.testing.assert_close(loss, ref, atol=2e-3, rtol=2e-3)
+            loss.backward()
+            ref.backward()
+            torch.testing.assert_close(
+                logits.grad.float(), ref_in.grad, atol=2e-3, rtol=2e-2
+            )
+
+    def test_deterministic(self):
+        ops.hip_ops()
+        torch.manual_seed(1)
+        logits = torch.randn(65536, device="cuda")
+        labels = torch.randint(0, 2, (65536,), device="cuda").float()
+        a = ops.fused_bce_with_logits(logits, labels)
+        b = ops.fused_bce_with_logits(logits, labels)
+        assert torch.equal(a, b)

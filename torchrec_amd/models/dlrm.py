@@ -325,5 +325,10 @@ class DLRMTrain(nn.Module):
     ) -> Tuple[torch.Tensor, Tuple[torch.Tensor, torch.Tensor, torch.Tensor]]:
         logits = self.model(batch.dense_features, batch.sparse_features)
         logits = logits.squeeze(-1)
-        loss = self.loss_fn(logits, batch.labels.float())
+        if not isinstance(logits, torch.fx.Proxy) and logits.is_cuda:
+            from torchrec_amd import ops as _ops
+
+            loss = _ops.fused_bce_with_logits(logits, batch.labels)
+        else:
+            loss = self.loss_fn(logits, batch.labels.float())
         return loss, (loss.detach(), logits.detach(), batch.labels.detach())

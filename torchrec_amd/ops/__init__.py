@@ -486,6 +486,30 @@ def fused_interaction(dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor
     return _FusedInteraction.apply(dense, sparse)
 
 
+class _FusedBceWithLogits(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels):  # type: ignore[override]
+        labels_f = labels.float()
+        ctx.save_for_backward(logits, labels_f)
+        return torch.ops.trec_amd.bce_with_logits_fwd(logits, labels_f)
+
+    @staticmethod
+    def backward(ctx, grad_out):  # type: ignore[override]
+        logits, labels_f = ctx.saved_tensors
+        dx = torch.ops.trec_amd.bce_with_logits_bwd(
+            logits, labels_f, grad_out.contiguous().float()
+        )
+        return dx, None
+
+
+def fused_bce_with_logits(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """Mean-reduced BCEWithLogits in two kernels (fwd deterministic two-level
+    reduce; bwd one elementwise pass) — torch's nn.BCEWithLogitsLoss chain is
+    ~12 launch-floor kernels inside a captured graph."""
+    hip_ops()
+    return _FusedBceWithLogits.apply(logits.contiguous(), labels)
+
+
 def jagged_index_select_2d(
     values: torch.Tensor, lengths: torch.Tensor, indices: torch.Tensor
 ) -> Tuple[torch.Tensor, torch.Tensor]:

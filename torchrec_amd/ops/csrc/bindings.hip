@@ -113,6 +113,9 @@ at::Tensor tbe_forward_seq_int8(const at::Tensor& qweights,
 // interaction.hip
 at::Tensor col_sum(const at::Tensor& input);
 // mlp_ops.hip
+at::Tensor bce_with_logits_fwd(const at::Tensor& logits, const at::Tensor& labels);
+at::Tensor bce_with_logits_bwd(const at::Tensor& logits, const at::Tensor& labels,
+                               const at::Tensor& grad_out);
 std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
                                                     const at::Tensor& y);
 // interaction_mfma.hip
@@ -219,6 +222,8 @@ TORCH_LIBRARY(trec_amd, m) {
   m.def("col_sum(Tensor input) -> Tensor");
   m.def("interaction_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
   m.def("relu_bwd_col_sum(Tensor grad_out, Tensor y) -> (Tensor, Tensor)");
+  m.def("bce_with_logits_fwd(Tensor logits, Tensor labels) -> Tensor");
+  m.def("bce_with_logits_bwd(Tensor logits, Tensor labels, Tensor grad_out) -> Tensor");
   m.def("interaction_mfma_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
   m.def(
       "interaction_mfma_backward(Tensor grad_out, Tensor dense, Tensor sparse, Tensor pair_col)"
@@ -261,6 +266,8 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("col_sum", trec_amd::col_sum);
   m.impl("interaction_forward", trec_amd::interaction_forward);
   m.impl("relu_bwd_col_sum", trec_amd::relu_bwd_col_sum);
+  m.impl("bce_with_logits_fwd", trec_amd::bce_with_logits_fwd);
+  m.impl("bce_with_logits_bwd", trec_amd::bce_with_logits_bwd);
   m.impl("interaction_mfma_forward", trec_amd::interaction_mfma_forward);
   m.impl("interaction_mfma_backward", trec_amd::interaction_mfma_backward);
   m.impl("interaction_backward", trec_amd::interaction_backward);

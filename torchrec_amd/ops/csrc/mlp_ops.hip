@@ -168,4 +168,114 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
   return {g, db};
 }
 
+// ---------------------------------------------------------------------------
+// fused BCE-with-logits (mean reduction). torch's BCEWithLogitsLoss expands
+// to ~12 launch-floor kernels per step inside the captured graph
+// (log_sigmoid fwd chain + sigmoid/sub/scale bwd); these two kernels do one
+// pass each. fwd: deterministic two-level mean of
+//   max(x,0) - x*y + log1p(exp(-|x|)); bwd: dx = go * (sigmoid(x) - y) / B.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ float bce_ldf(float v) { return v; }
+__device__ __forceinline__ float bce_ldf(__half v) { return __half2float(v); }
+__device__ __forceinline__ float bce_ldf(__hip_bfloat16 v) { return __bfloat162float(v); }
+
+template <typename x_t>
+__global__ void __launch_bounds__(kBlockThreads) bce_logits_partial_kernel(
+    const x_t* __restrict__ logits, const float* __restrict__ labels, int64_t B,
+    float* __restrict__ partial /* [G] */) {
+  __shared__ float red[kBlockThreads / kWaveSize];
+  float acc = 0.f;
+  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < B; i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    float x = bce_ldf(logits[i]);
+    float y = labels[i];
+    acc += fmaxf(x, 0.f) - x * y + log1pf(__expf(-fabsf(x)));
+  }
+  // wave then block reduce
+  for (int off = kWaveSize / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, kWaveSize);
+  int wave = threadIdx.x / kWaveSize;
+  if (threadIdx.x % kWaveSize == 0) red[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int w = 0; w < kBlockThreads / kWaveSize; ++w) s += red[w];
+    partial[blockIdx.x] = s;
+  }
+}
+
+__global__ void bce_logits_final_kernel(const float* __restrict__ partial, int G,
+                                        float inv_B, float* __restrict__ out) {
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int g = 0; g < G; ++g) s += partial[g];
+    *out = s * inv_B;
+  }
+}
+
+template <typename x_t>
+__global__ void __launch_bounds__(kBlockThreads) bce_logits_bwd_kernel(
+    const x_t* __restrict__ logits, const float* __restrict__ labels,
+    const float* __restrict__ grad_out, float inv_B, int64_t B,
+    x_t* __restrict__ dx) {
+  float g = *grad_out * inv_B;
+  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < B; i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    float x = bce_ldf(logits[i]);
+    float sig = 1.f / (1.f + __expf(-x));
+    dx[i] = float2emb(g * (sig - labels[i]), x_t{});
+  }
+}
+
+at::Tensor bce_with_logits_fwd(const at::Tensor& logits, const at::Tensor& labels) {
+  TORCH_CHECK(logits.is_cuda() && labels.scalar_type() == at::kFloat);
+  int64_t B = logits.numel();
+  auto out = at::empty({}, logits.options().dtype(at::kFloat));
+  auto stream = mlp_stream();
+  int G = std::min<int64_t>(512, (B + kBlockThreads - 1) / kBlockThreads);
+  G = std::max(G, 1);
+  auto partial = at::empty({G}, logits.options().dtype(at::kFloat));
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, logits.scalar_type(),
+                                  "bce_fwd", [&] {
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 unsupported");
+    } else {
+      using dev_t = typename DevType<scalar_t>::type;
+      hipLaunchKernelGGL((bce_logits_partial_kernel<dev_t>), dim3(G),
+                         dim3(kBlockThreads), 0, stream,
+                         reinterpret_cast<const dev_t*>(logits.data_ptr<scalar_t>()),
+                         labels.data_ptr<float>(), B, partial.data_ptr<float>());
+    }
+  });
+  hipLaunchKernelGGL(bce_logits_final_kernel, dim3(1), dim3(kWaveSize), 0, stream,
+                     partial.data_ptr<float>(), G, B > 0 ? 1.f / B : 0.f,
+                     out.data_ptr<float>());
+  return out;
+}
+
+at::Tensor bce_with_logits_bwd(const at::Tensor& logits, const at::Tensor& labels,
+                               const at::Tensor& grad_out) {
+  int64_t B = logits.numel();
+  auto dx = at::empty_like(logits);
+  if (B == 0) return dx;
+  auto stream = mlp_stream();
+  int grid = grid_for(B, kBlockThreads);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, logits.scalar_type(),
+                                  "bce_bwd", [&] {
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 unsupported");
+    } else {
+      using dev_t = typename DevType<scalar_t>::type;
+      hipLaunchKernelGGL((bce_logits_bwd_kernel<dev_t>), dim3(grid),
+                         dim3(kBlockThreads), 0, stream,
+                         reinterpret_cast<const dev_t*>(logits.data_ptr<scalar_t>()),
+                         labels.data_ptr<float>(),
+                         grad_out.data_ptr<float>(), 1.f / B, B,
+                         reinterpret_cast<dev_t*>(dx.data_ptr<scalar_t>()));
+    }
+  });
+  return dx;
+}
+
 }  // namespace trec_amd
