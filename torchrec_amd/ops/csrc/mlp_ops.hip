@@ -26,7 +26,12 @@ template <typename scalar_t, int VPT>
 __global__ void __launch_bounds__(kBlockThreads) relu_bwd_colsum_partial_kernel(
     const scalar_t* __restrict__ dy, const scalar_t* __restrict__ y, int64_t M,
     int64_t N, int rows_per_group, int n_groups, scalar_t* __restrict__ g,
-    float* __restrict__ partial /* [G, N] */) {
+    float* __restrict__ partial /* [G, N] */, int* __restrict__ counters,
+    int n_tiles) {
+  // zero the finish kernel's per-tile semaphores (it launches after us on
+  // the same stream, so ordering is guaranteed)
+  if (counters && blockIdx.x == 0)
+    for (int t = threadIdx.x; t < n_tiles; t += blockDim.x) counters[t] = 0;
   const int64_t col_chunks = (N + VPT - 1) / VPT;
   const int64_t items = col_chunks * n_groups;
   for (int64_t it = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
@@ -85,29 +90,51 @@ __global__ void __launch_bounds__(kBlockThreads) relu_bwd_colsum_partial_kernel(
   }
 }
 
-// two-level deterministic reduce of [G, N] -> [N]: level 1 collapses G into
-// S segments with a 2-D grid (full-chip parallel), level 2 folds S
-__global__ void __launch_bounds__(kBlockThreads) colsum_seg_f32_kernel(
+// single-kernel deterministic reduce of [G, N] -> [N] in the output dtype:
+// a 2-D grid ((column tile) x S) collapses G rows into S per-tile segment
+// sums; the LAST block to finish each column tile (device-scope semaphore,
+// the rocprim decoupled-lookback pattern) folds the S rows in FIXED order —
+// deterministic — and writes db, cast. Replaces three kernels (segment
+// reduce + final reduce + dtype cast) with one.
+template <typename o_t>
+__global__ void __launch_bounds__(kBlockThreads) colsum_finish_kernel(
     const float* __restrict__ partial, int G, int seg_rows, int64_t N,
-    float* __restrict__ seg_out /* [S, N] */) {
+    float* __restrict__ seg /* [S, N] */, int* __restrict__ counters,
+    int S, o_t* __restrict__ db) {
   int64_t c = static_cast<int64_t>(blockIdx.x) * kBlockThreads + threadIdx.x;
-  if (c >= N) return;
   int s = blockIdx.y;
-  int g0 = s * seg_rows;
-  int g1 = min(G, g0 + seg_rows);
-  float acc = 0.f;
-  for (int grp = g0; grp < g1; ++grp)
-    acc += partial[static_cast<int64_t>(grp) * N + c];
-  seg_out[static_cast<int64_t>(s) * N + c] = acc;
+  if (c < N) {
+    int g0 = s * seg_rows;
+    int g1 = min(G, g0 + seg_rows);
+    float acc = 0.f;
+    for (int grp = g0; grp < g1; ++grp)
+      acc += partial[static_cast<int64_t>(grp) * N + c];
+    __hip_atomic_store(&seg[static_cast<int64_t>(s) * N + c], acc,
+                       __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+  }
+  __syncthreads();
+  __shared__ int last;
+  if (threadIdx.x == 0)
+    last = (__hip_atomic_fetch_add(&counters[blockIdx.x], 1, __ATOMIC_ACQ_REL,
+                                   __HIP_MEMORY_SCOPE_AGENT) == S - 1);
+  __syncthreads();
+  if (last && c < N) {
+    float tot = 0.f;
+    for (int ss = 0; ss < S; ++ss)
+      tot += __hip_atomic_load(&seg[static_cast<int64_t>(ss) * N + c],
+                               __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+    db[c] = float2emb(tot, o_t{});
+  }
 }
 
-__global__ void __launch_bounds__(kBlockThreads) colsum_final_f32_kernel(
-    const float* __restrict__ partial, int G, int64_t N, float* __restrict__ out) {
+template <typename o_t>
+__global__ void __launch_bounds__(kBlockThreads) colsum_final_cast_kernel(
+    const float* __restrict__ partial, int G, int64_t N, o_t* __restrict__ out) {
   int64_t c = static_cast<int64_t>(blockIdx.x) * kBlockThreads + threadIdx.x;
   if (c >= N) return;
   float acc = 0.f;
   for (int grp = 0; grp < G; ++grp) acc += partial[static_cast<int64_t>(grp) * N + c];
-  out[c] = acc;
+  out[c] = float2emb(acc, o_t{});
 }
 
 std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
@@ -118,7 +145,7 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
   auto dy = grad_out.contiguous();
   auto yc = y.contiguous();
   auto g = at::empty_like(dy);
-  auto db = at::empty({N}, dy.options().dtype(at::kFloat));
+  auto db = at::empty({N}, dy.options());  // bias grad in the grad dtype
   if (M == 0 || N == 0) {
     db.zero_();
     return {g, db};
@@ -139,6 +166,15 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
       int rows_per_group = (int)((M + G - 1) / G);
       G = (int)((M + rows_per_group - 1) / rows_per_group);
       auto partial = at::empty({(int64_t)G * N}, dy.options().dtype(at::kFloat));
+      int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);
+      bool finish = G > 64;
+      at::Tensor counters, seg;
+      int* cnt_ptr = nullptr;
+      if (finish) {
+        counters = at::empty({ftiles}, dy.options().dtype(at::kInt));
+        seg = at::empty({(int64_t)32 * N}, dy.options().dtype(at::kFloat));
+        cnt_ptr = counters.data_ptr<int>();
+      }
       int grid = grid_for(col_chunks * G, kBlockThreads);
       hipLaunchKernelGGL((relu_bwd_colsum_partial_kernel<dev_t, VPT>), dim3(grid),
                          dim3(kBlockThreads), 0, stream,
@@ -146,22 +182,18 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
                          reinterpret_cast<const dev_t*>(yc.data_ptr<scalar_t>()), M, N,
                          rows_per_group, G,
                          reinterpret_cast<dev_t*>(g.data_ptr<scalar_t>()),
-                         partial.data_ptr<float>());
-      int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);
-      if (G > 64) {
+                         partial.data_ptr<float>(), cnt_ptr, ftiles);
+      if (finish) {
         constexpr int S = 32;
         int seg_rows = (G + S - 1) / S;
-        auto seg = at::empty({(int64_t)S * N}, dy.options().dtype(at::kFloat));
-        hipLaunchKernelGGL(colsum_seg_f32_kernel, dim3(ftiles, S),
+        hipLaunchKernelGGL((colsum_finish_kernel<dev_t>), dim3(ftiles, S),
                            dim3(kBlockThreads), 0, stream, partial.data_ptr<float>(),
-                           G, seg_rows, N, seg.data_ptr<float>());
-        hipLaunchKernelGGL(colsum_final_f32_kernel, dim3(ftiles), dim3(kBlockThreads),
-                           0, stream, seg.data_ptr<float>(), S, N,
-                           db.data_ptr<float>());
+                           G, seg_rows, N, seg.data_ptr<float>(), cnt_ptr, S,
+                           reinterpret_cast<dev_t*>(db.data_ptr<scalar_t>()));
       } else {
-        hipLaunchKernelGGL(colsum_final_f32_kernel, dim3(ftiles), dim3(kBlockThreads),
-                           0, stream, partial.data_ptr<float>(), G, N,
-                           db.data_ptr<float>());
+        hipLaunchKernelGGL((colsum_final_cast_kernel<dev_t>), dim3(ftiles),
+                           dim3(kBlockThreads), 0, stream, partial.data_ptr<float>(),
+                           G, N, reinterpret_cast<dev_t*>(db.data_ptr<scalar_t>()));
       }
     }
   });
