@@ -31,7 +31,8 @@ def _time_kernel(fn, iters=50, warmup=10) -> float:
     return (time.perf_counter() - t0) / iters * 1e6  # us
 
 
-def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1, precision="fp32"):
+def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1, precision="fp32",
+              id_pattern="random"):
     from torchrec_amd.ops.tbe import TableBatchedEmbeddingBags
 
     torch.manual_seed(0)
@@ -42,7 +43,10 @@ def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1, precision="fp32"):
     )
     F = tables
     lengths = torch.full((F * B,), L, dtype=torch.int64)
-    indices = torch.randint(0, rows, (F * B * L,)).cuda()
+    if id_pattern == "sequential":
+        indices = (torch.arange(F * B * L) % rows).cuda()
+    else:
+        indices = torch.randint(0, rows, (F * B * L,)).cuda()
     offsets = torch.zeros(F * B + 1, dtype=torch.int64)
     torch.cumsum(lengths, 0, out=offsets[1:])
     offsets = offsets.cuda()
@@ -61,7 +65,7 @@ def bench_tbe(B=8192, D=128, tables=26, rows=1_000_000, L=1, precision="fp32"):
     bytes_moved = F * B * L * D * elem + B * F * D * 4
     print(json.dumps({
         "bench": "tbe", "B": B, "D": D, "tables": tables, "rows_per_table": rows,
-        "precision": precision,
+        "precision": precision, "id_pattern": id_pattern,
         "fwd_us": round(fwd_us, 1), "fwd_bwd_opt_us": round(full_us, 1),
         "fwd_gb_s": round(bytes_moved / fwd_us / 1e3, 1),
     }))
@@ -159,12 +163,21 @@ def bench_quant_tbe(B=8192, D=128, tables=26, rows=100_000):
 if __name__ == "__main__":
     p = argparse.ArgumentParser()
     p.add_argument("--suite", default="all",
-                   choices=["all", "tbe", "interaction", "sort", "col_sum", "quant"])
+                   choices=["all", "tbe", "tbe_sweep", "interaction", "sort",
+                            "col_sum", "quant"])
     a = p.parse_args()
     assert torch.cuda.is_available(), "run on a GPU box"
     if a.suite in ("all", "tbe"):
         bench_tbe()
         bench_tbe(precision="bf16")
+    if a.suite == "tbe_sweep":
+        # backward access-pattern discriminator: random vs dense-coverage vs
+        # sequential ids localize whether tbe_bwd_fused is random-DRAM-bound
+        # or metadata-latency-bound
+        bench_tbe(rows=1_000_000, id_pattern="random")
+        bench_tbe(rows=131_072, id_pattern="random")
+        bench_tbe(rows=1_000_000, id_pattern="sequential")
+        bench_tbe(rows=1_000_000, id_pattern="random", B=65536)
     if a.suite in ("all", "interaction"):
         bench_interaction()
     if a.suite in ("all", "sort"):
