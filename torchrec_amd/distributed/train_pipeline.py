@@ -605,6 +605,35 @@ class TrainPipelineSemiSync(TrainPipelineSparseDist):
         return output
 
 
+class TrainPipelinePT2(TrainPipelineBase):
+    """torch.compile pipeline (reference train_pipelines.py:423): the model's
+    dense compute is compiled on first progress; sparse structures (KJT)
+    enter the graph through the fx-leaf tracer contract, so unsupported
+    constructs fall back to graph breaks rather than errors.
+
+    MI355X note: inductor on ROCm emits HIP via triton-cpu/cpp on CPU and is
+    primarily useful here for the dense arch; the HIP custom ops stay as
+    opaque calls."""
+
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        optimizer: torch.optim.Optimizer,
+        device: torch.device,
+        autocast_dtype: Optional[torch.dtype] = None,
+        compile_kwargs: Optional[dict] = None,
+    ) -> None:
+        super().__init__(model, optimizer, device, autocast_dtype)
+        self._compile_kwargs = dict(compile_kwargs or {})
+        self._compiled = False
+
+    def progress(self, dataloader_iter: Iterator[Pipelineable]) -> Any:
+        if not self._compiled:
+            self._model = torch.compile(self._model, **self._compile_kwargs)
+            self._compiled = True
+        return super().progress(dataloader_iter)
+
+
 class MicroBatchPipeline:
     """Micro-batch serving/eval pipeline (reference maglev/pipeline.py:20):
     splits each incoming batch into ``num_micro`` slices and runs them

@@ -988,14 +988,14 @@ __device__ __forceinline__ int upper_bound_segment_i32(const int32_t* offs, int 
 
 template <typename g_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
-    const g_t* __restrict__ grad, int64_t grad_stride,
-    const int64_t* __restrict__ sorted_linear, const int32_t* __restrict__ sort_perm,
-    const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
+    const g_t* __restrict__ grad,
+    const int4* __restrict__ desc_a,   // per run: {k0, k1, c0, c1}
+    const int* __restrict__ desc_d,    // per run: D
+    const int64_t* __restrict__ grow_off,   // per sorted position: grad row offset
+    const float* __restrict__ sorted_scale,  // nullable, sorted order
+    const int32_t* __restrict__ num_runs_ptr,
     const int32_t* __restrict__ chunk_offsets, const int32_t* __restrict__ total_chunks_ptr,
-    const int32_t* __restrict__ pos_row, const int64_t* __restrict__ pos_col,
-    const float* __restrict__ pos_scale, const int64_t* __restrict__ table_row_offsets,
-    const int32_t* __restrict__ dims, int T, int chunk_size, int64_t max_D,
-    float* __restrict__ scratch) {
+    int chunk_size, int64_t max_D, float* __restrict__ scratch) {
   int sl = threadIdx.x % LPS;
   int64_t slot = (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / LPS;
   int64_t n_slots = (static_cast<int64_t>(gridDim.x) * blockDim.x) / LPS;
@@ -1003,19 +1003,16 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
   int32_t num_runs = *num_runs_ptr;
   for (int64_t c = slot; c < total_chunks; c += n_slots) {
     int r = upper_bound_segment_i32(chunk_offsets, num_runs, (int32_t)c);
-    int32_t rk0 = seg_offsets[r], rk1 = seg_offsets[r + 1];
-    int32_t k0 = rk0 + (c - chunk_offsets[r]) * chunk_size;
-    int32_t k1 = min(rk1, k0 + chunk_size);
-    int64_t lin = sorted_linear[rk0];
-    int t = upper_bound_segment(table_row_offsets, T, lin);
-    int D = dims[t];
+    int4 da = desc_a[r];
+    int32_t k0 = da.x + (c - da.z) * chunk_size;
+    int32_t k1 = min(da.y, k0 + chunk_size);
+    int D = desc_d[r];
     float4 acc[CHUNKS];
 #pragma unroll
     for (int cc = 0; cc < CHUNKS; ++cc) acc[cc] = make_float4(0.f, 0.f, 0.f, 0.f);
     for (int32_t k = k0; k < k1; ++k) {
-      int32_t p = sort_perm[k];
-      const g_t* grow = grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p];
-      float s = pos_scale ? pos_scale[p] : 1.f;
+      const g_t* grow = grad + grow_off[k];
+      float s = sorted_scale ? sorted_scale[k] : 1.f;
 #pragma unroll
       for (int cc = 0; cc < CHUNKS; ++cc) {
         int col4 = cc * LPS + sl;
@@ -1037,6 +1034,45 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
   }
 }
 
+// one streaming pass builds (a) a 36-byte descriptor per duplicate-index run
+// — collapsing the fused kernel's 7-deep dependent scalar chain
+// (seg_offsets -> sorted_linear -> table binary search -> chunk_offsets) to
+// three independent vector loads — and (b) the grad-row offset (+ scale) per
+// SORTED position, so the gather chain perm -> (row, col) -> grad row
+// becomes one sequential read + one row load. Measured motivation: with
+// ids made fully sequential the fused kernel only sped up 8% => it is
+// latency-chain bound, not DRAM-randomness bound.
+__global__ void __launch_bounds__(kBlockThreads) tbe_bwd_build_desc_kernel(
+    const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ chunk_offsets,
+    const int32_t* __restrict__ num_runs_ptr, const int64_t* __restrict__ sorted_linear,
+    const int32_t* __restrict__ sort_perm, const int32_t* __restrict__ pos_row,
+    const int64_t* __restrict__ pos_col, const float* __restrict__ pos_scale,
+    int64_t grad_stride, const int64_t* __restrict__ table_row_offsets,
+    const int64_t* __restrict__ table_elem_offsets, const int32_t* __restrict__ dims,
+    int T, int64_t n, int4* __restrict__ desc_a, longlong2* __restrict__ desc_b,
+    int* __restrict__ desc_d, int64_t* __restrict__ grow_off,
+    float* __restrict__ sorted_scale) {
+  int32_t num_runs = *num_runs_ptr;
+  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; i < n;
+       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    int32_t p = sort_perm[i];
+    grow_off[i] = static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p];
+    if (sorted_scale) sorted_scale[i] = pos_scale[p];
+    if (i < num_runs) {
+      int32_t k0 = seg_offsets[i], k1 = seg_offsets[i + 1];
+      int64_t lin = sorted_linear[k0];
+      int t = upper_bound_segment(table_row_offsets, T, lin);
+      int D = dims[t];
+      desc_a[i] = make_int4(k0, k1, chunk_offsets[i], chunk_offsets[i + 1]);
+      longlong2 b;
+      b.x = lin;
+      b.y = table_elem_offsets[t] + (lin - table_row_offsets[t]) * static_cast<int64_t>(D);
+      desc_b[i] = b;
+      desc_d[i] = D;
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // backward + fused optimizer. One wave-slot per duplicate-index run; long
 // runs (chunk_offsets[r+1] > chunk_offsets[r]) sum pre-computed chunk
@@ -1047,16 +1083,15 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_long_partial_kernel(
 template <typename emb_t, typename g_t, int LPS, int CHUNKS>
 __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
     emb_t* __restrict__ weights, float* __restrict__ momentum,
-    const g_t* __restrict__ grad, int64_t grad_stride,
-    const int64_t* __restrict__ sorted_linear, const int32_t* __restrict__ sort_perm,
-    const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
-    const int32_t* __restrict__ chunk_offsets, const float* __restrict__ scratch,
-    int64_t max_D,
-    const int32_t* __restrict__ pos_row, const int64_t* __restrict__ pos_col,
-    const float* __restrict__ pos_scale,
-    const int64_t* __restrict__ table_row_offsets,   // [T+1] rows
-    const int64_t* __restrict__ table_elem_offsets,  // [T]
-    const int32_t* __restrict__ dims, int T, float lr, float eps, int mode,
+    const g_t* __restrict__ grad,
+    const int4* __restrict__ desc_a,        // per run: {k0, k1, c0, c1}
+    const longlong2* __restrict__ desc_b,   // per run: {lin, elem_base}
+    const int* __restrict__ desc_d,         // per run: D
+    const int64_t* __restrict__ grow_off,   // per sorted position
+    const float* __restrict__ sorted_scale, // nullable, sorted order
+    const int32_t* __restrict__ sort_perm,  // cache-locator path only
+    const int32_t* __restrict__ num_runs_ptr, const float* __restrict__ scratch,
+    int64_t max_D, float lr, float eps, int mode,
     emb_t* __restrict__ grad_weights, float* __restrict__ cache_weights,
     const int32_t* __restrict__ cache_loc, int64_t cache_stride,
     float* __restrict__ m1, float* __restrict__ m2, float beta1, float beta2,
@@ -1068,18 +1103,17 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
   int32_t num_runs = *num_runs_ptr;
   const uint64_t rng_base = rng_state ? (uint64_t)*rng_state : 0;
   for (int64_t r = slot; r < num_runs; r += n_slots) {
-    int32_t k0 = seg_offsets[r], k1 = seg_offsets[r + 1];
-    int64_t lin = sorted_linear[k0];
-    int t = upper_bound_segment(table_row_offsets, T, lin);
-    int64_t local = lin - table_row_offsets[t];
-    int D = dims[t];
+    int4 da = desc_a[r];
+    longlong2 dbv = desc_b[r];
+    int32_t k0 = da.x, k1 = da.y;
+    int64_t lin = dbv.x;
+    int D = desc_d[r];
     float4 acc[CHUNKS];
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) acc[c] = make_float4(0.f, 0.f, 0.f, 0.f);
-    int32_t c0 = chunk_offsets[r], c1 = chunk_offsets[r + 1];
-    if (c1 > c0) {
+    if (da.w > da.z) {
       // long run: sum phase-1 chunk partials in order (deterministic)
-      for (int32_t c = c0; c < c1; ++c) {
+      for (int32_t c = da.z; c < da.w; ++c) {
         const float4* srow = reinterpret_cast<const float4*>(scratch + c * max_D);
 #pragma unroll
         for (int cc = 0; cc < CHUNKS; ++cc) {
@@ -1095,9 +1129,8 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
       }
     } else {
       for (int32_t k = k0; k < k1; ++k) {
-        int32_t p = sort_perm[k];
-        const g_t* grow = grad + static_cast<int64_t>(pos_row[p]) * grad_stride + pos_col[p];
-        float s = pos_scale ? pos_scale[p] : 1.f;
+        const g_t* grow = grad + grow_off[k];
+        float s = sorted_scale ? sorted_scale[k] : 1.f;
 #pragma unroll
         for (int c = 0; c < CHUNKS; ++c) {
           int col4 = c * LPS + sl;
@@ -1112,7 +1145,7 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
       }
     }
     // weight row: emb_t in the table, fp32 if the row sits in the lxu cache
-    emb_t* wrow = weights + table_elem_offsets[t] + local * static_cast<int64_t>(D);
+    emb_t* wrow = weights + dbv.y;
     float* crow = nullptr;
     int32_t cloc = cache_loc ? cache_loc[sort_perm[k0]] : -1;
     if (cloc >= 0) crow = cache_weights + static_cast<int64_t>(cloc) * cache_stride;
@@ -1166,7 +1199,7 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
       float t_iter = iter_ptr ? *iter_ptr : 1.f;
       float bc1 = 1.f / (1.f - __powf(beta1, t_iter));
       float bc2 = 1.f / (1.f - __powf(beta2, t_iter));
-      int64_t ebase = table_elem_offsets[t] + local * static_cast<int64_t>(D);
+      int64_t ebase = dbv.y;
       float inv_sqrt_row = 0.f;
       if (mode == 4) {
         float gsq = 0.f;
@@ -1223,7 +1256,7 @@ __global__ void __launch_bounds__(kBlockThreads) tbe_bwd_fused_kernel(
         }
       }
     } else {
-      emb_t* gw = grad_weights + table_elem_offsets[t] + local * static_cast<int64_t>(D);
+      emb_t* gw = grad_weights + dbv.y;
 #pragma unroll
       for (int c = 0; c < CHUNKS; ++c) {
         int col4 = c * LPS + sl;
@@ -1250,27 +1283,48 @@ static void launch_tbe_bwd(
   const g_t* grad_ptr = reinterpret_cast<const g_t*>(grad.data_ptr());
   emb_t* gw_ptr = grad_weights.numel() > 0
       ? reinterpret_cast<emb_t*>(grad_weights.data_ptr<host_w_t>()) : nullptr;
+  // one streaming pass collapses the per-run metadata chain + per-position
+  // grad gather chain (see tbe_bwd_build_desc_kernel)
+  int64_t n = sorted_linear.numel();
+  auto opts_i = sorted_linear.options().dtype(at::kInt);
+  auto desc_a_t = at::empty({n * 4}, opts_i);
+  auto desc_b_t = at::empty({n * 2}, sorted_linear.options());
+  auto desc_d_t = at::empty({n}, opts_i);
+  auto grow_off_t = at::empty({n}, sorted_linear.options());
+  at::Tensor sorted_scale_t;
+  float* sscale_ptr = nullptr;
+  if (scale_ptr) {
+    sorted_scale_t = at::empty({n}, grad.options().dtype(at::kFloat));
+    sscale_ptr = sorted_scale_t.data_ptr<float>();
+  }
+  int4* desc_a_p = reinterpret_cast<int4*>(desc_a_t.data_ptr<int32_t>());
+  longlong2* desc_b_p = reinterpret_cast<longlong2*>(desc_b_t.data_ptr<int64_t>());
+  hipLaunchKernelGGL(tbe_bwd_build_desc_kernel, dim3(grid_for(n, kBlockThreads)),
+                     dim3(kBlockThreads), 0, stream,
+                     seg_offsets.data_ptr<int32_t>(), chunk_offsets.data_ptr<int32_t>(),
+                     num_runs.data_ptr<int32_t>(), sorted_linear.data_ptr<int64_t>(),
+                     sort_perm.data_ptr<int32_t>(), pos_row.data_ptr<int32_t>(),
+                     pos_col.data_ptr<int64_t>(), scale_ptr, grad.size(1),
+                     table_row_offsets.data_ptr<int64_t>(),
+                     table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,
+                     n, desc_a_p, desc_b_p, desc_d_t.data_ptr<int32_t>(),
+                     grow_off_t.data_ptr<int64_t>(), sscale_ptr);
 #define TBE_BWD_LAUNCH(LPS, CHUNKS)                                                          \
   do {                                                                                       \
     hipLaunchKernelGGL((tbe_bwd_long_partial_kernel<g_t, LPS, CHUNKS>), dim3(grid_long),     \
-                       dim3(kBlockThreads), 0, stream, grad_ptr, grad.size(1),               \
-                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
-                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
+                       dim3(kBlockThreads), 0, stream, grad_ptr, desc_a_p,                   \
+                       desc_d_t.data_ptr<int32_t>(), grow_off_t.data_ptr<int64_t>(),         \
+                       sscale_ptr, num_runs.data_ptr<int32_t>(),                             \
                        chunk_offsets.data_ptr<int32_t>(), total_chunks.data_ptr<int32_t>(),  \
-                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
-                       table_row_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,   \
                        chunk_size, max_D, scratch.data_ptr<float>());                        \
     hipLaunchKernelGGL((tbe_bwd_fused_kernel<emb_t, g_t, LPS, CHUNKS>), dim3(grid),          \
                        dim3(kBlockThreads),                                                  \
                        0, stream, reinterpret_cast<emb_t*>(uvm_ptr<host_w_t>(weights)),      \
                        uvm_ptr<float>(momentum),                                             \
-                       grad_ptr, grad.size(1),                                               \
-                       sorted_linear.data_ptr<int64_t>(), sort_perm.data_ptr<int32_t>(),     \
-                       seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),        \
-                       chunk_offsets.data_ptr<int32_t>(), scratch.data_ptr<float>(), max_D,  \
-                       pos_row.data_ptr<int32_t>(), pos_col.data_ptr<int64_t>(), scale_ptr,  \
-                       table_row_offsets.data_ptr<int64_t>(),                                \
-                       table_elem_offsets.data_ptr<int64_t>(), dims.data_ptr<int32_t>(), T,  \
+                       grad_ptr, desc_a_p, desc_b_p, desc_d_t.data_ptr<int32_t>(),           \
+                       grow_off_t.data_ptr<int64_t>(), sscale_ptr,                           \
+                       sort_perm.data_ptr<int32_t>(),                                        \
+                       num_runs.data_ptr<int32_t>(), scratch.data_ptr<float>(), max_D,       \
                        lr, eps, mode, gw_ptr, cache_w_ptr,                                   \
                        cache_loc_ptr, max_D, m1_ptr, m2_ptr, beta1, beta2, iter_ptr,         \
                        rng_ptr, stochastic);                                                 \
