@@ -104,6 +104,37 @@ def test_pipeline_base_cpu_single():
         dist.destroy_process_group()
 
 
+def test_pipeline_pt2_cpu_single():
+    """TrainPipelinePT2: torch.compile path (reference train_pipelines.py:423);
+    KJT enters via the fx-leaf contract, unsupported constructs graph-break."""
+    import os
+
+    from torchrec_amd.distributed.train_pipeline import TrainPipelinePT2
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29551")
+    created = not dist.is_initialized()
+    if created:
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        dmp, opt, tables = _build_dmp(1)
+        ref, ref_opt, _ = _build_dmp(1)
+        pipe = TrainPipelinePT2(dmp, opt, torch.device("cpu"))
+        it = iter(_batches(0, tables, n=3))
+        ref_it = iter(_batches(0, tables, n=3))
+        for _ in range(3):
+            out = pipe.progress(it)
+            batch = next(ref_it)
+            loss, _ = ref(batch)
+            ref_opt.zero_grad()
+            loss.backward()
+            ref_opt.step()
+            torch.testing.assert_close(out[0], loss.detach(), atol=1e-4, rtol=1e-4)
+    finally:
+        if created:
+            dist.destroy_process_group()
+
+
 def test_staged_pipeline_cpu():
     from torchrec_amd.distributed.train_pipeline import PipelineStage, StagedTrainPipeline
 
