@@ -203,7 +203,8 @@ def run_bench(gpus: int, steps: int, warmup: int, batch_size: int, scale: float,
     # TREC_FORCE_DIST=1: initialize the process group even at world 1 (the
     # driver-launch rehearsal exercises RCCL init + dist code paths on metal)
     if world > 1 or os.environ.get("TREC_FORCE_DIST") == "1":
-        dist.init_process_group("gloo" if cpu_mode else "nccl")
+        if not dist.is_initialized():
+            dist.init_process_group("gloo" if cpu_mode else "nccl")
         env = ShardingEnv.from_process_group(dist.group.WORLD)
         pg = dist.group.WORLD
     else:
@@ -544,6 +545,212 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
     print(json.dumps(result))
 
 
+def run_dist_graph_bench(
+    steps: int, warmup: int, batch_size: int, scale: float
+) -> None:
+    """Multi-rank hipGraph mode: the WHOLE per-rank train step — KJT a2a,
+    lookup, pooled output a2a, dense fwd/bwd, dense-grad all-reduce, both
+    optimizers — is stream-captured and replayed. RCCL collectives are
+    capture-safe, and with Criteo's fixed one-hot shapes the a2a splits are
+    constant, so the static-splits fast path (set_static_kjt_splits) removes
+    the per-step device->host splits sync that normally forbids capture.
+    Falls back to the eager pipeline if capture fails (raise to caller)."""
+    import torch.distributed as dist
+
+    from torchrec_amd.datasets.random import Batch
+    from torchrec_amd.distributed.dist_data import set_static_kjt_splits
+    from torchrec_amd.distributed.embeddingbag import EmbeddingBagCollectionSharder
+    from torchrec_amd.distributed.model_parallel import DistributedModelParallel
+    from torchrec_amd.distributed.planner.planners import EmbeddingShardingPlanner
+    from torchrec_amd.distributed.planner.types import Topology
+    from torchrec_amd.distributed.types import ShardingEnv
+    from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    dev_idx = local_rank % max(1, torch.cuda.device_count())
+    torch.cuda.set_device(dev_idx)
+    device = torch.device("cuda", dev_idx)
+    enable_tuned_gemms()
+    os.environ.setdefault("TREC_RELU_COLSUM", "1")
+    if not dist.is_initialized():
+        dist.init_process_group("nccl")
+    env = ShardingEnv.from_process_group(dist.group.WORLD)
+    emb_precision = os.environ.get("TREC_EMB_PRECISION", "fp32")
+    emb_out = os.environ.get("TREC_EMB_OUT", "bf16")
+    model = build_model(scale, emb_precision)
+    fused_params = {
+        "optimizer": "rowwise_adagrad",
+        "learning_rate": LR,
+        "fixed_bag_length": IDS_PER_FEATURE,
+        "output_dtype": emb_out,
+    }
+    sharder = EmbeddingBagCollectionSharder(fused_params=fused_params)
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(world_size=world, compute_device="cuda", batch_size=batch_size)
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, env=env, plan=plan, sharders=[sharder], device=device,
+        init_data_parallel=False,  # dense grads all-reduce INSIDE the graph
+    )
+    _dense_to_bf16(dmp)
+    dense_params = [p for p in dmp.parameters() if p.requires_grad]
+    dense_opt = torch.optim.SGD(dense_params, lr=LR, foreach=True)
+    if rank == 0:
+        counts = {}
+        for mplan in plan.plan.values():
+            for ps in mplan.values():
+                counts[ps.sharding_type] = counts.get(ps.sharding_type, 0) + 1
+        print(f"# dist-graph plan sharding mix: {counts}", flush=True)
+
+    host_batches = make_host_batches(8, batch_size, scale, seed=1234 + rank, pin=True)
+    host_labels_f = [hb.labels.float().pin_memory() for hb in host_batches]
+    b0 = host_batches[0].to(device)
+    static_values = b0.sparse_features.values().clone()
+    static_dense = b0.dense_features.clone()
+    static_labels = b0.labels.clone().float()
+    static_kjt = KeyedJaggedTensor(
+        keys=b0.sparse_features.keys(),
+        values=static_values,
+        lengths=b0.sparse_features.lengths().clone(),
+        stride=batch_size,
+    )
+    static_kjt.sync()
+    static_batch = Batch(static_dense, static_kjt, static_labels)
+
+    def one_step():
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            loss, _ = dmp(static_batch)
+        dense_opt.zero_grad(set_to_none=True)
+        loss.backward()
+        if world > 1:
+            grads = [p.grad for p in dense_params if p.grad is not None]
+            flat = torch.cat([g.reshape(-1) for g in grads])
+            dist.all_reduce(flat, op=dist.ReduceOp.AVG)
+            off = 0
+            for g in grads:
+                g.copy_(flat[off : off + g.numel()].view_as(g))
+                off += g.numel()
+        dense_opt.step()
+        return loss
+
+    # eager warmup on a side stream: RCCL communicators come up, the KJT a2a
+    # splits caches fill (first exchange), allocator pools stabilize
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            one_step()
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+    dist.barrier()
+    # static splits ON for capture: the cached-splits path is the one with no
+    # device->host sync, which stream capture requires
+    set_static_kjt_splits(True)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        static_loss = one_step()
+
+    copy_stream = torch.cuda.Stream()
+    stage = [
+        (
+            torch.empty_like(static_values),
+            torch.empty_like(static_dense),
+            torch.empty_like(static_labels),
+        )
+        for _ in range(2)
+    ]
+    ev_staged = [torch.cuda.Event() for _ in range(2)]
+    ev_consumed = [torch.cuda.Event() for _ in range(2)]
+    for ev in ev_consumed:
+        ev.record()
+
+    def preload(i: int) -> None:
+        slot = i % 2
+        hb = host_batches[i % len(host_batches)]
+        with torch.cuda.stream(copy_stream):
+            copy_stream.wait_event(ev_consumed[slot])
+            sv, sd, sl = stage[slot]
+            sv.copy_(hb.sparse_features.values(), non_blocking=True)
+            sd.copy_(hb.dense_features, non_blocking=True)
+            sl.copy_(host_labels_f[i % len(host_batches)], non_blocking=True)
+            ev_staged[slot].record(copy_stream)
+
+    def commit(i: int) -> None:
+        slot = i % 2
+        cur = torch.cuda.current_stream()
+        cur.wait_event(ev_staged[slot])
+        sv, sd, sl = stage[slot]
+        static_values.copy_(sv, non_blocking=True)
+        static_dense.copy_(sd, non_blocking=True)
+        static_labels.copy_(sl, non_blocking=True)
+        ev_consumed[slot].record(cur)
+
+    preload(0)
+    for i in range(warmup):
+        commit(i)
+        preload(i + 1)
+        g.replay()
+    torch.cuda.synchronize()
+    assert torch.isfinite(static_loss).all(), "non-finite loss in captured dist step"
+    dist.barrier()
+    torch.cuda.synchronize()
+    step_t = []
+    t0 = time.perf_counter()
+    tprev = t0
+    for i in range(steps):
+        commit(warmup + i)
+        preload(warmup + i + 1)
+        g.replay()
+        tnow = time.perf_counter()
+        step_t.append(tnow - tprev)
+        tprev = tnow
+    torch.cuda.synchronize()
+    dist.barrier()
+    dt = time.perf_counter() - t0
+    t = torch.tensor([dt], device=device)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    dt = float(t.item())
+    ms_per_step = dt / steps * 1e3
+    result = {
+        "metric": "samples/sec (whole node) DLRM Criteo-TB synthetic",
+        "value": batch_size * world * steps / dt,
+        "unit": "samples/s",
+        "n_gpus": world,
+        "steps": steps,
+        "warmup": warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic (random ids/dense/labels, Criteo-TB shapes)",
+        "config": {
+            "model": "DLRM (dot interaction, fused rowwise-Adagrad HIP TBE)",
+            "global_batch": batch_size * world,
+            "local_batch": batch_size,
+            "tables": len(DLRM_EMB_ROWS),
+            "embedding_dim": EMB_DIM,
+            "row_scale": scale,
+            "emb_dtype": emb_precision,
+            "emb_out_dtype": emb_out,
+            "dense_dtype": "bf16-autocast",
+            "parallelism": f"planner/dmp x{world} + hipGraph step capture (RCCL in-graph)",
+        },
+    }
+    if rank == 0:
+        st = sorted(step_t)
+        pct = lambda q: st[min(len(st) - 1, int(q * len(st)))] * 1e3  # noqa: E731
+        print(
+            f"# step-ms p10={pct(0.10):.3f} p50={pct(0.50):.3f} "
+            f"p90={pct(0.90):.3f} p99={pct(0.99):.3f} mean={ms_per_step:.3f}",
+            flush=True,
+        )
+        print(json.dumps(result))
+
+
 def run_smoke() -> None:
     """One tiny forward+backward of the flagship on cuda:0 (driver contract)."""
     from torchrec_amd.distributed.model_parallel import DistributedModelParallel
@@ -608,6 +815,18 @@ if __name__ == "__main__":
             and not args.smoke
         )
     )
+    # multi-rank (or forced-dist rehearsal) on GPU: capture the whole step —
+    # RCCL collectives included — unless opted out; fall back to the eager
+    # pipeline if capture fails
+    want_dist_graph = (
+        (int(os.environ.get("WORLD_SIZE", "1")) > 1
+         or os.environ.get("TREC_FORCE_DIST") == "1")
+        and os.environ.get("TREC_BENCH_CPU") != "1"
+        and os.environ.get("TREC_NO_HIPGRAPH") != "1"
+        and os.environ.get("TREC_DIST_GRAPH", "1") == "1"
+        and args.qcomm == "none"
+        and not args.smoke
+    )
     if args.smoke:
         run_smoke()
     elif want_graph and int(os.environ.get("WORLD_SIZE", "1")) == 1:
@@ -615,6 +834,17 @@ if __name__ == "__main__":
             run_graph_bench(args.steps, args.warmup, args.batch_size, args.scale)
         except Exception as exc:  # pragma: no cover — capture fallback
             print(f"# hipGraph capture failed ({exc!r}); eager pipeline fallback",
+                  flush=True)
+            run_bench(args.gpus, args.steps, args.warmup, args.batch_size,
+                      args.scale, qcomm=args.qcomm)
+    elif want_dist_graph:
+        try:
+            run_dist_graph_bench(args.steps, args.warmup, args.batch_size, args.scale)
+        except Exception as exc:  # pragma: no cover — capture fallback
+            from torchrec_amd.distributed.dist_data import set_static_kjt_splits
+
+            set_static_kjt_splits(False)
+            print(f"# dist hipGraph capture failed ({exc!r}); eager pipeline fallback",
                   flush=True)
             run_bench(args.gpus, args.steps, args.warmup, args.batch_size,
                       args.scale, qcomm=args.qcomm)

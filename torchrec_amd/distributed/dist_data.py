@@ -24,6 +24,23 @@ from torchrec_amd.distributed.comm_ops import (
 from torchrec_amd.distributed.types import Awaitable, LazyAwaitable, NoWait
 from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
 
+# static-shape mode: callers with fixed batch shapes (synthetic benches,
+# hipGraph-captured steps) promise every KJT a2a moves the same element
+# counts each step, letting the a2a reuse its first-exchange splits and run
+# with no device->host sync (a hard requirement for stream capture)
+_STATIC_KJT_SPLITS = False
+
+
+def set_static_kjt_splits(enabled: bool) -> None:
+    global _STATIC_KJT_SPLITS
+    _STATIC_KJT_SPLITS = enabled
+
+
+def static_kjt_splits_enabled() -> bool:
+    import os
+
+    return _STATIC_KJT_SPLITS or os.environ.get("TREC_STATIC_KJT_SPLITS") == "1"
+
 
 class KJTAllToAllTensorsAwaitable(Awaitable[KeyedJaggedTensor]):
     """Phase 2: async lengths/values/weights a2a -> recat -> KJT.
@@ -161,6 +178,7 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
         keys: List[str],  # my features post-exchange
         stagger: int = 1,
         rank_order: Optional[List[int]] = None,
+        splits_cache: Optional[object] = None,
     ) -> None:
         super().__init__()
         self._pg = pg
@@ -176,6 +194,20 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
         # per-dest-rank value counts (device, async)
         lengths = input.lengths()
         self._vbe = input.variable_stride_per_key()
+        # static-shape fast path (hipGraph capture): when the caller promises
+        # fixed splits (static_kjt_splits_enabled), reuse the host splits the
+        # FIRST exchange produced — no splits a2a, no device->host sync, so
+        # the whole a2a chain becomes stream-capturable. VBE stays dynamic.
+        self._splits_cache = splits_cache if not self._vbe else None
+        self._cached = None
+        if (
+            self._splits_cache is not None
+            and static_kjt_splits_enabled()
+            and getattr(self._splits_cache, "_cached_value_splits", None) is not None
+        ):
+            self._cached = self._splits_cache._cached_value_splits
+            self._validate = False
+            return
         if self._vbe:
             # VBE: per-key strides differ; key k occupies stride_per_key[k]
             # lengths entries (reference: variable-batch KJT a2a)
@@ -245,19 +277,27 @@ class KJTAllToAllSplitsAwaitable(Awaitable[KJTAllToAllTensorsAwaitable]):
         )
 
     def _wait_impl(self) -> KJTAllToAllTensorsAwaitable:
-        self._splits_work.wait()
-        if getattr(self, "_vbe", False):
+        if self._cached is not None:
+            in_value_splits, out_value_splits = self._cached
+        elif getattr(self, "_vbe", False):
+            self._splits_work.wait()
             return self._wait_impl_vbe()
-        if self._validate:
+        elif self._validate:
             from torchrec_amd.distributed.collective_utils import verify_tags
 
+            self._splits_work.wait()
             recv = self._out_splits_t.view(self._W, 2).cpu()
             verify_tags(recv[:, 1], self._tag, "kjt_a2a", self._pg)
             in_value_splits = self._in_splits_t.view(self._W, 2)[:, 0].cpu().tolist()
             out_value_splits = recv[:, 0].tolist()
         else:
+            self._splits_work.wait()
             in_value_splits = self._in_splits_t.cpu().tolist()  # sync (small)
             out_value_splits = self._out_splits_t.cpu().tolist()
+        if self._splits_cache is not None and self._cached is None:
+            self._splits_cache._cached_value_splits = (
+                list(in_value_splits), list(out_value_splits)
+            )
         kjt = self._input
         B = self._B
         W = self._W
@@ -389,12 +429,18 @@ class KJTAllToAll(nn.Module):
         splits: List[int],
         stagger: int = 1,
         rank_order: Optional[List[int]] = None,
+        allow_static: bool = False,
     ) -> None:
         super().__init__()
         self._pg = pg
         self._splits = splits
         self._stagger = stagger
         self._rank_order = rank_order
+        # static-splits caching is only sound when per-feature element counts
+        # are shape-determined (plain feature a2a, e.g. TW/CW). Bucketized
+        # inputs (RW/TWRW/MC) have DATA-dependent value splits and must keep
+        # the per-step exchange even in static mode.
+        self._allow_static = allow_static
         self._splits_cumsum = [0]
         for s in splits:
             self._splits_cumsum.append(self._splits_cumsum[-1] + s)
@@ -405,7 +451,8 @@ class KJTAllToAll(nn.Module):
             self._splits_cumsum[rank] : self._splits_cumsum[rank + 1]
         ]
         return KJTAllToAllSplitsAwaitable(
-            self._pg, input, self._splits, local_keys, self._stagger, self._rank_order
+            self._pg, input, self._splits, local_keys, self._stagger, self._rank_order,
+            splits_cache=self if self._allow_static else None,
         )
 
 

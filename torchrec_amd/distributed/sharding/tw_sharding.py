@@ -29,7 +29,7 @@ from torchrec_amd.sparse.jagged_tensor import KeyedJaggedTensor
 class TwSparseFeaturesDist(BaseSparseFeaturesDist):
     def __init__(self, pg, features_per_rank: List[int]) -> None:
         super().__init__()
-        self._a2a = KJTAllToAll(pg, splits=features_per_rank)
+        self._a2a = KJTAllToAll(pg, splits=features_per_rank, allow_static=True)
 
     def forward(self, sparse_features: KeyedJaggedTensor):
         return self._a2a(sparse_features)
