@@ -649,9 +649,21 @@ def run_dist_graph_bench(
     # static splits ON for capture: the cached-splits path is the one with no
     # device->host sync, which stream capture requires
     set_static_kjt_splits(True)
-    g = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(g):
-        static_loss = one_step()
+    static_loss = None
+    try:
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            static_loss = one_step()
+        ok = torch.ones(1, device=device)
+    except Exception as exc:
+        print(f"# rank {rank}: dist capture failed locally: {exc!r}", flush=True)
+        ok = torch.zeros(1, device=device)
+    # capture agreement: if ANY rank failed to capture, every rank falls back
+    # together (a lone faller would deadlock its peers' collectives)
+    dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+    if float(ok.item()) < 1.0:
+        set_static_kjt_splits(False)
+        raise RuntimeError("dist hipGraph capture failed on at least one rank")
 
     copy_stream = torch.cuda.Stream()
     stage = [
