@@ -64,7 +64,7 @@ def main():
             log.append({"step": step + 1, "loss": round(float(ld), 4),
                         "window_ne": round(r_ne, 4), "window_auc": round(r_auc, 4)})
             print(json.dumps(log[-1]), flush=True)
-    with open("gpurun_out/convergence_r01.json", "w") as f:
+    with open("gpurun_out/convergence_r02b.json", "w") as f:
         json.dump(log, f, indent=1)
 
 if __name__ == "__main__":

@@ -872,24 +872,78 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> seg_sort_pairs_large(
   return {k32_out.to(at::kLong), perm, overflow};
 }
 
-__global__ void mark_runs_kernel(const int64_t* __restrict__ sorted, int64_t n,
-                                 int32_t* __restrict__ flags) {
-  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; i < n;
-       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    flags[i] = (i == 0) || (sorted[i] != sorted[i - 1]);
-  }
+// ---------------------------------------------------------------------------
+// single-kernel run detection and chunk planning (decoupled-lookback scans).
+// The mark -> rocprim scan -> scatter pipeline was 10 launch-floor kernels
+// (~50 us inside the captured step); each phase is now ONE kernel: compute
+// the scan input inline, block-scan, decoupled lookback for the tile prefix
+// (rocprim pattern: publish aggregate, spin-read predecessors, publish
+// inclusive prefix), scatter the results. Tile ids come from a global
+// counter so tiles start in issue order (no deadlock).
+// ---------------------------------------------------------------------------
+
+namespace {
+constexpr int kScanItems = 8;
+constexpr int kScanTile = kBlockThreads * kScanItems;
+
+__device__ __forceinline__ void scan_publish(uint64_t* st, int tile, int32_t val,
+                                             uint64_t status) {
+  __hip_atomic_store(&st[tile], (status << 32) | (uint32_t)val, __ATOMIC_RELEASE,
+                     __HIP_MEMORY_SCOPE_AGENT);
 }
 
-__global__ void write_seg_offsets_kernel(const int32_t* __restrict__ flags,
-                                         const int32_t* __restrict__ run_ids /* inclusive */,
-                                         int64_t n, int32_t* __restrict__ seg_offsets,
-                                         int32_t* __restrict__ num_runs) {
-  for (int64_t i = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; i < n;
-       i += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    if (flags[i]) seg_offsets[run_ids[i] - 1] = static_cast<int32_t>(i);
-    if (i == n - 1) {
-      seg_offsets[run_ids[i]] = static_cast<int32_t>(n);
-      *num_runs = run_ids[i];
+__device__ __forceinline__ int32_t scan_lookback(const uint64_t* st, int tile) {
+  int32_t sum = 0;
+  for (int t = tile - 1; t >= 0; --t) {
+    uint64_t p;
+    do {
+      p = __hip_atomic_load(&st[t], __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+    } while ((p >> 32) == 0);
+    sum += (int32_t)(uint32_t)p;
+    if ((p >> 32) == 2) break;
+  }
+  return sum;
+}
+}  // namespace
+
+__global__ void __launch_bounds__(kBlockThreads) prep_runs_lookback_kernel(
+    const int64_t* __restrict__ sorted, int64_t n, int32_t* __restrict__ seg_offsets,
+    int32_t* __restrict__ num_runs, uint64_t* __restrict__ tile_state,
+    int* __restrict__ tile_counter) {
+  using BlockScan = hipcub::BlockScan<int32_t, kBlockThreads>;
+  __shared__ typename BlockScan::TempStorage temp;
+  __shared__ int s_tile;
+  __shared__ int32_t s_prefix;
+  if (threadIdx.x == 0) s_tile = atomicAdd(tile_counter, 1);
+  __syncthreads();
+  const int tile = s_tile;
+  const int64_t base = static_cast<int64_t>(tile) * kScanTile;
+  int32_t flags[kScanItems];
+  int32_t incl[kScanItems];
+#pragma unroll
+  for (int v = 0; v < kScanItems; ++v) {
+    int64_t i = base + static_cast<int64_t>(threadIdx.x) * kScanItems + v;
+    flags[v] = (i < n) ? ((i == 0) || (sorted[i] != sorted[i - 1])) : 0;
+  }
+  int32_t agg;
+  BlockScan(temp).InclusiveSum(flags, incl, agg);
+  if (threadIdx.x == 0) {
+    scan_publish(tile_state, tile, agg, 1);
+    s_prefix = scan_lookback(tile_state, tile);
+    scan_publish(tile_state, tile, s_prefix + agg, 2);
+  }
+  __syncthreads();
+  const int32_t prefix = s_prefix;
+#pragma unroll
+  for (int v = 0; v < kScanItems; ++v) {
+    int64_t i = base + static_cast<int64_t>(threadIdx.x) * kScanItems + v;
+    if (i < n) {
+      int32_t run_id = prefix + incl[v];  // inclusive count of run starts
+      if (flags[v]) seg_offsets[run_id - 1] = static_cast<int32_t>(i);
+      if (i == n - 1) {
+        seg_offsets[run_id] = static_cast<int32_t>(n);
+        *num_runs = run_id;
+      }
     }
   }
 }
@@ -898,23 +952,17 @@ std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_li
   int64_t n = sorted_linear.numel();
   auto opts = sorted_linear.options().dtype(at::kInt);
   auto seg_offsets = at::empty({n + 1}, opts);
-  auto num_runs = at::zeros({1}, opts);
-  if (n == 0) return {seg_offsets, num_runs};
-  auto flags = at::empty({n}, opts);
-  auto run_ids = at::empty({n}, opts);
+  auto num_runs = at::empty({1}, opts);
+  if (n == 0) return {seg_offsets, num_runs.zero_()};
   auto stream = tbe_stream();
-  int grid = grid_for(n, kBlockThreads);
-  hipLaunchKernelGGL(mark_runs_kernel, dim3(grid), dim3(kBlockThreads), 0, stream,
-                     sorted_linear.data_ptr<int64_t>(), n, flags.data_ptr<int32_t>());
-  size_t tmp_bytes = 0;
-  hipcub::DeviceScan::InclusiveSum(nullptr, tmp_bytes, flags.data_ptr<int32_t>(),
-                                   run_ids.data_ptr<int32_t>(), n, stream);
-  auto tmp = at::empty({static_cast<int64_t>(tmp_bytes)}, opts.dtype(at::kByte));
-  hipcub::DeviceScan::InclusiveSum(tmp.data_ptr(), tmp_bytes, flags.data_ptr<int32_t>(),
-                                   run_ids.data_ptr<int32_t>(), n, stream);
-  hipLaunchKernelGGL(write_seg_offsets_kernel, dim3(grid), dim3(kBlockThreads), 0, stream,
-                     flags.data_ptr<int32_t>(), run_ids.data_ptr<int32_t>(), n,
-                     seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>());
+  int64_t tiles = (n + kScanTile - 1) / kScanTile;
+  // workspace: [tiles] packed lookback states + 1 tile counter, zeroed once
+  auto ws = at::zeros({tiles + 1}, sorted_linear.options());  // int64 slots
+  hipLaunchKernelGGL(prep_runs_lookback_kernel, dim3((int)tiles), dim3(kBlockThreads),
+                     0, stream, sorted_linear.data_ptr<int64_t>(), n,
+                     seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),
+                     reinterpret_cast<uint64_t*>(ws.data_ptr<int64_t>()),
+                     reinterpret_cast<int*>(ws.data_ptr<int64_t>() + tiles));
   return {seg_offsets, num_runs};
 }
 
@@ -930,25 +978,50 @@ std::tuple<at::Tensor, at::Tensor> tbe_backward_prep(const at::Tensor& sorted_li
 // :787/:1292) redesigned for wave-slot scheduling.
 // ---------------------------------------------------------------------------
 
-__global__ void chunk_prep_kernel(const int32_t* __restrict__ seg_offsets,
-                                  const int32_t* __restrict__ num_runs_ptr, int64_t n,
-                                  int chunk_size, int32_t* __restrict__ nchunks) {
-  int32_t num_runs = *num_runs_ptr;
-  for (int64_t r = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x; r < n;
-       r += static_cast<int64_t>(gridDim.x) * blockDim.x) {
-    int32_t nc = 0;
+__global__ void __launch_bounds__(kBlockThreads) prep_chunks_lookback_kernel(
+    const int32_t* __restrict__ seg_offsets, const int32_t* __restrict__ num_runs_ptr,
+    int64_t n, int chunk_size, int32_t* __restrict__ chunk_offsets,
+    int32_t* __restrict__ total, uint64_t* __restrict__ tile_state,
+    int* __restrict__ tile_counter) {
+  using BlockScan = hipcub::BlockScan<int32_t, kBlockThreads>;
+  __shared__ typename BlockScan::TempStorage temp;
+  __shared__ int s_tile;
+  __shared__ int32_t s_prefix;
+  if (threadIdx.x == 0) s_tile = atomicAdd(tile_counter, 1);
+  __syncthreads();
+  const int tile = s_tile;
+  const int64_t base = static_cast<int64_t>(tile) * kScanTile;
+  const int32_t num_runs = *num_runs_ptr;
+  int32_t nc[kScanItems];
+  int32_t incl[kScanItems];
+#pragma unroll
+  for (int v = 0; v < kScanItems; ++v) {
+    int64_t r = base + static_cast<int64_t>(threadIdx.x) * kScanItems + v;
+    int32_t c = 0;
     if (r < num_runs) {
       int32_t len = seg_offsets[r + 1] - seg_offsets[r];
-      if (len > chunk_size) nc = (len + chunk_size - 1) / chunk_size;
+      if (len > chunk_size) c = (len + chunk_size - 1) / chunk_size;
     }
-    nchunks[r] = nc;
+    nc[v] = c;
   }
-}
-
-__global__ void total_chunks_kernel(const int32_t* __restrict__ chunk_offsets,
-                                    const int32_t* __restrict__ num_runs_ptr,
-                                    int32_t* __restrict__ total) {
-  *total = chunk_offsets[*num_runs_ptr];
+  int32_t agg;
+  BlockScan(temp).InclusiveSum(nc, incl, agg);
+  if (threadIdx.x == 0) {
+    scan_publish(tile_state, tile, agg, 1);
+    s_prefix = scan_lookback(tile_state, tile);
+    scan_publish(tile_state, tile, s_prefix + agg, 2);
+  }
+  __syncthreads();
+  const int32_t prefix = s_prefix;
+#pragma unroll
+  for (int v = 0; v < kScanItems; ++v) {
+    int64_t r = base + static_cast<int64_t>(threadIdx.x) * kScanItems + v;
+    if (r < n) {
+      chunk_offsets[r + 1] = prefix + incl[v];
+      if (r == 0) chunk_offsets[0] = 0;
+      if (r == num_runs - 1) *total = prefix + incl[v];
+    }
+  }
 }
 
 std::tuple<at::Tensor, at::Tensor> tbe_backward_chunk_prep(const at::Tensor& seg_offsets,
@@ -956,24 +1029,18 @@ std::tuple<at::Tensor, at::Tensor> tbe_backward_chunk_prep(const at::Tensor& seg
                                                            int64_t chunk_size) {
   int64_t n = seg_offsets.numel() - 1;  // max possible runs
   auto opts = seg_offsets.options();
-  auto chunk_offsets = at::zeros({n + 1}, opts);
-  auto total = at::zeros({1}, opts);
-  if (n == 0) return {chunk_offsets, total};
-  auto nchunks = at::empty({n}, opts);
+  auto chunk_offsets = at::empty({n + 1}, opts);
+  auto total = at::empty({1}, opts);  // writer guaranteed: n>0 => num_runs>=1
+  if (n == 0) return {chunk_offsets, total.zero_()};
   auto stream = tbe_stream();
-  int grid = grid_for(n, kBlockThreads);
-  hipLaunchKernelGGL(chunk_prep_kernel, dim3(grid), dim3(kBlockThreads), 0, stream,
-                     seg_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(), n,
-                     (int)chunk_size, nchunks.data_ptr<int32_t>());
-  size_t tmp_bytes = 0;
-  hipcub::DeviceScan::InclusiveSum(nullptr, tmp_bytes, nchunks.data_ptr<int32_t>(),
-                                   chunk_offsets.data_ptr<int32_t>() + 1, n, stream);
-  auto tmp = at::empty({static_cast<int64_t>(tmp_bytes)}, opts.dtype(at::kByte));
-  hipcub::DeviceScan::InclusiveSum(tmp.data_ptr(), tmp_bytes, nchunks.data_ptr<int32_t>(),
-                                   chunk_offsets.data_ptr<int32_t>() + 1, n, stream);
-  hipLaunchKernelGGL(total_chunks_kernel, dim3(1), dim3(1), 0, stream,
-                     chunk_offsets.data_ptr<int32_t>(), num_runs.data_ptr<int32_t>(),
-                     total.data_ptr<int32_t>());
+  int64_t tiles = (n + kScanTile - 1) / kScanTile;
+  auto ws = at::zeros({tiles + 1}, seg_offsets.options().dtype(at::kLong));
+  hipLaunchKernelGGL(prep_chunks_lookback_kernel, dim3((int)tiles), dim3(kBlockThreads),
+                     0, stream, seg_offsets.data_ptr<int32_t>(),
+                     num_runs.data_ptr<int32_t>(), n, (int)chunk_size,
+                     chunk_offsets.data_ptr<int32_t>(), total.data_ptr<int32_t>(),
+                     reinterpret_cast<uint64_t*>(ws.data_ptr<int64_t>()),
+                     reinterpret_cast<int*>(ws.data_ptr<int64_t>() + tiles));
   return {chunk_offsets, total};
 }
 
