@@ -200,14 +200,6 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kern
     io_t* ddrow = d_dense + b * D;
     io_t* dsrow = d_sparse + b * static_cast<int64_t>(F1 - 1) * D;
     const int zr = (l >> 4) * 4, zc = l & 15;
-    // phase 1: every wave's MFMA tiles accumulate in registers (<= 8 tiles
-    // per wave at D=256). Direct per-lane global stores measured
-    // store-wait-bound (SQ_WAIT_ANY 2.6M cycles vs the fwd's 0.5M: a wave
-    // covers only 32 contiguous bytes per output row), so the result is
-    // staged in LDS and written back 16 B per lane instead.
-    constexpr int kMaxTilesPerWave = 8;
-    f32x4 accs[kMaxTilesPerWave];
-    int ntile_mine = 0;
     for (int t = w; t < ntiles; t += kIWaves) {
       const int mi = t / (D / 16), ni = t - mi * (D / 16);
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -215,57 +207,17 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kern
       bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
           tt_tile + (ni * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-      accs[ntile_mine++] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
-    }
-    __syncthreads();  // all waves done READING g_tile/tt_tile
-    if constexpr (std::is_same_v<io_t, __hip_bfloat16>) {
-      // phase 2: scatter accs into the (now free) tt_tile region as a dense
-      // [F1, D] bf16 dT tile; the dense row folds in the copied-through grad
-      __bf16* dT = reinterpret_cast<__bf16*>(tt_tile);
-      ntile_mine = 0;
-      for (int t = w; t < ntiles; t += kIWaves) {
-        const int mi = t / (D / 16), ni = t - mi * (D / 16);
-        f32x4 acc = accs[ntile_mine++];
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int i = mi * 16 + zr + r;
-          int dcol = ni * 16 + zc;
-          if (i >= F1) continue;
-          float v = acc[r];
-          if (i == 0) v += emb2float(grow[dcol]);
-          dT[i * D + dcol] = static_cast<__bf16>(v);
-        }
-      }
-      __syncthreads();
-      // phase 3: vectorized write-back, 8 bf16 (16 B) per lane, rows contiguous
-      const int vecs = (F1 * D) / 8;
-      const bf16x8* dTv = reinterpret_cast<const bf16x8*>(dT);
-      for (int e = threadIdx.x; e < vecs; e += blockDim.x) {
-        int i = (e * 8) / D;
-        int dcol = (e * 8) - i * D;
-        bf16x8 pack = dTv[e];
-        io_t* dst = (i == 0) ? (ddrow + dcol)
-                             : (dsrow + static_cast<int64_t>(i - 1) * D + dcol);
-        *reinterpret_cast<bf16x8*>(dst) = pack;
-      }
-    } else {
-      // fp32 io: the 4 B/lane direct stores already cover 64 B per row-quad;
-      // an fp32 LDS staging tile would not fit in the Tt region
-      ntile_mine = 0;
-      for (int t = w; t < ntiles; t += kIWaves) {
-        const int mi = t / (D / 16), ni = t - mi * (D / 16);
-        f32x4 acc = accs[ntile_mine++];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int i = mi * 16 + zr + r;
-          int dcol = ni * 16 + zc;
-          if (i >= F1) continue;
-          float v = acc[r];
-          if (i == 0) {
-            ddrow[dcol] = float2emb(v + emb2float(grow[dcol]), io_t{});
-          } else {
-            dsrow[static_cast<int64_t>(i - 1) * D + dcol] = float2emb(v, io_t{});
-          }
+      for (int r = 0; r < 4; ++r) {
+        int i = mi * 16 + zr + r;     // T row (0 = dense)
+        int dcol = ni * 16 + zc;      // dim column
+        if (i >= F1) continue;
+        float v = acc[r];
+        if (i == 0) {
+          ddrow[dcol] = float2emb(v + emb2float(grow[dcol]), io_t{});
+        } else {
+          dsrow[static_cast<int64_t>(i - 1) * D + dcol] = float2emb(v, io_t{});
         }
       }
     }
