@@ -175,3 +175,50 @@ class TestInferenceSchema:
         from torchrec_amd.quant.embedding_modules import EmbeddingBagCollection
 
         assert hasattr(EmbeddingBagCollection, "from_float")
+
+
+class TestCheckpointSchema:
+    def test_dcp_helpers(self):
+        from torchrec_amd.distributed.checkpoint import (
+            load_checkpoint,
+            save_checkpoint,
+            state_dict_for_checkpoint,
+        )
+
+        assert_sig_prefix(save_checkpoint, ["model", "path"])
+        assert_sig_prefix(load_checkpoint, ["model", "path"])
+        assert_sig_prefix(state_dict_for_checkpoint, ["model"])
+
+
+class TestPsTransportSchema:
+    def test_ps_io(self):
+        from torchrec_amd.dynamic_embedding.ps import (
+            ParameterServer,
+            PSIO,
+            get_ps_io,
+            register_ps_io,
+        )
+        from torchrec_amd.dynamic_embedding.ps_net import PSNetServer, TcpPSIO
+
+        assert_sig_prefix(ParameterServer.__init__, ["dims", "io"])
+        assert_sig_prefix(get_ps_io, ["name", "dim"])
+        assert_sig_prefix(TcpPSIO.__init__, ["dim", "address"])
+        assert hasattr(PSNetServer, "address") and hasattr(PSNetServer, "close")
+        assert hasattr(PSIO, "push") and hasattr(PSIO, "pull")
+
+
+class TestStaticSplitsSchema:
+    def test_dist_data(self):
+        from torchrec_amd.distributed.dist_data import (
+            KJTAllToAll,
+            set_static_kjt_splits,
+            static_kjt_splits_enabled,
+        )
+
+        assert_sig_prefix(KJTAllToAll.__init__, ["pg", "splits"])
+        assert "allow_static" in [
+            p.name
+            for p in __import__("inspect").signature(KJTAllToAll.__init__).parameters.values()
+        ]
+        set_static_kjt_splits(False)
+        assert static_kjt_splits_enabled() in (True, False)
