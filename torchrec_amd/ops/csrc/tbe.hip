@@ -23,6 +23,9 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include <algorithm>
+#include <cstdlib>
 #include <hipcub/hipcub.hpp>
 #include <rocprim/block/block_radix_sort.hpp>
 #include <rocprim/device/device_segmented_radix_sort.hpp>
@@ -1448,7 +1451,13 @@ void tbe_backward_fused(
   auto stream = tbe_stream();
   int lps = (max_D <= 64) ? 16 : (max_D <= 128 ? 32 : 64);
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
-  int grid = grid_for(n * lps, kBlockThreads);  // upper bound: runs <= n
+  // full grid (one slot per run): the kernel is latency-chain bound, so a
+  // grid-strided slot serializing ~3 runs costs ~3 dependent-chain lengths;
+  // excess blocks just queue behind retiring ones (TREC_BWD_GRID_CAP to cap)
+  int64_t grid_cap = 131072;
+  if (const char* gc = std::getenv("TREC_BWD_GRID_CAP")) grid_cap = std::atoll(gc);
+  int grid = (int)std::min<int64_t>(
+      (n * lps + kBlockThreads - 1) / kBlockThreads, std::max<int64_t>(grid_cap, 1));
   const float* scale_ptr = pos_scale.numel() > 0 ? pos_scale.data_ptr<float>() : nullptr;
   float* cache_w_ptr = cache_weights.numel() > 0 ? cache_weights.data_ptr<float>() : nullptr;
   const int32_t* cache_loc_ptr =
