@@ -17,6 +17,11 @@ from torchrec_amd.distributed.types import (  # noqa: F401
     ShardingPlan,
     ShardingType,
 )
+from torchrec_amd.distributed.checkpoint import (  # noqa: F401
+    load_checkpoint,
+    save_checkpoint,
+    state_dict_for_checkpoint,
+)
 from torchrec_amd.distributed.embeddingbag import (  # noqa: F401
     EmbeddingBagCollectionSharder,
     ShardedEmbeddingBagCollection,
