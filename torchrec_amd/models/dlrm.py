@@ -220,8 +220,11 @@ class DLRM(nn.Module):
         self.over_arch = OverArch(over_in, over_arch_layer_sizes, device=dense_device)
 
     def forward(self, dense_features: torch.Tensor, sparse_features: KeyedJaggedTensor) -> torch.Tensor:
-        embedded_dense = self.dense_arch(dense_features)
+        # sparse FIRST: the sharded EBC's pooled-output all-to-all rides
+        # RCCL's comm stream and is waited lazily at the interaction, so the
+        # dense MLP overlaps the wire time at N>1 (free at N=1)
         embedded_sparse = self.sparse_arch(sparse_features)
+        embedded_dense = self.dense_arch(dense_features)
         concat = self.inter_arch(embedded_dense, embedded_sparse)
         return self.over_arch(concat)
 
@@ -252,8 +255,11 @@ class DLRM_DCN(nn.Module):
         self.over_arch = OverArch(cross_in, over_arch_layer_sizes, device=dense_device)
 
     def forward(self, dense_features: torch.Tensor, sparse_features: KeyedJaggedTensor) -> torch.Tensor:
-        embedded_dense = self.dense_arch(dense_features)
+        # sparse FIRST: the sharded EBC's pooled-output all-to-all rides
+        # RCCL's comm stream and is waited lazily at the interaction, so the
+        # dense MLP overlaps the wire time at N>1 (free at N=1)
         embedded_sparse = self.sparse_arch(sparse_features)
+        embedded_dense = self.dense_arch(dense_features)
         concat = self.inter_arch(embedded_dense, embedded_sparse)
         return self.over_arch(concat)
 
@@ -303,8 +309,11 @@ class DLRM_Projection(nn.Module):
     def forward(
         self, dense_features: torch.Tensor, sparse_features: KeyedJaggedTensor
     ) -> torch.Tensor:
-        embedded_dense = self.dense_arch(dense_features)
+        # sparse FIRST: the sharded EBC's pooled-output all-to-all rides
+        # RCCL's comm stream and is waited lazily at the interaction, so the
+        # dense MLP overlaps the wire time at N>1 (free at N=1)
         embedded_sparse = self.sparse_arch(sparse_features)
+        embedded_dense = self.dense_arch(dense_features)
         concat = self.inter_arch(embedded_dense, embedded_sparse)
         return self.over_arch(concat)
 
