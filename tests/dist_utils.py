@@ -42,7 +42,8 @@ def run_multi_process(
     fn: Callable, world_size: int = 2, backend: str = "gloo", *args: Any, **kwargs: Any
 ) -> None:
     last: Exception = RuntimeError("unreachable")
-    for attempt in range(2):
+    attempts = 3
+    for attempt in range(attempts):
         port = _free_port()
         try:
             mp.start_processes(
@@ -57,9 +58,10 @@ def run_multi_process(
             msg = str(e)
             rendezvous_race = any(
                 t in msg
-                for t in ("Address already in use", "EADDRINUSE", "Connection refused")
+                for t in ("Address already in use", "EADDRINUSE", "Connection refused",
+                          "Connection reset")
             )
-            if attempt == 0 and rendezvous_race:
+            if attempt < attempts - 1 and rendezvous_race:
                 last = e
                 continue
             raise
