@@ -228,7 +228,13 @@ at::Tensor tbe_forward_pooled(
   // pick lanes-per-slot and register chunks from max_D
   int lps = (max_D <= 64) ? 16 : (max_D <= 128 ? 32 : 64);
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
-  int grid = grid_for(n_bags * lps, kBlockThreads);
+  // full grid (one slot per bag): like the backward, the per-bag gather is
+  // a short dependent chain, so serializing ~3 bags per grid-strided slot
+  // wastes chain latency (TREC_FWD_GRID_CAP to cap)
+  int64_t fwd_cap = 131072;
+  if (const char* gc = std::getenv("TREC_FWD_GRID_CAP")) fwd_cap = std::atoll(gc);
+  int grid = (int)std::min<int64_t>(
+      (n_bags * lps + kBlockThreads - 1) / kBlockThreads, std::max<int64_t>(fwd_cap, 1));
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, weights.scalar_type(),
                                   "tbe_fwd_pooled", [&] {
     using dev_t = typename DevType<scalar_t>::type;
