@@ -895,23 +895,45 @@ namespace {
 constexpr int kScanItems = 8;
 constexpr int kScanTile = kBlockThreads * kScanItems;
 
+// single-word (status<<32 | value) protocol: the packed word is both the
+// flag and the payload, so RELAXED atomics suffice (no ordering against
+// other locations) and probe loads pipeline freely.
 __device__ __forceinline__ void scan_publish(uint64_t* st, int tile, int32_t val,
                                              uint64_t status) {
-  __hip_atomic_store(&st[tile], (status << 32) | (uint32_t)val, __ATOMIC_RELEASE,
+  __hip_atomic_store(&st[tile], (status << 32) | (uint32_t)val, __ATOMIC_RELAXED,
                      __HIP_MEMORY_SCOPE_AGENT);
 }
 
-__device__ __forceinline__ int32_t scan_lookback(const uint64_t* st, int tile) {
-  int32_t sum = 0;
-  for (int t = tile - 1; t >= 0; --t) {
-    uint64_t p;
-    do {
-      p = __hip_atomic_load(&st[t], __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
-    } while ((p >> 32) == 0);
-    sum += (int32_t)(uint32_t)p;
-    if ((p >> 32) == 2) break;
+// wave-parallel decoupled lookback (rocprim pattern): 64 predecessor tiles
+// probed at once; a serial walk costs ~0.7 us per probed tile and dominated
+// the kernel at >100 tiles. All 64 lanes of wave 0 must call this; every
+// lane returns the exclusive prefix.
+__device__ __forceinline__ int32_t scan_lookback_wave(const uint64_t* st, int tile,
+                                                      int lane) {
+  int32_t total = 0;
+  int base = tile;  // probe tiles [base-64, base-1]
+  while (base > 0) {
+    int t = base - 64 + lane;
+    uint64_t p = ((uint64_t)2) << 32;  // lanes below tile 0 act as prefix 0
+    if (t >= 0) {
+      do {
+        p = __hip_atomic_load(&st[t], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      } while ((p >> 32) == 0);
+    }
+    uint64_t pref_mask = __ballot((p >> 32) == 2);
+    // highest lane holding an inclusive prefix; cut = -1 (sum ALL aggregates,
+    // keep walking) when no prefix is visible in this window
+    int cut = pref_mask ? (63 - __clzll(pref_mask)) : -1;
+    int32_t contrib = (lane >= cut) ? (int32_t)(uint32_t)p : 0;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      contrib += __shfl_down(contrib, off, kWaveSize);
+    contrib = __shfl(contrib, 0, kWaveSize);
+    total += contrib;
+    if (cut >= 0) return total;  // prefix found: done
+    base -= 64;
   }
-  return sum;
+  return total;
 }
 }  // namespace
 
@@ -936,10 +958,14 @@ __global__ void __launch_bounds__(kBlockThreads) prep_runs_lookback_kernel(
   }
   int32_t agg;
   BlockScan(temp).InclusiveSum(flags, incl, agg);
-  if (threadIdx.x == 0) {
-    scan_publish(tile_state, tile, agg, 1);
-    s_prefix = scan_lookback(tile_state, tile);
-    scan_publish(tile_state, tile, s_prefix + agg, 2);
+  __syncthreads();  // broadcast agg via BlockScan done; wave 0 runs lookback
+  if (threadIdx.x < kWaveSize) {
+    if (threadIdx.x == 0) scan_publish(tile_state, tile, agg, 1);
+    int32_t pre = scan_lookback_wave(tile_state, tile, threadIdx.x);
+    if (threadIdx.x == 0) {
+      s_prefix = pre;
+      scan_publish(tile_state, tile, pre + agg, 2);
+    }
   }
   __syncthreads();
   const int32_t prefix = s_prefix;
@@ -1015,10 +1041,14 @@ __global__ void __launch_bounds__(kBlockThreads) prep_chunks_lookback_kernel(
   }
   int32_t agg;
   BlockScan(temp).InclusiveSum(nc, incl, agg);
-  if (threadIdx.x == 0) {
-    scan_publish(tile_state, tile, agg, 1);
-    s_prefix = scan_lookback(tile_state, tile);
-    scan_publish(tile_state, tile, s_prefix + agg, 2);
+  __syncthreads();
+  if (threadIdx.x < kWaveSize) {
+    if (threadIdx.x == 0) scan_publish(tile_state, tile, agg, 1);
+    int32_t pre = scan_lookback_wave(tile_state, tile, threadIdx.x);
+    if (threadIdx.x == 0) {
+      s_prefix = pre;
+      scan_publish(tile_state, tile, pre + agg, 2);
+    }
   }
   __syncthreads();
   const int32_t prefix = s_prefix;
