@@ -90,45 +90,22 @@ __global__ void __launch_bounds__(kBlockThreads) relu_bwd_colsum_partial_kernel(
   }
 }
 
-// single-kernel deterministic reduce of [G, N] -> [N] in the output dtype:
-// a 2-D grid ((column tile) x S) collapses G rows into S per-tile segment
-// sums; the LAST block to finish each column tile (device-scope semaphore,
-// the rocprim decoupled-lookback pattern) folds the S rows in FIXED order —
-// deterministic — and writes db, cast. Replaces three kernels (segment
-// reduce + final reduce + dtype cast) with one.
-template <typename o_t>
-__global__ void __launch_bounds__(kBlockThreads) colsum_finish_kernel(
+// level-1 segment reduce: 2-D grid ((column tile) x S) collapses G rows to
+// S per-tile sums. Measured better than a single-kernel semaphore finish
+// (21.8 us) as a pair with the cast-fused final (9 + 8.5 us): the last-block
+// fold's 32-row tail ran on too few blocks.
+__global__ void __launch_bounds__(kBlockThreads) colsum_seg_f32_kernel(
     const float* __restrict__ partial, int G, int seg_rows, int64_t N,
-    float* __restrict__ seg /* [S, N] */, int* __restrict__ counters,
-    int S, o_t* __restrict__ db) {
+    float* __restrict__ seg_out /* [S, N] */) {
   int64_t c = static_cast<int64_t>(blockIdx.x) * kBlockThreads + threadIdx.x;
+  if (c >= N) return;
   int s = blockIdx.y;
-  if (c < N) {
-    int g0 = s * seg_rows;
-    int g1 = min(G, g0 + seg_rows);
-    float acc = 0.f;
-    for (int grp = g0; grp < g1; ++grp)
-      acc += partial[static_cast<int64_t>(grp) * N + c];
-    seg[static_cast<int64_t>(s) * N + c] = acc;
-  }
-  // publish: plain stores + ONE device fence per thread, then the counter.
-  // (Per-element RELEASE stores / ACQUIRE loads serialize at the CU and cost
-  // ~0.7 us each — measured 30 us/kernel before this relaxation.)
-  __threadfence();
-  __syncthreads();
-  __shared__ int last;
-  if (threadIdx.x == 0)
-    last = (__hip_atomic_fetch_add(&counters[blockIdx.x], 1, __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT) == S - 1);
-  __syncthreads();
-  if (last && c < N) {
-    __threadfence();  // acquire side: predecessors' seg rows now visible
-    float tot = 0.f;
-    for (int ss = 0; ss < S; ++ss)
-      tot += __hip_atomic_load(&seg[static_cast<int64_t>(ss) * N + c],
-                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    db[c] = float2emb(tot, o_t{});
-  }
+  int g0 = s * seg_rows;
+  int g1 = min(G, g0 + seg_rows);
+  float acc = 0.f;
+  for (int grp = g0; grp < g1; ++grp)
+    acc += partial[static_cast<int64_t>(grp) * N + c];
+  seg_out[static_cast<int64_t>(s) * N + c] = acc;
 }
 
 template <typename o_t>
@@ -171,14 +148,6 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
       G = (int)((M + rows_per_group - 1) / rows_per_group);
       auto partial = at::empty({(int64_t)G * N}, dy.options().dtype(at::kFloat));
       int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);
-      bool finish = G > 64;
-      at::Tensor counters, seg;
-      int* cnt_ptr = nullptr;
-      if (finish) {
-        counters = at::empty({ftiles}, dy.options().dtype(at::kInt));
-        seg = at::empty({(int64_t)32 * N}, dy.options().dtype(at::kFloat));
-        cnt_ptr = counters.data_ptr<int>();
-      }
       int grid = grid_for(col_chunks * G, kBlockThreads);
       hipLaunchKernelGGL((relu_bwd_colsum_partial_kernel<dev_t, VPT>), dim3(grid),
                          dim3(kBlockThreads), 0, stream,
@@ -186,14 +155,17 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
                          reinterpret_cast<const dev_t*>(yc.data_ptr<scalar_t>()), M, N,
                          rows_per_group, G,
                          reinterpret_cast<dev_t*>(g.data_ptr<scalar_t>()),
-                         partial.data_ptr<float>(), cnt_ptr, ftiles);
-      if (finish) {
+                         partial.data_ptr<float>(), nullptr, 0);
+      if (G > 64) {
         constexpr int S = 32;
         int seg_rows = (G + S - 1) / S;
-        hipLaunchKernelGGL((colsum_finish_kernel<dev_t>), dim3(ftiles, S),
+        auto seg = at::empty({(int64_t)S * N}, dy.options().dtype(at::kFloat));
+        hipLaunchKernelGGL(colsum_seg_f32_kernel, dim3(ftiles, S),
                            dim3(kBlockThreads), 0, stream, partial.data_ptr<float>(),
-                           G, seg_rows, N, seg.data_ptr<float>(), cnt_ptr, S,
-                           reinterpret_cast<dev_t*>(db.data_ptr<scalar_t>()));
+                           G, seg_rows, N, seg.data_ptr<float>());
+        hipLaunchKernelGGL((colsum_final_cast_kernel<dev_t>), dim3(ftiles),
+                           dim3(kBlockThreads), 0, stream, seg.data_ptr<float>(),
+                           S, N, reinterpret_cast<dev_t*>(db.data_ptr<scalar_t>()));
       } else {
         hipLaunchKernelGGL((colsum_final_cast_kernel<dev_t>), dim3(ftiles),
                            dim3(kBlockThreads), 0, stream, partial.data_ptr<float>(),
