@@ -228,10 +228,11 @@ at::Tensor tbe_forward_pooled(
   // pick lanes-per-slot and register chunks from max_D
   int lps = (max_D <= 64) ? 16 : (max_D <= 128 ? 32 : 64);
   int chunks = (int)((max_D + lps * 4 - 1) / (lps * 4));
-  // full grid (one slot per bag): like the backward, the per-bag gather is
-  // a short dependent chain, so serializing ~3 bags per grid-strided slot
-  // wastes chain latency (TREC_FWD_GRID_CAP to cap)
-  int64_t fwd_cap = 131072;
+  // grid-strided (capped) launch: unlike the backward, the forward is
+  // BANDWIDTH-bound — same-box A/B showed the full one-slot-per-bag grid
+  // 4-10% SLOWER (55.1 vs 52.9 us fp32; block scheduling overhead with no
+  // latency chain to hide). TREC_FWD_GRID_CAP overrides.
+  int64_t fwd_cap = kMaxBlocks;
   if (const char* gc = std::getenv("TREC_FWD_GRID_CAP")) fwd_cap = std::atoll(gc);
   int grid = (int)std::min<int64_t>(
       (n_bags * lps + kBlockThreads - 1) / kBlockThreads, std::max<int64_t>(fwd_cap, 1));
