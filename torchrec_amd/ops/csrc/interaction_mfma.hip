@@ -137,8 +137,7 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_fwd_kern
   }
 }
 
-template <typename io_t, int kTPW>  // kTPW: compile-time tiles/wave cap
-                                    // (0 = direct-store path for fp32 io)
+template <typename io_t>
 __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kernel(
     const io_t* __restrict__ grad_out,  // [B, D+P]
     const io_t* __restrict__ dense, const io_t* __restrict__ sparse,
@@ -201,73 +200,24 @@ __global__ void __launch_bounds__(kIWaves * kWaveSize) interaction_mfma_bwd_kern
     io_t* ddrow = d_dense + b * D;
     io_t* dsrow = d_sparse + b * static_cast<int64_t>(F1 - 1) * D;
     const int zr = (l >> 4) * 4, zc = l & 15;
-    if constexpr (kTPW > 0) {
-      // bf16 io: accumulate this wave's tiles in (statically indexed)
-      // registers, stage the [F1, D] dT tile in the freed Tt LDS region,
-      // write back 16 B/lane. (A first attempt used a DYNAMICALLY indexed
-      // accumulator array — it spilled to scratch and ran 35% slower; the
-      // compile-time-unrolled retry below keeps everything in VGPRs.)
-      f32x4 accs[kTPW];
+    for (int t = w; t < ntiles; t += kIWaves) {
+      const int mi = t / (D / 16), ni = t - mi * (D / 16);
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          g_tile + (mi * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
+      bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+          tt_tile + (ni * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
 #pragma unroll
-      for (int it = 0; it < kTPW; ++it) {
-        int t = w + it * kIWaves;
-        if (t >= ntiles) break;
-        const int mi = t / (D / 16), ni = t - mi * (D / 16);
-        bf16x8 a = *reinterpret_cast<const bf16x8*>(
-            g_tile + (mi * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
-        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
-            tt_tile + (ni * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
-        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-        accs[it] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
-      }
-      __syncthreads();  // all waves done reading g_tile/tt_tile
-      __bf16* dT = reinterpret_cast<__bf16*>(tt_tile);
-#pragma unroll
-      for (int it = 0; it < kTPW; ++it) {
-        int t = w + it * kIWaves;
-        if (t >= ntiles) break;
-        const int mi = t / (D / 16), ni = t - mi * (D / 16);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int i = mi * 16 + zr + r;
-          int dcol = ni * 16 + zc;
-          if (i >= F1) continue;
-          float v = accs[it][r];
-          if (i == 0) v += emb2float(grow[dcol]);
-          dT[i * D + dcol] = static_cast<__bf16>(v);
-        }
-      }
-      __syncthreads();
-      const int vecs = (F1 * D) / 8;
-      const bf16x8* dTv = reinterpret_cast<const bf16x8*>(dT);
-      for (int e = threadIdx.x; e < vecs; e += blockDim.x) {
-        int i = (e * 8) / D;
-        int dcol = (e * 8) - i * D;
-        bf16x8 pack = dTv[e];
-        io_t* dst = (i == 0) ? (ddrow + dcol)
-                             : (dsrow + static_cast<int64_t>(i - 1) * D + dcol);
-        *reinterpret_cast<bf16x8*>(dst) = pack;
-      }
-    } else {
-      for (int t = w; t < ntiles; t += kIWaves) {
-        const int mi = t / (D / 16), ni = t - mi * (D / 16);
-        bf16x8 a = *reinterpret_cast<const bf16x8*>(
-            g_tile + (mi * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
-        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
-            tt_tile + (ni * 16 + (l & 15)) * (kStrideE * 2) + ((l >> 4) << 4));
-        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int i = mi * 16 + zr + r;     // T row (0 = dense)
-          int dcol = ni * 16 + zc;      // dim column
-          if (i >= F1) continue;
-          float v = acc[r];
-          if (i == 0) {
-            ddrow[dcol] = float2emb(v + emb2float(grow[dcol]), io_t{});
-          } else {
-            dsrow[static_cast<int64_t>(i - 1) * D + dcol] = float2emb(v, io_t{});
-          }
+      for (int r = 0; r < 4; ++r) {
+        int i = mi * 16 + zr + r;     // T row (0 = dense)
+        int dcol = ni * 16 + zc;      // dim column
+        if (i >= F1) continue;
+        float v = acc[r];
+        if (i == 0) {
+          ddrow[dcol] = float2emb(v + emb2float(grow[dcol]), io_t{});
+        } else {
+          dsrow[static_cast<int64_t>(i - 1) * D + dcol] = float2emb(v, io_t{});
         }
       }
     }
@@ -330,23 +280,14 @@ std::tuple<at::Tensor, at::Tensor> interaction_mfma_backward(
       TORCH_CHECK(false, "fp64 unsupported");
     } else {
       using dev_t = typename DevType<scalar_t>::type;
-      int ntiles_h = ((F1 + 15) / 16) * (D / 16);
-      int tpw = (ntiles_h + kIWaves - 1) / kIWaves;
-      auto launch = [&](auto tpw_tag) {
-        constexpr int TPW = decltype(tpw_tag)::value;
-        hipLaunchKernelGGL((interaction_mfma_bwd_kernel<dev_t, TPW>), dim3(grid),
-                           dim3(kIWaves * kWaveSize), lds_bytes, stream,
-                           reinterpret_cast<const dev_t*>(grad_out.contiguous().data_ptr<scalar_t>()),
-                           reinterpret_cast<const dev_t*>(dense.contiguous().data_ptr<scalar_t>()),
-                           reinterpret_cast<const dev_t*>(sparse.contiguous().data_ptr<scalar_t>()),
-                           pair_col.data_ptr<int32_t>(), B, F1, D, P,
-                           reinterpret_cast<dev_t*>(d_dense.data_ptr<scalar_t>()),
-                           reinterpret_cast<dev_t*>(d_sparse.data_ptr<scalar_t>()));
-      };
-      if (!std::is_same_v<dev_t, __hip_bfloat16>) launch(std::integral_constant<int, 0>{});
-      else if (tpw <= 2) launch(std::integral_constant<int, 2>{});
-      else if (tpw <= 4) launch(std::integral_constant<int, 4>{});
-      else launch(std::integral_constant<int, 8>{});
+      hipLaunchKernelGGL((interaction_mfma_bwd_kernel<dev_t>), dim3(grid),
+                         dim3(kIWaves * kWaveSize), lds_bytes, stream,
+                         reinterpret_cast<const dev_t*>(grad_out.contiguous().data_ptr<scalar_t>()),
+                         reinterpret_cast<const dev_t*>(dense.contiguous().data_ptr<scalar_t>()),
+                         reinterpret_cast<const dev_t*>(sparse.contiguous().data_ptr<scalar_t>()),
+                         pair_col.data_ptr<int32_t>(), B, F1, D, P,
+                         reinterpret_cast<dev_t*>(d_dense.data_ptr<scalar_t>()),
+                         reinterpret_cast<dev_t*>(d_sparse.data_ptr<scalar_t>()));
     }
   });
   return {d_dense, d_sparse};
