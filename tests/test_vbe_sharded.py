@@ -523,3 +523,70 @@ def test_vbe_weighted_tw():
 
 def test_vbe_weighted_rw():
     run_multi_process(_run_vbe_weighted, 2, "gloo", ShardingType.ROW_WISE.value)
+
+
+def _run_vbe_twrw_featureless_node(rank, world_size):
+    """TWRW VBE with ONE table: the non-owning node's ranks are featureless
+    and must still drive the cross a2a (stage-1 RS-v skipped group-wide)."""
+    import os
+
+    os.environ["LOCAL_WORLD_SIZE"] = "2"
+    torch.manual_seed(42)
+    one_table = [("t0", 40, 8, "f0")]
+    configs = [
+        EmbeddingBagConfig(num_embeddings=r, embedding_dim=d, name=n, feature_names=[f])
+        for (n, r, d, f) in one_table
+    ]
+    model = SparseModel(configs)
+    planner = EmbeddingShardingPlanner(
+        topology=Topology(
+            world_size=world_size, compute_device="cpu", hbm_cap=1 << 40,
+            local_world_size=2,
+        ),
+        constraints={
+            "t0": ParameterConstraints(
+                sharding_types=[ShardingType.TABLE_ROW_WISE.value]
+            )
+        },
+    )
+    sharder = EmbeddingBagCollectionSharder(
+        fused_params={"optimizer": "rowwise_adagrad", "learning_rate": LR}
+    )
+    plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    dmp = DistributedModelParallel(
+        model, plan=plan, sharders=[sharder], init_data_parallel=False
+    )
+    torch.manual_seed(42)
+    golden = FusedEmbeddingBagCollection(
+        configs, optimizer="rowwise_adagrad", learning_rate=LR
+    )
+    dmp.load_state_dict(
+        {"sparse.embedding_bags.t0.weight": golden.split_embedding_weights()[0]},
+        strict=False,
+    )
+    strides = {"f0": [2, 1, 3, 2]}
+    g = torch.Generator().manual_seed(11)
+    b_tot = sum(strides["f0"])
+    lengths = torch.randint(0, 4, (b_tot,), generator=g)
+    values = torch.randint(0, 40, (int(lengths.sum()),), generator=g)
+    kjt_global = KeyedJaggedTensor(
+        keys=["f0"], values=values, lengths=lengths,
+        stride_per_key_per_rank=[list(strides["f0"])],
+    )
+    kjt_local = _local_slice(kjt_global, rank)
+    kt = dmp(kjt_local)
+    vals = kt.values()
+    golden_vals = golden(kjt_global).values()
+    d = 8
+    block = golden_vals.view(b_tot, d)
+    r0 = sum(strides["f0"][:rank])
+    expected = block[r0 : r0 + strides["f0"][rank]].reshape(-1)
+    torch.testing.assert_close(vals, expected, atol=1e-5, rtol=1e-5)
+    vals.sum().backward()
+    golden_vals.sum().backward()
+
+
+def test_vbe_twrw_featureless_node():
+    run_multi_process(
+        _run_vbe_twrw_featureless_node, 4, "gloo"
+    )

@@ -581,7 +581,15 @@ class ShardedEmbeddingBagCollection(nn.Module):
             [sum(spr[f][l * NN : (l + 1) * NN]) for l in range(L)] for f in range(F)
         ]
         intra_pg, cross_pg = intra_and_cross_node_pg()
-        aw1 = reduce_scatter_v_per_feature_pooled(packed, spr_intra, dims, intra_pg)
+        if F == 0:
+            # featureless node: every intra peer is also featureless (tables
+            # assign per NODE), so the whole intra group skips stage 1; the
+            # cross a2a below still runs to receive this rank's bags
+            from torchrec_amd.distributed.types import NoWait
+
+            aw1 = NoWait(packed.reshape(0))
+        else:
+            aw1 = reduce_scatter_v_per_feature_pooled(packed, spr_intra, dims, intra_pg)
         if NN == 1:
             return (aw1, names_mine, dims, [0] * F)
         # stage 2: regroup the reduced feature-major pack node-major, a2a back
