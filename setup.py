@@ -24,9 +24,11 @@ ext = CUDAExtension(
         os.path.join(CSRC, "interaction.hip"),
         os.path.join(CSRC, "interaction_mfma.hip"),
         os.path.join(CSRC, "mlp_ops.hip"),
+        os.path.join(CSRC, "lt_gemm.hip"),
         os.path.join(CSRC, "cache.hip"),
         os.path.join(CSRC, "bindings.hip"),
     ],
+    libraries=["hipblaslt"],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
         "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],

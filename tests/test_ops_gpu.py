@@ -429,3 +429,69 @@ class TestFusedBce:
         a = ops.fused_bce_with_logits(logits, labels)
         b = ops.fused_bce_with_logits(logits, labels)
         assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+class TestLtEpilogues:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    def test_linear_relu_fwd(self, dtype):
+        ops.hip_ops()
+        torch.manual_seed(0)
+        M, K, N = 512, 96, 160
+        x = torch.randn(M, K, device="cuda").to(dtype)
+        w = torch.randn(N, K, device="cuda").to(dtype)
+        b = torch.randn(N, device="cuda").to(dtype)
+        y = torch.ops.trec_amd.lt_linear_relu_fwd(x, w, b)
+        ref = torch.relu(torch.nn.functional.linear(x.float(), w.float(), b.float()))
+        tol = 1e-4 if dtype == torch.float32 else 3e-2
+        torch.testing.assert_close(y.float(), ref, atol=tol, rtol=tol)
+
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    def test_wgrad_bgrad(self, dtype):
+        ops.hip_ops()
+        torch.manual_seed(1)
+        M, N, K = 4096, 128, 64
+        g = torch.randn(M, N, device="cuda").to(dtype)
+        x = torch.randn(M, K, device="cuda").to(dtype)
+        dW, db = torch.ops.trec_amd.lt_wgrad_bgrad(g, x)
+        refW = g.float().t() @ x.float()
+        refb = g.float().sum(0)
+        tol = 1e-3 if dtype == torch.float32 else 1.0
+        torch.testing.assert_close(dW.float(), refW, atol=tol, rtol=2e-2)
+        torch.testing.assert_close(db.float(), refb, atol=tol, rtol=2e-2)
+
+    def test_relu_bwd_mask(self):
+        ops.hip_ops()
+        y = torch.randn(1000, 257, device="cuda", dtype=torch.bfloat16).relu()
+        dy = torch.randn_like(y)
+        g = torch.ops.trec_amd.relu_bwd_mask(dy, y)
+        assert torch.equal(g, dy * (y > 0))
+
+    def test_fused_linear_lt_path_matches(self):
+        import os
+
+        from torchrec_amd.modules.mlp import _LinearReLUFused
+
+        os.environ["TREC_LT_MLP"] = "1"
+        try:
+            torch.manual_seed(0)
+            x = torch.randn(2048, 64, device="cuda", dtype=torch.bfloat16,
+                            requires_grad=True)
+            w = torch.randn(96, 64, device="cuda", dtype=torch.bfloat16,
+                            requires_grad=True)
+            b = torch.randn(96, device="cuda", dtype=torch.bfloat16,
+                            requires_grad=True)
+            y = _LinearReLUFused.apply(x, w, b)
+            gout = torch.randn_like(y)
+            y.backward(gout)
+            xr = x.detach().float().requires_grad_(True)
+            wr = w.detach().float().requires_grad_(True)
+            br = b.detach().float().requires_grad_(True)
+            yr = torch.relu(torch.nn.functional.linear(xr, wr, br))
+            yr.backward(gout.float())
+            torch.testing.assert_close(y.float(), yr, atol=5e-2, rtol=5e-2)
+            torch.testing.assert_close(x.grad.float(), xr.grad, atol=5e-2, rtol=5e-2)
+            torch.testing.assert_close(w.grad.float(), wr.grad, atol=1.0, rtol=5e-2)
+            torch.testing.assert_close(b.grad.float(), br.grad, atol=1.0, rtol=5e-2)
+        finally:
+            os.environ.pop("TREC_LT_MLP", None)

@@ -113,6 +113,11 @@ at::Tensor tbe_forward_seq_int8(const at::Tensor& qweights,
 // interaction.hip
 at::Tensor col_sum(const at::Tensor& input);
 // mlp_ops.hip
+at::Tensor lt_linear_relu_fwd(const at::Tensor& x, const at::Tensor& w,
+                              const at::Tensor& b);
+std::tuple<at::Tensor, at::Tensor> lt_wgrad_bgrad(const at::Tensor& g,
+                                                  const at::Tensor& x);
+at::Tensor relu_bwd_mask(const at::Tensor& grad_out, const at::Tensor& y);
 at::Tensor bce_with_logits_fwd(const at::Tensor& logits, const at::Tensor& labels);
 at::Tensor bce_with_logits_bwd(const at::Tensor& logits, const at::Tensor& labels,
                                const at::Tensor& grad_out);
@@ -222,6 +227,9 @@ TORCH_LIBRARY(trec_amd, m) {
   m.def("col_sum(Tensor input) -> Tensor");
   m.def("interaction_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
   m.def("relu_bwd_col_sum(Tensor grad_out, Tensor y) -> (Tensor, Tensor)");
+  m.def("lt_linear_relu_fwd(Tensor x, Tensor w, Tensor b) -> Tensor");
+  m.def("lt_wgrad_bgrad(Tensor g, Tensor x) -> (Tensor, Tensor)");
+  m.def("relu_bwd_mask(Tensor grad_out, Tensor y) -> Tensor");
   m.def("bce_with_logits_fwd(Tensor logits, Tensor labels) -> Tensor");
   m.def("bce_with_logits_bwd(Tensor logits, Tensor labels, Tensor grad_out) -> Tensor");
   m.def("interaction_mfma_forward(Tensor dense, Tensor sparse, Tensor pi, Tensor pj) -> Tensor");
@@ -266,6 +274,9 @@ TORCH_LIBRARY_IMPL(trec_amd, CUDA, m) {
   m.impl("col_sum", trec_amd::col_sum);
   m.impl("interaction_forward", trec_amd::interaction_forward);
   m.impl("relu_bwd_col_sum", trec_amd::relu_bwd_col_sum);
+  m.impl("lt_linear_relu_fwd", trec_amd::lt_linear_relu_fwd);
+  m.impl("lt_wgrad_bgrad", trec_amd::lt_wgrad_bgrad);
+  m.impl("relu_bwd_mask", trec_amd::relu_bwd_mask);
   m.impl("bce_with_logits_fwd", trec_amd::bce_with_logits_fwd);
   m.impl("bce_with_logits_bwd", trec_amd::bce_with_logits_bwd);
   m.impl("interaction_mfma_forward", trec_amd::interaction_mfma_forward);

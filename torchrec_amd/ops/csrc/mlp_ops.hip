@@ -176,6 +176,61 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
   return {g, db};
 }
 
+// relu mask only: g = dy * (y > 0) — one vectorized pass; the bias grad
+// comes out of the wgrad GEMM's BGRADB epilogue (lt_gemm.hip), so the
+// column-sum machinery is skipped entirely on that path.
+template <typename scalar_t, int VPT>
+__global__ void __launch_bounds__(kBlockThreads) relu_bwd_mask_kernel(
+    const scalar_t* __restrict__ dy, const scalar_t* __restrict__ y, int64_t n,
+    scalar_t* __restrict__ g) {
+  const int64_t chunks = (n + VPT - 1) / VPT;
+  for (int64_t it = static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+       it < chunks; it += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    const int64_t base = it * VPT;
+    if (base + VPT <= n) {
+      uint4 qy = *reinterpret_cast<const uint4*>(y + base);
+      uint4 qd = *reinterpret_cast<const uint4*>(dy + base);
+      const scalar_t* py = reinterpret_cast<const scalar_t*>(&qy);
+      const scalar_t* pd = reinterpret_cast<const scalar_t*>(&qd);
+      uint4 qo;
+      scalar_t* po = reinterpret_cast<scalar_t*>(&qo);
+#pragma unroll
+      for (int v = 0; v < VPT; ++v)
+        po[v] = (emb2float(py[v]) > 0.f) ? pd[v] : float2emb(0.f, scalar_t{});
+      *reinterpret_cast<uint4*>(g + base) = qo;
+    } else {
+      for (int64_t i = base; i < n; ++i)
+        g[i] = (emb2float(y[i]) > 0.f) ? dy[i] : float2emb(0.f, scalar_t{});
+    }
+  }
+}
+
+at::Tensor relu_bwd_mask(const at::Tensor& grad_out, const at::Tensor& y) {
+  TORCH_CHECK(grad_out.is_cuda() && grad_out.sizes() == y.sizes());
+  auto dy = grad_out.contiguous();
+  auto yc = y.contiguous();
+  auto g = at::empty_like(dy);
+  int64_t n = dy.numel();
+  if (n == 0) return g;
+  auto stream = mlp_stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kHalf, at::kBFloat16, dy.scalar_type(),
+                                  "relu_bwd_mask", [&] {
+    if constexpr (std::is_same_v<scalar_t, double>) {
+      TORCH_CHECK(false, "fp64 unsupported");
+    } else {
+      using dev_t = typename DevType<scalar_t>::type;
+      constexpr int VPT = sizeof(dev_t) == 2 ? 8 : 4;
+      int grid = grid_for((n + VPT - 1) / VPT, kBlockThreads);
+      hipLaunchKernelGGL((relu_bwd_mask_kernel<dev_t, VPT>), dim3(grid),
+                         dim3(kBlockThreads), 0, stream,
+                         reinterpret_cast<const dev_t*>(dy.data_ptr<scalar_t>()),
+                         reinterpret_cast<const dev_t*>(yc.data_ptr<scalar_t>()), n,
+                         reinterpret_cast<dev_t*>(g.data_ptr<scalar_t>()));
+    }
+  });
+  return g;
+}
+
 // ---------------------------------------------------------------------------
 // fused BCE-with-logits (mean reduction). torch's BCEWithLogitsLoss expands
 // to ~12 launch-floor kernels per step inside the captured graph
