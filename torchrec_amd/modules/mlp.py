@@ -61,10 +61,18 @@ def _splitk_wgrad(g: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
 
 
 def _use_lt_epilogues() -> bool:
-    """hipBLASLt epilogue fusion: relu+bias ride the forward GEMM and the
-    bias grad rides the wgrad GEMM (BGRADB), skipping the separate relu
-    pass and the whole column-sum chain."""
+    """hipBLASLt BGRADB wgrad path: measured SLOWER than split-K bmm +
+    colsum (the epilogue constraint excludes split-K algos for the deep-K
+    wgrad shapes) — off unless explicitly requested."""
     return os.environ.get("TREC_LT_MLP", "0") == "1"
+
+
+def _use_lt_fwd() -> bool:
+    """Forward-only hipBLASLt RELU_BIAS epilogue (relu rides the GEMM)."""
+    return (
+        os.environ.get("TREC_LT_MLP_FWD", "0") == "1"
+        or os.environ.get("TREC_LT_MLP", "0") == "1"
+    )
 
 
 class _LinearReLUFused(torch.autograd.Function):
@@ -73,7 +81,7 @@ class _LinearReLUFused(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b):  # type: ignore[override]
-        if _use_lt_epilogues():
+        if _use_lt_fwd():
             from torchrec_amd import ops
 
             ops.hip_ops()
