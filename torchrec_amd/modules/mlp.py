@@ -99,7 +99,7 @@ class _LinearReLUFused(torch.autograd.Function):
         if _use_lt_epilogues():
             ops.hip_ops()
             g = torch.ops.trec_amd.relu_bwd_mask(dy.contiguous(), y)
-            dx = g @ w
+            dx = g @ w if ctx.needs_input_grad[0] else None
             dw, db = torch.ops.trec_amd.lt_wgrad_bgrad(g, x)
             return dx, dw, db
         # relu_bwd_col_sum kernel: opt-in — the scalar-load version measured
@@ -112,7 +112,8 @@ class _LinearReLUFused(torch.autograd.Function):
         else:
             g = dy * (y > 0)
             db = g.sum(0)
-        dx = g @ w
+        # the first dense-arch layer's input is loader data: skip its igrad
+        dx = g @ w if ctx.needs_input_grad[0] else None
         dw = _splitk_wgrad(g, x)
         return dx, dw, db
 
