@@ -166,7 +166,8 @@ at::Tensor lt_linear_relu_fwd(const at::Tensor& x, const at::Tensor& w,
   hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_RELU_BIAS;
   LT_CHECK(hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_EPILOGUE,
                                            &epi, sizeof(epi)));
-  const void* bias_ptr = b.contiguous().data_ptr();
+  auto bc = b.contiguous();  // keep alive through the (possibly async) matmul
+  const void* bias_ptr = bc.data_ptr();
   LT_CHECK(hipblasLtMatmulDescSetAttribute(
       desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias_ptr, sizeof(bias_ptr)));
   int32_t bias_dt = (int32_t)dt;
