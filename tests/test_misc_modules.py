@@ -56,11 +56,6 @@ def test_batch_is_pipelineable():
     assert torch.equal(b2.dense_features, b.dense_features)
     # record_stream on CPU tensors is a no-op but must not raise on the
     # contract surface (GPU pipelines call it on every batch)
-    class _FakeStream:
-        pass
-
-    try:
-        b.record_stream(torch.cuda.current_stream() if torch.cuda.is_available()
-                        else _FakeStream())
-    except AttributeError:
-        pass  # CPU tensors have no record_stream; contract forwards only
+    assert callable(getattr(b, "record_stream"))
+    if torch.cuda.is_available():  # pragma: no cover — GPU-only contract call
+        b.record_stream(torch.cuda.current_stream())
