@@ -601,6 +601,12 @@ def run_dist_graph_bench(
     )
     _dense_to_bf16(dmp)
     dense_params = [p for p in dmp.parameters() if p.requires_grad]
+    if world > 1:
+        # no DDP wrap in this mode (the grad all-reduce lives INSIDE the
+        # captured step), so replicate rank 0's dense init explicitly — DDP
+        # normally does this broadcast at construction
+        for p in dense_params:
+            dist.broadcast(p.data, src=0)
     dense_opt = torch.optim.SGD(dense_params, lr=LR, foreach=True)
     if rank == 0:
         counts = {}
