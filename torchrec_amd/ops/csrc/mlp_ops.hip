@@ -10,6 +10,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
+#include <cstdlib>
 #include <type_traits>
 
 #include "common.h"
@@ -90,6 +91,69 @@ __global__ void __launch_bounds__(kBlockThreads) relu_bwd_colsum_partial_kernel(
   }
 }
 
+// one-kernel segment+final fold, take 2 (first try rejected at 21.8 us for
+// per-element ORDERED atomics): phase 1 writes per-segment sums with plain
+// float4 stores; publication is one device fence + a RELAXED counter; the
+// last block per column tile re-fences and folds the 32 segment rows with
+// PLAIN float4 loads (independent, pipelined). Fold order is fixed =>
+// deterministic.
+template <typename o_t>
+__global__ void __launch_bounds__(kBlockThreads) colsum_finish2_kernel(
+    const float* __restrict__ partial, int G, int seg_rows, int64_t N,
+    float* __restrict__ seg /* [S, N] */, int* __restrict__ counters,
+    int S, o_t* __restrict__ db) {
+  // each thread owns 4 consecutive columns (float4 lanes)
+  int64_t c4 = (static_cast<int64_t>(blockIdx.x) * kBlockThreads + threadIdx.x) * 4;
+  int sgm = blockIdx.y;
+  if (c4 + 4 <= N) {
+    int g0 = sgm * seg_rows;
+    int g1 = min(G, g0 + seg_rows);
+    float4 acc = make_float4(0.f, 0.f, 0.f, 0.f);
+    for (int grp = g0; grp < g1; ++grp) {
+      float4 v = *reinterpret_cast<const float4*>(
+          partial + static_cast<int64_t>(grp) * N + c4);
+      acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+    }
+    *reinterpret_cast<float4*>(seg + static_cast<int64_t>(sgm) * N + c4) = acc;
+  } else {
+    for (int64_t c = c4; c < N; ++c) {
+      int g0 = sgm * seg_rows, g1 = min(G, g0 + seg_rows);
+      float a = 0.f;
+      for (int grp = g0; grp < g1; ++grp)
+        a += partial[static_cast<int64_t>(grp) * N + c];
+      seg[static_cast<int64_t>(sgm) * N + c] = a;
+    }
+  }
+  __threadfence();
+  __syncthreads();
+  __shared__ int last;
+  if (threadIdx.x == 0)
+    last = (__hip_atomic_fetch_add(&counters[blockIdx.x], 1, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT) == S - 1);
+  __syncthreads();
+  if (!last) return;
+  __threadfence();
+  if (c4 + 4 <= N) {
+    float4 tot = make_float4(0.f, 0.f, 0.f, 0.f);
+    for (int ss = 0; ss < S; ++ss) {
+      float4 v = *reinterpret_cast<const float4*>(
+          seg + static_cast<int64_t>(ss) * N + c4);
+      tot.x += v.x; tot.y += v.y; tot.z += v.z; tot.w += v.w;
+    }
+    db[c4 + 0] = float2emb(tot.x, o_t{});
+    db[c4 + 1] = float2emb(tot.y, o_t{});
+    db[c4 + 2] = float2emb(tot.z, o_t{});
+    db[c4 + 3] = float2emb(tot.w, o_t{});
+  } else {
+    for (int64_t c = c4; c < N; ++c) {
+      float tot = 0.f;
+      for (int ss = 0; ss < S; ++ss)
+        tot += seg[static_cast<int64_t>(ss) * N + c];
+      db[c] = float2emb(tot, o_t{});
+    }
+  }
+}
+
 // level-1 segment reduce: 2-D grid ((column tile) x S) collapses G rows to
 // S per-tile sums. Measured better than a single-kernel semaphore finish
 // (21.8 us) as a pair with the cast-fused final (9 + 8.5 us): the last-block
@@ -149,14 +213,36 @@ std::tuple<at::Tensor, at::Tensor> relu_bwd_col_sum(const at::Tensor& grad_out,
       auto partial = at::empty({(int64_t)G * N}, dy.options().dtype(at::kFloat));
       int ftiles = (int)((N + kBlockThreads - 1) / kBlockThreads);
       int grid = grid_for(col_chunks * G, kBlockThreads);
+      static const bool use_finish2 = [] {
+        const char* e = std::getenv("TREC_COLSUM_FINISH2");
+        return !e || e[0] != '0';
+      }();
+      int tiles4 = (int)((N / 4 + kBlockThreads - 1) / kBlockThreads);
+      if (tiles4 < 1) tiles4 = 1;
+      at::Tensor counters;
+      int* cnt_ptr = nullptr;
+      if (G > 64 && use_finish2) {
+        // zeroed by block 0 of the partial kernel (same stream => ordered)
+        counters = at::empty({tiles4}, dy.options().dtype(at::kInt));
+        cnt_ptr = counters.data_ptr<int>();
+      }
       hipLaunchKernelGGL((relu_bwd_colsum_partial_kernel<dev_t, VPT>), dim3(grid),
                          dim3(kBlockThreads), 0, stream,
                          reinterpret_cast<const dev_t*>(dy.data_ptr<scalar_t>()),
                          reinterpret_cast<const dev_t*>(yc.data_ptr<scalar_t>()), M, N,
                          rows_per_group, G,
                          reinterpret_cast<dev_t*>(g.data_ptr<scalar_t>()),
-                         partial.data_ptr<float>(), nullptr, 0);
-      if (G > 64) {
+                         partial.data_ptr<float>(), cnt_ptr, tiles4);
+      if (G > 64 && use_finish2) {
+        constexpr int S = 32;
+        int seg_rows = (G + S - 1) / S;
+        auto seg = at::empty({(int64_t)S * N}, dy.options().dtype(at::kFloat));
+        hipLaunchKernelGGL((colsum_finish2_kernel<dev_t>), dim3(tiles4, S),
+                           dim3(kBlockThreads), 0, stream, partial.data_ptr<float>(),
+                           G, seg_rows, N, seg.data_ptr<float>(),
+                           cnt_ptr, S,
+                           reinterpret_cast<dev_t*>(db.data_ptr<scalar_t>()));
+      } else if (G > 64) {
         constexpr int S = 32;
         int seg_rows = (G + S - 1) / S;
         auto seg = at::empty({(int64_t)S * N}, dy.options().dtype(at::kFloat));
