@@ -406,6 +406,14 @@ def run_graph_bench(steps: int, warmup: int, batch_size: int, scale: float) -> N
         topology=Topology(world_size=1, compute_device="cuda", batch_size=batch_size)
     )
     plan = planner.plan(model, [sharder])
+    if any(
+        "uvm" in ps.compute_kernel
+        for mplan in plan.plan.values()
+        for ps in mplan.values()
+    ):
+        # UVM prefetch copies are not stream-capturable; fail BEFORE the
+        # table weights materialize so the eager fallback starts clean
+        raise RuntimeError("plan uses UVM kernels: hipGraph capture unsupported")
     dmp = DistributedModelParallel(
         model, env=ShardingEnv.from_local(1, 0), plan=plan, sharders=[sharder], device=device
     )
@@ -849,27 +857,42 @@ if __name__ == "__main__":
         and args.qcomm == "none"
         and not args.smoke
     )
+    def _release_and_fallback(msg: str) -> None:
+        # drop the failed attempt's references OUTSIDE the except block (a
+        # live traceback pins the frame and with it up to ~250 GB of table
+        # weights, which OOMed the fallback's own build)
+        import gc
+
+        from torchrec_amd.distributed.dist_data import set_static_kjt_splits
+
+        set_static_kjt_splits(False)
+        print(msg, flush=True)
+        gc.collect()
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+        run_bench(args.gpus, args.steps, args.warmup, args.batch_size,
+                  args.scale, qcomm=args.qcomm)
+
     if args.smoke:
         run_smoke()
     elif want_graph and int(os.environ.get("WORLD_SIZE", "1")) == 1:
+        failed = None
         try:
             run_graph_bench(args.steps, args.warmup, args.batch_size, args.scale)
         except Exception as exc:  # pragma: no cover — capture fallback
-            print(f"# hipGraph capture failed ({exc!r}); eager pipeline fallback",
-                  flush=True)
-            run_bench(args.gpus, args.steps, args.warmup, args.batch_size,
-                      args.scale, qcomm=args.qcomm)
+            failed = repr(exc)
+        if failed is not None:
+            _release_and_fallback(
+                f"# hipGraph capture failed ({failed}); eager pipeline fallback")
     elif want_dist_graph:
+        failed = None
         try:
             run_dist_graph_bench(args.steps, args.warmup, args.batch_size, args.scale)
         except Exception as exc:  # pragma: no cover — capture fallback
-            from torchrec_amd.distributed.dist_data import set_static_kjt_splits
-
-            set_static_kjt_splits(False)
-            print(f"# dist hipGraph capture failed ({exc!r}); eager pipeline fallback",
-                  flush=True)
-            run_bench(args.gpus, args.steps, args.warmup, args.batch_size,
-                      args.scale, qcomm=args.qcomm)
+            failed = repr(exc)
+        if failed is not None:
+            _release_and_fallback(
+                f"# dist hipGraph capture failed ({failed}); eager pipeline fallback")
     else:
         run_bench(args.gpus, args.steps, args.warmup, args.batch_size, args.scale,
                   qcomm=args.qcomm)
