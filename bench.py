@@ -603,6 +603,12 @@ def run_dist_graph_bench(
         topology=Topology(world_size=world, compute_device="cuda", batch_size=batch_size)
     )
     plan = planner.collective_plan(model, [sharder], dist.group.WORLD)
+    if any(
+        "uvm" in ps.compute_kernel
+        for mplan in plan.plan.values()
+        for ps in mplan.values()
+    ):
+        raise RuntimeError("plan uses UVM kernels: hipGraph capture unsupported")
     dmp = DistributedModelParallel(
         model, env=env, plan=plan, sharders=[sharder], device=device,
         init_data_parallel=False,  # dense grads all-reduce INSIDE the graph
